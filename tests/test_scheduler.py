@@ -1,0 +1,129 @@
+"""Scheduler logic tests against the FakeBackend (no processes, no GPUs).
+
+Covers the reference's behaviours: jobs normalization
+(tfmesos/__init__.py:9-16), first-fit offer matching
+(tfmesos/scheduler.py:252-266), revive-up-to-3 pre-start
+(:181,404-434), fail-fast post-start (:394-401), finished() semantics
+(:474-477).
+"""
+
+import pytest
+
+from tfmesos_amd import Job
+from tfmesos_amd.backend import (
+    FakeBackend, TASK_FAILED, TASK_FINISHED, TASK_RUNNING)
+from tfmesos_amd.scheduler import LocalScheduler, MAX_FAILURE_COUNT
+
+
+def make_sched(jobs, **kw):
+    be = FakeBackend(**kw.pop("resources", {}))
+    s = LocalScheduler(jobs, backend=be, quiet=True)
+    s.addr = "127.0.0.1:0"  # normally set by start()
+    be.listener = s
+    return s, be
+
+
+def test_job_normalization_forms():
+    from tfmesos_amd import cluster  # noqa: F401  (import works)
+    j = Job(name="w", num=3, start=1)
+    s, be = make_sched([j])
+    assert len(s.tasks) == 2  # indices 1,2 (start..num)
+    assert [t.task_index for t in s.tasks] == [1, 2]
+
+
+def test_first_fit_matching_with_gpus():
+    jobs = [Job(name="ps", num=1, cpus=1, mem=1024, gpus=1),
+            Job(name="worker", num=2, cpus=1, mem=1024, gpus=2)]
+    s, be = make_sched(jobs, resources={"cpus": 8.0, "mem": 65536.0, "gpus": 8})
+    be.offer()
+    assert len(be.launched) == 3
+    used_gpus = [spec.gpu_ids for spec in be.launched]
+    # distinct device ids, right counts
+    assert [len(g) for g in used_gpus] == [1, 2, 2]
+    flat = [g for gs in used_gpus for g in gs]
+    assert len(set(flat)) == 5
+
+
+def test_first_fit_insufficient_gpus_defers():
+    jobs = [Job(name="w", num=2, gpus=2)]
+    s, be = make_sched(jobs, resources={"gpus": 3})
+    be.offer()
+    assert len(be.launched) == 1  # only one task fits
+    assert not s.tasks[1].offered
+
+
+def test_oversubscribed_cpu_defers():
+    jobs = [Job(name="w", num=4, cpus=4.0)]
+    s, be = make_sched(jobs, resources={"cpus": 8.0})
+    be.offer()
+    assert sum(1 for t in s.tasks if t.offered) == 2
+
+
+def test_revive_pre_start_then_fatal():
+    jobs = [Job(name="w", num=1)]
+    s, be = make_sched(jobs)
+    be.offer()
+    first_id = s.tasks[0].task_id
+    be.send_update(first_id, TASK_FAILED, "boom")
+    # revived with a fresh id, not offered yet
+    assert s.tasks[0].task_id != first_id
+    assert not s.tasks[0].offered
+    be.offer()
+    assert s.tasks[0].offered
+    # two more failures -> fatal
+    be.send_update(s.tasks[0].task_id, TASK_FAILED, "boom2")
+    be.offer()
+    be.send_update(s.tasks[0].task_id, TASK_FAILED, "boom3")
+    assert s._error is not None
+    with pytest.raises(RuntimeError):
+        s.finished()
+    assert s.task_failure_count[("w", 0)] == MAX_FAILURE_COUNT
+
+
+def test_post_start_failure_is_fatal():
+    jobs = [Job(name="w", num=1)]
+    s, be = make_sched(jobs)
+    be.offer()
+    s.started = True
+    be.send_update(s.tasks[0].task_id, TASK_FAILED, "crash")
+    with pytest.raises(RuntimeError):
+        s.finished()
+
+
+def test_finished_when_any_job_done():
+    jobs = [Job(name="ps", num=1), Job(name="worker", num=2)]
+    s, be = make_sched(jobs)
+    be.offer()
+    s.started = True
+    workers = [t for t in s.tasks if t.job_name == "worker"]
+    be.send_update(workers[0].task_id, TASK_FINISHED)
+    assert not s.finished()
+    be.send_update(workers[1].task_id, TASK_FINISHED)
+    assert s.finished()  # worker job complete; ps still "running"
+
+
+def test_running_update_not_terminal():
+    jobs = [Job(name="w", num=1)]
+    s, be = make_sched(jobs)
+    be.offer()
+    be.send_update(s.tasks[0].task_id, TASK_RUNNING)
+    assert s._error is None
+    assert s.tasks[0].offered
+
+
+def test_rank_assignment_stable():
+    jobs = [Job(name="ps", num=2), Job(name="worker", num=3)]
+    s, be = make_sched(jobs)
+    assert s._rank_of[("ps", 0)] == 0
+    assert s._rank_of[("ps", 1)] == 1
+    assert s._rank_of[("worker", 0)] == 2
+    assert s._rank_of[("worker", 2)] == 4
+
+
+def test_suppress_when_all_placed():
+    jobs = [Job(name="w", num=1)]
+    s, be = make_sched(jobs)
+    be.offer()
+    assert len(be.launched) == 1
+    be.offer()  # everything placed -> suppress
+    assert be.suppressed

@@ -1,0 +1,98 @@
+"""The mnist_replica MLP: 784 -> hidden(100) relu -> 10 softmax-xent.
+
+Same architecture/hyperparameters as the reference benchmark workload
+(``examples/mnist/mnist_replica.py:122-145``: truncated-normal init with
+stddev 1/sqrt(784), hidden 100, batch 100, lr 0.01). Forward AND backward
+are written explicitly against ``tfmesos_amd.ops`` (bf16 MFMA GEMM +
+fused softmax-xent + fused relu-bwd on GPU) — no autograd in the hot
+path, so the step is a short fixed kernel sequence that hipGraph can
+capture.
+"""
+
+import math
+
+import torch
+
+from tfmesos_amd import ops
+
+
+class MnistMLP(object):
+
+    def __init__(self, hidden_units=100, image_pixels=784, classes=10,
+                 seed=1234):
+        self.hidden = hidden_units
+        self.inputs = image_pixels
+        self.classes = classes
+        self.seed = seed
+
+    def param_specs(self):
+        return [
+            ("hid_w", (self.inputs, self.hidden)),
+            ("hid_b", (self.hidden,)),
+            ("sm_w", (self.hidden, self.classes)),
+            ("sm_b", (self.classes,)),
+        ]
+
+    def init_params(self):
+        """Truncated-normal(stddev=1/sqrt(inputs)) weights, zero biases —
+        the reference's init (mnist_replica.py:122-137)."""
+        g = torch.Generator().manual_seed(self.seed)
+        std1 = 1.0 / math.sqrt(self.inputs)
+        std2 = 1.0 / math.sqrt(self.hidden)
+
+        def trunc(shape, std):
+            t = torch.empty(*shape)
+            torch.nn.init.trunc_normal_(t, std=std, a=-2 * std, b=2 * std,
+                                        generator=g)
+            return t
+
+        return [
+            ("hid_w", trunc((self.inputs, self.hidden), std1)),
+            ("hid_b", torch.zeros(self.hidden)),
+            ("sm_w", trunc((self.hidden, self.classes), std2)),
+            ("sm_b", torch.zeros(self.classes)),
+        ]
+
+    def fwd_bwd(self, p, x, y, g):
+        """One replica fwd+bwd.
+
+        p: callable name -> bf16 (GPU) / fp32 (CPU) param view
+        x: [B, 784] activations dtype; y: [B] int64 labels
+        g: callable name -> grad view to fill (fp32)
+        Returns mean loss (fp32 scalar tensor).
+        """
+        B = x.shape[0]
+        hid_w, hid_b = p("hid_w"), p("hid_b")
+        sm_w, sm_b = p("sm_w"), p("sm_b")
+
+        # forward
+        h = ops.gemm_bias_act(x, hid_w, hid_b, act="relu")       # [B,H]
+        logits = ops.gemm_bias_act(h, sm_w, sm_b, act="none")    # [B,C]
+        loss, probs = ops.softmax_xent_fwd(logits, y)
+
+        # backward
+        dlogits = ops.softmax_xent_bwd(probs, y)                  # [B,C] (/B)
+        # dW2 = h^T @ dlogits ; db2 = colsum(dlogits)
+        g("sm_w").copy_(ops.gemm_bias_act(h, dlogits, trans_a=True).float())
+        g("sm_b").copy_(dlogits.float().sum(0))
+        dh = ops.gemm_bias_act(dlogits, sm_w, trans_b=True)       # [B,H]
+        dh = ops.relu_bwd(dh, h)
+        g("hid_w").copy_(ops.gemm_bias_act(x, dh, trans_a=True).float())
+        g("hid_b").copy_(dh.float().sum(0))
+        return loss
+
+    def loss_only(self, p, x, y):
+        h = ops.gemm_bias_act(x, p("hid_w"), p("hid_b"), act="relu")
+        logits = ops.gemm_bias_act(h, p("sm_w"), p("sm_b"))
+        loss, _ = ops.softmax_xent_fwd(logits, y)
+        return loss
+
+
+def synthetic_batch(batch_size=100, image_pixels=784, classes=10,
+                    device="cpu", dtype=torch.float32, seed=0):
+    """Synthetic MNIST-shaped data (no network for datasets; BASELINE.md
+    prescribes synthetic data / random-init weights)."""
+    gen = torch.Generator().manual_seed(seed)
+    x = torch.rand(batch_size, image_pixels, generator=gen)
+    y = torch.randint(0, classes, (batch_size,), generator=gen)
+    return x.to(device=device, dtype=dtype), y.to(device)
