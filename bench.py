@@ -1,0 +1,162 @@
+#!/usr/bin/env python
+"""Flagship benchmark: mnist_replica 1-ps/N-worker sync SGD, bf16.
+
+Measures the BASELINE.json metric — global steps/sec of the
+mnist_replica-equivalent workload (784-100-10 MLP, batch 100/worker,
+sync gradient aggregation at the PS, SGD lr=0.01) on N MI355X GPUs:
+world==1 runs ps+worker colocated on one GPU; world>1 puts the PS on
+rank 0's GPU and workers on ranks 1..N-1, with grad push = RCCL reduce
+and param pull = RCCL broadcast over xGMI.
+
+Run directly (N=1) or under torch.distributed.run with one rank per GPU:
+  python bench.py --gpus 1 --steps 1000 --warmup 100
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
+      --master-addr 127.0.0.1 bench.py --gpus 8 --steps 1000 --warmup 100
+
+Rank 0 prints ONE JSON line with the whole-job metric.
+
+Extra workloads (same PS machinery): --workload nmf (sparse-embedding
+2-ps-shard NMF config), --workload inception (conv net, bf16).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def log(msg):
+    sys.stderr.write(msg + "\n")
+    sys.stderr.flush()
+
+
+def run_mnist(args, device, rank, world):
+    from tfmesos_amd.models.mlp import MnistMLP, synthetic_batch
+    from tfmesos_amd.ps.replica import SyncReplicaTrainer
+
+    model = MnistMLP(hidden_units=args.hidden)
+    trainer = SyncReplicaTrainer(
+        model.init_params(), optimizer=args.optimizer,
+        hparams={"lr": args.lr}, device=device)
+    roles = trainer.roles
+
+    act_dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+    x, y = synthetic_batch(args.batch, device=device, dtype=act_dtype,
+                           seed=1000 + rank)
+
+    pview = (lambda n: trainer.store.view(n, bf16=True)) \
+        if act_dtype == torch.bfloat16 else (lambda n: trainer.store.view(n))
+
+    def one_step():
+        if roles.is_worker:
+            model.fwd_bwd(pview, x, y, trainer.grad_view)
+        trainer.step()
+
+    return one_step, {
+        "model": "mnist_replica_mlp_784x%dx10" % args.hidden,
+        "global_batch": args.batch * roles.n_workers,
+        "parallelism": roles.describe(),
+        "optimizer": args.optimizer,
+        "lr": args.lr,
+        "sync": True,
+    }
+
+
+def run_nmf(args, device, rank, world):
+    """matrix_factorization config: rank-200 NMF with the factor tables
+    treated as embeddings (sparse push/pull path)."""
+    from tfmesos_amd.models.nmf import NMFWorkload
+    wl = NMFWorkload(n=args.nmf_n, rank=args.nmf_rank, device=device,
+                     lr=args.lr, seed=1234 + rank)
+    return wl.one_step, {
+        "model": "nmf_%dx%d_rank%d" % (args.nmf_n, args.nmf_n, args.nmf_rank),
+        "global_batch": args.nmf_n,
+        "parallelism": "local" if world == 1 else "ps%d+w%d" % (2, world - 2),
+        "optimizer": "sgd",
+        "lr": args.lr,
+        "sync": True,
+    }
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=1000)
+    p.add_argument("--warmup", type=int, default=100)
+    p.add_argument("--workload", default="mnist",
+                   choices=["mnist", "nmf"])
+    p.add_argument("--batch", type=int, default=100)
+    p.add_argument("--hidden", type=int, default=100)
+    p.add_argument("--optimizer", default="sgd")
+    p.add_argument("--lr", type=float, default=0.01)
+    p.add_argument("--nmf-n", type=int, default=1000)
+    p.add_argument("--nmf-rank", type=int, default=200)
+    args = p.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+        log("WARNING: no GPU visible; CPU dev run (not a valid benchmark)")
+
+    one_step, config = {"mnist": run_mnist, "nmf": run_nmf}[args.workload](
+        args, device, rank, world)
+
+    def barrier_sync():
+        if world > 1:
+            dist.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        one_step()
+    barrier_sync()
+
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks = whole-job time
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if dist.get_backend() == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        steps_per_sec = args.steps / elapsed
+        out = {
+            "metric": "global steps/sec, mnist_replica 1-ps/N-worker at "
+                      "1/2/4/8 MI355X" if args.workload == "mnist"
+                      else "global steps/sec, %s" % args.workload,
+            "value": steps_per_sec,
+            "unit": "steps/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed * 1000.0 / args.steps,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if device.type == "cuda" else "fp32-cpu-devrun",
+            "data": "synthetic",
+            "config": config,
+        }
+        print(json.dumps(out))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,54 @@
+"""Helper process for distributed replica tests (gloo, CPU).
+
+Usage: RANK/WORLD_SIZE/MASTER_ADDR/MASTER_PORT in env;
+argv: <mode sync|async> <steps> <out_path_prefix>
+Rank 0 (ps) saves final master params to <prefix>.pt after training.
+"""
+
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from tfmesos_amd.models.mlp import MnistMLP, synthetic_batch  # noqa: E402
+from tfmesos_amd.ps.replica import (  # noqa: E402
+    AsyncPSServer, AsyncPSWorker, SyncReplicaTrainer)
+
+
+def main():
+    mode, steps, prefix = sys.argv[1], int(sys.argv[2]), sys.argv[3]
+    model = MnistMLP()
+    trainer = SyncReplicaTrainer(model.init_params(), optimizer="sgd",
+                                 hparams={"lr": 0.1}, device="cpu")
+    roles = trainer.roles
+    # every worker gets the same batch as the single-process reference
+    x, y = synthetic_batch(50, seed=42)
+
+    if mode == "sync":
+        for _ in range(steps):
+            if roles.is_worker:
+                model.fwd_bwd(trainer.param, x, y, trainer.grad_view)
+            trainer.step()
+    elif mode == "async":
+        from tfmesos_amd.ps.replica import make_pair_groups
+        groups = make_pair_groups(roles)
+        if roles.is_ps:
+            AsyncPSServer(trainer, groups).serve(steps)
+        else:
+            w = AsyncPSWorker(trainer, groups)
+            for _ in range(steps):
+                model.fwd_bwd(trainer.param, x, y, trainer.grad_view)
+                w.step()
+    else:
+        raise SystemExit("bad mode")
+
+    if roles.is_ps:
+        torch.save({n: trainer.store.view(n).clone()
+                    for n in trainer.store.names}, prefix + ".pt")
+        torch.save(trainer.store.global_step, prefix + ".step")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,248 @@
+"""Numerics: every HIP kernel vs a plain PyTorch fp32 reference.
+
+All tests are @gpu — they run on the MI355X box (pytest -m gpu) against
+the in-tree gfx950 extension; inputs are random and asymmetric
+(transpose-detecting, guide §5.4 rule 16).
+"""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def bf(x):
+    return x.to(DEV, torch.bfloat16)
+
+
+# ------------------------------------------------------------------ apply
+
+def test_fused_sgd_matches_reference():
+    from tfmesos_amd import ops
+    torch.manual_seed(0)
+    n = 5000
+    p_ref = torch.randn(n)
+    g = torch.randn(n)
+    p = p_ref.to(DEV)
+    mb = torch.zeros(n, device=DEV)
+    out = torch.zeros(n, dtype=torch.bfloat16, device=DEV)
+    ops.fused_sgd(p, g.to(DEV), lr=0.1, momentum=0.9, weight_decay=0.01,
+                  momentum_buf=mb, bf16_out=out, grad_scale=0.5)
+    # cpu reference
+    mb_ref = torch.zeros(n)
+    ops.fused_sgd(p_ref, g, lr=0.1, momentum=0.9, weight_decay=0.01,
+                  momentum_buf=mb_ref, bf16_out=None, grad_scale=0.5)
+    assert torch.allclose(p.cpu(), p_ref, atol=1e-5)
+    assert torch.allclose(mb.cpu(), mb_ref, atol=1e-5)
+    assert torch.allclose(out.float().cpu(), p_ref, atol=0.02)
+
+
+def test_fused_sgd_bf16_grad():
+    from tfmesos_amd import ops
+    torch.manual_seed(1)
+    n = 1000
+    p_ref = torch.randn(n)
+    g = torch.randn(n).to(torch.bfloat16)
+    p = p_ref.to(DEV)
+    ops.fused_sgd(p, g.to(DEV), lr=0.1)
+    ops.fused_sgd(p_ref, g, lr=0.1)
+    assert torch.allclose(p.cpu(), p_ref, atol=1e-5)
+
+
+def test_fused_adam_matches_reference():
+    from tfmesos_amd import ops
+    torch.manual_seed(2)
+    n = 4096
+    p_ref = torch.randn(n)
+    p = p_ref.to(DEV)
+    m = torch.zeros(n, device=DEV)
+    v = torch.zeros(n, device=DEV)
+    m_ref = torch.zeros(n)
+    v_ref = torch.zeros(n)
+    for step in range(1, 4):
+        g = torch.randn(n)
+        ops.fused_adam(p, g.to(DEV), m, v, step, lr=0.01)
+        ops.fused_adam(p_ref, g, m_ref, v_ref, step, lr=0.01)
+    assert torch.allclose(p.cpu(), p_ref, atol=1e-5)
+    assert torch.allclose(v.cpu(), v_ref, atol=1e-5)
+
+
+def test_fused_adagrad_matches_reference():
+    from tfmesos_amd import ops
+    torch.manual_seed(3)
+    n = 3000
+    p_ref = torch.randn(n)
+    p = p_ref.to(DEV)
+    acc = torch.full((n,), 0.1, device=DEV)
+    acc_ref = torch.full((n,), 0.1)
+    for _ in range(3):
+        g = torch.randn(n)
+        ops.fused_adagrad(p, g.to(DEV), acc, lr=0.05)
+        ops.fused_adagrad(p_ref, g, acc_ref, lr=0.05)
+    assert torch.allclose(p.cpu(), p_ref, atol=1e-5)
+
+
+# ------------------------------------------------------------------- gemm
+
+@pytest.mark.parametrize("ta", [False, True])
+@pytest.mark.parametrize("tb", [False, True])
+@pytest.mark.parametrize("mnk", [(100, 100, 784), (100, 10, 100),
+                                 (33, 65, 17), (128, 128, 128),
+                                 (1000, 1000, 200), (784, 100, 100)])
+def test_gemm_transposes(ta, tb, mnk):
+    from tfmesos_amd import ops
+    M, N, K = mnk
+    torch.manual_seed(hash((ta, tb, mnk)) % 2**31)
+    A = torch.randn(K, M) if ta else torch.randn(M, K)
+    B = torch.randn(N, K) if tb else torch.randn(K, N)
+    ref = (A.t() if ta else A).float() @ (B.t() if tb else B).float()
+    out = ops.gemm_bias_act(bf(A), bf(B), trans_a=ta, trans_b=tb)
+    err = (out.float().cpu() - ref).abs().max()
+    tol = 0.02 * ref.abs().max() + 0.05
+    assert err < tol, (err, tol)
+
+
+def test_gemm_bias_relu():
+    from tfmesos_amd import ops
+    torch.manual_seed(10)
+    A, B = torch.randn(100, 784), torch.randn(784, 100)
+    bias = torch.randn(100)
+    ref = torch.relu(A.float() @ B.float() + bias)
+    out = ops.gemm_bias_act(bf(A), bf(B), bias.to(DEV), act="relu")
+    err = (out.float().cpu() - ref).abs().max()
+    assert err < 0.02 * ref.abs().max() + 0.05
+
+
+def test_gemm_out_fp32():
+    from tfmesos_amd import ops
+    torch.manual_seed(11)
+    A, B = torch.randn(64, 32), torch.randn(32, 48)
+    ref = A.float() @ B.float()
+    out = torch.zeros(64, 48, device=DEV)
+    ops.gemm_bias_act(bf(A), bf(B), out=out)
+    assert (out.cpu() - ref).abs().max() < 0.15
+
+
+# ----------------------------------------------------------- softmax-xent
+
+def test_softmax_xent_fwd_bwd():
+    from tfmesos_amd import ops
+    torch.manual_seed(20)
+    B, C = 100, 10
+    logits = torch.randn(B, C) * 3
+    labels = torch.randint(0, C, (B,))
+    loss, probs = ops.softmax_xent_fwd(bf(logits), labels.to(DEV))
+    ref_probs = torch.softmax(logits.float(), 1)
+    ref_loss = torch.nn.functional.cross_entropy(logits.float(), labels)
+    assert abs(float(loss) - float(ref_loss)) < 0.02
+    assert (probs.float().cpu() - ref_probs).abs().max() < 0.01
+    d = ops.softmax_xent_bwd(probs, labels.to(DEV))
+    ref_d = ref_probs.clone()
+    ref_d[torch.arange(B), labels] -= 1
+    ref_d /= B
+    assert (d.float().cpu() - ref_d).abs().max() < 1e-3
+
+
+def test_softmax_xent_wide():
+    from tfmesos_amd import ops
+    torch.manual_seed(21)
+    B, C = 32, 1000  # wider than one wave pass
+    logits = torch.randn(B, C) * 2
+    labels = torch.randint(0, C, (B,))
+    loss, probs = ops.softmax_xent_fwd(bf(logits), labels.to(DEV))
+    ref_loss = torch.nn.functional.cross_entropy(logits.float(), labels)
+    assert abs(float(loss) - float(ref_loss)) < 0.05
+
+
+# -------------------------------------------------------------- embedding
+
+def test_embedding_gather():
+    from tfmesos_amd import ops
+    torch.manual_seed(30)
+    V, D = 1000, 200
+    table = torch.randn(V, D)
+    ids = torch.randint(0, V, (77,))
+    out = ops.embedding_gather(bf(table), ids.to(DEV))
+    ref = table[ids]
+    assert (out.float().cpu() - ref).abs().max() < 0.02
+
+
+def test_embedding_scatter_add_duplicates():
+    from tfmesos_amd import ops
+    torch.manual_seed(31)
+    V, D = 50, 64
+    table = torch.zeros(V, D, device=DEV)
+    ids = torch.tensor([3, 7, 3, 3, 0])
+    rows = torch.randn(5, D)
+    ops.embedding_scatter_add(table, ids.to(DEV), rows.to(DEV))
+    ref = torch.zeros(V, D)
+    ref.index_add_(0, ids, rows)
+    assert torch.allclose(table.cpu(), ref, atol=1e-4)
+
+
+# ------------------------------------------------------------ elementwise
+
+def test_relu_bwd():
+    from tfmesos_amd import ops
+    torch.manual_seed(40)
+    dy = torch.randn(100, 100)
+    act = torch.randn(100, 100)
+    out = ops.relu_bwd(bf(dy), bf(act))
+    ref = dy.to(torch.bfloat16).float() * (act.to(torch.bfloat16).float() > 0)
+    assert (out.float().cpu() - ref).abs().max() < 1e-3
+
+
+def test_colsum():
+    from tfmesos_amd import ops
+    torch.manual_seed(41)
+    x = torch.randn(100, 110)
+    out = ops.colsum(bf(x))
+    ref = x.to(torch.bfloat16).float().sum(0)
+    assert (out.cpu() - ref).abs().max() < 0.05
+
+
+# ------------------------------------------------------------------- e2e
+
+def test_mlp_step_gpu_vs_cpu():
+    """Full fused-kernel train step on GPU tracks the fp32 CPU reference."""
+    from tfmesos_amd.models.mlp import MnistMLP, synthetic_batch
+    from tfmesos_amd.ps.store import PStore
+    model = MnistMLP()
+    params = model.init_params()
+
+    def run(device, dtype):
+        store = PStore(device=device)
+        store.init_params([(n, t.clone()) for n, t in params],
+                          optimizer="sgd", lr=0.01)
+        fg = torch.zeros_like(store.flat)
+
+        def gv(name):
+            s, c = store.offsets[name]
+            return fg[s:s + c].view(store.shapes[name])
+
+        x, y = synthetic_batch(100, device=device, dtype=dtype)
+        losses = []
+        pv = (lambda n: store.view(n, bf16=True)) if dtype == torch.bfloat16 \
+            else (lambda n: store.view(n))
+        for _ in range(10):
+            losses.append(float(model.fwd_bwd(pv, x, y, gv)))
+            store.apply_flat(fg)
+        return losses
+
+    cpu_losses = run("cpu", torch.float32)
+    gpu_losses = run(DEV, torch.bfloat16)
+    for c, g in zip(cpu_losses, gpu_losses):
+        assert abs(c - g) < 0.05 + 0.05 * abs(c), (cpu_losses, gpu_losses)
+    assert gpu_losses[-1] < gpu_losses[0]
+
+
+def test_native_ext_is_mandatory_on_gpu(monkeypatch):
+    """On GPU, ops must fail loudly if the extension is missing."""
+    import tfmesos_amd.ops as ops
+    monkeypatch.setattr(ops, "_EXT", None)
+    monkeypatch.setattr(ops, "_EXT_ERR", ImportError("simulated"))
+    x = torch.randn(4, 4, device=DEV, dtype=torch.bfloat16)
+    with pytest.raises(RuntimeError, match="extension not built"):
+        ops.gemm_bias_act(x, x)

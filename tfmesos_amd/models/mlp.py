@@ -70,15 +70,14 @@ class MnistMLP(object):
         logits = ops.gemm_bias_act(h, sm_w, sm_b, act="none")    # [B,C]
         loss, probs = ops.softmax_xent_fwd(logits, y)
 
-        # backward
+        # backward (fp32 grads written straight into the flat grad views)
         dlogits = ops.softmax_xent_bwd(probs, y)                  # [B,C] (/B)
-        # dW2 = h^T @ dlogits ; db2 = colsum(dlogits)
-        g("sm_w").copy_(ops.gemm_bias_act(h, dlogits, trans_a=True).float())
-        g("sm_b").copy_(dlogits.float().sum(0))
+        ops.gemm_bias_act(h, dlogits, trans_a=True, out=g("sm_w"))
+        ops.colsum(dlogits, out=g("sm_b"))
         dh = ops.gemm_bias_act(dlogits, sm_w, trans_b=True)       # [B,H]
         dh = ops.relu_bwd(dh, h)
-        g("hid_w").copy_(ops.gemm_bias_act(x, dh, trans_a=True).float())
-        g("hid_b").copy_(dh.float().sum(0))
+        ops.gemm_bias_act(x, dh, trans_a=True, out=g("hid_w"))
+        ops.colsum(dh, out=g("hid_b"))
         return loss
 
     def loss_only(self, p, x, y):

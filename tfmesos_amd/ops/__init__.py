@@ -47,20 +47,22 @@ def ext_available():
 
 @torch.no_grad()
 def fused_sgd(param, grad, lr, momentum=0.0, weight_decay=0.0, momentum_buf=None,
-              bf16_out=None):
+              bf16_out=None, grad_scale=1.0):
     """In-place SGD(+momentum, +wd) on an fp32 master param.
 
-    grad may be bf16 or fp32. If bf16_out is given, also writes the
-    updated param as bf16 (the broadcast copy) in the same pass.
+    grad may be bf16 or fp32 and is multiplied by grad_scale (worker-mean
+    for sync replicas). If bf16_out is given, also writes the updated
+    param as bf16 (the broadcast copy) in the same pass.
     """
     if param.is_cuda:
         _ext().fused_sgd(param, grad, momentum_buf if momentum_buf is not None
                          else torch.empty(0, device=param.device),
                          bf16_out if bf16_out is not None
                          else torch.empty(0, dtype=torch.bfloat16, device=param.device),
-                         float(lr), float(momentum), float(weight_decay))
+                         float(lr), float(momentum), float(weight_decay),
+                         float(grad_scale))
         return
-    g = grad.float()
+    g = grad.float() * grad_scale
     if weight_decay:
         g = g + weight_decay * param
     if momentum and momentum_buf is not None:
@@ -73,7 +75,8 @@ def fused_sgd(param, grad, lr, momentum=0.0, weight_decay=0.0, momentum_buf=None
 
 @torch.no_grad()
 def fused_adam(param, grad, exp_avg, exp_avg_sq, step, lr, beta1=0.9,
-               beta2=0.999, eps=1e-8, weight_decay=0.0, bf16_out=None):
+               beta2=0.999, eps=1e-8, weight_decay=0.0, bf16_out=None,
+               grad_scale=1.0):
     """In-place Adam on fp32 master param (bias-corrected, as
     tf.train.AdamOptimizer used by mnist_replica.py:147)."""
     if param.is_cuda:
@@ -81,9 +84,9 @@ def fused_adam(param, grad, exp_avg, exp_avg_sq, step, lr, beta1=0.9,
                           bf16_out if bf16_out is not None
                           else torch.empty(0, dtype=torch.bfloat16, device=param.device),
                           int(step), float(lr), float(beta1), float(beta2),
-                          float(eps), float(weight_decay))
+                          float(eps), float(weight_decay), float(grad_scale))
         return
-    g = grad.float()
+    g = grad.float() * grad_scale
     if weight_decay:
         g = g + weight_decay * param
     exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
@@ -98,14 +101,15 @@ def fused_adam(param, grad, exp_avg, exp_avg_sq, step, lr, beta1=0.9,
 
 @torch.no_grad()
 def fused_adagrad(param, grad, accum, lr, eps=1e-10, weight_decay=0.0,
-                  bf16_out=None):
+                  bf16_out=None, grad_scale=1.0):
     if param.is_cuda:
         _ext().fused_adagrad(param, grad, accum,
                              bf16_out if bf16_out is not None
                              else torch.empty(0, dtype=torch.bfloat16, device=param.device),
-                             float(lr), float(eps), float(weight_decay))
+                             float(lr), float(eps), float(weight_decay),
+                             float(grad_scale))
         return
-    g = grad.float()
+    g = grad.float() * grad_scale
     if weight_decay:
         g = g + weight_decay * param
     accum.addcmul_(g, g, value=1.0)
@@ -116,17 +120,23 @@ def fused_adagrad(param, grad, accum, lr, eps=1e-10, weight_decay=0.0,
 
 # --------------------------------------------------------------------- gemm
 
-def gemm_bias_act(a, b, bias=None, act="none", trans_a=False, trans_b=False):
-    """C = act(op(A) @ op(B) + bias), bf16 in, fp32 accumulate, bf16 out.
+def gemm_bias_act(a, b, bias=None, act="none", trans_a=False, trans_b=False,
+                  out=None):
+    """C = act(op(A) @ op(B) + bias), bf16 in, fp32 accumulate.
 
-    GPU: hand-written MFMA kernel (csrc/gemm.hip). CPU: torch reference
-    in fp32.
+    Output dtype: bf16 by default; pass ``out`` (bf16 or fp32) to write
+    in place — backward GEMMs write fp32 straight into the flat gradient
+    buffer, skipping a cast+copy. GPU: hand-written MFMA kernel
+    (csrc/gemm.hip). CPU: torch reference in fp32.
     """
+    act_code = {"none": 0, "relu": 1}[act]
     if a.is_cuda:
-        return _ext().gemm_bias_act(
-            a, b,
-            bias if bias is not None else torch.empty(0, device=a.device),
-            {"none": 0, "relu": 1}[act], bool(trans_a), bool(trans_b))
+        ebias = bias if bias is not None else torch.empty(0, device=a.device)
+        if out is not None:
+            return _ext().gemm_bias_act_out(a, b, ebias, act_code,
+                                            bool(trans_a), bool(trans_b), out)
+        return _ext().gemm_bias_act(a, b, ebias, act_code, bool(trans_a),
+                                    bool(trans_b))
     x = a.float().t() if trans_a else a.float()
     y = b.float().t() if trans_b else b.float()
     c = x @ y
@@ -134,7 +144,23 @@ def gemm_bias_act(a, b, bias=None, act="none", trans_a=False, trans_b=False):
         c = c + bias.float()
     if act == "relu":
         c = torch.relu(c)
+    if out is not None:
+        out.copy_(c.to(out.dtype))
+        return out
     return c.to(a.dtype)
+
+
+def colsum(x, out=None):
+    """out[n] = sum_m x[m,n] in fp32 (bias gradients)."""
+    if x.is_cuda:
+        return _ext().colsum(x, out if out is not None else
+                             torch.empty(0, dtype=torch.float32,
+                                         device=x.device))
+    s = x.float().sum(0)
+    if out is not None:
+        out.copy_(s)
+        return out
+    return s
 
 
 # ------------------------------------------------------------- softmax-xent

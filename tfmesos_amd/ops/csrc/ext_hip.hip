@@ -1,0 +1,273 @@
+// Python bindings for the tfmesos_amd CDNA4 kernels (torch extension).
+// Pure dispatch + shape checking; all device code lives in the .hip files.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+typedef __hip_bfloat16 bf16_t;
+
+// launchers from the .hip translation units
+void launch_sgd(float*, const void*, bool, float*, bf16_t*, long, float,
+                float, float, float, hipStream_t);
+void launch_adam(float*, const void*, bool, float*, float*, bf16_t*, long,
+                 long, float, float, float, float, float, float, hipStream_t);
+void launch_adagrad(float*, const void*, bool, float*, bf16_t*, long, float,
+                    float, float, float, hipStream_t);
+void launch_gemm(const bf16_t*, const bf16_t*, const float*, void*, bool,
+                 int, int, int, int, int, int, bool, bool, int, hipStream_t);
+void launch_softmax_xent_fwd(const bf16_t*, const long*, bf16_t*, float*,
+                             int, int, hipStream_t);
+void launch_softmax_xent_bwd(const bf16_t*, const long*, bf16_t*, float,
+                             int, int, hipStream_t);
+void launch_gather_bf16(const bf16_t*, const long*, bf16_t*, long, int, long,
+                        hipStream_t);
+void launch_gather_f32(const float*, const long*, float*, long, int, long,
+                       hipStream_t);
+void launch_scatter_add_bf16(float*, const long*, const bf16_t*, long, int,
+                             long, hipStream_t);
+void launch_scatter_add_f32(float*, const long*, const float*, long, int,
+                            long, hipStream_t);
+void launch_relu_bwd(const bf16_t*, const bf16_t*, bf16_t*, long,
+                     hipStream_t);
+void launch_colsum(const bf16_t*, float*, int, int, hipStream_t);
+
+namespace {
+
+hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+const void* grad_ptr(const torch::Tensor& g, bool* is_bf16) {
+  TORCH_CHECK(g.is_contiguous(), "grad must be contiguous");
+  if (g.scalar_type() == torch::kBFloat16) {
+    *is_bf16 = true;
+    return g.data_ptr();
+  }
+  TORCH_CHECK(g.scalar_type() == torch::kFloat32, "grad must be fp32 or bf16");
+  *is_bf16 = false;
+  return g.data_ptr();
+}
+
+float* opt_f32(torch::Tensor& t, long n, const char* name) {
+  if (t.numel() == 0) return nullptr;
+  TORCH_CHECK(t.numel() == n && t.scalar_type() == torch::kFloat32,
+              name, " must be fp32 with same numel as param");
+  return t.data_ptr<float>();
+}
+
+bf16_t* opt_bf16(torch::Tensor& t, long n, const char* name) {
+  if (t.numel() == 0) return nullptr;
+  TORCH_CHECK(t.numel() == n && t.scalar_type() == torch::kBFloat16,
+              name, " must be bf16 with same numel as param");
+  return (bf16_t*)t.data_ptr();
+}
+
+void fused_sgd(torch::Tensor param, torch::Tensor grad,
+               torch::Tensor momentum_buf, torch::Tensor bf16_out, double lr,
+               double momentum, double weight_decay, double grad_scale) {
+  TORCH_CHECK(param.is_cuda() && param.is_contiguous() &&
+              param.scalar_type() == torch::kFloat32,
+              "param must be contiguous fp32 on GPU");
+  long n = param.numel();
+  TORCH_CHECK(grad.numel() == n, "grad/param numel mismatch");
+  bool gb;
+  const void* g = grad_ptr(grad, &gb);
+  launch_sgd(param.data_ptr<float>(), g, gb,
+             opt_f32(momentum_buf, n, "momentum_buf"),
+             opt_bf16(bf16_out, n, "bf16_out"), n, (float)lr,
+             (float)momentum, (float)weight_decay, (float)grad_scale,
+             cur_stream());
+}
+
+void fused_adam(torch::Tensor param, torch::Tensor grad, torch::Tensor m,
+                torch::Tensor v, torch::Tensor bf16_out, long step, double lr,
+                double beta1, double beta2, double eps, double weight_decay,
+                double grad_scale) {
+  long n = param.numel();
+  TORCH_CHECK(param.is_cuda() && param.scalar_type() == torch::kFloat32 &&
+              grad.numel() == n && m.numel() == n && v.numel() == n,
+              "adam tensor mismatch");
+  bool gb;
+  const void* g = grad_ptr(grad, &gb);
+  launch_adam(param.data_ptr<float>(), g, gb, m.data_ptr<float>(),
+              v.data_ptr<float>(), opt_bf16(bf16_out, n, "bf16_out"), n, step,
+              (float)lr, (float)beta1, (float)beta2, (float)eps,
+              (float)weight_decay, (float)grad_scale, cur_stream());
+}
+
+void fused_adagrad(torch::Tensor param, torch::Tensor grad,
+                   torch::Tensor accum, torch::Tensor bf16_out, double lr,
+                   double eps, double weight_decay, double grad_scale) {
+  long n = param.numel();
+  TORCH_CHECK(param.is_cuda() && param.scalar_type() == torch::kFloat32 &&
+              grad.numel() == n && accum.numel() == n,
+              "adagrad tensor mismatch");
+  bool gb;
+  const void* g = grad_ptr(grad, &gb);
+  launch_adagrad(param.data_ptr<float>(), g, gb, accum.data_ptr<float>(),
+                 opt_bf16(bf16_out, n, "bf16_out"), n, (float)lr, (float)eps,
+                 (float)weight_decay, (float)grad_scale, cur_stream());
+}
+
+torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
+                                torch::Tensor bias, long act, bool trans_a,
+                                bool trans_b, torch::Tensor out) {
+  TORCH_CHECK(a.is_cuda() && b.is_cuda(), "gemm: tensors must be on GPU");
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+              b.scalar_type() == torch::kBFloat16, "gemm: bf16 inputs only");
+  TORCH_CHECK(a.is_contiguous() && b.is_contiguous() && a.dim() == 2 &&
+              b.dim() == 2, "gemm: contiguous 2-D inputs");
+  int M = trans_a ? a.size(1) : a.size(0);
+  int Ka = trans_a ? a.size(0) : a.size(1);
+  int Kb = trans_b ? b.size(1) : b.size(0);
+  int N = trans_b ? b.size(0) : b.size(1);
+  TORCH_CHECK(Ka == Kb, "gemm: inner dims mismatch ", Ka, " vs ", Kb);
+  const float* bias_p = nullptr;
+  if (bias.numel() > 0) {
+    TORCH_CHECK(bias.scalar_type() == torch::kFloat32 && bias.numel() == N,
+                "bias must be fp32 [N]");
+    bias_p = bias.data_ptr<float>();
+  }
+  bool out_f32;
+  if (out.numel() == 0) {
+    out = torch::empty({M, N}, a.options());
+    out_f32 = false;
+  } else {
+    TORCH_CHECK(out.is_contiguous() && out.dim() == 2 && out.size(0) == M &&
+                out.size(1) == N, "out shape mismatch");
+    out_f32 = out.scalar_type() == torch::kFloat32;
+    TORCH_CHECK(out_f32 || out.scalar_type() == torch::kBFloat16,
+                "out must be fp32 or bf16");
+  }
+  launch_gemm((const bf16_t*)a.data_ptr(), (const bf16_t*)b.data_ptr(),
+              bias_p, out.data_ptr(), out_f32, M, N, Ka, a.size(1), b.size(1),
+              N, trans_a, trans_b, (int)act, cur_stream());
+  return out;
+}
+
+torch::Tensor gemm_bias_act(torch::Tensor a, torch::Tensor b,
+                            torch::Tensor bias, long act, bool trans_a,
+                            bool trans_b) {
+  auto empty = torch::empty({0}, a.options());
+  return gemm_bias_act_out(a, b, bias, act, trans_a, trans_b, empty);
+}
+
+std::tuple<torch::Tensor, torch::Tensor> softmax_xent_fwd(
+    torch::Tensor logits, torch::Tensor labels) {
+  TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == torch::kBFloat16 &&
+              logits.is_contiguous() && logits.dim() == 2,
+              "logits must be contiguous bf16 [B,C] on GPU");
+  TORCH_CHECK(labels.scalar_type() == torch::kInt64 &&
+              labels.numel() == logits.size(0), "labels must be i64 [B]");
+  int B = logits.size(0), C = logits.size(1);
+  auto probs = torch::empty_like(logits);
+  auto loss = torch::zeros({}, logits.options().dtype(torch::kFloat32));
+  launch_softmax_xent_fwd((const bf16_t*)logits.data_ptr(),
+                          labels.data_ptr<long>(), (bf16_t*)probs.data_ptr(),
+                          loss.data_ptr<float>(), B, C, cur_stream());
+  return {loss, probs};
+}
+
+torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
+                               double scale) {
+  TORCH_CHECK(probs.is_cuda() && probs.scalar_type() == torch::kBFloat16 &&
+              probs.is_contiguous() && probs.dim() == 2, "probs bf16 [B,C]");
+  int B = probs.size(0), C = probs.size(1);
+  auto d = torch::empty_like(probs);
+  launch_softmax_xent_bwd((const bf16_t*)probs.data_ptr(),
+                          labels.data_ptr<long>(), (bf16_t*)d.data_ptr(),
+                          (float)scale, B, C, cur_stream());
+  return d;
+}
+
+torch::Tensor embedding_gather(torch::Tensor table, torch::Tensor ids) {
+  TORCH_CHECK(table.is_cuda() && table.dim() == 2 && table.is_contiguous(),
+              "table must be contiguous [V,D] on GPU");
+  TORCH_CHECK(ids.scalar_type() == torch::kInt64 && ids.dim() == 1,
+              "ids must be i64 [N]");
+  long n = ids.numel(), V = table.size(0);
+  int D = table.size(1);
+  auto out = torch::empty({n, (long)D}, table.options());
+  if (n == 0) return out;
+  if (table.scalar_type() == torch::kBFloat16)
+    launch_gather_bf16((const bf16_t*)table.data_ptr(), ids.data_ptr<long>(),
+                       (bf16_t*)out.data_ptr(), n, D, V, cur_stream());
+  else if (table.scalar_type() == torch::kFloat32)
+    launch_gather_f32(table.data_ptr<float>(), ids.data_ptr<long>(),
+                      out.data_ptr<float>(), n, D, V, cur_stream());
+  else
+    TORCH_CHECK(false, "table must be bf16 or fp32");
+  return out;
+}
+
+void embedding_scatter_add(torch::Tensor table, torch::Tensor ids,
+                           torch::Tensor rows) {
+  TORCH_CHECK(table.is_cuda() && table.scalar_type() == torch::kFloat32 &&
+              table.is_contiguous(), "scatter-add table must be fp32 [V,D]");
+  TORCH_CHECK(ids.scalar_type() == torch::kInt64, "ids must be i64");
+  long n = ids.numel(), V = table.size(0);
+  int D = table.size(1);
+  if (n == 0) return;
+  TORCH_CHECK(rows.is_contiguous() && rows.numel() == n * D,
+              "rows shape mismatch");
+  if (rows.scalar_type() == torch::kBFloat16)
+    launch_scatter_add_bf16(table.data_ptr<float>(), ids.data_ptr<long>(),
+                            (const bf16_t*)rows.data_ptr(), n, D, V,
+                            cur_stream());
+  else
+    launch_scatter_add_f32(table.data_ptr<float>(), ids.data_ptr<long>(),
+                           rows.data_ptr<float>(), n, D, V, cur_stream());
+}
+
+torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor act) {
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16 &&
+              dy.is_contiguous() && act.is_contiguous() &&
+              act.numel() == dy.numel(), "relu_bwd: bf16 contiguous");
+  auto dx = torch::empty_like(dy);
+  launch_relu_bwd((const bf16_t*)dy.data_ptr(), (const bf16_t*)act.data_ptr(),
+                  (bf16_t*)dx.data_ptr(), dy.numel(), cur_stream());
+  return dx;
+}
+
+torch::Tensor colsum(torch::Tensor x, torch::Tensor out) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 &&
+              x.is_contiguous() && x.dim() == 2, "colsum: bf16 [M,N]");
+  int M = x.size(0), N = x.size(1);
+  if (out.numel() == 0)
+    out = torch::empty({N}, x.options().dtype(torch::kFloat32));
+  else
+    TORCH_CHECK(out.scalar_type() == torch::kFloat32 && out.numel() == N &&
+                out.is_contiguous(), "colsum out must be fp32 [N]");
+  launch_colsum((const bf16_t*)x.data_ptr(), out.data_ptr<float>(), M, N,
+                cur_stream());
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_sgd", &fused_sgd, "fused SGD apply + bf16 shadow",
+        py::arg("param"), py::arg("grad"), py::arg("momentum_buf"),
+        py::arg("bf16_out"), py::arg("lr"), py::arg("momentum") = 0.0,
+        py::arg("weight_decay") = 0.0, py::arg("grad_scale") = 1.0);
+  m.def("fused_adam", &fused_adam, "fused Adam apply + bf16 shadow",
+        py::arg("param"), py::arg("grad"), py::arg("m"), py::arg("v"),
+        py::arg("bf16_out"), py::arg("step"), py::arg("lr"),
+        py::arg("beta1") = 0.9, py::arg("beta2") = 0.999,
+        py::arg("eps") = 1e-8, py::arg("weight_decay") = 0.0,
+        py::arg("grad_scale") = 1.0);
+  m.def("fused_adagrad", &fused_adagrad, "fused Adagrad apply + bf16 shadow",
+        py::arg("param"), py::arg("grad"), py::arg("accum"),
+        py::arg("bf16_out"), py::arg("lr"), py::arg("eps") = 1e-10,
+        py::arg("weight_decay") = 0.0, py::arg("grad_scale") = 1.0);
+  m.def("gemm_bias_act", &gemm_bias_act, "bf16 MFMA GEMM + bias + act");
+  m.def("gemm_bias_act_out", &gemm_bias_act_out,
+        "bf16 MFMA GEMM + bias + act into out (bf16 or fp32)");
+  m.def("softmax_xent_fwd", &softmax_xent_fwd);
+  m.def("softmax_xent_bwd", &softmax_xent_bwd);
+  m.def("embedding_gather", &embedding_gather);
+  m.def("embedding_scatter_add", &embedding_scatter_add);
+  m.def("relu_bwd", &relu_bwd);
+  m.def("colsum", &colsum, py::arg("x"), py::arg("out"));
+}

@@ -128,13 +128,13 @@ class SyncReplicaTrainer(object):
         scale = grad_scale if grad_scale is not None \
             else 1.0 / self.roles.n_workers
         if self.world > 1:
+            if self.roles.is_ps and not self.roles.is_worker:
+                # dist.reduce is in-place: the PS buffer holds last step's
+                # sum and would be re-added — contribute zeros instead
+                self.flat_grad.zero_()
             dist.reduce(self.flat_grad, dst=self.roles.ps_rank, op=dist.ReduceOp.SUM)
         if self.roles.is_ps:
-            g = self.flat_grad
-            if scale != 1.0:
-                g = g.mul_(scale)
-            self.store.apply_flat(g.to(self.store.flat.dtype)
-                                  if g.dtype != self.store.flat.dtype else g)
+            self.store.apply_flat(self.flat_grad, grad_scale=scale)
         if self.world > 1:
             dist.broadcast(self.flat_params_bf16, src=self.roles.ps_rank)
         return self.store.global_step
@@ -149,41 +149,68 @@ class SyncReplicaTrainer(object):
             dist.broadcast(self.flat_params_bf16, src=self.roles.ps_rank)
 
 
-class AsyncPSWorker(object):
-    """Worker side of async (apply-on-arrival) PS exchange."""
+def make_pair_groups(roles):
+    """One process group per (ps, worker) pair — every rank must call
+    new_group for every pair (it is collective). These give async mode
+    independent channels: each PS serving thread blocks on its own
+    group, which is safe on both gloo and RCCL (one communicator per
+    thread)."""
+    groups = {}
+    for w in range(1, roles.world):
+        groups[w] = dist.new_group([roles.ps_rank, w])
+    return groups
 
-    def __init__(self, trainer):
+
+class AsyncPSWorker(object):
+    """Worker side of async (apply-on-arrival) PS exchange.
+
+    Replicates the reference's default async mode (each worker pushes
+    grads and pulls params on its own clock, no aggregation barrier —
+    README.rst:68-72)."""
+
+    def __init__(self, trainer, pair_groups):
         self.t = trainer
         assert trainer.world > 1, "async mode needs a separate PS rank"
+        self.group = pair_groups[trainer.rank]
 
     def step(self):
         t = self.t
-        dist.send(t.flat_grad, dst=t.roles.ps_rank)
-        dist.recv(t.flat_params_bf16, src=t.roles.ps_rank)
+        dist.send(t.flat_grad, dst=t.roles.ps_rank, group=self.group)
+        dist.recv(t.flat_params_bf16, src=t.roles.ps_rank, group=self.group)
         return True
+
+    def stop(self):
+        """Tell the PS this worker is done (a zero-length sentinel is not
+        expressible; the server counts steps instead)."""
 
 
 class AsyncPSServer(object):
-    """PS side: serve gradient pushes until each worker sent `steps`."""
+    """PS side: one serving thread per worker, apply-on-arrival."""
 
-    def __init__(self, trainer):
+    def __init__(self, trainer, pair_groups):
         self.t = trainer
+        self.groups = pair_groups
         assert trainer.roles.is_ps and trainer.world > 1
 
-    def serve(self, steps_per_worker):
+    def _serve_one(self, worker_rank, steps):
         t = self.t
-        n = t.roles.n_workers
-        bufs = [torch.zeros_like(t.flat_grad) for _ in range(n)]
-        reqs = [dist.irecv(bufs[i], src=i + 1) for i in range(n)]
-        remaining = [steps_per_worker] * n
-        while any(r > 0 for r in remaining):
-            for i in range(n):
-                if remaining[i] > 0 and reqs[i] is not None and reqs[i].is_completed():
-                    t.store.apply_flat(
-                        bufs[i].to(t.store.flat.dtype)
-                        if bufs[i].dtype != t.store.flat.dtype else bufs[i].clone())
-                    dist.send(t.store.flat_bf16, dst=i + 1)
-                    remaining[i] -= 1
-                    reqs[i] = dist.irecv(bufs[i], src=i + 1) \
-                        if remaining[i] > 0 else None
-        return t.store.global_step
+        buf = torch.zeros_like(t.flat_grad)
+        group = self.groups[worker_rank]
+        for _ in range(steps):
+            dist.recv(buf, src=worker_rank, group=group)
+            with t.store.lock:
+                t.store.apply_flat(buf)
+                params = t.store.flat_bf16.clone()
+            dist.send(params, dst=worker_rank, group=group)
+
+    def serve(self, steps_per_worker):
+        import threading
+        threads = [
+            threading.Thread(target=self._serve_one, args=(w, steps_per_worker))
+            for w in range(1, self.t.world)
+        ]
+        for th in threads:
+            th.start()
+        for th in threads:
+            th.join()
+        return self.t.store.global_step

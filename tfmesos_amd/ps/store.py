@@ -121,9 +121,10 @@ class PStore(object):
             self.apply_flat(flat_g)
             return self.global_step
 
-    def apply_flat(self, flat_grad):
+    def apply_flat(self, flat_grad, grad_scale=1.0):
         """One fused optimizer apply over the whole flat buffer; also
-        refreshes the bf16 shadow in the same kernel."""
+        refreshes the bf16 shadow in the same kernel. grad_scale folds
+        the sync-replica worker-mean into the kernel."""
         with self.lock:
             self.global_step += 1
             hp = self.hparams
@@ -133,7 +134,7 @@ class PStore(object):
                               momentum=hp.get("momentum", 0.0),
                               weight_decay=hp.get("weight_decay", 0.0),
                               momentum_buf=self.state.get("momentum_buf"),
-                              bf16_out=self.flat_bf16)
+                              bf16_out=self.flat_bf16, grad_scale=grad_scale)
             elif self.opt == "adam":
                 ops.fused_adam(self.flat, flat_grad,
                                self.state["exp_avg"], self.state["exp_avg_sq"],
@@ -142,12 +143,12 @@ class PStore(object):
                                beta2=hp.get("beta2", 0.999),
                                eps=hp.get("eps", 1e-8),
                                weight_decay=hp.get("weight_decay", 0.0),
-                               bf16_out=self.flat_bf16)
+                               bf16_out=self.flat_bf16, grad_scale=grad_scale)
             elif self.opt == "adagrad":
                 ops.fused_adagrad(self.flat, flat_grad, self.state["accum"],
                                   lr, eps=hp.get("eps", 1e-10),
                                   weight_decay=hp.get("weight_decay", 0.0),
-                                  bf16_out=self.flat_bf16)
+                                  bf16_out=self.flat_bf16, grad_scale=grad_scale)
             else:
                 raise ValueError("unknown optimizer %r" % self.opt)
 
