@@ -14,8 +14,9 @@ void launch_adam(float*, const void*, bool, float*, float*, bf16_t*, long,
                  long, float, float, float, float, float, float, hipStream_t);
 void launch_adagrad(float*, const void*, bool, float*, bf16_t*, long, float,
                     float, float, float, hipStream_t);
-void launch_gemm(const bf16_t*, const bf16_t*, const float*, void*, bool,
-                 int, int, int, int, int, int, bool, bool, int, hipStream_t);
+void launch_gemm(const bf16_t*, const bf16_t*, const void*, bool, void*,
+                 bool, int, int, int, int, int, int, bool, bool, int,
+                 hipStream_t);
 void launch_softmax_xent_fwd(const bf16_t*, const long*, bf16_t*, float*,
                              int, int, hipStream_t);
 void launch_softmax_xent_bwd(const bf16_t*, const long*, bf16_t*, float,
@@ -123,11 +124,14 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
   int Kb = trans_b ? b.size(1) : b.size(0);
   int N = trans_b ? b.size(0) : b.size(1);
   TORCH_CHECK(Ka == Kb, "gemm: inner dims mismatch ", Ka, " vs ", Kb);
-  const float* bias_p = nullptr;
+  const void* bias_p = nullptr;
+  bool bias_bf16 = false;
   if (bias.numel() > 0) {
-    TORCH_CHECK(bias.scalar_type() == torch::kFloat32 && bias.numel() == N,
-                "bias must be fp32 [N]");
-    bias_p = bias.data_ptr<float>();
+    TORCH_CHECK((bias.scalar_type() == torch::kFloat32 ||
+                 bias.scalar_type() == torch::kBFloat16) && bias.numel() == N,
+                "bias must be fp32 or bf16 [N]");
+    bias_bf16 = bias.scalar_type() == torch::kBFloat16;
+    bias_p = bias.data_ptr();
   }
   bool out_f32;
   if (out.numel() == 0) {
@@ -141,8 +145,9 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
                 "out must be fp32 or bf16");
   }
   launch_gemm((const bf16_t*)a.data_ptr(), (const bf16_t*)b.data_ptr(),
-              bias_p, out.data_ptr(), out_f32, M, N, Ka, a.size(1), b.size(1),
-              N, trans_a, trans_b, (int)act, cur_stream());
+              bias_p, bias_bf16, out.data_ptr(), out_f32, M, N, Ka,
+              a.size(1), b.size(1), N, trans_a, trans_b, (int)act,
+              cur_stream());
   return out;
 }
 

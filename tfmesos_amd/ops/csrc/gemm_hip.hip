@@ -24,7 +24,8 @@ constexpr int APAD = 8;  // +16B: keeps b128 fragment reads aligned
 template <bool TA, bool TB, int ACT, bool BIAS, bool OUTF32>
 __global__ __launch_bounds__(256)
 void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
-                 const float* __restrict__ bias, void* __restrict__ Cout,
+                 const void* __restrict__ bias, bool bias_bf16,
+                 void* __restrict__ Cout,
                  int M, int N, int K, int lda, int ldb, int ldc) {
   __shared__ bf16_t As[BM][BK + APAD];   // [m][k]
   __shared__ bf16_t Bs[BN][BK + APAD];   // [n][k] (B^T tile)
@@ -91,7 +92,9 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     for (int fn = 0; fn < 2; ++fn) {
       const int col = tn0 + wc * 32 + fn * 16 + (lane & 15);
       if (col >= N) continue;
-      const float bv = BIAS ? bias[col] : 0.f;
+      const float bv = !BIAS ? 0.f
+          : (bias_bf16 ? bf2f(((const bf16_t*)bias)[col])
+                       : ((const float*)bias)[col]);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + r;
@@ -109,16 +112,17 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
 
 }  // namespace
 
-void launch_gemm(const bf16_t* A, const bf16_t* B, const float* bias,
-                 void* C, bool out_f32, int M, int N, int K, int lda,
-                 int ldb, int ldc, bool ta, bool tb, int act,
+void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
+                 bool bias_bf16, void* C, bool out_f32, int M, int N, int K,
+                 int lda, int ldb, int ldc, bool ta, bool tb, int act,
                  hipStream_t stream) {
   dim3 grid(ceil_div(N, BN), ceil_div(M, BM));
   dim3 block(256);
   const bool has_bias = bias != nullptr;
 #define DISP(TAv, TBv, ACTv, BIASv, OUTv)                                  \
   hipLaunchKernelGGL((gemm_kernel<TAv, TBv, ACTv, BIASv, OUTv>), grid,     \
-                     block, 0, stream, A, B, bias, C, M, N, K, lda, ldb, ldc)
+                     block, 0, stream, A, B, bias, bias_bf16, C, M, N, K,   \
+                     lda, ldb, ldc)
 #define DISP_OUT(TAv, TBv, ACTv, BIASv)                                    \
   do { if (out_f32) DISP(TAv, TBv, ACTv, BIASv, true);                     \
        else DISP(TAv, TBv, ACTv, BIASv, false); } while (0)
