@@ -125,6 +125,65 @@ def test_gemm_out_fp32():
     assert (out.cpu() - ref).abs().max() < 0.15
 
 
+def test_gemm_splitk_repeatable():
+    """(100,100,784) nn hits the split-K path; the last-arriver epilogue
+    must leave the workspace zeroed, so repeated launches agree."""
+    from tfmesos_amd import ops
+    torch.manual_seed(12)
+    A, B = bf(torch.randn(100, 784)), bf(torch.randn(784, 100))
+    ref = A.float() @ B.float()
+    outs = [ops.gemm_bias_act(A, B) for _ in range(5)]
+    for o in outs:
+        assert (o.float() - ref).abs().max() < 0.02 * ref.abs().max() + 0.05
+        assert torch.equal(o, outs[0])
+
+
+def test_gemm_splitk_bias_relu_deep():
+    from tfmesos_amd import ops
+    torch.manual_seed(13)
+    A, B = torch.randn(64, 2000), torch.randn(2000, 64)
+    bias = torch.randn(64)
+    ref = torch.relu(A.float() @ B.float() + bias)
+    out = ops.gemm_bias_act(bf(A), bf(B), bias.to(DEV), act="relu")
+    assert (out.float().cpu() - ref).abs().max() < 0.02 * ref.abs().max() + 0.1
+
+
+def test_gemm_relu_bwd_epilogue():
+    """dX = (dY @ W^T) * (h > 0) fused into the GEMM."""
+    from tfmesos_amd import ops
+    torch.manual_seed(14)
+    dy, w, h = torch.randn(100, 10), torch.randn(100, 10), torch.randn(100, 100)
+    ref = (dy.float() @ w.float().t()) * (h.to(torch.bfloat16).float() > 0)
+    out = ops.gemm_bias_act(bf(dy), bf(w), trans_b=True, act="relu_bwd",
+                            aux=bf(h))
+    assert (out.float().cpu() - ref).abs().max() < 0.05
+
+
+def test_gemm_colsum_epilogue():
+    """dW = X^T dY with the bias gradient colsum(dY) fused in."""
+    from tfmesos_amd import ops
+    torch.manual_seed(15)
+    x, dy = torch.randn(100, 784), torch.randn(100, 100)
+    xb, dyb = bf(x), bf(dy)
+    ref_w = xb.float().t() @ dyb.float()
+    ref_b = dyb.float().sum(0)
+    out = torch.zeros(784, 100, device=DEV)
+    cs = torch.zeros(100, device=DEV)
+    ops.gemm_bias_act(xb, dyb, trans_a=True, out=out, colsum_out=cs)
+    assert (out.cpu() - ref_w.cpu()).abs().max() < 0.3
+    assert (cs.cpu() - ref_b.cpu()).abs().max() < 0.05
+
+
+def test_gemm_odd_leading_dim():
+    """ld not divisible by the vector widths falls back to scalar staging."""
+    from tfmesos_amd import ops
+    torch.manual_seed(16)
+    A, B = torch.randn(33, 17), torch.randn(17, 21)
+    ref = A.float() @ B.float()
+    out = ops.gemm_bias_act(bf(A), bf(B))
+    assert (out.float().cpu() - ref).abs().max() < 0.15
+
+
 # ----------------------------------------------------------- softmax-xent
 
 def test_softmax_xent_fwd_bwd():
@@ -143,6 +202,35 @@ def test_softmax_xent_fwd_bwd():
     ref_d[torch.arange(B), labels] -= 1
     ref_d /= B
     assert (d.float().cpu() - ref_d).abs().max() < 1e-3
+
+
+def test_softmax_xent_fused_one_kernel():
+    from tfmesos_amd import ops
+    torch.manual_seed(22)
+    B, C = 100, 10
+    logits = torch.randn(B, C) * 3
+    labels = torch.randint(0, C, (B,))
+    loss, d = ops.softmax_xent_fused(bf(logits), labels.to(DEV))
+    ref_probs = torch.softmax(bf(logits).float().cpu(), 1)
+    ref_loss = torch.nn.functional.cross_entropy(logits.float(), labels)
+    ref_d = ref_probs.clone()
+    ref_d[torch.arange(B), labels] -= 1
+    ref_d /= B
+    assert abs(float(loss) - float(ref_loss)) < 0.02
+    assert (d.float().cpu() - ref_d).abs().max() < 1e-3
+
+
+def test_softmax_xent_fused_large_falls_back():
+    """Shapes beyond the one-workgroup kernel compose fwd+bwd correctly."""
+    from tfmesos_amd import ops
+    torch.manual_seed(23)
+    B, C = 512, 100
+    logits = torch.randn(B, C) * 2
+    labels = torch.randint(0, C, (B,))
+    loss, d = ops.softmax_xent_fused(bf(logits), labels.to(DEV))
+    ref_loss = torch.nn.functional.cross_entropy(logits.float(), labels)
+    assert abs(float(loss) - float(ref_loss)) < 0.05
+    assert d.shape == (B, C)
 
 
 def test_softmax_xent_wide():

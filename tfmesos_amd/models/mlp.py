@@ -61,23 +61,24 @@ class MnistMLP(object):
         g: callable name -> grad view to fill (fp32)
         Returns mean loss (fp32 scalar tensor).
         """
-        B = x.shape[0]
         hid_w, hid_b = p("hid_w"), p("hid_b")
         sm_w, sm_b = p("sm_w"), p("sm_b")
 
-        # forward
+        # forward (2 kernels: split-K GEMM+bias+relu, GEMM+bias)
         h = ops.gemm_bias_act(x, hid_w, hid_b, act="relu")       # [B,H]
         logits = ops.gemm_bias_act(h, sm_w, sm_b, act="none")    # [B,C]
-        loss, probs = ops.softmax_xent_fwd(logits, y)
+        # loss fwd + bwd fused into one kernel
+        loss, dlogits = ops.softmax_xent_fused(logits, y)         # [B,C] (/B)
 
-        # backward (fp32 grads written straight into the flat grad views)
-        dlogits = ops.softmax_xent_bwd(probs, y)                  # [B,C] (/B)
-        ops.gemm_bias_act(h, dlogits, trans_a=True, out=g("sm_w"))
-        ops.colsum(dlogits, out=g("sm_b"))
-        dh = ops.gemm_bias_act(dlogits, sm_w, trans_b=True)       # [B,H]
-        dh = ops.relu_bwd(dh, h)
-        ops.gemm_bias_act(x, dh, trans_a=True, out=g("hid_w"))
-        ops.colsum(dh, out=g("hid_b"))
+        # backward: 3 GEMMs with fused epilogues, fp32 grads written
+        # straight into the flat grad views (bias colsums ride the dW
+        # GEMMs; relu-bwd masking rides the dh GEMM)
+        ops.gemm_bias_act(h, dlogits, trans_a=True, out=g("sm_w"),
+                          colsum_out=g("sm_b"))
+        dh = ops.gemm_bias_act(dlogits, sm_w, trans_b=True,
+                               act="relu_bwd", aux=h)             # [B,H]
+        ops.gemm_bias_act(x, dh, trans_a=True, out=g("hid_w"),
+                          colsum_out=g("hid_b"))
         return loss
 
     def loss_only(self, p, x, y):

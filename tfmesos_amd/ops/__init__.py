@@ -121,22 +121,35 @@ def fused_adagrad(param, grad, accum, lr, eps=1e-10, weight_decay=0.0,
 # --------------------------------------------------------------------- gemm
 
 def gemm_bias_act(a, b, bias=None, act="none", trans_a=False, trans_b=False,
-                  out=None):
+                  out=None, aux=None, colsum_out=None):
     """C = act(op(A) @ op(B) + bias), bf16 in, fp32 accumulate.
 
     Output dtype: bf16 by default; pass ``out`` (bf16 or fp32) to write
     in place — backward GEMMs write fp32 straight into the flat gradient
-    buffer, skipping a cast+copy. GPU: hand-written MFMA kernel
-    (csrc/gemm.hip). CPU: torch reference in fp32.
+    buffer, skipping a cast+copy. Fused epilogues (one kernel each):
+
+    * ``act="relu_bwd"`` with ``aux``: mask the product by ``aux > 0``
+      (the saved forward activation) — the dX GEMM absorbs relu_bwd.
+    * ``colsum_out``: also write ``sum_k op(B)[k, n]`` (fp32 [N]) — the
+      dW GEMM absorbs the bias gradient's column sum.
+
+    GPU: hand-written MFMA kernel (csrc/gemm.hip) with split-K for deep
+    skinny shapes. CPU: torch reference in fp32.
     """
-    act_code = {"none": 0, "relu": 1}[act]
+    act_code = {"none": 0, "relu": 1, "relu_bwd": 2}[act]
     if a.is_cuda:
-        ebias = bias if bias is not None else torch.empty(0, device=a.device)
-        if out is not None:
-            return _ext().gemm_bias_act_out(a, b, ebias, act_code,
-                                            bool(trans_a), bool(trans_b), out)
-        return _ext().gemm_bias_act(a, b, ebias, act_code, bool(trans_a),
-                                    bool(trans_b))
+        empty = torch.empty(0, device=a.device)
+        ebias = bias if bias is not None else empty
+        eaux = aux if aux is not None else empty
+        ecs = colsum_out if colsum_out is not None else empty
+        if out is None:
+            out = torch.empty(
+                (a.shape[1] if trans_a else a.shape[0],
+                 b.shape[0] if trans_b else b.shape[1]),
+                device=a.device, dtype=a.dtype)
+        return _ext().gemm_bias_act_out(a, b, ebias, act_code,
+                                        bool(trans_a), bool(trans_b), out,
+                                        eaux, ecs)
     x = a.float().t() if trans_a else a.float()
     y = b.float().t() if trans_b else b.float()
     c = x @ y
@@ -144,6 +157,10 @@ def gemm_bias_act(a, b, bias=None, act="none", trans_a=False, trans_b=False,
         c = c + bias.float()
     if act == "relu":
         c = torch.relu(c)
+    elif act == "relu_bwd":
+        c = c * (aux.float() > 0)
+    if colsum_out is not None:
+        colsum_out.copy_(y.sum(0))
     if out is not None:
         out.copy_(c.to(out.dtype))
         return out
@@ -177,6 +194,22 @@ def softmax_xent_fwd(logits, labels):
     probs = torch.softmax(lg, dim=1)
     loss = torch.nn.functional.nll_loss(torch.log_softmax(lg, 1), labels)
     return loss, probs.to(logits.dtype)
+
+
+def softmax_xent_fused(logits, labels, scale=None):
+    """Returns (mean_loss fp32 scalar, dlogits bf16 [B,C]) in one fused
+    kernel (fwd softmax + loss + bwd (p - onehot)*scale); scale defaults
+    to 1/B. The mnist step's loss path is this single launch."""
+    B = logits.shape[0]
+    s = float(scale if scale is not None else 1.0 / B)
+    if logits.is_cuda:
+        return _ext().softmax_xent_fused(logits, labels, s)
+    lg = logits.float()
+    probs = torch.softmax(lg, dim=1)
+    loss = torch.nn.functional.nll_loss(torch.log_softmax(lg, 1), labels)
+    d = probs.clone()
+    d[torch.arange(B), labels] -= 1.0
+    return loss, (d * s).to(logits.dtype)
 
 
 def softmax_xent_bwd(probs, labels, scale=None):

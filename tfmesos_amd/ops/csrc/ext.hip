@@ -15,10 +15,13 @@ void launch_adam(float*, const void*, bool, float*, float*, bf16_t*, long,
 void launch_adagrad(float*, const void*, bool, float*, bf16_t*, long, float,
                     float, float, float, hipStream_t);
 void launch_gemm(const bf16_t*, const bf16_t*, const void*, bool, void*,
-                 bool, int, int, int, int, int, int, bool, bool, int,
+                 bool, const bf16_t*, float*, float*, int*, int, int, int,
+                 int, int, int, int, int, bool, bool, int, int, int,
                  hipStream_t);
 void launch_softmax_xent_fwd(const bf16_t*, const long*, bf16_t*, float*,
                              int, int, hipStream_t);
+void launch_softmax_xent_fused(const bf16_t*, const long*, bf16_t*, float*,
+                               float, int, int, hipStream_t);
 void launch_softmax_xent_bwd(const bf16_t*, const long*, bf16_t*, float,
                              int, int, hipStream_t);
 void launch_gather_bf16(const bf16_t*, const long*, bf16_t*, long, int, long,
@@ -111,9 +114,34 @@ void fused_adagrad(torch::Tensor param, torch::Tensor grad,
                  (float)weight_decay, (float)grad_scale, cur_stream());
 }
 
+// staging vector width for an operand: b128 when every row start stays
+// 16B-aligned, b32 when 4B, else scalar
+int vec_level(const void* base, long ld) {
+  if (((uintptr_t)base % 16 == 0) && (ld % 8 == 0)) return 8;
+  if (((uintptr_t)base % 4 == 0) && (ld % 2 == 0)) return 2;
+  return 1;
+}
+
+// persistent per-device split-K workspace + tile arrival counters. The
+// gemm kernel's last-arriver epilogue re-zeroes exactly what it added,
+// so the buffers stay zeroed between (stream-ordered) launches.
+float* splitk_ws(const torch::Device& dev, long n, long ntiles, int** cnt) {
+  static std::unordered_map<int, torch::Tensor> wsmap, cntmap;
+  const int idx = dev.index();
+  auto& w = wsmap[idx];
+  if (!w.defined() || w.numel() < n)
+    w = torch::zeros({n}, torch::dtype(torch::kFloat32).device(dev));
+  auto& c = cntmap[idx];
+  if (!c.defined() || c.numel() < ntiles)
+    c = torch::zeros({ntiles}, torch::dtype(torch::kInt32).device(dev));
+  *cnt = c.data_ptr<int>();
+  return w.data_ptr<float>();
+}
+
 torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
                                 torch::Tensor bias, long act, bool trans_a,
-                                bool trans_b, torch::Tensor out) {
+                                bool trans_b, torch::Tensor out,
+                                torch::Tensor aux, torch::Tensor colsum_out) {
   TORCH_CHECK(a.is_cuda() && b.is_cuda(), "gemm: tensors must be on GPU");
   TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
               b.scalar_type() == torch::kBFloat16, "gemm: bf16 inputs only");
@@ -144,10 +172,45 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
     TORCH_CHECK(out_f32 || out.scalar_type() == torch::kBFloat16,
                 "out must be fp32 or bf16");
   }
+  const bf16_t* aux_p = nullptr;
+  if (act == 2) {
+    TORCH_CHECK(!trans_a && trans_b && !out_f32 && bias_p == nullptr,
+                "act=relu_bwd needs nt, bf16 out, no bias");
+    TORCH_CHECK(aux.scalar_type() == torch::kBFloat16 && aux.is_contiguous()
+                && aux.numel() == (long)M * N, "aux must be bf16 [M,N]");
+    aux_p = (const bf16_t*)aux.data_ptr();
+  }
+  float* colsum_p = nullptr;
+  if (colsum_out.numel() > 0) {
+    TORCH_CHECK(trans_a && !trans_b && bias_p == nullptr && act == 0,
+                "colsum fusion needs tn, no bias, no act");
+    TORCH_CHECK(colsum_out.scalar_type() == torch::kFloat32 &&
+                colsum_out.is_contiguous() && colsum_out.numel() == N,
+                "colsum_out must be fp32 [N]");
+    colsum_p = colsum_out.data_ptr<float>();
+  }
+  // split-K when the plain tile grid can't feed the 256-CU chip and K
+  // has enough depth to slice
+  const int nx = (N + 63) / 64, ny = (M + 63) / 64;
+  int nslice = 1, kc = 0;
+  float* ws = nullptr;
+  int* cnt = nullptr;
+  if (!trans_a && !trans_b && act != 2 && nx * ny < 64 && Ka >= 256) {
+    int want = std::min((int)(Ka / 64), 256 / (nx * ny));
+    if (want > 16) want = 16;
+    if (want > 1) {
+      kc = ((Ka + want - 1) / want + 31) / 32 * 32;
+      nslice = (Ka + kc - 1) / kc;
+      if (nslice > 1)
+        ws = splitk_ws(a.device(), (long)M * N, (long)nx * ny, &cnt);
+    }
+  }
   launch_gemm((const bf16_t*)a.data_ptr(), (const bf16_t*)b.data_ptr(),
-              bias_p, bias_bf16, out.data_ptr(), out_f32, M, N, Ka,
-              a.size(1), b.size(1), N, trans_a, trans_b, (int)act,
-              cur_stream());
+              bias_p, bias_bf16, out.data_ptr(), out_f32, aux_p, colsum_p,
+              ws, cnt, kc, nslice, M, N, Ka, a.size(1), b.size(1), N,
+              trans_a, trans_b, (int)act,
+              vec_level(a.data_ptr(), a.size(1)),
+              vec_level(b.data_ptr(), b.size(1)), cur_stream());
   return out;
 }
 
@@ -155,7 +218,8 @@ torch::Tensor gemm_bias_act(torch::Tensor a, torch::Tensor b,
                             torch::Tensor bias, long act, bool trans_a,
                             bool trans_b) {
   auto empty = torch::empty({0}, a.options());
-  return gemm_bias_act_out(a, b, bias, act, trans_a, trans_b, empty);
+  return gemm_bias_act_out(a, b, bias, act, trans_a, trans_b, empty, empty,
+                           empty);
 }
 
 std::tuple<torch::Tensor, torch::Tensor> softmax_xent_fwd(
@@ -172,6 +236,38 @@ std::tuple<torch::Tensor, torch::Tensor> softmax_xent_fwd(
                           labels.data_ptr<long>(), (bf16_t*)probs.data_ptr(),
                           loss.data_ptr<float>(), B, C, cur_stream());
   return {loss, probs};
+}
+
+// fully fused fwd+bwd: returns (mean loss, dlogits=(softmax-onehot)*scale)
+// in ONE kernel for small shapes; composes the two-kernel path otherwise
+std::tuple<torch::Tensor, torch::Tensor> softmax_xent_fused(
+    torch::Tensor logits, torch::Tensor labels, double scale) {
+  TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == torch::kBFloat16 &&
+              logits.is_contiguous() && logits.dim() == 2,
+              "logits must be contiguous bf16 [B,C] on GPU");
+  TORCH_CHECK(labels.scalar_type() == torch::kInt64 &&
+              labels.numel() == logits.size(0), "labels must be i64 [B]");
+  int B = logits.size(0), C = logits.size(1);
+  auto dlogits = torch::empty_like(logits);
+  if (B <= 256 && C <= 32) {
+    auto loss = torch::empty({}, logits.options().dtype(torch::kFloat32));
+    launch_softmax_xent_fused((const bf16_t*)logits.data_ptr(),
+                              labels.data_ptr<long>(),
+                              (bf16_t*)dlogits.data_ptr(),
+                              loss.data_ptr<float>(), (float)scale, B, C,
+                              cur_stream());
+    return {loss, dlogits};
+  }
+  auto loss = torch::zeros({}, logits.options().dtype(torch::kFloat32));
+  auto probs = torch::empty_like(logits);
+  launch_softmax_xent_fwd((const bf16_t*)logits.data_ptr(),
+                          labels.data_ptr<long>(), (bf16_t*)probs.data_ptr(),
+                          loss.data_ptr<float>(), B, C, cur_stream());
+  launch_softmax_xent_bwd((const bf16_t*)probs.data_ptr(),
+                          labels.data_ptr<long>(),
+                          (bf16_t*)dlogits.data_ptr(), (float)scale, B, C,
+                          cur_stream());
+  return {loss, dlogits};
 }
 
 torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
@@ -268,7 +364,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("weight_decay") = 0.0, py::arg("grad_scale") = 1.0);
   m.def("gemm_bias_act", &gemm_bias_act, "bf16 MFMA GEMM + bias + act");
   m.def("gemm_bias_act_out", &gemm_bias_act_out,
-        "bf16 MFMA GEMM + bias + act into out (bf16 or fp32)");
+        "bf16 MFMA GEMM + fused epilogue (bias/act/relu_bwd/colsum) into "
+        "out (bf16 or fp32), split-K for deep skinny shapes");
+  m.def("softmax_xent_fused", &softmax_xent_fused);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("embedding_gather", &embedding_gather);
