@@ -126,8 +126,8 @@ def test_gemm_out_fp32():
 
 
 def test_gemm_splitk_repeatable():
-    """(100,100,784) nn hits the split-K path; the last-arriver epilogue
-    must leave the workspace zeroed, so repeated launches agree."""
+    """(100,100,784) nn hits the split-K path; per-slice workspace
+    stripes summed in fixed order make it bitwise deterministic."""
     from tfmesos_amd import ops
     torch.manual_seed(12)
     A, B = bf(torch.randn(100, 784)), bf(torch.randn(784, 100))
@@ -153,10 +153,10 @@ def test_gemm_relu_bwd_epilogue():
     from tfmesos_amd import ops
     torch.manual_seed(14)
     dy, w, h = torch.randn(100, 10), torch.randn(100, 10), torch.randn(100, 100)
-    ref = (dy.float() @ w.float().t()) * (h.to(torch.bfloat16).float() > 0)
-    out = ops.gemm_bias_act(bf(dy), bf(w), trans_b=True, act="relu_bwd",
-                            aux=bf(h))
-    assert (out.float().cpu() - ref).abs().max() < 0.05
+    dyb, wb, hb = bf(dy), bf(w), bf(h)
+    ref = (dyb.float() @ wb.float().t()) * (hb.float() > 0)
+    out = ops.gemm_bias_act(dyb, wb, trans_b=True, act="relu_bwd", aux=hb)
+    assert (out.float() - ref).abs().max() < 0.02 * ref.abs().max() + 0.05
 
 
 def test_gemm_colsum_epilogue():

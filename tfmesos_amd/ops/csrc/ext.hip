@@ -122,9 +122,10 @@ int vec_level(const void* base, long ld) {
   return 1;
 }
 
-// persistent per-device split-K workspace + tile arrival counters. The
-// gemm kernel's last-arriver epilogue re-zeroes exactly what it added,
-// so the buffers stay zeroed between (stream-ordered) launches.
+// persistent per-device split-K workspace (one fp32 stripe per K-slice,
+// fully overwritten by every launch) + tile arrival counters (the
+// last-arriver epilogue resets them, so they stay zeroed between
+// stream-ordered launches).
 float* splitk_ws(const torch::Device& dev, long n, long ntiles, int** cnt) {
   static std::unordered_map<int, torch::Tensor> wsmap, cntmap;
   const int idx = dev.index();
@@ -202,7 +203,8 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
       kc = ((Ka + want - 1) / want + 31) / 32 * 32;
       nslice = (Ka + kc - 1) / kc;
       if (nslice > 1)
-        ws = splitk_ws(a.device(), (long)M * N, (long)nx * ny, &cnt);
+        ws = splitk_ws(a.device(), (long)M * N * nslice, (long)nx * ny,
+                       &cnt);
     }
   }
   launch_gemm((const bf16_t*)a.data_ptr(), (const bf16_t*)b.data_ptr(),

@@ -10,11 +10,12 @@
 //
 // Performance features (driven by profiles/r01_mnist_n1_kernel_stats.txt,
 // where the un-split fwd GEMM ran 4 workgroups on a 256-CU chip):
-//  * split-K: grid.z slices accumulate fp32 partials into a persistent
-//    workspace with native fp32 atomics; the LAST workgroup to arrive at
-//    each output tile (agent-scope arrival counter) runs the fused
-//    bias/activation epilogue and re-zeroes the workspace for the next
-//    launch — one kernel, no separate zero/reduce launches.
+//  * split-K: grid.z slices each store their fp32 partial tile into a
+//    per-slice workspace stripe; the LAST workgroup to arrive at each
+//    output tile (agent-scope arrival counter) sums the stripes in fixed
+//    slice order and runs the fused bias/activation epilogue — one
+//    kernel, no separate zero/reduce launches, and DETERMINISTIC
+//    (fixed-order fp32 sums, no atomics on the data path).
 //  * vectorized LDS staging: b128 loads when the operand's leading dim
 //    and base allow, b32 otherwise; both [X,K] and [K,X] storage orders
 //    have contiguous global access patterns.
@@ -143,9 +144,11 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     colsum_out[tn0 + t] = cs_acc;
 
   __shared__ int is_last;
+  const long slice_stride = (long)M * ldc;
   if (SK) {
-    // accumulate fp32 partials; ws stays zeroed between launches because
-    // the epilogue workgroup re-zeroes what it reads
+    // store this slice's fp32 partial tile into its workspace stripe
+    // (plain per-slice stores -> fixed-order summation is deterministic)
+    float* wslice = ws + (long)blockIdx.z * slice_stride;
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
@@ -155,8 +158,10 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + r;
-          if (row < M) unsafeAtomicAdd(&ws[(long)row * ldc + col],
-                                       acc[fm][fn][r]);
+          if (row < M)
+            __hip_atomic_store(&wslice[(long)row * ldc + col],
+                               acc[fm][fn][r], __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
         }
       }
     __threadfence();
@@ -188,10 +193,11 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
         const long idx = (long)row * ldc + col;
         float v;
         if (SK) {
-          v = __hip_atomic_load(&ws[idx], __ATOMIC_RELAXED,
-                                __HIP_MEMORY_SCOPE_AGENT);
-          __hip_atomic_store(&ws[idx], 0.f, __ATOMIC_RELAXED,
-                             __HIP_MEMORY_SCOPE_AGENT);
+          v = 0.f;
+          for (int z = 0; z < (int)gridDim.z; ++z)
+            v += __hip_atomic_load(&ws[z * slice_stride + idx],
+                                   __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
         } else {
           v = acc[fm][fn][r];
         }
