@@ -11,11 +11,13 @@
 // Performance features (driven by profiles/r01_mnist_n1_kernel_stats.txt,
 // where the un-split fwd GEMM ran 4 workgroups on a 256-CU chip):
 //  * split-K: grid.z slices each store their fp32 partial tile into a
-//    per-slice workspace stripe; the LAST workgroup to arrive at each
-//    output tile (agent-scope arrival counter) sums the stripes in fixed
-//    slice order and runs the fused bias/activation epilogue — one
-//    kernel, no separate zero/reduce launches, and DETERMINISTIC
-//    (fixed-order fp32 sums, no atomics on the data path).
+//    per-slice workspace stripe (plain stores, no atomics); a small
+//    vectorized reduce kernel then sums the stripes in fixed slice order
+//    and applies the bias/activation epilogue. Two launches, but the
+//    reduce is coalesced cached f32x4 traffic and the kernel boundary is
+//    the acquire fence — DETERMINISTIC and memory-model-clean (an
+//    in-kernel last-arriver tail measured 3x slower: its uncached
+//    per-element loads serialize).
 //  * vectorized LDS staging: b128 loads when the operand's leading dim
 //    and base allow, b32 otherwise; both [X,K] and [K,X] storage orders
 //    have contiguous global access patterns.
@@ -143,12 +145,10 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
   if (CS && blockIdx.y == 0 && t < BN && tn0 + t < N)
     colsum_out[tn0 + t] = cs_acc;
 
-  __shared__ int is_last;
-  const long slice_stride = (long)M * ldc;
   if (SK) {
-    // store this slice's fp32 partial tile into its workspace stripe
-    // (plain per-slice stores -> fixed-order summation is deterministic)
-    float* wslice = ws + (long)blockIdx.z * slice_stride;
+    // store this slice's fp32 partial tile into its workspace stripe;
+    // the follow-up reduce kernel sums stripes in fixed slice order
+    float* wslice = ws + (long)blockIdx.z * (long)M * ldc;
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
@@ -158,21 +158,10 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + r;
-          if (row < M)
-            __hip_atomic_store(&wslice[(long)row * ldc + col],
-                               acc[fm][fn][r], __ATOMIC_RELAXED,
-                               __HIP_MEMORY_SCOPE_AGENT);
+          if (row < M) wslice[(long)row * ldc + col] = acc[fm][fn][r];
         }
       }
-    __threadfence();
-    if (t == 0) {
-      const int tile = blockIdx.y * gridDim.x + blockIdx.x;
-      int old = __hip_atomic_fetch_add(&cnt[tile], 1, __ATOMIC_ACQ_REL,
-                                       __HIP_MEMORY_SCOPE_AGENT);
-      is_last = (old == (int)gridDim.z - 1);
-    }
-    __syncthreads();
-    if (!is_last) return;
+    return;
   }
 
   // epilogue: (last WG per tile when SK) bias + activation + store
@@ -191,16 +180,7 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
         const int row = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + r;
         if (row >= M) continue;
         const long idx = (long)row * ldc + col;
-        float v;
-        if (SK) {
-          v = 0.f;
-          for (int z = 0; z < (int)gridDim.z; ++z)
-            v += __hip_atomic_load(&ws[z * slice_stride + idx],
-                                   __ATOMIC_RELAXED,
-                                   __HIP_MEMORY_SCOPE_AGENT);
-        } else {
-          v = acc[fm][fn][r];
-        }
+        float v = acc[fm][fn][r];
         v += bv;
         if (ACT == 1) v = v > 0.f ? v : 0.f;
         if (ACT == 2) v = (float)aux[idx] > 0.f ? v : 0.f;
@@ -211,9 +191,42 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
       }
     }
   }
-  if (SK && t == 0)
-    __hip_atomic_store(&cnt[blockIdx.y * gridDim.x + blockIdx.x], 0,
-                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+// split-K phase 2: out[i] = act(sum_z ws[z][i] + bias), f32x4-vectorized
+// over the flattened [M,ldc] output (ldc == N, contiguous).
+template <int ACT, bool BIAS, bool OUTF32>
+__global__ __launch_bounds__(256)
+void splitk_reduce_kernel(const float* __restrict__ ws, const void* __restrict__ bias,
+                          bool bias_bf16, void* __restrict__ Cout, long mn,
+                          int ldc, int nslice) {
+  const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (i0 >= mn) return;
+  f32x4 v = {};
+  for (int z = 0; z < nslice; ++z) {
+    const float* s = ws + (long)z * mn + i0;
+    if (i0 + 4 <= mn) {
+      const f32x4 sv = *(const f32x4*)s;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] += sv[j];
+    } else {
+      for (int j = 0; i0 + j < mn; ++j) v[j] += s[j];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const long i = i0 + j;
+    if (i >= mn) break;
+    float x = v[j];
+    if (BIAS) {
+      const int col = (int)(i % ldc);
+      x += bias_bf16 ? (float)((const __bf16*)bias)[col]
+                     : ((const float*)bias)[col];
+    }
+    if (ACT == 1) x = x > 0.f ? x : 0.f;
+    if (OUTF32) ((float*)Cout)[i] = x;
+    else ((__bf16*)Cout)[i] = (__bf16)x;
+  }
 }
 
 }  // namespace
@@ -246,19 +259,27 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
     return;
   }
   if (sk) {
-#define SK_ACT(ACTv)                                                        \
-    do {                                                                    \
-      if (has_bias) { if (out_f32) LAUNCH(false, false, ACTv, true, true,   \
-                                          true, false);                     \
-                      else LAUNCH(false, false, ACTv, true, false, true,    \
-                                  false); }                                 \
-      else          { if (out_f32) LAUNCH(false, false, ACTv, false, true,  \
-                                          true, false);                     \
-                      else LAUNCH(false, false, ACTv, false, false, true,   \
-                                  false); }                                 \
-    } while (0)
-    if (act == 1) SK_ACT(1); else SK_ACT(0);
-#undef SK_ACT
+    // phase 1: partials (epilogue template args unused in SK stores)
+    LAUNCH(false, false, 0, false, false, true, false);
+    // phase 2: fixed-order stripe reduce + fused epilogue
+    const long mn = (long)M * ldc;
+    dim3 rgrid((unsigned)((mn / 4 + 255) / 256)), rblock(256);
+#define RLAUNCH(ACTv, BIASv, OUTv)                                          \
+    hipLaunchKernelGGL((splitk_reduce_kernel<ACTv, BIASv, OUTv>), rgrid,    \
+                       rblock, 0, stream, ws, bias, bias_bf16, C, mn, ldc,  \
+                       nslice)
+    if (act == 1) {
+      if (has_bias) { if (out_f32) RLAUNCH(1, true, true);
+                      else RLAUNCH(1, true, false); }
+      else          { if (out_f32) RLAUNCH(1, false, true);
+                      else RLAUNCH(1, false, false); }
+    } else {
+      if (has_bias) { if (out_f32) RLAUNCH(0, true, true);
+                      else RLAUNCH(0, true, false); }
+      else          { if (out_f32) RLAUNCH(0, false, true);
+                      else RLAUNCH(0, false, false); }
+    }
+#undef RLAUNCH
     return;
   }
 #define DISP_OUT(TAv, TBv, ACTv, BIASv)                                     \
