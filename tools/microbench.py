@@ -6,7 +6,11 @@ synced once) and compares the GEMMs against torch.matmul (hipBLASLt).
 Run: gpurun -- 'python tools/microbench.py > gpurun_out/micro.txt'
 """
 
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
