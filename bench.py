@@ -41,7 +41,7 @@ def run_mnist(args, device, rank, world):
     model = MnistMLP(hidden_units=args.hidden)
     trainer = SyncReplicaTrainer(
         model.init_params(), optimizer=args.optimizer,
-        hparams={"lr": args.lr}, device=device)
+        hparams={"lr": args.lr}, device=device, n_ps=args.num_ps)
     roles = trainer.roles
 
     act_dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
@@ -92,6 +92,8 @@ def main():
     p.add_argument("--batch", type=int, default=100)
     p.add_argument("--hidden", type=int, default=100)
     p.add_argument("--optimizer", default="sgd")
+    p.add_argument("--num-ps", type=int, default=1,
+                   help="PS shards (ranks 0..n_ps-1) when world>1")
     p.add_argument("--lr", type=float, default=0.01)
     p.add_argument("--nmf-n", type=int, default=1000)
     p.add_argument("--nmf-rank", type=int, default=200)

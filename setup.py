@@ -28,6 +28,9 @@ setup(
     name="tfmesos_amd",
     version="0.1.0",
     packages=find_packages(exclude=["tests"]),
+    entry_points={
+        "console_scripts": ["tfa_run = tfmesos_amd.cli:main"],
+    },
     ext_modules=[
         CUDAExtension(
             name="tfmesos_amd._C",

@@ -1,8 +1,8 @@
 """Helper process for distributed replica tests (gloo, CPU).
 
 Usage: RANK/WORLD_SIZE/MASTER_ADDR/MASTER_PORT in env;
-argv: <mode sync|async> <steps> <out_path_prefix>
-Rank 0 (ps) saves final master params to <prefix>.pt after training.
+argv: <mode sync|async> <steps> <out_path_prefix> [n_ps]
+Rank 0 saves final (shard-synced) master params to <prefix>.pt.
 """
 
 import os
@@ -19,9 +19,11 @@ from tfmesos_amd.ps.replica import (  # noqa: E402
 
 def main():
     mode, steps, prefix = sys.argv[1], int(sys.argv[2]), sys.argv[3]
+    n_ps = int(sys.argv[4]) if len(sys.argv) > 4 else 1
     model = MnistMLP()
     trainer = SyncReplicaTrainer(model.init_params(), optimizer="sgd",
-                                 hparams={"lr": 0.1}, device="cpu")
+                                 hparams={"lr": 0.1}, device="cpu",
+                                 n_ps=n_ps)
     roles = trainer.roles
     # every worker gets the same batch as the single-process reference
     x, y = synthetic_batch(50, seed=42)
@@ -44,10 +46,15 @@ def main():
     else:
         raise SystemExit("bad mode")
 
-    if roles.is_ps:
+    trainer.sync_masters()
+    if trainer.rank == 0:
         torch.save({n: trainer.store.view(n).clone()
                     for n in trainer.store.names}, prefix + ".pt")
         torch.save(trainer.store.global_step, prefix + ".step")
+    if trainer.world > 1:
+        import torch.distributed as dist
+        dist.barrier()
+        dist.destroy_process_group()
 
 
 if __name__ == "__main__":

@@ -16,7 +16,7 @@ HERE = os.path.dirname(os.path.abspath(__file__))
 REPO = os.path.dirname(HERE)
 
 
-def _spawn_world(world, mode, steps, prefix):
+def _spawn_world(world, mode, steps, prefix, n_ps=1):
     from tfmesos_amd.utils import free_port
     port = free_port()
     procs = []
@@ -29,7 +29,7 @@ def _spawn_world(world, mode, steps, prefix):
         })
         procs.append(subprocess.Popen(
             [sys.executable, os.path.join(HERE, "_replica_proc.py"),
-             mode, str(steps), prefix], env=env))
+             mode, str(steps), prefix, str(n_ps)], env=env))
     for p in procs:
         assert p.wait(timeout=180) == 0
 
@@ -85,6 +85,29 @@ def test_async_world2_trains(tmp_path):
     assert step == 5
     got = torch.load(prefix + ".pt", weights_only=True)
     # async with 1 worker == sync with 1 worker == single-process ref
+    want = _single_process_reference(5)
+    for n in want:
+        assert torch.allclose(got[n], want[n], atol=1e-6), n
+
+
+@pytest.mark.timeout(240)
+def test_sync_two_ps_shards_match_single_process(tmp_path):
+    """2 PS shards + 1 worker: sharded reduce/apply/broadcast must equal
+    the single-process reference exactly (multi-PS sharding, SURVEY.md
+    §2c strategy 5)."""
+    prefix = str(tmp_path / "w3ps2")
+    _spawn_world(3, "sync", 5, prefix, n_ps=2)
+    got = torch.load(prefix + ".pt", weights_only=True)
+    want = _single_process_reference(5)
+    for n in want:
+        assert torch.allclose(got[n], want[n], atol=1e-6), n
+
+
+@pytest.mark.timeout(240)
+def test_async_two_ps_shards(tmp_path):
+    prefix = str(tmp_path / "waps2")
+    _spawn_world(3, "async", 5, prefix, n_ps=2)
+    got = torch.load(prefix + ".pt", weights_only=True)
     want = _single_process_reference(5)
     for n in want:
         assert torch.allclose(got[n], want[n], atol=1e-6), n

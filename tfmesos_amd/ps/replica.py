@@ -4,13 +4,16 @@ Replaces the reference's TF PS data plane for replica mode (grad push +
 param pull per step over gRPC, reference ``examples/mnist/
 mnist_replica.py:116-157``) with collectives sized for the xGMI fabric:
 
-* **push** = one ``dist.reduce`` of a single flat fp32 gradient buffer
-  onto the PS rank (every worker has a DIRECT xGMI link to the PS GPU,
-  so the asymmetric PS pattern maps well to the point-to-point fabric);
-* **apply** = ONE fused HIP kernel over the flat master buffer
+* **push** = one ``dist.reduce`` per PS shard of the flat fp32 gradient
+  buffer onto that shard's PS rank (every worker has a DIRECT xGMI link
+  to every PS GPU, so the asymmetric PS pattern maps well to the
+  point-to-point fabric, and multiple shards spread root bandwidth
+  across links — the xGMI-idiomatic form of the reference's multi-ps
+  variable sharding, ``examples/mnist/mnist.py:43``);
+* **apply** = ONE fused HIP kernel per shard over the flat master buffer
   (optimizer update + bf16 shadow refresh in the same pass);
-* **pull** = one ``dist.broadcast`` of the flat bf16 parameter buffer
-  from the PS rank.
+* **pull** = one ``dist.broadcast`` per shard of the flat bf16 parameter
+  buffer from that shard's PS rank.
 
 Sync mode replicates ``tf.train.SyncReplicasOptimizer`` semantics
 (aggregate exactly N worker gradients, apply once, all workers see the
@@ -20,11 +23,12 @@ worker per step, stale gradients impossible by construction), and the
 broadcast is the token release.
 
 Async mode (the reference's default ``tfrun`` mode) uses point-to-point
-isend/irecv pairs per worker with apply-on-arrival at the PS.
+send/recv pairs per (ps, worker) with apply-on-arrival at the PS.
 
 Role mapping when launched by torchrun/bench (one rank per GPU):
-world==1 -> colocated ps+worker; world>1 -> rank 0 = ps, 1..W-1 workers.
-When launched by the tfmesos_amd launcher, roles come from TFA_JOB_NAME.
+world==1 -> colocated ps+worker; world>1 -> ranks [0, n_ps) are PS
+shards, the rest workers. When launched by the tfmesos_amd launcher,
+roles come from TFA_JOB_NAME/TFA_NUM_PS.
 """
 
 import os
@@ -59,53 +63,94 @@ def init_distributed(device=None):
 
 
 class Roles(object):
-    def __init__(self, rank, world):
+    """Rank -> role mapping. Ranks [0, n_ps) are PS shards (reference:
+    ps tasks come first in the jobs_def and hold param slices), the rest
+    are workers. world==1 colocates one ps and one worker."""
+
+    def __init__(self, rank, world, n_ps=1):
         self.rank = rank
         self.world = world
         if world == 1:
+            self.n_ps = 1
             self.is_ps = True
             self.is_worker = True
             self.n_workers = 1
             self.ps_rank = 0
+            self.ps_ranks = [0]
             self.worker_index = 0
         else:
+            n_ps = max(1, min(int(n_ps), world - 1))
+            self.n_ps = n_ps
+            self.ps_ranks = list(range(n_ps))
             self.ps_rank = 0
-            self.is_ps = rank == 0
-            self.is_worker = rank != 0
-            self.n_workers = world - 1
-            self.worker_index = rank - 1 if rank > 0 else -1
+            self.is_ps = rank < n_ps
+            self.is_worker = rank >= n_ps
+            self.n_workers = world - n_ps
+            self.worker_index = rank - n_ps if rank >= n_ps else -1
+
+    @property
+    def worker_ranks(self):
+        if self.world == 1:
+            return [0]
+        return list(range(self.n_ps, self.world))
 
     def describe(self):
         if self.world == 1:
             return "ps1+w1(colocated)"
-        return "ps1+w%d" % self.n_workers
+        return "ps%d+w%d" % (self.n_ps, self.n_workers)
+
+
+def _shard_ranges(total, n_ps, align=256):
+    """Even, 256-element-aligned contiguous slices of the flat buffer."""
+    bounds = [0]
+    for i in range(1, n_ps):
+        b = (total * i // n_ps + align - 1) // align * align
+        bounds.append(min(b, total))
+    bounds.append(total)
+    return [(bounds[i], bounds[i + 1]) for i in range(n_ps)]
 
 
 class SyncReplicaTrainer(object):
-    """Dense sync PS trainer over one flat buffer."""
+    """Dense sync PS trainer over one flat buffer, sharded across n_ps."""
 
     def __init__(self, params, optimizer="sgd", hparams=None, device="cpu",
-                 grad_dtype=torch.float32):
+                 grad_dtype=torch.float32, n_ps=None):
         """params: dict/list of (name, fp32 tensor) — identical on all
         ranks (same seed)."""
         self.device = torch.device(device)
         self.rank, self.world = init_distributed(device)
-        self.roles = Roles(self.rank, self.world)
+        if n_ps is None:
+            n_ps = int(_env("TFA_NUM_PS", "1"))
+        self.roles = Roles(self.rank, self.world, n_ps=n_ps)
         hparams = dict(hparams or {})
 
-        # every rank computes the same flat layout; only the PS rank
-        # keeps optimizer state
+        # every rank computes the same flat layout; only PS ranks apply
+        # (and then only their own shard)
         self.store = PStore(device=device)
         self.store.init_params(params, optimizer=optimizer, **hparams)
         if not self.roles.is_ps:
             self.store.state = {}  # workers don't need optimizer state
 
+        self.shards = _shard_ranges(self.store.flat.numel(), self.roles.n_ps)
+        if self.roles.is_ps and self.world > 1:
+            self.my_shard = self.shards[self.rank]
+        elif self.world == 1:
+            self.my_shard = self.shards[0]
+        else:
+            self.my_shard = None
+
         self.flat_params_bf16 = self.store.flat_bf16
         self.flat_grad = torch.zeros_like(self.store.flat, dtype=grad_dtype)
 
         # initial pull so every worker starts from the PS masters
-        if self.world > 1:
-            dist.broadcast(self.flat_params_bf16, src=self.roles.ps_rank)
+        self._broadcast_params()
+
+    def _broadcast_params(self):
+        if self.world == 1:
+            return
+        for i, (lo, hi) in enumerate(self.shards):
+            dist.broadcast(self.flat_params_bf16[lo:hi],
+                           src=self.roles.ps_ranks[i])
 
     # parameter views (bf16 working copies, refreshed in-place by pull)
     def param(self, name):
@@ -119,7 +164,7 @@ class SyncReplicaTrainer(object):
         self.flat_grad.zero_()
 
     def step(self, grad_scale=None):
-        """One global step: reduce grads -> PS apply -> broadcast params.
+        """One global step: reduce grads per shard -> PS apply -> broadcast.
 
         The worker must have filled ``flat_grad`` (via ``grad_view``)
         before calling. grad_scale defaults to 1/n_workers (grad mean,
@@ -128,25 +173,51 @@ class SyncReplicaTrainer(object):
         scale = grad_scale if grad_scale is not None \
             else 1.0 / self.roles.n_workers
         if self.world > 1:
-            if self.roles.is_ps and not self.roles.is_worker:
+            if self.roles.is_ps:
                 # dist.reduce is in-place: the PS buffer holds last step's
                 # sum and would be re-added — contribute zeros instead
                 self.flat_grad.zero_()
-            dist.reduce(self.flat_grad, dst=self.roles.ps_rank, op=dist.ReduceOp.SUM)
-        if self.roles.is_ps:
-            self.store.apply_flat(self.flat_grad, grad_scale=scale)
+            for i, (lo, hi) in enumerate(self.shards):
+                dist.reduce(self.flat_grad[lo:hi],
+                            dst=self.roles.ps_ranks[i], op=dist.ReduceOp.SUM)
+        if self.my_shard is not None:
+            lo, hi = self.my_shard
+            self.store.apply_flat(self.flat_grad, grad_scale=scale,
+                                  lo=lo, hi=hi)
         if self.world > 1:
-            dist.broadcast(self.flat_params_bf16, src=self.roles.ps_rank)
+            for i, (lo, hi) in enumerate(self.shards):
+                dist.broadcast(self.flat_params_bf16[lo:hi],
+                               src=self.roles.ps_ranks[i])
         return self.store.global_step
 
+    def sync_masters(self):
+        """Broadcast every shard's fp32 masters so ALL ranks hold the
+        complete master buffer (e.g. for a full checkpoint or eval)."""
+        if self.world > 1:
+            for i, (lo, hi) in enumerate(self.shards):
+                dist.broadcast(self.store.flat[lo:hi],
+                               src=self.roles.ps_ranks[i])
+        return self.store.flat
+
     def save(self, path):
-        if self.roles.is_ps:
+        """PS-side checkpoint. Single PS writes ``path``; each shard of a
+        multi-PS run writes ``path.ps<i>`` (its own masters + opt state).
+        The reference delegated this to tf.train.Supervisor's chief
+        (``mnist_replica.py:165-183``); here the PS owns its state."""
+        if self.world == 1 or (self.roles.is_ps and self.roles.n_ps == 1):
             self.store.save(path)
+        elif self.roles.is_ps:
+            self.store.save("%s.ps%d" % (path, self.rank), *self.my_shard)
 
     def load(self, path):
-        self.store.load(path)
-        if self.world > 1:
-            dist.broadcast(self.flat_params_bf16, src=self.roles.ps_rank)
+        """Restore and re-broadcast (others wait at the broadcast — the
+        chief-inits/others-wait semantic of the reference Supervisor)."""
+        if self.world == 1 or self.roles.n_ps == 1:
+            if self.roles.is_ps or self.world == 1:
+                self.store.load(path)
+        elif self.roles.is_ps:
+            self.store.load("%s.ps%d" % (path, self.rank), *self.my_shard)
+        self._broadcast_params()
 
 
 def make_pair_groups(roles):
@@ -156,8 +227,9 @@ def make_pair_groups(roles):
     group, which is safe on both gloo and RCCL (one communicator per
     thread)."""
     groups = {}
-    for w in range(1, roles.world):
-        groups[w] = dist.new_group([roles.ps_rank, w])
+    for p in roles.ps_ranks:
+        for w in roles.worker_ranks:
+            groups[(p, w)] = dist.new_group([p, w])
     return groups
 
 
@@ -170,13 +242,19 @@ class AsyncPSWorker(object):
 
     def __init__(self, trainer, pair_groups):
         self.t = trainer
-        assert trainer.world > 1, "async mode needs a separate PS rank"
-        self.group = pair_groups[trainer.rank]
+        assert trainer.roles.is_worker and trainer.world > 1, \
+            "async mode needs separate PS rank(s)"
+        self.groups = pair_groups
 
     def step(self):
         t = self.t
-        dist.send(t.flat_grad, dst=t.roles.ps_rank, group=self.group)
-        dist.recv(t.flat_params_bf16, src=t.roles.ps_rank, group=self.group)
+        for i, (lo, hi) in enumerate(t.shards):
+            p = t.roles.ps_ranks[i]
+            dist.send(t.flat_grad[lo:hi], dst=p, group=self.groups[(p, t.rank)])
+        for i, (lo, hi) in enumerate(t.shards):
+            p = t.roles.ps_ranks[i]
+            dist.recv(t.flat_params_bf16[lo:hi], src=p,
+                      group=self.groups[(p, t.rank)])
         return True
 
     def stop(self):
@@ -185,7 +263,8 @@ class AsyncPSWorker(object):
 
 
 class AsyncPSServer(object):
-    """PS side: one serving thread per worker, apply-on-arrival."""
+    """PS-shard side: one serving thread per worker, apply-on-arrival
+    over this rank's shard only."""
 
     def __init__(self, trainer, pair_groups):
         self.t = trainer
@@ -194,20 +273,24 @@ class AsyncPSServer(object):
 
     def _serve_one(self, worker_rank, steps):
         t = self.t
-        buf = torch.zeros_like(t.flat_grad)
-        group = self.groups[worker_rank]
+        lo, hi = t.my_shard
+        buf = torch.zeros(hi - lo, dtype=t.flat_grad.dtype,
+                          device=t.flat_grad.device)
+        group = self.groups[(t.rank, worker_rank)]
+        full = torch.zeros_like(t.flat_grad)
         for _ in range(steps):
             dist.recv(buf, src=worker_rank, group=group)
             with t.store.lock:
-                t.store.apply_flat(buf)
-                params = t.store.flat_bf16.clone()
+                full[lo:hi].copy_(buf)
+                t.store.apply_flat(full, lo=lo, hi=hi)
+                params = t.store.flat_bf16[lo:hi].clone()
             dist.send(params, dst=worker_rank, group=group)
 
     def serve(self, steps_per_worker):
         import threading
         threads = [
             threading.Thread(target=self._serve_one, args=(w, steps_per_worker))
-            for w in range(1, self.t.world)
+            for w in self.t.roles.worker_ranks
         ]
         for th in threads:
             th.start()

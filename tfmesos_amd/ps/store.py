@@ -121,64 +121,89 @@ class PStore(object):
             self.apply_flat(flat_g)
             return self.global_step
 
-    def apply_flat(self, flat_grad, grad_scale=1.0):
-        """One fused optimizer apply over the whole flat buffer; also
-        refreshes the bf16 shadow in the same kernel. grad_scale folds
-        the sync-replica worker-mean into the kernel."""
+    def apply_flat(self, flat_grad, grad_scale=1.0, lo=0, hi=None):
+        """One fused optimizer apply over the flat buffer (or the [lo,hi)
+        slice of it — a multi-PS shard); also refreshes the bf16 shadow
+        in the same kernel. grad_scale folds the sync-replica worker-mean
+        into the kernel."""
         with self.lock:
             self.global_step += 1
             hp = self.hparams
             lr = hp.get("lr", 0.01)
+            hi = self.flat.numel() if hi is None else hi
+            p = self.flat[lo:hi]
+            g = flat_grad[lo:hi]
+            bf = self.flat_bf16[lo:hi]
+            st = {k: v[lo:hi] for k, v in self.state.items()}
             if self.opt == "sgd":
-                ops.fused_sgd(self.flat, flat_grad, lr,
+                ops.fused_sgd(p, g, lr,
                               momentum=hp.get("momentum", 0.0),
                               weight_decay=hp.get("weight_decay", 0.0),
-                              momentum_buf=self.state.get("momentum_buf"),
-                              bf16_out=self.flat_bf16, grad_scale=grad_scale)
+                              momentum_buf=st.get("momentum_buf"),
+                              bf16_out=bf, grad_scale=grad_scale)
             elif self.opt == "adam":
-                ops.fused_adam(self.flat, flat_grad,
-                               self.state["exp_avg"], self.state["exp_avg_sq"],
+                ops.fused_adam(p, g, st["exp_avg"], st["exp_avg_sq"],
                                self.global_step, lr,
                                beta1=hp.get("beta1", 0.9),
                                beta2=hp.get("beta2", 0.999),
                                eps=hp.get("eps", 1e-8),
                                weight_decay=hp.get("weight_decay", 0.0),
-                               bf16_out=self.flat_bf16, grad_scale=grad_scale)
+                               bf16_out=bf, grad_scale=grad_scale)
             elif self.opt == "adagrad":
-                ops.fused_adagrad(self.flat, flat_grad, self.state["accum"],
+                ops.fused_adagrad(p, g, st["accum"],
                                   lr, eps=hp.get("eps", 1e-10),
                                   weight_decay=hp.get("weight_decay", 0.0),
-                                  bf16_out=self.flat_bf16, grad_scale=grad_scale)
+                                  bf16_out=bf, grad_scale=grad_scale)
             else:
                 raise ValueError("unknown optimizer %r" % self.opt)
 
     # ---------------------------------------------------------- checkpoint
 
-    def save(self, path):
+    def save(self, path, lo=0, hi=None):
         """PS-side checkpoint: params + optimizer state + step (the
         reference delegated this to tf.train.Supervisor's chief; here the
-        PS owns it — SURVEY.md §5 checkpoint/resume)."""
+        PS owns it — SURVEY.md §5 checkpoint/resume). With lo/hi, saves
+        only that shard's slice (multi-PS: each shard owns its slice)."""
         with self.lock:
+            full = lo == 0 and (hi is None or hi == self.flat.numel())
+            hi = self.flat.numel() if hi is None else hi
             torch.save({
                 "names": self.names,
                 "shapes": self.shapes,
                 "offsets": self.offsets,
-                "flat": self.flat.cpu(),
+                "shard": None if full else (lo, hi),
+                "flat": self.flat[lo:hi].cpu(),
                 "opt": self.opt,
                 "hparams": self.hparams,
-                "state": {k: v.cpu() for k, v in self.state.items()},
+                "state": {k: v[lo:hi].cpu() for k, v in self.state.items()},
                 "global_step": self.global_step,
             }, path)
 
-    def load(self, path):
+    def load(self, path, lo=0, hi=None):
+        """Restore a full checkpoint, or copy a shard checkpoint's slice
+        into the (already initialized) buffers."""
         with self.lock:
             ck = torch.load(path, map_location="cpu", weights_only=True)
-            self.names = ck["names"]
-            self.shapes = ck["shapes"]
-            self.offsets = ck["offsets"]
-            self.flat = ck["flat"].to(self.device)
-            self.flat_bf16 = self.flat.to(torch.bfloat16)
-            self.opt = ck["opt"]
-            self.hparams = ck["hparams"]
-            self.state = {k: v.to(self.device) for k, v in ck["state"].items()}
+            shard = ck.get("shard")
+            if shard is None and lo == 0 and hi is None:
+                self.names = ck["names"]
+                self.shapes = ck["shapes"]
+                self.offsets = ck["offsets"]
+                self.flat = ck["flat"].to(self.device)
+                self.flat_bf16 = self.flat.to(torch.bfloat16)
+                self.opt = ck["opt"]
+                self.hparams = ck["hparams"]
+                self.state = {k: v.to(self.device)
+                              for k, v in ck["state"].items()}
+                self.global_step = ck["global_step"]
+                return
+            cklo, ckhi = shard if shard is not None else (lo, hi)
+            if self.flat is None:
+                raise RuntimeError("shard checkpoint needs an initialized "
+                                   "store (call init_params first)")
+            self.flat[cklo:ckhi].copy_(ck["flat"].to(self.device))
+            self.flat_bf16[cklo:ckhi].copy_(
+                self.flat[cklo:ckhi].to(torch.bfloat16))
+            for k, v in ck["state"].items():
+                self.state[k][cklo:ckhi].copy_(v.to(self.device))
             self.global_step = ck["global_step"]
