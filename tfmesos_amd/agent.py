@@ -120,6 +120,7 @@ def main(argv):
         "TFA_JOB_NAME": job_name,
         "TFA_TASK_INDEX": str(task_index),
         "TFA_DISTRIBUTED": "yes",
+        "TFA_NUM_PS": str(len(cluster_def.get("ps", [])) or 1),
         "TFA_RANK": str(config.get("rank", 0)),
         "TFA_WORLD_SIZE": str(config.get("world_size", 1)),
         "TFA_CLUSTER_DEF": json.dumps(cluster_def),
