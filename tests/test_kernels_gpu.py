@@ -334,3 +334,78 @@ def test_native_ext_is_mandatory_on_gpu(monkeypatch):
     x = torch.randn(4, 4, device=DEV, dtype=torch.bfloat16)
     with pytest.raises(RuntimeError, match="extension not built"):
         ops.gemm_bias_act(x, x)
+
+
+# ------------------------------------------------------------------- conv
+
+@pytest.mark.parametrize("shape", [
+    # (N, C, H, W, K, R, S, stride, pad)
+    (2, 3, 16, 16, 8, 3, 3, 1, 1),
+    (2, 8, 15, 15, 16, 3, 3, 2, 0),      # stride 2, odd size
+    (2, 16, 8, 8, 32, 1, 1, 1, 0),       # pointwise
+    (1, 4, 12, 12, 6, 1, 7, 1, 3),       # asymmetric 1x7 (Inception)
+    (1, 4, 12, 12, 6, 7, 1, 1, 3),       # asymmetric 7x1 — pad clamps
+    (2, 3, 31, 31, 8, 3, 3, 2, 0),       # Inception stem-ish
+])
+def test_conv2d_fwd_matches_torch(shape):
+    from tfmesos_amd import ops
+    N, C, H, W, K, R, S, st, pd = shape
+    pads = (pd if R > 1 else 0, pd if S > 1 else 0)
+    torch.manual_seed(hash(shape) % 2**31)
+    x = torch.randn(N, C, H, W)
+    w = torch.randn(K, C, R, S) * 0.2
+    b = torch.randn(K)
+    xb, wb = bf(x), bf(w)
+    y = ops.conv2d(xb, wb, b.to(DEV), stride=st, padding=pads)
+    ref = torch.nn.functional.conv2d(xb.float().cpu(), wb.float().cpu(), b,
+                                     stride=st, padding=pads)
+    err = (y.float().cpu() - ref).abs().max()
+    tol = 0.03 * ref.abs().max() + 0.06
+    assert err < tol, (err, tol)
+
+
+def test_conv2d_bwd_matches_torch():
+    from tfmesos_amd import ops
+    torch.manual_seed(77)
+    N, C, H, W, K, R, S, st, pd = 2, 8, 14, 14, 16, 3, 3, 2, 1
+    x = bf(torch.randn(N, C, H, W)).requires_grad_(True)
+    w = bf(torch.randn(K, C, R, S) * 0.2).requires_grad_(True)
+    y = ops.conv2d(x, w, None, stride=st, padding=pd)
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    xf = x.detach().float().cpu().requires_grad_(True)
+    wf = w.detach().float().cpu().requires_grad_(True)
+    ref = torch.nn.functional.conv2d(xf, wf, stride=st, padding=pd)
+    ref.backward(g.float().cpu())
+
+    dxe = (x.grad.float().cpu() - xf.grad).abs().max()
+    dwe = (w.grad.float().cpu() - wf.grad).abs().max()
+    assert dxe < 0.03 * xf.grad.abs().max() + 0.06, dxe
+    assert dwe < 0.03 * wf.grad.abs().max() + 0.3, dwe
+
+
+def test_conv2d_train_step_decreases_loss():
+    from tfmesos_amd import ops
+    torch.manual_seed(78)
+    x = bf(torch.randn(4, 3, 16, 16))
+    tgt = torch.randint(0, 4, (4,), device=DEV)
+    w1 = bf(torch.randn(8, 3, 3, 3) * 0.2).requires_grad_(True)
+    wf = bf(torch.randn(4, 8 * 8 * 8) * 0.05).requires_grad_(True)
+
+    def loss_fn():
+        h = torch.relu(ops.conv2d(x, w1, None, stride=2, padding=1))
+        logits = h.reshape(4, -1).float() @ wf.float().t()
+        return torch.nn.functional.cross_entropy(logits, tgt)
+
+    l0 = None
+    for _ in range(12):
+        loss = loss_fn()
+        if l0 is None:
+            l0 = float(loss)
+        loss.backward()
+        with torch.no_grad():
+            for p in (w1, wf):
+                p -= 0.1 * p.grad
+                p.grad = None
+    assert float(loss_fn()) < l0

@@ -248,3 +248,61 @@ def relu_bwd(grad_out, act):
     if grad_out.is_cuda:
         return _ext().relu_bwd(grad_out, act)
     return (grad_out.float() * (act.float() > 0)).to(grad_out.dtype)
+
+
+# --------------------------------------------------------------------- conv
+
+class _Conv2dFn(torch.autograd.Function):
+    """Autograd wrapper over the implicit-GEMM HIP conv kernels (NCHW,
+    bf16 activations/weights, fp32 weight grads). The Inception path
+    (BASELINE.json conv config) builds on this; CPU falls back to the
+    torch fp32 reference so model code runs in CI."""
+
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, padding):
+        ctx.save_for_backward(x, w)
+        ctx.stride = stride
+        ctx.padding = padding
+        ctx.has_bias = bias is not None
+        if x.is_cuda:
+            eb = bias.float() if bias is not None else \
+                torch.empty(0, device=x.device)
+            return _ext().conv2d_fwd(x, w, eb, stride[0], stride[1],
+                                     padding[0], padding[1], False)
+        y = torch.nn.functional.conv2d(
+            x.float(), w.float(), bias.float() if bias is not None else None,
+            stride=stride, padding=padding)
+        return y.to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        if x.is_cuda:
+            dx = _ext().conv2d_bwd_data(dy, w, x.shape[2], x.shape[3],
+                                        ctx.stride[0], ctx.stride[1],
+                                        ctx.padding[0], ctx.padding[1])
+            dw = _ext().conv2d_bwd_weight(dy, x, w.shape[2], w.shape[3],
+                                          ctx.stride[0], ctx.stride[1],
+                                          ctx.padding[0], ctx.padding[1])
+            dw = dw.to(w.dtype)
+        else:
+            dyf, xf, wf = dy.float(), x.float(), w.float()
+            dx = torch.nn.grad.conv2d_input(
+                x.shape, wf, dyf, stride=ctx.stride, padding=ctx.padding
+            ).to(x.dtype)
+            dw = torch.nn.grad.conv2d_weight(
+                xf, w.shape, dyf, stride=ctx.stride, padding=ctx.padding
+            ).to(w.dtype)
+        db = dy.float().sum(dim=(0, 2, 3)) if ctx.has_bias else None
+        return dx, dw, db, None, None
+
+
+def conv2d(x, w, bias=None, stride=1, padding=0):
+    """2-D convolution (NCHW): hand-written implicit-GEMM MFMA kernels on
+    GPU (csrc/conv.hip), torch fp32 reference on CPU. Differentiable."""
+    if isinstance(stride, int):
+        stride = (stride, stride)
+    if isinstance(padding, int):
+        padding = (padding, padding)
+    return _Conv2dFn.apply(x, w, bias, tuple(stride), tuple(padding))

@@ -35,6 +35,15 @@ void launch_scatter_add_f32(float*, const long*, const float*, long, int,
 void launch_relu_bwd(const bf16_t*, const bf16_t*, bf16_t*, long,
                      hipStream_t);
 void launch_colsum(const bf16_t*, float*, int, int, hipStream_t);
+void launch_conv_fwd(const bf16_t*, const bf16_t*, const float*, bf16_t*,
+                     int, int, int, int, int, int, int, int, int, int, int,
+                     int, int, bool, hipStream_t);
+void launch_conv_bwd_data(const bf16_t*, const bf16_t*, bf16_t*, int, int,
+                          int, int, int, int, int, int, int, int, int, int,
+                          int, hipStream_t);
+void launch_conv_bwd_weight(const bf16_t*, const bf16_t*, float*, int, int,
+                            int, int, int, int, int, int, int, int, int,
+                            int, int, hipStream_t);
 
 namespace {
 
@@ -347,6 +356,69 @@ torch::Tensor colsum(torch::Tensor x, torch::Tensor out) {
   return out;
 }
 
+// ------------------------------------------------------------------ conv
+
+static void conv_out_shape(const torch::Tensor& x, const torch::Tensor& w,
+                           long stride_h, long stride_w, long pad_h,
+                           long pad_w, int* Ho, int* Wo) {
+  *Ho = (int)((x.size(2) + 2 * pad_h - w.size(2)) / stride_h + 1);
+  *Wo = (int)((x.size(3) + 2 * pad_w - w.size(3)) / stride_w + 1);
+}
+
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
+                         torch::Tensor bias, long stride_h, long stride_w,
+                         long pad_h, long pad_w, bool relu) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous() &&
+              x.scalar_type() == torch::kBFloat16, "x must be bf16 NCHW");
+  TORCH_CHECK(w.is_cuda() && w.dim() == 4 && w.is_contiguous() &&
+              w.scalar_type() == torch::kBFloat16 && w.size(1) == x.size(1),
+              "w must be bf16 KCRS matching x channels");
+  const float* bias_p = nullptr;
+  if (bias.numel() > 0) {
+    TORCH_CHECK(bias.scalar_type() == torch::kFloat32 &&
+                bias.numel() == w.size(0), "bias must be fp32 [K]");
+    bias_p = bias.data_ptr<float>();
+  }
+  int Ho, Wo;
+  conv_out_shape(x, w, stride_h, stride_w, pad_h, pad_w, &Ho, &Wo);
+  auto y = torch::empty({x.size(0), w.size(0), Ho, Wo}, x.options());
+  launch_conv_fwd((const bf16_t*)x.data_ptr(), (const bf16_t*)w.data_ptr(),
+                  bias_p, (bf16_t*)y.data_ptr(), x.size(0), x.size(1),
+                  x.size(2), x.size(3), w.size(0), w.size(2), w.size(3),
+                  Ho, Wo, stride_h, stride_w, pad_h, pad_w, relu,
+                  cur_stream());
+  return y;
+}
+
+torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor w,
+                              long H, long W, long stride_h, long stride_w,
+                              long pad_h, long pad_w) {
+  TORCH_CHECK(dy.is_cuda() && dy.dim() == 4 && dy.is_contiguous() &&
+              dy.scalar_type() == torch::kBFloat16, "dy must be bf16 NCHW");
+  auto dx = torch::empty({dy.size(0), w.size(1), H, W}, dy.options());
+  launch_conv_bwd_data((const bf16_t*)dy.data_ptr(),
+                       (const bf16_t*)w.data_ptr(), (bf16_t*)dx.data_ptr(),
+                       dy.size(0), w.size(1), H, W, w.size(0), w.size(2),
+                       w.size(3), dy.size(2), dy.size(3), stride_h, stride_w,
+                       pad_h, pad_w, cur_stream());
+  return dx;
+}
+
+torch::Tensor conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
+                                long R, long S, long stride_h, long stride_w,
+                                long pad_h, long pad_w) {
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda() && dy.is_contiguous() &&
+              x.is_contiguous(), "dy/x must be contiguous");
+  auto dw = torch::zeros({dy.size(1), x.size(1), R, S},
+                         x.options().dtype(torch::kFloat32));
+  launch_conv_bwd_weight((const bf16_t*)dy.data_ptr(),
+                         (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(),
+                         x.size(0), x.size(1), x.size(2), x.size(3),
+                         dy.size(1), R, S, dy.size(2), dy.size(3), stride_h,
+                         stride_w, pad_h, pad_w, cur_stream());
+  return dw;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -371,6 +443,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_xent_fused", &softmax_xent_fused);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
+  m.def("conv2d_fwd", &conv2d_fwd, "implicit-GEMM conv fwd (MFMA, bf16)");
+  m.def("conv2d_bwd_data", &conv2d_bwd_data);
+  m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
   m.def("embedding_gather", &embedding_gather);
   m.def("embedding_scatter_add", &embedding_scatter_add);
   m.def("relu_bwd", &relu_bwd);
