@@ -82,13 +82,48 @@ def run_nmf(args, device, rank, world):
     }
 
 
+def run_inception(args, device, rank, world):
+    """Inception-v3 distributed train config (BASELINE.json): autograd
+    model on the implicit-GEMM MFMA conv kernels, PS replica data plane
+    (zero-copy flat store wiring)."""
+    import torch.nn.functional  # noqa: F401
+
+    from tfmesos_amd import ops
+    from tfmesos_amd.models.inception import InceptionV3, synthetic_images
+    from tfmesos_amd.ps.module_trainer import ModuleReplicaTrainer
+
+    model = InceptionV3(num_classes=args.classes)
+    trainer = ModuleReplicaTrainer(model, optimizer=args.optimizer,
+                                   hparams={"lr": args.lr}, device=device,
+                                   n_ps=args.num_ps)
+    x, y = synthetic_images(args.inc_batch, size=args.inc_size,
+                            classes=args.classes, device=device,
+                            dtype=torch.bfloat16, seed=1000 + rank)
+
+    def one_step():
+        if trainer.roles.is_worker:
+            trainer.zero_grad()
+            loss = ops.softmax_xent_loss(model(x).contiguous(), y)
+            loss.backward()
+        trainer.step()
+
+    return one_step, {
+        "model": "inception_v3_%d" % args.inc_size,
+        "global_batch": args.inc_batch * trainer.roles.n_workers,
+        "parallelism": trainer.roles.describe(),
+        "optimizer": args.optimizer,
+        "lr": args.lr,
+        "sync": True,
+    }
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=1000)
     p.add_argument("--warmup", type=int, default=100)
     p.add_argument("--workload", default="mnist",
-                   choices=["mnist", "nmf"])
+                   choices=["mnist", "nmf", "inception"])
     p.add_argument("--batch", type=int, default=100)
     p.add_argument("--hidden", type=int, default=100)
     p.add_argument("--optimizer", default="sgd")
@@ -97,6 +132,9 @@ def main():
     p.add_argument("--lr", type=float, default=0.01)
     p.add_argument("--nmf-n", type=int, default=1000)
     p.add_argument("--nmf-rank", type=int, default=200)
+    p.add_argument("--inc-batch", type=int, default=32)
+    p.add_argument("--inc-size", type=int, default=299)
+    p.add_argument("--classes", type=int, default=1000)
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -110,7 +148,8 @@ def main():
         device = torch.device("cpu")
         log("WARNING: no GPU visible; CPU dev run (not a valid benchmark)")
 
-    one_step, config = {"mnist": run_mnist, "nmf": run_nmf}[args.workload](
+    one_step, config = {"mnist": run_mnist, "nmf": run_nmf,
+                        "inception": run_inception}[args.workload](
         args, device, rank, world)
 
     def barrier_sync():

@@ -1,0 +1,48 @@
+"""Inception-v3 on MI355X: full fwd+bwd+PS step through the implicit-GEMM
+MFMA conv kernels (pytest -m gpu)."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.mark.timeout(420)
+def test_inception_train_step_gpu():
+    from tfmesos_amd import ops
+    from tfmesos_amd.models.inception import InceptionV3, synthetic_images
+    from tfmesos_amd.ps.module_trainer import ModuleReplicaTrainer
+
+    model = InceptionV3(num_classes=100)
+    tr = ModuleReplicaTrainer(model, optimizer="sgd",
+                              hparams={"lr": 0.05}, device="cuda:0")
+    x, y = synthetic_images(8, size=299, classes=100, device="cuda:0",
+                            dtype=torch.bfloat16, seed=3)
+    losses = []
+    for _ in range(6):
+        tr.zero_grad()
+        loss = ops.softmax_xent_loss(model(x).contiguous(), y)
+        loss.backward()
+        tr.step()
+        losses.append(float(loss.detach()))
+    torch.cuda.synchronize()
+    assert all(l == l for l in losses), losses           # finite
+    assert losses[-1] < losses[0], losses                # learning
+
+
+@pytest.mark.timeout(300)
+def test_inception_gpu_matches_cpu_forward():
+    """GPU bf16 forward (MFMA convs) tracks the CPU fp32 reference of
+    the same random-init model within bf16 tolerance."""
+    from tfmesos_amd.models.inception import InceptionV3
+
+    m = InceptionV3(num_classes=10)
+    x = torch.rand(2, 3, 128, 128)
+    with torch.no_grad():
+        ref = m(x.clone())
+        m.to("cuda:0", torch.bfloat16)
+        out = m(x.to("cuda:0", torch.bfloat16))
+    # BN amplifies small numeric differences; check correlation not equality
+    r = torch.corrcoef(torch.stack([
+        out.float().cpu().flatten(), ref.flatten()]))[0, 1]
+    assert r > 0.98, r

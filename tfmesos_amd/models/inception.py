@@ -1,0 +1,225 @@
+"""Inception-v3 on the hand-written MFMA conv kernels (BASELINE.json
+config "Inception-v3 distributed train 1-ps/8-worker bf16").
+
+The reference cites inception as its canonical heavyweight workload
+(``README.rst:72``: distributed inception over tfmesos). Architecture
+follows the standard Inception-v3 layout (stem 299x299 -> 5 InceptionA/
+B/C/D/E stages -> 8x8 avgpool -> fc 1000); convolutions run on
+``tfmesos_amd.ops.conv2d`` (implicit-GEMM MFMA, csrc/conv.hip), the
+classifier GEMM on the MFMA GEMM kernel, batch-norm as plain fp32
+tensor math (train-mode batch stats) — no MIOpen, no cuDNN shims.
+"""
+
+import torch
+import torch.nn as nn
+
+from tfmesos_amd import ops
+
+
+class Conv2d(nn.Module):
+    """Conv on the hand-written implicit-GEMM kernel (bias folded into
+    the following BN, as in Inception-v3)."""
+
+    def __init__(self, cin, cout, kernel_size, stride=1, padding=0):
+        super().__init__()
+        if isinstance(kernel_size, int):
+            kernel_size = (kernel_size, kernel_size)
+        self.stride = stride
+        self.padding = padding
+        fan_in = cin * kernel_size[0] * kernel_size[1]
+        self.weight = nn.Parameter(
+            torch.randn(cout, cin, *kernel_size) * (2.0 / fan_in) ** 0.5)
+
+    def forward(self, x):
+        return ops.conv2d(x, self.weight, None, stride=self.stride,
+                          padding=self.padding)
+
+
+class BatchNorm2d(nn.Module):
+    """Train-mode batch-norm in fp32 tensor math (stats over N,H,W)."""
+
+    def __init__(self, num_features, eps=1e-3):
+        super().__init__()
+        self.eps = eps
+        self.weight = nn.Parameter(torch.ones(num_features))
+        self.bias = nn.Parameter(torch.zeros(num_features))
+
+    def forward(self, x):
+        xf = x.float()
+        mean = xf.mean(dim=(0, 2, 3), keepdim=True)
+        var = xf.var(dim=(0, 2, 3), unbiased=False, keepdim=True)
+        y = (xf - mean) * torch.rsqrt(var + self.eps)
+        y = y * self.weight.float().view(1, -1, 1, 1) \
+            + self.bias.float().view(1, -1, 1, 1)
+        return y.to(x.dtype)
+
+
+class BasicConv2d(nn.Module):
+    def __init__(self, cin, cout, **kw):
+        super().__init__()
+        self.conv = Conv2d(cin, cout, **kw)
+        self.bn = BatchNorm2d(cout)
+
+    def forward(self, x):
+        return torch.relu(self.bn(self.conv(x)))
+
+
+def _avg_pool(x, k, stride=1, padding=1):
+    return torch.nn.functional.avg_pool2d(x.float(), k, stride=stride,
+                                          padding=padding).to(x.dtype)
+
+
+def _max_pool(x, k, stride):
+    return torch.nn.functional.max_pool2d(x, k, stride=stride)
+
+
+class InceptionA(nn.Module):
+    def __init__(self, cin, pool_features):
+        super().__init__()
+        self.b1x1 = BasicConv2d(cin, 64, kernel_size=1)
+        self.b5x5_1 = BasicConv2d(cin, 48, kernel_size=1)
+        self.b5x5_2 = BasicConv2d(48, 64, kernel_size=5, padding=2)
+        self.b3x3_1 = BasicConv2d(cin, 64, kernel_size=1)
+        self.b3x3_2 = BasicConv2d(64, 96, kernel_size=3, padding=1)
+        self.b3x3_3 = BasicConv2d(96, 96, kernel_size=3, padding=1)
+        self.bpool = BasicConv2d(cin, pool_features, kernel_size=1)
+
+    def forward(self, x):
+        return torch.cat([
+            self.b1x1(x),
+            self.b5x5_2(self.b5x5_1(x)),
+            self.b3x3_3(self.b3x3_2(self.b3x3_1(x))),
+            self.bpool(_avg_pool(x, 3)),
+        ], 1)
+
+
+class InceptionB(nn.Module):
+    def __init__(self, cin):
+        super().__init__()
+        self.b3x3 = BasicConv2d(cin, 384, kernel_size=3, stride=2)
+        self.b3x3dbl_1 = BasicConv2d(cin, 64, kernel_size=1)
+        self.b3x3dbl_2 = BasicConv2d(64, 96, kernel_size=3, padding=1)
+        self.b3x3dbl_3 = BasicConv2d(96, 96, kernel_size=3, stride=2)
+
+    def forward(self, x):
+        return torch.cat([
+            self.b3x3(x),
+            self.b3x3dbl_3(self.b3x3dbl_2(self.b3x3dbl_1(x))),
+            _max_pool(x, 3, 2),
+        ], 1)
+
+
+class InceptionC(nn.Module):
+    def __init__(self, cin, c7):
+        super().__init__()
+        self.b1x1 = BasicConv2d(cin, 192, kernel_size=1)
+        self.b7_1 = BasicConv2d(cin, c7, kernel_size=1)
+        self.b7_2 = BasicConv2d(c7, c7, kernel_size=(1, 7), padding=(0, 3))
+        self.b7_3 = BasicConv2d(c7, 192, kernel_size=(7, 1), padding=(3, 0))
+        self.b7d_1 = BasicConv2d(cin, c7, kernel_size=1)
+        self.b7d_2 = BasicConv2d(c7, c7, kernel_size=(7, 1), padding=(3, 0))
+        self.b7d_3 = BasicConv2d(c7, c7, kernel_size=(1, 7), padding=(0, 3))
+        self.b7d_4 = BasicConv2d(c7, c7, kernel_size=(7, 1), padding=(3, 0))
+        self.b7d_5 = BasicConv2d(c7, 192, kernel_size=(1, 7), padding=(0, 3))
+        self.bpool = BasicConv2d(cin, 192, kernel_size=1)
+
+    def forward(self, x):
+        return torch.cat([
+            self.b1x1(x),
+            self.b7_3(self.b7_2(self.b7_1(x))),
+            self.b7d_5(self.b7d_4(self.b7d_3(self.b7d_2(self.b7d_1(x))))),
+            self.bpool(_avg_pool(x, 3)),
+        ], 1)
+
+
+class InceptionD(nn.Module):
+    def __init__(self, cin):
+        super().__init__()
+        self.b3_1 = BasicConv2d(cin, 192, kernel_size=1)
+        self.b3_2 = BasicConv2d(192, 320, kernel_size=3, stride=2)
+        self.b7_1 = BasicConv2d(cin, 192, kernel_size=1)
+        self.b7_2 = BasicConv2d(192, 192, kernel_size=(1, 7), padding=(0, 3))
+        self.b7_3 = BasicConv2d(192, 192, kernel_size=(7, 1), padding=(3, 0))
+        self.b7_4 = BasicConv2d(192, 192, kernel_size=3, stride=2)
+
+    def forward(self, x):
+        return torch.cat([
+            self.b3_2(self.b3_1(x)),
+            self.b7_4(self.b7_3(self.b7_2(self.b7_1(x)))),
+            _max_pool(x, 3, 2),
+        ], 1)
+
+
+class InceptionE(nn.Module):
+    def __init__(self, cin):
+        super().__init__()
+        self.b1x1 = BasicConv2d(cin, 320, kernel_size=1)
+        self.b3_1 = BasicConv2d(cin, 384, kernel_size=1)
+        self.b3_2a = BasicConv2d(384, 384, kernel_size=(1, 3), padding=(0, 1))
+        self.b3_2b = BasicConv2d(384, 384, kernel_size=(3, 1), padding=(1, 0))
+        self.b3d_1 = BasicConv2d(cin, 448, kernel_size=1)
+        self.b3d_2 = BasicConv2d(448, 384, kernel_size=3, padding=1)
+        self.b3d_3a = BasicConv2d(384, 384, kernel_size=(1, 3), padding=(0, 1))
+        self.b3d_3b = BasicConv2d(384, 384, kernel_size=(3, 1), padding=(1, 0))
+        self.bpool = BasicConv2d(cin, 192, kernel_size=1)
+
+    def forward(self, x):
+        b3 = self.b3_1(x)
+        b3d = self.b3d_2(self.b3d_1(x))
+        return torch.cat([
+            self.b1x1(x),
+            torch.cat([self.b3_2a(b3), self.b3_2b(b3)], 1),
+            torch.cat([self.b3d_3a(b3d), self.b3d_3b(b3d)], 1),
+            self.bpool(_avg_pool(x, 3)),
+        ], 1)
+
+
+class InceptionV3(nn.Module):
+    """Standard Inception-v3 (no aux head; train-mode BN)."""
+
+    def __init__(self, num_classes=1000, seed=0):
+        super().__init__()
+        torch.manual_seed(seed)
+        self.stem = nn.ModuleList([
+            BasicConv2d(3, 32, kernel_size=3, stride=2),
+            BasicConv2d(32, 32, kernel_size=3),
+            BasicConv2d(32, 64, kernel_size=3, padding=1),
+        ])
+        self.stem2 = nn.ModuleList([
+            BasicConv2d(64, 80, kernel_size=1),
+            BasicConv2d(80, 192, kernel_size=3),
+        ])
+        self.mixed = nn.ModuleList([
+            InceptionA(192, 32), InceptionA(256, 64), InceptionA(288, 64),
+            InceptionB(288),
+            InceptionC(768, 128), InceptionC(768, 160),
+            InceptionC(768, 160), InceptionC(768, 192),
+            InceptionD(768),
+            InceptionE(1280), InceptionE(2048),
+        ])
+        self.fc_w = nn.Parameter(torch.randn(2048, num_classes) * 0.01)
+        self.fc_b = nn.Parameter(torch.zeros(num_classes))
+
+    def forward(self, x):
+        for m in self.stem:
+            x = m(x)
+        x = _max_pool(x, 3, 2)
+        for m in self.stem2:
+            x = m(x)
+        x = _max_pool(x, 3, 2)
+        for m in self.mixed:
+            x = m(x)
+        x = torch.nn.functional.adaptive_avg_pool2d(x.float(), 1)
+        x = x.flatten(1).to(self.fc_w.dtype)
+        if x.is_cuda:
+            return ops.gemm_bias_act(x.contiguous(), self.fc_w,
+                                     self.fc_b.float())
+        return x.float() @ self.fc_w.float() + self.fc_b.float()
+
+
+def synthetic_images(batch, size=299, classes=1000, device="cpu",
+                     dtype=torch.float32, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    x = torch.rand(batch, 3, size, size, generator=g)
+    y = torch.randint(0, classes, (batch,), generator=g)
+    return x.to(device=device, dtype=dtype), y.to(device)

@@ -306,3 +306,33 @@ def conv2d(x, w, bias=None, stride=1, padding=0):
     if isinstance(padding, int):
         padding = (padding, padding)
     return _Conv2dFn.apply(x, w, bias, tuple(stride), tuple(padding))
+
+
+class _SoftmaxXentFn(torch.autograd.Function):
+    """Differentiable fused softmax cross-entropy (mean over batch) on
+    the HIP kernels; used by autograd models (Inception classifier)."""
+
+    @staticmethod
+    def forward(ctx, logits, labels):
+        if logits.is_cuda:
+            loss, probs = _ext().softmax_xent_fwd(logits, labels)
+        else:
+            lg = logits.float()
+            probs = torch.softmax(lg, 1).to(logits.dtype)
+            loss = torch.nn.functional.cross_entropy(lg, labels)
+        ctx.save_for_backward(probs, labels)
+        return loss
+
+    @staticmethod
+    def backward(ctx, gl):
+        probs, labels = ctx.saved_tensors
+        B = probs.shape[0]
+        d = softmax_xent_bwd(probs, labels, scale=1.0 / B)
+        if not torch.equal(gl, torch.ones_like(gl)):
+            d = d * gl.to(d.dtype)
+        return d, None
+
+
+def softmax_xent_loss(logits, labels):
+    """Mean softmax cross-entropy, differentiable w.r.t. logits."""
+    return _SoftmaxXentFn.apply(logits, labels)
