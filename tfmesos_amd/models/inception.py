@@ -210,11 +210,8 @@ class InceptionV3(nn.Module):
         for m in self.mixed:
             x = m(x)
         x = torch.nn.functional.adaptive_avg_pool2d(x.float(), 1)
-        x = x.flatten(1).to(self.fc_w.dtype)
-        if x.is_cuda:
-            return ops.gemm_bias_act(x.contiguous(), self.fc_w,
-                                     self.fc_b.float())
-        return x.float() @ self.fc_w.float() + self.fc_b.float()
+        x = x.flatten(1).to(self.fc_w.dtype).contiguous()
+        return ops.linear(x, self.fc_w, self.fc_b)
 
 
 def synthetic_images(batch, size=299, classes=1000, device="cpu",

@@ -336,3 +336,43 @@ class _SoftmaxXentFn(torch.autograd.Function):
 def softmax_xent_loss(logits, labels):
     """Mean softmax cross-entropy, differentiable w.r.t. logits."""
     return _SoftmaxXentFn.apply(logits, labels)
+
+
+class _LinearFn(torch.autograd.Function):
+    """Differentiable y = x @ w + b on the MFMA GEMM kernels (bf16).
+    Backward reuses the fused epilogues: dx = dy w^T, dw = x^T dy with
+    the bias-grad colsum fused into the dw GEMM."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = b is not None
+        if x.is_cuda:
+            return gemm_bias_act(x, w, b)
+        y = x.float() @ w.float()
+        if b is not None:
+            y = y + b.float()
+        return y.to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        if x.is_cuda:
+            db = torch.empty(w.shape[1], dtype=torch.float32,
+                             device=x.device) if ctx.has_bias else None
+            dw = torch.empty(w.shape, dtype=torch.float32, device=x.device)
+            gemm_bias_act(x, dy, trans_a=True, out=dw, colsum_out=db)
+            dx = gemm_bias_act(dy, w, trans_b=True)
+            dw = dw.to(w.dtype)
+        else:
+            dyf = dy.float()
+            dx = (dyf @ w.float().t()).to(x.dtype)
+            dw = (x.float().t() @ dyf).to(w.dtype)
+            db = dyf.sum(0) if ctx.has_bias else None
+        return dx, dw, db
+
+
+def linear(x, w, b=None):
+    """Differentiable linear layer on the hand-written MFMA GEMM."""
+    return _LinearFn.apply(x, w, b)
