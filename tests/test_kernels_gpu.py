@@ -409,3 +409,38 @@ def test_conv2d_train_step_decreases_loss():
                 p -= 0.1 * p.grad
                 p.grad = None
     assert float(loss_fn()) < l0
+
+
+# --------------------------------------------------------------------- bn
+
+@pytest.mark.parametrize("relu", [False, True])
+@pytest.mark.parametrize("shape", [(4, 16, 14, 14), (2, 64, 35, 35),
+                                   (8, 32, 7, 9)])
+def test_batch_norm_act_fwd_bwd(shape, relu):
+    from tfmesos_amd import ops
+    torch.manual_seed(hash((shape, relu)) % 2**31)
+    N, C, H, W = shape
+    x = bf(torch.randn(N, C, H, W) * 2 + 0.5).requires_grad_(True)
+    g = torch.rand(C, device=DEV) + 0.5
+    b = torch.randn(C, device=DEV) * 0.2
+    g = g.requires_grad_(True)
+    b = b.requires_grad_(True)
+    y = ops.batch_norm_act(x, g, b, eps=1e-3, relu=relu)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xf = x.detach().float().cpu().requires_grad_(True)
+    gf = g.detach().float().cpu().requires_grad_(True)
+    bff = b.detach().float().cpu().requires_grad_(True)
+    ref = torch.nn.functional.batch_norm(
+        xf, None, None, gf, bff, training=True, eps=1e-3)
+    if relu:
+        ref = torch.relu(ref)
+    ref.backward(dy.float().cpu())
+
+    assert (y.float().cpu() - ref).abs().max() < 0.05
+    assert (x.grad.float().cpu() - xf.grad).abs().max() < 0.05
+    assert (g.grad.cpu() - gf.grad).abs().max() < \
+        0.02 * gf.grad.abs().max() + 0.1
+    assert (b.grad.cpu() - bff.grad).abs().max() < \
+        0.02 * bff.grad.abs().max() + 0.1

@@ -36,32 +36,31 @@ class Conv2d(nn.Module):
 
 
 class BatchNorm2d(nn.Module):
-    """Train-mode batch-norm in fp32 tensor math (stats over N,H,W)."""
+    """Train-mode batch-norm on the fused HIP kernels (csrc/bn.hip);
+    optional fused relu."""
 
-    def __init__(self, num_features, eps=1e-3):
+    def __init__(self, num_features, eps=1e-3, relu=False):
         super().__init__()
         self.eps = eps
+        self.relu = relu
         self.weight = nn.Parameter(torch.ones(num_features))
         self.bias = nn.Parameter(torch.zeros(num_features))
 
     def forward(self, x):
-        xf = x.float()
-        mean = xf.mean(dim=(0, 2, 3), keepdim=True)
-        var = xf.var(dim=(0, 2, 3), unbiased=False, keepdim=True)
-        y = (xf - mean) * torch.rsqrt(var + self.eps)
-        y = y * self.weight.float().view(1, -1, 1, 1) \
-            + self.bias.float().view(1, -1, 1, 1)
-        return y.to(x.dtype)
+        return ops.batch_norm_act(x, self.weight, self.bias, eps=self.eps,
+                                  relu=self.relu)
 
 
 class BasicConv2d(nn.Module):
+    """conv -> fused BN+relu (one autograd node, 3 kernels fwd)."""
+
     def __init__(self, cin, cout, **kw):
         super().__init__()
         self.conv = Conv2d(cin, cout, **kw)
-        self.bn = BatchNorm2d(cout)
+        self.bn = BatchNorm2d(cout, relu=True)
 
     def forward(self, x):
-        return torch.relu(self.bn(self.conv(x)))
+        return self.bn(self.conv(x))
 
 
 def _avg_pool(x, k, stride=1, padding=1):

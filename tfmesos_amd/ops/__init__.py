@@ -376,3 +376,56 @@ class _LinearFn(torch.autograd.Function):
 def linear(x, w, b=None):
     """Differentiable linear layer on the hand-written MFMA GEMM."""
     return _LinearFn.apply(x, w, b)
+
+
+class _BatchNormActFn(torch.autograd.Function):
+    """Fused train-mode batch-norm (+ optional relu) on the HIP kernels
+    (csrc/bn.hip); fp32 torch reference on CPU."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps, relu):
+        if x.is_cuda:
+            y, mean, invstd = _ext().bn_fwd(x, weight.float(), bias.float(),
+                                            eps, relu)
+        else:
+            xf = x.float()
+            mean = xf.mean(dim=(0, 2, 3))
+            var = xf.var(dim=(0, 2, 3), unbiased=False)
+            invstd = torch.rsqrt(var + eps)
+            y = (xf - mean.view(1, -1, 1, 1)) * invstd.view(1, -1, 1, 1)
+            y = y * weight.float().view(1, -1, 1, 1) \
+                + bias.float().view(1, -1, 1, 1)
+            if relu:
+                y = torch.relu(y)
+            y = y.to(x.dtype)
+        ctx.save_for_backward(x, y, weight, mean, invstd)
+        ctx.relu = relu
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, y, weight, mean, invstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        if x.is_cuda:
+            dx, dgamma, dbeta = _ext().bn_bwd(x, dy, y, weight.float(),
+                                              mean, invstd, ctx.relu)
+        else:
+            dyf = dy.float()
+            if ctx.relu:
+                dyf = dyf * (y.float() > 0)
+            xf = x.float()
+            nhw = x.numel() / x.shape[1]
+            xhat = (xf - mean.view(1, -1, 1, 1)) * invstd.view(1, -1, 1, 1)
+            s1 = dyf.sum(dim=(0, 2, 3))
+            s2 = (dyf * xhat).sum(dim=(0, 2, 3))
+            dgamma, dbeta = s2, s1
+            dx = (weight.float() * invstd).view(1, -1, 1, 1) * (
+                dyf - (s1 / nhw).view(1, -1, 1, 1)
+                - xhat * (s2 / nhw).view(1, -1, 1, 1))
+            dx = dx.to(x.dtype)
+        return dx, dgamma.to(weight.dtype), dbeta.to(weight.dtype), None, None
+
+
+def batch_norm_act(x, weight, bias, eps=1e-3, relu=False):
+    """Differentiable fused train-mode BN (+relu), NCHW."""
+    return _BatchNormActFn.apply(x.contiguous(), weight, bias, eps, relu)

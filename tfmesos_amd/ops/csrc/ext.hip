@@ -44,6 +44,14 @@ void launch_conv_bwd_data(const bf16_t*, const bf16_t*, bf16_t*, int, int,
 void launch_conv_bwd_weight(const bf16_t*, const bf16_t*, float*, int, int,
                             int, int, int, int, int, int, int, int, int,
                             int, int, hipStream_t);
+void launch_bn_fwd(const bf16_t*, const float*, const float*, bf16_t*,
+                   float*, float*, float*, int, int, long, int, float, bool,
+                   hipStream_t);
+void launch_bn_bwd(const bf16_t*, const bf16_t*, const bf16_t*, const float*,
+                   const float*, const float*, bf16_t*, float*, float*,
+                   float*, float*, float*, int, int, long, int, bool,
+                   hipStream_t);
+int bn_stats_slices(int, int, long);
 
 namespace {
 
@@ -419,6 +427,55 @@ torch::Tensor conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
   return dw;
 }
 
+// -------------------------------------------------------------------- bn
+
+std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor g,
+                                  torch::Tensor b, double eps, bool relu) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous() &&
+              x.scalar_type() == torch::kBFloat16, "x must be bf16 NCHW");
+  const int N = x.size(0), C = x.size(1);
+  const long HW = (long)x.size(2) * x.size(3);
+  TORCH_CHECK(g.numel() == C && b.numel() == C &&
+              g.scalar_type() == torch::kFloat32 &&
+              b.scalar_type() == torch::kFloat32, "g/b must be fp32 [C]");
+  const int Z = bn_stats_slices(N, C, HW);
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({C}, opts);
+  auto invstd = torch::empty({C}, opts);
+  auto part = torch::empty({(long)Z * C * 2}, opts);
+  launch_bn_fwd((const bf16_t*)x.data_ptr(), g.data_ptr<float>(),
+                b.data_ptr<float>(), (bf16_t*)y.data_ptr(),
+                mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                part.data_ptr<float>(), N, C, HW, Z, (float)eps, relu,
+                cur_stream());
+  return {y, mean, invstd};
+}
+
+std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
+                                  torch::Tensor y, torch::Tensor g,
+                                  torch::Tensor mean, torch::Tensor invstd,
+                                  bool relu) {
+  const int N = x.size(0), C = x.size(1);
+  const long HW = (long)x.size(2) * x.size(3);
+  const int Z = bn_stats_slices(N, C, HW);
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto dx = torch::empty_like(x);
+  auto dgamma = torch::empty({C}, opts);
+  auto dbeta = torch::empty({C}, opts);
+  auto part = torch::empty({(long)Z * C * 2}, opts);
+  auto s1n = torch::empty({C}, opts);
+  auto s2n = torch::empty({C}, opts);
+  launch_bn_bwd((const bf16_t*)x.data_ptr(), (const bf16_t*)dy.data_ptr(),
+                (const bf16_t*)y.data_ptr(), g.data_ptr<float>(),
+                mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                (bf16_t*)dx.data_ptr(), dgamma.data_ptr<float>(),
+                dbeta.data_ptr<float>(), part.data_ptr<float>(),
+                s1n.data_ptr<float>(), s2n.data_ptr<float>(), N, C, HW, Z,
+                relu, cur_stream());
+  return {dx, dgamma, dbeta};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -444,6 +501,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("conv2d_fwd", &conv2d_fwd, "implicit-GEMM conv fwd (MFMA, bf16)");
+  m.def("bn_fwd", &bn_fwd, "fused train-mode batch-norm (+relu) fwd");
+  m.def("bn_bwd", &bn_bwd, "fused batch-norm (+relu mask) bwd");
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
   m.def("embedding_gather", &embedding_gather);
