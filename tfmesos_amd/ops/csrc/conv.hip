@@ -15,9 +15,12 @@
 //   bwd-w:  dW[k,c,r,s]   = sum_{n,ho,wo} dY[n,k,ho,wo] X[n,c,ho*U+r-P,...]
 //           GEMM  M = K,       Ncol = C*R*S, Kdim = N*Ho*Wo
 //
-// Correctness-first: gathers are scalar (predicated) but LDS/MFMA use
-// the same vectorized fragment reads as the GEMM; tuning follows
-// rocprof (profiles/).
+// Staging performance: every gather decodes its pixel index ONCE per
+// 8-element strip (integer division is the expensive op — the first,
+// correctness-only version did 8 long divisions per strip and ran at
+// <1% MFMA peak, profiles/r01_inception_n1_kernel_stats.txt), then
+// walks contiguous addresses; full-row interior strips take a straight
+// pointer walk, boundary strips an incremental-carry path.
 #include "common.h"
 
 namespace {
@@ -36,35 +39,57 @@ struct ConvShape {
 // ---------------------------------------------------------------- forward
 
 // Stage a 64(m) x 32(k) patch tile: m = output pixel, k = (c,r,s) tap.
-// Thread t loads 8 consecutive m for one tap -> consecutive wo ->
-// mostly-consecutive wi (coalesced within rows).
+// Thread t loads 8 consecutive m (-> consecutive wo) for one tap.
 DEVINL void stage_patch_fwd(const __bf16* __restrict__ X, __bf16 (*Sm)[BK + APAD],
                             const ConvShape cs, long m0, int k0, long M,
                             int CRS, int t) {
   const int kk = t >> 3;          // 0..31
   const int mm0 = (t & 7) * 8;
   const int q = k0 + kk;
-  int c = 0, r = 0, s = 0;
-  bool tap_ok = q < CRS;
-  if (tap_ok) {
-    c = q / (cs.R * cs.S);
-    r = (q / cs.S) % cs.R;
-    s = q % cs.S;
+  const long pm0 = m0 + mm0;
+  if (q >= CRS || pm0 >= M) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) Sm[mm0 + j][kk] = (__bf16)0.f;
+    return;
   }
+  const int c = q / (cs.R * cs.S);
+  const int rs = q - c * (cs.R * cs.S);
+  const int r = rs / cs.S;
+  const int s = rs - r * cs.S;
+  // decode pixel once (32-bit divisions; M < 2^31 enforced by host)
+  const int HoWo = cs.Ho * cs.Wo;
+  int n = (int)(pm0 / HoWo);
+  int rem = (int)(pm0 - (long)n * HoWo);
+  int ho = rem / cs.Wo;
+  int wo = rem - ho * cs.Wo;
+  const int hi = ho * cs.U + r - cs.P;
+  const int wi0 = wo * cs.V + s - cs.Q;
+
+  if (wo + 8 <= cs.Wo && pm0 + 8 <= M &&
+      hi >= 0 && hi < cs.H && wi0 >= 0 && wi0 + 7 * cs.V < cs.W) {
+    // fast path: one row, fully interior — straight pointer walk
+    const __bf16* src = X + (((long)n * cs.C + c) * cs.H + hi) * cs.W + wi0;
+    if (cs.V == 1) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) Sm[mm0 + j][kk] = src[j];
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) Sm[mm0 + j][kk] = src[j * cs.V];
+    }
+    return;
+  }
+  // slow path: incremental carry across wo/ho/n, per-element bounds
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    const long pm = m0 + mm0 + j;
     float v = 0.f;
-    if (tap_ok && pm < M) {
-      const int n = (int)(pm / (cs.Ho * cs.Wo));
-      const int rem = (int)(pm % (cs.Ho * cs.Wo));
-      const int ho = rem / cs.Wo, wo = rem % cs.Wo;
-      const int hi = ho * cs.U + r - cs.P;
-      const int wi = wo * cs.V + s - cs.Q;
-      if (hi >= 0 && hi < cs.H && wi >= 0 && wi < cs.W)
-        v = (float)X[(((long)n * cs.C + c) * cs.H + hi) * cs.W + wi];
+    if (pm0 + j < M) {
+      const int hij = ho * cs.U + r - cs.P;
+      const int wij = wo * cs.V + s - cs.Q;
+      if (hij >= 0 && hij < cs.H && wij >= 0 && wij < cs.W)
+        v = (float)X[(((long)n * cs.C + c) * cs.H + hij) * cs.W + wij];
     }
     Sm[mm0 + j][kk] = (__bf16)v;
+    if (++wo == cs.Wo) { wo = 0; if (++ho == cs.Ho) { ho = 0; ++n; } }
   }
 }
 
@@ -76,9 +101,14 @@ DEVINL void stage_wtile(const __bf16* __restrict__ Wt, __bf16 (*Sn)[BK + APAD],
   const int gx = n0 + x;
   const int gk = k0 + kk0;
   const __bf16* src = Wt + (long)gx * CRS + gk;
+  if (gx < K && gk + 8 <= CRS) {
 #pragma unroll
-  for (int j = 0; j < 8; ++j)
-    Sn[x][kk0 + j] = (gx < K && gk + j < CRS) ? src[j] : (__bf16)0.f;
+    for (int j = 0; j < 8; ++j) Sn[x][kk0 + j] = src[j];
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      Sn[x][kk0 + j] = (gx < K && gk + j < CRS) ? src[j] : (__bf16)0.f;
+  }
 }
 
 template <bool BIAS, bool RELU>
@@ -141,60 +171,94 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
 // --------------------------------------------------------------- bwd-data
 
 // m = input pixel (n,h,w); tap q = (k,r,s); contributes when
-// (h+P-r) % U == 0 and in range (same for w).
+// (h+P-r) % U == 0 and in range (same for w). STRIDE1 specializes the
+// common U==V==1 case (no divisibility tests, contiguous fast path).
+template <bool STRIDE1>
 DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
                              __bf16 (*Sm)[BK + APAD], const ConvShape cs,
                              long m0, int k0, long M, int KRS, int t) {
   const int kk = t >> 3;
   const int mm0 = (t & 7) * 8;
   const int q = k0 + kk;
-  int k = 0, r = 0, s = 0;
-  const bool tap_ok = q < KRS;
-  if (tap_ok) {
-    k = q / (cs.R * cs.S);
-    r = (q / cs.S) % cs.R;
-    s = q % cs.S;
+  const long pm0 = m0 + mm0;
+  if (q >= KRS || pm0 >= M) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) Sm[mm0 + j][kk] = (__bf16)0.f;
+    return;
+  }
+  const int k = q / (cs.R * cs.S);
+  const int rs = q - k * (cs.R * cs.S);
+  const int r = rs / cs.S;
+  const int s = rs - r * cs.S;
+  const int HWi = cs.H * cs.W;
+  int n = (int)(pm0 / HWi);
+  int rem = (int)(pm0 - (long)n * HWi);
+  int h = rem / cs.W;
+  int w = rem - h * cs.W;
+
+  if (STRIDE1) {
+    const int ho = h + cs.P - r;
+    const int wo0 = w + cs.Q - s;
+    if (w + 8 <= cs.W && pm0 + 8 <= M &&
+        ho >= 0 && ho < cs.Ho && wo0 >= 0 && wo0 + 8 <= cs.Wo) {
+      const __bf16* src =
+          dY + (((long)n * cs.K + k) * cs.Ho + ho) * cs.Wo + wo0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) Sm[mm0 + j][kk] = src[j];
+      return;
+    }
   }
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    const long pm = m0 + mm0 + j;
     float v = 0.f;
-    if (tap_ok && pm < M) {
-      const int n = (int)(pm / ((long)cs.H * cs.W));
-      const int rem = (int)(pm % ((long)cs.H * cs.W));
-      const int h = rem / cs.W, w = rem % cs.W;
+    if (pm0 + j < M) {
       const int hn = h + cs.P - r, wn = w + cs.Q - s;
-      if (hn >= 0 && wn >= 0 && hn % cs.U == 0 && wn % cs.V == 0) {
-        const int ho = hn / cs.U, wo = wn / cs.V;
-        if (ho < cs.Ho && wo < cs.Wo)
-          v = (float)dY[(((long)n * cs.K + k) * cs.Ho + ho) * cs.Wo + wo];
+      if (STRIDE1) {
+        if (hn >= 0 && hn < cs.Ho && wn >= 0 && wn < cs.Wo)
+          v = (float)dY[(((long)n * cs.K + k) * cs.Ho + hn) * cs.Wo + wn];
+      } else {
+        if (hn >= 0 && wn >= 0 && hn % cs.U == 0 && wn % cs.V == 0) {
+          const int ho = hn / cs.U, wo = wn / cs.V;
+          if (ho < cs.Ho && wo < cs.Wo)
+            v = (float)dY[(((long)n * cs.K + k) * cs.Ho + ho) * cs.Wo + wo];
+        }
       }
     }
     Sm[mm0 + j][kk] = (__bf16)v;
+    if (++w == cs.W) { w = 0; if (++h == cs.H) { h = 0; ++n; } }
   }
 }
 
-// Weight tile for bwd-data: Bs[c][tap(k,r,s)] = W[k,c,r,s] (strided gather).
+// Weight tile for bwd-data: Bs[c][tap(k,r,s)] = W[k,c,r,s] (strided gather;
+// taps decode incrementally — consecutive q walk s, then r, then k).
 DEVINL void stage_wtile_bwdd(const __bf16* __restrict__ Wt,
                              __bf16 (*Sn)[BK + APAD], const ConvShape cs,
                              int n0, int k0, int KRS, int t) {
   const int x = t >> 2;          // c offset 0..63
   const int kk0 = (t & 3) * 8;
   const int c = n0 + x;
+  const int q0 = k0 + kk0;
+  if (c >= cs.C || q0 >= KRS) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) Sn[x][kk0 + j] = (__bf16)0.f;
+    return;
+  }
+  const int RS = cs.R * cs.S;
+  int k = q0 / RS;
+  int rs = q0 - k * RS;
+  int r = rs / cs.S;
+  int s = rs - r * cs.S;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    const int q = k0 + kk0 + j;
     float v = 0.f;
-    if (c < cs.C && q < KRS) {
-      const int k = q / (cs.R * cs.S);
-      const int r = (q / cs.S) % cs.R;
-      const int s = q % cs.S;
+    if (q0 + j < KRS)
       v = (float)Wt[(((long)k * cs.C + c) * cs.R + r) * cs.S + s];
-    }
     Sn[x][kk0 + j] = (__bf16)v;
+    if (++s == cs.S) { s = 0; if (++r == cs.R) { r = 0; ++k; } }
   }
 }
 
+template <bool STRIDE1>
 __global__ __launch_bounds__(256)
 void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ Wt,
                       __bf16* __restrict__ dX, ConvShape cs) {
@@ -211,7 +275,7 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
 
   f32x4 acc[2][2] = {};
   for (int k0 = 0; k0 < KRS; k0 += BK) {
-    stage_patch_bwdd(dY, As, cs, tm0, k0, M, KRS, t);
+    stage_patch_bwdd<STRIDE1>(dY, As, cs, tm0, k0, M, KRS, t);
     stage_wtile_bwdd(Wt, Bs, cs, tn0, k0, KRS, t);
     __syncthreads();
     const int kfrag = (lane >> 4) * 8;
@@ -249,57 +313,86 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
 // -------------------------------------------------------------- bwd-weight
 
 // GEMM: rows m = output channel k (M=K), cols = (c,r,s), Kdim = N*Ho*Wo.
-// As[k][p] = dY[n,k,ho,wo] for reduction pixel p; Bs[crs][p] = X patch.
+// As[k][p] = dY[n,k,ho,wo]; thread t loads 8 consecutive p for one k
+// (contiguous in dY within an image).
 DEVINL void stage_dy_bwdw(const __bf16* __restrict__ dY, __bf16 (*Sm)[BK + APAD],
-                          const ConvShape cs, int m0, long p0, int t) {
-  const int kk = t >> 3;          // reduction pixel offset 0..31
-  const int mm0 = (t & 7) * 8;    // k-channel offset
-  const long p = p0 + kk;
+                          const ConvShape cs, int m0, long p0, long Ptot,
+                          int t) {
+  const int pp0 = (t & 3) * 8;     // p offset 0..24
+  const int kx = t >> 2;           // k-channel row 0..63
+  const int k = m0 + kx;
+  const long p = p0 + pp0;
   const long HoWo = (long)cs.Ho * cs.Wo;
-  const bool p_ok = p < (long)cs.N * HoWo;
-  int n = 0, rem = 0;
-  if (p_ok) {
-    n = (int)(p / HoWo);
-    rem = (int)(p % HoWo);
+  if (k >= cs.K || p >= Ptot) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) Sm[kx][pp0 + j] = (__bf16)0.f;
+    return;
+  }
+  int n = (int)(p / HoWo);
+  int rem = (int)(p - (long)n * HoWo);
+  if (p + 8 <= Ptot && rem + 8 <= (int)HoWo) {
+    const __bf16* src = dY + ((long)n * cs.K + k) * HoWo + rem;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) Sm[kx][pp0 + j] = src[j];
+    return;
   }
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    const int k = m0 + mm0 + j;
     float v = 0.f;
-    if (p_ok && k < cs.K)
+    if (p + j < Ptot)
       v = (float)dY[((long)n * cs.K + k) * HoWo + rem];
-    Sm[mm0 + j][kk] = (__bf16)v;
+    Sm[kx][pp0 + j] = (__bf16)v;
+    if (++rem == (int)HoWo) { rem = 0; ++n; }
   }
 }
 
+// Bs[crs][p] = X[n, c, ho*U+r-P, wo*V+s-Q]; 8 consecutive p for one tap.
 DEVINL void stage_x_bwdw(const __bf16* __restrict__ X, __bf16 (*Sn)[BK + APAD],
-                         const ConvShape cs, int n0, long p0, int CRS, int t) {
-  const int kk = t >> 3;          // reduction pixel offset
-  const int xx0 = (t & 7) * 8;    // crs offset
-  const long p = p0 + kk;
+                         const ConvShape cs, int n0, long p0, int CRS,
+                         long Ptot, int t) {
+  const int pp0 = (t & 3) * 8;
+  const int qx = t >> 2;           // tap row 0..63
+  const int q = n0 + qx;
+  const long p = p0 + pp0;
+  if (q >= CRS || p >= Ptot) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) Sn[qx][pp0 + j] = (__bf16)0.f;
+    return;
+  }
+  const int c = q / (cs.R * cs.S);
+  const int rs = q - c * (cs.R * cs.S);
+  const int r = rs / cs.S;
+  const int s = rs - r * cs.S;
   const long HoWo = (long)cs.Ho * cs.Wo;
-  const bool p_ok = p < (long)cs.N * HoWo;
-  int n = 0, ho = 0, wo = 0;
-  if (p_ok) {
-    n = (int)(p / HoWo);
-    const int rem = (int)(p % HoWo);
-    ho = rem / cs.Wo;
-    wo = rem % cs.Wo;
+  int n = (int)(p / HoWo);
+  int rem = (int)(p - (long)n * HoWo);
+  int ho = rem / cs.Wo;
+  int wo = rem - ho * cs.Wo;
+  const int hi = ho * cs.U + r - cs.P;
+  const int wi0 = wo * cs.V + s - cs.Q;
+  if (p + 8 <= Ptot && wo + 8 <= cs.Wo &&
+      hi >= 0 && hi < cs.H && wi0 >= 0 && wi0 + 7 * cs.V < cs.W) {
+    const __bf16* src = X + (((long)n * cs.C + c) * cs.H + hi) * cs.W + wi0;
+    if (cs.V == 1) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) Sn[qx][pp0 + j] = src[j];
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) Sn[qx][pp0 + j] = src[j * cs.V];
+    }
+    return;
   }
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    const int q = n0 + xx0 + j;
     float v = 0.f;
-    if (p_ok && q < CRS) {
-      const int c = q / (cs.R * cs.S);
-      const int r = (q / cs.S) % cs.R;
-      const int s = q % cs.S;
-      const int hi = ho * cs.U + r - cs.P;
-      const int wi = wo * cs.V + s - cs.Q;
-      if (hi >= 0 && hi < cs.H && wi >= 0 && wi < cs.W)
-        v = (float)X[(((long)n * cs.C + c) * cs.H + hi) * cs.W + wi];
+    if (p + j < Ptot) {
+      const int hij = ho * cs.U + r - cs.P;
+      const int wij = wo * cs.V + s - cs.Q;
+      if (hij >= 0 && hij < cs.H && wij >= 0 && wij < cs.W)
+        v = (float)X[(((long)n * cs.C + c) * cs.H + hij) * cs.W + wij];
     }
-    Sn[xx0 + j][kk] = (__bf16)v;
+    Sn[qx][pp0 + j] = (__bf16)v;
+    if (++wo == cs.Wo) { wo = 0; if (++ho == cs.Ho) { ho = 0; ++n; } }
   }
 }
 
@@ -323,8 +416,8 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
 
   f32x4 acc[2][2] = {};
   for (long p0 = ps; p0 < pe; p0 += BK) {
-    stage_dy_bwdw(dY, As, cs, tm0, p0, t);
-    stage_x_bwdw(X, Bs, cs, tn0, p0, CRS, t);
+    stage_dy_bwdw(dY, As, cs, tm0, p0, Ptot, t);
+    stage_x_bwdw(X, Bs, cs, tn0, p0, CRS, Ptot, t);
     __syncthreads();
     const int kfrag = (lane >> 4) * 8;
 #pragma unroll
@@ -397,8 +490,12 @@ void launch_conv_bwd_data(const bf16_t* dY, const bf16_t* W, bf16_t* dX,
   const long M = (long)N * H * Wd;
   dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(C, BN));
   dim3 block(256);
-  hipLaunchKernelGGL(conv_bwdd_kernel, grid, block, 0, stream,
-                     (const __bf16*)dY, (const __bf16*)W, (__bf16*)dX, cs);
+  if (U == 1 && V == 1)
+    hipLaunchKernelGGL((conv_bwdd_kernel<true>), grid, block, 0, stream,
+                       (const __bf16*)dY, (const __bf16*)W, (__bf16*)dX, cs);
+  else
+    hipLaunchKernelGGL((conv_bwdd_kernel<false>), grid, block, 0, stream,
+                       (const __bf16*)dY, (const __bf16*)W, (__bf16*)dX, cs);
 }
 
 void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
@@ -408,7 +505,7 @@ void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
   ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q};
   const int CRS = C * R * S;
   const long Ptot = (long)N * Ho * Wo;
-  // slice the reduction so the grid can fill the chip (>=256 WGs)
+  // slice the reduction so the grid can fill the chip (>=512 WGs)
   const long tiles = (long)ceil_div(K, BM) * ceil_div(CRS, BN);
   long zmax = (Ptot + BK - 1) / BK;
   long zwant = 512 / tiles;
