@@ -260,15 +260,20 @@ class _Conv2dFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, bias, stride, padding):
-        ctx.save_for_backward(x, w)
         ctx.stride = stride
         ctx.padding = padding
         ctx.has_bias = bias is not None
         if x.is_cuda:
+            # channels-last memory: gathers become contiguous channel
+            # runs (csrc/conv.hip); weights go tap-major [K,R,S,C]
+            xm = x.contiguous(memory_format=torch.channels_last)
+            wm = w.permute(0, 2, 3, 1).contiguous()
+            ctx.save_for_backward(xm, w)
             eb = bias.float() if bias is not None else \
                 torch.empty(0, device=x.device)
-            return _ext().conv2d_fwd(x, w, eb, stride[0], stride[1],
+            return _ext().conv2d_fwd(xm, wm, eb, stride[0], stride[1],
                                      padding[0], padding[1], False)
+        ctx.save_for_backward(x, w)
         y = torch.nn.functional.conv2d(
             x.float(), w.float(), bias.float() if bias is not None else None,
             stride=stride, padding=padding)
@@ -277,16 +282,18 @@ class _Conv2dFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, w = ctx.saved_tensors
-        dy = dy.contiguous()
         if x.is_cuda:
-            dx = _ext().conv2d_bwd_data(dy, w, x.shape[2], x.shape[3],
+            dy = dy.contiguous(memory_format=torch.channels_last)
+            wt = w.permute(1, 2, 3, 0).contiguous()   # [C,R,S,K] = W^T
+            dx = _ext().conv2d_bwd_data(dy, wt, x.shape[2], x.shape[3],
                                         ctx.stride[0], ctx.stride[1],
                                         ctx.padding[0], ctx.padding[1])
-            dw = _ext().conv2d_bwd_weight(dy, x, w.shape[2], w.shape[3],
-                                          ctx.stride[0], ctx.stride[1],
-                                          ctx.padding[0], ctx.padding[1])
-            dw = dw.to(w.dtype)
+            dwm = _ext().conv2d_bwd_weight(dy, x, w.shape[2], w.shape[3],
+                                           ctx.stride[0], ctx.stride[1],
+                                           ctx.padding[0], ctx.padding[1])
+            dw = dwm.permute(0, 3, 1, 2).to(w.dtype)
         else:
+            dy = dy.contiguous()
             dyf, xf, wf = dy.float(), x.float(), w.float()
             dx = torch.nn.grad.conv2d_input(
                 x.shape, wf, dyf, stride=ctx.stride, padding=ctx.padding
@@ -385,6 +392,7 @@ class _BatchNormActFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, eps, relu):
         if x.is_cuda:
+            x = x.contiguous(memory_format=torch.channels_last)
             y, mean, invstd = _ext().bn_fwd(x, weight.float(), bias.float(),
                                             eps, relu)
         else:
@@ -405,12 +413,12 @@ class _BatchNormActFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, y, weight, mean, invstd = ctx.saved_tensors
-        dy = dy.contiguous()
         if x.is_cuda:
+            dy = dy.contiguous(memory_format=torch.channels_last)
             dx, dgamma, dbeta = _ext().bn_bwd(x, dy, y, weight.float(),
                                               mean, invstd, ctx.relu)
         else:
-            dyf = dy.float()
+            dyf = dy.contiguous().float()
             if ctx.relu:
                 dyf = dyf * (y.float() > 0)
             xf = x.float()
@@ -427,5 +435,7 @@ class _BatchNormActFn(torch.autograd.Function):
 
 
 def batch_norm_act(x, weight, bias, eps=1e-3, relu=False):
-    """Differentiable fused train-mode BN (+relu), NCHW."""
-    return _BatchNormActFn.apply(x.contiguous(), weight, bias, eps, relu)
+    """Differentiable fused train-mode BN (+relu); channels-last on GPU."""
+    if not x.is_cuda:
+        x = x.contiguous()
+    return _BatchNormActFn.apply(x, weight, bias, eps, relu)

@@ -1,51 +1,51 @@
-// Fused train-mode batch-norm (+ReLU) for NCHW bf16 activations.
+// Fused train-mode batch-norm (+ReLU) for channels-last bf16 activations.
 //
 // The Inception blocks are conv -> BN -> relu; profiling the torch
 // fallback (profiles/r01_inception_n1_kernel_stats.txt) showed the BN
 // tensor-math soup (fp32 casts + separate mean/var/normalize/affine/relu
-// kernels) at ~36% of the training step. Here each direction is three
-// hand-written kernels, all bf16-in/bf16-out with fp32 math:
+// kernels) at ~36% of the training step. The activation layout is the
+// conv kernels' channels-last ([P, C] with P = N*H*W pixels, channels
+// innermost), so every wave reads 64 consecutive channels — perfectly
+// coalesced — and per-channel scale/shift live in LDS for the
+// elementwise passes. Three kernels per direction:
 //
-//   fwd: stats-partial (grid C x Z, plane-coalesced reduction, no
-//        atomics -> per-slice partials) -> finalize (mean/invstd) ->
-//        apply (normalize+affine+ReLU fused, one pass)
-//   bwd: stats-partial (sum dy_eff, sum dy_eff*xhat; the ReLU mask
-//        y>0 is fused in) -> finalize (dgamma/dbeta + normalized sums)
-//        -> apply (dx in one pass)
-//
-// Reductions read each element once; scalar loads coalesce across the
-// 64 lanes (planes are contiguous), so the kernels are bandwidth-bound.
+//   fwd: stats-partial (grid C/64 x Z, no atomics -> per-slice partials)
+//        -> finalize (mean/invstd) -> apply (normalize+affine+ReLU, one
+//        pass, b128 vectorized when C % 8 == 0)
+//   bwd: stats-partial (sum dy_eff, sum dy_eff*xhat; ReLU mask y>0
+//        fused) -> finalize (dgamma/dbeta + normalized sums) -> apply
+//        (dx in one pass)
 #include "common.h"
 
 namespace {
 
+constexpr int MAXC = 2048;   // LDS scale/shift staging bound (40 KB worst case)
+
+// grid (ceil(C/64), Z); block 256 = 64 channel lanes x 4 pixel rows
 __global__ __launch_bounds__(256)
 void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict__ part,
-                     int N, int C, long HW, int Z) {
-  const int c = blockIdx.x;
+                     long P, int C, int Z) {
+  const int cl = threadIdx.x & 63;
+  const int pr = threadIdx.x >> 6;           // 0..3
+  const int c = blockIdx.x * 64 + cl;
   const int z = blockIdx.y;
-  const int t = threadIdx.x;
   float sum = 0.f, sq = 0.f;
-  for (int n = 0; n < N; ++n) {
-    const __bf16* plane = x + ((long)n * C + c) * HW;
-    for (long i = (long)z * 256 + t; i < HW; i += (long)Z * 256) {
-      const float v = (float)plane[i];
+  if (c < C) {
+    for (long p = (long)z * 4 + pr; p < P; p += (long)Z * 4) {
+      const float v = (float)x[p * C + c];
       sum += v;
       sq += v * v;
     }
   }
-  __shared__ float ls[256], lq[256];
-  ls[t] = sum;
-  lq[t] = sq;
+  __shared__ float ls[4][64], lq[4][64];
+  ls[pr][cl] = sum;
+  lq[pr][cl] = sq;
   __syncthreads();
-#pragma unroll
-  for (int s = 128; s > 0; s >>= 1) {
-    if (t < s) { ls[t] += ls[t + s]; lq[t] += lq[t + s]; }
-    __syncthreads();
-  }
-  if (t == 0) {
-    part[((long)z * C + c) * 2] = ls[0];
-    part[((long)z * C + c) * 2 + 1] = lq[0];
+  if (pr == 0 && c < C) {
+    sum = ls[0][cl] + ls[1][cl] + ls[2][cl] + ls[3][cl];
+    sq = lq[0][cl] + lq[1][cl] + lq[2][cl] + lq[3][cl];
+    part[((long)z * C + c) * 2] = sum;
+    part[((long)z * C + c) * 2 + 1] = sq;
   }
 }
 
@@ -67,23 +67,44 @@ __global__ void bn_finalize_kernel(const float* __restrict__ part,
   invstd[c] = __frsqrt_rn(var + eps);
 }
 
-template <bool RELU>
+template <bool RELU, bool VEC>
 __global__ __launch_bounds__(256)
 void bn_apply_kernel(const __bf16* __restrict__ x,
                      const float* __restrict__ mean,
                      const float* __restrict__ invstd,
                      const float* __restrict__ g, const float* __restrict__ b,
-                     __bf16* __restrict__ y, int C, long HW) {
-  const long p = blockIdx.x;           // plane n*C + c
-  const int c = (int)(p % C);
-  const float mu = mean[c], is = invstd[c];
-  const float sc = g[c] * is, sh = b[c] - mu * sc;
-  const __bf16* xin = x + p * HW;
-  __bf16* yout = y + p * HW;
-  for (long i = threadIdx.x; i < HW; i += 256) {
-    float v = (float)xin[i] * sc + sh;
-    if (RELU) v = v > 0.f ? v : 0.f;
-    yout[i] = (__bf16)v;
+                     __bf16* __restrict__ y, long P, int C) {
+  __shared__ float sc[MAXC], sh[MAXC];
+  for (int c = threadIdx.x; c < C; c += 256) {
+    const float s = g[c] * invstd[c];
+    sc[c] = s;
+    sh[c] = b[c] - mean[c] * s;
+  }
+  __syncthreads();
+  const long total = P * C;
+  const long stride = (long)gridDim.x * 256 * 8;
+  for (long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8; i < total;
+       i += stride) {
+    const int c0 = (int)(i % C);
+    if (VEC && i + 8 <= total) {
+      bf16x8 v = *(const bf16x8*)&x[i];
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = (float)v[j] * sc[c0 + j] + sh[c0 + j];
+        if (RELU) f = f > 0.f ? f : 0.f;
+        o[j] = (__bf16)f;
+      }
+      *(bf16x8*)&y[i] = o;
+    } else {
+      int c = c0;
+      for (int j = 0; j < 8 && i + j < total; ++j) {
+        float f = (float)x[i + j] * sc[c] + sh[c];
+        if (RELU) f = f > 0.f ? f : 0.f;
+        y[i + j] = (__bf16)f;
+        if (++c == C) c = 0;
+      }
+    }
   }
 }
 
@@ -94,35 +115,32 @@ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
                          const __bf16* __restrict__ y,
                          const float* __restrict__ mean,
                          const float* __restrict__ invstd,
-                         float* __restrict__ part, int N, int C, long HW,
-                         int Z) {
-  const int c = blockIdx.x;
+                         float* __restrict__ part, long P, int C, int Z) {
+  const int cl = threadIdx.x & 63;
+  const int pr = threadIdx.x >> 6;
+  const int c = blockIdx.x * 64 + cl;
   const int z = blockIdx.y;
-  const int t = threadIdx.x;
-  const float mu = mean[c], is = invstd[c];
   float s1 = 0.f, s2 = 0.f;
-  for (int n = 0; n < N; ++n) {
-    const long base = ((long)n * C + c) * HW;
-    for (long i = (long)z * 256 + t; i < HW; i += (long)Z * 256) {
-      float d = (float)dy[base + i];
-      if (RELU && (float)y[base + i] <= 0.f) d = 0.f;
-      const float xh = ((float)x[base + i] - mu) * is;
+  if (c < C) {
+    const float mu = mean[c], is = invstd[c];
+    for (long p = (long)z * 4 + pr; p < P; p += (long)Z * 4) {
+      const long i = p * C + c;
+      float d = (float)dy[i];
+      if (RELU && (float)y[i] <= 0.f) d = 0.f;
+      const float xh = ((float)x[i] - mu) * is;
       s1 += d;
       s2 += d * xh;
     }
   }
-  __shared__ float l1[256], l2[256];
-  l1[t] = s1;
-  l2[t] = s2;
+  __shared__ float l1[4][64], l2[4][64];
+  l1[pr][cl] = s1;
+  l2[pr][cl] = s2;
   __syncthreads();
-#pragma unroll
-  for (int s = 128; s > 0; s >>= 1) {
-    if (t < s) { l1[t] += l1[t + s]; l2[t] += l2[t + s]; }
-    __syncthreads();
-  }
-  if (t == 0) {
-    part[((long)z * C + c) * 2] = l1[0];
-    part[((long)z * C + c) * 2 + 1] = l2[0];
+  if (pr == 0 && c < C) {
+    s1 = l1[0][cl] + l1[1][cl] + l1[2][cl] + l1[3][cl];
+    s2 = l2[0][cl] + l2[1][cl] + l2[2][cl] + l2[3][cl];
+    part[((long)z * C + c) * 2] = s1;
+    part[((long)z * C + c) * 2 + 1] = s2;
   }
 }
 
@@ -145,7 +163,7 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ part,
   s2n[c] = s2 * inv_count;
 }
 
-template <bool RELU>
+template <bool RELU, bool VEC>
 __global__ __launch_bounds__(256)
 void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
                          const __bf16* __restrict__ dy,
@@ -155,79 +173,119 @@ void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
                          const float* __restrict__ g,
                          const float* __restrict__ s1n,
                          const float* __restrict__ s2n,
-                         __bf16* __restrict__ dx, int C, long HW) {
-  const long p = blockIdx.x;
-  const int c = (int)(p % C);
-  const float mu = mean[c], is = invstd[c];
-  const float gs = g[c] * is;
-  const float a = s1n[c], bb = s2n[c];
-  const long base = p * HW;
-  for (long i = threadIdx.x; i < HW; i += 256) {
-    float d = (float)dy[base + i];
-    if (RELU && (float)y[base + i] <= 0.f) d = 0.f;
-    const float xh = ((float)x[base + i] - mu) * is;
-    dx[base + i] = (__bf16)(gs * (d - a - xh * bb));
+                         __bf16* __restrict__ dx, long P, int C) {
+  // per-channel constants staged in LDS: dx = gs*(dy_eff - a - xhat*bb),
+  // xhat = (x - mu)*is
+  __shared__ float lgs[MAXC], la[MAXC], lbb[MAXC], lmu[MAXC], lis[MAXC];
+  for (int c = threadIdx.x; c < C; c += 256) {
+    lgs[c] = g[c] * invstd[c];
+    la[c] = s1n[c];
+    lbb[c] = s2n[c];
+    lmu[c] = mean[c];
+    lis[c] = invstd[c];
+  }
+  __syncthreads();
+  const long total = P * C;
+  const long stride = (long)gridDim.x * 256 * 8;
+  for (long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8; i < total;
+       i += stride) {
+    const int c0 = (int)(i % C);
+    if (VEC && i + 8 <= total) {
+      bf16x8 xv = *(const bf16x8*)&x[i];
+      bf16x8 dv = *(const bf16x8*)&dy[i];
+      bf16x8 yv;
+      if (RELU) yv = *(const bf16x8*)&y[i];
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int c = c0 + j;
+        float d = (float)dv[j];
+        if (RELU && (float)yv[j] <= 0.f) d = 0.f;
+        const float xh = ((float)xv[j] - lmu[c]) * lis[c];
+        o[j] = (__bf16)(lgs[c] * (d - la[c] - xh * lbb[c]));
+      }
+      *(bf16x8*)&dx[i] = o;
+    } else {
+      int c = c0;
+      for (int j = 0; j < 8 && i + j < total; ++j) {
+        float d = (float)dy[i + j];
+        if (RELU && (float)y[i + j] <= 0.f) d = 0.f;
+        const float xh = ((float)x[i + j] - lmu[c]) * lis[c];
+        dx[i + j] = (__bf16)(lgs[c] * (d - la[c] - xh * lbb[c]));
+        if (++c == C) c = 0;
+      }
+    }
   }
 }
 
-inline int stats_slices(int N, int C, long HW) {
-  // target >=1024 workgroups across the C x Z grid, but keep each
-  // slice >=4 round-trips of 256 threads
-  long per = ((long)N * HW) / (256 * 4);
-  long want = (1024 + C - 1) / C;
+inline int stats_slices(long P, int C) {
+  // target >=1024 workgroups across the (C/64) x Z grid, each slice
+  // covering >=8 pixel rounds of 4 rows
+  long cb = (C + 63) / 64;
+  long want = (1024 + cb - 1) / cb;
+  long per = P / (4 * 8);
   long z = want < per ? want : per;
   if (z < 1) z = 1;
-  if (z > 64) z = 64;
+  if (z > 128) z = 128;
   return (int)z;
+}
+
+inline unsigned ew_grid(long total) {
+  long wgs = (total / 8 + 255) / 256;
+  if (wgs > 8192) wgs = 8192;
+  if (wgs < 1) wgs = 1;
+  return (unsigned)wgs;
 }
 
 }  // namespace
 
 void launch_bn_fwd(const bf16_t* x, const float* g, const float* b,
                    bf16_t* y, float* mean, float* invstd, float* part,
-                   int N, int C, long HW, int Z, float eps, bool relu,
+                   long P, int C, int Z, float eps, bool relu,
                    hipStream_t stream) {
-  dim3 sg(C, Z), sb(256);
+  dim3 sg(ceil_div(C, 64), Z), sb(256);
   hipLaunchKernelGGL(bn_stats_kernel, sg, sb, 0, stream, (const __bf16*)x,
-                     part, N, C, HW, Z);
+                     part, P, C, Z);
   hipLaunchKernelGGL(bn_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256),
-                     0, stream, part, mean, invstd, C, Z,
-                     1.f / ((float)N * HW), eps);
-  dim3 ag((unsigned)((long)N * C)), ab(256);
-  if (relu)
-    hipLaunchKernelGGL((bn_apply_kernel<true>), ag, ab, 0, stream,
-                       (const __bf16*)x, mean, invstd, g, b, (__bf16*)y, C, HW);
-  else
-    hipLaunchKernelGGL((bn_apply_kernel<false>), ag, ab, 0, stream,
-                       (const __bf16*)x, mean, invstd, g, b, (__bf16*)y, C, HW);
+                     0, stream, part, mean, invstd, C, Z, 1.f / (float)P,
+                     eps);
+  dim3 ag(ew_grid(P * (long)C)), ab(256);
+  const bool vec = (C & 7) == 0;
+#define APPLY(RELUv, VECv)                                                  \
+  hipLaunchKernelGGL((bn_apply_kernel<RELUv, VECv>), ag, ab, 0, stream,     \
+                     (const __bf16*)x, mean, invstd, g, b, (__bf16*)y, P, C)
+  if (relu) { if (vec) APPLY(true, true); else APPLY(true, false); }
+  else      { if (vec) APPLY(false, true); else APPLY(false, false); }
+#undef APPLY
 }
 
 void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, const bf16_t* y,
                    const float* g, const float* mean, const float* invstd,
                    bf16_t* dx, float* dgamma, float* dbeta, float* part,
-                   float* s1n, float* s2n, int N, int C, long HW, int Z,
+                   float* s1n, float* s2n, long P, int C, int Z,
                    bool relu, hipStream_t stream) {
-  dim3 sg(C, Z), sb(256);
+  dim3 sg(ceil_div(C, 64), Z), sb(256);
   if (relu)
     hipLaunchKernelGGL((bn_bwd_stats_kernel<true>), sg, sb, 0, stream,
                        (const __bf16*)x, (const __bf16*)dy, (const __bf16*)y,
-                       mean, invstd, part, N, C, HW, Z);
+                       mean, invstd, part, P, C, Z);
   else
     hipLaunchKernelGGL((bn_bwd_stats_kernel<false>), sg, sb, 0, stream,
                        (const __bf16*)x, (const __bf16*)dy, (const __bf16*)y,
-                       mean, invstd, part, N, C, HW, Z);
+                       mean, invstd, part, P, C, Z);
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(ceil_div(C, 256)),
                      dim3(256), 0, stream, part, dgamma, dbeta, s1n, s2n, C,
-                     Z, 1.f / ((float)N * HW));
-  dim3 ag((unsigned)((long)N * C)), ab(256);
-  if (relu)
-    hipLaunchKernelGGL((bn_bwd_apply_kernel<true>), ag, ab, 0, stream,
-                       (const __bf16*)x, (const __bf16*)dy, (const __bf16*)y,
-                       mean, invstd, g, s1n, s2n, (__bf16*)dx, C, HW);
-  else
-    hipLaunchKernelGGL((bn_bwd_apply_kernel<false>), ag, ab, 0, stream,
-                       (const __bf16*)x, (const __bf16*)dy, (const __bf16*)y,
-                       mean, invstd, g, s1n, s2n, (__bf16*)dx, C, HW);
+                     Z, 1.f / (float)P);
+  dim3 ag(ew_grid(P * (long)C)), ab(256);
+  const bool vec = (C & 7) == 0;
+#define APPLY(RELUv, VECv)                                                  \
+  hipLaunchKernelGGL((bn_bwd_apply_kernel<RELUv, VECv>), ag, ab, 0, stream, \
+                     (const __bf16*)x, (const __bf16*)dy, (const __bf16*)y,  \
+                     mean, invstd, g, s1n, s2n, (__bf16*)dx, P, C)
+  if (relu) { if (vec) APPLY(true, true); else APPLY(true, false); }
+  else      { if (vec) APPLY(false, true); else APPLY(false, false); }
+#undef APPLY
 }
 
-int bn_stats_slices(int N, int C, long HW) { return stats_slices(N, C, HW); }
+int bn_stats_slices(long P, int C) { return stats_slices(P, C); }
+int bn_max_channels() { return MAXC; }

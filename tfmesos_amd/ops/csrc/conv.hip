@@ -1,26 +1,27 @@
-// Implicit-GEMM 2-D convolution on MFMA (bf16 in, fp32 accumulate).
+// Implicit-GEMM 2-D convolution on MFMA, channels-last (NHWC), bf16 in,
+// fp32 accumulate.
 //
 // The Inception-v3 config of BASELINE.json needs conv fwd/bwd on the
 // worker compute path (the reference delegated convs to TF kernels;
 // SURVEY.md §2b row "conv stack"). CDNA4-native design: the conv is a
-// GEMM over [output pixels] x [filter taps], with the im2col patch
-// gather fused into the LDS staging (no materialized im2col buffer —
-// Inception's mid layers would need >100 GB of col data), and the same
-// v_mfma_f32_16x16x32_bf16 64x64 core as gemm.hip. NCHW layout.
+// GEMM over [output pixels] x [filter taps] with the im2col gather fused
+// into LDS staging (no materialized col buffer), the same
+// v_mfma_f32_16x16x32_bf16 64x64 core as gemm.hip, and CHANNELS-LAST
+// memory so every gather is a contiguous run of C (the profile of the
+// NCHW version, profiles/r01_inception_n1_kernel_stats.txt, showed the
+// channel-strided scalar gathers capping the kernels at ~1.5% MFMA
+// peak). Tap order is (r, s, c) — c innermost — so an 8-element strip
+// is ONE aligned b128 load whenever C % 8 == 0 (every Inception layer
+// except the 3-channel stem input; scalar fallback otherwise), and the
+// output store is coalesced across channels.
 //
-//   fwd:    Y[n,k,ho,wo]  = sum_{c,r,s} X[n,c,ho*U+r-P, wo*V+s-Q] W[k,c,r,s]
-//           GEMM  M = N*Ho*Wo, Ncol = K,     Kdim = C*R*S
-//   bwd-d:  dX[n,c,h,w]   = sum_{k,r,s} dY[n,k,(h+P-r)/U,(w+Q-s)/V] W[k,c,r,s]
-//           GEMM  M = N*H*W,   Ncol = C,     Kdim = K*R*S   (U,V-divisible taps)
-//   bwd-w:  dW[k,c,r,s]   = sum_{n,ho,wo} dY[n,k,ho,wo] X[n,c,ho*U+r-P,...]
-//           GEMM  M = K,       Ncol = C*R*S, Kdim = N*Ho*Wo
-//
-// Staging performance: every gather decodes its pixel index ONCE per
-// 8-element strip (integer division is the expensive op — the first,
-// correctness-only version did 8 long divisions per strip and ran at
-// <1% MFMA peak, profiles/r01_inception_n1_kernel_stats.txt), then
-// walks contiguous addresses; full-row interior strips take a straight
-// pointer walk, boundary strips an incremental-carry path.
+//   fwd:    Y[n,ho,wo,k]  = sum_{r,s,c} X[n, ho*U+r-P, wo*V+s-Q, c] W[k,r,s,c]
+//           GEMM  M = N*Ho*Wo, Ncol = K,     Kdim = R*S*C
+//   bwd-d:  dX[n,h,w,c]   = sum_{r,s,k} dY[n,(h+P-r)/U,(w+Q-s)/V,k] W[k,r,s,c]
+//           GEMM  M = N*H*W,   Ncol = C,     Kdim = R*S*K  (wants W^T memory
+//           [c][r][s][k], built by the host wrapper)
+//   bwd-w:  dW[k,r,s,c]   = sum_{n,ho,wo} dY[n,ho,wo,k] X[n,ho*U+r-P,...,c]
+//           GEMM  M = K,       Ncol = R*S*C, Kdim = N*Ho*Wo
 #include "common.h"
 
 namespace {
@@ -29,85 +30,92 @@ constexpr int BM = 64, BN = 64, BK = 32;
 constexpr int APAD = 8;
 
 struct ConvShape {
-  int N, C, H, W;     // input
+  int N, C, H, W;     // input (NHWC memory)
   int K, R, S;        // filter
   int Ho, Wo;         // output
   int U, V;           // stride
   int P, Q;           // pad
 };
 
-// ---------------------------------------------------------------- forward
-
-// Stage a 64(m) x 32(k) patch tile: m = output pixel, k = (c,r,s) tap.
-// Thread t loads 8 consecutive m (-> consecutive wo) for one tap.
-DEVINL void stage_patch_fwd(const __bf16* __restrict__ X, __bf16 (*Sm)[BK + APAD],
-                            const ConvShape cs, long m0, int k0, long M,
-                            int CRS, int t) {
-  const int kk = t >> 3;          // 0..31
-  const int mm0 = (t & 7) * 8;
-  const int q = k0 + kk;
-  const long pm0 = m0 + mm0;
-  if (q >= CRS || pm0 >= M) {
+// Load an 8-element channel run X[pix_base + c0 .. c0+8) into S[row][..],
+// vectorized when aligned.
+DEVINL void put8(__bf16* dstrow, const __bf16* src, bool vec) {
+  if (vec) {
+    *(bf16x8*)dstrow = *(const bf16x8*)src;
+  } else {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) Sm[mm0 + j][kk] = (__bf16)0.f;
-    return;
-  }
-  const int c = q / (cs.R * cs.S);
-  const int rs = q - c * (cs.R * cs.S);
-  const int r = rs / cs.S;
-  const int s = rs - r * cs.S;
-  // decode pixel once (32-bit divisions; M < 2^31 enforced by host)
-  const int HoWo = cs.Ho * cs.Wo;
-  int n = (int)(pm0 / HoWo);
-  int rem = (int)(pm0 - (long)n * HoWo);
-  int ho = rem / cs.Wo;
-  int wo = rem - ho * cs.Wo;
-  const int hi = ho * cs.U + r - cs.P;
-  const int wi0 = wo * cs.V + s - cs.Q;
-
-  if (wo + 8 <= cs.Wo && pm0 + 8 <= M &&
-      hi >= 0 && hi < cs.H && wi0 >= 0 && wi0 + 7 * cs.V < cs.W) {
-    // fast path: one row, fully interior — straight pointer walk
-    const __bf16* src = X + (((long)n * cs.C + c) * cs.H + hi) * cs.W + wi0;
-    if (cs.V == 1) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) Sm[mm0 + j][kk] = src[j];
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) Sm[mm0 + j][kk] = src[j * cs.V];
-    }
-    return;
-  }
-  // slow path: incremental carry across wo/ho/n, per-element bounds
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    float v = 0.f;
-    if (pm0 + j < M) {
-      const int hij = ho * cs.U + r - cs.P;
-      const int wij = wo * cs.V + s - cs.Q;
-      if (hij >= 0 && hij < cs.H && wij >= 0 && wij < cs.W)
-        v = (float)X[(((long)n * cs.C + c) * cs.H + hij) * cs.W + wij];
-    }
-    Sm[mm0 + j][kk] = (__bf16)v;
-    if (++wo == cs.Wo) { wo = 0; if (++ho == cs.Ho) { ho = 0; ++n; } }
+    for (int j = 0; j < 8; ++j) dstrow[j] = src[j];
   }
 }
 
-// Stage the weight tile Bs[kchan][tap]: W flat [K][CRS], row-contiguous.
-DEVINL void stage_wtile(const __bf16* __restrict__ Wt, __bf16 (*Sn)[BK + APAD],
-                        int n0, int k0, int K, int CRS, int t) {
+// ---------------------------------------------------------------- forward
+
+// As[m][kk]: m = output pixel, kk = tap (r,s,c) with c innermost.
+// Thread t: pixel row m = t>>2, 8-tap chunk kk0 = (t&3)*8 -> one b128.
+DEVINL void stage_patch_fwd(const __bf16* __restrict__ X, __bf16 (*Sm)[BK + APAD],
+                            const ConvShape cs, long m0, int k0, long M,
+                            int KD, int t, bool cvec) {
+  const int mx = t >> 2;
+  const int kk0 = (t & 3) * 8;
+  const int q0 = k0 + kk0;
+  const long pm = m0 + mx;
+  if (q0 >= KD || pm >= M) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) Sm[mx][kk0 + j] = (__bf16)0.f;
+    return;
+  }
+  const int rs = q0 / cs.C;            // tap (r,s) block (c0 % 8 == 0 when
+  const int c0 = q0 - rs * cs.C;       //  C % 8 == 0, so the run stays
+  const int r = rs / cs.S;             //  inside one (r,s))
+  const int s = rs - r * cs.S;
+  const int HoWo = cs.Ho * cs.Wo;
+  const int n = (int)(pm / HoWo);
+  const int rem = (int)(pm - (long)n * HoWo);
+  const int ho = rem / cs.Wo;
+  const int wo = rem - ho * cs.Wo;
+  const int hi = ho * cs.U + r - cs.P;
+  const int wi = wo * cs.V + s - cs.Q;
+
+  if (cvec && c0 + 8 <= cs.C) {
+    if (hi >= 0 && hi < cs.H && wi >= 0 && wi < cs.W) {
+      const __bf16* src = X + (((long)n * cs.H + hi) * cs.W + wi) * cs.C + c0;
+      put8(&Sm[mx][kk0], src, true);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) Sm[mx][kk0 + j] = (__bf16)0.f;
+    }
+    return;
+  }
+  // generic path: the 8 taps may straddle (r,s) blocks (small C)
+  int c = c0, rr = r, ss = s;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    float v = 0.f;
+    if (q0 + j < KD) {
+      const int hij = ho * cs.U + rr - cs.P;
+      const int wij = wo * cs.V + ss - cs.Q;
+      if (hij >= 0 && hij < cs.H && wij >= 0 && wij < cs.W)
+        v = (float)X[(((long)n * cs.H + hij) * cs.W + wij) * cs.C + c];
+    }
+    Sm[mx][kk0 + j] = (__bf16)v;
+    if (++c == cs.C) { c = 0; if (++ss == cs.S) { ss = 0; ++rr; } }
+  }
+}
+
+// Bs[k][kk]: weight memory [K][R*S*C] rows contiguous in tap order.
+DEVINL void stage_wrows(const __bf16* __restrict__ Wt, __bf16 (*Sn)[BK + APAD],
+                        int n0, int k0, int NROWS, int KD, int t, bool vec) {
   const int x = t >> 2;
   const int kk0 = (t & 3) * 8;
   const int gx = n0 + x;
   const int gk = k0 + kk0;
-  const __bf16* src = Wt + (long)gx * CRS + gk;
-  if (gx < K && gk + 8 <= CRS) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) Sn[x][kk0 + j] = src[j];
+  const __bf16* src = Wt + (long)gx * KD + gk;
+  if (gx < NROWS && gk + 8 <= KD) {
+    put8(&Sn[x][kk0], src, vec && ((gk & 7) == 0) && ((KD & 7) == 0));
   } else {
 #pragma unroll
     for (int j = 0; j < 8; ++j)
-      Sn[x][kk0 + j] = (gx < K && gk + j < CRS) ? src[j] : (__bf16)0.f;
+      Sn[x][kk0 + j] = (gx < NROWS && gk + j < KD) ? src[j] : (__bf16)0.f;
   }
 }
 
@@ -119,7 +127,8 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
   __shared__ __align__(16) __bf16 As[BM][BK + APAD];
   __shared__ __align__(16) __bf16 Bs[BN][BK + APAD];
   const long M = (long)cs.N * cs.Ho * cs.Wo;
-  const int CRS = cs.C * cs.R * cs.S;
+  const int KD = cs.R * cs.S * cs.C;
+  const bool cvec = (cs.C & 7) == 0;
   const long tm0 = (long)blockIdx.x * BM;   // pixel tiles ride grid.x (2^31)
   const int tn0 = blockIdx.y * BN;
   const int t = threadIdx.x;
@@ -128,9 +137,9 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
   const int wr = wave >> 1, wc = wave & 1;
 
   f32x4 acc[2][2] = {};
-  for (int k0 = 0; k0 < CRS; k0 += BK) {
-    stage_patch_fwd(X, As, cs, tm0, k0, M, CRS, t);
-    stage_wtile(Wt, Bs, tn0, k0, cs.K, CRS, t);
+  for (int k0 = 0; k0 < KD; k0 += BK) {
+    stage_patch_fwd(X, As, cs, tm0, k0, M, KD, t, cvec);
+    stage_wrows(Wt, Bs, tn0, k0, cs.K, KD, t, true);
     __syncthreads();
     const int kfrag = (lane >> 4) * 8;
 #pragma unroll
@@ -146,7 +155,7 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
     __syncthreads();
   }
 
-  // epilogue: Y[n,k,ho,wo] (col = k channel, row = output pixel)
+  // epilogue: Y[n,ho,wo,k] — k contiguous across lanes (coalesced)
 #pragma unroll
   for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
@@ -158,103 +167,85 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
       for (int rr = 0; rr < 4; ++rr) {
         const long pm = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
         if (pm >= M) continue;
-        const int n = (int)(pm / (cs.Ho * cs.Wo));
-        const int rem = (int)(pm % (cs.Ho * cs.Wo));
         float v = acc[fm][fn][rr] + bv;
         if (RELU) v = v > 0.f ? v : 0.f;
-        Y[(((long)n * cs.K + k) * cs.Ho + rem / cs.Wo) * cs.Wo + rem % cs.Wo] =
-            (__bf16)v;
+        Y[pm * cs.K + k] = (__bf16)v;
       }
     }
 }
 
 // --------------------------------------------------------------- bwd-data
 
-// m = input pixel (n,h,w); tap q = (k,r,s); contributes when
-// (h+P-r) % U == 0 and in range (same for w). STRIDE1 specializes the
-// common U==V==1 case (no divisibility tests, contiguous fast path).
+// As[m][kk]: m = input pixel, kk = tap (r,s,k) with k innermost, reading
+// dY[n, (h+P-r)/U, (w+Q-s)/V, k] (contiguous in k). STRIDE1 removes the
+// divisibility tests.
 template <bool STRIDE1>
 DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
                              __bf16 (*Sm)[BK + APAD], const ConvShape cs,
-                             long m0, int k0, long M, int KRS, int t) {
-  const int kk = t >> 3;
-  const int mm0 = (t & 7) * 8;
-  const int q = k0 + kk;
-  const long pm0 = m0 + mm0;
-  if (q >= KRS || pm0 >= M) {
+                             long m0, int k0, long M, int KD, int t,
+                             bool kvec) {
+  const int mx = t >> 2;
+  const int kk0 = (t & 3) * 8;
+  const int q0 = k0 + kk0;
+  const long pm = m0 + mx;
+  if (q0 >= KD || pm >= M) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) Sm[mm0 + j][kk] = (__bf16)0.f;
+    for (int j = 0; j < 8; ++j) Sm[mx][kk0 + j] = (__bf16)0.f;
     return;
   }
-  const int k = q / (cs.R * cs.S);
-  const int rs = q - k * (cs.R * cs.S);
+  const int rs = q0 / cs.K;
+  const int kc0 = q0 - rs * cs.K;
   const int r = rs / cs.S;
   const int s = rs - r * cs.S;
   const int HWi = cs.H * cs.W;
-  int n = (int)(pm0 / HWi);
-  int rem = (int)(pm0 - (long)n * HWi);
-  int h = rem / cs.W;
-  int w = rem - h * cs.W;
+  const int n = (int)(pm / HWi);
+  const int rem = (int)(pm - (long)n * HWi);
+  const int h = rem / cs.W;
+  const int w = rem - h * cs.W;
+  const int hn = h + cs.P - r, wn = w + cs.Q - s;
 
-  if (STRIDE1) {
-    const int ho = h + cs.P - r;
-    const int wo0 = w + cs.Q - s;
-    if (w + 8 <= cs.W && pm0 + 8 <= M &&
-        ho >= 0 && ho < cs.Ho && wo0 >= 0 && wo0 + 8 <= cs.Wo) {
+  if (kvec && kc0 + 8 <= cs.K) {
+    bool ok;
+    int ho, wo;
+    if (STRIDE1) {
+      ho = hn; wo = wn;
+      ok = hn >= 0 && hn < cs.Ho && wn >= 0 && wn < cs.Wo;
+    } else {
+      ok = hn >= 0 && wn >= 0 && hn % cs.U == 0 && wn % cs.V == 0;
+      ho = hn / cs.U; wo = wn / cs.V;
+      ok = ok && ho < cs.Ho && wo < cs.Wo;
+    }
+    if (ok) {
       const __bf16* src =
-          dY + (((long)n * cs.K + k) * cs.Ho + ho) * cs.Wo + wo0;
+          dY + (((long)n * cs.Ho + ho) * cs.Wo + wo) * cs.K + kc0;
+      put8(&Sm[mx][kk0], src, true);
+    } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) Sm[mm0 + j][kk] = src[j];
-      return;
+      for (int j = 0; j < 8; ++j) Sm[mx][kk0 + j] = (__bf16)0.f;
     }
-  }
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    float v = 0.f;
-    if (pm0 + j < M) {
-      const int hn = h + cs.P - r, wn = w + cs.Q - s;
-      if (STRIDE1) {
-        if (hn >= 0 && hn < cs.Ho && wn >= 0 && wn < cs.Wo)
-          v = (float)dY[(((long)n * cs.K + k) * cs.Ho + hn) * cs.Wo + wn];
-      } else {
-        if (hn >= 0 && wn >= 0 && hn % cs.U == 0 && wn % cs.V == 0) {
-          const int ho = hn / cs.U, wo = wn / cs.V;
-          if (ho < cs.Ho && wo < cs.Wo)
-            v = (float)dY[(((long)n * cs.K + k) * cs.Ho + ho) * cs.Wo + wo];
-        }
-      }
-    }
-    Sm[mm0 + j][kk] = (__bf16)v;
-    if (++w == cs.W) { w = 0; if (++h == cs.H) { h = 0; ++n; } }
-  }
-}
-
-// Weight tile for bwd-data: Bs[c][tap(k,r,s)] = W[k,c,r,s] (strided gather;
-// taps decode incrementally — consecutive q walk s, then r, then k).
-DEVINL void stage_wtile_bwdd(const __bf16* __restrict__ Wt,
-                             __bf16 (*Sn)[BK + APAD], const ConvShape cs,
-                             int n0, int k0, int KRS, int t) {
-  const int x = t >> 2;          // c offset 0..63
-  const int kk0 = (t & 3) * 8;
-  const int c = n0 + x;
-  const int q0 = k0 + kk0;
-  if (c >= cs.C || q0 >= KRS) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) Sn[x][kk0 + j] = (__bf16)0.f;
     return;
   }
-  const int RS = cs.R * cs.S;
-  int k = q0 / RS;
-  int rs = q0 - k * RS;
-  int r = rs / cs.S;
-  int s = rs - r * cs.S;
+  int kc = kc0, rr2 = r, ss2 = s;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     float v = 0.f;
-    if (q0 + j < KRS)
-      v = (float)Wt[(((long)k * cs.C + c) * cs.R + r) * cs.S + s];
-    Sn[x][kk0 + j] = (__bf16)v;
-    if (++s == cs.S) { s = 0; if (++r == cs.R) { r = 0; ++k; } }
+    if (q0 + j < KD) {
+      const int hnj = h + cs.P - rr2, wnj = w + cs.Q - ss2;
+      bool ok;
+      int ho, wo;
+      if (STRIDE1) {
+        ho = hnj; wo = wnj;
+        ok = hnj >= 0 && hnj < cs.Ho && wnj >= 0 && wnj < cs.Wo;
+      } else {
+        ok = hnj >= 0 && wnj >= 0 && hnj % cs.U == 0 && wnj % cs.V == 0;
+        ho = hnj / cs.U; wo = wnj / cs.V;
+        ok = ok && ho < cs.Ho && wo < cs.Wo;
+      }
+      if (ok)
+        v = (float)dY[(((long)n * cs.Ho + ho) * cs.Wo + wo) * cs.K + kc];
+    }
+    Sm[mx][kk0 + j] = (__bf16)v;
+    if (++kc == cs.K) { kc = 0; if (++ss2 == cs.S) { ss2 = 0; ++rr2; } }
   }
 }
 
@@ -262,11 +253,13 @@ template <bool STRIDE1>
 __global__ __launch_bounds__(256)
 void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ Wt,
                       __bf16* __restrict__ dX, ConvShape cs) {
+  // Wt memory: [C][R*S*K] (host-permuted W^T)
   __shared__ __align__(16) __bf16 As[BM][BK + APAD];
   __shared__ __align__(16) __bf16 Bs[BN][BK + APAD];
   const long M = (long)cs.N * cs.H * cs.W;
-  const int KRS = cs.K * cs.R * cs.S;
-  const long tm0 = (long)blockIdx.x * BM;   // pixel tiles ride grid.x (2^31)
+  const int KD = cs.R * cs.S * cs.K;
+  const bool kvec = (cs.K & 7) == 0;
+  const long tm0 = (long)blockIdx.x * BM;
   const int tn0 = blockIdx.y * BN;
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -274,9 +267,9 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
   const int wr = wave >> 1, wc = wave & 1;
 
   f32x4 acc[2][2] = {};
-  for (int k0 = 0; k0 < KRS; k0 += BK) {
-    stage_patch_bwdd<STRIDE1>(dY, As, cs, tm0, k0, M, KRS, t);
-    stage_wtile_bwdd(Wt, Bs, cs, tn0, k0, KRS, t);
+  for (int k0 = 0; k0 < KD; k0 += BK) {
+    stage_patch_bwdd<STRIDE1>(dY, As, cs, tm0, k0, M, KD, t, kvec);
+    stage_wrows(Wt, Bs, tn0, k0, cs.C, KD, t, kvec);
     __syncthreads();
     const int kfrag = (lane >> 4) * 8;
 #pragma unroll
@@ -302,109 +295,102 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
       for (int rr = 0; rr < 4; ++rr) {
         const long pm = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
         if (pm >= M) continue;
-        const int n = (int)(pm / ((long)cs.H * cs.W));
-        const int rem = (int)(pm % ((long)cs.H * cs.W));
-        dX[(((long)n * cs.C + c) * cs.H + rem / cs.W) * cs.W + rem % cs.W] =
-            (__bf16)acc[fm][fn][rr];
+        dX[pm * cs.C + c] = (__bf16)acc[fm][fn][rr];
       }
     }
 }
 
 // -------------------------------------------------------------- bwd-weight
 
-// GEMM: rows m = output channel k (M=K), cols = (c,r,s), Kdim = N*Ho*Wo.
-// As[k][p] = dY[n,k,ho,wo]; thread t loads 8 consecutive p for one k
-// (contiguous in dY within an image).
+// As[k][p]: dY[n,ho,wo,k] — thread t loads 8 consecutive k (one b128)
+// for one reduction pixel p, transposing into LDS.
 DEVINL void stage_dy_bwdw(const __bf16* __restrict__ dY, __bf16 (*Sm)[BK + APAD],
                           const ConvShape cs, int m0, long p0, long Ptot,
-                          int t) {
-  const int pp0 = (t & 3) * 8;     // p offset 0..24
-  const int kx = t >> 2;           // k-channel row 0..63
-  const int k = m0 + kx;
-  const long p = p0 + pp0;
-  const long HoWo = (long)cs.Ho * cs.Wo;
-  if (k >= cs.K || p >= Ptot) {
+                          int t, bool kvec) {
+  const int px = t >> 3;           // 0..31 reduction pixel
+  const int kk0 = (t & 7) * 8;     // k chunk
+  const long p = p0 + px;
+  const int k0 = m0 + kk0;
+  if (p >= Ptot || k0 >= cs.K) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) Sm[kx][pp0 + j] = (__bf16)0.f;
+    for (int j = 0; j < 8; ++j) Sm[kk0 + j][px] = (__bf16)0.f;
     return;
   }
-  int n = (int)(p / HoWo);
-  int rem = (int)(p - (long)n * HoWo);
-  if (p + 8 <= Ptot && rem + 8 <= (int)HoWo) {
-    const __bf16* src = dY + ((long)n * cs.K + k) * HoWo + rem;
+  const __bf16* src = dY + p * cs.K + k0;
+  if (kvec && k0 + 8 <= cs.K) {
+    bf16x8 v = *(const bf16x8*)src;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) Sm[kx][pp0 + j] = src[j];
-    return;
-  }
+    for (int j = 0; j < 8; ++j) Sm[kk0 + j][px] = v[j];
+  } else {
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    float v = 0.f;
-    if (p + j < Ptot)
-      v = (float)dY[((long)n * cs.K + k) * HoWo + rem];
-    Sm[kx][pp0 + j] = (__bf16)v;
-    if (++rem == (int)HoWo) { rem = 0; ++n; }
+    for (int j = 0; j < 8; ++j)
+      Sm[kk0 + j][px] = (k0 + j < cs.K) ? src[j] : (__bf16)0.f;
   }
 }
 
-// Bs[crs][p] = X[n, c, ho*U+r-P, wo*V+s-Q]; 8 consecutive p for one tap.
+// Bs[tap(r,s,c)][p]: X[n, ho*U+r-P, wo*V+s-Q, c] — thread t loads 8
+// consecutive c (one b128) for one pixel, transposing into LDS.
 DEVINL void stage_x_bwdw(const __bf16* __restrict__ X, __bf16 (*Sn)[BK + APAD],
-                         const ConvShape cs, int n0, long p0, int CRS,
-                         long Ptot, int t) {
-  const int pp0 = (t & 3) * 8;
-  const int qx = t >> 2;           // tap row 0..63
-  const int q = n0 + qx;
-  const long p = p0 + pp0;
-  if (q >= CRS || p >= Ptot) {
+                         const ConvShape cs, int n0, long p0, int KD,
+                         long Ptot, int t, bool cvec) {
+  const int px = t >> 3;
+  const int qq0 = (t & 7) * 8;
+  const long p = p0 + px;
+  const int q0 = n0 + qq0;
+  if (p >= Ptot || q0 >= KD) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) Sn[qx][pp0 + j] = (__bf16)0.f;
+    for (int j = 0; j < 8; ++j) Sn[qq0 + j][px] = (__bf16)0.f;
     return;
   }
-  const int c = q / (cs.R * cs.S);
-  const int rs = q - c * (cs.R * cs.S);
+  const int rs = q0 / cs.C;
+  const int c0 = q0 - rs * cs.C;
   const int r = rs / cs.S;
   const int s = rs - r * cs.S;
   const long HoWo = (long)cs.Ho * cs.Wo;
-  int n = (int)(p / HoWo);
-  int rem = (int)(p - (long)n * HoWo);
-  int ho = rem / cs.Wo;
-  int wo = rem - ho * cs.Wo;
+  const int n = (int)(p / HoWo);
+  const int rem = (int)(p - (long)n * HoWo);
+  const int ho = rem / cs.Wo;
+  const int wo = rem - ho * cs.Wo;
   const int hi = ho * cs.U + r - cs.P;
-  const int wi0 = wo * cs.V + s - cs.Q;
-  if (p + 8 <= Ptot && wo + 8 <= cs.Wo &&
-      hi >= 0 && hi < cs.H && wi0 >= 0 && wi0 + 7 * cs.V < cs.W) {
-    const __bf16* src = X + (((long)n * cs.C + c) * cs.H + hi) * cs.W + wi0;
-    if (cs.V == 1) {
+  const int wi = wo * cs.V + s - cs.Q;
+  if (cvec && c0 + 8 <= cs.C) {
+    if (hi >= 0 && hi < cs.H && wi >= 0 && wi < cs.W) {
+      const __bf16* src = X + (((long)n * cs.H + hi) * cs.W + wi) * cs.C + c0;
+      bf16x8 v = *(const bf16x8*)src;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) Sn[qx][pp0 + j] = src[j];
+      for (int j = 0; j < 8; ++j) Sn[qq0 + j][px] = v[j];
     } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) Sn[qx][pp0 + j] = src[j * cs.V];
+      for (int j = 0; j < 8; ++j) Sn[qq0 + j][px] = (__bf16)0.f;
     }
     return;
   }
+  int c = c0, rr2 = r, ss2 = s;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     float v = 0.f;
-    if (p + j < Ptot) {
-      const int hij = ho * cs.U + r - cs.P;
-      const int wij = wo * cs.V + s - cs.Q;
+    if (q0 + j < KD) {
+      const int hij = ho * cs.U + rr2 - cs.P;
+      const int wij = wo * cs.V + ss2 - cs.Q;
       if (hij >= 0 && hij < cs.H && wij >= 0 && wij < cs.W)
-        v = (float)X[(((long)n * cs.C + c) * cs.H + hij) * cs.W + wij];
+        v = (float)X[(((long)n * cs.H + hij) * cs.W + wij) * cs.C + c];
     }
-    Sn[qx][pp0 + j] = (__bf16)v;
-    if (++wo == cs.Wo) { wo = 0; if (++ho == cs.Ho) { ho = 0; ++n; } }
+    Sn[qq0 + j][px] = (__bf16)v;
+    if (++c == cs.C) { c = 0; if (++ss2 == cs.S) { ss2 = 0; ++rr2; } }
   }
 }
 
-// dW fp32 out [K][C*R*S]; grid.z slices the huge N*Ho*Wo reduction and
+// dW fp32 out [K][R*S*C]; grid.z slices the huge N*Ho*Wo reduction and
 // accumulates with fp32 atomics (dW is zeroed by the launcher).
 __global__ __launch_bounds__(256)
 void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ X,
                       float* __restrict__ dW, ConvShape cs, long pc) {
   __shared__ __align__(16) __bf16 As[BM][BK + APAD];
   __shared__ __align__(16) __bf16 Bs[BN][BK + APAD];
-  const int CRS = cs.C * cs.R * cs.S;
+  const int KD = cs.R * cs.S * cs.C;
   const long Ptot = (long)cs.N * cs.Ho * cs.Wo;
+  const bool cvec = (cs.C & 7) == 0;
+  const bool kvec = (cs.K & 7) == 0;
   const int tm0 = blockIdx.y * BM;
   const int tn0 = blockIdx.x * BN;
   const long ps = (long)blockIdx.z * pc;
@@ -416,8 +402,8 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
 
   f32x4 acc[2][2] = {};
   for (long p0 = ps; p0 < pe; p0 += BK) {
-    stage_dy_bwdw(dY, As, cs, tm0, p0, Ptot, t);
-    stage_x_bwdw(X, Bs, cs, tn0, p0, CRS, Ptot, t);
+    stage_dy_bwdw(dY, As, cs, tm0, p0, Ptot, t, kvec);
+    stage_x_bwdw(X, Bs, cs, tn0, p0, KD, Ptot, t, cvec);
     __syncthreads();
     const int kfrag = (lane >> 4) * 8;
 #pragma unroll
@@ -438,15 +424,15 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
 #pragma unroll
     for (int fn = 0; fn < 2; ++fn) {
       const int q = tn0 + wc * 32 + fn * 16 + (lane & 15);
-      if (q >= CRS) continue;
+      if (q >= KD) continue;
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
         const int k = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
         if (k >= cs.K) continue;
         if (gridDim.z == 1)
-          dW[(long)k * CRS + q] = acc[fm][fn][rr];
+          dW[(long)k * KD + q] = acc[fm][fn][rr];
         else
-          unsafeAtomicAdd(&dW[(long)k * CRS + q], acc[fm][fn][rr]);
+          unsafeAtomicAdd(&dW[(long)k * KD + q], acc[fm][fn][rr]);
       }
     }
 }
@@ -482,20 +468,21 @@ void launch_conv_fwd(const bf16_t* X, const bf16_t* W, const float* bias,
   }
 }
 
-void launch_conv_bwd_data(const bf16_t* dY, const bf16_t* W, bf16_t* dX,
+void launch_conv_bwd_data(const bf16_t* dY, const bf16_t* Wt, bf16_t* dX,
                           int N, int C, int H, int Wd, int K, int R, int S,
                           int Ho, int Wo, int U, int V, int P, int Q,
                           hipStream_t stream) {
+  // Wt: host-permuted W^T, memory [C][R*S*K]
   ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q};
   const long M = (long)N * H * Wd;
   dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(C, BN));
   dim3 block(256);
   if (U == 1 && V == 1)
     hipLaunchKernelGGL((conv_bwdd_kernel<true>), grid, block, 0, stream,
-                       (const __bf16*)dY, (const __bf16*)W, (__bf16*)dX, cs);
+                       (const __bf16*)dY, (const __bf16*)Wt, (__bf16*)dX, cs);
   else
     hipLaunchKernelGGL((conv_bwdd_kernel<false>), grid, block, 0, stream,
-                       (const __bf16*)dY, (const __bf16*)W, (__bf16*)dX, cs);
+                       (const __bf16*)dY, (const __bf16*)Wt, (__bf16*)dX, cs);
 }
 
 void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
@@ -503,10 +490,10 @@ void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
                             int Ho, int Wo, int U, int V, int P, int Q,
                             hipStream_t stream) {
   ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q};
-  const int CRS = C * R * S;
+  const int KD = C * R * S;
   const long Ptot = (long)N * Ho * Wo;
   // slice the reduction so the grid can fill the chip (>=512 WGs)
-  const long tiles = (long)ceil_div(K, BM) * ceil_div(CRS, BN);
+  const long tiles = (long)ceil_div(K, BM) * ceil_div(KD, BN);
   long zmax = (Ptot + BK - 1) / BK;
   long zwant = 512 / tiles;
   if (zwant < 1) zwant = 1;
@@ -514,7 +501,7 @@ void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
   long pc = (Ptot + z - 1) / z;
   pc = (pc + BK - 1) / BK * BK;
   z = (int)((Ptot + pc - 1) / pc);
-  dim3 grid(ceil_div(CRS, BN), ceil_div(K, BM), z);
+  dim3 grid(ceil_div(KD, BN), ceil_div(K, BM), z);
   dim3 block(256);
   hipLaunchKernelGGL(conv_bwdw_kernel, grid, block, 0, stream,
                      (const __bf16*)dY, (const __bf16*)X, dW, cs, pc);

@@ -45,13 +45,14 @@ void launch_conv_bwd_weight(const bf16_t*, const bf16_t*, float*, int, int,
                             int, int, int, int, int, int, int, int, int,
                             int, int, hipStream_t);
 void launch_bn_fwd(const bf16_t*, const float*, const float*, bf16_t*,
-                   float*, float*, float*, int, int, long, int, float, bool,
+                   float*, float*, float*, long, int, int, float, bool,
                    hipStream_t);
 void launch_bn_bwd(const bf16_t*, const bf16_t*, const bf16_t*, const float*,
                    const float*, const float*, bf16_t*, float*, float*,
-                   float*, float*, float*, int, int, long, int, bool,
+                   float*, float*, float*, long, int, int, bool,
                    hipStream_t);
-int bn_stats_slices(int, int, long);
+int bn_stats_slices(long, int);
+int bn_max_channels();
 
 namespace {
 
@@ -366,58 +367,70 @@ torch::Tensor colsum(torch::Tensor x, torch::Tensor out) {
 
 // ------------------------------------------------------------------ conv
 
-static void conv_out_shape(const torch::Tensor& x, const torch::Tensor& w,
-                           long stride_h, long stride_w, long pad_h,
-                           long pad_w, int* Ho, int* Wo) {
-  *Ho = (int)((x.size(2) + 2 * pad_h - w.size(2)) / stride_h + 1);
-  *Wo = (int)((x.size(3) + 2 * pad_w - w.size(3)) / stride_w + 1);
+static bool is_cl(const torch::Tensor& t) {
+  return t.is_contiguous(torch::MemoryFormat::ChannelsLast);
 }
 
-torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
+// x: logical NCHW in channels-last memory; wm: [K,R,S,C] contiguous
+// (tap-major weight copy built by the Python wrapper). Returns y in
+// channels-last memory.
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor wm,
                          torch::Tensor bias, long stride_h, long stride_w,
                          long pad_h, long pad_w, bool relu) {
-  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous() &&
-              x.scalar_type() == torch::kBFloat16, "x must be bf16 NCHW");
-  TORCH_CHECK(w.is_cuda() && w.dim() == 4 && w.is_contiguous() &&
-              w.scalar_type() == torch::kBFloat16 && w.size(1) == x.size(1),
-              "w must be bf16 KCRS matching x channels");
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && is_cl(x) &&
+              x.scalar_type() == torch::kBFloat16,
+              "x must be bf16 NCHW in channels-last memory");
+  TORCH_CHECK(wm.is_cuda() && wm.dim() == 4 && wm.is_contiguous() &&
+              wm.scalar_type() == torch::kBFloat16 &&
+              wm.size(3) == x.size(1), "wm must be bf16 [K,R,S,C]");
+  const long K = wm.size(0), R = wm.size(1), S = wm.size(2);
   const float* bias_p = nullptr;
   if (bias.numel() > 0) {
-    TORCH_CHECK(bias.scalar_type() == torch::kFloat32 &&
-                bias.numel() == w.size(0), "bias must be fp32 [K]");
+    TORCH_CHECK(bias.scalar_type() == torch::kFloat32 && bias.numel() == K,
+                "bias must be fp32 [K]");
     bias_p = bias.data_ptr<float>();
   }
-  int Ho, Wo;
-  conv_out_shape(x, w, stride_h, stride_w, pad_h, pad_w, &Ho, &Wo);
-  auto y = torch::empty({x.size(0), w.size(0), Ho, Wo}, x.options());
-  launch_conv_fwd((const bf16_t*)x.data_ptr(), (const bf16_t*)w.data_ptr(),
+  const int Ho = (int)((x.size(2) + 2 * pad_h - R) / stride_h + 1);
+  const int Wo = (int)((x.size(3) + 2 * pad_w - S) / stride_w + 1);
+  auto y = torch::empty({x.size(0), K, Ho, Wo},
+                        x.options().memory_format(
+                            torch::MemoryFormat::ChannelsLast));
+  launch_conv_fwd((const bf16_t*)x.data_ptr(), (const bf16_t*)wm.data_ptr(),
                   bias_p, (bf16_t*)y.data_ptr(), x.size(0), x.size(1),
-                  x.size(2), x.size(3), w.size(0), w.size(2), w.size(3),
-                  Ho, Wo, stride_h, stride_w, pad_h, pad_w, relu,
-                  cur_stream());
+                  x.size(2), x.size(3), K, R, S, Ho, Wo, stride_h, stride_w,
+                  pad_h, pad_w, relu, cur_stream());
   return y;
 }
 
-torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor w,
+// dy channels-last; wt: [C,R,S,K] contiguous (W^T copy). Returns dx
+// channels-last.
+torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor wt,
                               long H, long W, long stride_h, long stride_w,
                               long pad_h, long pad_w) {
-  TORCH_CHECK(dy.is_cuda() && dy.dim() == 4 && dy.is_contiguous() &&
-              dy.scalar_type() == torch::kBFloat16, "dy must be bf16 NCHW");
-  auto dx = torch::empty({dy.size(0), w.size(1), H, W}, dy.options());
+  TORCH_CHECK(dy.is_cuda() && dy.dim() == 4 && is_cl(dy) &&
+              dy.scalar_type() == torch::kBFloat16,
+              "dy must be bf16 channels-last");
+  TORCH_CHECK(wt.is_contiguous() && wt.dim() == 4 &&
+              wt.size(3) == dy.size(1), "wt must be [C,R,S,K]");
+  const long C = wt.size(0), R = wt.size(1), S = wt.size(2);
+  auto dx = torch::empty({dy.size(0), C, H, W},
+                         dy.options().memory_format(
+                             torch::MemoryFormat::ChannelsLast));
   launch_conv_bwd_data((const bf16_t*)dy.data_ptr(),
-                       (const bf16_t*)w.data_ptr(), (bf16_t*)dx.data_ptr(),
-                       dy.size(0), w.size(1), H, W, w.size(0), w.size(2),
-                       w.size(3), dy.size(2), dy.size(3), stride_h, stride_w,
+                       (const bf16_t*)wt.data_ptr(), (bf16_t*)dx.data_ptr(),
+                       dy.size(0), C, H, W, dy.size(1), R, S,
+                       dy.size(2), dy.size(3), stride_h, stride_w,
                        pad_h, pad_w, cur_stream());
   return dx;
 }
 
+// Returns dW in tap-major [K,R,S,C] fp32 (wrapper permutes to KCRS).
 torch::Tensor conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
                                 long R, long S, long stride_h, long stride_w,
                                 long pad_h, long pad_w) {
-  TORCH_CHECK(dy.is_cuda() && x.is_cuda() && dy.is_contiguous() &&
-              x.is_contiguous(), "dy/x must be contiguous");
-  auto dw = torch::zeros({dy.size(1), x.size(1), R, S},
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda() && is_cl(dy) && is_cl(x),
+              "dy/x must be channels-last");
+  auto dw = torch::zeros({dy.size(1), R, S, x.size(1)},
                          x.options().dtype(torch::kFloat32));
   launch_conv_bwd_weight((const bf16_t*)dy.data_ptr(),
                          (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(),
@@ -431,14 +444,16 @@ torch::Tensor conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
 
 std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor g,
                                   torch::Tensor b, double eps, bool relu) {
-  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous() &&
-              x.scalar_type() == torch::kBFloat16, "x must be bf16 NCHW");
-  const int N = x.size(0), C = x.size(1);
-  const long HW = (long)x.size(2) * x.size(3);
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && is_cl(x) &&
+              x.scalar_type() == torch::kBFloat16,
+              "x must be bf16 channels-last");
+  const int C = x.size(1);
+  const long P = x.numel() / C;
+  TORCH_CHECK(C <= bn_max_channels(), "bn: C > LDS staging bound");
   TORCH_CHECK(g.numel() == C && b.numel() == C &&
               g.scalar_type() == torch::kFloat32 &&
               b.scalar_type() == torch::kFloat32, "g/b must be fp32 [C]");
-  const int Z = bn_stats_slices(N, C, HW);
+  const int Z = bn_stats_slices(P, C);
   auto opts = x.options().dtype(torch::kFloat32);
   auto y = torch::empty_like(x);
   auto mean = torch::empty({C}, opts);
@@ -447,7 +462,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor g,
   launch_bn_fwd((const bf16_t*)x.data_ptr(), g.data_ptr<float>(),
                 b.data_ptr<float>(), (bf16_t*)y.data_ptr(),
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                part.data_ptr<float>(), N, C, HW, Z, (float)eps, relu,
+                part.data_ptr<float>(), P, C, Z, (float)eps, relu,
                 cur_stream());
   return {y, mean, invstd};
 }
@@ -456,9 +471,11 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
                                   torch::Tensor y, torch::Tensor g,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   bool relu) {
-  const int N = x.size(0), C = x.size(1);
-  const long HW = (long)x.size(2) * x.size(3);
-  const int Z = bn_stats_slices(N, C, HW);
+  TORCH_CHECK(is_cl(x) && is_cl(dy) && is_cl(y),
+              "bn_bwd: channels-last tensors required");
+  const int C = x.size(1);
+  const long P = x.numel() / C;
+  const int Z = bn_stats_slices(P, C);
   auto opts = x.options().dtype(torch::kFloat32);
   auto dx = torch::empty_like(x);
   auto dgamma = torch::empty({C}, opts);
@@ -471,7 +488,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
                 (bf16_t*)dx.data_ptr(), dgamma.data_ptr<float>(),
                 dbeta.data_ptr<float>(), part.data_ptr<float>(),
-                s1n.data_ptr<float>(), s2n.data_ptr<float>(), N, C, HW, Z,
+                s1n.data_ptr<float>(), s2n.data_ptr<float>(), P, C, Z,
                 relu, cur_stream());
   return {dx, dgamma, dbeta};
 }
