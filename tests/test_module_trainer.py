@@ -24,10 +24,14 @@ class TinyCNN(nn.Module):
 def test_module_trainer_zero_copy_views():
     m = TinyCNN()
     tr = ModuleReplicaTrainer(m, hparams={"lr": 0.05})
-    # parameters alias the bf16 shadow; grads alias the flat grad buffer
+    # parameters alias the bf16 shadow; step() batch-copies autograd
+    # grads into flat grad views aligned with the store layout
     w = dict(m.named_parameters())["c1.conv.weight"]
     assert w.data.data_ptr() >= tr.store.flat_bf16.data_ptr()
-    assert w.grad.data_ptr() >= tr.t.flat_grad.data_ptr()
+    names = [n for n, _ in m.named_parameters()]
+    gv = tr._gviews[names.index("c1.conv.weight")]
+    assert gv.data_ptr() >= tr.t.flat_grad.data_ptr()
+    assert gv.shape == w.shape
     assert w.dtype == torch.bfloat16
 
 

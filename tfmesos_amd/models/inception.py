@@ -64,8 +64,9 @@ class BasicConv2d(nn.Module):
 
 
 def _avg_pool(x, k, stride=1, padding=1):
-    return torch.nn.functional.avg_pool2d(x.float(), k, stride=stride,
-                                          padding=padding).to(x.dtype)
+    # bf16 native (channels-last aten kernel); a 3x3 mean is safe in bf16
+    return torch.nn.functional.avg_pool2d(x, k, stride=stride,
+                                          padding=padding)
 
 
 def _max_pool(x, k, stride):
@@ -209,7 +210,7 @@ class InceptionV3(nn.Module):
         for m in self.mixed:
             x = m(x)
         x = torch.nn.functional.adaptive_avg_pool2d(x.float(), 1)
-        x = x.flatten(1).to(self.fc_w.dtype).contiguous()
+        x = x.flatten(1).to(self.fc_w.dtype).contiguous()  # [B, 2048]
         return ops.linear(x, self.fc_w, self.fc_b)
 
 

@@ -49,22 +49,32 @@ void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict__ part,
   }
 }
 
-__global__ void bn_finalize_kernel(const float* __restrict__ part,
-                                   float* __restrict__ mean,
-                                   float* __restrict__ invstd,
-                                   int C, int Z, float inv_count, float eps) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+// one wave per channel: lanes cover the Z partial slices in parallel
+// (a single-workgroup loop over Z was latency-bound at ~28 us)
+__global__ __launch_bounds__(64)
+void bn_finalize_kernel(const float* __restrict__ part,
+                        float* __restrict__ mean,
+                        float* __restrict__ invstd,
+                        int C, int Z, float inv_count, float eps) {
+  const int c = blockIdx.x;
+  const int l = threadIdx.x;
   float s = 0.f, q = 0.f;
-  for (int z = 0; z < Z; ++z) {
+  for (int z = l; z < Z; z += 64) {
     s += part[((long)z * C + c) * 2];
     q += part[((long)z * C + c) * 2 + 1];
   }
-  const float m = s * inv_count;
-  float var = q * inv_count - m * m;
-  if (var < 0.f) var = 0.f;
-  mean[c] = m;
-  invstd[c] = __frsqrt_rn(var + eps);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s += __shfl_xor(s, off, 64);
+    q += __shfl_xor(q, off, 64);
+  }
+  if (l == 0) {
+    const float m = s * inv_count;
+    float var = q * inv_count - m * m;
+    if (var < 0.f) var = 0.f;
+    mean[c] = m;
+    invstd[c] = __frsqrt_rn(var + eps);
+  }
 }
 
 template <bool RELU, bool VEC>
@@ -144,23 +154,31 @@ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
   }
 }
 
-__global__ void bn_bwd_finalize_kernel(const float* __restrict__ part,
-                                       float* __restrict__ dgamma,
-                                       float* __restrict__ dbeta,
-                                       float* __restrict__ s1n,
-                                       float* __restrict__ s2n,
-                                       int C, int Z, float inv_count) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+__global__ __launch_bounds__(64)
+void bn_bwd_finalize_kernel(const float* __restrict__ part,
+                            float* __restrict__ dgamma,
+                            float* __restrict__ dbeta,
+                            float* __restrict__ s1n,
+                            float* __restrict__ s2n,
+                            int C, int Z, float inv_count) {
+  const int c = blockIdx.x;
+  const int l = threadIdx.x;
   float s1 = 0.f, s2 = 0.f;
-  for (int z = 0; z < Z; ++z) {
+  for (int z = l; z < Z; z += 64) {
     s1 += part[((long)z * C + c) * 2];
     s2 += part[((long)z * C + c) * 2 + 1];
   }
-  dbeta[c] = s1;
-  dgamma[c] = s2;
-  s1n[c] = s1 * inv_count;
-  s2n[c] = s2 * inv_count;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s1 += __shfl_xor(s1, off, 64);
+    s2 += __shfl_xor(s2, off, 64);
+  }
+  if (l == 0) {
+    dbeta[c] = s1;
+    dgamma[c] = s2;
+    s1n[c] = s1 * inv_count;
+    s2n[c] = s2 * inv_count;
+  }
 }
 
 template <bool RELU, bool VEC>
@@ -246,7 +264,7 @@ void launch_bn_fwd(const bf16_t* x, const float* g, const float* b,
   dim3 sg(ceil_div(C, 64), Z), sb(256);
   hipLaunchKernelGGL(bn_stats_kernel, sg, sb, 0, stream, (const __bf16*)x,
                      part, P, C, Z);
-  hipLaunchKernelGGL(bn_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256),
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64),
                      0, stream, part, mean, invstd, C, Z, 1.f / (float)P,
                      eps);
   dim3 ag(ew_grid(P * (long)C)), ab(256);
@@ -273,9 +291,8 @@ void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, const bf16_t* y,
     hipLaunchKernelGGL((bn_bwd_stats_kernel<false>), sg, sb, 0, stream,
                        (const __bf16*)x, (const __bf16*)dy, (const __bf16*)y,
                        mean, invstd, part, P, C, Z);
-  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(ceil_div(C, 256)),
-                     dim3(256), 0, stream, part, dgamma, dbeta, s1n, s2n, C,
-                     Z, 1.f / (float)P);
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(64), 0, stream,
+                     part, dgamma, dbeta, s1n, s2n, C, Z, 1.f / (float)P);
   dim3 ag(ew_grid(P * (long)C)), ab(256);
   const bool vec = (C & 7) == 0;
 #define APPLY(RELUv, VECv)                                                  \

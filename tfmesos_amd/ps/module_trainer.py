@@ -31,16 +31,34 @@ class ModuleReplicaTrainer(object):
         self.roles = self.t.roles
         self.module = module
         module.to(device=self.device, dtype=torch.bfloat16)
+        self._params = []
+        self._gviews = []
         for name, p in module.named_parameters():
             p.data = self.t.store.view(name, bf16=True)
-            p.grad = self.t.grad_view(name)
-
-    def zero_grad(self):
+            self._params.append(p)
+            self._gviews.append(self.t.grad_view(name))
+        # grads stay None between steps: autograd then ASSIGNS (no
+        # per-tensor accumulate kernels); step() batch-copies them into
+        # the flat reduce buffer with a handful of foreach kernels
         self.t.flat_grad.zero_()
 
+    def zero_grad(self):
+        for p in self._params:
+            p.grad = None
+
     def step(self, grad_scale=None):
-        """Reduce (sharded) -> fused apply on PS -> broadcast shadows.
-        Module weights update in place via the shadow views."""
+        """Gather grads into the flat buffer (foreach copy), reduce
+        (sharded) -> fused apply on PS -> broadcast shadows. Module
+        weights update in place via the shadow views."""
+        views, grads = [], []
+        for p, gv in zip(self._params, self._gviews):
+            if p.grad is not None:
+                views.append(gv)
+                grads.append(p.grad)
+        if grads:
+            torch._foreach_copy_(views, grads)
+            for p in self._params:
+                p.grad = None
         return self.t.step(grad_scale=grad_scale)
 
     @property
