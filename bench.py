@@ -56,6 +56,12 @@ def run_mnist(args, device, rank, world):
             model.fwd_bwd(pview, x, y, trainer.grad_view)
         trainer.step()
 
+    if world == 1 and device.type == "cuda" and not args.no_graph:
+        # single-GPU hot path: capture the whole 8-kernel step as ONE
+        # hipGraph replay (no collectives inside)
+        from tfmesos_amd.utils.graphstep import GraphedStep
+        one_step = GraphedStep(one_step)
+
     return one_step, {
         "model": "mnist_replica_mlp_784x%dx10" % args.hidden,
         "global_batch": args.batch * roles.n_workers,
@@ -67,18 +73,33 @@ def run_mnist(args, device, rank, world):
 
 
 def run_nmf(args, device, rank, world):
-    """matrix_factorization config: rank-200 NMF with the factor tables
-    treated as embeddings (sparse push/pull path)."""
-    from tfmesos_amd.models.nmf import NMFWorkload
-    wl = NMFWorkload(n=args.nmf_n, rank=args.nmf_rank, device=device,
-                     lr=args.lr, seed=1234 + rank)
-    return wl.one_step, {
+    """matrix_factorization config: rank-200 NMF. world==1 runs the
+    dense local workload; world>1 runs the sparse-embedding PS path
+    (W rows on ps:0, H rows on ps:1, minibatch pull/push)."""
+    if world > 1:
+        from tfmesos_amd.models.nmf import SparseNMF
+        wl = SparseNMF(rank, world, device=device, n=args.nmf_n,
+                       factor_rank=args.nmf_rank, batch=args.nmf_batch,
+                       lr=args.lr)
+        n_ps = 2 if world > 2 else 1
+        par = "ps%d+w%d(sparse)" % (n_ps, world - n_ps)
+        gbatch = args.nmf_batch * (world - n_ps)
+    else:
+        from tfmesos_amd.models.nmf import NMFWorkload
+        wl = NMFWorkload(n=args.nmf_n, rank=args.nmf_rank, device=device,
+                         lr=args.lr, seed=1234 + rank)
+        par = "local"
+        gbatch = args.nmf_n
+    def one_step():
+        wl.one_step()
+    one_step.finalize = getattr(wl, "finalize", lambda: None)
+    return one_step, {
         "model": "nmf_%dx%d_rank%d" % (args.nmf_n, args.nmf_n, args.nmf_rank),
-        "global_batch": args.nmf_n,
-        "parallelism": "local" if world == 1 else "ps%d+w%d" % (2, world - 2),
+        "global_batch": gbatch,
+        "parallelism": par,
         "optimizer": "sgd",
         "lr": args.lr,
-        "sync": True,
+        "sync": False if world > 1 else True,
     }
 
 
@@ -129,9 +150,12 @@ def main():
     p.add_argument("--optimizer", default="sgd")
     p.add_argument("--num-ps", type=int, default=1,
                    help="PS shards (ranks 0..n_ps-1) when world>1")
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable hipGraph step capture (world==1)")
     p.add_argument("--lr", type=float, default=0.01)
     p.add_argument("--nmf-n", type=int, default=1000)
     p.add_argument("--nmf-rank", type=int, default=200)
+    p.add_argument("--nmf-batch", type=int, default=256)
     p.add_argument("--inc-batch", type=int, default=32)
     p.add_argument("--inc-size", type=int, default=299)
     p.add_argument("--classes", type=int, default=1000)
@@ -195,6 +219,7 @@ def main():
             "config": config,
         }
         print(json.dumps(out))
+    getattr(one_step, "finalize", lambda: None)()
     if world > 1:
         dist.destroy_process_group()
 

@@ -57,3 +57,90 @@ class NMFWorkload(object):
         self.W -= self.lr * dW
         self.H -= self.lr * dH
         return None
+
+
+class SparseNMF(object):
+    """Distributed minibatch NMF on the sparse-embedding PS
+    (BASELINE.json config "matrix_factorization sparse-embedding PS,
+    2-ps/6-worker"): W rows on ps:0, H rows on ps:1 (manual model
+    parallelism as in the reference, matrix_factorization.py:21-28);
+    each worker step pulls only the minibatch's factor rows, computes
+    the minibatch Frobenius gradient on MFMA GEMMs, and pushes sparse
+    row grads back (HIP gather/scatter-add on a GPU PS).
+    """
+
+    def __init__(self, rank, world, device="cpu", n=1000, factor_rank=200,
+                 batch=256, lr=0.05, seed=0):
+        from tfmesos_amd.ps.replica import init_distributed
+        init_distributed(device)
+        from tfmesos_amd.ps.sparse import (
+            EmbeddingTable, SparsePSServer, SparseWorkerClient,
+            make_sparse_pair_groups)
+
+        assert world > 1
+        self.rank, self.world = rank, world
+        self.device = torch.device(device)
+        self.n, self.r, self.batch, self.lr = n, factor_rank, batch, lr
+        n_ps = 2 if world > 2 else 1
+        ps_ranks = list(range(n_ps))
+        worker_ranks = list(range(n_ps, world))
+        self.is_ps = rank < n_ps
+        homes = {"W": 0, "H": ps_ranks[-1]}
+        groups = make_sparse_pair_groups(ps_ranks, worker_ranks)
+
+        # planted low-rank target factors (same seed everywhere)
+        g = torch.Generator().manual_seed(seed)
+        self.w0 = torch.rand(n, factor_rank, generator=g)
+        self.h0 = torch.rand(n, factor_rank, generator=g)
+        self.scale = 1.0 / factor_rank
+
+        if self.is_ps:
+            tables = []
+            for name, home in homes.items():
+                if home == rank:
+                    tables.append(EmbeddingTable(
+                        name, n, factor_rank, device=device, lr=lr,
+                        seed=seed + hash(name) % 1000))
+            self.server = SparsePSServer(rank, tables, worker_ranks, groups)
+            import threading
+            self._srv_thread = threading.Thread(target=self.server.serve,
+                                                daemon=True)
+            self._srv_thread.start()
+        else:
+            self.client = SparseWorkerClient(
+                rank, homes, {"W": factor_rank, "H": factor_rank}, groups,
+                device=device)
+            self._step_gen = torch.Generator().manual_seed(777 + rank)
+        self.losses = []
+
+    def one_step(self):
+        if self.is_ps:
+            return None
+        from tfmesos_amd import ops
+        b = self.batch
+        I = torch.randint(0, self.n, (b,), generator=self._step_gen)
+        J = torch.randint(0, self.n, (b,), generator=self._step_gen)
+        WI = self.client.pull("W", I)           # [b, r] bf16
+        HJ = self.client.pull("H", J)           # [b, r] bf16
+        X = (self.w0[I] @ self.h0[J].t() * self.scale).to(self.device)
+        if self.device.type == "cuda":
+            E = (ops.gemm_bias_act(WI, HJ, trans_b=True).float() - X)
+            Eb = E.to(torch.bfloat16)
+            dW = ops.gemm_bias_act(Eb, HJ).float() * 2.0
+            dH = ops.gemm_bias_act(Eb, WI, trans_a=True).float() * 2.0
+        else:
+            WIf, HJf = WI.float(), HJ.float()
+            E = WIf @ HJf.t() - X
+            dW = 2.0 * (E @ HJf)
+            dH = 2.0 * (E.t() @ WIf)
+        self.client.push("W", I, dW / self.batch)
+        self.client.push("H", J, dH / self.batch)
+        loss = float((E * E).mean())
+        self.losses.append(loss)
+        return loss
+
+    def finalize(self):
+        if self.is_ps:
+            self._srv_thread.join(timeout=60)
+        else:
+            self.client.done_all()
