@@ -444,3 +444,21 @@ def test_batch_norm_act_fwd_bwd(shape, relu):
         0.02 * gf.grad.abs().max() + 0.1
     assert (b.grad.cpu() - bff.grad).abs().max() < \
         0.02 * bff.grad.abs().max() + 0.1
+
+
+def test_mlp_head_fused_matches_composed():
+    """One-kernel classifier head == gemm + softmax_xent + masked dh."""
+    from tfmesos_amd import ops
+    torch.manual_seed(50)
+    B, H, C = 100, 100, 10
+    h = torch.relu(bf(torch.randn(B, H)))
+    w = bf(torch.randn(H, C) * 0.1)
+    b = bf(torch.randn(C) * 0.1)
+    y = torch.randint(0, C, (B,), device=DEV)
+    loss, dl, dh = ops.mlp_head_fused(h, w, b, y)
+    logits = ops.gemm_bias_act(h, w, b)
+    loss2, dl2 = ops.softmax_xent_fused(logits, y)
+    dh2 = ops.gemm_bias_act(dl2, w, trans_b=True, act="relu_bwd", aux=h)
+    assert abs(float(loss) - float(loss2)) < 1e-2
+    assert (dl.float() - dl2.float()).abs().max() < 1e-3
+    assert (dh.float() - dh2.float()).abs().max() < 2e-3

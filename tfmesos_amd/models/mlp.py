@@ -61,22 +61,26 @@ class MnistMLP(object):
         g: callable name -> grad view to fill (fp32)
         Returns mean loss (fp32 scalar tensor).
         """
+        B = x.shape[0]
         hid_w, hid_b = p("hid_w"), p("hid_b")
         sm_w, sm_b = p("sm_w"), p("sm_b")
 
-        # forward (2 kernels: split-K GEMM+bias+relu, GEMM+bias)
+        # fwd hidden layer: split-K GEMM + bias + relu
         h = ops.gemm_bias_act(x, hid_w, hid_b, act="relu")       # [B,H]
-        logits = ops.gemm_bias_act(h, sm_w, sm_b, act="none")    # [B,C]
-        # loss fwd + bwd fused into one kernel
-        loss, dlogits = ops.softmax_xent_fused(logits, y)         # [B,C] (/B)
-
-        # backward: 3 GEMMs with fused epilogues, fp32 grads written
-        # straight into the flat grad views (bias colsums ride the dW
-        # GEMMs; relu-bwd masking rides the dh GEMM)
+        if x.is_cuda and self.classes <= 16 and self.hidden <= 512 \
+                and B <= 4096:
+            # whole classifier head (fwd GEMM + softmax-xent fwd/bwd +
+            # dh GEMM with relu mask) in ONE kernel
+            loss, dlogits, dh = ops.mlp_head_fused(h, sm_w, sm_b, y)
+        else:
+            logits = ops.gemm_bias_act(h, sm_w, sm_b, act="none")
+            loss, dlogits = ops.softmax_xent_fused(logits, y)
+            dh = ops.gemm_bias_act(dlogits, sm_w, trans_b=True,
+                                   act="relu_bwd", aux=h)
+        # dW GEMMs with the bias-grad colsums fused in; fp32 grads
+        # written straight into the flat grad views
         ops.gemm_bias_act(h, dlogits, trans_a=True, out=g("sm_w"),
                           colsum_out=g("sm_b"))
-        dh = ops.gemm_bias_act(dlogits, sm_w, trans_b=True,
-                               act="relu_bwd", aux=h)             # [B,H]
         ops.gemm_bias_act(x, dh, trans_a=True, out=g("hid_w"),
                           colsum_out=g("hid_b"))
         return loss

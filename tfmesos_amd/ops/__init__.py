@@ -439,3 +439,23 @@ def batch_norm_act(x, weight, bias, eps=1e-3, relu=False):
     if not x.is_cuda:
         x = x.contiguous()
     return _BatchNormActFn.apply(x, weight, bias, eps, relu)
+
+
+def mlp_head_fused(h, w, b, labels, scale=None):
+    """Fused classifier head (one kernel): logits = h@w+b, softmax,
+    mean xent loss, dlogits = (p-onehot)*scale, and dh = (dlogits@w^T)
+    masked by h>0 (h is a relu output). Returns (loss, dlogits, dh).
+    GPU limits C<=16, H<=512, B<=4096; CPU reference otherwise."""
+    B = h.shape[0]
+    s = float(scale if scale is not None else 1.0 / B)
+    if h.is_cuda:
+        return _ext().mlp_head_fused(h, w, b, labels, s)
+    hf, wf = h.float(), w.float()
+    logits = hf @ wf + b.float()
+    probs = torch.softmax(logits, 1)
+    loss = torch.nn.functional.cross_entropy(logits, labels)
+    d = probs.clone()
+    d[torch.arange(B), labels] -= 1.0
+    d *= s
+    dh = (d @ wf.t()) * (hf > 0)
+    return loss, d.to(h.dtype), dh.to(h.dtype)

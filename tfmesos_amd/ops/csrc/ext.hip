@@ -24,6 +24,9 @@ void launch_softmax_xent_fused(const bf16_t*, const long*, bf16_t*, float*,
                                float, int, int, hipStream_t);
 void launch_softmax_xent_bwd(const bf16_t*, const long*, bf16_t*, float,
                              int, int, hipStream_t);
+void launch_mlp_head_fused(const bf16_t*, const bf16_t*, const bf16_t*,
+                           const long*, bf16_t*, bf16_t*, float*, float,
+                           int, int, int, hipStream_t);
 void launch_gather_bf16(const bf16_t*, const long*, bf16_t*, long, int, long,
                         hipStream_t);
 void launch_gather_f32(const float*, const long*, float*, long, int, long,
@@ -290,6 +293,33 @@ std::tuple<torch::Tensor, torch::Tensor> softmax_xent_fused(
   return {loss, dlogits};
 }
 
+// fused classifier head: logits=h@w+b, softmax-xent, dlogits, and
+// dh = (dlogits @ w^T) * (h>0), all in ONE single-workgroup kernel
+std::vector<torch::Tensor> mlp_head_fused(torch::Tensor h, torch::Tensor w,
+                                          torch::Tensor b, torch::Tensor labels,
+                                          double scale) {
+  TORCH_CHECK(h.is_cuda() && h.dim() == 2 && h.is_contiguous() &&
+              h.scalar_type() == torch::kBFloat16, "h must be bf16 [B,H]");
+  TORCH_CHECK(w.is_contiguous() && w.dim() == 2 && w.size(0) == h.size(1) &&
+              w.scalar_type() == torch::kBFloat16, "w must be bf16 [H,C]");
+  const int B = h.size(0), H = h.size(1), C = w.size(1);
+  TORCH_CHECK(C <= 16 && H <= 512 && B <= 4096,
+              "mlp_head_fused limits: C<=16, H<=512, B<=4096");
+  TORCH_CHECK(b.scalar_type() == torch::kBFloat16 && b.numel() == C,
+              "bias must be bf16 [C]");
+  TORCH_CHECK(labels.scalar_type() == torch::kInt64 && labels.numel() == B);
+  auto dlogits = torch::empty({B, C}, h.options());
+  auto dh = torch::empty_like(h);
+  auto loss = torch::empty({}, h.options().dtype(torch::kFloat32));
+  launch_mlp_head_fused((const bf16_t*)h.data_ptr(),
+                        (const bf16_t*)w.data_ptr(),
+                        (const bf16_t*)b.data_ptr(), labels.data_ptr<long>(),
+                        (bf16_t*)dlogits.data_ptr(), (bf16_t*)dh.data_ptr(),
+                        loss.data_ptr<float>(), (float)scale, B, H, C,
+                        cur_stream());
+  return {loss, dlogits, dh};
+}
+
 torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
                                double scale) {
   TORCH_CHECK(probs.is_cuda() && probs.scalar_type() == torch::kBFloat16 &&
@@ -515,6 +545,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "bf16 MFMA GEMM + fused epilogue (bias/act/relu_bwd/colsum) into "
         "out (bf16 or fp32), split-K for deep skinny shapes");
   m.def("softmax_xent_fused", &softmax_xent_fused);
+  m.def("mlp_head_fused", &mlp_head_fused);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("conv2d_fwd", &conv2d_fwd, "implicit-GEMM conv fwd (MFMA, bf16)");

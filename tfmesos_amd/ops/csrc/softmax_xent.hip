@@ -133,3 +133,106 @@ void launch_softmax_xent_bwd(const bf16_t* probs, const long* labels,
   hipLaunchKernelGGL(softmax_xent_bwd_kernel, dim3(B), dim3(WAVE), 0, stream,
                      probs, labels, dlogits, scale, B, C);
 }
+
+namespace {
+
+// Fully fused MLP classifier head for the mnist_replica hot step:
+//   logits = h @ w + b ; softmax ; mean xent loss ;
+//   dlogits = (p - onehot)/B ; dh = (dlogits @ w^T) * (h > 0)
+// ONE single-workgroup kernel (8 waves; each wave owns rows round-robin,
+// w staged in LDS once) replacing three ~5-6 us launches — the per-
+// kernel execution floor dominates at these sizes (see profiles/).
+// Constraints: C <= 16, H <= 512, B <= 4096.
+__global__ __launch_bounds__(512)
+void mlp_head_fused_kernel(const bf16_t* __restrict__ h,
+                           const bf16_t* __restrict__ w,
+                           const bf16_t* __restrict__ bias,
+                           const long* __restrict__ labels,
+                           bf16_t* __restrict__ dlogits,
+                           bf16_t* __restrict__ dh,
+                           float* __restrict__ loss_out,
+                           float scale, int B, int H, int C) {
+  __shared__ float ws[512][17];       // [e][j], padded stride 17
+  __shared__ float bs[16];
+  __shared__ float lsum[8];
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;            // 0..7
+  // stage w (H*C) and bias into LDS as fp32
+  for (int i = t; i < H * C; i += 512)
+    ws[i / C][i % C] = bf2f(w[i]);
+  for (int i = t; i < C; i += 512) bs[i] = bf2f(bias[i]);
+  __syncthreads();
+
+  const int nch = (H + 63) / 64;      // h-chunks per lane (<=8)
+  float neglogp_acc = 0.f;
+  for (int row = wave; row < B; row += 8) {
+    const bf16_t* hrow = h + (long)row * H;
+    float hv[8];
+#pragma unroll
+    for (int cch = 0; cch < 8; ++cch) {
+      const int e = cch * 64 + lane;
+      hv[cch] = (cch < nch && e < H) ? bf2f(hrow[e]) : 0.f;
+    }
+    // logits: per-class wave reductions (butterfly -> all lanes hold all)
+    float logit[16];
+    for (int j = 0; j < C; ++j) {
+      float p = 0.f;
+#pragma unroll
+      for (int cch = 0; cch < 8; ++cch) {
+        const int e = cch * 64 + lane;
+        if (cch < nch && e < H) p += hv[cch] * ws[e][j];
+      }
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) p += __shfl_xor(p, off, 64);
+      logit[j] = p + bs[j];
+    }
+    // softmax + xent + dlogits (every lane redundantly; registers only)
+    float mx = -1e30f;
+    for (int j = 0; j < C; ++j) mx = fmaxf(mx, logit[j]);
+    float sum = 0.f;
+    for (int j = 0; j < C; ++j) {
+      logit[j] = __expf(logit[j] - mx);
+      sum += logit[j];
+    }
+    const float inv = 1.f / sum;
+    const int label = (int)labels[row];
+    float dl[16];
+    for (int j = 0; j < C; ++j) {
+      const float p = logit[j] * inv;
+      dl[j] = (p - (j == label ? 1.f : 0.f)) * scale;
+      if (j == label && lane == 0) neglogp_acc += -__logf(fmaxf(p, 1e-30f));
+    }
+    if (lane < C) dlogits[(long)row * C + lane] = f2bf(dl[lane]);
+    // dh = (dl @ w^T) masked by h > 0
+    bf16_t* dhrow = dh + (long)row * H;
+#pragma unroll
+    for (int cch = 0; cch < 8; ++cch) {
+      const int e = cch * 64 + lane;
+      if (cch < nch && e < H) {
+        float v = 0.f;
+        for (int j = 0; j < C; ++j) v += dl[j] * ws[e][j];
+        dhrow[e] = f2bf(hv[cch] > 0.f ? v : 0.f);
+      }
+    }
+  }
+  if (lane == 0) lsum[wave] = neglogp_acc;
+  __syncthreads();
+  if (t == 0) {
+    float s = 0.f;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) s += lsum[i];
+    *loss_out = s / B;
+  }
+}
+
+}  // namespace
+
+void launch_mlp_head_fused(const bf16_t* h, const bf16_t* w,
+                           const bf16_t* bias, const long* labels,
+                           bf16_t* dlogits, bf16_t* dh, float* loss,
+                           float scale, int B, int H, int C,
+                           hipStream_t stream) {
+  hipLaunchKernelGGL(mlp_head_fused_kernel, dim3(1), dim3(512), 0, stream,
+                     h, w, bias, labels, dlogits, dh, loss, scale, B, H, C);
+}
