@@ -174,9 +174,14 @@ void mlp_head_fused_kernel(const bf16_t* __restrict__ h,
       const int e = cch * 64 + lane;
       hv[cch] = (cch < nch && e < H) ? bf2f(hrow[e]) : 0.f;
     }
-    // logits: per-class wave reductions (butterfly -> all lanes hold all)
+    // logits: per-class wave reductions (butterfly -> all lanes hold
+    // all). Every loop over classes is fully unrolled to the 16-class
+    // cap with an early break: runtime-bounded loops over register
+    // arrays would demote logit[]/dl[] to scratch memory.
     float logit[16];
-    for (int j = 0; j < C; ++j) {
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      if (j >= C) break;
       float p = 0.f;
 #pragma unroll
       for (int cch = 0; cch < 8; ++cch) {
@@ -189,21 +194,29 @@ void mlp_head_fused_kernel(const bf16_t* __restrict__ h,
     }
     // softmax + xent + dlogits (every lane redundantly; registers only)
     float mx = -1e30f;
-    for (int j = 0; j < C; ++j) mx = fmaxf(mx, logit[j]);
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      if (j >= C) break;
+      mx = fmaxf(mx, logit[j]);
+    }
     float sum = 0.f;
-    for (int j = 0; j < C; ++j) {
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      if (j >= C) break;
       logit[j] = __expf(logit[j] - mx);
       sum += logit[j];
     }
     const float inv = 1.f / sum;
     const int label = (int)labels[row];
     float dl[16];
-    for (int j = 0; j < C; ++j) {
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      if (j >= C) break;
       const float p = logit[j] * inv;
       dl[j] = (p - (j == label ? 1.f : 0.f)) * scale;
       if (j == label && lane == 0) neglogp_acc += -__logf(fmaxf(p, 1e-30f));
+      if (lane == j) dlogits[(long)row * C + j] = f2bf(dl[j]);
     }
-    if (lane < C) dlogits[(long)row * C + lane] = f2bf(dl[lane]);
     // dh = (dl @ w^T) masked by h > 0
     bf16_t* dhrow = dh + (long)row * H;
 #pragma unroll
@@ -211,7 +224,11 @@ void mlp_head_fused_kernel(const bf16_t* __restrict__ h,
       const int e = cch * 64 + lane;
       if (cch < nch && e < H) {
         float v = 0.f;
-        for (int j = 0; j < C; ++j) v += dl[j] * ws[e][j];
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          if (j >= C) break;
+          v += dl[j] * ws[e][j];
+        }
         dhrow[e] = f2bf(hv[cch] > 0.f ? v : 0.f);
       }
     }
