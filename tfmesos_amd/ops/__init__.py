@@ -445,7 +445,7 @@ def mlp_head_fused(h, w, b, labels, scale=None):
     """Fused classifier head (one kernel): logits = h@w+b, softmax,
     mean xent loss, dlogits = (p-onehot)*scale, and dh = (dlogits@w^T)
     masked by h>0 (h is a relu output). Returns (loss, dlogits, dh).
-    GPU limits C<=16, H<=512, B<=4096; CPU reference otherwise."""
+    GPU limits C<=16, H<=512, B<=512; CPU reference otherwise."""
     B = h.shape[0]
     s = float(scale if scale is not None else 1.0 / B)
     if h.is_cuda:

@@ -139,10 +139,13 @@ namespace {
 // Fully fused MLP classifier head for the mnist_replica hot step:
 //   logits = h @ w + b ; softmax ; mean xent loss ;
 //   dlogits = (p - onehot)/B ; dh = (dlogits @ w^T) * (h > 0)
-// ONE single-workgroup kernel (8 waves; each wave owns rows round-robin,
-// w staged in LDS once) replacing three ~5-6 us launches — the per-
-// kernel execution floor dominates at these sizes (see profiles/).
-// Constraints: C <= 16, H <= 512, B <= 4096.
+// ONE single-workgroup kernel replacing three ~5-6 us launches (the
+// per-kernel execution floor dominates at these sizes — see profiles/).
+// One THREAD per row, ~2*H*C unrolled MACs each, weights broadcast from
+// LDS; no cross-lane reductions (a wave-shuffle variant serialized on
+// the DS pipe and ran 10x slower). Every class loop is unrolled to the
+// 16-class cap: runtime-bounded loops over register arrays would demote
+// acc[]/dl[] to scratch. Constraints: C <= 16, H <= 512, B <= 512.
 __global__ __launch_bounds__(512)
 void mlp_head_fused_kernel(const bf16_t* __restrict__ h,
                            const bf16_t* __restrict__ w,
@@ -153,94 +156,70 @@ void mlp_head_fused_kernel(const bf16_t* __restrict__ h,
                            float* __restrict__ loss_out,
                            float scale, int B, int H, int C) {
   __shared__ float ws[512][17];       // [e][j], padded stride 17
-  __shared__ float bs[16];
-  __shared__ float lsum[8];
+  __shared__ float lsum[512];
   const int t = threadIdx.x;
-  const int lane = t & 63;
-  const int wave = t >> 6;            // 0..7
-  // stage w (H*C) and bias into LDS as fp32
   for (int i = t; i < H * C; i += 512)
     ws[i / C][i % C] = bf2f(w[i]);
-  for (int i = t; i < C; i += 512) bs[i] = bf2f(bias[i]);
   __syncthreads();
 
-  const int nch = (H + 63) / 64;      // h-chunks per lane (<=8)
-  float neglogp_acc = 0.f;
-  for (int row = wave; row < B; row += 8) {
-    const bf16_t* hrow = h + (long)row * H;
-    float hv[8];
+  float neglogp = 0.f;
+  if (t < B) {
+    const bf16_t* hrow = h + (long)t * H;
+    float acc[16];
 #pragma unroll
-    for (int cch = 0; cch < 8; ++cch) {
-      const int e = cch * 64 + lane;
-      hv[cch] = (cch < nch && e < H) ? bf2f(hrow[e]) : 0.f;
-    }
-    // logits: per-class wave reductions (butterfly -> all lanes hold
-    // all). Every loop over classes is fully unrolled to the 16-class
-    // cap with an early break: runtime-bounded loops over register
-    // arrays would demote logit[]/dl[] to scratch memory.
-    float logit[16];
+    for (int j = 0; j < 16; ++j) acc[j] = (j < C) ? bf2f(bias[j]) : 0.f;
+    for (int e = 0; e < H; ++e) {
+      const float hv = bf2f(hrow[e]);
 #pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      if (j >= C) break;
-      float p = 0.f;
-#pragma unroll
-      for (int cch = 0; cch < 8; ++cch) {
-        const int e = cch * 64 + lane;
-        if (cch < nch && e < H) p += hv[cch] * ws[e][j];
+      for (int j = 0; j < 16; ++j) {
+        if (j >= C) break;
+        acc[j] += hv * ws[e][j];
       }
-#pragma unroll
-      for (int off = 32; off > 0; off >>= 1) p += __shfl_xor(p, off, 64);
-      logit[j] = p + bs[j];
     }
-    // softmax + xent + dlogits (every lane redundantly; registers only)
     float mx = -1e30f;
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
       if (j >= C) break;
-      mx = fmaxf(mx, logit[j]);
+      mx = fmaxf(mx, acc[j]);
     }
     float sum = 0.f;
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
       if (j >= C) break;
-      logit[j] = __expf(logit[j] - mx);
-      sum += logit[j];
+      acc[j] = __expf(acc[j] - mx);
+      sum += acc[j];
     }
     const float inv = 1.f / sum;
-    const int label = (int)labels[row];
+    const int label = (int)labels[t];
     float dl[16];
+    bf16_t* drow = dlogits + (long)t * C;
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
       if (j >= C) break;
-      const float p = logit[j] * inv;
+      const float p = acc[j] * inv;
       dl[j] = (p - (j == label ? 1.f : 0.f)) * scale;
-      if (j == label && lane == 0) neglogp_acc += -__logf(fmaxf(p, 1e-30f));
-      if (lane == j) dlogits[(long)row * C + j] = f2bf(dl[j]);
+      drow[j] = f2bf(dl[j]);
+      if (j == label) neglogp = -__logf(fmaxf(p, 1e-30f));
     }
-    // dh = (dl @ w^T) masked by h > 0
-    bf16_t* dhrow = dh + (long)row * H;
+    bf16_t* dhrow = dh + (long)t * H;
+    for (int e = 0; e < H; ++e) {
+      float v = 0.f;
 #pragma unroll
-    for (int cch = 0; cch < 8; ++cch) {
-      const int e = cch * 64 + lane;
-      if (cch < nch && e < H) {
-        float v = 0.f;
-#pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          if (j >= C) break;
-          v += dl[j] * ws[e][j];
-        }
-        dhrow[e] = f2bf(hv[cch] > 0.f ? v : 0.f);
+      for (int j = 0; j < 16; ++j) {
+        if (j >= C) break;
+        v += dl[j] * ws[e][j];
       }
+      dhrow[e] = f2bf(bf2f(hrow[e]) > 0.f ? v : 0.f);
     }
   }
-  if (lane == 0) lsum[wave] = neglogp_acc;
+  lsum[t] = neglogp;
   __syncthreads();
-  if (t == 0) {
-    float s = 0.f;
 #pragma unroll
-    for (int i = 0; i < 8; ++i) s += lsum[i];
-    *loss_out = s / B;
+  for (int s = 256; s > 0; s >>= 1) {
+    if (t < s) lsum[t] += lsum[t + s];
+    __syncthreads();
   }
+  if (t == 0) *loss_out = lsum[0] / B;
 }
 
 }  // namespace

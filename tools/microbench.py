@@ -67,6 +67,8 @@ def main():
         ("dW1 gemm tn + colsum [100x784]^T@[100x100]",
          lambda: ops.gemm_bias_act(x, h, trans_a=True, out=gw1,
                                    colsum_out=gb1)),
+        ("mlp_head_fused (fwd2+softmax+dh in one kernel)",
+         lambda: ops.mlp_head_fused(h, sm_w, sm_b, y)),
         ("sgd fused apply (79510 params)",
          lambda: ops.fused_sgd(flat, flat_g, lr=0.01, bf16_out=flat_bf)),
         ("torch.matmul ref fwd1 (hipBLASLt)",
