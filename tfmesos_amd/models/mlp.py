@@ -68,7 +68,7 @@ class MnistMLP(object):
         # fwd hidden layer: split-K GEMM + bias + relu
         h = ops.gemm_bias_act(x, hid_w, hid_b, act="relu")       # [B,H]
         if x.is_cuda and self.classes <= 16 and self.hidden <= 512 \
-                and B <= 512:
+                and B <= 512 and B * (self.hidden + 8) <= 15000:
             # whole classifier head (fwd GEMM + softmax-xent fwd/bwd +
             # dh GEMM with relu mask) in ONE kernel
             loss, dlogits, dh = ops.mlp_head_fused(h, sm_w, sm_b, y)

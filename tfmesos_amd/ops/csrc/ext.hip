@@ -303,8 +303,10 @@ std::vector<torch::Tensor> mlp_head_fused(torch::Tensor h, torch::Tensor w,
   TORCH_CHECK(w.is_contiguous() && w.dim() == 2 && w.size(0) == h.size(1) &&
               w.scalar_type() == torch::kBFloat16, "w must be bf16 [H,C]");
   const int B = h.size(0), H = h.size(1), C = w.size(1);
-  TORCH_CHECK(C <= 16 && H <= 512 && B <= 512,
-              "mlp_head_fused limits: C<=16, H<=512, B<=512");
+  TORCH_CHECK(C <= 16 && H <= 512 && B <= 512 &&
+              (long)B * (H + 8) <= 15000,
+              "mlp_head_fused limits: C<=16, H<=512, B<=512, "
+              "B*(H+8)<=15000");
   TORCH_CHECK(b.scalar_type() == torch::kBFloat16 && b.numel() == C,
               "bias must be bf16 [C]");
   TORCH_CHECK(labels.scalar_type() == torch::kInt64 && labels.numel() == B);

@@ -141,11 +141,13 @@ namespace {
 //   dlogits = (p - onehot)/B ; dh = (dlogits @ w^T) * (h > 0)
 // ONE single-workgroup kernel replacing three ~5-6 us launches (the
 // per-kernel execution floor dominates at these sizes — see profiles/).
-// One THREAD per row, ~2*H*C unrolled MACs each, weights broadcast from
-// LDS; no cross-lane reductions (a wave-shuffle variant serialized on
-// the DS pipe and ran 10x slower). Every class loop is unrolled to the
-// 16-class cap: runtime-bounded loops over register arrays would demote
-// acc[]/dl[] to scratch. Constraints: C <= 16, H <= 512, B <= 512.
+// One THREAD per row; h is staged through LDS with COALESCED
+// cooperative copies (per-thread row walks over global memory are
+// 64-way divergent and ran 10x slower), weights are read as f32x4
+// broadcasts, dh is written back through the same LDS staging. Every
+// class loop is unrolled to the 16-class cap: runtime-bounded loops
+// over register arrays would demote acc[]/dl[] to scratch.
+// Constraints: C <= 16, H <= 512, B <= 512, B*(H+8) <= 15000.
 __global__ __launch_bounds__(512)
 void mlp_head_fused_kernel(const bf16_t* __restrict__ h,
                            const bf16_t* __restrict__ w,
@@ -155,25 +157,33 @@ void mlp_head_fused_kernel(const bf16_t* __restrict__ h,
                            bf16_t* __restrict__ dh,
                            float* __restrict__ loss_out,
                            float scale, int B, int H, int C) {
-  __shared__ float ws[512][17];       // [e][j], padded stride 17
+  __shared__ __align__(16) float ws[512][16];    // 32 KB, f32x4 reads
+  __shared__ __align__(16) __bf16 hs[15000];     // [B][H+8] staged rows
   __shared__ float lsum[512];
   const int t = threadIdx.x;
+  const int hp = H + 8;
   for (int i = t; i < H * C; i += 512)
     ws[i / C][i % C] = bf2f(w[i]);
+  for (int i = t; i < B * H; i += 512)
+    hs[(i / H) * hp + (i % H)] = *(const __bf16*)&h[i];
   __syncthreads();
 
   float neglogp = 0.f;
   if (t < B) {
-    const bf16_t* hrow = h + (long)t * H;
+    __bf16* hrow = &hs[t * hp];
     float acc[16];
 #pragma unroll
     for (int j = 0; j < 16; ++j) acc[j] = (j < C) ? bf2f(bias[j]) : 0.f;
     for (int e = 0; e < H; ++e) {
-      const float hv = bf2f(hrow[e]);
+      const float hv = (float)hrow[e];
+      const f32x4* wrow = (const f32x4*)ws[e];
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        if (j >= C) break;
-        acc[j] += hv * ws[e][j];
+      for (int q = 0; q < 4; ++q) {
+        if (q * 4 >= C) break;
+        const f32x4 wv = wrow[q];
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          if (q * 4 + j < C) acc[q * 4 + j] += hv * wv[j];
       }
     }
     float mx = -1e30f;
@@ -201,19 +211,27 @@ void mlp_head_fused_kernel(const bf16_t* __restrict__ h,
       drow[j] = f2bf(dl[j]);
       if (j == label) neglogp = -__logf(fmaxf(p, 1e-30f));
     }
-    bf16_t* dhrow = dh + (long)t * H;
+    // dh into the LDS staging (overwrite h rows in place)
     for (int e = 0; e < H; ++e) {
+      const float hv = (float)hrow[e];
+      const f32x4* wrow = (const f32x4*)ws[e];
       float v = 0.f;
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        if (j >= C) break;
-        v += dl[j] * ws[e][j];
+      for (int q = 0; q < 4; ++q) {
+        if (q * 4 >= C) break;
+        const f32x4 wv = wrow[q];
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          if (q * 4 + j < C) v += dl[q * 4 + j] * wv[j];
       }
-      dhrow[e] = f2bf(bf2f(hrow[e]) > 0.f ? v : 0.f);
+      hrow[e] = (__bf16)(hv > 0.f ? v : 0.f);
     }
   }
   lsum[t] = neglogp;
   __syncthreads();
+  // coalesced writeback of dh
+  for (int i = t; i < B * H; i += 512)
+    dh[i] = *(bf16_t*)&hs[(i / H) * hp + (i % H)];
 #pragma unroll
   for (int s = 256; s > 0; s >>= 1) {
     if (t < s) lsum[t] += lsum[t + s];
