@@ -65,18 +65,16 @@ class MnistMLP(object):
         hid_w, hid_b = p("hid_w"), p("hid_b")
         sm_w, sm_b = p("sm_w"), p("sm_b")
 
-        # fwd hidden layer: split-K GEMM + bias + relu
+        # fwd hidden layer: split-K GEMM + bias + relu. (A fully fused
+        # single-workgroup classifier head exists — ops.mlp_head_fused —
+        # but exposes only B threads of parallelism and measured slower
+        # than these three MFMA kernels; the composed path is the
+        # default.)
         h = ops.gemm_bias_act(x, hid_w, hid_b, act="relu")       # [B,H]
-        if x.is_cuda and self.classes <= 16 and self.hidden <= 512 \
-                and B <= 512 and B * (self.hidden + 8) <= 15000:
-            # whole classifier head (fwd GEMM + softmax-xent fwd/bwd +
-            # dh GEMM with relu mask) in ONE kernel
-            loss, dlogits, dh = ops.mlp_head_fused(h, sm_w, sm_b, y)
-        else:
-            logits = ops.gemm_bias_act(h, sm_w, sm_b, act="none")
-            loss, dlogits = ops.softmax_xent_fused(logits, y)
-            dh = ops.gemm_bias_act(dlogits, sm_w, trans_b=True,
-                                   act="relu_bwd", aux=h)
+        logits = ops.gemm_bias_act(h, sm_w, sm_b, act="none")
+        loss, dlogits = ops.softmax_xent_fused(logits, y)
+        dh = ops.gemm_bias_act(dlogits, sm_w, trans_b=True,
+                               act="relu_bwd", aux=h)
         # dW GEMMs with the bias-grad colsums fused in; fp32 grads
         # written straight into the flat grad views
         ops.gemm_bias_act(h, dlogits, trans_a=True, out=g("sm_w"),
