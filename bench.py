@@ -39,9 +39,13 @@ def run_mnist(args, device, rank, world):
     from tfmesos_amd.ps.replica import SyncReplicaTrainer
 
     model = MnistMLP(hidden_units=args.hidden)
+    # PS shards colocate with workers: all N GPUs compute replicas and
+    # the sharded reduce lands on ranks 0..n_ps-1 ("1-ps/N-worker" on N
+    # devices, the BASELINE.json config)
     trainer = SyncReplicaTrainer(
         model.init_params(), optimizer=args.optimizer,
-        hparams={"lr": args.lr}, device=device, n_ps=args.num_ps)
+        hparams={"lr": args.lr}, device=device, n_ps=args.num_ps,
+        colocate_ps=True)
     roles = trainer.roles
 
     act_dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
@@ -116,7 +120,7 @@ def run_inception(args, device, rank, world):
     model = InceptionV3(num_classes=args.classes)
     trainer = ModuleReplicaTrainer(model, optimizer=args.optimizer,
                                    hparams={"lr": args.lr}, device=device,
-                                   n_ps=args.num_ps)
+                                   n_ps=args.num_ps, colocate_ps=True)
     x, y = synthetic_images(args.inc_batch, size=args.inc_size,
                             classes=args.classes, device=device,
                             dtype=torch.bfloat16, seed=1000 + rank)

@@ -20,10 +20,11 @@ from tfmesos_amd.ps.replica import (  # noqa: E402
 def main():
     mode, steps, prefix = sys.argv[1], int(sys.argv[2]), sys.argv[3]
     n_ps = int(sys.argv[4]) if len(sys.argv) > 4 else 1
+    colocate = len(sys.argv) > 5 and sys.argv[5] == "colocate"
     model = MnistMLP()
     trainer = SyncReplicaTrainer(model.init_params(), optimizer="sgd",
                                  hparams={"lr": 0.1}, device="cpu",
-                                 n_ps=n_ps)
+                                 n_ps=n_ps, colocate_ps=colocate)
     roles = trainer.roles
     # every worker gets the same batch as the single-process reference
     x, y = synthetic_batch(50, seed=42)

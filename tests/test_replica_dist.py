@@ -16,7 +16,7 @@ HERE = os.path.dirname(os.path.abspath(__file__))
 REPO = os.path.dirname(HERE)
 
 
-def _spawn_world(world, mode, steps, prefix, n_ps=1):
+def _spawn_world(world, mode, steps, prefix, n_ps=1, colocate=False):
     from tfmesos_amd.utils import free_port
     port = free_port()
     procs = []
@@ -27,9 +27,11 @@ def _spawn_world(world, mode, steps, prefix, n_ps=1):
             "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
             "PYTHONPATH": REPO,
         })
-        procs.append(subprocess.Popen(
-            [sys.executable, os.path.join(HERE, "_replica_proc.py"),
-             mode, str(steps), prefix, str(n_ps)], env=env))
+        argv = [sys.executable, os.path.join(HERE, "_replica_proc.py"),
+                mode, str(steps), prefix, str(n_ps)]
+        if colocate:
+            argv.append("colocate")
+        procs.append(subprocess.Popen(argv, env=env))
     for p in procs:
         assert p.wait(timeout=180) == 0
 
@@ -111,3 +113,16 @@ def test_async_two_ps_shards(tmp_path):
     want = _single_process_reference(5)
     for n in want:
         assert torch.allclose(got[n], want[n], atol=1e-6), n
+
+
+@pytest.mark.timeout(240)
+def test_sync_colocated_ps_matches_single_process(tmp_path):
+    """colocate_ps: every rank is a worker AND rank 0 applies — the
+    bench's 1-ps/N-worker-on-N-GPUs mapping. Identical batches => mean
+    grad == single grad => masters match the reference exactly."""
+    prefix = str(tmp_path / "wc")
+    _spawn_world(2, "sync", 5, prefix, colocate=True)
+    got = torch.load(prefix + ".pt", weights_only=True)
+    want = _single_process_reference(5)
+    for n in want:
+        assert torch.allclose(got[n], want[n], atol=1e-5), n

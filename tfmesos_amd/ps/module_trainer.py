@@ -21,13 +21,14 @@ from tfmesos_amd.ps.replica import SyncReplicaTrainer
 class ModuleReplicaTrainer(object):
 
     def __init__(self, module, optimizer="sgd", hparams=None, device="cpu",
-                 n_ps=None):
+                 n_ps=None, colocate_ps=False):
         self.device = torch.device(device)
         params = [(name, p.detach().float().cpu())
                   for name, p in module.named_parameters()]
         self.t = SyncReplicaTrainer(params, optimizer=optimizer,
                                     hparams=hparams, device=device,
-                                    grad_dtype=torch.bfloat16, n_ps=n_ps)
+                                    grad_dtype=torch.bfloat16, n_ps=n_ps,
+                                    colocate_ps=colocate_ps)
         self.roles = self.t.roles
         self.module = module
         module.to(device=self.device, dtype=torch.bfloat16)

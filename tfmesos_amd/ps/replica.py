@@ -65,11 +65,17 @@ def init_distributed(device=None):
 class Roles(object):
     """Rank -> role mapping. Ranks [0, n_ps) are PS shards (reference:
     ps tasks come first in the jobs_def and hold param slices), the rest
-    are workers. world==1 colocates one ps and one worker."""
+    are workers. world==1 colocates one ps and one worker.
 
-    def __init__(self, rank, world, n_ps=1):
+    ``colocate_ps=True`` makes the PS-shard ranks ALSO workers (the
+    launcher packs a ps task and a worker task onto the same GPU when
+    resources allow): every rank computes a replica, so N GPUs give N
+    workers — "1-ps/N-worker" on N devices."""
+
+    def __init__(self, rank, world, n_ps=1, colocate_ps=False):
         self.rank = rank
         self.world = world
+        self.colocated = colocate_ps or world == 1
         if world == 1:
             self.n_ps = 1
             self.is_ps = True
@@ -78,6 +84,15 @@ class Roles(object):
             self.ps_rank = 0
             self.ps_ranks = [0]
             self.worker_index = 0
+        elif colocate_ps:
+            n_ps = max(1, min(int(n_ps), world))
+            self.n_ps = n_ps
+            self.ps_ranks = list(range(n_ps))
+            self.ps_rank = 0
+            self.is_ps = rank < n_ps
+            self.is_worker = True
+            self.n_workers = world
+            self.worker_index = rank
         else:
             n_ps = max(1, min(int(n_ps), world - 1))
             self.n_ps = n_ps
@@ -92,11 +107,15 @@ class Roles(object):
     def worker_ranks(self):
         if self.world == 1:
             return [0]
+        if self.colocated:
+            return list(range(self.world))
         return list(range(self.n_ps, self.world))
 
     def describe(self):
         if self.world == 1:
             return "ps1+w1(colocated)"
+        if self.colocated:
+            return "ps%d+w%d(colocated)" % (self.n_ps, self.n_workers)
         return "ps%d+w%d" % (self.n_ps, self.n_workers)
 
 
@@ -114,14 +133,15 @@ class SyncReplicaTrainer(object):
     """Dense sync PS trainer over one flat buffer, sharded across n_ps."""
 
     def __init__(self, params, optimizer="sgd", hparams=None, device="cpu",
-                 grad_dtype=torch.float32, n_ps=None):
+                 grad_dtype=torch.float32, n_ps=None, colocate_ps=False):
         """params: dict/list of (name, fp32 tensor) — identical on all
         ranks (same seed)."""
         self.device = torch.device(device)
         self.rank, self.world = init_distributed(device)
         if n_ps is None:
             n_ps = int(_env("TFA_NUM_PS", "1"))
-        self.roles = Roles(self.rank, self.world, n_ps=n_ps)
+        self.roles = Roles(self.rank, self.world, n_ps=n_ps,
+                           colocate_ps=colocate_ps)
         hparams = dict(hparams or {})
 
         # every rank computes the same flat layout; only PS ranks apply
@@ -173,9 +193,10 @@ class SyncReplicaTrainer(object):
         scale = grad_scale if grad_scale is not None \
             else 1.0 / self.roles.n_workers
         if self.world > 1:
-            if self.roles.is_ps:
-                # dist.reduce is in-place: the PS buffer holds last step's
-                # sum and would be re-added — contribute zeros instead
+            if self.roles.is_ps and not self.roles.is_worker:
+                # dist.reduce is in-place: a PURE ps buffer holds last
+                # step's sum and would be re-added — contribute zeros
+                # (colocated ps ranks contribute their own worker grads)
                 self.flat_grad.zero_()
             for i, (lo, hi) in enumerate(self.shards):
                 dist.reduce(self.flat_grad[lo:hi],
