@@ -20,11 +20,13 @@ from tfmesos_amd.ps.replica import (  # noqa: E402
 def main():
     mode, steps, prefix = sys.argv[1], int(sys.argv[2]), sys.argv[3]
     n_ps = int(sys.argv[4]) if len(sys.argv) > 4 else 1
-    colocate = len(sys.argv) > 5 and sys.argv[5] == "colocate"
+    variant = sys.argv[5] if len(sys.argv) > 5 else ""
     model = MnistMLP()
     trainer = SyncReplicaTrainer(model.init_params(), optimizer="sgd",
                                  hparams={"lr": 0.1}, device="cpu",
-                                 n_ps=n_ps, colocate_ps=colocate)
+                                 n_ps=n_ps, colocate_ps=variant == "colocate",
+                                 mode="allreduce" if variant == "allreduce"
+                                 else "ps")
     roles = trainer.roles
     # every worker gets the same batch as the single-process reference
     x, y = synthetic_batch(50, seed=42)

@@ -30,7 +30,7 @@ def _spawn_world(world, mode, steps, prefix, n_ps=1, colocate=False):
         argv = [sys.executable, os.path.join(HERE, "_replica_proc.py"),
                 mode, str(steps), prefix, str(n_ps)]
         if colocate:
-            argv.append("colocate")
+            argv.append(colocate if isinstance(colocate, str) else "colocate")
         procs.append(subprocess.Popen(argv, env=env))
     for p in procs:
         assert p.wait(timeout=180) == 0
@@ -122,6 +122,18 @@ def test_sync_colocated_ps_matches_single_process(tmp_path):
     grad == single grad => masters match the reference exactly."""
     prefix = str(tmp_path / "wc")
     _spawn_world(2, "sync", 5, prefix, colocate=True)
+    got = torch.load(prefix + ".pt", weights_only=True)
+    want = _single_process_reference(5)
+    for n in want:
+        assert torch.allclose(got[n], want[n], atol=1e-5), n
+
+
+@pytest.mark.timeout(240)
+def test_sync_allreduce_mode_matches_single_process(tmp_path):
+    """allreduce mode (one collective + replicated deterministic apply)
+    must be bit-identical to the PS path and the single-process ref."""
+    prefix = str(tmp_path / "war")
+    _spawn_world(2, "sync", 5, prefix, colocate="allreduce")
     got = torch.load(prefix + ".pt", weights_only=True)
     want = _single_process_reference(5)
     for n in want:

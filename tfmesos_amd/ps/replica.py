@@ -133,22 +133,35 @@ class SyncReplicaTrainer(object):
     """Dense sync PS trainer over one flat buffer, sharded across n_ps."""
 
     def __init__(self, params, optimizer="sgd", hparams=None, device="cpu",
-                 grad_dtype=torch.float32, n_ps=None, colocate_ps=False):
+                 grad_dtype=torch.float32, n_ps=None, colocate_ps=False,
+                 mode="ps"):
         """params: dict/list of (name, fp32 tensor) — identical on all
-        ranks (same seed)."""
+        ranks (same seed).
+
+        mode="ps" (default): sharded reduce -> PS apply -> broadcast.
+        mode="allreduce": when every rank hosts a replica, the PS
+        exchange degenerates to ONE all_reduce + a replicated
+        deterministic apply on every rank (identical masters everywhere,
+        no broadcast) — half the collectives, the xGMI-efficient option
+        at scale. Only valid with colocated roles.
+        """
         self.device = torch.device(device)
         self.rank, self.world = init_distributed(device)
         if n_ps is None:
             n_ps = int(_env("TFA_NUM_PS", "1"))
+        self.mode = mode
+        if mode == "allreduce":
+            colocate_ps = True
         self.roles = Roles(self.rank, self.world, n_ps=n_ps,
                            colocate_ps=colocate_ps)
         hparams = dict(hparams or {})
 
         # every rank computes the same flat layout; only PS ranks apply
-        # (and then only their own shard)
+        # (and then only their own shard) — except allreduce mode, where
+        # every rank applies the whole buffer
         self.store = PStore(device=device)
         self.store.init_params(params, optimizer=optimizer, **hparams)
-        if not self.roles.is_ps:
+        if not self.roles.is_ps and mode != "allreduce":
             self.store.state = {}  # workers don't need optimizer state
 
         self.shards = _shard_ranges(self.store.flat.numel(), self.roles.n_ps)
@@ -192,6 +205,11 @@ class SyncReplicaTrainer(object):
         """
         scale = grad_scale if grad_scale is not None \
             else 1.0 / self.roles.n_workers
+        if self.mode == "allreduce":
+            if self.world > 1:
+                dist.all_reduce(self.flat_grad, op=dist.ReduceOp.SUM)
+            self.store.apply_flat(self.flat_grad, grad_scale=scale)
+            return self.store.global_step
         if self.world > 1:
             if self.roles.is_ps and not self.roles.is_worker:
                 # dist.reduce is in-place: a PURE ps buffer holds last
