@@ -56,10 +56,39 @@ def init_distributed(device=None):
     if dist.is_initialized():
         return dist.get_rank(), dist.get_world_size()
     rank = int(_env("RANK", _env("TFA_RANK", "0")))
-    backend = "nccl" if (device and str(device).startswith("cuda")) else "gloo"
+    # launcher-provided backend wins (a mixed CPU-ps/GPU-worker cluster
+    # must agree on gloo); otherwise pick by device
+    backend = _env("TFA_DIST_BACKEND") or (
+        "nccl" if (device and str(device).startswith("cuda")) else "gloo")
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     dist.init_process_group(backend=backend, rank=rank, world_size=world)
     return rank, world
+
+
+def _staged(t):
+    """gloo collectives need CPU tensors; nccl wants device-resident."""
+    return t.cpu() if (t.is_cuda and dist.get_backend() == "gloo") else t
+
+
+def reduce_(t, dst):
+    s = _staged(t)
+    dist.reduce(s, dst=dst, op=dist.ReduceOp.SUM)
+    if s is not t:
+        t.copy_(s)
+
+
+def broadcast_(t, src):
+    s = _staged(t)
+    dist.broadcast(s, src=src)
+    if s is not t:
+        t.copy_(s)
+
+
+def all_reduce_(t):
+    s = _staged(t)
+    dist.all_reduce(s, op=dist.ReduceOp.SUM)
+    if s is not t:
+        t.copy_(s)
 
 
 class Roles(object):
@@ -182,8 +211,8 @@ class SyncReplicaTrainer(object):
         if self.world == 1:
             return
         for i, (lo, hi) in enumerate(self.shards):
-            dist.broadcast(self.flat_params_bf16[lo:hi],
-                           src=self.roles.ps_ranks[i])
+            broadcast_(self.flat_params_bf16[lo:hi],
+                       src=self.roles.ps_ranks[i])
 
     # parameter views (bf16 working copies, refreshed in-place by pull)
     def param(self, name):
@@ -207,7 +236,7 @@ class SyncReplicaTrainer(object):
             else 1.0 / self.roles.n_workers
         if self.mode == "allreduce":
             if self.world > 1:
-                dist.all_reduce(self.flat_grad, op=dist.ReduceOp.SUM)
+                all_reduce_(self.flat_grad)
             self.store.apply_flat(self.flat_grad, grad_scale=scale)
             return self.store.global_step
         if self.world > 1:
@@ -217,16 +246,15 @@ class SyncReplicaTrainer(object):
                 # (colocated ps ranks contribute their own worker grads)
                 self.flat_grad.zero_()
             for i, (lo, hi) in enumerate(self.shards):
-                dist.reduce(self.flat_grad[lo:hi],
-                            dst=self.roles.ps_ranks[i], op=dist.ReduceOp.SUM)
+                reduce_(self.flat_grad[lo:hi], dst=self.roles.ps_ranks[i])
         if self.my_shard is not None:
             lo, hi = self.my_shard
             self.store.apply_flat(self.flat_grad, grad_scale=scale,
                                   lo=lo, hi=hi)
         if self.world > 1:
             for i, (lo, hi) in enumerate(self.shards):
-                dist.broadcast(self.flat_params_bf16[lo:hi],
-                               src=self.roles.ps_ranks[i])
+                broadcast_(self.flat_params_bf16[lo:hi],
+                           src=self.roles.ps_ranks[i])
         return self.store.global_step
 
     def sync_masters(self):
@@ -234,8 +262,8 @@ class SyncReplicaTrainer(object):
         complete master buffer (e.g. for a full checkpoint or eval)."""
         if self.world > 1:
             for i, (lo, hi) in enumerate(self.shards):
-                dist.broadcast(self.store.flat[lo:hi],
-                               src=self.roles.ps_ranks[i])
+                broadcast_(self.store.flat[lo:hi],
+                           src=self.roles.ps_ranks[i])
         return self.store.flat
 
     def save(self, path):
@@ -287,13 +315,21 @@ class AsyncPSWorker(object):
 
     def step(self):
         t = self.t
+        gloo = dist.get_backend() == "gloo"
         for i, (lo, hi) in enumerate(t.shards):
             p = t.roles.ps_ranks[i]
-            dist.send(t.flat_grad[lo:hi], dst=p, group=self.groups[(p, t.rank)])
-        for i, (lo, hi) in enumerate(t.shards):
-            p = t.roles.ps_ranks[i]
-            dist.recv(t.flat_params_bf16[lo:hi], src=p,
+            g = t.flat_grad[lo:hi]
+            dist.send(g.cpu() if (gloo and g.is_cuda) else g, dst=p,
                       group=self.groups[(p, t.rank)])
+        for i, (lo, hi) in enumerate(t.shards):
+            p = t.roles.ps_ranks[i]
+            dst = t.flat_params_bf16[lo:hi]
+            if gloo and dst.is_cuda:
+                buf = torch.empty_like(dst, device="cpu")
+                dist.recv(buf, src=p, group=self.groups[(p, t.rank)])
+                dst.copy_(buf)
+            else:
+                dist.recv(dst, src=p, group=self.groups[(p, t.rank)])
         return True
 
     def stop(self):

@@ -72,6 +72,12 @@ class LocalScheduler(object):
         self.task_failure_count = {}
         self.job_finished = {}
 
+        # data-plane backend for launched tasks: RCCL only when every
+        # task owns a GPU; mixed CPU-ps/GPU-worker clusters use gloo
+        # with CPU-staged collectives (ps/replica.py)
+        all_gpu = all(j.gpus >= 1 for j in self.jobs if j.num > j.start)
+        self.dist_backend = "nccl" if all_gpu else "gloo"
+
         self.tasks = []
         for job in self.jobs:
             for task_index in range(job.start, job.num):
@@ -124,6 +130,7 @@ class LocalScheduler(object):
         env = dict(task.env)
         env["TFA_SECRET"] = self.secret.hex()
         env["TFA_TASK_ID"] = task.task_id
+        env["TFA_DIST_BACKEND"] = self.dist_backend
         spec = LaunchSpec(
             task.task_id,
             argv=[sys.executable, "-m", "tfmesos_amd.agent",
