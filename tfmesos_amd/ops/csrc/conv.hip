@@ -124,8 +124,10 @@ __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt,
                      const float* __restrict__ bias, __bf16* __restrict__ Y,
                      ConvShape cs) {
-  __shared__ __align__(16) __bf16 As[BM][BK + APAD];
-  __shared__ __align__(16) __bf16 Bs[BN][BK + APAD];
+  // double-buffered LDS: stage tile i+1 while MFMA consumes tile i —
+  // one barrier per K-iteration, global-load latency overlapped
+  __shared__ __align__(16) __bf16 As[2][BM][BK + APAD];
+  __shared__ __align__(16) __bf16 Bs[2][BN][BK + APAD];
   const long M = (long)cs.N * cs.Ho * cs.Wo;
   const int KD = cs.R * cs.S * cs.C;
   const bool cvec = (cs.C & 7) == 0;
@@ -137,17 +139,24 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
   const int wr = wave >> 1, wc = wave & 1;
 
   f32x4 acc[2][2] = {};
-  for (int k0 = 0; k0 < KD; k0 += BK) {
-    stage_patch_fwd(X, As, cs, tm0, k0, M, KD, t, cvec);
-    stage_wrows(Wt, Bs, tn0, k0, cs.K, KD, t, true);
-    __syncthreads();
+  stage_patch_fwd(X, As[0], cs, tm0, 0, M, KD, t, cvec);
+  stage_wrows(Wt, Bs[0], tn0, 0, cs.K, KD, t, true);
+  __syncthreads();
+  int cur = 0;
+  for (int k0 = 0; k0 < KD; k0 += BK, cur ^= 1) {
+    if (k0 + BK < KD) {
+      stage_patch_fwd(X, As[cur ^ 1], cs, tm0, k0 + BK, M, KD, t, cvec);
+      stage_wrows(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs.K, KD, t, true);
+    }
     const int kfrag = (lane >> 4) * 8;
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
-      bf16x8 a = *(const bf16x8*)&As[wr * 32 + fm * 16 + (lane & 15)][kfrag];
+      bf16x8 a =
+          *(const bf16x8*)&As[cur][wr * 32 + fm * 16 + (lane & 15)][kfrag];
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn) {
-        bf16x8 b = *(const bf16x8*)&Bs[wc * 32 + fn * 16 + (lane & 15)][kfrag];
+        bf16x8 b =
+            *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
         acc[fm][fn] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm][fn], 0, 0, 0);
       }
@@ -254,8 +263,8 @@ __global__ __launch_bounds__(256)
 void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ Wt,
                       __bf16* __restrict__ dX, ConvShape cs) {
   // Wt memory: [C][R*S*K] (host-permuted W^T)
-  __shared__ __align__(16) __bf16 As[BM][BK + APAD];
-  __shared__ __align__(16) __bf16 Bs[BN][BK + APAD];
+  __shared__ __align__(16) __bf16 As[2][BM][BK + APAD];
+  __shared__ __align__(16) __bf16 Bs[2][BN][BK + APAD];
   const long M = (long)cs.N * cs.H * cs.W;
   const int KD = cs.R * cs.S * cs.K;
   const bool kvec = (cs.K & 7) == 0;
@@ -267,17 +276,25 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
   const int wr = wave >> 1, wc = wave & 1;
 
   f32x4 acc[2][2] = {};
-  for (int k0 = 0; k0 < KD; k0 += BK) {
-    stage_patch_bwdd<STRIDE1>(dY, As, cs, tm0, k0, M, KD, t, kvec);
-    stage_wrows(Wt, Bs, tn0, k0, cs.C, KD, t, kvec);
-    __syncthreads();
+  stage_patch_bwdd<STRIDE1>(dY, As[0], cs, tm0, 0, M, KD, t, kvec);
+  stage_wrows(Wt, Bs[0], tn0, 0, cs.C, KD, t, kvec);
+  __syncthreads();
+  int cur = 0;
+  for (int k0 = 0; k0 < KD; k0 += BK, cur ^= 1) {
+    if (k0 + BK < KD) {
+      stage_patch_bwdd<STRIDE1>(dY, As[cur ^ 1], cs, tm0, k0 + BK, M, KD, t,
+                                kvec);
+      stage_wrows(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs.C, KD, t, kvec);
+    }
     const int kfrag = (lane >> 4) * 8;
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
-      bf16x8 a = *(const bf16x8*)&As[wr * 32 + fm * 16 + (lane & 15)][kfrag];
+      bf16x8 a =
+          *(const bf16x8*)&As[cur][wr * 32 + fm * 16 + (lane & 15)][kfrag];
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn) {
-        bf16x8 b = *(const bf16x8*)&Bs[wc * 32 + fn * 16 + (lane & 15)][kfrag];
+        bf16x8 b =
+            *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
         acc[fm][fn] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm][fn], 0, 0, 0);
       }
@@ -385,8 +402,8 @@ DEVINL void stage_x_bwdw(const __bf16* __restrict__ X, __bf16 (*Sn)[BK + APAD],
 __global__ __launch_bounds__(256)
 void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ X,
                       float* __restrict__ dW, ConvShape cs, long pc) {
-  __shared__ __align__(16) __bf16 As[BM][BK + APAD];
-  __shared__ __align__(16) __bf16 Bs[BN][BK + APAD];
+  __shared__ __align__(16) __bf16 As[2][BM][BK + APAD];
+  __shared__ __align__(16) __bf16 Bs[2][BN][BK + APAD];
   const int KD = cs.R * cs.S * cs.C;
   const long Ptot = (long)cs.N * cs.Ho * cs.Wo;
   const bool cvec = (cs.C & 7) == 0;
@@ -401,17 +418,24 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
   const int wr = wave >> 1, wc = wave & 1;
 
   f32x4 acc[2][2] = {};
-  for (long p0 = ps; p0 < pe; p0 += BK) {
-    stage_dy_bwdw(dY, As, cs, tm0, p0, Ptot, t, kvec);
-    stage_x_bwdw(X, Bs, cs, tn0, p0, KD, Ptot, t, cvec);
-    __syncthreads();
+  stage_dy_bwdw(dY, As[0], cs, tm0, ps, Ptot, t, kvec);
+  stage_x_bwdw(X, Bs[0], cs, tn0, ps, KD, Ptot, t, cvec);
+  __syncthreads();
+  int cur = 0;
+  for (long p0 = ps; p0 < pe; p0 += BK, cur ^= 1) {
+    if (p0 + BK < pe) {
+      stage_dy_bwdw(dY, As[cur ^ 1], cs, tm0, p0 + BK, Ptot, t, kvec);
+      stage_x_bwdw(X, Bs[cur ^ 1], cs, tn0, p0 + BK, KD, Ptot, t, cvec);
+    }
     const int kfrag = (lane >> 4) * 8;
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
-      bf16x8 a = *(const bf16x8*)&As[wr * 32 + fm * 16 + (lane & 15)][kfrag];
+      bf16x8 a =
+          *(const bf16x8*)&As[cur][wr * 32 + fm * 16 + (lane & 15)][kfrag];
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn) {
-        bf16x8 b = *(const bf16x8*)&Bs[wc * 32 + fn * 16 + (lane & 15)][kfrag];
+        bf16x8 b =
+            *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
         acc[fm][fn] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm][fn], 0, 0, 0);
       }
