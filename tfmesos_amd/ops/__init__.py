@@ -395,6 +395,8 @@ class _BatchNormActFn(torch.autograd.Function):
             x = x.contiguous(memory_format=torch.channels_last)
             y, mean, invstd = _ext().bn_fwd(x, weight.float(), bias.float(),
                                             eps, relu)
+            # y is NOT saved for backward: the relu mask is recomputed
+            # from sign(g*xhat + b) inside the bwd kernels
         else:
             xf = x.float()
             mean = xf.mean(dim=(0, 2, 3))
@@ -406,17 +408,18 @@ class _BatchNormActFn(torch.autograd.Function):
             if relu:
                 y = torch.relu(y)
             y = y.to(x.dtype)
-        ctx.save_for_backward(x, y, weight, mean, invstd)
+        ctx.save_for_backward(x, y, weight, bias, mean, invstd)
         ctx.relu = relu
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, weight, mean, invstd = ctx.saved_tensors
+        x, y, weight, bias, mean, invstd = ctx.saved_tensors
         if x.is_cuda:
             dy = dy.contiguous(memory_format=torch.channels_last)
-            dx, dgamma, dbeta = _ext().bn_bwd(x, dy, y, weight.float(),
-                                              mean, invstd, ctx.relu)
+            dx, dgamma, dbeta = _ext().bn_bwd(x, dy, weight.float(),
+                                              bias.float(), mean, invstd,
+                                              ctx.relu)
         else:
             dyf = dy.contiguous().float()
             if ctx.relu:

@@ -118,11 +118,14 @@ void bn_apply_kernel(const __bf16* __restrict__ x,
   }
 }
 
+// The ReLU mask is recomputed as sign(g*xhat + b) == sign(y) instead
+// of reading the y stream — one fewer full activation read.
 template <bool RELU>
 __global__ __launch_bounds__(256)
 void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
                          const __bf16* __restrict__ dy,
-                         const __bf16* __restrict__ y,
+                         const float* __restrict__ g,
+                         const float* __restrict__ b,
                          const float* __restrict__ mean,
                          const float* __restrict__ invstd,
                          float* __restrict__ part, long P, int C, int Z) {
@@ -133,11 +136,12 @@ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
   float s1 = 0.f, s2 = 0.f;
   if (c < C) {
     const float mu = mean[c], is = invstd[c];
+    const float gc = RELU ? g[c] : 0.f, bc = RELU ? b[c] : 0.f;
     for (long p = (long)z * 4 + pr; p < P; p += (long)Z * 4) {
       const long i = p * C + c;
       float d = (float)dy[i];
-      if (RELU && (float)y[i] <= 0.f) d = 0.f;
       const float xh = ((float)x[i] - mu) * is;
+      if (RELU && gc * xh + bc <= 0.f) d = 0.f;
       s1 += d;
       s2 += d * xh;
     }
@@ -185,22 +189,25 @@ template <bool RELU, bool VEC>
 __global__ __launch_bounds__(256)
 void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
                          const __bf16* __restrict__ dy,
-                         const __bf16* __restrict__ y,
+                         const float* __restrict__ g,
+                         const float* __restrict__ b,
                          const float* __restrict__ mean,
                          const float* __restrict__ invstd,
-                         const float* __restrict__ g,
                          const float* __restrict__ s1n,
                          const float* __restrict__ s2n,
                          __bf16* __restrict__ dx, long P, int C) {
   // per-channel constants staged in LDS: dx = gs*(dy_eff - a - xhat*bb),
-  // xhat = (x - mu)*is
+  // xhat = (x - mu)*is; relu mask recomputed as g*xh + b > 0 (no y read)
   __shared__ float lgs[MAXC], la[MAXC], lbb[MAXC], lmu[MAXC], lis[MAXC];
+  __shared__ float lg[MAXC], lb[MAXC];
   for (int c = threadIdx.x; c < C; c += 256) {
     lgs[c] = g[c] * invstd[c];
     la[c] = s1n[c];
     lbb[c] = s2n[c];
     lmu[c] = mean[c];
     lis[c] = invstd[c];
+    lg[c] = g[c];
+    lb[c] = b[c];
   }
   __syncthreads();
   const long total = P * C;
@@ -211,15 +218,13 @@ void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
     if (VEC && i + 8 <= total) {
       bf16x8 xv = *(const bf16x8*)&x[i];
       bf16x8 dv = *(const bf16x8*)&dy[i];
-      bf16x8 yv;
-      if (RELU) yv = *(const bf16x8*)&y[i];
       bf16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int c = c0 + j;
         float d = (float)dv[j];
-        if (RELU && (float)yv[j] <= 0.f) d = 0.f;
         const float xh = ((float)xv[j] - lmu[c]) * lis[c];
+        if (RELU && lg[c] * xh + lb[c] <= 0.f) d = 0.f;
         o[j] = (__bf16)(lgs[c] * (d - la[c] - xh * lbb[c]));
       }
       *(bf16x8*)&dx[i] = o;
@@ -227,8 +232,8 @@ void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
       int c = c0;
       for (int j = 0; j < 8 && i + j < total; ++j) {
         float d = (float)dy[i + j];
-        if (RELU && (float)y[i + j] <= 0.f) d = 0.f;
         const float xh = ((float)x[i + j] - lmu[c]) * lis[c];
+        if (RELU && lg[c] * xh + lb[c] <= 0.f) d = 0.f;
         dx[i + j] = (__bf16)(lgs[c] * (d - la[c] - xh * lbb[c]));
         if (++c == C) c = 0;
       }
@@ -237,14 +242,14 @@ void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
 }
 
 inline int stats_slices(long P, int C) {
-  // target >=1024 workgroups across the (C/64) x Z grid, each slice
+  // target >=2048 workgroups across the (C/64) x Z grid, each slice
   // covering >=8 pixel rounds of 4 rows
   long cb = (C + 63) / 64;
-  long want = (1024 + cb - 1) / cb;
+  long want = (2048 + cb - 1) / cb;
   long per = P / (4 * 8);
   long z = want < per ? want : per;
   if (z < 1) z = 1;
-  if (z > 128) z = 128;
+  if (z > 256) z = 256;
   return (int)z;
 }
 
@@ -277,19 +282,19 @@ void launch_bn_fwd(const bf16_t* x, const float* g, const float* b,
 #undef APPLY
 }
 
-void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, const bf16_t* y,
-                   const float* g, const float* mean, const float* invstd,
+void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, const float* g,
+                   const float* b, const float* mean, const float* invstd,
                    bf16_t* dx, float* dgamma, float* dbeta, float* part,
                    float* s1n, float* s2n, long P, int C, int Z,
                    bool relu, hipStream_t stream) {
   dim3 sg(ceil_div(C, 64), Z), sb(256);
   if (relu)
     hipLaunchKernelGGL((bn_bwd_stats_kernel<true>), sg, sb, 0, stream,
-                       (const __bf16*)x, (const __bf16*)dy, (const __bf16*)y,
+                       (const __bf16*)x, (const __bf16*)dy, g, b,
                        mean, invstd, part, P, C, Z);
   else
     hipLaunchKernelGGL((bn_bwd_stats_kernel<false>), sg, sb, 0, stream,
-                       (const __bf16*)x, (const __bf16*)dy, (const __bf16*)y,
+                       (const __bf16*)x, (const __bf16*)dy, g, b,
                        mean, invstd, part, P, C, Z);
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(64), 0, stream,
                      part, dgamma, dbeta, s1n, s2n, C, Z, 1.f / (float)P);
@@ -297,8 +302,8 @@ void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, const bf16_t* y,
   const bool vec = (C & 7) == 0;
 #define APPLY(RELUv, VECv)                                                  \
   hipLaunchKernelGGL((bn_bwd_apply_kernel<RELUv, VECv>), ag, ab, 0, stream, \
-                     (const __bf16*)x, (const __bf16*)dy, (const __bf16*)y,  \
-                     mean, invstd, g, s1n, s2n, (__bf16*)dx, P, C)
+                     (const __bf16*)x, (const __bf16*)dy, g, b,              \
+                     mean, invstd, s1n, s2n, (__bf16*)dx, P, C)
   if (relu) { if (vec) APPLY(true, true); else APPLY(true, false); }
   else      { if (vec) APPLY(false, true); else APPLY(false, false); }
 #undef APPLY

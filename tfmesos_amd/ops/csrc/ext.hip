@@ -50,7 +50,7 @@ void launch_conv_bwd_weight(const bf16_t*, const bf16_t*, float*, int, int,
 void launch_bn_fwd(const bf16_t*, const float*, const float*, bf16_t*,
                    float*, float*, float*, long, int, int, float, bool,
                    hipStream_t);
-void launch_bn_bwd(const bf16_t*, const bf16_t*, const bf16_t*, const float*,
+void launch_bn_bwd(const bf16_t*, const bf16_t*, const float*, const float*,
                    const float*, const float*, bf16_t*, float*, float*,
                    float*, float*, float*, long, int, int, bool,
                    hipStream_t);
@@ -500,10 +500,10 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor g,
 }
 
 std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
-                                  torch::Tensor y, torch::Tensor g,
+                                  torch::Tensor g, torch::Tensor b,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   bool relu) {
-  TORCH_CHECK(is_cl(x) && is_cl(dy) && is_cl(y),
+  TORCH_CHECK(is_cl(x) && is_cl(dy),
               "bn_bwd: channels-last tensors required");
   const int C = x.size(1);
   const long P = x.numel() / C;
@@ -516,7 +516,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   auto s1n = torch::empty({C}, opts);
   auto s2n = torch::empty({C}, opts);
   launch_bn_bwd((const bf16_t*)x.data_ptr(), (const bf16_t*)dy.data_ptr(),
-                (const bf16_t*)y.data_ptr(), g.data_ptr<float>(),
+                g.data_ptr<float>(), b.data_ptr<float>(),
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
                 (bf16_t*)dx.data_ptr(), dgamma.data_ptr<float>(),
                 dbeta.data_ptr<float>(), part.data_ptr<float>(),
