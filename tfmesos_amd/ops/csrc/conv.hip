@@ -119,54 +119,47 @@ DEVINL void stage_wrows(const __bf16* __restrict__ Wt, __bf16 (*Sn)[BK + APAD],
   }
 }
 
-// 128(m) x 64(k-chan) block tile: 4 waves stacked along m (32 rows
-// each), 2x4 fragments per wave — twice the MFMA work per staged byte
-// of the 64x64 tile, with the 4 B-fragments hoisted and reused across
-// both A-fragments (PMC showed the 64x64 version at ~1% MFMA issue
-// density: the staging instruction stream dominated).
 template <bool BIAS, bool RELU>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt,
                      const float* __restrict__ bias, __bf16* __restrict__ Y,
                      ConvShape cs) {
-  __shared__ __align__(16) __bf16 As[2][2 * BM][BK + APAD];
+  // double-buffered LDS: stage tile i+1 while MFMA consumes tile i —
+  // one barrier per K-iteration, global-load latency overlapped
+  __shared__ __align__(16) __bf16 As[2][BM][BK + APAD];
   __shared__ __align__(16) __bf16 Bs[2][BN][BK + APAD];
   const long M = (long)cs.N * cs.Ho * cs.Wo;
   const int KD = cs.R * cs.S * cs.C;
   const bool cvec = (cs.C & 7) == 0;
-  const long tm0 = (long)blockIdx.x * (2 * BM);
+  const long tm0 = (long)blockIdx.x * BM;   // pixel tiles ride grid.x (2^31)
   const int tn0 = blockIdx.y * BN;
   const int t = threadIdx.x;
   const int lane = t & 63;
-  const int wave = t >> 6;          // 0..3, stacked along m
+  const int wave = t >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
 
-  f32x4 acc[2][4] = {};
+  f32x4 acc[2][2] = {};
   stage_patch_fwd(X, As[0], cs, tm0, 0, M, KD, t, cvec);
-  stage_patch_fwd(X, As[0] + BM, cs, tm0 + BM, 0, M, KD, t, cvec);
   stage_wrows(Wt, Bs[0], tn0, 0, cs.K, KD, t, true);
   __syncthreads();
   int cur = 0;
   for (int k0 = 0; k0 < KD; k0 += BK, cur ^= 1) {
     if (k0 + BK < KD) {
       stage_patch_fwd(X, As[cur ^ 1], cs, tm0, k0 + BK, M, KD, t, cvec);
-      stage_patch_fwd(X, As[cur ^ 1] + BM, cs, tm0 + BM, k0 + BK, M, KD, t,
-                      cvec);
       stage_wrows(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs.K, KD, t, true);
     }
     const int kfrag = (lane >> 4) * 8;
-    bf16x8 b[4];
-#pragma unroll
-    for (int fn = 0; fn < 4; ++fn)
-      b[fn] = *(const bf16x8*)&Bs[cur][fn * 16 + (lane & 15)][kfrag];
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
       bf16x8 a =
-          *(const bf16x8*)&As[cur][wave * 32 + fm * 16 + (lane & 15)][kfrag];
+          *(const bf16x8*)&As[cur][wr * 32 + fm * 16 + (lane & 15)][kfrag];
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn)
+      for (int fn = 0; fn < 2; ++fn) {
+        bf16x8 b =
+            *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
         acc[fm][fn] =
-            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b[fn], acc[fm][fn],
-                                                    0, 0, 0);
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm][fn], 0, 0, 0);
+      }
     }
     __syncthreads();
   }
@@ -175,13 +168,13 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
 #pragma unroll
   for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
-    for (int fn = 0; fn < 4; ++fn) {
-      const int k = tn0 + fn * 16 + (lane & 15);
+    for (int fn = 0; fn < 2; ++fn) {
+      const int k = tn0 + wc * 32 + fn * 16 + (lane & 15);
       if (k >= cs.K) continue;
       const float bv = BIAS ? bias[k] : 0.f;
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
-        const long pm = tm0 + wave * 32 + fm * 16 + (lane >> 4) * 4 + rr;
+        const long pm = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
         if (pm >= M) continue;
         float v = acc[fm][fn][rr] + bv;
         if (RELU) v = v > 0.f ? v : 0.f;
@@ -476,7 +469,7 @@ void launch_conv_fwd(const bf16_t* X, const bf16_t* W, const float* bias,
                      bool relu, hipStream_t stream) {
   ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q};
   const long M = (long)N * Ho * Wo;
-  dim3 grid((unsigned)((M + 2 * BM - 1) / (2 * BM)), ceil_div(K, BN));
+  dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(K, BN));
   dim3 block(256);
   if (bias) {
     if (relu)
