@@ -18,7 +18,8 @@ from tfmesos_amd import ops
 
 class Conv2d(nn.Module):
     """Conv on the hand-written implicit-GEMM kernel (bias folded into
-    the following BN, as in Inception-v3)."""
+    the following BN, as in Inception-v3). Weights live in the kernels'
+    native tap-major [K, R, S, C] layout — no per-call permute."""
 
     def __init__(self, cin, cout, kernel_size, stride=1, padding=0):
         super().__init__()
@@ -28,11 +29,12 @@ class Conv2d(nn.Module):
         self.padding = padding
         fan_in = cin * kernel_size[0] * kernel_size[1]
         self.weight = nn.Parameter(
-            torch.randn(cout, cin, *kernel_size) * (2.0 / fan_in) ** 0.5)
+            torch.randn(cout, kernel_size[0], kernel_size[1], cin)
+            * (2.0 / fan_in) ** 0.5)
 
     def forward(self, x):
         return ops.conv2d(x, self.weight, None, stride=self.stride,
-                          padding=self.padding)
+                          padding=self.padding, weight_format="krsc")
 
 
 class BatchNorm2d(nn.Module):

@@ -259,60 +259,79 @@ class _Conv2dFn(torch.autograd.Function):
     torch fp32 reference so model code runs in CI."""
 
     @staticmethod
-    def forward(ctx, x, w, bias, stride, padding):
+    def forward(ctx, x, w, bias, stride, padding, krsc):
         ctx.stride = stride
         ctx.padding = padding
         ctx.has_bias = bias is not None
+        ctx.krsc = krsc
         if x.is_cuda:
             # channels-last memory: gathers become contiguous channel
-            # runs (csrc/conv.hip); weights go tap-major [K,R,S,C]
+            # runs (csrc/conv.hip); weights are tap-major [K,R,S,C] —
+            # either natively (krsc params skip the per-call permute)
+            # or permuted here from KCRS
             xm = x.contiguous(memory_format=torch.channels_last)
-            wm = w.permute(0, 2, 3, 1).contiguous()
+            wm = w.contiguous() if krsc else \
+                w.permute(0, 2, 3, 1).contiguous()
             ctx.save_for_backward(xm, w)
             eb = bias.float() if bias is not None else \
                 torch.empty(0, device=x.device)
             return _ext().conv2d_fwd(xm, wm, eb, stride[0], stride[1],
                                      padding[0], padding[1], False)
         ctx.save_for_backward(x, w)
+        wf = w.permute(0, 3, 1, 2).float() if krsc else w.float()
         y = torch.nn.functional.conv2d(
-            x.float(), w.float(), bias.float() if bias is not None else None,
+            x.float(), wf, bias.float() if bias is not None else None,
             stride=stride, padding=padding)
         return y.to(x.dtype)
 
     @staticmethod
     def backward(ctx, dy):
         x, w = ctx.saved_tensors
+        krsc = ctx.krsc
         if x.is_cuda:
             dy = dy.contiguous(memory_format=torch.channels_last)
-            wt = w.permute(1, 2, 3, 0).contiguous()   # [C,R,S,K] = W^T
+            # [C,R,S,K] = W^T memory for the bwd-data gather
+            wt = (w.permute(3, 1, 2, 0) if krsc
+                  else w.permute(1, 2, 3, 0)).contiguous()
+            R, S = (w.shape[1], w.shape[2]) if krsc \
+                else (w.shape[2], w.shape[3])
             dx = _ext().conv2d_bwd_data(dy, wt, x.shape[2], x.shape[3],
                                         ctx.stride[0], ctx.stride[1],
                                         ctx.padding[0], ctx.padding[1])
-            dwm = _ext().conv2d_bwd_weight(dy, x, w.shape[2], w.shape[3],
+            dwm = _ext().conv2d_bwd_weight(dy, x, R, S,
                                            ctx.stride[0], ctx.stride[1],
                                            ctx.padding[0], ctx.padding[1])
-            dw = dwm.permute(0, 3, 1, 2).to(w.dtype)
+            # kernel emits tap-major [K,R,S,C]: native for krsc params
+            dw = dwm.to(w.dtype) if krsc \
+                else dwm.permute(0, 3, 1, 2).to(w.dtype)
         else:
             dy = dy.contiguous()
-            dyf, xf, wf = dy.float(), x.float(), w.float()
+            wk = w.permute(0, 3, 1, 2) if krsc else w
+            dyf, xf, wf = dy.float(), x.float(), wk.float()
             dx = torch.nn.grad.conv2d_input(
                 x.shape, wf, dyf, stride=ctx.stride, padding=ctx.padding
             ).to(x.dtype)
             dw = torch.nn.grad.conv2d_weight(
-                xf, w.shape, dyf, stride=ctx.stride, padding=ctx.padding
-            ).to(w.dtype)
+                xf, wk.shape, dyf, stride=ctx.stride, padding=ctx.padding)
+            if krsc:
+                dw = dw.permute(0, 2, 3, 1)
+            dw = dw.contiguous().to(w.dtype)
         db = dy.float().sum(dim=(0, 2, 3)) if ctx.has_bias else None
-        return dx, dw, db, None, None
+        return dx, dw, db, None, None, None
 
 
-def conv2d(x, w, bias=None, stride=1, padding=0):
-    """2-D convolution (NCHW): hand-written implicit-GEMM MFMA kernels on
-    GPU (csrc/conv.hip), torch fp32 reference on CPU. Differentiable."""
+def conv2d(x, w, bias=None, stride=1, padding=0, weight_format="kcrs"):
+    """2-D convolution (NCHW activations): hand-written implicit-GEMM
+    MFMA kernels on GPU (csrc/conv.hip), torch fp32 reference on CPU.
+    Differentiable. weight_format "kcrs" (torch layout) or "krsc"
+    (tap-major — the kernels' native layout; parameters stored this way
+    skip a permute+copy per call in fwd AND in the weight-grad path)."""
     if isinstance(stride, int):
         stride = (stride, stride)
     if isinstance(padding, int):
         padding = (padding, padding)
-    return _Conv2dFn.apply(x, w, bias, tuple(stride), tuple(padding))
+    return _Conv2dFn.apply(x, w, bias, tuple(stride), tuple(padding),
+                           weight_format == "krsc")
 
 
 class _SoftmaxXentFn(torch.autograd.Function):

@@ -516,10 +516,10 @@ void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
   ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q};
   const int KD = C * R * S;
   const long Ptot = (long)N * Ho * Wo;
-  // slice the reduction so the grid can fill the chip (>=512 WGs)
+  // slice the reduction so the grid can fill the chip (>=1024 WGs)
   const long tiles = (long)ceil_div(K, BM) * ceil_div(KD, BN);
   long zmax = (Ptot + BK - 1) / BK;
-  long zwant = 512 / tiles;
+  long zwant = 1024 / tiles;
   if (zwant < 1) zwant = 1;
   int z = (int)(zmax < zwant ? zmax : zwant);
   long pc = (Ptot + z - 1) / z;
