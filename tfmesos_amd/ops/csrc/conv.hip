@@ -29,12 +29,38 @@ namespace {
 constexpr int BM = 64, BN = 64, BK = 32;
 constexpr int APAD = 8;
 
+// Magic-number unsigned division (libdivide-style): integer divide on
+// CDNA is ~30 VALU cycles and the gathers decode several indices per
+// 8-strip; x / d == (x * m) >> 32 >> s for x < 2^31, m,s host-computed.
+struct FastDiv {
+  unsigned m;
+  int s;
+  int d;
+};
+
+inline FastDiv make_fdiv(int d) {
+  FastDiv f;
+  f.d = d;
+  if (d == 1) { f.m = 0; f.s = 0; return f; }
+  int s = 0;
+  while ((1LL << s) < d) ++s;
+  f.m = (unsigned)(((1ULL << (32 + s)) + d - 1) / d);
+  f.s = s;
+  return f;
+}
+
+DEVINL unsigned fdiv(unsigned x, FastDiv f) {
+  if (f.m == 0) return x;               // d == 1
+  return (unsigned)(((unsigned long long)x * f.m) >> 32) >> f.s;
+}
+
 struct ConvShape {
   int N, C, H, W;     // input (NHWC memory)
   int K, R, S;        // filter
   int Ho, Wo;         // output
   int U, V;           // stride
   int P, Q;           // pad
+  FastDiv dC, dS, dK, dWo, dW2, dHoWo, dHW;   // divisors for gather decode
 };
 
 // Load an 8-element channel run X[pix_base + c0 .. c0+8) into S[row][..],
@@ -64,14 +90,14 @@ DEVINL void stage_patch_fwd(const __bf16* __restrict__ X, __bf16 (*Sm)[BK + APAD
     for (int j = 0; j < 8; ++j) Sm[mx][kk0 + j] = (__bf16)0.f;
     return;
   }
-  const int rs = q0 / cs.C;            // tap (r,s) block (c0 % 8 == 0 when
+  const int rs = fdiv(q0, cs.dC);      // tap (r,s) block (c0 % 8 == 0 when
   const int c0 = q0 - rs * cs.C;       //  C % 8 == 0, so the run stays
-  const int r = rs / cs.S;             //  inside one (r,s))
+  const int r = fdiv(rs, cs.dS);       //  inside one (r,s))
   const int s = rs - r * cs.S;
   const int HoWo = cs.Ho * cs.Wo;
-  const int n = (int)(pm / HoWo);
+  const int n = (int)fdiv((unsigned)pm, cs.dHoWo);
   const int rem = (int)(pm - (long)n * HoWo);
-  const int ho = rem / cs.Wo;
+  const int ho = fdiv(rem, cs.dWo);
   const int wo = rem - ho * cs.Wo;
   const int hi = ho * cs.U + r - cs.P;
   const int wi = wo * cs.V + s - cs.Q;
@@ -202,14 +228,14 @@ DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
     for (int j = 0; j < 8; ++j) Sm[mx][kk0 + j] = (__bf16)0.f;
     return;
   }
-  const int rs = q0 / cs.K;
+  const int rs = fdiv(q0, cs.dK);
   const int kc0 = q0 - rs * cs.K;
-  const int r = rs / cs.S;
+  const int r = fdiv(rs, cs.dS);
   const int s = rs - r * cs.S;
   const int HWi = cs.H * cs.W;
-  const int n = (int)(pm / HWi);
+  const int n = (int)fdiv((unsigned)pm, cs.dHW);
   const int rem = (int)(pm - (long)n * HWi);
-  const int h = rem / cs.W;
+  const int h = fdiv(rem, cs.dW2);
   const int w = rem - h * cs.W;
   const int hn = h + cs.P - r, wn = w + cs.Q - s;
 
@@ -359,14 +385,14 @@ DEVINL void stage_x_bwdw(const __bf16* __restrict__ X, __bf16 (*Sn)[BK + APAD],
     for (int j = 0; j < 8; ++j) Sn[qq0 + j][px] = (__bf16)0.f;
     return;
   }
-  const int rs = q0 / cs.C;
+  const int rs = fdiv(q0, cs.dC);
   const int c0 = q0 - rs * cs.C;
-  const int r = rs / cs.S;
+  const int r = fdiv(rs, cs.dS);
   const int s = rs - r * cs.S;
   const long HoWo = (long)cs.Ho * cs.Wo;
-  const int n = (int)(p / HoWo);
+  const int n = (int)fdiv((unsigned)p, cs.dHoWo);
   const int rem = (int)(p - (long)n * HoWo);
-  const int ho = rem / cs.Wo;
+  const int ho = fdiv(rem, cs.dWo);
   const int wo = rem - ho * cs.Wo;
   const int hi = ho * cs.U + r - cs.P;
   const int wi = wo * cs.V + s - cs.Q;
@@ -467,7 +493,9 @@ void launch_conv_fwd(const bf16_t* X, const bf16_t* W, const float* bias,
                      bf16_t* Y, int N, int C, int H, int Wd, int K, int R,
                      int S, int Ho, int Wo, int U, int V, int P, int Q,
                      bool relu, hipStream_t stream) {
-  ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q};
+  ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q,
+               make_fdiv(C), make_fdiv(S), make_fdiv(K), make_fdiv(Wo),
+               make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd)};
   const long M = (long)N * Ho * Wo;
   dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(K, BN));
   dim3 block(256);
@@ -497,7 +525,9 @@ void launch_conv_bwd_data(const bf16_t* dY, const bf16_t* Wt, bf16_t* dX,
                           int Ho, int Wo, int U, int V, int P, int Q,
                           hipStream_t stream) {
   // Wt: host-permuted W^T, memory [C][R*S*K]
-  ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q};
+  ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q,
+               make_fdiv(C), make_fdiv(S), make_fdiv(K), make_fdiv(Wo),
+               make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd)};
   const long M = (long)N * H * Wd;
   dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(C, BN));
   dim3 block(256);
@@ -513,7 +543,9 @@ void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
                             int N, int C, int H, int Wd, int K, int R, int S,
                             int Ho, int Wo, int U, int V, int P, int Q,
                             hipStream_t stream) {
-  ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q};
+  ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q,
+               make_fdiv(C), make_fdiv(S), make_fdiv(K), make_fdiv(Wo),
+               make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd)};
   const int KD = C * R * S;
   const long Ptot = (long)N * Ho * Wo;
   // slice the reduction so the grid can fill the chip (>=1024 WGs)
