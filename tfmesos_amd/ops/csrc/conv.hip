@@ -29,29 +29,28 @@ namespace {
 constexpr int BM = 64, BN = 64, BK = 32;
 constexpr int APAD = 8;
 
-// Magic-number unsigned division (libdivide-style): integer divide on
-// CDNA is ~30 VALU cycles and the gathers decode several indices per
-// 8-strip; x / d == (x * m) >> 32 >> s for x < 2^31, m,s host-computed.
+// Magic-number unsigned division (Granlund-Montgomery): integer divide
+// on CDNA is ~30 VALU cycles and the gathers decode several indices per
+// 8-strip. With L = ceil(log2 d) and M = ceil(2^(31+L) / d),
+// floor(x/d) == (x * M) >> (31+L) exactly for all 0 <= x < 2^31.
 struct FastDiv {
-  unsigned m;
-  int s;
+  unsigned long long m;
+  int s;                 // 31 + L
   int d;
 };
 
 inline FastDiv make_fdiv(int d) {
   FastDiv f;
   f.d = d;
-  if (d == 1) { f.m = 0; f.s = 0; return f; }
-  int s = 0;
-  while ((1LL << s) < d) ++s;
-  f.m = (unsigned)(((1ULL << (32 + s)) + d - 1) / d);
-  f.s = s;
+  int L = 0;
+  while ((1LL << L) < d) ++L;
+  f.s = 31 + L;
+  f.m = ((1ULL << f.s) + d - 1) / d;
   return f;
 }
 
 DEVINL unsigned fdiv(unsigned x, FastDiv f) {
-  if (f.m == 0) return x;               // d == 1
-  return (unsigned)(((unsigned long long)x * f.m) >> 32) >> f.s;
+  return (unsigned)(((unsigned long long)x * f.m) >> f.s);
 }
 
 struct ConvShape {
