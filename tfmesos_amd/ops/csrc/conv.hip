@@ -211,9 +211,10 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
 // --------------------------------------------------------------- bwd-data
 
 // As[m][kk]: m = input pixel, kk = tap (r,s,k) with k innermost, reading
-// dY[n, (h+P-r)/U, (w+Q-s)/V, k] (contiguous in k). STRIDE1 removes the
-// divisibility tests.
-template <bool STRIDE1>
+// dY[n, (h+P-r)/U, (w+Q-s)/V, k] (contiguous in k). STRIDE specializes
+// the common cases: 1 removes the divisibility tests, 2 turns them into
+// bit ops (runtime %/ by the stride costs ~30 VALU cycles each).
+template <int STRIDE>
 DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
                              __bf16 (*Sm)[BK + APAD], const ConvShape cs,
                              long m0, int k0, long M, int KD, int t,
@@ -241,9 +242,13 @@ DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
   if (kvec && kc0 + 8 <= cs.K) {
     bool ok;
     int ho, wo;
-    if (STRIDE1) {
+    if (STRIDE == 1) {
       ho = hn; wo = wn;
       ok = hn >= 0 && hn < cs.Ho && wn >= 0 && wn < cs.Wo;
+    } else if (STRIDE == 2) {
+      ok = hn >= 0 && wn >= 0 && !(hn & 1) && !(wn & 1);
+      ho = hn >> 1; wo = wn >> 1;
+      ok = ok && ho < cs.Ho && wo < cs.Wo;
     } else {
       ok = hn >= 0 && wn >= 0 && hn % cs.U == 0 && wn % cs.V == 0;
       ho = hn / cs.U; wo = wn / cs.V;
@@ -267,9 +272,13 @@ DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
       const int hnj = h + cs.P - rr2, wnj = w + cs.Q - ss2;
       bool ok;
       int ho, wo;
-      if (STRIDE1) {
+      if (STRIDE == 1) {
         ho = hnj; wo = wnj;
         ok = hnj >= 0 && hnj < cs.Ho && wnj >= 0 && wnj < cs.Wo;
+      } else if (STRIDE == 2) {
+        ok = hnj >= 0 && wnj >= 0 && !(hnj & 1) && !(wnj & 1);
+        ho = hnj >> 1; wo = wnj >> 1;
+        ok = ok && ho < cs.Ho && wo < cs.Wo;
       } else {
         ok = hnj >= 0 && wnj >= 0 && hnj % cs.U == 0 && wnj % cs.V == 0;
         ho = hnj / cs.U; wo = wnj / cs.V;
@@ -283,7 +292,7 @@ DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
   }
 }
 
-template <bool STRIDE1>
+template <int STRIDE>
 __global__ __launch_bounds__(256)
 void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ Wt,
                       __bf16* __restrict__ dX, ConvShape cs) {
@@ -301,14 +310,14 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
   const int wr = wave >> 1, wc = wave & 1;
 
   f32x4 acc[2][2] = {};
-  stage_patch_bwdd<STRIDE1>(dY, As[0], cs, tm0, 0, M, KD, t, kvec);
+  stage_patch_bwdd<STRIDE>(dY, As[0], cs, tm0, 0, M, KD, t, kvec);
   stage_wrows(Wt, Bs[0], tn0, 0, cs.C, KD, t, kvec);
   __syncthreads();
   int cur = 0;
   for (int k0 = 0; k0 < KD; k0 += BK, cur ^= 1) {
     if (k0 + BK < KD) {
-      stage_patch_bwdd<STRIDE1>(dY, As[cur ^ 1], cs, tm0, k0 + BK, M, KD, t,
-                                kvec);
+      stage_patch_bwdd<STRIDE>(dY, As[cur ^ 1], cs, tm0, k0 + BK, M, KD, t,
+                               kvec);
       stage_wrows(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs.C, KD, t, kvec);
     }
     const int kfrag = (lane >> 4) * 8;
@@ -531,10 +540,13 @@ void launch_conv_bwd_data(const bf16_t* dY, const bf16_t* Wt, bf16_t* dX,
   dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(C, BN));
   dim3 block(256);
   if (U == 1 && V == 1)
-    hipLaunchKernelGGL((conv_bwdd_kernel<true>), grid, block, 0, stream,
+    hipLaunchKernelGGL((conv_bwdd_kernel<1>), grid, block, 0, stream,
+                       (const __bf16*)dY, (const __bf16*)Wt, (__bf16*)dX, cs);
+  else if (U == 2 && V == 2)
+    hipLaunchKernelGGL((conv_bwdd_kernel<2>), grid, block, 0, stream,
                        (const __bf16*)dY, (const __bf16*)Wt, (__bf16*)dX, cs);
   else
-    hipLaunchKernelGGL((conv_bwdd_kernel<false>), grid, block, 0, stream,
+    hipLaunchKernelGGL((conv_bwdd_kernel<0>), grid, block, 0, stream,
                        (const __bf16*)dY, (const __bf16*)Wt, (__bf16*)dX, cs);
 }
 
