@@ -412,8 +412,9 @@ class _BatchNormActFn(torch.autograd.Function):
     def forward(ctx, x, weight, bias, eps, relu):
         if x.is_cuda:
             x = x.contiguous(memory_format=torch.channels_last)
-            y, mean, invstd = _ext().bn_fwd(x, weight.float(), bias.float(),
-                                            eps, relu)
+            y, mean, invstd = _ext().bn_fwd(
+                x, weight.to(torch.bfloat16), bias.to(torch.bfloat16),
+                eps, relu)
             # y is NOT saved for backward: the relu mask is recomputed
             # from sign(g*xhat + b) inside the bwd kernels
         else:
@@ -436,9 +437,9 @@ class _BatchNormActFn(torch.autograd.Function):
         x, y, weight, bias, mean, invstd = ctx.saved_tensors
         if x.is_cuda:
             dy = dy.contiguous(memory_format=torch.channels_last)
-            dx, dgamma, dbeta = _ext().bn_bwd(x, dy, weight.float(),
-                                              bias.float(), mean, invstd,
-                                              ctx.relu)
+            dx, dgamma, dbeta = _ext().bn_bwd(
+                x, dy, weight.to(torch.bfloat16), bias.to(torch.bfloat16),
+                mean, invstd, ctx.relu)
         else:
             dyf = dy.contiguous().float()
             if ctx.relu:

@@ -82,13 +82,13 @@ __global__ __launch_bounds__(256)
 void bn_apply_kernel(const __bf16* __restrict__ x,
                      const float* __restrict__ mean,
                      const float* __restrict__ invstd,
-                     const float* __restrict__ g, const float* __restrict__ b,
+                     const __bf16* __restrict__ g, const __bf16* __restrict__ b,
                      __bf16* __restrict__ y, long P, int C) {
   __shared__ float sc[MAXC], sh[MAXC];
   for (int c = threadIdx.x; c < C; c += 256) {
-    const float s = g[c] * invstd[c];
+    const float s = (float)g[c] * invstd[c];
     sc[c] = s;
-    sh[c] = b[c] - mean[c] * s;
+    sh[c] = (float)b[c] - mean[c] * s;
   }
   __syncthreads();
   const long total = P * C;
@@ -124,8 +124,8 @@ template <bool RELU>
 __global__ __launch_bounds__(256)
 void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
                          const __bf16* __restrict__ dy,
-                         const float* __restrict__ g,
-                         const float* __restrict__ b,
+                         const __bf16* __restrict__ g,
+                         const __bf16* __restrict__ b,
                          const float* __restrict__ mean,
                          const float* __restrict__ invstd,
                          float* __restrict__ part, long P, int C, int Z) {
@@ -136,7 +136,8 @@ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
   float s1 = 0.f, s2 = 0.f;
   if (c < C) {
     const float mu = mean[c], is = invstd[c];
-    const float gc = RELU ? g[c] : 0.f, bc = RELU ? b[c] : 0.f;
+    const float gc = RELU ? (float)g[c] : 0.f;
+    const float bc = RELU ? (float)b[c] : 0.f;
     for (long p = (long)z * 4 + pr; p < P; p += (long)Z * 4) {
       const long i = p * C + c;
       float d = (float)dy[i];
@@ -160,8 +161,8 @@ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
 
 __global__ __launch_bounds__(64)
 void bn_bwd_finalize_kernel(const float* __restrict__ part,
-                            float* __restrict__ dgamma,
-                            float* __restrict__ dbeta,
+                            __bf16* __restrict__ dgamma,
+                            __bf16* __restrict__ dbeta,
                             float* __restrict__ s1n,
                             float* __restrict__ s2n,
                             int C, int Z, float inv_count) {
@@ -178,8 +179,8 @@ void bn_bwd_finalize_kernel(const float* __restrict__ part,
     s2 += __shfl_xor(s2, off, 64);
   }
   if (l == 0) {
-    dbeta[c] = s1;
-    dgamma[c] = s2;
+    dbeta[c] = (__bf16)s1;
+    dgamma[c] = (__bf16)s2;
     s1n[c] = s1 * inv_count;
     s2n[c] = s2 * inv_count;
   }
@@ -189,8 +190,8 @@ template <bool RELU, bool VEC>
 __global__ __launch_bounds__(256)
 void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
                          const __bf16* __restrict__ dy,
-                         const float* __restrict__ g,
-                         const float* __restrict__ b,
+                         const __bf16* __restrict__ g,
+                         const __bf16* __restrict__ b,
                          const float* __restrict__ mean,
                          const float* __restrict__ invstd,
                          const float* __restrict__ s1n,
@@ -201,13 +202,13 @@ void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
   __shared__ float lgs[MAXC], la[MAXC], lbb[MAXC], lmu[MAXC], lis[MAXC];
   __shared__ float lg[MAXC], lb[MAXC];
   for (int c = threadIdx.x; c < C; c += 256) {
-    lgs[c] = g[c] * invstd[c];
+    lgs[c] = (float)g[c] * invstd[c];
     la[c] = s1n[c];
     lbb[c] = s2n[c];
     lmu[c] = mean[c];
     lis[c] = invstd[c];
-    lg[c] = g[c];
-    lb[c] = b[c];
+    lg[c] = (float)g[c];
+    lb[c] = (float)b[c];
   }
   __syncthreads();
   const long total = P * C;
@@ -262,7 +263,7 @@ inline unsigned ew_grid(long total) {
 
 }  // namespace
 
-void launch_bn_fwd(const bf16_t* x, const float* g, const float* b,
+void launch_bn_fwd(const bf16_t* x, const bf16_t* g, const bf16_t* b,
                    bf16_t* y, float* mean, float* invstd, float* part,
                    long P, int C, int Z, float eps, bool relu,
                    hipStream_t stream) {
@@ -276,34 +277,37 @@ void launch_bn_fwd(const bf16_t* x, const float* g, const float* b,
   const bool vec = (C & 7) == 0;
 #define APPLY(RELUv, VECv)                                                  \
   hipLaunchKernelGGL((bn_apply_kernel<RELUv, VECv>), ag, ab, 0, stream,     \
-                     (const __bf16*)x, mean, invstd, g, b, (__bf16*)y, P, C)
+                     (const __bf16*)x, mean, invstd, (const __bf16*)g,       \
+                     (const __bf16*)b, (__bf16*)y, P, C)
   if (relu) { if (vec) APPLY(true, true); else APPLY(true, false); }
   else      { if (vec) APPLY(false, true); else APPLY(false, false); }
 #undef APPLY
 }
 
-void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, const float* g,
-                   const float* b, const float* mean, const float* invstd,
-                   bf16_t* dx, float* dgamma, float* dbeta, float* part,
+void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, const bf16_t* g,
+                   const bf16_t* b, const float* mean, const float* invstd,
+                   bf16_t* dx, bf16_t* dgamma, bf16_t* dbeta, float* part,
                    float* s1n, float* s2n, long P, int C, int Z,
                    bool relu, hipStream_t stream) {
   dim3 sg(ceil_div(C, 64), Z), sb(256);
   if (relu)
     hipLaunchKernelGGL((bn_bwd_stats_kernel<true>), sg, sb, 0, stream,
-                       (const __bf16*)x, (const __bf16*)dy, g, b,
-                       mean, invstd, part, P, C, Z);
+                       (const __bf16*)x, (const __bf16*)dy, (const __bf16*)g,
+                       (const __bf16*)b, mean, invstd, part, P, C, Z);
   else
     hipLaunchKernelGGL((bn_bwd_stats_kernel<false>), sg, sb, 0, stream,
-                       (const __bf16*)x, (const __bf16*)dy, g, b,
-                       mean, invstd, part, P, C, Z);
+                       (const __bf16*)x, (const __bf16*)dy, (const __bf16*)g,
+                       (const __bf16*)b, mean, invstd, part, P, C, Z);
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(64), 0, stream,
-                     part, dgamma, dbeta, s1n, s2n, C, Z, 1.f / (float)P);
+                     part, (__bf16*)dgamma, (__bf16*)dbeta, s1n, s2n, C, Z,
+                     1.f / (float)P);
   dim3 ag(ew_grid(P * (long)C)), ab(256);
   const bool vec = (C & 7) == 0;
 #define APPLY(RELUv, VECv)                                                  \
   hipLaunchKernelGGL((bn_bwd_apply_kernel<RELUv, VECv>), ag, ab, 0, stream, \
-                     (const __bf16*)x, (const __bf16*)dy, g, b,              \
-                     mean, invstd, s1n, s2n, (__bf16*)dx, P, C)
+                     (const __bf16*)x, (const __bf16*)dy, (const __bf16*)g,  \
+                     (const __bf16*)b, mean, invstd, s1n, s2n, (__bf16*)dx,  \
+                     P, C)
   if (relu) { if (vec) APPLY(true, true); else APPLY(true, false); }
   else      { if (vec) APPLY(false, true); else APPLY(false, false); }
 #undef APPLY

@@ -47,11 +47,11 @@ void launch_conv_bwd_data(const bf16_t*, const bf16_t*, bf16_t*, int, int,
 void launch_conv_bwd_weight(const bf16_t*, const bf16_t*, float*, int, int,
                             int, int, int, int, int, int, int, int, int,
                             int, int, hipStream_t);
-void launch_bn_fwd(const bf16_t*, const float*, const float*, bf16_t*,
+void launch_bn_fwd(const bf16_t*, const bf16_t*, const bf16_t*, bf16_t*,
                    float*, float*, float*, long, int, int, float, bool,
                    hipStream_t);
-void launch_bn_bwd(const bf16_t*, const bf16_t*, const float*, const float*,
-                   const float*, const float*, bf16_t*, float*, float*,
+void launch_bn_bwd(const bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*,
+                   const float*, const float*, bf16_t*, bf16_t*, bf16_t*,
                    float*, float*, float*, long, int, int, bool,
                    hipStream_t);
 int bn_stats_slices(long, int);
@@ -462,8 +462,15 @@ torch::Tensor conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
                                 long pad_h, long pad_w) {
   TORCH_CHECK(dy.is_cuda() && x.is_cuda() && is_cl(dy) && is_cl(x),
               "dy/x must be channels-last");
-  auto dw = torch::zeros({dy.size(1), R, S, x.size(1)},
-                         x.options().dtype(torch::kFloat32));
+  // zero-init only when the reduction is z-sliced (atomic accumulate);
+  // single-slice grids overwrite every element
+  const long KD = x.size(1) * R * S;
+  const long tiles = ((dy.size(1) + 63) / 64) * ((KD + 63) / 64);
+  const bool sliced = tiles < 1024 &&
+      (long)x.size(0) * dy.size(2) * dy.size(3) > 32;
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto dw = sliced ? torch::zeros({dy.size(1), R, S, x.size(1)}, opts)
+                   : torch::empty({dy.size(1), R, S, x.size(1)}, opts);
   launch_conv_bwd_weight((const bf16_t*)dy.data_ptr(),
                          (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(),
                          x.size(0), x.size(1), x.size(2), x.size(3),
@@ -483,16 +490,16 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor g,
   const long P = x.numel() / C;
   TORCH_CHECK(C <= bn_max_channels(), "bn: C > LDS staging bound");
   TORCH_CHECK(g.numel() == C && b.numel() == C &&
-              g.scalar_type() == torch::kFloat32 &&
-              b.scalar_type() == torch::kFloat32, "g/b must be fp32 [C]");
+              g.scalar_type() == torch::kBFloat16 &&
+              b.scalar_type() == torch::kBFloat16, "g/b must be bf16 [C]");
   const int Z = bn_stats_slices(P, C);
   auto opts = x.options().dtype(torch::kFloat32);
   auto y = torch::empty_like(x);
   auto mean = torch::empty({C}, opts);
   auto invstd = torch::empty({C}, opts);
   auto part = torch::empty({(long)Z * C * 2}, opts);
-  launch_bn_fwd((const bf16_t*)x.data_ptr(), g.data_ptr<float>(),
-                b.data_ptr<float>(), (bf16_t*)y.data_ptr(),
+  launch_bn_fwd((const bf16_t*)x.data_ptr(), (const bf16_t*)g.data_ptr(),
+                (const bf16_t*)b.data_ptr(), (bf16_t*)y.data_ptr(),
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
                 part.data_ptr<float>(), P, C, Z, (float)eps, relu,
                 cur_stream());
@@ -510,16 +517,16 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   const int Z = bn_stats_slices(P, C);
   auto opts = x.options().dtype(torch::kFloat32);
   auto dx = torch::empty_like(x);
-  auto dgamma = torch::empty({C}, opts);
-  auto dbeta = torch::empty({C}, opts);
+  auto dgamma = torch::empty({C}, x.options());
+  auto dbeta = torch::empty({C}, x.options());
   auto part = torch::empty({(long)Z * C * 2}, opts);
   auto s1n = torch::empty({C}, opts);
   auto s2n = torch::empty({C}, opts);
   launch_bn_bwd((const bf16_t*)x.data_ptr(), (const bf16_t*)dy.data_ptr(),
-                g.data_ptr<float>(), b.data_ptr<float>(),
+                (const bf16_t*)g.data_ptr(), (const bf16_t*)b.data_ptr(),
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                (bf16_t*)dx.data_ptr(), dgamma.data_ptr<float>(),
-                dbeta.data_ptr<float>(), part.data_ptr<float>(),
+                (bf16_t*)dx.data_ptr(), (bf16_t*)dgamma.data_ptr(),
+                (bf16_t*)dbeta.data_ptr(), part.data_ptr<float>(),
                 s1n.data_ptr<float>(), s2n.data_ptr<float>(), P, C, Z,
                 relu, cur_stream());
   return {dx, dgamma, dbeta};
