@@ -439,7 +439,12 @@ def test_batch_norm_act_fwd_bwd(shape, relu):
     ref.backward(dy.float().cpu())
 
     assert (y.float().cpu() - ref).abs().max() < 0.05
-    assert (x.grad.float().cpu() - xf.grad).abs().max() < 0.05
+    # at exact relu boundaries (|y| within bf16 rounding of 0) the
+    # subgradient choice may differ from the fp32 reference: compare dx
+    # only where both agree on the mask
+    agree = ((y.float().cpu() > 0) == (ref > 0)) | (~torch.tensor(relu))
+    dxe = ((x.grad.float().cpu() - xf.grad).abs() * agree).max()
+    assert dxe < 0.05, dxe
     assert (g.grad.cpu() - gf.grad).abs().max() < \
         0.02 * gf.grad.abs().max() + 0.1
     assert (b.grad.cpu() - bff.grad).abs().max() < \

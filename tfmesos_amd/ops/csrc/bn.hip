@@ -118,8 +118,10 @@ void bn_apply_kernel(const __bf16* __restrict__ x,
   }
 }
 
-// The ReLU mask is recomputed as sign(g*xhat + b) == sign(y) instead
-// of reading the y stream — one fewer full activation read.
+// The ReLU mask is recomputed instead of reading the y stream (one
+// fewer full activation read), using the SAME expression and bf16
+// rounding as the forward apply (y = bf16(x*sc + sh)) so the mask is
+// bit-exact with what the forward produced.
 template <bool RELU>
 __global__ __launch_bounds__(256)
 void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
@@ -136,13 +138,14 @@ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
   float s1 = 0.f, s2 = 0.f;
   if (c < C) {
     const float mu = mean[c], is = invstd[c];
-    const float gc = RELU ? (float)g[c] : 0.f;
-    const float bc = RELU ? (float)b[c] : 0.f;
+    const float sc = RELU ? (float)g[c] * is : 0.f;
+    const float sh = RELU ? (float)b[c] - mu * sc : 0.f;
     for (long p = (long)z * 4 + pr; p < P; p += (long)Z * 4) {
       const long i = p * C + c;
+      const float xv = (float)x[i];
       float d = (float)dy[i];
-      const float xh = ((float)x[i] - mu) * is;
-      if (RELU && gc * xh + bc <= 0.f) d = 0.f;
+      const float xh = (xv - mu) * is;
+      if (RELU && (float)(__bf16)(xv * sc + sh) <= 0.f) d = 0.f;
       s1 += d;
       s2 += d * xh;
     }
@@ -198,17 +201,18 @@ void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
                          const float* __restrict__ s2n,
                          __bf16* __restrict__ dx, long P, int C) {
   // per-channel constants staged in LDS: dx = gs*(dy_eff - a - xhat*bb),
-  // xhat = (x - mu)*is; relu mask recomputed as g*xh + b > 0 (no y read)
+  // xhat = (x - mu)*is; relu mask recomputed EXACTLY as the forward
+  // wrote it: y = bf16(x*sc + sh) (no y read)
   __shared__ float lgs[MAXC], la[MAXC], lbb[MAXC], lmu[MAXC], lis[MAXC];
-  __shared__ float lg[MAXC], lb[MAXC];
+  __shared__ float lsh[MAXC];
   for (int c = threadIdx.x; c < C; c += 256) {
-    lgs[c] = (float)g[c] * invstd[c];
+    const float sc = (float)g[c] * invstd[c];
+    lgs[c] = sc;
     la[c] = s1n[c];
     lbb[c] = s2n[c];
     lmu[c] = mean[c];
     lis[c] = invstd[c];
-    lg[c] = (float)g[c];
-    lb[c] = (float)b[c];
+    lsh[c] = (float)b[c] - mean[c] * sc;
   }
   __syncthreads();
   const long total = P * C;
@@ -223,18 +227,20 @@ void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int c = c0 + j;
+        const float x1 = (float)xv[j];
         float d = (float)dv[j];
-        const float xh = ((float)xv[j] - lmu[c]) * lis[c];
-        if (RELU && lg[c] * xh + lb[c] <= 0.f) d = 0.f;
+        const float xh = (x1 - lmu[c]) * lis[c];
+        if (RELU && (float)(__bf16)(x1 * lgs[c] + lsh[c]) <= 0.f) d = 0.f;
         o[j] = (__bf16)(lgs[c] * (d - la[c] - xh * lbb[c]));
       }
       *(bf16x8*)&dx[i] = o;
     } else {
       int c = c0;
       for (int j = 0; j < 8 && i + j < total; ++j) {
+        const float x1 = (float)x[i + j];
         float d = (float)dy[i + j];
-        const float xh = ((float)x[i + j] - lmu[c]) * lis[c];
-        if (RELU && lg[c] * xh + lb[c] <= 0.f) d = 0.f;
+        const float xh = (x1 - lmu[c]) * lis[c];
+        if (RELU && (float)(__bf16)(x1 * lgs[c] + lsh[c]) <= 0.f) d = 0.f;
         dx[i + j] = (__bf16)(lgs[c] * (d - la[c] - xh * lbb[c]));
         if (++c == C) c = 0;
       }
