@@ -21,6 +21,7 @@ sources = [
     os.path.join(CSRC, "gemm.hip"),
     os.path.join(CSRC, "conv.hip"),
     os.path.join(CSRC, "bn.hip"),
+    os.path.join(CSRC, "pool.hip"),
     os.path.join(CSRC, "softmax_xent.hip"),
     os.path.join(CSRC, "embedding.hip"),
     os.path.join(CSRC, "elementwise.hip"),

@@ -477,3 +477,20 @@ def test_mlp_head_fused_matches_composed():
     assert abs(float(loss) - float(loss2)) < 1e-2
     assert (dl.float() - dl2.float()).abs().max() < 1e-3
     assert (dh.float() - dh2.float()).abs().max() < 2e-3
+
+
+def test_avg_pool3x3_fwd_bwd():
+    from tfmesos_amd import ops
+    torch.manual_seed(60)
+    for shape in [(2, 16, 9, 11), (1, 7, 8, 8)]:   # vec + scalar-C paths
+        x = bf(torch.randn(*shape)).requires_grad_(True)
+        y = ops.avg_pool3x3(x)
+        ref = torch.nn.functional.avg_pool2d(
+            x.detach().float().cpu(), 3, stride=1, padding=1)
+        assert (y.float().cpu() - ref).abs().max() < 0.02, shape
+        dy = bf(torch.randn(*shape))
+        y.backward(dy)
+        xf = x.detach().float().cpu().requires_grad_(True)
+        r2 = torch.nn.functional.avg_pool2d(xf, 3, stride=1, padding=1)
+        r2.backward(dy.float().cpu())
+        assert (x.grad.float().cpu() - xf.grad).abs().max() < 0.02, shape

@@ -66,7 +66,8 @@ class BasicConv2d(nn.Module):
 
 
 def _avg_pool(x, k, stride=1, padding=1):
-    # bf16 native (channels-last aten kernel); a 3x3 mean is safe in bf16
+    if k == 3 and stride == 1 and padding == 1 and x.is_cuda:
+        return ops.avg_pool3x3(x)    # hand-written stencil (csrc/pool.hip)
     return torch.nn.functional.avg_pool2d(x, k, stride=stride,
                                           padding=padding)
 

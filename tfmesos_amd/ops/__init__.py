@@ -482,3 +482,30 @@ def mlp_head_fused(h, w, b, labels, scale=None):
     d *= s
     dh = (d @ wf.t()) * (hf > 0)
     return loss, d.to(h.dtype), dh.to(h.dtype)
+
+
+class _AvgPool3x3Fn(torch.autograd.Function):
+    """3x3 stride-1 pad-1 average pool (the Inception block pool) on a
+    channels-last HIP stencil kernel; the stencil is symmetric so the
+    backward is the same kernel applied to dy."""
+
+    @staticmethod
+    def forward(ctx, x):
+        if x.is_cuda:
+            return _ext().avg_pool3x3(
+                x.contiguous(memory_format=torch.channels_last))
+        return torch.nn.functional.avg_pool2d(
+            x.float(), 3, stride=1, padding=1).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        if dy.is_cuda:
+            return _ext().avg_pool3x3(
+                dy.contiguous(memory_format=torch.channels_last))
+        return torch.nn.functional.avg_pool2d(
+            dy.float(), 3, stride=1, padding=1).to(dy.dtype)
+
+
+def avg_pool3x3(x):
+    """Differentiable 3x3/s1/p1 average pool (count_include_pad)."""
+    return _AvgPool3x3Fn.apply(x)

@@ -56,6 +56,7 @@ void launch_bn_bwd(const bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*,
                    hipStream_t);
 int bn_stats_slices(long, int);
 int bn_max_channels();
+void launch_avg3x3(const bf16_t*, bf16_t*, int, int, int, int, hipStream_t);
 
 namespace {
 
@@ -532,6 +533,16 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   return {dx, dgamma, dbeta};
 }
 
+torch::Tensor avg_pool3x3(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && is_cl(x) &&
+              x.scalar_type() == torch::kBFloat16,
+              "avg_pool3x3: bf16 channels-last NCHW");
+  auto y = torch::empty_like(x);
+  launch_avg3x3((const bf16_t*)x.data_ptr(), (bf16_t*)y.data_ptr(),
+                x.size(0), x.size(2), x.size(3), x.size(1), cur_stream());
+  return y;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -559,6 +570,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("conv2d_fwd", &conv2d_fwd, "implicit-GEMM conv fwd (MFMA, bf16)");
   m.def("bn_fwd", &bn_fwd, "fused train-mode batch-norm (+relu) fwd");
+  m.def("avg_pool3x3", &avg_pool3x3, "3x3 s1 p1 avg pool, channels-last");
   m.def("bn_bwd", &bn_bwd, "fused batch-norm (+relu mask) bwd");
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
