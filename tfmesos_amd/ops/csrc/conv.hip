@@ -174,17 +174,19 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
       stage_wrows(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs.K, KD, t, true);
     }
     const int kfrag = (lane >> 4) * 8;
+    bf16x8 bfrag[2];
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn)
+      bfrag[fn] =
+          *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
       bf16x8 a =
           *(const bf16x8*)&As[cur][wr * 32 + fm * 16 + (lane & 15)][kfrag];
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        bf16x8 b =
-            *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
-        acc[fm][fn] =
-            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm][fn], 0, 0, 0);
-      }
+      for (int fn = 0; fn < 2; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, bfrag[fn], acc[fm][fn], 0, 0, 0);
     }
     __syncthreads();
   }
@@ -321,17 +323,19 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
       stage_wrows(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs.C, KD, t, kvec);
     }
     const int kfrag = (lane >> 4) * 8;
+    bf16x8 bfrag[2];
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn)
+      bfrag[fn] =
+          *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
       bf16x8 a =
           *(const bf16x8*)&As[cur][wr * 32 + fm * 16 + (lane & 15)][kfrag];
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        bf16x8 b =
-            *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
-        acc[fm][fn] =
-            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm][fn], 0, 0, 0);
-      }
+      for (int fn = 0; fn < 2; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, bfrag[fn], acc[fm][fn], 0, 0, 0);
     }
     __syncthreads();
   }
@@ -462,17 +466,19 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
       stage_x_bwdw(X, Bs[cur ^ 1], cs, tn0, p0 + BK, KD, Ptot, t, cvec);
     }
     const int kfrag = (lane >> 4) * 8;
+    bf16x8 bfrag[2];
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn)
+      bfrag[fn] =
+          *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
       bf16x8 a =
           *(const bf16x8*)&As[cur][wr * 32 + fm * 16 + (lane & 15)][kfrag];
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        bf16x8 b =
-            *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
-        acc[fm][fn] =
-            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm][fn], 0, 0, 0);
-      }
+      for (int fn = 0; fn < 2; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, bfrag[fn], acc[fm][fn], 0, 0, 0);
     }
     __syncthreads();
   }

@@ -129,15 +129,17 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     }
 
     const int kfrag = (lane >> 4) * 8;
+    bf16x8 bfrag[2];
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn)
+      bfrag[fn] = *(const bf16x8*)&Bs[wc * 32 + fn * 16 + (lane & 15)][kfrag];
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
       bf16x8 a = *(const bf16x8*)&As[wr * 32 + fm * 16 + (lane & 15)][kfrag];
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        bf16x8 b = *(const bf16x8*)&Bs[wc * 32 + fn * 16 + (lane & 15)][kfrag];
-        acc[fm][fn] =
-            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm][fn], 0, 0, 0);
-      }
+      for (int fn = 0; fn < 2; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, bfrag[fn], acc[fm][fn], 0, 0, 0);
     }
     __syncthreads();
   }
