@@ -42,10 +42,11 @@ def run_mnist(args, device, rank, world):
     # PS shards colocate with workers: all N GPUs compute replicas and
     # the sharded reduce lands on ranks 0..n_ps-1 ("1-ps/N-worker" on N
     # devices, the BASELINE.json config)
+    grad_dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
     trainer = SyncReplicaTrainer(
         model.init_params(), optimizer=args.optimizer,
         hparams={"lr": args.lr}, device=device, n_ps=args.num_ps,
-        colocate_ps=True)
+        colocate_ps=True, grad_dtype=grad_dtype)
     roles = trainer.roles
 
     act_dtype = torch.bfloat16 if device.type == "cuda" else torch.float32

@@ -494,3 +494,18 @@ def test_avg_pool3x3_fwd_bwd():
         r2 = torch.nn.functional.avg_pool2d(xf, 3, stride=1, padding=1)
         r2.backward(dy.float().cpu())
         assert (x.grad.float().cpu() - xf.grad).abs().max() < 0.02, shape
+
+
+def test_gemm_colsum_epilogue_bf16_out():
+    """bf16 grad buffers (half the xGMI reduce bytes): dW and the fused
+    colsum both land in bf16."""
+    from tfmesos_amd import ops
+    torch.manual_seed(61)
+    x, dy = bf(torch.randn(100, 784)), bf(torch.randn(100, 100))
+    out = torch.zeros(784, 100, device=DEV, dtype=torch.bfloat16)
+    cs = torch.zeros(100, device=DEV, dtype=torch.bfloat16)
+    ops.gemm_bias_act(x, dy, trans_a=True, out=out, colsum_out=cs)
+    ref_w = x.float().t() @ dy.float()
+    ref_b = dy.float().sum(0)
+    assert (out.float() - ref_w).abs().max() < 0.02 * ref_w.abs().max() + 0.3
+    assert (cs.float() - ref_b).abs().max() < 0.02 * ref_b.abs().max() + 0.1

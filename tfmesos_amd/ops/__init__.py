@@ -160,7 +160,7 @@ def gemm_bias_act(a, b, bias=None, act="none", trans_a=False, trans_b=False,
     elif act == "relu_bwd":
         c = c * (aux.float() > 0)
     if colsum_out is not None:
-        colsum_out.copy_(y.sum(0))
+        colsum_out.copy_(y.sum(0).to(colsum_out.dtype))
     if out is not None:
         out.copy_(c.to(out.dtype))
         return out

@@ -15,7 +15,7 @@ void launch_adam(float*, const void*, bool, float*, float*, bf16_t*, long,
 void launch_adagrad(float*, const void*, bool, float*, bf16_t*, long, float,
                     float, float, float, hipStream_t);
 void launch_gemm(const bf16_t*, const bf16_t*, const void*, bool, void*,
-                 bool, const bf16_t*, float*, float*, int*, int, int, int,
+                 bool, const bf16_t*, void*, float*, int*, int, int, int,
                  int, int, int, int, int, bool, bool, int, int, int,
                  hipStream_t);
 void launch_softmax_xent_fwd(const bf16_t*, const long*, bf16_t*, float*,
@@ -203,14 +203,14 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
                 && aux.numel() == (long)M * N, "aux must be bf16 [M,N]");
     aux_p = (const bf16_t*)aux.data_ptr();
   }
-  float* colsum_p = nullptr;
+  void* colsum_p = nullptr;
   if (colsum_out.numel() > 0) {
     TORCH_CHECK(trans_a && !trans_b && bias_p == nullptr && act == 0,
                 "colsum fusion needs tn, no bias, no act");
-    TORCH_CHECK(colsum_out.scalar_type() == torch::kFloat32 &&
-                colsum_out.is_contiguous() && colsum_out.numel() == N,
-                "colsum_out must be fp32 [N]");
-    colsum_p = colsum_out.data_ptr<float>();
+    TORCH_CHECK(colsum_out.is_contiguous() && colsum_out.numel() == N &&
+                colsum_out.scalar_type() == out.scalar_type(),
+                "colsum_out must match out dtype, [N]");
+    colsum_p = colsum_out.data_ptr();
   }
   // split-K when the plain tile grid can't feed the 256-CU chip and K
   // has enough depth to slice

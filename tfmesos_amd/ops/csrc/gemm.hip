@@ -97,7 +97,7 @@ __global__ __launch_bounds__(256)
 void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
                  const void* __restrict__ bias, bool bias_bf16,
                  void* __restrict__ Cout, const __bf16* __restrict__ aux,
-                 float* __restrict__ colsum_out, float* __restrict__ ws,
+                 void* __restrict__ colsum_out, float* __restrict__ ws,
                  int* __restrict__ cnt, int M, int N, int K, int lda, int ldb,
                  int ldc, int kc, int veca, int vecb) {
   __shared__ __align__(16) __bf16 As[BM][BK + APAD];   // [m][k]
@@ -144,8 +144,10 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     __syncthreads();
   }
 
-  if (CS && blockIdx.y == 0 && t < BN && tn0 + t < N)
-    colsum_out[tn0 + t] = cs_acc;
+  if (CS && blockIdx.y == 0 && t < BN && tn0 + t < N) {
+    if (OUTF32) ((float*)colsum_out)[tn0 + t] = cs_acc;
+    else ((__bf16*)colsum_out)[tn0 + t] = (__bf16)cs_acc;
+  }
 
   if (SK) {
     // store this slice's fp32 partial tile into its workspace stripe;
@@ -235,7 +237,7 @@ void splitk_reduce_kernel(const float* __restrict__ ws, const void* __restrict__
 
 void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
                  bool bias_bf16, void* C, bool out_f32, const bf16_t* aux,
-                 float* colsum_out, float* ws, int* cnt, int kc, int nslice,
+                 void* colsum_out, float* ws, int* cnt, int kc, int nslice,
                  int M, int N, int K, int lda, int ldb, int ldc, bool ta,
                  bool tb, int act, int veca, int vecb, hipStream_t stream) {
   const bool sk = nslice > 1;
