@@ -144,11 +144,15 @@ DEVINL void stage_wrows(const __bf16* __restrict__ Wt, __bf16 (*Sn)[BK + APAD],
   }
 }
 
-template <bool BIAS, bool RELU>
+// SK: split the tap reduction over grid.z (small-spatial deep layers
+// otherwise leave most of the 256 CUs idle); slices store fp32 partial
+// stripes, conv_reduce_kernel sums them in fixed order + applies the
+// epilogue — same two-phase pattern as gemm.hip split-K.
+template <bool BIAS, bool RELU, bool SK>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt,
                      const float* __restrict__ bias, __bf16* __restrict__ Y,
-                     ConvShape cs) {
+                     float* __restrict__ ws, int kc, ConvShape cs) {
   // double-buffered LDS: stage tile i+1 while MFMA consumes tile i —
   // one barrier per K-iteration, global-load latency overlapped
   __shared__ __align__(16) __bf16 As[2][BM][BK + APAD];
@@ -163,13 +167,16 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
   const int wave = t >> 6;
   const int wr = wave >> 1, wc = wave & 1;
 
+  const int ks = SK ? blockIdx.z * kc : 0;
+  const int ke = SK ? min(ks + kc, KD) : KD;
+
   f32x4 acc[2][2] = {};
-  stage_patch_fwd(X, As[0], cs, tm0, 0, M, KD, t, cvec);
-  stage_wrows(Wt, Bs[0], tn0, 0, cs.K, KD, t, true);
+  stage_patch_fwd(X, As[0], cs, tm0, ks, M, KD, t, cvec);
+  stage_wrows(Wt, Bs[0], tn0, ks, cs.K, KD, t, true);
   __syncthreads();
   int cur = 0;
-  for (int k0 = 0; k0 < KD; k0 += BK, cur ^= 1) {
-    if (k0 + BK < KD) {
+  for (int k0 = ks; k0 < ke; k0 += BK, cur ^= 1) {
+    if (k0 + BK < ke) {
       stage_patch_fwd(X, As[cur ^ 1], cs, tm0, k0 + BK, M, KD, t, cvec);
       stage_wrows(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs.K, KD, t, true);
     }
@@ -191,6 +198,23 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
     __syncthreads();
   }
 
+  if (SK) {
+    // store this slice's fp32 partial stripe
+    float* wsl = ws + (long)blockIdx.z * M * cs.K;
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+        const int k = tn0 + wc * 32 + fn * 16 + (lane & 15);
+        if (k >= cs.K) continue;
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+          const long pm = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
+          if (pm < M) wsl[pm * cs.K + k] = acc[fm][fn][rr];
+        }
+      }
+    return;
+  }
   // epilogue: Y[n,ho,wo,k] — k contiguous across lanes (coalesced)
 #pragma unroll
   for (int fm = 0; fm < 2; ++fm)
@@ -208,6 +232,36 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
         Y[pm * cs.K + k] = (__bf16)v;
       }
     }
+}
+
+// split-KD phase 2: y[i] = act(sum_z ws[z][i] + bias[i % K])
+template <bool BIAS, bool RELU>
+__global__ __launch_bounds__(256)
+void conv_reduce_kernel(const float* __restrict__ ws,
+                        const float* __restrict__ bias,
+                        __bf16* __restrict__ Y, long mk, int K, int z) {
+  const long i0 = ((long)blockIdx.x * 256 + threadIdx.x) * 4;
+  if (i0 >= mk) return;
+  f32x4 v = {};
+  for (int s = 0; s < z; ++s) {
+    const float* p = ws + (long)s * mk + i0;
+    if (i0 + 4 <= mk) {
+      const f32x4 sv = *(const f32x4*)p;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] += sv[j];
+    } else {
+      for (int j = 0; i0 + j < mk; ++j) v[j] += p[j];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const long i = i0 + j;
+    if (i >= mk) break;
+    float x = v[j];
+    if (BIAS) x += bias[i % K];
+    if (RELU) x = x > 0.f ? x : 0.f;
+    Y[i] = (__bf16)x;
+  }
 }
 
 // --------------------------------------------------------------- bwd-data
@@ -503,35 +557,56 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
 
 }  // namespace
 
+int conv_fwd_slices(int N, int K, int Ho, int Wo, int C, int R, int S) {
+  // binding uses this to size the split workspace; must match launch
+  const long M = (long)N * Ho * Wo;
+  const long tiles = ((M + BM - 1) / BM) * ceil_div(K, BN);
+  const int KD = R * S * C;
+  if (tiles >= 256 || KD < 1024) return 1;
+  long zwant = 512 / tiles;
+  long zmax = KD / (2 * BK);
+  long z = zwant < zmax ? zwant : zmax;
+  if (z > 16) z = 16;
+  if (z < 1) z = 1;
+  return (int)z;
+}
+
 void launch_conv_fwd(const bf16_t* X, const bf16_t* W, const float* bias,
-                     bf16_t* Y, int N, int C, int H, int Wd, int K, int R,
-                     int S, int Ho, int Wo, int U, int V, int P, int Q,
-                     bool relu, hipStream_t stream) {
+                     bf16_t* Y, float* ws, int N, int C, int H, int Wd,
+                     int K, int R, int S, int Ho, int Wo, int U, int V,
+                     int P, int Q, bool relu, hipStream_t stream) {
   ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q,
                make_fdiv(C), make_fdiv(S), make_fdiv(K), make_fdiv(Wo),
                make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd)};
   const long M = (long)N * Ho * Wo;
-  dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(K, BN));
+  const int KD = R * S * C;
+  const int z = conv_fwd_slices(N, K, Ho, Wo, C, R, S);
   dim3 block(256);
-  if (bias) {
-    if (relu)
-      hipLaunchKernelGGL((conv_fwd_kernel<true, true>), grid, block, 0,
-                         stream, (const __bf16*)X, (const __bf16*)W, bias,
-                         (__bf16*)Y, cs);
-    else
-      hipLaunchKernelGGL((conv_fwd_kernel<true, false>), grid, block, 0,
-                         stream, (const __bf16*)X, (const __bf16*)W, bias,
-                         (__bf16*)Y, cs);
-  } else {
-    if (relu)
-      hipLaunchKernelGGL((conv_fwd_kernel<false, true>), grid, block, 0,
-                         stream, (const __bf16*)X, (const __bf16*)W, bias,
-                         (__bf16*)Y, cs);
-    else
-      hipLaunchKernelGGL((conv_fwd_kernel<false, false>), grid, block, 0,
-                         stream, (const __bf16*)X, (const __bf16*)W, bias,
-                         (__bf16*)Y, cs);
+  if (z > 1 && ws) {
+    int kc = ceil_div(ceil_div(KD, z), BK) * BK;
+    const int zr = ceil_div(KD, kc);
+    dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(K, BN), zr);
+    hipLaunchKernelGGL((conv_fwd_kernel<false, false, true>), grid, block,
+                       0, stream, (const __bf16*)X, (const __bf16*)W, bias,
+                       (__bf16*)Y, ws, kc, cs);
+    const long mk = M * K;
+    dim3 rgrid((unsigned)((mk / 4 + 255) / 256)), rblock(256);
+#define RL(BIASv, RELUv)                                                    \
+    hipLaunchKernelGGL((conv_reduce_kernel<BIASv, RELUv>), rgrid, rblock,   \
+                       0, stream, ws, bias, (__bf16*)Y, mk, K, zr)
+    if (bias) { if (relu) RL(true, true); else RL(true, false); }
+    else      { if (relu) RL(false, true); else RL(false, false); }
+#undef RL
+    return;
   }
+  dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(K, BN));
+#define FL(BIASv, RELUv)                                                    \
+  hipLaunchKernelGGL((conv_fwd_kernel<BIASv, RELUv, false>), grid, block,   \
+                     0, stream, (const __bf16*)X, (const __bf16*)W, bias,   \
+                     (__bf16*)Y, nullptr, 0, cs)
+  if (bias) { if (relu) FL(true, true); else FL(true, false); }
+  else      { if (relu) FL(false, true); else FL(false, false); }
+#undef FL
 }
 
 void launch_conv_bwd_data(const bf16_t* dY, const bf16_t* Wt, bf16_t* dX,

@@ -39,8 +39,9 @@ void launch_relu_bwd(const bf16_t*, const bf16_t*, bf16_t*, long,
                      hipStream_t);
 void launch_colsum(const bf16_t*, float*, int, int, hipStream_t);
 void launch_conv_fwd(const bf16_t*, const bf16_t*, const float*, bf16_t*,
-                     int, int, int, int, int, int, int, int, int, int, int,
-                     int, int, bool, hipStream_t);
+                     float*, int, int, int, int, int, int, int, int, int,
+                     int, int, int, int, bool, hipStream_t);
+int conv_fwd_slices(int, int, int, int, int, int, int);
 void launch_conv_bwd_data(const bf16_t*, const bf16_t*, bf16_t*, int, int,
                           int, int, int, int, int, int, int, int, int, int,
                           int, hipStream_t);
@@ -428,8 +429,15 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor wm,
   auto y = torch::empty({x.size(0), K, Ho, Wo},
                         x.options().memory_format(
                             torch::MemoryFormat::ChannelsLast));
+  float* ws = nullptr;
+  const int z = conv_fwd_slices(x.size(0), K, Ho, Wo, x.size(1), R, S);
+  if (z > 1) {
+    int* cnt_unused;
+    ws = splitk_ws(x.device(), (long)x.size(0) * Ho * Wo * K * z, 1,
+                   &cnt_unused);
+  }
   launch_conv_fwd((const bf16_t*)x.data_ptr(), (const bf16_t*)wm.data_ptr(),
-                  bias_p, (bf16_t*)y.data_ptr(), x.size(0), x.size(1),
+                  bias_p, (bf16_t*)y.data_ptr(), ws, x.size(0), x.size(1),
                   x.size(2), x.size(3), K, R, S, Ho, Wo, stride_h, stride_w,
                   pad_h, pad_w, relu, cur_stream());
   return y;
