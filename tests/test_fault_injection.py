@@ -55,13 +55,16 @@ def test_initializer_finalizer_hooks(tmp_path):
     assert ini.exists() and out.exists() and fin.exists()
 
 
-@pytest.mark.timeout(120)
+@pytest.mark.timeout(180)
 def test_killed_task_process_reported(tmp_path):
-    """Killing the task's process group mid-run surfaces as a fatal
-    non-FINISHED status (the 'slave lost' analogue on one node)."""
+    """Killing the task's process mid-run surfaces as a fatal
+    non-FINISHED status (the 'slave lost' analogue on one node).
+    The sleeper's stdio is detached so the orphan can't hold the
+    agent's log pipe open after the kill."""
     pidfile = tmp_path / "pid"
     jobs = [Job(name="worker", num=1,
-                cmd="echo $$ > %s; sleep 60" % pidfile)]
+                cmd="sleep 60 >/dev/null 2>&1 & echo $! > %s; wait"
+                    % pidfile)]
     s = LocalScheduler(jobs, quiet=True)
     s.start()
     deadline = time.time() + 20
