@@ -63,7 +63,7 @@ def test_killed_task_process_reported(tmp_path):
     agent's log pipe open after the kill."""
     pidfile = tmp_path / "pid"
     jobs = [Job(name="worker", num=1,
-                cmd="sleep 60 >/dev/null 2>&1 & echo $! > %s; wait"
+                cmd="sleep 60 >/dev/null 2>&1 & echo $! > %s; wait $!"
                     % pidfile)]
     s = LocalScheduler(jobs, quiet=True)
     s.start()
