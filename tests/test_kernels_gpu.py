@@ -509,3 +509,22 @@ def test_gemm_colsum_epilogue_bf16_out():
     ref_b = dy.float().sum(0)
     assert (out.float() - ref_w).abs().max() < 0.02 * ref_w.abs().max() + 0.3
     assert (cs.float() - ref_b).abs().max() < 0.02 * ref_b.abs().max() + 0.1
+
+
+def test_maxpool3x3s2_fwd_bwd():
+    from tfmesos_amd import ops
+    torch.manual_seed(62)
+    for shape in [(2, 16, 15, 15), (1, 7, 9, 11)]:
+        x = bf(torch.randn(*shape)).requires_grad_(True)
+        y = ops.max_pool3x3s2(x)
+        ref = torch.nn.functional.max_pool2d(
+            x.detach().float().cpu(), 3, stride=2)
+        assert torch.equal(y.float().cpu(), ref.to(torch.bfloat16).float()
+                           .to(torch.float32)) or \
+            (y.float().cpu() - ref).abs().max() < 1e-6, shape
+        dy = bf(torch.randn(*y.shape))
+        y.backward(dy)
+        xf = x.detach().float().cpu().requires_grad_(True)
+        r2 = torch.nn.functional.max_pool2d(xf, 3, stride=2)
+        r2.backward(dy.float().cpu())
+        assert (x.grad.float().cpu() - xf.grad).abs().max() < 1e-5, shape

@@ -73,6 +73,8 @@ def _avg_pool(x, k, stride=1, padding=1):
 
 
 def _max_pool(x, k, stride):
+    if k == 3 and stride == 2 and x.is_cuda:
+        return ops.max_pool3x3s2(x)   # hand-written (csrc/pool.hip)
     return torch.nn.functional.max_pool2d(x, k, stride=stride)
 
 

@@ -509,3 +509,39 @@ class _AvgPool3x3Fn(torch.autograd.Function):
 def avg_pool3x3(x):
     """Differentiable 3x3/s1/p1 average pool (count_include_pad)."""
     return _AvgPool3x3Fn.apply(x)
+
+
+class _MaxPool3x3s2Fn(torch.autograd.Function):
+    """3x3 stride-2 max pool (the Inception reduction pool) — forward
+    saves the winning tap; backward gathers deterministically."""
+
+    @staticmethod
+    def forward(ctx, x):
+        if x.is_cuda:
+            x = x.contiguous(memory_format=torch.channels_last)
+            y, idx = _ext().maxpool3x3s2_fwd(x)
+            ctx.save_for_backward(idx)
+            ctx.hw = (x.shape[2], x.shape[3])
+            ctx.gpu = True
+            return y
+        ctx.gpu = False
+        y, idx = torch.nn.functional.max_pool2d(
+            x.float(), 3, stride=2, return_indices=True)
+        ctx.save_for_backward(idx)
+        ctx.xshape = x.shape
+        return y.to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        if ctx.gpu:
+            dy = dy.contiguous(memory_format=torch.channels_last)
+            return _ext().maxpool3x3s2_bwd(dy, idx, *ctx.hw)
+        return torch.nn.functional.max_unpool2d(
+            dy.float(), idx, 3, stride=2,
+            output_size=ctx.xshape[2:]).to(dy.dtype)
+
+
+def max_pool3x3s2(x):
+    """Differentiable 3x3/s2 max pool (no padding)."""
+    return _MaxPool3x3s2Fn.apply(x)

@@ -58,6 +58,10 @@ void launch_bn_bwd(const bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*,
 int bn_stats_slices(long, int);
 int bn_max_channels();
 void launch_avg3x3(const bf16_t*, bf16_t*, int, int, int, int, hipStream_t);
+void launch_maxpool3x3s2_fwd(const bf16_t*, bf16_t*, unsigned char*, int,
+                             int, int, int, int, int, hipStream_t);
+void launch_maxpool3x3s2_bwd(const bf16_t*, const unsigned char*, bf16_t*,
+                             int, int, int, int, int, int, hipStream_t);
 
 namespace {
 
@@ -551,6 +555,38 @@ torch::Tensor avg_pool3x3(torch::Tensor x) {
   return y;
 }
 
+std::vector<torch::Tensor> maxpool3x3s2_fwd(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && is_cl(x) &&
+              x.scalar_type() == torch::kBFloat16,
+              "maxpool3x3s2: bf16 channels-last NCHW");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int Ho = (H - 3) / 2 + 1, Wo = (W - 3) / 2 + 1;
+  auto y = torch::empty({N, C, Ho, Wo},
+                        x.options().memory_format(
+                            torch::MemoryFormat::ChannelsLast));
+  auto idx = torch::empty({N, C, Ho, Wo},
+                          x.options().dtype(torch::kUInt8).memory_format(
+                              torch::MemoryFormat::ChannelsLast));
+  launch_maxpool3x3s2_fwd((const bf16_t*)x.data_ptr(), (bf16_t*)y.data_ptr(),
+                          idx.data_ptr<unsigned char>(), N, H, W, C, Ho, Wo,
+                          cur_stream());
+  return {y, idx};
+}
+
+torch::Tensor maxpool3x3s2_bwd(torch::Tensor dy, torch::Tensor idx,
+                               long H, long W) {
+  TORCH_CHECK(dy.is_cuda() && is_cl(dy) && is_cl(idx));
+  const int N = dy.size(0), C = dy.size(1);
+  auto dx = torch::empty({N, C, H, W},
+                         dy.options().memory_format(
+                             torch::MemoryFormat::ChannelsLast));
+  launch_maxpool3x3s2_bwd((const bf16_t*)dy.data_ptr(),
+                          idx.data_ptr<unsigned char>(),
+                          (bf16_t*)dx.data_ptr(), N, H, W, C, dy.size(2),
+                          dy.size(3), cur_stream());
+  return dx;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -579,6 +615,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &conv2d_fwd, "implicit-GEMM conv fwd (MFMA, bf16)");
   m.def("bn_fwd", &bn_fwd, "fused train-mode batch-norm (+relu) fwd");
   m.def("avg_pool3x3", &avg_pool3x3, "3x3 s1 p1 avg pool, channels-last");
+  m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd);
+  m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
   m.def("bn_bwd", &bn_bwd, "fused batch-norm (+relu mask) bwd");
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);

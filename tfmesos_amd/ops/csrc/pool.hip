@@ -85,3 +85,117 @@ void launch_avg3x3(const bf16_t* X, bf16_t* Y, int N, int H, int W, int C,
     hipLaunchKernelGGL((avg3x3_kernel<false>), grid, block, 0, stream,
                        (const __bf16*)X, (__bf16*)Y, N, H, W, C);
 }
+
+namespace {
+
+// 3x3 stride-2 max pool (no padding -> every window is fully in
+// bounds), channels-last. Forward saves the winning tap (0..8, u8);
+// backward GATHERS: each input pixel sums the dy of the <=2x2 windows
+// whose saved tap points at it — deterministic, no atomics.
+template <bool VEC>
+__global__ __launch_bounds__(256)
+void maxpool3x3s2_fwd_kernel(const __bf16* __restrict__ X,
+                             __bf16* __restrict__ Y,
+                             unsigned char* __restrict__ idx,
+                             int N, int H, int W, int C, int Ho, int Wo) {
+  const int nho = blockIdx.x;
+  const int n = nho / Ho;
+  const int ho = nho - n * Ho;
+  const __bf16* base = X + ((long)n * H + ho * 2) * W * C;
+  __bf16* out = Y + ((long)n * Ho + ho) * Wo * C;
+  unsigned char* oi = idx + ((long)n * Ho + ho) * Wo * C;
+  const int WoC = Wo * C;
+  const int step = VEC ? 8 : 1;
+  for (int i = threadIdx.x * step; i < WoC; i += 256 * step) {
+    const int wo = i / C;
+    const int c = i - wo * C;
+    const long col0 = (long)wo * 2 * C + c;
+    if (VEC) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float best = -3.4e38f;
+        int bi = 0;
+#pragma unroll
+        for (int r = 0; r < 3; ++r)
+#pragma unroll
+          for (int s = 0; s < 3; ++s) {
+            const float v =
+                (float)base[(long)r * W * C + col0 + (long)s * C + j];
+            if (v > best) { best = v; bi = r * 3 + s; }
+          }
+        out[i + j] = (__bf16)best;
+        oi[i + j] = (unsigned char)bi;
+      }
+    } else {
+      float best = -3.4e38f;
+      int bi = 0;
+#pragma unroll
+      for (int r = 0; r < 3; ++r)
+#pragma unroll
+        for (int s = 0; s < 3; ++s) {
+          const float v = (float)base[(long)r * W * C + col0 + (long)s * C];
+          if (v > best) { best = v; bi = r * 3 + s; }
+        }
+      out[i] = (__bf16)best;
+      oi[i] = (unsigned char)bi;
+    }
+  }
+}
+
+__global__ __launch_bounds__(256)
+void maxpool3x3s2_bwd_kernel(const __bf16* __restrict__ dY,
+                             const unsigned char* __restrict__ idx,
+                             __bf16* __restrict__ dX,
+                             int N, int H, int W, int C, int Ho, int Wo) {
+  const int nh = blockIdx.x;
+  const int n = nh / H;
+  const int h = nh - n * H;
+  const int WC = W * C;
+  __bf16* out = dX + ((long)n * H + h) * WC;
+  for (int i = threadIdx.x; i < WC; i += 256) {
+    const int w = i / C;
+    const int c = i - w * C;
+    float acc = 0.f;
+    // windows (ho, wo) covering (h, w): ho*2 <= h <= ho*2+2
+    const int ho_lo = (h - 2 + 1) / 2 < 0 ? 0 : (h - 2 + 1) / 2;
+    const int ho_hi = h / 2 < Ho - 1 ? h / 2 : Ho - 1;
+    const int wo_lo = (w - 2 + 1) / 2 < 0 ? 0 : (w - 2 + 1) / 2;
+    const int wo_hi = w / 2 < Wo - 1 ? w / 2 : Wo - 1;
+    for (int ho = ho_lo; ho <= ho_hi; ++ho) {
+      const int r = h - ho * 2;
+      if (r < 0 || r > 2) continue;
+      for (int wo = wo_lo; wo <= wo_hi; ++wo) {
+        const int s = w - wo * 2;
+        if (s < 0 || s > 2) continue;
+        const long oidx = (((long)n * Ho + ho) * Wo + wo) * C + c;
+        if (idx[oidx] == (unsigned char)(r * 3 + s))
+          acc += (float)dY[oidx];
+      }
+    }
+    out[i] = (__bf16)acc;
+  }
+}
+
+}  // namespace
+
+void launch_maxpool3x3s2_fwd(const bf16_t* X, bf16_t* Y, unsigned char* idx,
+                             int N, int H, int W, int C, int Ho, int Wo,
+                             hipStream_t stream) {
+  dim3 grid((unsigned)(N * Ho)), block(256);
+  if ((C & 7) == 0)
+    hipLaunchKernelGGL((maxpool3x3s2_fwd_kernel<true>), grid, block, 0,
+                       stream, (const __bf16*)X, (__bf16*)Y, idx, N, H, W, C,
+                       Ho, Wo);
+  else
+    hipLaunchKernelGGL((maxpool3x3s2_fwd_kernel<false>), grid, block, 0,
+                       stream, (const __bf16*)X, (__bf16*)Y, idx, N, H, W, C,
+                       Ho, Wo);
+}
+
+void launch_maxpool3x3s2_bwd(const bf16_t* dY, const unsigned char* idx,
+                             bf16_t* dX, int N, int H, int W, int C, int Ho,
+                             int Wo, hipStream_t stream) {
+  dim3 grid((unsigned)(N * H)), block(256);
+  hipLaunchKernelGGL(maxpool3x3s2_bwd_kernel, grid, block, 0, stream,
+                     (const __bf16*)dY, idx, (__bf16*)dX, N, H, W, C, Ho, Wo);
+}
