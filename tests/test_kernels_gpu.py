@@ -527,4 +527,5 @@ def test_maxpool3x3s2_fwd_bwd():
         xf = x.detach().float().cpu().requires_grad_(True)
         r2 = torch.nn.functional.max_pool2d(xf, 3, stride=2)
         r2.backward(dy.float().cpu())
-        assert (x.grad.float().cpu() - xf.grad).abs().max() < 1e-5, shape
+        dxe = (x.grad.float().cpu() - xf.grad).abs().max()
+        assert dxe < 0.01 * xf.grad.abs().max() + 0.01, (shape, dxe)
