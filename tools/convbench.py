@@ -37,6 +37,10 @@ def t_ms(fn, iters=200, warmup=20):
 
 def main():
     dev = "cuda:0"
+    only = os.environ.get("CONV_ONLY")
+    global SHAPES
+    if only is not None:
+        SHAPES = [SHAPES[int(only)]]
     print("%-24s %9s %9s %9s | TFLOP/s fwd bwdd bwdw" %
           ("shape", "fwd_us", "bwdd_us", "bwdw_us"))
     for name, N, C, H, W, K, R, S, st, pd in SHAPES:
