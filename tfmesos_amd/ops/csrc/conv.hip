@@ -28,6 +28,19 @@ namespace {
 
 constexpr int BM = 64, BN = 64, BK = 32;
 constexpr int APAD = 8;
+// LDS tile rows are stride 40 bf16 (80 B = 20 dwords); banks repeat
+// every 8 rows, so the transposing stagers' 8-rows-per-instruction
+// writes were 8-way bank-conflicted (PMC: 80% of LDS cycles were
+// conflicts in bwd-weight). A 16 B skew per 8-row group keeps b128
+// fragment reads aligned and spreads the groups across banks.
+constexpr int TILE_ELEMS = BM * (BK + APAD) + (BM / 8) * 8;
+
+DEVINL __bf16* srow(__bf16* S, int r) {
+  return S + r * (BK + APAD) + ((r >> 3) << 3);
+}
+DEVINL const __bf16* srow(const __bf16* S, int r) {
+  return S + r * (BK + APAD) + ((r >> 3) << 3);
+}
 
 // Magic-number unsigned division (Granlund-Montgomery): integer divide
 // on CDNA is ~30 VALU cycles and the gathers decode several indices per
@@ -77,7 +90,7 @@ DEVINL void put8(__bf16* dstrow, const __bf16* src, bool vec) {
 
 // As[m][kk]: m = output pixel, kk = tap (r,s,c) with c innermost.
 // Thread t: pixel row m = t>>2, 8-tap chunk kk0 = (t&3)*8 -> one b128.
-DEVINL void stage_patch_fwd(const __bf16* __restrict__ X, __bf16 (*Sm)[BK + APAD],
+DEVINL void stage_patch_fwd(const __bf16* __restrict__ X, __bf16* Sm,
                             const ConvShape cs, long m0, int k0, long M,
                             int KD, int t, bool cvec) {
   const int mx = t >> 2;
@@ -86,7 +99,7 @@ DEVINL void stage_patch_fwd(const __bf16* __restrict__ X, __bf16 (*Sm)[BK + APAD
   const long pm = m0 + mx;
   if (q0 >= KD || pm >= M) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) Sm[mx][kk0 + j] = (__bf16)0.f;
+    for (int j = 0; j < 8; ++j) srow(Sm, mx)[kk0 + j] = (__bf16)0.f;
     return;
   }
   const int rs = fdiv(q0, cs.dC);      // tap (r,s) block (c0 % 8 == 0 when
@@ -104,10 +117,10 @@ DEVINL void stage_patch_fwd(const __bf16* __restrict__ X, __bf16 (*Sm)[BK + APAD
   if (cvec && c0 + 8 <= cs.C) {
     if (hi >= 0 && hi < cs.H && wi >= 0 && wi < cs.W) {
       const __bf16* src = X + (((long)n * cs.H + hi) * cs.W + wi) * cs.C + c0;
-      put8(&Sm[mx][kk0], src, true);
+      put8(&srow(Sm, mx)[kk0], src, true);
     } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) Sm[mx][kk0 + j] = (__bf16)0.f;
+      for (int j = 0; j < 8; ++j) srow(Sm, mx)[kk0 + j] = (__bf16)0.f;
     }
     return;
   }
@@ -122,13 +135,13 @@ DEVINL void stage_patch_fwd(const __bf16* __restrict__ X, __bf16 (*Sm)[BK + APAD
       if (hij >= 0 && hij < cs.H && wij >= 0 && wij < cs.W)
         v = (float)X[(((long)n * cs.H + hij) * cs.W + wij) * cs.C + c];
     }
-    Sm[mx][kk0 + j] = (__bf16)v;
+    srow(Sm, mx)[kk0 + j] = (__bf16)v;
     if (++c == cs.C) { c = 0; if (++ss == cs.S) { ss = 0; ++rr; } }
   }
 }
 
 // Bs[k][kk]: weight memory [K][R*S*C] rows contiguous in tap order.
-DEVINL void stage_wrows(const __bf16* __restrict__ Wt, __bf16 (*Sn)[BK + APAD],
+DEVINL void stage_wrows(const __bf16* __restrict__ Wt, __bf16* Sn,
                         int n0, int k0, int NROWS, int KD, int t, bool vec) {
   const int x = t >> 2;
   const int kk0 = (t & 3) * 8;
@@ -136,11 +149,11 @@ DEVINL void stage_wrows(const __bf16* __restrict__ Wt, __bf16 (*Sn)[BK + APAD],
   const int gk = k0 + kk0;
   const __bf16* src = Wt + (long)gx * KD + gk;
   if (gx < NROWS && gk + 8 <= KD) {
-    put8(&Sn[x][kk0], src, vec && ((gk & 7) == 0) && ((KD & 7) == 0));
+    put8(&srow(Sn, x)[kk0], src, vec && ((gk & 7) == 0) && ((KD & 7) == 0));
   } else {
 #pragma unroll
     for (int j = 0; j < 8; ++j)
-      Sn[x][kk0 + j] = (gx < NROWS && gk + j < KD) ? src[j] : (__bf16)0.f;
+      srow(Sn, x)[kk0 + j] = (gx < NROWS && gk + j < KD) ? src[j] : (__bf16)0.f;
   }
 }
 
@@ -155,8 +168,8 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
                      float* __restrict__ ws, int kc, ConvShape cs) {
   // double-buffered LDS: stage tile i+1 while MFMA consumes tile i —
   // one barrier per K-iteration, global-load latency overlapped
-  __shared__ __align__(16) __bf16 As[2][BM][BK + APAD];
-  __shared__ __align__(16) __bf16 Bs[2][BN][BK + APAD];
+  __shared__ __align__(16) __bf16 As[2][TILE_ELEMS];
+  __shared__ __align__(16) __bf16 Bs[2][TILE_ELEMS];
   const long M = (long)cs.N * cs.Ho * cs.Wo;
   const int KD = cs.R * cs.S * cs.C;
   const bool cvec = (cs.C & 7) == 0;
@@ -185,11 +198,11 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
 #pragma unroll
     for (int fn = 0; fn < 2; ++fn)
       bfrag[fn] =
-          *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
+          *(const bf16x8*)&srow(Bs[cur], wc * 32 + fn * 16 + (lane & 15))[kfrag];
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
       bf16x8 a =
-          *(const bf16x8*)&As[cur][wr * 32 + fm * 16 + (lane & 15)][kfrag];
+          *(const bf16x8*)&srow(As[cur], wr * 32 + fm * 16 + (lane & 15))[kfrag];
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -272,7 +285,7 @@ void conv_reduce_kernel(const float* __restrict__ ws,
 // bit ops (runtime %/ by the stride costs ~30 VALU cycles each).
 template <int STRIDE>
 DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
-                             __bf16 (*Sm)[BK + APAD], const ConvShape cs,
+                             __bf16* Sm, const ConvShape cs,
                              long m0, int k0, long M, int KD, int t,
                              bool kvec) {
   const int mx = t >> 2;
@@ -281,7 +294,7 @@ DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
   const long pm = m0 + mx;
   if (q0 >= KD || pm >= M) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) Sm[mx][kk0 + j] = (__bf16)0.f;
+    for (int j = 0; j < 8; ++j) srow(Sm, mx)[kk0 + j] = (__bf16)0.f;
     return;
   }
   const int rs = fdiv(q0, cs.dK);
@@ -313,10 +326,10 @@ DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
     if (ok) {
       const __bf16* src =
           dY + (((long)n * cs.Ho + ho) * cs.Wo + wo) * cs.K + kc0;
-      put8(&Sm[mx][kk0], src, true);
+      put8(&srow(Sm, mx)[kk0], src, true);
     } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) Sm[mx][kk0 + j] = (__bf16)0.f;
+      for (int j = 0; j < 8; ++j) srow(Sm, mx)[kk0 + j] = (__bf16)0.f;
     }
     return;
   }
@@ -343,7 +356,7 @@ DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
       if (ok)
         v = (float)dY[(((long)n * cs.Ho + ho) * cs.Wo + wo) * cs.K + kc];
     }
-    Sm[mx][kk0 + j] = (__bf16)v;
+    srow(Sm, mx)[kk0 + j] = (__bf16)v;
     if (++kc == cs.K) { kc = 0; if (++ss2 == cs.S) { ss2 = 0; ++rr2; } }
   }
 }
@@ -353,8 +366,8 @@ __global__ __launch_bounds__(256)
 void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ Wt,
                       __bf16* __restrict__ dX, ConvShape cs) {
   // Wt memory: [C][R*S*K] (host-permuted W^T)
-  __shared__ __align__(16) __bf16 As[2][BM][BK + APAD];
-  __shared__ __align__(16) __bf16 Bs[2][BN][BK + APAD];
+  __shared__ __align__(16) __bf16 As[2][TILE_ELEMS];
+  __shared__ __align__(16) __bf16 Bs[2][TILE_ELEMS];
   const long M = (long)cs.N * cs.H * cs.W;
   const int KD = cs.R * cs.S * cs.K;
   const bool kvec = (cs.K & 7) == 0;
@@ -381,11 +394,11 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
 #pragma unroll
     for (int fn = 0; fn < 2; ++fn)
       bfrag[fn] =
-          *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
+          *(const bf16x8*)&srow(Bs[cur], wc * 32 + fn * 16 + (lane & 15))[kfrag];
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
       bf16x8 a =
-          *(const bf16x8*)&As[cur][wr * 32 + fm * 16 + (lane & 15)][kfrag];
+          *(const bf16x8*)&srow(As[cur], wr * 32 + fm * 16 + (lane & 15))[kfrag];
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -413,7 +426,7 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
 
 // As[k][p]: dY[n,ho,wo,k] — thread t loads 8 consecutive k (one b128)
 // for one reduction pixel p, transposing into LDS.
-DEVINL void stage_dy_bwdw(const __bf16* __restrict__ dY, __bf16 (*Sm)[BK + APAD],
+DEVINL void stage_dy_bwdw(const __bf16* __restrict__ dY, __bf16* Sm,
                           const ConvShape cs, int m0, long p0, long Ptot,
                           int t, bool kvec) {
   const int px = t >> 3;           // 0..31 reduction pixel
@@ -422,24 +435,24 @@ DEVINL void stage_dy_bwdw(const __bf16* __restrict__ dY, __bf16 (*Sm)[BK + APAD]
   const int k0 = m0 + kk0;
   if (p >= Ptot || k0 >= cs.K) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) Sm[kk0 + j][px] = (__bf16)0.f;
+    for (int j = 0; j < 8; ++j) srow(Sm, kk0 + j)[px] = (__bf16)0.f;
     return;
   }
   const __bf16* src = dY + p * cs.K + k0;
   if (kvec && k0 + 8 <= cs.K) {
     bf16x8 v = *(const bf16x8*)src;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) Sm[kk0 + j][px] = v[j];
+    for (int j = 0; j < 8; ++j) srow(Sm, kk0 + j)[px] = v[j];
   } else {
 #pragma unroll
     for (int j = 0; j < 8; ++j)
-      Sm[kk0 + j][px] = (k0 + j < cs.K) ? src[j] : (__bf16)0.f;
+      srow(Sm, kk0 + j)[px] = (k0 + j < cs.K) ? src[j] : (__bf16)0.f;
   }
 }
 
 // Bs[tap(r,s,c)][p]: X[n, ho*U+r-P, wo*V+s-Q, c] — thread t loads 8
 // consecutive c (one b128) for one pixel, transposing into LDS.
-DEVINL void stage_x_bwdw(const __bf16* __restrict__ X, __bf16 (*Sn)[BK + APAD],
+DEVINL void stage_x_bwdw(const __bf16* __restrict__ X, __bf16* Sn,
                          const ConvShape cs, int n0, long p0, int KD,
                          long Ptot, int t, bool cvec) {
   const int px = t >> 3;
@@ -448,7 +461,7 @@ DEVINL void stage_x_bwdw(const __bf16* __restrict__ X, __bf16 (*Sn)[BK + APAD],
   const int q0 = n0 + qq0;
   if (p >= Ptot || q0 >= KD) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) Sn[qq0 + j][px] = (__bf16)0.f;
+    for (int j = 0; j < 8; ++j) srow(Sn, qq0 + j)[px] = (__bf16)0.f;
     return;
   }
   const int rs = fdiv(q0, cs.dC);
@@ -467,10 +480,10 @@ DEVINL void stage_x_bwdw(const __bf16* __restrict__ X, __bf16 (*Sn)[BK + APAD],
       const __bf16* src = X + (((long)n * cs.H + hi) * cs.W + wi) * cs.C + c0;
       bf16x8 v = *(const bf16x8*)src;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) Sn[qq0 + j][px] = v[j];
+      for (int j = 0; j < 8; ++j) srow(Sn, qq0 + j)[px] = v[j];
     } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) Sn[qq0 + j][px] = (__bf16)0.f;
+      for (int j = 0; j < 8; ++j) srow(Sn, qq0 + j)[px] = (__bf16)0.f;
     }
     return;
   }
@@ -484,7 +497,7 @@ DEVINL void stage_x_bwdw(const __bf16* __restrict__ X, __bf16 (*Sn)[BK + APAD],
       if (hij >= 0 && hij < cs.H && wij >= 0 && wij < cs.W)
         v = (float)X[(((long)n * cs.H + hij) * cs.W + wij) * cs.C + c];
     }
-    Sn[qq0 + j][px] = (__bf16)v;
+    srow(Sn, qq0 + j)[px] = (__bf16)v;
     if (++c == cs.C) { c = 0; if (++ss2 == cs.S) { ss2 = 0; ++rr2; } }
   }
 }
@@ -494,8 +507,8 @@ DEVINL void stage_x_bwdw(const __bf16* __restrict__ X, __bf16 (*Sn)[BK + APAD],
 __global__ __launch_bounds__(256)
 void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ X,
                       float* __restrict__ dW, ConvShape cs, long pc) {
-  __shared__ __align__(16) __bf16 As[2][BM][BK + APAD];
-  __shared__ __align__(16) __bf16 Bs[2][BN][BK + APAD];
+  __shared__ __align__(16) __bf16 As[2][TILE_ELEMS];
+  __shared__ __align__(16) __bf16 Bs[2][TILE_ELEMS];
   const int KD = cs.R * cs.S * cs.C;
   const long Ptot = (long)cs.N * cs.Ho * cs.Wo;
   const bool cvec = (cs.C & 7) == 0;
@@ -524,11 +537,11 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
 #pragma unroll
     for (int fn = 0; fn < 2; ++fn)
       bfrag[fn] =
-          *(const bf16x8*)&Bs[cur][wc * 32 + fn * 16 + (lane & 15)][kfrag];
+          *(const bf16x8*)&srow(Bs[cur], wc * 32 + fn * 16 + (lane & 15))[kfrag];
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
       bf16x8 a =
-          *(const bf16x8*)&As[cur][wr * 32 + fm * 16 + (lane & 15)][kfrag];
+          *(const bf16x8*)&srow(As[cur], wr * 32 + fm * 16 + (lane & 15))[kfrag];
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
