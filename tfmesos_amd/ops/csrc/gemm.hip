@@ -33,12 +33,24 @@ namespace {
 
 constexpr int BM = 64, BN = 64, BK = 32;
 constexpr int APAD = 8;  // +16B: keeps b128 fragment reads aligned
+// 16B skew per 8-row group: the row stride (40 bf16 = 20 dwords) makes
+// banks repeat every 8 rows, so the transposing stagers' writes were
+// 8-way bank-conflicted (measured via PMC on the conv twins of these
+// stagers; same layout here).
+constexpr int TILE_ELEMS = BM * (BK + APAD) + (BM / 8) * 8;
+
+DEVINL __bf16* srow(__bf16* S, int r) {
+  return S + r * (BK + APAD) + ((r >> 3) << 3);
+}
+DEVINL const __bf16* srow(const __bf16* S, int r) {
+  return S + r * (BK + APAD) + ((r >> 3) << 3);
+}
 
 typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
 
 // Stage a [X,K]-stored operand tile (row x contiguous in k) into S[x][k].
 // Used for A when !TA and B when TB. 256 threads x 8 elems = 64x32 tile.
-DEVINL void stage_xk(const __bf16* __restrict__ P, __bf16 (*S)[BK + APAD],
+DEVINL void stage_xk(const __bf16* __restrict__ P, __bf16* S,
                      int x0, int k0, int X, int K, int ld, int t, int vec) {
   const int x = t >> 2;
   const int kk0 = (t & 3) * 8;
@@ -47,26 +59,26 @@ DEVINL void stage_xk(const __bf16* __restrict__ P, __bf16 (*S)[BK + APAD],
   const __bf16* src = P + (long)gx * ld + gk;
   if (gx < X && gk + 8 <= K) {
     if (vec == 8) {
-      *(bf16x8*)&S[x][kk0] = *(const bf16x8*)src;
+      *(bf16x8*)&srow(S, x)[kk0] = *(const bf16x8*)src;
     } else if (vec == 2) {
 #pragma unroll
       for (int j = 0; j < 8; j += 2)
-        *(bf16x2*)&S[x][kk0 + j] = *(const bf16x2*)(src + j);
+        *(bf16x2*)&srow(S, x)[kk0 + j] = *(const bf16x2*)(src + j);
     } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) S[x][kk0 + j] = src[j];
+      for (int j = 0; j < 8; ++j) srow(S, x)[kk0 + j] = src[j];
     }
   } else {
 #pragma unroll
     for (int j = 0; j < 8; ++j)
-      S[x][kk0 + j] = (gx < X && gk + j < K) ? src[j] : (__bf16)0.f;
+      srow(S, x)[kk0 + j] = (gx < X && gk + j < K) ? src[j] : (__bf16)0.f;
   }
 }
 
 // Stage a [K,X]-stored operand tile (row k contiguous in x) into S[x][k].
 // Used for A when TA and B when !TB. Global access stays contiguous
 // (8 consecutive x per thread); the transpose happens on the LDS write.
-DEVINL void stage_kx(const __bf16* __restrict__ P, __bf16 (*S)[BK + APAD],
+DEVINL void stage_kx(const __bf16* __restrict__ P, __bf16* S,
                      int x0, int k0, int X, int K, int ld, int t, int vec) {
   const int k = t >> 3;          // 0..31 == BK
   const int xx0 = (t & 7) * 8;   // 0..56
@@ -77,15 +89,15 @@ DEVINL void stage_kx(const __bf16* __restrict__ P, __bf16 (*S)[BK + APAD],
     if (vec == 8) {
       bf16x8 v = *(const bf16x8*)src;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) S[xx0 + j][k] = v[j];
+      for (int j = 0; j < 8; ++j) srow(S, xx0 + j)[k] = v[j];
     } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) S[xx0 + j][k] = src[j];
+      for (int j = 0; j < 8; ++j) srow(S, xx0 + j)[k] = src[j];
     }
   } else {
 #pragma unroll
     for (int j = 0; j < 8; ++j)
-      S[xx0 + j][k] = (gk < K && gx + j < X) ? src[j] : (__bf16)0.f;
+      srow(S, xx0 + j)[k] = (gk < K && gx + j < X) ? src[j] : (__bf16)0.f;
   }
 }
 
@@ -100,8 +112,8 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
                  void* __restrict__ colsum_out, float* __restrict__ ws,
                  int* __restrict__ cnt, int M, int N, int K, int lda, int ldb,
                  int ldc, int kc, int veca, int vecb) {
-  __shared__ __align__(16) __bf16 As[BM][BK + APAD];   // [m][k]
-  __shared__ __align__(16) __bf16 Bs[BN][BK + APAD];   // [n][k] (B^T tile)
+  __shared__ __align__(16) __bf16 As[TILE_ELEMS];   // [m][k], skewed
+  __shared__ __align__(16) __bf16 Bs[TILE_ELEMS];   // [n][k] (B^T), skewed
 
   const int tm0 = blockIdx.y * BM;
   const int tn0 = blockIdx.x * BN;
@@ -125,17 +137,19 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
 
     if (CS && blockIdx.y == 0 && t < BN) {
 #pragma unroll
-      for (int kk = 0; kk < BK; ++kk) cs_acc += (float)Bs[t][kk];
+      for (int kk = 0; kk < BK; ++kk) cs_acc += (float)srow(Bs, t)[kk];
     }
 
     const int kfrag = (lane >> 4) * 8;
     bf16x8 bfrag[2];
 #pragma unroll
     for (int fn = 0; fn < 2; ++fn)
-      bfrag[fn] = *(const bf16x8*)&Bs[wc * 32 + fn * 16 + (lane & 15)][kfrag];
+      bfrag[fn] =
+          *(const bf16x8*)&srow(Bs, wc * 32 + fn * 16 + (lane & 15))[kfrag];
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
-      bf16x8 a = *(const bf16x8*)&As[wr * 32 + fm * 16 + (lane & 15)][kfrag];
+      bf16x8 a =
+          *(const bf16x8*)&srow(As, wr * 32 + fm * 16 + (lane & 15))[kfrag];
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
