@@ -361,10 +361,11 @@ DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
   }
 }
 
-template <int STRIDE>
+template <int STRIDE, bool SK>
 __global__ __launch_bounds__(256)
 void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ Wt,
-                      __bf16* __restrict__ dX, ConvShape cs) {
+                      __bf16* __restrict__ dX, float* __restrict__ ws, int kc,
+                      ConvShape cs) {
   // Wt memory: [C][R*S*K] (host-permuted W^T)
   __shared__ __align__(16) __bf16 As[2][TILE_ELEMS];
   __shared__ __align__(16) __bf16 Bs[2][TILE_ELEMS];
@@ -378,13 +379,16 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
   const int wave = t >> 6;
   const int wr = wave >> 1, wc = wave & 1;
 
+  const int ks = SK ? blockIdx.z * kc : 0;
+  const int ke = SK ? min(ks + kc, KD) : KD;
+
   f32x4 acc[2][2] = {};
-  stage_patch_bwdd<STRIDE>(dY, As[0], cs, tm0, 0, M, KD, t, kvec);
-  stage_wrows(Wt, Bs[0], tn0, 0, cs.C, KD, t, kvec);
+  stage_patch_bwdd<STRIDE>(dY, As[0], cs, tm0, ks, M, KD, t, kvec);
+  stage_wrows(Wt, Bs[0], tn0, ks, cs.C, KD, t, kvec);
   __syncthreads();
   int cur = 0;
-  for (int k0 = 0; k0 < KD; k0 += BK, cur ^= 1) {
-    if (k0 + BK < KD) {
+  for (int k0 = ks; k0 < ke; k0 += BK, cur ^= 1) {
+    if (k0 + BK < ke) {
       stage_patch_bwdd<STRIDE>(dY, As[cur ^ 1], cs, tm0, k0 + BK, M, KD, t,
                                kvec);
       stage_wrows(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs.C, KD, t, kvec);
@@ -407,6 +411,22 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
     __syncthreads();
   }
 
+  if (SK) {
+    float* wsl = ws + (long)blockIdx.z * M * cs.C;
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+        const int c = tn0 + wc * 32 + fn * 16 + (lane & 15);
+        if (c >= cs.C) continue;
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+          const long pm = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
+          if (pm < M) wsl[pm * cs.C + c] = acc[fm][fn][rr];
+        }
+      }
+    return;
+  }
 #pragma unroll
   for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
@@ -575,8 +595,21 @@ int conv_fwd_slices(int N, int K, int Ho, int Wo, int C, int R, int S) {
   const long M = (long)N * Ho * Wo;
   const long tiles = ((M + BM - 1) / BM) * ceil_div(K, BN);
   const int KD = R * S * C;
-  if (tiles >= 256 || KD < 1024) return 1;
-  long zwant = 512 / tiles;
+  if (tiles >= 512 || KD < 512) return 1;
+  long zwant = (1024 + tiles - 1) / tiles;
+  long zmax = KD / (2 * BK);
+  long z = zwant < zmax ? zwant : zmax;
+  if (z > 16) z = 16;
+  if (z < 1) z = 1;
+  return (int)z;
+}
+
+int conv_bwdd_slices(int N, int H, int W, int C, int K, int R, int S) {
+  const long M = (long)N * H * W;
+  const long tiles = ((M + BM - 1) / BM) * ceil_div(C, BN);
+  const int KD = R * S * K;
+  if (tiles >= 512 || KD < 512) return 1;
+  long zwant = (1024 + tiles - 1) / tiles;
   long zmax = KD / (2 * BK);
   long z = zwant < zmax ? zwant : zmax;
   if (z > 16) z = 16;
@@ -623,25 +656,44 @@ void launch_conv_fwd(const bf16_t* X, const bf16_t* W, const float* bias,
 }
 
 void launch_conv_bwd_data(const bf16_t* dY, const bf16_t* Wt, bf16_t* dX,
-                          int N, int C, int H, int Wd, int K, int R, int S,
-                          int Ho, int Wo, int U, int V, int P, int Q,
-                          hipStream_t stream) {
+                          float* ws, int N, int C, int H, int Wd, int K,
+                          int R, int S, int Ho, int Wo, int U, int V, int P,
+                          int Q, hipStream_t stream) {
   // Wt: host-permuted W^T, memory [C][R*S*K]
   ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q,
                make_fdiv(C), make_fdiv(S), make_fdiv(K), make_fdiv(Wo),
                make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd)};
   const long M = (long)N * H * Wd;
-  dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(C, BN));
+  const int KD = R * S * K;
+  const int z = conv_bwdd_slices(N, H, Wd, C, K, R, S);
   dim3 block(256);
-  if (U == 1 && V == 1)
-    hipLaunchKernelGGL((conv_bwdd_kernel<1>), grid, block, 0, stream,
-                       (const __bf16*)dY, (const __bf16*)Wt, (__bf16*)dX, cs);
-  else if (U == 2 && V == 2)
-    hipLaunchKernelGGL((conv_bwdd_kernel<2>), grid, block, 0, stream,
-                       (const __bf16*)dY, (const __bf16*)Wt, (__bf16*)dX, cs);
-  else
-    hipLaunchKernelGGL((conv_bwdd_kernel<0>), grid, block, 0, stream,
-                       (const __bf16*)dY, (const __bf16*)Wt, (__bf16*)dX, cs);
+  if (z > 1 && ws) {
+    int kc = ceil_div(ceil_div(KD, z), BK) * BK;
+    const int zr = ceil_div(KD, kc);
+    dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(C, BN), zr);
+#define BL(STRIDEv)                                                         \
+    hipLaunchKernelGGL((conv_bwdd_kernel<STRIDEv, true>), grid, block, 0,   \
+                       stream, (const __bf16*)dY, (const __bf16*)Wt,        \
+                       (__bf16*)dX, ws, kc, cs)
+    if (U == 1 && V == 1) BL(1);
+    else if (U == 2 && V == 2) BL(2);
+    else BL(0);
+#undef BL
+    const long mk = M * C;
+    dim3 rgrid((unsigned)((mk / 4 + 255) / 256)), rblock(256);
+    hipLaunchKernelGGL((conv_reduce_kernel<false, false>), rgrid, rblock, 0,
+                       stream, ws, nullptr, (__bf16*)dX, mk, C, zr);
+    return;
+  }
+  dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(C, BN));
+#define BL(STRIDEv)                                                         \
+  hipLaunchKernelGGL((conv_bwdd_kernel<STRIDEv, false>), grid, block, 0,    \
+                     stream, (const __bf16*)dY, (const __bf16*)Wt,          \
+                     (__bf16*)dX, nullptr, 0, cs)
+  if (U == 1 && V == 1) BL(1);
+  else if (U == 2 && V == 2) BL(2);
+  else BL(0);
+#undef BL
 }
 
 void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,

@@ -42,9 +42,10 @@ void launch_conv_fwd(const bf16_t*, const bf16_t*, const float*, bf16_t*,
                      float*, int, int, int, int, int, int, int, int, int,
                      int, int, int, int, bool, hipStream_t);
 int conv_fwd_slices(int, int, int, int, int, int, int);
-void launch_conv_bwd_data(const bf16_t*, const bf16_t*, bf16_t*, int, int,
+void launch_conv_bwd_data(const bf16_t*, const bf16_t*, bf16_t*, float*,
                           int, int, int, int, int, int, int, int, int, int,
-                          int, hipStream_t);
+                          int, int, int, hipStream_t);
+int conv_bwdd_slices(int, int, int, int, int, int, int);
 void launch_conv_bwd_weight(const bf16_t*, const bf16_t*, float*, int, int,
                             int, int, int, int, int, int, int, int, int,
                             int, int, hipStream_t);
@@ -461,9 +462,16 @@ torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor wt,
   auto dx = torch::empty({dy.size(0), C, H, W},
                          dy.options().memory_format(
                              torch::MemoryFormat::ChannelsLast));
+  float* ws = nullptr;
+  const int z = conv_bwdd_slices(dy.size(0), H, W, C, dy.size(1), R, S);
+  if (z > 1) {
+    int* cnt_unused;
+    ws = splitk_ws(dy.device(), (long)dy.size(0) * H * W * C * z, 1,
+                   &cnt_unused);
+  }
   launch_conv_bwd_data((const bf16_t*)dy.data_ptr(),
                        (const bf16_t*)wt.data_ptr(), (bf16_t*)dx.data_ptr(),
-                       dy.size(0), C, H, W, dy.size(1), R, S,
+                       ws, dy.size(0), C, H, W, dy.size(1), R, S,
                        dy.size(2), dy.size(3), stride_h, stride_w,
                        pad_h, pad_w, cur_stream());
   return dx;
