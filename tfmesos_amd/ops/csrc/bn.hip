@@ -31,7 +31,19 @@ void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict__ part,
   const int z = blockIdx.y;
   float sum = 0.f, sq = 0.f;
   if (c < C) {
-    for (long p = (long)z * 4 + pr; p < P; p += (long)Z * 4) {
+    // 4 rows in flight per thread: these are independent strided loads
+    // and the kernel is latency-bound without the explicit MLP
+    const long step = (long)Z * 4;
+    long p = (long)z * 4 + pr;
+    for (; p + 3 * step < P; p += 4 * step) {
+      const float v0 = (float)x[p * C + c];
+      const float v1 = (float)x[(p + step) * C + c];
+      const float v2 = (float)x[(p + 2 * step) * C + c];
+      const float v3 = (float)x[(p + 3 * step) * C + c];
+      sum += v0 + v1 + v2 + v3;
+      sq += v0 * v0 + v1 * v1 + v2 * v2 + v3 * v3;
+    }
+    for (; p < P; p += step) {
       const float v = (float)x[p * C + c];
       sum += v;
       sq += v * v;
@@ -140,7 +152,19 @@ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
     const float mu = mean[c], is = invstd[c];
     const float sc = RELU ? (float)g[c] * is : 0.f;
     const float sh = RELU ? (float)b[c] - mu * sc : 0.f;
-    for (long p = (long)z * 4 + pr; p < P; p += (long)Z * 4) {
+    const long step = (long)Z * 4;
+    long p = (long)z * 4 + pr;
+    for (; p + step < P; p += 2 * step) {
+      const long i0 = p * C + c, i1 = (p + step) * C + c;
+      const float x0 = (float)x[i0], x1 = (float)x[i1];
+      float d0 = (float)dy[i0], d1 = (float)dy[i1];
+      const float h0 = (x0 - mu) * is, h1 = (x1 - mu) * is;
+      if (RELU && (float)(__bf16)(x0 * sc + sh) <= 0.f) d0 = 0.f;
+      if (RELU && (float)(__bf16)(x1 * sc + sh) <= 0.f) d1 = 0.f;
+      s1 += d0 + d1;
+      s2 += d0 * h0 + d1 * h1;
+    }
+    for (; p < P; p += step) {
       const long i = p * C + c;
       const float xv = (float)x[i];
       float d = (float)dy[i];
