@@ -248,11 +248,13 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
 }
 
 // split-KD phase 2: y[i] = act(sum_z ws[z][i] + bias[i % K])
-template <bool BIAS, bool RELU>
+template <bool BIAS, bool RELU, bool OUTF32 = false>
 __global__ __launch_bounds__(256)
 void conv_reduce_kernel(const float* __restrict__ ws,
                         const float* __restrict__ bias,
-                        __bf16* __restrict__ Y, long mk, int K, int z) {
+                        void* __restrict__ Yv, long mk, int K, int z) {
+  __bf16* Y = (__bf16*)Yv;
+  float* Yf = (float*)Yv;
   const long i0 = ((long)blockIdx.x * 256 + threadIdx.x) * 4;
   if (i0 >= mk) return;
   f32x4 v = {};
@@ -273,7 +275,8 @@ void conv_reduce_kernel(const float* __restrict__ ws,
     float x = v[j];
     if (BIAS) x += bias[i % K];
     if (RELU) x = x > 0.f ? x : 0.f;
-    Y[i] = (__bf16)x;
+    if (OUTF32) Yf[i] = x;
+    else Y[i] = (__bf16)x;
   }
 }
 
@@ -522,11 +525,13 @@ DEVINL void stage_x_bwdw(const __bf16* __restrict__ X, __bf16* Sn,
   }
 }
 
-// dW fp32 out [K][R*S*C]; grid.z slices the huge N*Ho*Wo reduction and
-// accumulates with fp32 atomics (dW is zeroed by the launcher).
+// dW fp32 out [K][R*S*C]; grid.z slices the huge N*Ho*Wo reduction.
+// Slices store fp32 partial stripes (no atomics, no zero-fill), summed
+// in fixed order by conv_reduce_kernel — deterministic.
 __global__ __launch_bounds__(256)
 void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ X,
-                      float* __restrict__ dW, ConvShape cs, long pc) {
+                      float* __restrict__ dW, float* __restrict__ ws,
+                      ConvShape cs, long pc) {
   __shared__ __align__(16) __bf16 As[2][TILE_ELEMS];
   __shared__ __align__(16) __bf16 Bs[2][TILE_ELEMS];
   const int KD = cs.R * cs.S * cs.C;
@@ -570,6 +575,8 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
     __syncthreads();
   }
 
+  float* out = gridDim.z == 1 ? dW
+      : ws + (long)blockIdx.z * cs.K * KD;
 #pragma unroll
   for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
@@ -580,10 +587,7 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
       for (int rr = 0; rr < 4; ++rr) {
         const int k = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
         if (k >= cs.K) continue;
-        if (gridDim.z == 1)
-          dW[(long)k * KD + q] = acc[fm][fn][rr];
-        else
-          unsafeAtomicAdd(&dW[(long)k * KD + q], acc[fm][fn][rr]);
+        out[(long)k * KD + q] = acc[fm][fn][rr];
       }
     }
 }
@@ -696,26 +700,42 @@ void launch_conv_bwd_data(const bf16_t* dY, const bf16_t* Wt, bf16_t* dX,
 #undef BL
 }
 
+int conv_bwdw_slices(int N, int C, int K, int R, int S, int Ho, int Wo) {
+  const int KD = C * R * S;
+  const long Ptot = (long)N * Ho * Wo;
+  const long tiles = (long)ceil_div(K, BM) * ceil_div(KD, BN);
+  long zmax = (Ptot + BK - 1) / BK;
+  long zwant = 1024 / tiles;
+  if (zwant < 1) zwant = 1;
+  if (zwant > 32) zwant = 32;
+  int z = (int)(zmax < zwant ? zmax : zwant);
+  long pc = (Ptot + z - 1) / z;
+  pc = (pc + BK - 1) / BK * BK;
+  return (int)((Ptot + pc - 1) / pc);
+}
+
 void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
-                            int N, int C, int H, int Wd, int K, int R, int S,
-                            int Ho, int Wo, int U, int V, int P, int Q,
-                            hipStream_t stream) {
+                            float* ws, int N, int C, int H, int Wd, int K,
+                            int R, int S, int Ho, int Wo, int U, int V,
+                            int P, int Q, hipStream_t stream) {
   ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q,
                make_fdiv(C), make_fdiv(S), make_fdiv(K), make_fdiv(Wo),
                make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd)};
   const int KD = C * R * S;
   const long Ptot = (long)N * Ho * Wo;
-  // slice the reduction so the grid can fill the chip (>=1024 WGs)
-  const long tiles = (long)ceil_div(K, BM) * ceil_div(KD, BN);
-  long zmax = (Ptot + BK - 1) / BK;
-  long zwant = 1024 / tiles;
-  if (zwant < 1) zwant = 1;
-  int z = (int)(zmax < zwant ? zmax : zwant);
+  int z = conv_bwdw_slices(N, C, K, R, S, Ho, Wo);
+  if (ws == nullptr) z = 1;
   long pc = (Ptot + z - 1) / z;
   pc = (pc + BK - 1) / BK * BK;
   z = (int)((Ptot + pc - 1) / pc);
   dim3 grid(ceil_div(KD, BN), ceil_div(K, BM), z);
   dim3 block(256);
   hipLaunchKernelGGL(conv_bwdw_kernel, grid, block, 0, stream,
-                     (const __bf16*)dY, (const __bf16*)X, dW, cs, pc);
+                     (const __bf16*)dY, (const __bf16*)X, dW, ws, cs, pc);
+  if (z > 1) {
+    const long mk = (long)K * KD;
+    dim3 rgrid((unsigned)((mk / 4 + 255) / 256)), rblock(256);
+    hipLaunchKernelGGL((conv_reduce_kernel<false, false, true>), rgrid,
+                       rblock, 0, stream, ws, nullptr, dW, mk, KD, z);
+  }
 }

@@ -142,6 +142,9 @@ void maxpool3x3s2_fwd_kernel(const __bf16* __restrict__ X,
   }
 }
 
+typedef __attribute__((ext_vector_type(8))) unsigned char u8x8;
+
+template <bool VEC>
 __global__ __launch_bounds__(256)
 void maxpool3x3s2_bwd_kernel(const __bf16* __restrict__ dY,
                              const unsigned char* __restrict__ idx,
@@ -152,27 +155,56 @@ void maxpool3x3s2_bwd_kernel(const __bf16* __restrict__ dY,
   const int h = nh - n * H;
   const int WC = W * C;
   __bf16* out = dX + ((long)n * H + h) * WC;
-  for (int i = threadIdx.x; i < WC; i += 256) {
+  const int step = VEC ? 8 : 1;
+  for (int i = threadIdx.x * step; i < WC; i += 256 * step) {
     const int w = i / C;
     const int c = i - w * C;
-    float acc = 0.f;
     // windows (ho, wo) covering (h, w): ho*2 <= h <= ho*2+2
     const int ho_lo = (h - 2 + 1) / 2 < 0 ? 0 : (h - 2 + 1) / 2;
     const int ho_hi = h / 2 < Ho - 1 ? h / 2 : Ho - 1;
     const int wo_lo = (w - 2 + 1) / 2 < 0 ? 0 : (w - 2 + 1) / 2;
     const int wo_hi = w / 2 < Wo - 1 ? w / 2 : Wo - 1;
-    for (int ho = ho_lo; ho <= ho_hi; ++ho) {
-      const int r = h - ho * 2;
-      if (r < 0 || r > 2) continue;
-      for (int wo = wo_lo; wo <= wo_hi; ++wo) {
-        const int s = w - wo * 2;
-        if (s < 0 || s > 2) continue;
-        const long oidx = (((long)n * Ho + ho) * Wo + wo) * C + c;
-        if (idx[oidx] == (unsigned char)(r * 3 + s))
-          acc += (float)dY[oidx];
+    if (VEC) {
+      f32x4 a0 = {}, a1 = {};
+      for (int ho = ho_lo; ho <= ho_hi; ++ho) {
+        const int r = h - ho * 2;
+        if (r < 0 || r > 2) continue;
+        for (int wo = wo_lo; wo <= wo_hi; ++wo) {
+          const int s = w - wo * 2;
+          if (s < 0 || s > 2) continue;
+          const long oidx = (((long)n * Ho + ho) * Wo + wo) * C + c;
+          const u8x8 iv = *(const u8x8*)&idx[oidx];
+          const bf16x8 dv = *(const bf16x8*)&dY[oidx];
+          const unsigned char want = (unsigned char)(r * 3 + s);
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            if (iv[j] == want) a0[j] += (float)dv[j];
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            if (iv[4 + j] == want) a1[j] += (float)dv[4 + j];
+        }
       }
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o[j] = (__bf16)a0[j];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o[4 + j] = (__bf16)a1[j];
+      *(bf16x8*)(out + i) = o;
+    } else {
+      float acc = 0.f;
+      for (int ho = ho_lo; ho <= ho_hi; ++ho) {
+        const int r = h - ho * 2;
+        if (r < 0 || r > 2) continue;
+        for (int wo = wo_lo; wo <= wo_hi; ++wo) {
+          const int s = w - wo * 2;
+          if (s < 0 || s > 2) continue;
+          const long oidx = (((long)n * Ho + ho) * Wo + wo) * C + c;
+          if (idx[oidx] == (unsigned char)(r * 3 + s))
+            acc += (float)dY[oidx];
+        }
+      }
+      out[i] = (__bf16)acc;
     }
-    out[i] = (__bf16)acc;
   }
 }
 
@@ -196,6 +228,12 @@ void launch_maxpool3x3s2_bwd(const bf16_t* dY, const unsigned char* idx,
                              bf16_t* dX, int N, int H, int W, int C, int Ho,
                              int Wo, hipStream_t stream) {
   dim3 grid((unsigned)(N * H)), block(256);
-  hipLaunchKernelGGL(maxpool3x3s2_bwd_kernel, grid, block, 0, stream,
-                     (const __bf16*)dY, idx, (__bf16*)dX, N, H, W, C, Ho, Wo);
+  if ((C & 7) == 0)
+    hipLaunchKernelGGL((maxpool3x3s2_bwd_kernel<true>), grid, block, 0,
+                       stream, (const __bf16*)dY, idx, (__bf16*)dX, N, H, W,
+                       C, Ho, Wo);
+  else
+    hipLaunchKernelGGL((maxpool3x3s2_bwd_kernel<false>), grid, block, 0,
+                       stream, (const __bf16*)dY, idx, (__bf16*)dX, N, H, W,
+                       C, Ho, Wo);
 }
