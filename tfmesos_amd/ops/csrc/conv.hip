@@ -707,7 +707,7 @@ int conv_bwdw_slices(int N, int C, int K, int R, int S, int Ho, int Wo) {
   long zmax = (Ptot + BK - 1) / BK;
   long zwant = 1024 / tiles;
   if (zwant < 1) zwant = 1;
-  if (zwant > 32) zwant = 32;
+  if (zwant > 256) zwant = 256;   // ws stays bounded: z*tiles ~ 1024
   int z = (int)(zmax < zwant ? zmax : zwant);
   long pc = (Ptot + z - 1) / z;
   pc = (pc + BK - 1) / BK * BK;
