@@ -525,9 +525,11 @@ DEVINL void stage_x_bwdw(const __bf16* __restrict__ X, __bf16* Sn,
   }
 }
 
-// dW fp32 out [K][R*S*C]; grid.z slices the huge N*Ho*Wo reduction.
-// Slices store fp32 partial stripes (no atomics, no zero-fill), summed
-// in fixed order by conv_reduce_kernel — deterministic.
+// dW fp32 out [K][R*S*C]; grid.z slices the huge N*Ho*Wo reduction and
+// accumulates with fp32 atomics (dW zero-filled by the binding when
+// sliced). A fixed-order stripe-reduce variant measured SLOWER: dW is
+// small (9-300K elements), so the z-deep serial reduce loop had too
+// little parallelism, while the atomic contention here is negligible.
 __global__ __launch_bounds__(256)
 void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ X,
                       float* __restrict__ dW, float* __restrict__ ws,
@@ -575,8 +577,6 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
     __syncthreads();
   }
 
-  float* out = gridDim.z == 1 ? dW
-      : ws + (long)blockIdx.z * cs.K * KD;
 #pragma unroll
   for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
@@ -587,7 +587,10 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
       for (int rr = 0; rr < 4; ++rr) {
         const int k = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
         if (k >= cs.K) continue;
-        out[(long)k * KD + q] = acc[fm][fn][rr];
+        if (gridDim.z == 1)
+          dW[(long)k * KD + q] = acc[fm][fn][rr];
+        else
+          unsafeAtomicAdd(&dW[(long)k * KD + q], acc[fm][fn][rr]);
       }
     }
 }
@@ -715,16 +718,17 @@ int conv_bwdw_slices(int N, int C, int K, int R, int S, int Ho, int Wo) {
 }
 
 void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
-                            float* ws, int N, int C, int H, int Wd, int K,
+                            int N, int C, int H, int Wd, int K,
                             int R, int S, int Ho, int Wo, int U, int V,
                             int P, int Q, hipStream_t stream) {
+  float* ws = nullptr;  // unused (atomic accumulate)
+  (void)ws;
   ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q,
                make_fdiv(C), make_fdiv(S), make_fdiv(K), make_fdiv(Wo),
                make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd)};
   const int KD = C * R * S;
   const long Ptot = (long)N * Ho * Wo;
   int z = conv_bwdw_slices(N, C, K, R, S, Ho, Wo);
-  if (ws == nullptr) z = 1;
   long pc = (Ptot + z - 1) / z;
   pc = (pc + BK - 1) / BK * BK;
   z = (int)((Ptot + pc - 1) / pc);
@@ -732,10 +736,4 @@ void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
   dim3 block(256);
   hipLaunchKernelGGL(conv_bwdw_kernel, grid, block, 0, stream,
                      (const __bf16*)dY, (const __bf16*)X, dW, ws, cs, pc);
-  if (z > 1) {
-    const long mk = (long)K * KD;
-    dim3 rgrid((unsigned)((mk / 4 + 255) / 256)), rblock(256);
-    hipLaunchKernelGGL((conv_reduce_kernel<false, false, true>), rgrid,
-                       rblock, 0, stream, ws, nullptr, dW, mk, KD, z);
-  }
 }

@@ -46,7 +46,7 @@ void launch_conv_bwd_data(const bf16_t*, const bf16_t*, bf16_t*, float*,
                           int, int, int, int, int, int, int, int, int, int,
                           int, int, int, hipStream_t);
 int conv_bwdd_slices(int, int, int, int, int, int, int);
-void launch_conv_bwd_weight(const bf16_t*, const bf16_t*, float*, float*,
+void launch_conv_bwd_weight(const bf16_t*, const bf16_t*, float*,
                             int, int, int, int, int, int, int, int, int,
                             int, int, int, int, hipStream_t);
 int conv_bwdw_slices(int, int, int, int, int, int, int);
@@ -484,21 +484,15 @@ torch::Tensor conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
                                 long pad_h, long pad_w) {
   TORCH_CHECK(dy.is_cuda() && x.is_cuda() && is_cl(dy) && is_cl(x),
               "dy/x must be channels-last");
-  // z-sliced reductions store per-slice stripes and reduce in fixed
-  // order (deterministic, no atomics, no zero-init)
-  const long KD = x.size(1) * R * S;
+  // zero-init only when the reduction is z-sliced (atomic accumulate)
   auto opts = x.options().dtype(torch::kFloat32);
-  auto dw = torch::empty({dy.size(1), R, S, x.size(1)}, opts);
-  float* ws = nullptr;
   const int z = conv_bwdw_slices(x.size(0), x.size(1), dy.size(1), R, S,
                                  dy.size(2), dy.size(3));
-  if (z > 1) {
-    int* cnt_unused;
-    ws = splitk_ws(x.device(), (long)dy.size(1) * KD * z, 1, &cnt_unused);
-  }
+  auto dw = z > 1 ? torch::zeros({dy.size(1), R, S, x.size(1)}, opts)
+                  : torch::empty({dy.size(1), R, S, x.size(1)}, opts);
   launch_conv_bwd_weight((const bf16_t*)dy.data_ptr(),
                          (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(),
-                         ws, x.size(0), x.size(1), x.size(2), x.size(3),
+                         x.size(0), x.size(1), x.size(2), x.size(3),
                          dy.size(1), R, S, dy.size(2), dy.size(3), stride_h,
                          stride_w, pad_h, pad_w, cur_stream());
   return dw;
