@@ -35,6 +35,8 @@ constexpr int APAD = 8;
 // fragment reads aligned and spreads the groups across banks.
 constexpr int TILE_ELEMS = BM * (BK + APAD) + (BM / 8) * 8;
 
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
 DEVINL __bf16* srow(__bf16* S, int r) {
   return S + r * (BK + APAD) + ((r >> 3) << 3);
 }
@@ -183,7 +185,11 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
   const int ks = SK ? blockIdx.z * kc : 0;
   const int ke = SK ? min(ks + kc, KD) : KD;
 
-  f32x4 acc[2][2] = {};
+  // 32x32x16 MFMA: one 32x32 wave tile, 2 MFMA + 4 fragment loads per
+  // BK=32 iteration (vs 4 MFMA + 6 loads with 16x16x32 fragments).
+  // A/B: lane l holds row/col (l&31), k = (l>>5)*8+j.
+  // D: reg v -> row 8*(v>>2) + 4*(l>>5) + (v&3), col (l&31).
+  f32x16 acc = {};
   stage_patch_fwd(X, As[0], cs, tm0, ks, M, KD, t, cvec);
   stage_wrows(Wt, Bs[0], tn0, ks, cs.K, KD, t, true);
   __syncthreads();
@@ -193,58 +199,43 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
       stage_patch_fwd(X, As[cur ^ 1], cs, tm0, k0 + BK, M, KD, t, cvec);
       stage_wrows(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs.K, KD, t, true);
     }
-    const int kfrag = (lane >> 4) * 8;
-    bf16x8 bfrag[2];
 #pragma unroll
-    for (int fn = 0; fn < 2; ++fn)
-      bfrag[fn] =
-          *(const bf16x8*)&srow(Bs[cur], wc * 32 + fn * 16 + (lane & 15))[kfrag];
-#pragma unroll
-    for (int fm = 0; fm < 2; ++fm) {
-      bf16x8 a =
-          *(const bf16x8*)&srow(As[cur], wr * 32 + fm * 16 + (lane & 15))[kfrag];
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
-        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a, bfrag[fn], acc[fm][fn], 0, 0, 0);
+    for (int kh = 0; kh < 2; ++kh) {
+      const int ko = kh * 16 + ((lane >> 5) << 3);
+      bf16x8 a = *(const bf16x8*)&srow(As[cur], wr * 32 + (lane & 31))[ko];
+      bf16x8 b = *(const bf16x8*)&srow(Bs[cur], wc * 32 + (lane & 31))[ko];
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
     }
     __syncthreads();
   }
 
+  const int k = tn0 + wc * 32 + (lane & 31);
   if (SK) {
     // store this slice's fp32 partial stripe
-    float* wsl = ws + (long)blockIdx.z * M * cs.K;
+    if (k < cs.K) {
+      float* wsl = ws + (long)blockIdx.z * M * cs.K;
 #pragma unroll
-    for (int fm = 0; fm < 2; ++fm)
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        const int k = tn0 + wc * 32 + fn * 16 + (lane & 15);
-        if (k >= cs.K) continue;
-#pragma unroll
-        for (int rr = 0; rr < 4; ++rr) {
-          const long pm = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
-          if (pm < M) wsl[pm * cs.K + k] = acc[fm][fn][rr];
-        }
+      for (int v = 0; v < 16; ++v) {
+        const long pm = tm0 + wr * 32 + ((v >> 2) << 3) +
+                        ((lane >> 5) << 2) + (v & 3);
+        if (pm < M) wsl[pm * cs.K + k] = acc[v];
       }
+    }
     return;
   }
   // epilogue: Y[n,ho,wo,k] — k contiguous across lanes (coalesced)
+  if (k < cs.K) {
+    const float bv = BIAS ? bias[k] : 0.f;
 #pragma unroll
-  for (int fm = 0; fm < 2; ++fm)
-#pragma unroll
-    for (int fn = 0; fn < 2; ++fn) {
-      const int k = tn0 + wc * 32 + fn * 16 + (lane & 15);
-      if (k >= cs.K) continue;
-      const float bv = BIAS ? bias[k] : 0.f;
-#pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const long pm = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
-        if (pm >= M) continue;
-        float v = acc[fm][fn][rr] + bv;
-        if (RELU) v = v > 0.f ? v : 0.f;
-        Y[pm * cs.K + k] = (__bf16)v;
-      }
+    for (int v = 0; v < 16; ++v) {
+      const long pm = tm0 + wr * 32 + ((v >> 2) << 3) +
+                      ((lane >> 5) << 2) + (v & 3);
+      if (pm >= M) continue;
+      float val = acc[v] + bv;
+      if (RELU) val = val > 0.f ? val : 0.f;
+      Y[pm * cs.K + k] = (__bf16)val;
     }
+  }
 }
 
 // split-KD phase 2: y[i] = act(sum_z ws[z][i] + bias[i % K])
