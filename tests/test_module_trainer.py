@@ -86,3 +86,43 @@ def test_inception_v3_cpu_forward():
     y = m(x)
     assert y.shape == (1, 10)
     assert torch.isfinite(y).all()
+
+
+@pytest.mark.timeout(240)
+def test_module_trainer_world2_matches_single_process(tmp_path):
+    """2-rank gloo run of the nn.Module adapter (bf16 grads reduced,
+    colocated PS) vs the single-process run: identical batches => the
+    mean grad equals the single grad => identical masters."""
+    import os
+    import subprocess
+    import sys
+
+    from tfmesos_amd.utils import free_port
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    repo = os.path.dirname(here)
+    out2 = str(tmp_path / "w2.pt")
+    port = free_port()
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ, RANK=str(rank), WORLD_SIZE="2",
+                   MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                   PYTHONPATH=repo)
+        procs.append(subprocess.Popen(
+            [sys.executable, os.path.join(here, "_module_trainer_proc.py"),
+             "4", out2], env=env))
+    for p in procs:
+        assert p.wait(timeout=180) == 0
+
+    # single-process reference with the same seeds
+    env = dict(os.environ, PYTHONPATH=repo)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    out1 = str(tmp_path / "w1.pt")
+    r = subprocess.run(
+        [sys.executable, os.path.join(here, "_module_trainer_proc.py"),
+         "4", out1], env=env, timeout=180)
+    assert r.returncode == 0
+    got2 = torch.load(out2, weights_only=True)
+    got1 = torch.load(out1, weights_only=True)
+    assert torch.allclose(got1, got2, atol=1e-5)
