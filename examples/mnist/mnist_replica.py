@@ -24,7 +24,6 @@ or standalone under torchrun (one rank per GPU).
 import argparse
 import os
 import sys
-import time
 
 REPO = os.path.dirname(os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))))
@@ -35,6 +34,7 @@ import torch  # noqa: E402
 from tfmesos_amd.models.mlp import MnistMLP, synthetic_batch  # noqa: E402
 from tfmesos_amd.ps.replica import (  # noqa: E402
     AsyncPSServer, AsyncPSWorker, SyncReplicaTrainer, make_pair_groups)
+from tfmesos_amd.utils.metrics import StepTimer  # noqa: E402
 
 
 def parse_args(argv):
@@ -98,13 +98,15 @@ def main(argv=None):
           % (job, trainer.rank, trainer.world, roles.describe(), device,
              "sync" if args.sync_replicas else "async"))
 
-    t0 = time.time()
+    timer = StepTimer(report_every=max(20, args.train_steps // 10),
+                      prefix="[%s:%d] " % (job, roles.worker_index)).start()
     if args.sync_replicas or trainer.world == 1:
         for step in range(args.train_steps):
             loss = None
             if roles.is_worker:
                 loss = model.fwd_bwd(pview, x, y, trainer.grad_view)
             trainer.step()
+            timer.step()
             if roles.is_worker and step % 20 == 0:
                 print("step %d loss %.4f" % (step, float(loss)))
     else:
@@ -116,13 +118,14 @@ def main(argv=None):
             for step in range(args.train_steps):
                 loss = model.fwd_bwd(pview, x, y, trainer.grad_view)
                 w.step()
+                timer.step()
                 if step % 20 == 0:
                     print("step %d loss %.4f" % (step, float(loss)))
-    elapsed = time.time() - t0
     if device != "cpu":
         torch.cuda.synchronize()
+    s = timer.summary()
     print("training done in %.3fs (%.1f steps/s)"
-          % (elapsed, args.train_steps / max(elapsed, 1e-9)))
+          % (s["elapsed_s"], s["steps_per_sec"]))
 
     # validation xent on the chief (reference prints it at the end,
     # mnist_replica.py:213-226)

@@ -21,9 +21,13 @@ def main():
     mode, steps, prefix = sys.argv[1], int(sys.argv[2]), sys.argv[3]
     n_ps = int(sys.argv[4]) if len(sys.argv) > 4 else 1
     variant = sys.argv[5] if len(sys.argv) > 5 else ""
+    optimizer = "adam" if mode == "sync-adam" else "sgd"
+    lr = 0.01 if optimizer == "adam" else 0.1
+    if mode == "sync-adam":
+        mode = "sync"
     model = MnistMLP()
-    trainer = SyncReplicaTrainer(model.init_params(), optimizer="sgd",
-                                 hparams={"lr": 0.1}, device="cpu",
+    trainer = SyncReplicaTrainer(model.init_params(), optimizer=optimizer,
+                                 hparams={"lr": lr}, device="cpu",
                                  n_ps=n_ps, colocate_ps=variant == "colocate",
                                  mode="allreduce" if variant == "allreduce"
                                  else "ps")

@@ -138,3 +138,33 @@ def test_sync_allreduce_mode_matches_single_process(tmp_path):
     want = _single_process_reference(5)
     for n in want:
         assert torch.allclose(got[n], want[n], atol=1e-5), n
+
+
+@pytest.mark.timeout(240)
+def test_sync_adam_world2_matches_single_process(tmp_path):
+    """Adam optimizer state lives on the PS; the 2-rank run must match
+    single-process exactly (same seed/batch)."""
+    import torch as T
+
+    from tfmesos_amd.models.mlp import MnistMLP, synthetic_batch
+    from tfmesos_amd.ps.store import PStore
+
+    prefix = str(tmp_path / "wadam")
+    _spawn_world(2, "sync-adam", 4, prefix)
+    got = T.load(prefix + ".pt", weights_only=True)
+
+    model = MnistMLP()
+    store = PStore()
+    store.init_params(model.init_params(), optimizer="adam", lr=0.01)
+    fg = T.zeros_like(store.flat)
+
+    def gv(n):
+        s, c = store.offsets[n]
+        return fg[s:s + c].view(store.shapes[n])
+
+    x, y = synthetic_batch(50, seed=42)
+    for _ in range(4):
+        model.fwd_bwd(lambda n: store.view(n, bf16=True), x, y, gv)
+        store.apply_flat(fg)
+    for n in store.names:
+        assert T.allclose(got[n], store.view(n), atol=1e-6), n
