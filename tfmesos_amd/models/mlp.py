@@ -24,6 +24,9 @@ class MnistMLP(object):
         self.inputs = image_pixels
         self.classes = classes
         self.seed = seed
+        self._side = None       # lazy side stream for the dW2 branch
+        self._ev_fork = None
+        self._ev_join = None
 
     def param_specs(self):
         return [
