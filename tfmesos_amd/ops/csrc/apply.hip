@@ -11,6 +11,8 @@
 
 namespace {
 
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+
 // grads arrive fp32 (collective-reduced) or bf16; templated load
 template <typename G>
 DEVINL float load_g(const G* g, long i);
@@ -19,6 +21,22 @@ DEVINL float load_g<float>(const float* g, long i) { return g[i]; }
 template <>
 DEVINL float load_g<bf16_t>(const bf16_t* g, long i) { return bf2f(g[i]); }
 
+// vectorized 4-wide grad load (the flat buffers are 256-element aligned)
+template <typename G>
+DEVINL f32x4 load_g4(const G* g, long i);
+template <>
+DEVINL f32x4 load_g4<float>(const float* g, long i) {
+  return *(const f32x4*)&g[i];
+}
+template <>
+DEVINL f32x4 load_g4<bf16_t>(const bf16_t* g, long i) {
+  const bf16x4 v = *(const bf16x4*)&g[i];
+  f32x4 o;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) o[j] = (float)v[j];
+  return o;
+}
+
 template <typename G, bool MOM, bool BF16OUT>
 __global__ void sgd_kernel(float* __restrict__ p, const G* __restrict__ g,
                            float* __restrict__ mbuf,
@@ -26,23 +44,39 @@ __global__ void sgd_kernel(float* __restrict__ p, const G* __restrict__ g,
                            float momentum, float wd, float gscale) {
   long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 4;
   long stride = (long)gridDim.x * blockDim.x * 4;
-  for (long i = i0; i < n; i += stride) {
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      long k = i + j;
-      if (k >= n) break;
-      float gv = load_g(g, k) * gscale;
-      float pv = p[k];
-      if (wd != 0.f) gv += wd * pv;
-      if (MOM) {
-        float m = mbuf[k] * momentum + gv;
-        mbuf[k] = m;
-        gv = m;
-      }
-      pv -= lr * gv;
-      p[k] = pv;
-      if (BF16OUT) pbf[k] = f2bf(pv);
+  long i = i0;
+  for (; i + 4 <= n; i += stride) {
+    // b128 loads/stores on the aligned bulk (one HBM round trip for
+    // master + grad + shadow)
+    f32x4 gv = load_g4(g, i) * gscale;
+    f32x4 pv = *(const f32x4*)&p[i];
+    if (wd != 0.f) gv += wd * pv;
+    if (MOM) {
+      f32x4 m = *(const f32x4*)&mbuf[i] * momentum + gv;
+      *(f32x4*)&mbuf[i] = m;
+      gv = m;
     }
+    pv -= lr * gv;
+    *(f32x4*)&p[i] = pv;
+    if (BF16OUT) {
+      bf16x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o[j] = (__bf16)pv[j];
+      *(bf16x4*)&((__bf16*)pbf)[i] = o;
+    }
+  }
+  for (; i < n; ++i) {      // tail (n % 4)
+    float gv = load_g(g, i) * gscale;
+    float pv = p[i];
+    if (wd != 0.f) gv += wd * pv;
+    if (MOM) {
+      float m = mbuf[i] * momentum + gv;
+      mbuf[i] = m;
+      gv = m;
+    }
+    pv -= lr * gv;
+    p[i] = pv;
+    if (BF16OUT) pbf[i] = f2bf(pv);
   }
 }
 
