@@ -88,23 +88,37 @@ __global__ void adam_kernel(float* __restrict__ p, const G* __restrict__ g,
                             float bc1, float bc2, float gscale) {
   long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 4;
   long stride = (long)gridDim.x * blockDim.x * 4;
-  for (long i = i0; i < n; i += stride) {
+  long i = i0;
+  for (; i + 4 <= n; i += stride) {
+    f32x4 gv = load_g4(g, i) * gscale;
+    f32x4 pv = *(const f32x4*)&p[i];
+    if (wd != 0.f) gv += wd * pv;
+    f32x4 mv = beta1 * *(const f32x4*)&m[i] + (1.f - beta1) * gv;
+    f32x4 vv = beta2 * *(const f32x4*)&v[i] + (1.f - beta2) * gv * gv;
+    *(f32x4*)&m[i] = mv;
+    *(f32x4*)&v[i] = vv;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      long k = i + j;
-      if (k >= n) break;
-      float gv = load_g(g, k) * gscale;
-      float pv = p[k];
-      if (wd != 0.f) gv += wd * pv;
-      float mv = beta1 * m[k] + (1.f - beta1) * gv;
-      float vv = beta2 * v[k] + (1.f - beta2) * gv * gv;
-      m[k] = mv;
-      v[k] = vv;
-      float denom = __builtin_sqrtf(vv / bc2) + eps;
-      pv -= lr * (mv / bc1) / denom;
-      p[k] = pv;
-      if (BF16OUT) pbf[k] = f2bf(pv);
+    for (int j = 0; j < 4; ++j)
+      pv[j] -= lr * (mv[j] / bc1) / (__builtin_sqrtf(vv[j] / bc2) + eps);
+    *(f32x4*)&p[i] = pv;
+    if (BF16OUT) {
+      bf16x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o[j] = (__bf16)pv[j];
+      *(bf16x4*)&((__bf16*)pbf)[i] = o;
     }
+  }
+  for (; i < n; ++i) {
+    float gv = load_g(g, i) * gscale;
+    float pv = p[i];
+    if (wd != 0.f) gv += wd * pv;
+    float mv = beta1 * m[i] + (1.f - beta1) * gv;
+    float vv = beta2 * v[i] + (1.f - beta2) * gv * gv;
+    m[i] = mv;
+    v[i] = vv;
+    pv -= lr * (mv / bc1) / (__builtin_sqrtf(vv / bc2) + eps);
+    p[i] = pv;
+    if (BF16OUT) pbf[i] = f2bf(pv);
   }
 }
 
@@ -115,20 +129,33 @@ __global__ void adagrad_kernel(float* __restrict__ p, const G* __restrict__ g,
                                float eps, float wd, float gscale) {
   long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 4;
   long stride = (long)gridDim.x * blockDim.x * 4;
-  for (long i = i0; i < n; i += stride) {
+  long i = i0;
+  for (; i + 4 <= n; i += stride) {
+    f32x4 gv = load_g4(g, i) * gscale;
+    f32x4 pv = *(const f32x4*)&p[i];
+    if (wd != 0.f) gv += wd * pv;
+    f32x4 a = *(const f32x4*)&acc[i] + gv * gv;
+    *(f32x4*)&acc[i] = a;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      long k = i + j;
-      if (k >= n) break;
-      float gv = load_g(g, k) * gscale;
-      float pv = p[k];
-      if (wd != 0.f) gv += wd * pv;
-      float a = acc[k] + gv * gv;
-      acc[k] = a;
-      pv -= lr * gv / (__builtin_sqrtf(a) + eps);
-      p[k] = pv;
-      if (BF16OUT) pbf[k] = f2bf(pv);
+    for (int j = 0; j < 4; ++j)
+      pv[j] -= lr * gv[j] / (__builtin_sqrtf(a[j]) + eps);
+    *(f32x4*)&p[i] = pv;
+    if (BF16OUT) {
+      bf16x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o[j] = (__bf16)pv[j];
+      *(bf16x4*)&((__bf16*)pbf)[i] = o;
     }
+  }
+  for (; i < n; ++i) {
+    float gv = load_g(g, i) * gscale;
+    float pv = p[i];
+    if (wd != 0.f) gv += wd * pv;
+    float a = acc[i] + gv * gv;
+    acc[i] = a;
+    pv -= lr * gv / (__builtin_sqrtf(a) + eps);
+    p[i] = pv;
+    if (BF16OUT) pbf[i] = f2bf(pv);
   }
 }
 
