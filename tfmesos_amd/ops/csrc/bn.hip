@@ -154,15 +154,20 @@ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
     const float sh = RELU ? (float)b[c] - mu * sc : 0.f;
     const long step = (long)Z * 4;
     long p = (long)z * 4 + pr;
-    for (; p + step < P; p += 2 * step) {
-      const long i0 = p * C + c, i1 = (p + step) * C + c;
+    for (; p + 3 * step < P; p += 4 * step) {
+      const long i0 = p * C + c, i1 = i0 + step * C;
+      const long i2 = i1 + step * C, i3 = i2 + step * C;
       const float x0 = (float)x[i0], x1 = (float)x[i1];
+      const float x2 = (float)x[i2], x3 = (float)x[i3];
       float d0 = (float)dy[i0], d1 = (float)dy[i1];
-      const float h0 = (x0 - mu) * is, h1 = (x1 - mu) * is;
+      float d2 = (float)dy[i2], d3 = (float)dy[i3];
       if (RELU && (float)(__bf16)(x0 * sc + sh) <= 0.f) d0 = 0.f;
       if (RELU && (float)(__bf16)(x1 * sc + sh) <= 0.f) d1 = 0.f;
-      s1 += d0 + d1;
-      s2 += d0 * h0 + d1 * h1;
+      if (RELU && (float)(__bf16)(x2 * sc + sh) <= 0.f) d2 = 0.f;
+      if (RELU && (float)(__bf16)(x3 * sc + sh) <= 0.f) d3 = 0.f;
+      s1 += d0 + d1 + d2 + d3;
+      s2 += d0 * ((x0 - mu) * is) + d1 * ((x1 - mu) * is) +
+            d2 * ((x2 - mu) * is) + d3 * ((x3 - mu) * is);
     }
     for (; p < P; p += step) {
       const long i = p * C + c;
