@@ -61,3 +61,16 @@ def test_cmd_mode_env_contract(tmp_path):
         c.join(timeout=60)
     lines = sorted(out.read_text().strip().splitlines())
     assert lines == ["ps:0:1:ps:0", "ps:1:1:ps:1", "worker:0:1:worker:0"]
+
+
+@pytest.mark.timeout(240)
+def test_cluster_env_kwarg_reaches_tasks(tmp_path):
+    """cluster(env=...) is exported into every task's environment
+    (reference tfmesos/scheduler.py:186 env kwarg)."""
+    out = tmp_path / "envdump"
+    jobs = [dict(name="worker", num=1,
+                 cmd="echo $MY_CUSTOM_VAR > %s" % out)]
+    from tfmesos_amd import cluster
+    with cluster(jobs, quiet=True, env={"MY_CUSTOM_VAR": "hello42"}) as c:
+        c.join(timeout=60)
+    assert out.read_text().strip() == "hello42"
