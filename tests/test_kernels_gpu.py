@@ -529,3 +529,26 @@ def test_maxpool3x3s2_fwd_bwd():
         r2.backward(dy.float().cpu())
         dxe = (x.grad.float().cpu() - xf.grad).abs().max()
         assert dxe < 0.01 * xf.grad.abs().max() + 0.01, (shape, dxe)
+
+
+def test_bn_bwd_accepts_channel_narrow_dy():
+    """torch.cat backward hands each branch a channel-narrow VIEW of the
+    block gradient; the BN backward reads it in place (no copy)."""
+    from tfmesos_amd import ops
+    torch.manual_seed(70)
+    N, C, H, W, CT = 2, 32, 9, 9, 96
+    x = bf(torch.randn(N, C, H, W)).requires_grad_(True)
+    g = (torch.rand(C, device=DEV) + 0.5).requires_grad_(True)
+    b = (torch.randn(C, device=DEV) * 0.2).requires_grad_(True)
+    y = ops.batch_norm_act(x, g, b, relu=True)
+    big = bf(torch.randn(N, CT, H, W)).contiguous(
+        memory_format=torch.channels_last)
+    dy_view = big.narrow(1, 32, C)          # strided channel slice
+    assert not dy_view.is_contiguous(memory_format=torch.channels_last)
+    y.backward(dy_view)
+    gx1 = x.grad.clone()
+    x.grad = None
+
+    y2 = ops.batch_norm_act(x, g, b, relu=True)
+    y2.backward(dy_view.contiguous(memory_format=torch.channels_last))
+    assert torch.equal(gx1, x.grad)

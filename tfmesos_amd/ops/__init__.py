@@ -404,6 +404,17 @@ def linear(x, w, b=None):
     return _LinearFn.apply(x, w, b)
 
 
+def _is_cl_narrow(t):
+    """Channels-last tensor OR a channel-narrow view of one (torch.cat's
+    backward hands out such views; the HIP BN/pool backward kernels read
+    them in place instead of copying)."""
+    if t.dim() != 4:
+        return False
+    s = t.stride()
+    return (s[1] == 1 and s[3] >= t.shape[1] and s[2] == t.shape[3] * s[3]
+            and s[0] == t.shape[2] * s[2])
+
+
 class _BatchNormActFn(torch.autograd.Function):
     """Fused train-mode batch-norm (+ optional relu) on the HIP kernels
     (csrc/bn.hip); fp32 torch reference on CPU."""
@@ -436,7 +447,8 @@ class _BatchNormActFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, y, weight, bias, mean, invstd = ctx.saved_tensors
         if x.is_cuda:
-            dy = dy.contiguous(memory_format=torch.channels_last)
+            if not _is_cl_narrow(dy):
+                dy = dy.contiguous(memory_format=torch.channels_last)
             dx, dgamma, dbeta = _ext().bn_bwd(
                 x, dy, weight.to(torch.bfloat16), bias.to(torch.bfloat16),
                 mean, invstd, ctx.relu)
@@ -535,7 +547,8 @@ class _MaxPool3x3s2Fn(torch.autograd.Function):
     def backward(ctx, dy):
         (idx,) = ctx.saved_tensors
         if ctx.gpu:
-            dy = dy.contiguous(memory_format=torch.channels_last)
+            if not _is_cl_narrow(dy):
+                dy = dy.contiguous(memory_format=torch.channels_last)
             return _ext().maxpool3x3s2_bwd(dy, idx, *ctx.hw)
         return torch.nn.functional.max_unpool2d(
             dy.float(), idx, 3, stride=2,

@@ -21,6 +21,24 @@ namespace {
 
 constexpr int MAXC = 2048;   // LDS scale/shift staging bound (40 KB worst case)
 
+// magic division for the flat-index -> channel decode in the strided-dy
+// apply path (see conv.hip for the derivation)
+struct FDiv {
+  unsigned long long m;
+  int s;
+};
+inline FDiv make_fd(int d) {
+  FDiv f;
+  int L = 0;
+  while ((1LL << L) < d) ++L;
+  f.s = 31 + L;
+  f.m = ((1ULL << f.s) + d - 1) / d;
+  return f;
+}
+DEVINL unsigned fd(unsigned x, FDiv f) {
+  return (unsigned)(((unsigned long long)x * f.m) >> f.s);
+}
+
 // grid (ceil(C/64), Z); block 256 = 64 channel lanes x 4 pixel rows
 __global__ __launch_bounds__(256)
 void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict__ part,
@@ -137,7 +155,7 @@ void bn_apply_kernel(const __bf16* __restrict__ x,
 template <bool RELU>
 __global__ __launch_bounds__(256)
 void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
-                         const __bf16* __restrict__ dy,
+                         const __bf16* __restrict__ dy, long ldy,
                          const __bf16* __restrict__ g,
                          const __bf16* __restrict__ b,
                          const float* __restrict__ mean,
@@ -157,10 +175,12 @@ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
     for (; p + 3 * step < P; p += 4 * step) {
       const long i0 = p * C + c, i1 = i0 + step * C;
       const long i2 = i1 + step * C, i3 = i2 + step * C;
+      const long j0 = p * ldy + c, j1 = j0 + step * ldy;
+      const long j2 = j1 + step * ldy, j3 = j2 + step * ldy;
       const float x0 = (float)x[i0], x1 = (float)x[i1];
       const float x2 = (float)x[i2], x3 = (float)x[i3];
-      float d0 = (float)dy[i0], d1 = (float)dy[i1];
-      float d2 = (float)dy[i2], d3 = (float)dy[i3];
+      float d0 = (float)dy[j0], d1 = (float)dy[j1];
+      float d2 = (float)dy[j2], d3 = (float)dy[j3];
       if (RELU && (float)(__bf16)(x0 * sc + sh) <= 0.f) d0 = 0.f;
       if (RELU && (float)(__bf16)(x1 * sc + sh) <= 0.f) d1 = 0.f;
       if (RELU && (float)(__bf16)(x2 * sc + sh) <= 0.f) d2 = 0.f;
@@ -170,9 +190,8 @@ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
             d2 * ((x2 - mu) * is) + d3 * ((x3 - mu) * is);
     }
     for (; p < P; p += step) {
-      const long i = p * C + c;
-      const float xv = (float)x[i];
-      float d = (float)dy[i];
+      const float xv = (float)x[p * C + c];
+      float d = (float)dy[p * ldy + c];
       const float xh = (xv - mu) * is;
       if (RELU && (float)(__bf16)(xv * sc + sh) <= 0.f) d = 0.f;
       s1 += d;
@@ -221,7 +240,7 @@ void bn_bwd_finalize_kernel(const float* __restrict__ part,
 template <bool RELU, bool VEC>
 __global__ __launch_bounds__(256)
 void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
-                         const __bf16* __restrict__ dy,
+                         const __bf16* __restrict__ dy, long ldy, FDiv dC,
                          const __bf16* __restrict__ g,
                          const __bf16* __restrict__ b,
                          const float* __restrict__ mean,
@@ -248,10 +267,12 @@ void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
   const long stride = (long)gridDim.x * 256 * 8;
   for (long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8; i < total;
        i += stride) {
-    const int c0 = (int)(i % C);
-    if (VEC && i + 8 <= total) {
+    const long prow = fd((unsigned)i, dC);           // i / C (i < 2^31)
+    const int c0 = (int)(i - prow * C);
+    const long jbase = prow * ldy + c0;
+    if (VEC && i + 8 <= total && c0 + 8 <= C) {
       bf16x8 xv = *(const bf16x8*)&x[i];
-      bf16x8 dv = *(const bf16x8*)&dy[i];
+      bf16x8 dv = *(const bf16x8*)&dy[jbase];
       bf16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -265,13 +286,14 @@ void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
       *(bf16x8*)&dx[i] = o;
     } else {
       int c = c0;
+      long jrow = prow;
       for (int j = 0; j < 8 && i + j < total; ++j) {
         const float x1 = (float)x[i + j];
-        float d = (float)dy[i + j];
+        float d = (float)dy[jrow * ldy + c];
         const float xh = (x1 - lmu[c]) * lis[c];
         if (RELU && (float)(__bf16)(x1 * lgs[c] + lsh[c]) <= 0.f) d = 0.f;
         dx[i + j] = (__bf16)(lgs[c] * (d - la[c] - xh * lbb[c]));
-        if (++c == C) c = 0;
+        if (++c == C) { c = 0; ++jrow; }
       }
     }
   }
@@ -319,30 +341,33 @@ void launch_bn_fwd(const bf16_t* x, const bf16_t* g, const bf16_t* b,
 #undef APPLY
 }
 
-void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, const bf16_t* g,
-                   const bf16_t* b, const float* mean, const float* invstd,
-                   bf16_t* dx, bf16_t* dgamma, bf16_t* dbeta, float* part,
-                   float* s1n, float* s2n, long P, int C, int Z,
-                   bool relu, hipStream_t stream) {
+void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, long ldy,
+                   const bf16_t* g, const bf16_t* b, const float* mean,
+                   const float* invstd, bf16_t* dx, bf16_t* dgamma,
+                   bf16_t* dbeta, float* part, float* s1n, float* s2n,
+                   long P, int C, int Z, bool relu, hipStream_t stream) {
   dim3 sg(ceil_div(C, 64), Z), sb(256);
   if (relu)
     hipLaunchKernelGGL((bn_bwd_stats_kernel<true>), sg, sb, 0, stream,
-                       (const __bf16*)x, (const __bf16*)dy, (const __bf16*)g,
-                       (const __bf16*)b, mean, invstd, part, P, C, Z);
+                       (const __bf16*)x, (const __bf16*)dy, ldy,
+                       (const __bf16*)g, (const __bf16*)b, mean, invstd,
+                       part, P, C, Z);
   else
     hipLaunchKernelGGL((bn_bwd_stats_kernel<false>), sg, sb, 0, stream,
-                       (const __bf16*)x, (const __bf16*)dy, (const __bf16*)g,
-                       (const __bf16*)b, mean, invstd, part, P, C, Z);
+                       (const __bf16*)x, (const __bf16*)dy, ldy,
+                       (const __bf16*)g, (const __bf16*)b, mean, invstd,
+                       part, P, C, Z);
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(64), 0, stream,
                      part, (__bf16*)dgamma, (__bf16*)dbeta, s1n, s2n, C, Z,
                      1.f / (float)P);
   dim3 ag(ew_grid(P * (long)C)), ab(256);
   const bool vec = (C & 7) == 0;
+  const FDiv dC = make_fd(C);
 #define APPLY(RELUv, VECv)                                                  \
   hipLaunchKernelGGL((bn_bwd_apply_kernel<RELUv, VECv>), ag, ab, 0, stream, \
-                     (const __bf16*)x, (const __bf16*)dy, (const __bf16*)g,  \
-                     (const __bf16*)b, mean, invstd, s1n, s2n, (__bf16*)dx,  \
-                     P, C)
+                     (const __bf16*)x, (const __bf16*)dy, ldy, dC,           \
+                     (const __bf16*)g, (const __bf16*)b, mean, invstd, s1n,  \
+                     s2n, (__bf16*)dx, P, C)
   if (relu) { if (vec) APPLY(true, true); else APPLY(true, false); }
   else      { if (vec) APPLY(false, true); else APPLY(false, false); }
 #undef APPLY

@@ -53,17 +53,18 @@ int conv_bwdw_slices(int, int, int, int, int, int, int);
 void launch_bn_fwd(const bf16_t*, const bf16_t*, const bf16_t*, bf16_t*,
                    float*, float*, float*, long, int, int, float, bool,
                    hipStream_t);
-void launch_bn_bwd(const bf16_t*, const bf16_t*, const bf16_t*, const bf16_t*,
-                   const float*, const float*, bf16_t*, bf16_t*, bf16_t*,
-                   float*, float*, float*, long, int, int, bool,
-                   hipStream_t);
+void launch_bn_bwd(const bf16_t*, const bf16_t*, long, const bf16_t*,
+                   const bf16_t*, const float*, const float*, bf16_t*,
+                   bf16_t*, bf16_t*, float*, float*, float*, long, int,
+                   int, bool, hipStream_t);
 int bn_stats_slices(long, int);
 int bn_max_channels();
 void launch_avg3x3(const bf16_t*, bf16_t*, int, int, int, int, hipStream_t);
 void launch_maxpool3x3s2_fwd(const bf16_t*, bf16_t*, unsigned char*, int,
                              int, int, int, int, int, hipStream_t);
-void launch_maxpool3x3s2_bwd(const bf16_t*, const unsigned char*, bf16_t*,
-                             int, int, int, int, int, int, hipStream_t);
+void launch_maxpool3x3s2_bwd(const bf16_t*, long, const unsigned char*,
+                             bf16_t*, int, int, int, int, int, int,
+                             hipStream_t);
 
 namespace {
 
@@ -525,12 +526,27 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor g,
   return {y, mean, invstd};
 }
 
+// dy may be a channel-narrow view of a wider channels-last tensor (the
+// backward of torch.cat): detected by its strides and read in place —
+// no contiguous() copy.
+static bool cl_narrow(const torch::Tensor& t, long* ldy) {
+  if (t.dim() != 4) return false;
+  const auto s = t.strides();
+  const long C = t.size(1), H = t.size(2), W = t.size(3);
+  if (s[1] != 1 || s[3] < C || s[2] != W * s[3] || s[0] != H * s[2])
+    return false;
+  *ldy = s[3];
+  return true;
+}
+
 std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
                                   torch::Tensor g, torch::Tensor b,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   bool relu) {
-  TORCH_CHECK(is_cl(x) && is_cl(dy),
-              "bn_bwd: channels-last tensors required");
+  long ldy = 0;
+  TORCH_CHECK(is_cl(x), "bn_bwd: channels-last x required");
+  TORCH_CHECK(cl_narrow(dy, &ldy), "bn_bwd: dy must be channels-last or a "
+              "channel-narrow view of a channels-last tensor");
   const int C = x.size(1);
   const long P = x.numel() / C;
   const int Z = bn_stats_slices(P, C);
@@ -542,7 +558,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   auto s1n = torch::empty({C}, opts);
   auto s2n = torch::empty({C}, opts);
   launch_bn_bwd((const bf16_t*)x.data_ptr(), (const bf16_t*)dy.data_ptr(),
-                (const bf16_t*)g.data_ptr(), (const bf16_t*)b.data_ptr(),
+                ldy, (const bf16_t*)g.data_ptr(), (const bf16_t*)b.data_ptr(),
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
                 (bf16_t*)dx.data_ptr(), (bf16_t*)dgamma.data_ptr(),
                 (bf16_t*)dbeta.data_ptr(), part.data_ptr<float>(),
@@ -581,12 +597,15 @@ std::vector<torch::Tensor> maxpool3x3s2_fwd(torch::Tensor x) {
 
 torch::Tensor maxpool3x3s2_bwd(torch::Tensor dy, torch::Tensor idx,
                                long H, long W) {
-  TORCH_CHECK(dy.is_cuda() && is_cl(dy) && is_cl(idx));
+  long ldy = 0;
+  TORCH_CHECK(dy.is_cuda() && cl_narrow(dy, &ldy) && is_cl(idx),
+              "maxpool bwd: dy must be channels-last or a channel-narrow "
+              "view of one");
   const int N = dy.size(0), C = dy.size(1);
   auto dx = torch::empty({N, C, H, W},
                          dy.options().memory_format(
                              torch::MemoryFormat::ChannelsLast));
-  launch_maxpool3x3s2_bwd((const bf16_t*)dy.data_ptr(),
+  launch_maxpool3x3s2_bwd((const bf16_t*)dy.data_ptr(), ldy,
                           idx.data_ptr<unsigned char>(),
                           (bf16_t*)dx.data_ptr(), N, H, W, C, dy.size(2),
                           dy.size(3), cur_stream());

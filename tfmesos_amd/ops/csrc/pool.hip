@@ -146,7 +146,7 @@ typedef __attribute__((ext_vector_type(8))) unsigned char u8x8;
 
 template <bool VEC>
 __global__ __launch_bounds__(256)
-void maxpool3x3s2_bwd_kernel(const __bf16* __restrict__ dY,
+void maxpool3x3s2_bwd_kernel(const __bf16* __restrict__ dY, long ldy,
                              const unsigned char* __restrict__ idx,
                              __bf16* __restrict__ dX,
                              int N, int H, int W, int C, int Ho, int Wo) {
@@ -172,9 +172,9 @@ void maxpool3x3s2_bwd_kernel(const __bf16* __restrict__ dY,
         for (int wo = wo_lo; wo <= wo_hi; ++wo) {
           const int s = w - wo * 2;
           if (s < 0 || s > 2) continue;
-          const long oidx = (((long)n * Ho + ho) * Wo + wo) * C + c;
-          const u8x8 iv = *(const u8x8*)&idx[oidx];
-          const bf16x8 dv = *(const bf16x8*)&dY[oidx];
+          const long op = ((long)n * Ho + ho) * Wo + wo;
+          const u8x8 iv = *(const u8x8*)&idx[op * C + c];
+          const bf16x8 dv = *(const bf16x8*)&dY[op * ldy + c];
           const unsigned char want = (unsigned char)(r * 3 + s);
 #pragma unroll
           for (int j = 0; j < 4; ++j)
@@ -198,9 +198,9 @@ void maxpool3x3s2_bwd_kernel(const __bf16* __restrict__ dY,
         for (int wo = wo_lo; wo <= wo_hi; ++wo) {
           const int s = w - wo * 2;
           if (s < 0 || s > 2) continue;
-          const long oidx = (((long)n * Ho + ho) * Wo + wo) * C + c;
-          if (idx[oidx] == (unsigned char)(r * 3 + s))
-            acc += (float)dY[oidx];
+          const long op = ((long)n * Ho + ho) * Wo + wo;
+          if (idx[op * C + c] == (unsigned char)(r * 3 + s))
+            acc += (float)dY[op * ldy + c];
         }
       }
       out[i] = (__bf16)acc;
@@ -224,16 +224,17 @@ void launch_maxpool3x3s2_fwd(const bf16_t* X, bf16_t* Y, unsigned char* idx,
                        Ho, Wo);
 }
 
-void launch_maxpool3x3s2_bwd(const bf16_t* dY, const unsigned char* idx,
+void launch_maxpool3x3s2_bwd(const bf16_t* dY, long ldy,
+                             const unsigned char* idx,
                              bf16_t* dX, int N, int H, int W, int C, int Ho,
                              int Wo, hipStream_t stream) {
   dim3 grid((unsigned)(N * H)), block(256);
   if ((C & 7) == 0)
     hipLaunchKernelGGL((maxpool3x3s2_bwd_kernel<true>), grid, block, 0,
-                       stream, (const __bf16*)dY, idx, (__bf16*)dX, N, H, W,
-                       C, Ho, Wo);
+                       stream, (const __bf16*)dY, ldy, idx, (__bf16*)dX, N,
+                       H, W, C, Ho, Wo);
   else
     hipLaunchKernelGGL((maxpool3x3s2_bwd_kernel<false>), grid, block, 0,
-                       stream, (const __bf16*)dY, idx, (__bf16*)dX, N, H, W,
-                       C, Ho, Wo);
+                       stream, (const __bf16*)dY, ldy, idx, (__bf16*)dX, N,
+                       H, W, C, Ho, Wo);
 }
