@@ -97,9 +97,6 @@ def run_nmf(args, device, rank, world):
         gbatch = args.nmf_n
     def one_step():
         wl.one_step()
-    if world == 1 and device.type == "cuda" and not args.no_graph:
-        from tfmesos_amd.utils.graphstep import GraphedStep
-        one_step = GraphedStep(one_step)
     one_step.finalize = getattr(wl, "finalize", lambda: None)
     return one_step, {
         "model": "nmf_%dx%d_rank%d" % (args.nmf_n, args.nmf_n, args.nmf_rank),
