@@ -171,8 +171,9 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
 
     if torch.cuda.is_available():
-        torch.cuda.set_device(local_rank)
-        device = torch.device("cuda", local_rank)
+        dev_idx = local_rank % torch.cuda.device_count()
+        torch.cuda.set_device(dev_idx)
+        device = torch.device("cuda", dev_idx)
     else:
         device = torch.device("cpu")
         log("WARNING: no GPU visible; CPU dev run (not a valid benchmark)")
