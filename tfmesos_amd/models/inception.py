@@ -31,10 +31,12 @@ class Conv2d(nn.Module):
         self.weight = nn.Parameter(
             torch.randn(cout, kernel_size[0], kernel_size[1], cin)
             * (2.0 / fan_in) ** 0.5)
+        self._dw_buf = None    # fp32 view into the model grad arena (GPU)
 
     def forward(self, x):
         return ops.conv2d(x, self.weight, None, stride=self.stride,
-                          padding=self.padding, weight_format="krsc")
+                          padding=self.padding, weight_format="krsc",
+                          dw_out=self._dw_buf if x.is_cuda else None)
 
 
 class BatchNorm2d(nn.Module):
@@ -204,8 +206,33 @@ class InceptionV3(nn.Module):
         ])
         self.fc_w = nn.Parameter(torch.randn(2048, num_classes) * 0.01)
         self.fc_b = nn.Parameter(torch.zeros(num_classes))
+        self._arena = None
+
+    def _ensure_arena(self, device):
+        """One fp32 grad arena for every conv weight: each layer's
+        bwd-weight kernel atomically accumulates into its slice, so the
+        whole backward needs ONE bulk zero (here, at forward start)
+        instead of ~94 per-layer fills, and the trainer gathers the
+        fp32 grads with one batched bf16 copy (no per-layer casts).
+        Constraint: one forward per backward (no cross-step grad
+        accumulation) — which is the replica-trainer step pattern."""
+        if self._arena is not None and self._arena.device == device:
+            self._arena.zero_()
+            return
+        convs = [m for m in self.modules() if isinstance(m, Conv2d)]
+        total = sum(m.weight.numel() for m in convs)
+        self._arena = torch.zeros(total, dtype=torch.float32,
+                                  device=device)
+        off = 0
+        for m in convs:
+            n = m.weight.numel()
+            m._dw_buf = self._arena[off:off + n].view(m.weight.shape)
+            m.weight._tfa_raw_grad = m._dw_buf
+            off += n
 
     def forward(self, x):
+        if x.is_cuda:
+            self._ensure_arena(x.device)
         for m in self.stem:
             x = m(x)
         x = _max_pool(x, 3, 2)

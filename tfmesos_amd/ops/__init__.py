@@ -259,11 +259,12 @@ class _Conv2dFn(torch.autograd.Function):
     torch fp32 reference so model code runs in CI."""
 
     @staticmethod
-    def forward(ctx, x, w, bias, stride, padding, krsc):
+    def forward(ctx, x, w, bias, stride, padding, krsc, dw_out=None):
         ctx.stride = stride
         ctx.padding = padding
         ctx.has_bias = bias is not None
         ctx.krsc = krsc
+        ctx.dw_out = dw_out
         if x.is_cuda:
             # channels-last memory: gathers become contiguous channel
             # runs (csrc/conv.hip); weights are tap-major [K,R,S,C] —
@@ -290,48 +291,73 @@ class _Conv2dFn(torch.autograd.Function):
         krsc = ctx.krsc
         if x.is_cuda:
             dy = dy.contiguous(memory_format=torch.channels_last)
-            # [C,R,S,K] = W^T memory for the bwd-data gather
-            wt = (w.permute(3, 1, 2, 0) if krsc
-                  else w.permute(1, 2, 3, 0)).contiguous()
             R, S = (w.shape[1], w.shape[2]) if krsc \
                 else (w.shape[2], w.shape[3])
-            dx = _ext().conv2d_bwd_data(dy, wt, x.shape[2], x.shape[3],
-                                        ctx.stride[0], ctx.stride[1],
-                                        ctx.padding[0], ctx.padding[1])
-            dwm = _ext().conv2d_bwd_weight(dy, x, R, S,
-                                           ctx.stride[0], ctx.stride[1],
-                                           ctx.padding[0], ctx.padding[1])
-            # kernel emits tap-major [K,R,S,C]: native for krsc params
-            dw = dwm.to(w.dtype) if krsc \
-                else dwm.permute(0, 3, 1, 2).to(w.dtype)
+            dx = None
+            if ctx.needs_input_grad[0]:
+                # [C,R,S,K] = W^T memory for the bwd-data gather
+                wt = (w.permute(3, 1, 2, 0) if krsc
+                      else w.permute(1, 2, 3, 0)).contiguous()
+                dx = _ext().conv2d_bwd_data(
+                    dy, wt, x.shape[2], x.shape[3],
+                    ctx.stride[0], ctx.stride[1],
+                    ctx.padding[0], ctx.padding[1])
+            if ctx.dw_out is not None:
+                # grad-arena path (krsc only): atomically accumulate
+                # into the model's pre-zeroed fp32 buffer; the trainer
+                # gathers it with one batched bf16 copy — no per-layer
+                # fill / cast kernels, and autograd sees no w grad.
+                _ext().conv2d_bwd_weight_out(
+                    dy, x, R, S, ctx.stride[0], ctx.stride[1],
+                    ctx.padding[0], ctx.padding[1], ctx.dw_out)
+                dw = None
+            else:
+                dwm = _ext().conv2d_bwd_weight(
+                    dy, x, R, S, ctx.stride[0], ctx.stride[1],
+                    ctx.padding[0], ctx.padding[1])
+                # kernel emits tap-major [K,R,S,C]: native for krsc
+                dw = dwm.to(w.dtype) if krsc \
+                    else dwm.permute(0, 3, 1, 2).to(w.dtype)
         else:
             dy = dy.contiguous()
             wk = w.permute(0, 3, 1, 2) if krsc else w
             dyf, xf, wf = dy.float(), x.float(), wk.float()
-            dx = torch.nn.grad.conv2d_input(
-                x.shape, wf, dyf, stride=ctx.stride, padding=ctx.padding
-            ).to(x.dtype)
+            dx = None
+            if ctx.needs_input_grad[0]:
+                dx = torch.nn.grad.conv2d_input(
+                    x.shape, wf, dyf, stride=ctx.stride,
+                    padding=ctx.padding).to(x.dtype)
             dw = torch.nn.grad.conv2d_weight(
                 xf, wk.shape, dyf, stride=ctx.stride, padding=ctx.padding)
             if krsc:
                 dw = dw.permute(0, 2, 3, 1)
             dw = dw.contiguous().to(w.dtype)
         db = dy.float().sum(dim=(0, 2, 3)) if ctx.has_bias else None
-        return dx, dw, db, None, None, None
+        return dx, dw, db, None, None, None, None
 
 
-def conv2d(x, w, bias=None, stride=1, padding=0, weight_format="kcrs"):
+def conv2d(x, w, bias=None, stride=1, padding=0, weight_format="kcrs",
+           dw_out=None):
     """2-D convolution (NCHW activations): hand-written implicit-GEMM
     MFMA kernels on GPU (csrc/conv.hip), torch fp32 reference on CPU.
     Differentiable. weight_format "kcrs" (torch layout) or "krsc"
     (tap-major — the kernels' native layout; parameters stored this way
-    skip a permute+copy per call in fwd AND in the weight-grad path)."""
+    skip a permute+copy per call in fwd AND in the weight-grad path).
+
+    dw_out (GPU + krsc only): a pre-zeroed fp32 [K,R,S,C] buffer the
+    weight grad is atomically accumulated into instead of being returned
+    through autograd (w.grad stays None) — the model grad-arena path
+    that batches ~90 per-layer fill/cast kernels into one bulk zero and
+    one foreach gather copy per step."""
     if isinstance(stride, int):
         stride = (stride, stride)
     if isinstance(padding, int):
         padding = (padding, padding)
+    if dw_out is not None:
+        assert weight_format == "krsc" and x.is_cuda, \
+            "dw_out requires krsc weights on GPU"
     return _Conv2dFn.apply(x, w, bias, tuple(stride), tuple(padding),
-                           weight_format == "krsc")
+                           weight_format == "krsc", dw_out)
 
 
 class _SoftmaxXentFn(torch.autograd.Function):

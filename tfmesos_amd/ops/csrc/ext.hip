@@ -499,6 +499,27 @@ torch::Tensor conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
   return dw;
 }
 
+// Accumulates dW into a caller-owned PRE-ZEROED fp32 [K,R,S,C] buffer
+// (the per-model grad arena: one bulk zero per backward replaces ~90
+// per-layer fills, and the fp32 result feeds the trainer's batched
+// bf16 gather copy — no per-layer cast kernels either).
+void conv2d_bwd_weight_out(torch::Tensor dy, torch::Tensor x,
+                           long R, long S, long stride_h, long stride_w,
+                           long pad_h, long pad_w, torch::Tensor dw) {
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda() && is_cl(dy) && is_cl(x),
+              "dy/x must be channels-last");
+  TORCH_CHECK(dw.is_cuda() && dw.is_contiguous() &&
+              dw.scalar_type() == torch::kFloat32 && dw.dim() == 4 &&
+              dw.size(0) == dy.size(1) && dw.size(1) == R &&
+              dw.size(2) == S && dw.size(3) == x.size(1),
+              "dw must be contiguous fp32 [K,R,S,C]");
+  launch_conv_bwd_weight((const bf16_t*)dy.data_ptr(),
+                         (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(),
+                         x.size(0), x.size(1), x.size(2), x.size(3),
+                         dy.size(1), R, S, dy.size(2), dy.size(3), stride_h,
+                         stride_w, pad_h, pad_w, cur_stream());
+}
+
 // -------------------------------------------------------------------- bn
 
 std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor g,
@@ -645,6 +666,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd", &bn_bwd, "fused batch-norm (+relu mask) bwd");
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
+  m.def("conv2d_bwd_weight_out", &conv2d_bwd_weight_out);
   m.def("embedding_gather", &embedding_gather);
   m.def("embedding_scatter_add", &embedding_scatter_add);
   m.def("relu_bwd", &relu_bwd);

@@ -6,9 +6,12 @@ mnist MLP) to autograd models (the Inception config of BASELINE.json):
 * every parameter's ``.data`` IS a named view of the PS flat **bf16
   shadow** — the per-step broadcast refreshes module weights in place,
   no per-tensor copies;
-* every parameter's ``.grad`` IS a view of the flat **bf16 gradient
-  buffer** — autograd accumulates straight into the reduce buffer, so
-  push is one (sharded) ``dist.reduce`` with zero gather kernels;
+* gradients are gathered into the flat **bf16 gradient buffer** with
+  one batched ``_foreach_copy_`` per step (``.grad`` stays ``None``
+  between steps so autograd assigns instead of accumulating; params
+  with an out-of-band fp32 ``_tfa_raw_grad`` — the conv grad arena —
+  are converted in the same batched pass), so push is one (sharded)
+  ``dist.reduce``;
 * apply stays ONE fused HIP kernel per PS shard over fp32 masters
   (bf16 gradients are converted inside the kernel, csrc/apply.hip).
 """
@@ -53,9 +56,16 @@ class ModuleReplicaTrainer(object):
         weights update in place via the shadow views."""
         views, grads = [], []
         for p, gv in zip(self._params, self._gviews):
-            if p.grad is not None:
+            # grad-arena params (conv weights on GPU) publish their
+            # fp32 grad out-of-band: autograd never materializes a
+            # per-layer bf16 grad and the foreach copy below does the
+            # fp32->bf16 conversion in the same batched pass
+            g = getattr(p, "_tfa_raw_grad", None)
+            if g is None:
+                g = p.grad
+            if g is not None:
                 views.append(gv)
-                grads.append(p.grad)
+                grads.append(g)
         if grads:
             torch._foreach_copy_(views, grads)
             for p in self._params:
