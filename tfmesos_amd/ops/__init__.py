@@ -295,11 +295,11 @@ class _Conv2dFn(torch.autograd.Function):
                 else (w.shape[2], w.shape[3])
             dx = None
             if ctx.needs_input_grad[0]:
-                # [C,R,S,K] = W^T memory for the bwd-data gather
-                wt = (w.permute(3, 1, 2, 0) if krsc
-                      else w.permute(1, 2, 3, 0)).contiguous()
+                # kernel reads W in its native [K,R,S,C] layout (the
+                # fwd tensor) — krsc params pass straight through
+                wk = w if krsc else w.permute(0, 2, 3, 1).contiguous()
                 dx = _ext().conv2d_bwd_data(
-                    dy, wt, x.shape[2], x.shape[3],
+                    dy, wk, x.shape[2], x.shape[3],
                     ctx.stride[0], ctx.stride[1],
                     ctx.padding[0], ctx.padding[1])
             if ctx.dw_out is not None:

@@ -18,8 +18,8 @@
 //   fwd:    Y[n,ho,wo,k]  = sum_{r,s,c} X[n, ho*U+r-P, wo*V+s-Q, c] W[k,r,s,c]
 //           GEMM  M = N*Ho*Wo, Ncol = K,     Kdim = R*S*C
 //   bwd-d:  dX[n,h,w,c]   = sum_{r,s,k} dY[n,(h+P-r)/U,(w+Q-s)/V,k] W[k,r,s,c]
-//           GEMM  M = N*H*W,   Ncol = C,     Kdim = R*S*K  (wants W^T memory
-//           [c][r][s][k], built by the host wrapper)
+//           GEMM  M = N*H*W,   Ncol = C,     Kdim = R*S*K  (W read in its
+//           native [k][r][s][c] layout by a transposing stager)
 //   bwd-w:  dW[k,r,s,c]   = sum_{n,ho,wo} dY[n,ho,wo,k] X[n,ho*U+r-P,...,c]
 //           GEMM  M = K,       Ncol = R*S*C, Kdim = N*Ho*Wo
 #include "common.h"
@@ -355,12 +355,44 @@ DEVINL void stage_patch_bwdd(const __bf16* __restrict__ dY,
   }
 }
 
+// Bs[c][q]: weight read in its NATIVE [K,R,S,C] layout (the forward
+// tensor) — 8 contiguous c for one tap q=(r,s,k), transposed into LDS.
+// Kills the host-side W^T permute+copy the backward used to pay per
+// layer per step (~94 launch-bound copies across Inception).
+DEVINL void stage_w_krsc(const __bf16* __restrict__ W, __bf16* Sn,
+                         int c0blk, int q0, const ConvShape cs, int KD,
+                         int t) {
+  const int px = t >> 3;          // tap q within the BK tile (0..31)
+  const int kk0 = (t & 7) * 8;    // c chunk
+  const int q = q0 + px;
+  const int c = c0blk + kk0;
+  if (q >= KD || c >= cs.C) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) srow(Sn, kk0 + j)[px] = (__bf16)0.f;
+    return;
+  }
+  const int rs = fdiv(q, cs.dK);
+  const int k = q - rs * cs.K;
+  const int r = fdiv(rs, cs.dS);
+  const int s = rs - r * cs.S;
+  const __bf16* src = W + (((long)k * cs.R + r) * cs.S + s) * cs.C + c;
+  if (((cs.C & 7) == 0) && c + 8 <= cs.C) {
+    bf16x8 v = *(const bf16x8*)src;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) srow(Sn, kk0 + j)[px] = v[j];
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      srow(Sn, kk0 + j)[px] = (c + j < cs.C) ? src[j] : (__bf16)0.f;
+  }
+}
+
 template <int STRIDE, bool SK>
 __global__ __launch_bounds__(256)
 void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ Wt,
                       __bf16* __restrict__ dX, float* __restrict__ ws, int kc,
                       ConvShape cs) {
-  // Wt memory: [C][R*S*K] (host-permuted W^T)
+  // Wt memory: NATIVE [K][R*S*C] (the forward weight tensor)
   __shared__ __align__(16) __bf16 As[2][TILE_ELEMS];
   __shared__ __align__(16) __bf16 Bs[2][TILE_ELEMS];
   const long M = (long)cs.N * cs.H * cs.W;
@@ -378,14 +410,14 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
 
   f32x4 acc[2][2] = {};
   stage_patch_bwdd<STRIDE>(dY, As[0], cs, tm0, ks, M, KD, t, kvec);
-  stage_wrows(Wt, Bs[0], tn0, ks, cs.C, KD, t, kvec);
+  stage_w_krsc(Wt, Bs[0], tn0, ks, cs, KD, t);
   __syncthreads();
   int cur = 0;
   for (int k0 = ks; k0 < ke; k0 += BK, cur ^= 1) {
     if (k0 + BK < ke) {
       stage_patch_bwdd<STRIDE>(dY, As[cur ^ 1], cs, tm0, k0 + BK, M, KD, t,
                                kvec);
-      stage_wrows(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs.C, KD, t, kvec);
+      stage_w_krsc(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs, KD, t);
     }
     const int kfrag = (lane >> 4) * 8;
     bf16x8 bfrag[2];

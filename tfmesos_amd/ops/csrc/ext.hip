@@ -458,9 +458,11 @@ torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor wt,
   TORCH_CHECK(dy.is_cuda() && dy.dim() == 4 && is_cl(dy) &&
               dy.scalar_type() == torch::kBFloat16,
               "dy must be bf16 channels-last");
+  // weight in its NATIVE [K,R,S,C] layout — the kernel's transposing
+  // stager reads it directly (no host-side W^T permute per call)
   TORCH_CHECK(wt.is_contiguous() && wt.dim() == 4 &&
-              wt.size(3) == dy.size(1), "wt must be [C,R,S,K]");
-  const long C = wt.size(0), R = wt.size(1), S = wt.size(2);
+              wt.size(0) == dy.size(1), "w must be [K,R,S,C]");
+  const long C = wt.size(3), R = wt.size(1), S = wt.size(2);
   auto dx = torch::empty({dy.size(0), C, H, W},
                          dy.options().memory_format(
                              torch::MemoryFormat::ChannelsLast));

@@ -26,8 +26,21 @@ class ModuleReplicaTrainer(object):
     def __init__(self, module, optimizer="sgd", hparams=None, device="cpu",
                  n_ps=None, colocate_ps=False):
         self.device = torch.device(device)
-        params = [(name, p.detach().float().cpu())
-                  for name, p in module.named_parameters()]
+        named = list(module.named_parameters())
+        # models with a grad arena (Inception convs) get their arena
+        # params laid out FIRST in the flat buffers, so the fp32 arena
+        # maps onto one contiguous flat_grad region and the per-step
+        # gather is a single fused cast-copy
+        self._arena = None
+        arena_names = []
+        if self.device.type == "cuda" and hasattr(module,
+                                                  "wire_grad_arena"):
+            arena_names, self._arena = module.wire_grad_arena(self.device)
+        pdict = dict(named)
+        order = arena_names + [n for n, _ in named
+                               if n not in set(arena_names)]
+        params = [(name, pdict[name].detach().float().cpu())
+                  for name in order]
         self.t = SyncReplicaTrainer(params, optimizer=optimizer,
                                     hparams=hparams, device=device,
                                     grad_dtype=torch.bfloat16, n_ps=n_ps,
@@ -54,18 +67,18 @@ class ModuleReplicaTrainer(object):
         """Gather grads into the flat buffer (foreach copy), reduce
         (sharded) -> fused apply on PS -> broadcast shadows. Module
         weights update in place via the shadow views."""
+        if self._arena is not None:
+            # grad-arena params (conv weights) occupy flat_grad[0:Na]
+            # in arena order: ONE fused fp32->bf16 cast-copy gathers
+            # all of them (their .grad stays None — the bwd-weight
+            # kernels accumulated into the arena directly)
+            na = self._arena.numel()
+            self.t.flat_grad[:na].copy_(self._arena)
         views, grads = [], []
         for p, gv in zip(self._params, self._gviews):
-            # grad-arena params (conv weights on GPU) publish their
-            # fp32 grad out-of-band: autograd never materializes a
-            # per-layer bf16 grad and the foreach copy below does the
-            # fp32->bf16 conversion in the same batched pass
-            g = getattr(p, "_tfa_raw_grad", None)
-            if g is None:
-                g = p.grad
-            if g is not None:
+            if p.grad is not None:
                 views.append(gv)
-                grads.append(g)
+                grads.append(p.grad)
         if grads:
             torch._foreach_copy_(views, grads)
             for p in self._params:
