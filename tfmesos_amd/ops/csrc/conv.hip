@@ -408,7 +408,10 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
   const int ks = SK ? blockIdx.z * kc : 0;
   const int ke = SK ? min(ks + kc, KD) : KD;
 
-  f32x4 acc[2][2] = {};
+  // 32x32x16 MFMA core (same scheme as conv_fwd: one 32x32 wave tile,
+  // 2 MFMA + 2 fragment loads per BK=32 iteration — ~1.7x the issue
+  // efficiency of the earlier 2x2 16x16x32 fragment scheme here).
+  f32x16 acc = {};
   stage_patch_bwdd<STRIDE>(dY, As[0], cs, tm0, ks, M, KD, t, kvec);
   stage_w_krsc(Wt, Bs[0], tn0, ks, cs, KD, t);
   __syncthreads();
@@ -419,53 +422,38 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
                                kvec);
       stage_w_krsc(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs, KD, t);
     }
-    const int kfrag = (lane >> 4) * 8;
-    bf16x8 bfrag[2];
 #pragma unroll
-    for (int fn = 0; fn < 2; ++fn)
-      bfrag[fn] =
-          *(const bf16x8*)&srow(Bs[cur], wc * 32 + fn * 16 + (lane & 15))[kfrag];
-#pragma unroll
-    for (int fm = 0; fm < 2; ++fm) {
-      bf16x8 a =
-          *(const bf16x8*)&srow(As[cur], wr * 32 + fm * 16 + (lane & 15))[kfrag];
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
-        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a, bfrag[fn], acc[fm][fn], 0, 0, 0);
+    for (int kh = 0; kh < 2; ++kh) {
+      const int ko = kh * 16 + ((lane >> 5) << 3);
+      bf16x8 a = *(const bf16x8*)&srow(As[cur], wr * 32 + (lane & 31))[ko];
+      bf16x8 b = *(const bf16x8*)&srow(Bs[cur], wc * 32 + (lane & 31))[ko];
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
     }
     __syncthreads();
   }
 
+  const int c = tn0 + wc * 32 + (lane & 31);
   if (SK) {
-    float* wsl = ws + (long)blockIdx.z * M * cs.C;
+    if (c < cs.C) {
+      float* wsl = ws + (long)blockIdx.z * M * cs.C;
 #pragma unroll
-    for (int fm = 0; fm < 2; ++fm)
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        const int c = tn0 + wc * 32 + fn * 16 + (lane & 15);
-        if (c >= cs.C) continue;
-#pragma unroll
-        for (int rr = 0; rr < 4; ++rr) {
-          const long pm = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
-          if (pm < M) wsl[pm * cs.C + c] = acc[fm][fn][rr];
-        }
-      }
-    return;
-  }
-#pragma unroll
-  for (int fm = 0; fm < 2; ++fm)
-#pragma unroll
-    for (int fn = 0; fn < 2; ++fn) {
-      const int c = tn0 + wc * 32 + fn * 16 + (lane & 15);
-      if (c >= cs.C) continue;
-#pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const long pm = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
-        if (pm >= M) continue;
-        dX[pm * cs.C + c] = (__bf16)acc[fm][fn][rr];
+      for (int v = 0; v < 16; ++v) {
+        const long pm = tm0 + wr * 32 + ((v >> 2) << 3) +
+                        ((lane >> 5) << 2) + (v & 3);
+        if (pm < M) wsl[pm * cs.C + c] = acc[v];
       }
     }
+    return;
+  }
+  if (c < cs.C) {
+#pragma unroll
+    for (int v = 0; v < 16; ++v) {
+      const long pm = tm0 + wr * 32 + ((v >> 2) << 3) +
+                      ((lane >> 5) << 2) + (v & 3);
+      if (pm >= M) continue;
+      dX[pm * cs.C + c] = (__bf16)acc[v];
+    }
+  }
 }
 
 // -------------------------------------------------------------- bwd-weight
