@@ -9,19 +9,12 @@
 // coalesced — and per-channel scale/shift live in LDS for the
 // elementwise passes. Three kernels per direction:
 //
-//   fwd: stats (grid C/64 x Z, per-slice partials; the LAST-ARRIVING
-//        workgroup of each 64-channel column reduces the partials and
-//        writes mean/invstd in the same launch — a separate finalize
-//        kernel was ~4 us of near-zero work x 94 layers x 2 directions,
-//        ~0.75 ms of pure kernel floor per Inception step)
-//        -> apply (normalize+affine+ReLU, one pass, b128 when C%8==0)
-//   bwd: stats-partial (sum dy_eff, sum dy_eff*xhat; ReLU mask fused,
-//        last-arriver writes dgamma/dbeta + normalized sums) -> apply
+//   fwd: stats-partial (grid C/64 x Z, no atomics -> per-slice partials)
+//        -> finalize (mean/invstd) -> apply (normalize+affine+ReLU, one
+//        pass, b128 vectorized when C % 8 == 0)
+//   bwd: stats-partial (sum dy_eff, sum dy_eff*xhat; ReLU mask y>0
+//        fused) -> finalize (dgamma/dbeta + normalized sums) -> apply
 //        (dx in one pass)
-//
-// The arrival counters come from the shared persistent splitk_ws pool
-// (ext.hip): every user resets its counters to zero before finishing,
-// so they can be shared by all stream-ordered launches on one stream.
 #include "common.h"
 
 namespace {
@@ -46,30 +39,10 @@ DEVINL unsigned fd(unsigned x, FDiv f) {
   return (unsigned)(((unsigned long long)x * f.m) >> f.s);
 }
 
-// Last-arriver handoff: returns true in EVERY thread of exactly one
-// workgroup per counter — the one whose atomicAdd observes Z-1 — after
-// making all earlier partial writes visible (release fence before the
-// add, acquire fence in the winner). The winner resets the counter so
-// the shared pool stays zeroed for the next stream-ordered launch.
-DEVINL bool last_arriver(int* cnt, int idx, int z_total) {
-  __shared__ bool last;
-  if (threadIdx.x == 0) {
-    __threadfence();
-    last = (atomicAdd(&cnt[idx], 1) == z_total - 1);
-    if (last) cnt[idx] = 0;
-  }
-  __syncthreads();
-  if (last) __threadfence();
-  return last;
-}
-
-// grid (ceil(C/64), Z); block 256 = 64 channel lanes x 4 pixel rows.
-// The last workgroup of each channel column also finalizes mean/invstd.
+// grid (ceil(C/64), Z); block 256 = 64 channel lanes x 4 pixel rows
 __global__ __launch_bounds__(256)
 void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict__ part,
-                     float* __restrict__ mean, float* __restrict__ invstd,
-                     int* __restrict__ cnt, long P, int C, int Z,
-                     float inv_count, float eps) {
+                     long P, int C, int Z) {
   const int cl = threadIdx.x & 63;
   const int pr = threadIdx.x >> 6;           // 0..3
   const int c = blockIdx.x * 64 + cl;
@@ -104,27 +77,28 @@ void bn_stats_kernel(const __bf16* __restrict__ x, float* __restrict__ part,
     part[((long)z * C + c) * 2] = sum;
     part[((long)z * C + c) * 2 + 1] = sq;
   }
-  if (Z > 1 && !last_arriver(cnt, blockIdx.x, Z)) return;
-  // finalize this 64-channel column: 4 z-strips per channel, LDS reduce
+}
+
+// one wave per channel: lanes cover the Z partial slices in parallel
+// (a single-workgroup loop over Z was latency-bound at ~28 us)
+__global__ __launch_bounds__(64)
+void bn_finalize_kernel(const float* __restrict__ part,
+                        float* __restrict__ mean,
+                        float* __restrict__ invstd,
+                        int C, int Z, float inv_count, float eps) {
+  const int c = blockIdx.x;
+  const int l = threadIdx.x;
   float s = 0.f, q = 0.f;
-  if (c < C) {
-    if (Z > 1) {
-      for (int zz = pr; zz < Z; zz += 4) {
-        s += part[((long)zz * C + c) * 2];
-        q += part[((long)zz * C + c) * 2 + 1];
-      }
-    } else if (pr == 0) {
-      s = sum;
-      q = sq;
-    }
+  for (int z = l; z < Z; z += 64) {
+    s += part[((long)z * C + c) * 2];
+    q += part[((long)z * C + c) * 2 + 1];
   }
-  __syncthreads();
-  ls[pr][cl] = s;
-  lq[pr][cl] = q;
-  __syncthreads();
-  if (pr == 0 && c < C) {
-    s = ls[0][cl] + ls[1][cl] + ls[2][cl] + ls[3][cl];
-    q = lq[0][cl] + lq[1][cl] + lq[2][cl] + lq[3][cl];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s += __shfl_xor(s, off, 64);
+    q += __shfl_xor(q, off, 64);
+  }
+  if (l == 0) {
     const float m = s * inv_count;
     float var = q * inv_count - m * m;
     if (var < 0.f) var = 0.f;
@@ -186,13 +160,7 @@ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
                          const __bf16* __restrict__ b,
                          const float* __restrict__ mean,
                          const float* __restrict__ invstd,
-                         float* __restrict__ part,
-                         __bf16* __restrict__ dgamma,
-                         __bf16* __restrict__ dbeta,
-                         float* __restrict__ s1nv,
-                         float* __restrict__ s2nv,
-                         int* __restrict__ cnt, long P, int C, int Z,
-                         float inv_count) {
+                         float* __restrict__ part, long P, int C, int Z) {
   const int cl = threadIdx.x & 63;
   const int pr = threadIdx.x >> 6;
   const int c = blockIdx.x * 64 + cl;
@@ -240,32 +208,32 @@ void bn_bwd_stats_kernel(const __bf16* __restrict__ x,
     part[((long)z * C + c) * 2] = s1;
     part[((long)z * C + c) * 2 + 1] = s2;
   }
-  if (Z > 1 && !last_arriver(cnt, blockIdx.x, Z)) return;
-  // finalize: dbeta = sum dy_eff, dgamma = sum dy_eff*xhat, plus the
-  // normalized sums the apply pass needs
-  float f1 = 0.f, f2 = 0.f;
-  if (c < C) {
-    if (Z > 1) {
-      for (int zz = pr; zz < Z; zz += 4) {
-        f1 += part[((long)zz * C + c) * 2];
-        f2 += part[((long)zz * C + c) * 2 + 1];
-      }
-    } else if (pr == 0) {
-      f1 = s1;
-      f2 = s2;
-    }
+}
+
+__global__ __launch_bounds__(64)
+void bn_bwd_finalize_kernel(const float* __restrict__ part,
+                            __bf16* __restrict__ dgamma,
+                            __bf16* __restrict__ dbeta,
+                            float* __restrict__ s1n,
+                            float* __restrict__ s2n,
+                            int C, int Z, float inv_count) {
+  const int c = blockIdx.x;
+  const int l = threadIdx.x;
+  float s1 = 0.f, s2 = 0.f;
+  for (int z = l; z < Z; z += 64) {
+    s1 += part[((long)z * C + c) * 2];
+    s2 += part[((long)z * C + c) * 2 + 1];
   }
-  __syncthreads();
-  l1[pr][cl] = f1;
-  l2[pr][cl] = f2;
-  __syncthreads();
-  if (pr == 0 && c < C) {
-    f1 = l1[0][cl] + l1[1][cl] + l1[2][cl] + l1[3][cl];
-    f2 = l2[0][cl] + l2[1][cl] + l2[2][cl] + l2[3][cl];
-    dbeta[c] = (__bf16)f1;
-    dgamma[c] = (__bf16)f2;
-    s1nv[c] = f1 * inv_count;
-    s2nv[c] = f2 * inv_count;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s1 += __shfl_xor(s1, off, 64);
+    s2 += __shfl_xor(s2, off, 64);
+  }
+  if (l == 0) {
+    dbeta[c] = (__bf16)s1;
+    dgamma[c] = (__bf16)s2;
+    s1n[c] = s1 * inv_count;
+    s2n[c] = s2 * inv_count;
   }
 }
 
@@ -354,11 +322,14 @@ inline unsigned ew_grid(long total) {
 
 void launch_bn_fwd(const bf16_t* x, const bf16_t* g, const bf16_t* b,
                    bf16_t* y, float* mean, float* invstd, float* part,
-                   int* cnt, long P, int C, int Z, float eps, bool relu,
+                   long P, int C, int Z, float eps, bool relu,
                    hipStream_t stream) {
   dim3 sg(ceil_div(C, 64), Z), sb(256);
   hipLaunchKernelGGL(bn_stats_kernel, sg, sb, 0, stream, (const __bf16*)x,
-                     part, mean, invstd, cnt, P, C, Z, 1.f / (float)P, eps);
+                     part, P, C, Z);
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64),
+                     0, stream, part, mean, invstd, C, Z, 1.f / (float)P,
+                     eps);
   dim3 ag(ew_grid(P * (long)C)), ab(256);
   const bool vec = (C & 7) == 0;
 #define APPLY(RELUv, VECv)                                                  \
@@ -374,21 +345,21 @@ void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, long ldy,
                    const bf16_t* g, const bf16_t* b, const float* mean,
                    const float* invstd, bf16_t* dx, bf16_t* dgamma,
                    bf16_t* dbeta, float* part, float* s1n, float* s2n,
-                   int* cnt, long P, int C, int Z, bool relu,
-                   hipStream_t stream) {
+                   long P, int C, int Z, bool relu, hipStream_t stream) {
   dim3 sg(ceil_div(C, 64), Z), sb(256);
   if (relu)
     hipLaunchKernelGGL((bn_bwd_stats_kernel<true>), sg, sb, 0, stream,
                        (const __bf16*)x, (const __bf16*)dy, ldy,
                        (const __bf16*)g, (const __bf16*)b, mean, invstd,
-                       part, (__bf16*)dgamma, (__bf16*)dbeta, s1n, s2n,
-                       cnt, P, C, Z, 1.f / (float)P);
+                       part, P, C, Z);
   else
     hipLaunchKernelGGL((bn_bwd_stats_kernel<false>), sg, sb, 0, stream,
                        (const __bf16*)x, (const __bf16*)dy, ldy,
                        (const __bf16*)g, (const __bf16*)b, mean, invstd,
-                       part, (__bf16*)dgamma, (__bf16*)dbeta, s1n, s2n,
-                       cnt, P, C, Z, 1.f / (float)P);
+                       part, P, C, Z);
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(64), 0, stream,
+                     part, (__bf16*)dgamma, (__bf16*)dbeta, s1n, s2n, C, Z,
+                     1.f / (float)P);
   dim3 ag(ew_grid(P * (long)C)), ab(256);
   const bool vec = (C & 7) == 0;
   const FDiv dC = make_fd(C);

@@ -51,12 +51,12 @@ void launch_conv_bwd_weight(const bf16_t*, const bf16_t*, float*,
                             int, int, int, int, hipStream_t);
 int conv_bwdw_slices(int, int, int, int, int, int, int);
 void launch_bn_fwd(const bf16_t*, const bf16_t*, const bf16_t*, bf16_t*,
-                   float*, float*, float*, int*, long, int, int, float,
-                   bool, hipStream_t);
+                   float*, float*, float*, long, int, int, float, bool,
+                   hipStream_t);
 void launch_bn_bwd(const bf16_t*, const bf16_t*, long, const bf16_t*,
                    const bf16_t*, const float*, const float*, bf16_t*,
-                   bf16_t*, bf16_t*, float*, float*, float*, int*, long,
-                   int, int, bool, hipStream_t);
+                   bf16_t*, bf16_t*, float*, float*, float*, long, int,
+                   int, bool, hipStream_t);
 int bn_stats_slices(long, int);
 int bn_max_channels();
 void launch_avg3x3(const bf16_t*, bf16_t*, int, int, int, int, hipStream_t);
@@ -541,12 +541,10 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor g,
   auto mean = torch::empty({C}, opts);
   auto invstd = torch::empty({C}, opts);
   auto part = torch::empty({(long)Z * C * 2}, opts);
-  int* cnt;
-  splitk_ws(x.device(), 0, (C + 63) / 64, &cnt);
   launch_bn_fwd((const bf16_t*)x.data_ptr(), (const bf16_t*)g.data_ptr(),
                 (const bf16_t*)b.data_ptr(), (bf16_t*)y.data_ptr(),
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                part.data_ptr<float>(), cnt, P, C, Z, (float)eps, relu,
+                part.data_ptr<float>(), P, C, Z, (float)eps, relu,
                 cur_stream());
   return {y, mean, invstd};
 }
@@ -582,15 +580,13 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   auto part = torch::empty({(long)Z * C * 2}, opts);
   auto s1n = torch::empty({C}, opts);
   auto s2n = torch::empty({C}, opts);
-  int* cnt;
-  splitk_ws(x.device(), 0, (C + 63) / 64, &cnt);
   launch_bn_bwd((const bf16_t*)x.data_ptr(), (const bf16_t*)dy.data_ptr(),
                 ldy, (const bf16_t*)g.data_ptr(), (const bf16_t*)b.data_ptr(),
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
                 (bf16_t*)dx.data_ptr(), (bf16_t*)dgamma.data_ptr(),
                 (bf16_t*)dbeta.data_ptr(), part.data_ptr<float>(),
-                s1n.data_ptr<float>(), s2n.data_ptr<float>(), cnt, P, C,
-                Z, relu, cur_stream());
+                s1n.data_ptr<float>(), s2n.data_ptr<float>(), P, C, Z,
+                relu, cur_stream());
   return {dx, dgamma, dbeta};
 }
 
