@@ -265,6 +265,7 @@ class _Conv2dFn(torch.autograd.Function):
         ctx.has_bias = bias is not None
         ctx.krsc = krsc
         ctx.dw_out = dw_out
+        ctx.pw = False
         if x.is_cuda:
             # channels-last memory: gathers become contiguous channel
             # runs (csrc/conv.hip); weights are tap-major [K,R,S,C] —
@@ -274,6 +275,23 @@ class _Conv2dFn(torch.autograd.Function):
             wm = w.contiguous() if krsc else \
                 w.permute(0, 2, 3, 1).contiguous()
             ctx.save_for_backward(xm, w)
+            ctx.pw = (wm.shape[1] == 1 and wm.shape[2] == 1
+                      and stride == (1, 1) and padding == (0, 0))
+            if ctx.pw:
+                # pointwise conv IS a GEMM over pixels; the tuned GEMM
+                # kernel (csrc/gemm.hip) beats the implicit-GEMM conv
+                # path on the large-P Inception shapes (tools/
+                # bench_1x1.py: 7.3 vs 9.2 us fwd at 35^2) but loses on
+                # 8^2 (its tile grid is too small without split-tap),
+                # hence the P gate
+                N, C, H, W = xm.shape
+                P = N * H * W
+                if P >= 4096:
+                    x2 = xm.permute(0, 2, 3, 1).reshape(P, C)
+                    eb = bias.float() if bias is not None else None
+                    y2 = gemm_bias_act(x2, wm.view(wm.shape[0], C),
+                                       bias=eb, trans_b=True)
+                    return y2.view(N, H, W, -1).permute(0, 3, 1, 2)
             eb = bias.float() if bias is not None else \
                 torch.empty(0, device=x.device)
             return _ext().conv2d_fwd(xm, wm, eb, stride[0], stride[1],
@@ -298,10 +316,19 @@ class _Conv2dFn(torch.autograd.Function):
                 # kernel reads W in its native [K,R,S,C] layout (the
                 # fwd tensor) — krsc params pass straight through
                 wk = w if krsc else w.permute(0, 2, 3, 1).contiguous()
-                dx = _ext().conv2d_bwd_data(
-                    dy, wk, x.shape[2], x.shape[3],
-                    ctx.stride[0], ctx.stride[1],
-                    ctx.padding[0], ctx.padding[1])
+                if ctx.pw:
+                    # pointwise: dX = dY @ W on the tuned GEMM kernel
+                    # (faster than the conv bwd-data path on every
+                    # Inception 1x1 shape — tools/bench_1x1.py)
+                    N, K, Ho, Wo = dy.shape
+                    dy2 = dy.permute(0, 2, 3, 1).reshape(N * Ho * Wo, K)
+                    dx2 = gemm_bias_act(dy2, wk.reshape(K, -1))
+                    dx = dx2.view(N, Ho, Wo, -1).permute(0, 3, 1, 2)
+                else:
+                    dx = _ext().conv2d_bwd_data(
+                        dy, wk, x.shape[2], x.shape[3],
+                        ctx.stride[0], ctx.stride[1],
+                        ctx.padding[0], ctx.padding[1])
             if ctx.dw_out is not None:
                 # grad-arena path (krsc only): atomically accumulate
                 # into the model's pre-zeroed fp32 buffer; the trainer
