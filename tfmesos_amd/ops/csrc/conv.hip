@@ -806,11 +806,7 @@ void launch_conv_bwd_data(const bf16_t* dY, const bf16_t* Wt, bf16_t* dX,
 int conv_bwdw_slices(int N, int C, int K, int R, int S, int Ho, int Wo) {
   const int KD = C * R * S;
   const long Ptot = (long)N * Ho * Wo;
-  // MT=2 only where the tap-amplified X re-read argument holds: multi-
-  // tap filters with enough K rows to fill the 128-row tile (1x1 layers
-  // measured SLOWER with MT=2: no amplification to save, and the z/
-  // occupancy costs dominate)
-  const int mtile = (K >= 2 * BM && R * S > 1) ? 2 * BM : BM;
+  const int mtile = K > BM ? 2 * BM : BM;   // must match the launch
   const long tiles = (long)ceil_div(K, mtile) * ceil_div(KD, BN);
   long zmax = (Ptot + BK - 1) / BK;
   long zwant = 1024 / tiles;
@@ -837,7 +833,7 @@ void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
   long pc = (Ptot + z - 1) / z;
   pc = (pc + BK - 1) / BK * BK;
   z = (int)((Ptot + pc - 1) / pc);
-  const int mtile = (K >= 2 * BM && R * S > 1) ? 2 * BM : BM;
+  const int mtile = K > BM ? 2 * BM : BM;
   dim3 grid(ceil_div(KD, BN), ceil_div(K, mtile), z);
   dim3 block(256);
   if (mtile == 2 * BM)
