@@ -609,23 +609,17 @@ struct XBwdwStage {
 // sliced). A fixed-order stripe-reduce variant measured SLOWER: dW is
 // small (9-300K elements), so the z-deep serial reduce loop had too
 // little parallelism, while the atomic contention here is negligible.
-//
-// MT = 64-row m-subtiles per workgroup (1 or 2). The kernel is bound
-// by the tap-amplified X traffic (each m-tile re-reads the full tap-
-// expanded X strip: 9x for a 3x3 layer), so K>64 layers use MT=2 —
-// one 128-row tile halves the X re-reads for ~10 KB more LDS.
-template <int MT>
 __global__ __launch_bounds__(256)
 void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ X,
                       float* __restrict__ dW, float* __restrict__ ws,
                       ConvShape cs, long pc) {
-  __shared__ __align__(16) __bf16 As[2][MT * TILE_ELEMS];
+  __shared__ __align__(16) __bf16 As[2][TILE_ELEMS];
   __shared__ __align__(16) __bf16 Bs[2][TILE_ELEMS];
   const int KD = cs.R * cs.S * cs.C;
   const long Ptot = (long)cs.N * cs.Ho * cs.Wo;
   const bool cvec = (cs.C & 7) == 0;
   const bool kvec = (cs.K & 7) == 0;
-  const int tm0 = blockIdx.y * (MT * BM);
+  const int tm0 = blockIdx.y * BM;
   const int tn0 = blockIdx.x * BN;
   const long ps = (long)blockIdx.z * pc;
   const long pe = min(ps + pc, Ptot);
@@ -634,24 +628,18 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
   const int wave = t >> 6;
   const int wr = wave >> 1, wc = wave & 1;
 
-  f32x4 acc[MT][2][2] = {};
-  DyBwdwStage dst[MT];
+  f32x4 acc[2][2] = {};
+  DyBwdwStage dst;
   XBwdwStage xst;
-#pragma unroll
-  for (int mt = 0; mt < MT; ++mt)
-    dst[mt].init(dY, cs, tm0 + mt * BM, ps, t, kvec);
+  dst.init(dY, cs, tm0, ps, t, kvec);
   xst.init(cs, tn0, ps, KD, t, cvec);
-#pragma unroll
-  for (int mt = 0; mt < MT; ++mt)
-    dst[mt].stage(As[0] + mt * TILE_ELEMS, cs, Ptot);
+  dst.stage(As[0], cs, Ptot);
   xst.stage(X, Bs[0], cs, KD, Ptot);
   __syncthreads();
   int cur = 0;
   for (long p0 = ps; p0 < pe; p0 += BK, cur ^= 1) {
     if (p0 + BK < pe) {
-#pragma unroll
-      for (int mt = 0; mt < MT; ++mt)
-        dst[mt].stage(As[cur ^ 1] + mt * TILE_ELEMS, cs, Ptot);
+      dst.stage(As[cur ^ 1], cs, Ptot);
       xst.stage(X, Bs[cur ^ 1], cs, KD, Ptot);
     }
     const int kfrag = (lane >> 4) * 8;
@@ -661,38 +649,33 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
       bfrag[fn] =
           *(const bf16x8*)&srow(Bs[cur], wc * 32 + fn * 16 + (lane & 15))[kfrag];
 #pragma unroll
-    for (int mt = 0; mt < MT; ++mt)
+    for (int fm = 0; fm < 2; ++fm) {
+      bf16x8 a =
+          *(const bf16x8*)&srow(As[cur], wr * 32 + fm * 16 + (lane & 15))[kfrag];
 #pragma unroll
-      for (int fm = 0; fm < 2; ++fm) {
-        bf16x8 a = *(const bf16x8*)&srow(As[cur] + mt * TILE_ELEMS,
-                                         wr * 32 + fm * 16 + (lane & 15))[kfrag];
-#pragma unroll
-        for (int fn = 0; fn < 2; ++fn)
-          acc[mt][fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a, bfrag[fn], acc[mt][fm][fn], 0, 0, 0);
-      }
+      for (int fn = 0; fn < 2; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, bfrag[fn], acc[fm][fn], 0, 0, 0);
+    }
     __syncthreads();
   }
 
 #pragma unroll
-  for (int mt = 0; mt < MT; ++mt)
+  for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
-    for (int fm = 0; fm < 2; ++fm)
+    for (int fn = 0; fn < 2; ++fn) {
+      const int q = tn0 + wc * 32 + fn * 16 + (lane & 15);
+      if (q >= KD) continue;
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        const int q = tn0 + wc * 32 + fn * 16 + (lane & 15);
-        if (q >= KD) continue;
-#pragma unroll
-        for (int rr = 0; rr < 4; ++rr) {
-          const int k = tm0 + mt * BM + wr * 32 + fm * 16 +
-                        (lane >> 4) * 4 + rr;
-          if (k >= cs.K) continue;
-          if (gridDim.z == 1)
-            dW[(long)k * KD + q] = acc[mt][fm][fn][rr];
-          else
-            unsafeAtomicAdd(&dW[(long)k * KD + q], acc[mt][fm][fn][rr]);
-        }
+      for (int rr = 0; rr < 4; ++rr) {
+        const int k = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
+        if (k >= cs.K) continue;
+        if (gridDim.z == 1)
+          dW[(long)k * KD + q] = acc[fm][fn][rr];
+        else
+          unsafeAtomicAdd(&dW[(long)k * KD + q], acc[fm][fn][rr]);
       }
+    }
 }
 
 }  // namespace
@@ -806,8 +789,7 @@ void launch_conv_bwd_data(const bf16_t* dY, const bf16_t* Wt, bf16_t* dX,
 int conv_bwdw_slices(int N, int C, int K, int R, int S, int Ho, int Wo) {
   const int KD = C * R * S;
   const long Ptot = (long)N * Ho * Wo;
-  const int mtile = K > BM ? 2 * BM : BM;   // must match the launch
-  const long tiles = (long)ceil_div(K, mtile) * ceil_div(KD, BN);
+  const long tiles = (long)ceil_div(K, BM) * ceil_div(KD, BN);
   long zmax = (Ptot + BK - 1) / BK;
   long zwant = 1024 / tiles;
   if (zwant < 1) zwant = 1;
@@ -833,13 +815,8 @@ void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
   long pc = (Ptot + z - 1) / z;
   pc = (pc + BK - 1) / BK * BK;
   z = (int)((Ptot + pc - 1) / pc);
-  const int mtile = K > BM ? 2 * BM : BM;
-  dim3 grid(ceil_div(KD, BN), ceil_div(K, mtile), z);
+  dim3 grid(ceil_div(KD, BN), ceil_div(K, BM), z);
   dim3 block(256);
-  if (mtile == 2 * BM)
-    hipLaunchKernelGGL((conv_bwdw_kernel<2>), grid, block, 0, stream,
-                       (const __bf16*)dY, (const __bf16*)X, dW, ws, cs, pc);
-  else
-    hipLaunchKernelGGL((conv_bwdw_kernel<1>), grid, block, 0, stream,
-                       (const __bf16*)dY, (const __bf16*)X, dW, ws, cs, pc);
+  hipLaunchKernelGGL(conv_bwdw_kernel, grid, block, 0, stream,
+                     (const __bf16*)dY, (const __bf16*)X, dW, ws, cs, pc);
 }
