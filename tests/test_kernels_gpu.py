@@ -552,3 +552,59 @@ def test_bn_bwd_accepts_channel_narrow_dy():
     y2 = ops.batch_norm_act(x, g, b, relu=True)
     y2.backward(dy_view.contiguous(memory_format=torch.channels_last))
     assert torch.equal(gx1, x.grad)
+
+
+def test_bn_strided_out_matches_dense():
+    """bn_fwd with a channel-narrow out view (the Inception fused-cat
+    path) writes exactly what the dense apply produces."""
+    from tfmesos_amd import ops
+    torch.manual_seed(11)
+    x = bf(torch.randn(4, 48, 9, 9)).contiguous(
+        memory_format=torch.channels_last)
+    g = bf(torch.randn(48).abs() + 0.4)
+    b = bf(torch.randn(48))
+    dense = ops.batch_norm_act(x, g, b, relu=True)
+    buf = torch.zeros(4, 128, 9, 9, device=DEV, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    out = ops.batch_norm_act(x, g, b, relu=True, out=buf.narrow(1, 40, 48))
+    torch.cuda.synchronize()
+    assert torch.equal(out, dense)
+    assert torch.equal(buf.narrow(1, 40, 48), dense)
+    # untouched slices stay zero
+    assert buf.narrow(1, 0, 40).abs().sum().item() == 0.0
+
+
+def test_fused_cat_block_matches_cat():
+    """InceptionA with the zero-copy fused concat is bit-identical
+    (forward AND gradients) to the same block running plain torch.cat."""
+    import tfmesos_amd.models.inception as inc
+
+    def run(block, x, fuse):
+        orig = inc._fused_cat
+        if not fuse:
+            inc._fused_cat = \
+                lambda xx, specs: torch.cat([b(None) for _, b in specs], 1)
+        try:
+            xx = x.detach().clone().requires_grad_(True)
+            y = block(xx)
+            loss = y.float().square().mean()
+            for p in block.parameters():
+                p.grad = None
+            loss.backward()
+            return (y.detach().clone(), xx.grad.detach().clone(),
+                    [p.grad.detach().clone() if p.grad is not None else
+                     getattr(p, "_tfa_raw_grad", torch.zeros(1)).clone()
+                     for p in block.parameters()])
+        finally:
+            inc._fused_cat = orig
+
+    torch.manual_seed(5)
+    block = inc.InceptionA(192, 32).to(DEV, torch.bfloat16)
+    x = bf(torch.randn(2, 192, 17, 17)).contiguous(
+        memory_format=torch.channels_last)
+    y1, dx1, g1 = run(block, x, fuse=True)
+    y2, dx2, g2 = run(block, x, fuse=False)
+    assert torch.equal(y1, y2)
+    assert torch.equal(dx1, dx2)
+    for a, b in zip(g1, g2):
+        assert torch.equal(a, b)

@@ -50,9 +50,9 @@ class BatchNorm2d(nn.Module):
         self.weight = nn.Parameter(torch.ones(num_features))
         self.bias = nn.Parameter(torch.zeros(num_features))
 
-    def forward(self, x):
+    def forward(self, x, out=None):
         return ops.batch_norm_act(x, self.weight, self.bias, eps=self.eps,
-                                  relu=self.relu)
+                                  relu=self.relu, out=out)
 
 
 class BasicConv2d(nn.Module):
@@ -63,8 +63,8 @@ class BasicConv2d(nn.Module):
         self.conv = Conv2d(cin, cout, **kw)
         self.bn = BatchNorm2d(cout, relu=True)
 
-    def forward(self, x):
-        return self.bn(self.conv(x))
+    def forward(self, x, out=None):
+        return self.bn(self.conv(x), out=out)
 
 
 def _avg_pool(x, k, stride=1, padding=1):
@@ -80,6 +80,45 @@ def _max_pool(x, k, stride):
     return torch.nn.functional.max_pool2d(x, k, stride=stride)
 
 
+class _JoinViews(torch.autograd.Function):
+    """Zero-copy block concat: the branches already wrote their outputs
+    into channel slices of one pre-allocated channels-last buffer (the
+    BN apply's strided store), so forward just returns the buffer and
+    backward hands each branch its channel-narrow gradient VIEW (the
+    BN/conv backward kernels read strided dy in place). Replaces the
+    aten cat copy per Inception block."""
+
+    @staticmethod
+    def forward(ctx, buf, *views):
+        ctx.chans = [v.shape[1] for v in views]
+        return buf.detach()
+
+    @staticmethod
+    def backward(ctx, dout):
+        grads, off = [], 0
+        for c in ctx.chans:
+            grads.append(dout.narrow(1, off, c))
+            off += c
+        return (None, *grads)
+
+
+def _fused_cat(x, specs):
+    """specs: [(channels, builder)] — builder(out_view_or_None) returns
+    the branch output. GPU: each branch's terminal BN writes into its
+    slice of one buffer; CPU: plain torch.cat."""
+    if not x.is_cuda:
+        return torch.cat([b(None) for _, b in specs], 1)
+    N, _, H, W = x.shape
+    ctot = sum(c for c, _ in specs)
+    buf = torch.empty((N, ctot, H, W), device=x.device, dtype=x.dtype,
+                      memory_format=torch.channels_last)
+    views, off = [], 0
+    for c, b in specs:
+        views.append(b(buf.narrow(1, off, c)))
+        off += c
+    return _JoinViews.apply(buf, *views)
+
+
 class InceptionA(nn.Module):
     def __init__(self, cin, pool_features):
         super().__init__()
@@ -90,14 +129,16 @@ class InceptionA(nn.Module):
         self.b3x3_2 = BasicConv2d(64, 96, kernel_size=3, padding=1)
         self.b3x3_3 = BasicConv2d(96, 96, kernel_size=3, padding=1)
         self.bpool = BasicConv2d(cin, pool_features, kernel_size=1)
+        self.pf = pool_features
 
     def forward(self, x):
-        return torch.cat([
-            self.b1x1(x),
-            self.b5x5_2(self.b5x5_1(x)),
-            self.b3x3_3(self.b3x3_2(self.b3x3_1(x))),
-            self.bpool(_avg_pool(x, 3)),
-        ], 1)
+        return _fused_cat(x, [
+            (64, lambda o: self.b1x1(x, out=o)),
+            (64, lambda o: self.b5x5_2(self.b5x5_1(x), out=o)),
+            (96, lambda o: self.b3x3_3(self.b3x3_2(self.b3x3_1(x)),
+                                       out=o)),
+            (self.pf, lambda o: self.bpool(_avg_pool(x, 3), out=o)),
+        ])
 
 
 class InceptionB(nn.Module):
@@ -131,12 +172,13 @@ class InceptionC(nn.Module):
         self.bpool = BasicConv2d(cin, 192, kernel_size=1)
 
     def forward(self, x):
-        return torch.cat([
-            self.b1x1(x),
-            self.b7_3(self.b7_2(self.b7_1(x))),
-            self.b7d_5(self.b7d_4(self.b7d_3(self.b7d_2(self.b7d_1(x))))),
-            self.bpool(_avg_pool(x, 3)),
-        ], 1)
+        return _fused_cat(x, [
+            (192, lambda o: self.b1x1(x, out=o)),
+            (192, lambda o: self.b7_3(self.b7_2(self.b7_1(x)), out=o)),
+            (192, lambda o: self.b7d_5(self.b7d_4(self.b7d_3(
+                self.b7d_2(self.b7d_1(x)))), out=o)),
+            (192, lambda o: self.bpool(_avg_pool(x, 3), out=o)),
+        ])
 
 
 class InceptionD(nn.Module):
@@ -173,12 +215,16 @@ class InceptionE(nn.Module):
     def forward(self, x):
         b3 = self.b3_1(x)
         b3d = self.b3d_2(self.b3d_1(x))
-        return torch.cat([
-            self.b1x1(x),
-            torch.cat([self.b3_2a(b3), self.b3_2b(b3)], 1),
-            torch.cat([self.b3d_3a(b3d), self.b3d_3b(b3d)], 1),
-            self.bpool(_avg_pool(x, 3)),
-        ], 1)
+        # the nested cats flatten: sub-branch slices are adjacent, so
+        # one buffer serves the whole block
+        return _fused_cat(x, [
+            (320, lambda o: self.b1x1(x, out=o)),
+            (384, lambda o: self.b3_2a(b3, out=o)),
+            (384, lambda o: self.b3_2b(b3, out=o)),
+            (384, lambda o: self.b3d_3a(b3d, out=o)),
+            (384, lambda o: self.b3d_3b(b3d, out=o)),
+            (192, lambda o: self.bpool(_avg_pool(x, 3), out=o)),
+        ])
 
 
 class InceptionV3(nn.Module):

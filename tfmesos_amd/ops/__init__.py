@@ -473,12 +473,22 @@ class _BatchNormActFn(torch.autograd.Function):
     (csrc/bn.hip); fp32 torch reference on CPU."""
 
     @staticmethod
-    def forward(ctx, x, weight, bias, eps, relu):
+    def forward(ctx, x, weight, bias, eps, relu, out=None):
         if x.is_cuda:
             x = x.contiguous(memory_format=torch.channels_last)
+            eout = out if out is not None \
+                else torch.empty(0, device=x.device, dtype=torch.bfloat16)
             y, mean, invstd = _ext().bn_fwd(
                 x, weight.to(torch.bfloat16), bias.to(torch.bfloat16),
-                eps, relu)
+                eps, relu, eout)
+            if out is not None:
+                # the apply wrote into the caller's concat-buffer view;
+                # detach severs the view metadata so autograd attaches
+                # this Function's backward instead of rejecting the
+                # aliased output (see _JoinViews in models/inception.py)
+                ctx.save_for_backward(x, None, weight, bias, mean, invstd)
+                ctx.relu = relu
+                return y.detach()
             # y is NOT saved for backward: the relu mask is recomputed
             # from sign(g*xhat + b) inside the bwd kernels
         else:
@@ -492,6 +502,12 @@ class _BatchNormActFn(torch.autograd.Function):
             if relu:
                 y = torch.relu(y)
             y = y.to(x.dtype)
+            if out is not None:
+                with torch.no_grad():
+                    out.copy_(y)
+                ctx.save_for_backward(x, out, weight, bias, mean, invstd)
+                ctx.relu = relu
+                return out.detach()
         ctx.save_for_backward(x, y, weight, bias, mean, invstd)
         ctx.relu = relu
         return y
@@ -519,14 +535,20 @@ class _BatchNormActFn(torch.autograd.Function):
                 dyf - (s1 / nhw).view(1, -1, 1, 1)
                 - xhat * (s2 / nhw).view(1, -1, 1, 1))
             dx = dx.to(x.dtype)
-        return dx, dgamma.to(weight.dtype), dbeta.to(weight.dtype), None, None
+        return (dx, dgamma.to(weight.dtype), dbeta.to(weight.dtype), None,
+                None, None)
 
 
-def batch_norm_act(x, weight, bias, eps=1e-3, relu=False):
-    """Differentiable fused train-mode BN (+relu); channels-last on GPU."""
+def batch_norm_act(x, weight, bias, eps=1e-3, relu=False, out=None):
+    """Differentiable fused train-mode BN (+relu); channels-last on GPU.
+
+    out (optional): a channel-narrow channels-last view the normalized
+    output is written into (strided store) — the Inception blocks pass
+    slices of a pre-allocated concat buffer so the block concat costs
+    zero copies (see models/inception.py _fused_cat)."""
     if not x.is_cuda:
         x = x.contiguous()
-    return _BatchNormActFn.apply(x, weight, bias, eps, relu)
+    return _BatchNormActFn.apply(x, weight, bias, eps, relu, out)
 
 
 def mlp_head_fused(h, w, b, labels, scale=None):

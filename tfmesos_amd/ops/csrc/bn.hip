@@ -107,13 +107,18 @@ void bn_finalize_kernel(const float* __restrict__ part,
   }
 }
 
-template <bool RELU, bool VEC>
+// STRIDED: y is a channel-narrow view of a wider channels-last tensor
+// (leading dim ldo > C) — the Inception block writes each branch's BN
+// output straight into its slice of the pre-allocated concat buffer,
+// eliminating the aten cat copy per block.
+template <bool RELU, bool VEC, bool STRIDED>
 __global__ __launch_bounds__(256)
 void bn_apply_kernel(const __bf16* __restrict__ x,
                      const float* __restrict__ mean,
                      const float* __restrict__ invstd,
                      const __bf16* __restrict__ g, const __bf16* __restrict__ b,
-                     __bf16* __restrict__ y, long P, int C) {
+                     __bf16* __restrict__ y, long ldo, FDiv dC,
+                     long P, int C) {
   __shared__ float sc[MAXC], sh[MAXC];
   for (int c = threadIdx.x; c < C; c += 256) {
     const float s = (float)g[c] * invstd[c];
@@ -125,7 +130,9 @@ void bn_apply_kernel(const __bf16* __restrict__ x,
   const long stride = (long)gridDim.x * 256 * 8;
   for (long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8; i < total;
        i += stride) {
-    const int c0 = (int)(i % C);
+    const long prow = STRIDED ? (long)fd((unsigned)i, dC) : 0;
+    const int c0 = STRIDED ? (int)(i - prow * C) : (int)(i % C);
+    __bf16* yout = STRIDED ? y + prow * ldo + c0 : y + i;
     if (VEC && i + 8 <= total) {
       bf16x8 v = *(const bf16x8*)&x[i];
       bf16x8 o;
@@ -135,14 +142,16 @@ void bn_apply_kernel(const __bf16* __restrict__ x,
         if (RELU) f = f > 0.f ? f : 0.f;
         o[j] = (__bf16)f;
       }
-      *(bf16x8*)&y[i] = o;
+      *(bf16x8*)yout = o;
     } else {
       int c = c0;
+      long jrow = prow;
       for (int j = 0; j < 8 && i + j < total; ++j) {
         float f = (float)x[i + j] * sc[c] + sh[c];
         if (RELU) f = f > 0.f ? f : 0.f;
-        y[i + j] = (__bf16)f;
-        if (++c == C) c = 0;
+        if (STRIDED) y[jrow * ldo + c] = (__bf16)f;
+        else y[i + j] = (__bf16)f;
+        if (++c == C) { c = 0; ++jrow; }
       }
     }
   }
@@ -321,8 +330,8 @@ inline unsigned ew_grid(long total) {
 }  // namespace
 
 void launch_bn_fwd(const bf16_t* x, const bf16_t* g, const bf16_t* b,
-                   bf16_t* y, float* mean, float* invstd, float* part,
-                   long P, int C, int Z, float eps, bool relu,
+                   bf16_t* y, long ldo, float* mean, float* invstd,
+                   float* part, long P, int C, int Z, float eps, bool relu,
                    hipStream_t stream) {
   dim3 sg(ceil_div(C, 64), Z), sb(256);
   hipLaunchKernelGGL(bn_stats_kernel, sg, sb, 0, stream, (const __bf16*)x,
@@ -332,12 +341,19 @@ void launch_bn_fwd(const bf16_t* x, const bf16_t* g, const bf16_t* b,
                      eps);
   dim3 ag(ew_grid(P * (long)C)), ab(256);
   const bool vec = (C & 7) == 0;
-#define APPLY(RELUv, VECv)                                                  \
-  hipLaunchKernelGGL((bn_apply_kernel<RELUv, VECv>), ag, ab, 0, stream,     \
-                     (const __bf16*)x, mean, invstd, (const __bf16*)g,       \
-                     (const __bf16*)b, (__bf16*)y, P, C)
-  if (relu) { if (vec) APPLY(true, true); else APPLY(true, false); }
-  else      { if (vec) APPLY(false, true); else APPLY(false, false); }
+  const bool strided = ldo != C;
+  const FDiv dC = make_fd(C);
+#define APPLY(RELUv, VECv, STRv)                                            \
+  hipLaunchKernelGGL((bn_apply_kernel<RELUv, VECv, STRv>), ag, ab, 0,       \
+                     stream, (const __bf16*)x, mean, invstd,                 \
+                     (const __bf16*)g, (const __bf16*)b, (__bf16*)y, ldo,    \
+                     dC, P, C)
+#define APPLY2(RELUv, VECv)                                                 \
+  do { if (strided) APPLY(RELUv, VECv, true);                               \
+       else APPLY(RELUv, VECv, false); } while (0)
+  if (relu) { if (vec) APPLY2(true, true); else APPLY2(true, false); }
+  else      { if (vec) APPLY2(false, true); else APPLY2(false, false); }
+#undef APPLY2
 #undef APPLY
 }
 

@@ -51,8 +51,8 @@ void launch_conv_bwd_weight(const bf16_t*, const bf16_t*, float*,
                             int, int, int, int, hipStream_t);
 int conv_bwdw_slices(int, int, int, int, int, int, int);
 void launch_bn_fwd(const bf16_t*, const bf16_t*, const bf16_t*, bf16_t*,
-                   float*, float*, float*, long, int, int, float, bool,
-                   hipStream_t);
+                   long, float*, float*, float*, long, int, int, float,
+                   bool, hipStream_t);
 void launch_bn_bwd(const bf16_t*, const bf16_t*, long, const bf16_t*,
                    const bf16_t*, const float*, const float*, bf16_t*,
                    bf16_t*, bf16_t*, float*, float*, float*, long, int,
@@ -524,31 +524,6 @@ void conv2d_bwd_weight_out(torch::Tensor dy, torch::Tensor x,
 
 // -------------------------------------------------------------------- bn
 
-std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor g,
-                                  torch::Tensor b, double eps, bool relu) {
-  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && is_cl(x) &&
-              x.scalar_type() == torch::kBFloat16,
-              "x must be bf16 channels-last");
-  const int C = x.size(1);
-  const long P = x.numel() / C;
-  TORCH_CHECK(C <= bn_max_channels(), "bn: C > LDS staging bound");
-  TORCH_CHECK(g.numel() == C && b.numel() == C &&
-              g.scalar_type() == torch::kBFloat16 &&
-              b.scalar_type() == torch::kBFloat16, "g/b must be bf16 [C]");
-  const int Z = bn_stats_slices(P, C);
-  auto opts = x.options().dtype(torch::kFloat32);
-  auto y = torch::empty_like(x);
-  auto mean = torch::empty({C}, opts);
-  auto invstd = torch::empty({C}, opts);
-  auto part = torch::empty({(long)Z * C * 2}, opts);
-  launch_bn_fwd((const bf16_t*)x.data_ptr(), (const bf16_t*)g.data_ptr(),
-                (const bf16_t*)b.data_ptr(), (bf16_t*)y.data_ptr(),
-                mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                part.data_ptr<float>(), P, C, Z, (float)eps, relu,
-                cur_stream());
-  return {y, mean, invstd};
-}
-
 // dy may be a channel-narrow view of a wider channels-last tensor (the
 // backward of torch.cat): detected by its strides and read in place —
 // no contiguous() copy.
@@ -560,6 +535,44 @@ static bool cl_narrow(const torch::Tensor& t, long* ldy) {
     return false;
   *ldy = s[3];
   return true;
+}
+
+std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor g,
+                                  torch::Tensor b, double eps, bool relu,
+                                  torch::Tensor out) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && is_cl(x) &&
+              x.scalar_type() == torch::kBFloat16,
+              "x must be bf16 channels-last");
+  const int C = x.size(1);
+  const long P = x.numel() / C;
+  TORCH_CHECK(C <= bn_max_channels(), "bn: C > LDS staging bound");
+  TORCH_CHECK(g.numel() == C && b.numel() == C &&
+              g.scalar_type() == torch::kBFloat16 &&
+              b.scalar_type() == torch::kBFloat16, "g/b must be bf16 [C]");
+  const int Z = bn_stats_slices(P, C);
+  auto opts = x.options().dtype(torch::kFloat32);
+  // out (optional): a channel-narrow channels-last view — the apply
+  // writes straight into the caller's concat buffer (strided store)
+  long ldo = C;
+  torch::Tensor y;
+  if (out.numel() > 0) {
+    TORCH_CHECK(cl_narrow(out, &ldo) && out.sizes() == x.sizes() &&
+                out.scalar_type() == torch::kBFloat16,
+                "bn_fwd: out must be a bf16 channels-last (or channel-"
+                "narrow) tensor with x's shape");
+    y = out;
+  } else {
+    y = torch::empty_like(x);
+  }
+  auto mean = torch::empty({C}, opts);
+  auto invstd = torch::empty({C}, opts);
+  auto part = torch::empty({(long)Z * C * 2}, opts);
+  launch_bn_fwd((const bf16_t*)x.data_ptr(), (const bf16_t*)g.data_ptr(),
+                (const bf16_t*)b.data_ptr(), (bf16_t*)y.data_ptr(), ldo,
+                mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                part.data_ptr<float>(), P, C, Z, (float)eps, relu,
+                cur_stream());
+  return {y, mean, invstd};
 }
 
 std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
