@@ -606,5 +606,8 @@ def test_fused_cat_block_matches_cat():
     y2, dx2, g2 = run(block, x, fuse=False)
     assert torch.equal(y1, y2)
     assert torch.equal(dx1, dx2)
+    # conv weight grads accumulate with z-sliced fp32 atomics, so the
+    # reduction order (and the final bf16 rounding) varies between
+    # runs — compare within a few ulps instead of bitwise
     for a, b in zip(g1, g2):
-        assert torch.equal(a, b)
+        assert torch.allclose(a.float(), b.float(), rtol=5e-2, atol=1e-6)
