@@ -493,36 +493,45 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
 
 // As[k][p]: dY[n,ho,wo,k] — thread t loads 8 consecutive k (one b128)
 // for one reduction pixel p, transposing into LDS.
+// COLUMN-wise transpose-free staging: thread t owns ONE k row
+// (t & 63) and an 8-pixel chunk ((t >> 6) * 8); the 8 global loads are
+// lane-coalesced across k (consecutive addresses within an
+// instruction) and the LDS store is a single aligned b128 into row k —
+// the same consecutive-row/same-column pattern as the MFMA fragment
+// reads, which the 8-row skew already makes conflict-free. The earlier
+// row-wise variant (one b128 load, 8 transposed b16 stores) serialized
+// on sub-dword LDS write conflicts (PMC: conflict count exceeded the
+// LDS instruction count in this kernel).
 struct DyBwdwStage {
-  int px, kk0, k0;
-  bool vec8, kok;
-  long p;
+  int k, pxc;
+  bool kok;
+  long p;              // first pixel of this thread's chunk
   const __bf16* src;   // advances by BK*K per tile
 
   DEVINL void init(const __bf16* __restrict__ dY, const ConvShape& cs,
                    int m0, long p0, int t, bool kvec) {
-    px = t >> 3;           // 0..31 reduction pixel
-    kk0 = (t & 7) * 8;     // k chunk
-    k0 = m0 + kk0;
-    kok = k0 < cs.K;
-    vec8 = kvec && k0 + 8 <= cs.K;
-    p = p0 + px;
-    src = dY + p * cs.K + k0;
+    (void)kvec;
+    k = t & 63;            // LDS row (k within the tile)
+    pxc = (t >> 6) * 8;    // 8-pixel chunk
+    kok = m0 + k < cs.K;
+    p = p0 + pxc;
+    src = dY + p * cs.K + (m0 + k);
   }
 
   DEVINL void stage(__bf16* Sm, const ConvShape cs, long Ptot) {
-    if (p >= Ptot || !kok) {
+    bf16x8 v = {};
+    if (kok) {
+      const long left = Ptot - p;
+      if (left >= 8) {
+        const __bf16* s = src;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) srow(Sm, kk0 + j)[px] = (__bf16)0.f;
-    } else if (vec8) {
-      bf16x8 v = *(const bf16x8*)src;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) srow(Sm, kk0 + j)[px] = v[j];
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        srow(Sm, kk0 + j)[px] = (k0 + j < cs.K) ? src[j] : (__bf16)0.f;
+        for (int j = 0; j < 8; ++j) { v[j] = *s; s += cs.K; }
+      } else if (left > 0) {
+        const __bf16* s = src;
+        for (int j = 0; j < (int)left; ++j) { v[j] = *s; s += cs.K; }
+      }
     }
+    *(bf16x8*)&srow(Sm, k)[pxc] = v;
     p += BK;
     src += (long)BK * cs.K;
   }
