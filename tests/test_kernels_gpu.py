@@ -343,6 +343,7 @@ def test_native_ext_is_mandatory_on_gpu(monkeypatch):
     (2, 3, 16, 16, 8, 3, 3, 1, 1),
     (2, 8, 15, 15, 16, 3, 3, 2, 0),      # stride 2, odd size
     (2, 16, 8, 8, 32, 1, 1, 1, 0),       # pointwise
+    (2, 16, 48, 48, 24, 1, 1, 1, 0),     # pointwise, P>=4096 (GEMM route)
     (1, 4, 12, 12, 6, 1, 7, 1, 3),       # asymmetric 1x7 (Inception)
     (1, 4, 12, 12, 6, 7, 1, 1, 3),       # asymmetric 7x1 — pad clamps
     (2, 3, 31, 31, 8, 3, 3, 2, 0),       # Inception stem-ish
