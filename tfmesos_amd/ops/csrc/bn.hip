@@ -310,13 +310,16 @@ void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
 
 inline int stats_slices(long P, int C) {
   // target >=2048 workgroups across the (C/64) x Z grid, each slice
-  // covering >=8 pixel rounds of 4 rows
+  // covering >=8 pixel rounds of 4 rows. The cap matters: narrow
+  // layers (C=64 -> one channel column) need Z ~ 2048 to cover the
+  // 256-CU chip; the old 256 cap left them 1 workgroup/CU and the PMC
+  // wait counters dominated everything else in the stats kernels.
   long cb = (C + 63) / 64;
   long want = (2048 + cb - 1) / cb;
   long per = P / (4 * 8);
   long z = want < per ? want : per;
   if (z < 1) z = 1;
-  if (z > 256) z = 256;
+  if (z > 2048) z = 2048;
   return (int)z;
 }
 
