@@ -26,7 +26,8 @@ void avg3x3_kernel(const __bf16* __restrict__ X, __bf16* __restrict__ Y,
 
   if (VEC) {
     const int WC = W * C;
-    for (int i = threadIdx.x * 8; i < WC; i += 256 * 8) {
+    for (int i = ((int)blockIdx.y * 256 + threadIdx.x) * 8; i < WC;
+         i += (int)gridDim.y * 256 * 8) {
       const int w = i / C;          // strip stays inside one w (C%8==0)
       const int c = i - w * C;
       f32x4 s0 = {}, s1 = {};
@@ -55,7 +56,8 @@ void avg3x3_kernel(const __bf16* __restrict__ X, __bf16* __restrict__ Y,
     }
   } else {
     const int WC = W * C;
-    for (int i = threadIdx.x; i < WC; i += 256) {
+    for (int i = (int)blockIdx.y * 256 + threadIdx.x; i < WC;
+         i += (int)gridDim.y * 256) {
       const int w = i / C;
       const int c = i - w * C;
       float s = 0.f;
@@ -77,7 +79,14 @@ void avg3x3_kernel(const __bf16* __restrict__ X, __bf16* __restrict__ Y,
 
 void launch_avg3x3(const bf16_t* X, bf16_t* Y, int N, int H, int W, int C,
                    hipStream_t stream) {
-  dim3 grid((unsigned)(N * H)), block(256);
+  // split each (n,h) row over grid.y so small-spatial deep layers
+  // (8^2 x 2048: N*H = 256 rows = 1 WG/CU) still fill the chip
+  const long rows = (long)N * H;
+  long want = (2048 + rows - 1) / rows;
+  long per = ((long)W * C + 256 * 8 - 1) / (256 * 8);
+  long ychunks = want < per ? want : per;
+  if (ychunks < 1) ychunks = 1;
+  dim3 grid((unsigned)rows, (unsigned)ychunks), block(256);
   if ((C & 7) == 0)
     hipLaunchKernelGGL((avg3x3_kernel<true>), grid, block, 0, stream,
                        (const __bf16*)X, (__bf16*)Y, N, H, W, C);
