@@ -115,7 +115,8 @@ void maxpool3x3s2_fwd_kernel(const __bf16* __restrict__ X,
   unsigned char* oi = idx + ((long)n * Ho + ho) * Wo * C;
   const int WoC = Wo * C;
   const int step = VEC ? 8 : 1;
-  for (int i = threadIdx.x * step; i < WoC; i += 256 * step) {
+  for (int i = ((int)blockIdx.y * 256 + threadIdx.x) * step; i < WoC;
+       i += (int)gridDim.y * 256 * step) {
     const int wo = i / C;
     const int c = i - wo * C;
     const long col0 = (long)wo * 2 * C + c;
@@ -165,7 +166,8 @@ void maxpool3x3s2_bwd_kernel(const __bf16* __restrict__ dY, long ldy,
   const int WC = W * C;
   __bf16* out = dX + ((long)n * H + h) * WC;
   const int step = VEC ? 8 : 1;
-  for (int i = threadIdx.x * step; i < WC; i += 256 * step) {
+  for (int i = ((int)blockIdx.y * 256 + threadIdx.x) * step; i < WC;
+       i += (int)gridDim.y * 256 * step) {
     const int w = i / C;
     const int c = i - w * C;
     // windows (ho, wo) covering (h, w): ho*2 <= h <= ho*2+2
@@ -222,7 +224,12 @@ void maxpool3x3s2_bwd_kernel(const __bf16* __restrict__ dY, long ldy,
 void launch_maxpool3x3s2_fwd(const bf16_t* X, bf16_t* Y, unsigned char* idx,
                              int N, int H, int W, int C, int Ho, int Wo,
                              hipStream_t stream) {
-  dim3 grid((unsigned)(N * Ho)), block(256);
+  const long mrows = (long)N * Ho;
+  long mwant = (2048 + mrows - 1) / mrows;
+  long mper = ((long)Wo * C + 256L * 8 - 1) / (256 * 8);
+  long mych = mwant < mper ? mwant : mper;
+  if (mych < 1) mych = 1;
+  dim3 grid((unsigned)mrows, (unsigned)mych), block(256);
   if ((C & 7) == 0)
     hipLaunchKernelGGL((maxpool3x3s2_fwd_kernel<true>), grid, block, 0,
                        stream, (const __bf16*)X, (__bf16*)Y, idx, N, H, W, C,
@@ -237,7 +244,12 @@ void launch_maxpool3x3s2_bwd(const bf16_t* dY, long ldy,
                              const unsigned char* idx,
                              bf16_t* dX, int N, int H, int W, int C, int Ho,
                              int Wo, hipStream_t stream) {
-  dim3 grid((unsigned)(N * H)), block(256);
+  const long mrows = (long)N * H;
+  long mwant = (2048 + mrows - 1) / mrows;
+  long mper = ((long)W * C + 256L * 8 - 1) / (256 * 8);
+  long mych = mwant < mper ? mwant : mper;
+  if (mych < 1) mych = 1;
+  dim3 grid((unsigned)mrows, (unsigned)mych), block(256);
   if ((C & 7) == 0)
     hipLaunchKernelGGL((maxpool3x3s2_bwd_kernel<true>), grid, block, 0,
                        stream, (const __bf16*)dY, ldy, idx, (__bf16*)dX, N,
