@@ -53,14 +53,13 @@ def main():
         y = ops.conv2d(x, w, stride=st, padding=pads, weight_format="krsc")
         dy = torch.randn_like(y).contiguous(memory_format=torch.channels_last)
         ext = ops._ext()
-        wt = w.permute(3, 1, 2, 0).contiguous()
         flops = 2.0 * N * Ho * Wo * K * C * R * S
 
         f = t_ms(lambda: ext.conv2d_fwd(
             x, w, torch.empty(0, device=dev), st, st, pads[0], pads[1],
             False)) * 1e3
         d = t_ms(lambda: ext.conv2d_bwd_data(
-            dy, wt, H, W, st, st, pads[0], pads[1])) * 1e3
+            dy, w, H, W, st, st, pads[0], pads[1])) * 1e3
         g = t_ms(lambda: ext.conv2d_bwd_weight(
             dy, x, R, S, st, st, pads[0], pads[1])) * 1e3
         print("%-24s %9.1f %9.1f %9.1f | %7.0f %7.0f %7.0f" %
