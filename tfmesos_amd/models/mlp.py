@@ -68,12 +68,33 @@ class MnistMLP(object):
         hid_w, hid_b = p("hid_w"), p("hid_b")
         sm_w, sm_b = p("sm_w"), p("sm_b")
 
-        # fwd hidden layer: split-K GEMM + bias + relu. (A fully fused
-        # single-workgroup classifier head exists — ops.mlp_head_fused —
-        # but exposes only B threads of parallelism and measured slower
-        # than these three MFMA kernels; the composed path is the
-        # default.)
+        # fwd hidden layer: split-K GEMM + bias + relu
         h = ops.gemm_bias_act(x, hid_w, hid_b, act="relu")       # [B,H]
+        if x.is_cuda and B <= 128 and self.hidden <= 128 \
+                and self.classes <= 16:
+            # the whole classifier head in ONE kernel (MFMA rewrite of
+            # mlp_head_fused, csrc/softmax_xent.hip): logits GEMM +
+            # softmax + mean loss + dlogits + relu-masked dh. At this
+            # size each launch is ~5-7 us of execution floor, so the
+            # step's critical path is kernel COUNT — this replaces
+            # three launches with one.
+            loss, dlogits, dh = ops.mlp_head_fused(h, sm_w, sm_b, y)
+            # dW2 is off the critical path: fork it onto a side stream
+            # so it overlaps the dW1 GEMM (hipGraph keeps the fork)
+            if self._side is None:
+                self._side = torch.cuda.Stream()
+                self._ev_fork = torch.cuda.Event()
+                self._ev_join = torch.cuda.Event()
+            self._ev_fork.record()
+            with torch.cuda.stream(self._side):
+                self._ev_fork.wait()
+                ops.gemm_bias_act(h, dlogits, trans_a=True,
+                                  out=g("sm_w"), colsum_out=g("sm_b"))
+                self._ev_join.record()
+            ops.gemm_bias_act(x, dh, trans_a=True, out=g("hid_w"),
+                              colsum_out=g("hid_b"))
+            self._ev_join.wait()
+            return loss
         logits = ops.gemm_bias_act(h, sm_w, sm_b, act="none")
         loss, dlogits = ops.softmax_xent_fused(logits, y)
         dh = ops.gemm_bias_act(dlogits, sm_w, trans_b=True,
