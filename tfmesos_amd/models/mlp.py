@@ -79,21 +79,13 @@ class MnistMLP(object):
             # step's critical path is kernel COUNT — this replaces
             # three launches with one.
             loss, dlogits, dh = ops.mlp_head_fused(h, sm_w, sm_b, y)
-            # dW2 is off the critical path: fork it onto a side stream
-            # so it overlaps the dW1 GEMM (hipGraph keeps the fork)
-            if self._side is None:
-                self._side = torch.cuda.Stream()
-                self._ev_fork = torch.cuda.Event()
-                self._ev_join = torch.cuda.Event()
-            self._ev_fork.record()
-            with torch.cuda.stream(self._side):
-                self._ev_fork.wait()
-                ops.gemm_bias_act(h, dlogits, trans_a=True,
-                                  out=g("sm_w"), colsum_out=g("sm_b"))
-                self._ev_join.record()
+            # (a side-stream fork for dW2 measured SLOWER here: the
+            # event fork/join edges cost more than the 7 us GEMM they
+            # overlap, both eager and inside the captured graph)
+            ops.gemm_bias_act(h, dlogits, trans_a=True, out=g("sm_w"),
+                              colsum_out=g("sm_b"))
             ops.gemm_bias_act(x, dh, trans_a=True, out=g("hid_w"),
                               colsum_out=g("hid_b"))
-            self._ev_join.wait()
             return loss
         logits = ops.gemm_bias_act(h, sm_w, sm_b, act="none")
         loss, dlogits = ops.softmax_xent_fused(logits, y)
