@@ -18,7 +18,7 @@ class GraphedStep(object):
     into a hipGraph after ``warmup`` eager runs; call the instance to
     replay. Falls back to eager on CPU."""
 
-    def __init__(self, fn, warmup=3):
+    def __init__(self, fn, warmup=3, tune_iters=200):
         self.fn = fn
         self.graph = None
         if not torch.cuda.is_available():
@@ -33,6 +33,27 @@ class GraphedStep(object):
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
             fn()
+        # self-tune: replay has a fixed per-graph submission cost, so
+        # for SHORT steps (few kernels) eager can win — measure both
+        # and keep the faster mode (the graph captures state either
+        # way, so dropping it is safe)
+        if tune_iters:
+            import time
+
+            def clock(step):
+                for _ in range(20):
+                    step()
+                torch.cuda.synchronize()
+                t0 = time.perf_counter()
+                for _ in range(tune_iters):
+                    step()
+                torch.cuda.synchronize()
+                return time.perf_counter() - t0
+
+            t_graph = clock(self.graph.replay)
+            t_eager = clock(fn)
+            if t_eager < t_graph * 0.98:
+                self.graph = None
 
     def __call__(self):
         if self.graph is not None:
