@@ -78,12 +78,12 @@ class MnistMLP(object):
             # size each launch is ~5-7 us of execution floor, so the
             # step's critical path is kernel COUNT — this replaces
             # three launches with one.
-            loss, dlogits, dh = ops.mlp_head_fused(h, sm_w, sm_b, y)
-            # (a side-stream fork for dW2 measured SLOWER here: the
-            # event fork/join edges cost more than the 7 us GEMM they
-            # overlap, both eager and inside the captured graph)
-            ops.gemm_bias_act(h, dlogits, trans_a=True, out=g("sm_w"),
-                              colsum_out=g("sm_b"))
+            # dW2/db2 are computed inside the head too (operands are
+            # already in its LDS); a side-stream fork for a separate
+            # dW2 GEMM measured SLOWER (event fork/join edges cost more
+            # than the 7 us GEMM they overlap)
+            loss, dlogits, dh = ops.mlp_head_fused(
+                h, sm_w, sm_b, y, dw2=g("sm_w"), db2=g("sm_b"))
             ops.gemm_bias_act(x, dh, trans_a=True, out=g("hid_w"),
                               colsum_out=g("hid_b"))
             return loss

@@ -551,15 +551,21 @@ def batch_norm_act(x, weight, bias, eps=1e-3, relu=False, out=None):
     return _BatchNormActFn.apply(x, weight, bias, eps, relu, out)
 
 
-def mlp_head_fused(h, w, b, labels, scale=None):
+def mlp_head_fused(h, w, b, labels, scale=None, dw2=None, db2=None):
     """Fused classifier head (one kernel): logits = h@w+b, softmax,
-    mean xent loss, dlogits = (p-onehot)*scale, and dh = (dlogits@w^T)
-    masked by h>0 (h is a relu output). Returns (loss, dlogits, dh).
-    GPU limits C<=16, H<=512, B<=512; CPU reference otherwise."""
+    mean xent loss, dlogits = (p-onehot)*scale, dh = (dlogits@w^T)
+    masked by h>0 (h is a relu output), and — when dw2/db2 (fp32 or
+    bf16 grad views) are given on the MFMA path (B<=128, H<=128) —
+    dW2 = h^T@dlogits and db2 = colsum(dlogits) in the SAME kernel.
+    Returns (loss, dlogits, dh). GPU limits C<=16, H<=512, B<=512;
+    CPU reference otherwise."""
     B = h.shape[0]
     s = float(scale if scale is not None else 1.0 / B)
     if h.is_cuda:
-        return _ext().mlp_head_fused(h, w, b, labels, s)
+        e = torch.empty(0, device=h.device)
+        return _ext().mlp_head_fused(h, w, b, labels, s,
+                                     dw2 if dw2 is not None else e,
+                                     db2 if db2 is not None else e)
     hf, wf = h.float(), w.float()
     logits = hf @ wf + b.float()
     probs = torch.softmax(logits, 1)
@@ -568,7 +574,11 @@ def mlp_head_fused(h, w, b, labels, scale=None):
     d[torch.arange(B), labels] -= 1.0
     d *= s
     dh = (d @ wf.t()) * (hf > 0)
-    return loss, d.to(h.dtype), dh.to(h.dtype)
+    db = d.to(h.dtype)
+    if dw2 is not None:
+        dw2.copy_((hf.t() @ db.float()).reshape(dw2.shape).to(dw2.dtype))
+        db2.copy_(db.float().sum(0).to(db2.dtype))
+    return loss, db, dh.to(h.dtype)
 
 
 class _AvgPool3x3Fn(torch.autograd.Function):

@@ -25,8 +25,8 @@ void launch_softmax_xent_fused(const bf16_t*, const long*, bf16_t*, float*,
 void launch_softmax_xent_bwd(const bf16_t*, const long*, bf16_t*, float,
                              int, int, hipStream_t);
 void launch_mlp_head_fused(const bf16_t*, const bf16_t*, const bf16_t*,
-                           const long*, bf16_t*, bf16_t*, float*, float,
-                           int, int, int, hipStream_t);
+                           const long*, bf16_t*, bf16_t*, float*, void*,
+                           void*, bool, float, int, int, int, hipStream_t);
 void launch_gather_bf16(const bf16_t*, const long*, bf16_t*, long, int, long,
                         hipStream_t);
 void launch_gather_f32(const float*, const long*, float*, long, int, long,
@@ -304,9 +304,13 @@ std::tuple<torch::Tensor, torch::Tensor> softmax_xent_fused(
 
 // fused classifier head: logits=h@w+b, softmax-xent, dlogits, and
 // dh = (dlogits @ w^T) * (h>0), all in ONE single-workgroup kernel
+// dw2/db2 (optional, pass empty to skip): classifier weight/bias grads
+// computed inside the fused kernel (fp32 or bf16 — the flat grad
+// buffer's dtype). Only supported on the MFMA path (B<=128, H<=128).
 std::vector<torch::Tensor> mlp_head_fused(torch::Tensor h, torch::Tensor w,
                                           torch::Tensor b, torch::Tensor labels,
-                                          double scale) {
+                                          double scale, torch::Tensor dw2,
+                                          torch::Tensor db2) {
   TORCH_CHECK(h.is_cuda() && h.dim() == 2 && h.is_contiguous() &&
               h.scalar_type() == torch::kBFloat16, "h must be bf16 [B,H]");
   TORCH_CHECK(w.is_contiguous() && w.dim() == 2 && w.size(0) == h.size(1) &&
@@ -322,12 +326,28 @@ std::vector<torch::Tensor> mlp_head_fused(torch::Tensor h, torch::Tensor w,
   auto dlogits = torch::empty({B, C}, h.options());
   auto dh = torch::empty_like(h);
   auto loss = torch::empty({}, h.options().dtype(torch::kFloat32));
+  void* dw2p = nullptr;
+  void* db2p = nullptr;
+  bool gf32 = true;
+  if (dw2.numel() > 0) {
+    TORCH_CHECK(B <= 128 && H <= 128,
+                "fused dw2 requires the MFMA head path (B<=128, H<=128)");
+    TORCH_CHECK(dw2.is_contiguous() && dw2.numel() == (long)H * C &&
+                db2.is_contiguous() && db2.numel() == C &&
+                dw2.scalar_type() == db2.scalar_type() &&
+                (dw2.scalar_type() == torch::kFloat32 ||
+                 dw2.scalar_type() == torch::kBFloat16),
+                "dw2/db2 must be contiguous fp32 or bf16 [H*C]/[C]");
+    gf32 = dw2.scalar_type() == torch::kFloat32;
+    dw2p = dw2.data_ptr();
+    db2p = db2.data_ptr();
+  }
   launch_mlp_head_fused((const bf16_t*)h.data_ptr(),
                         (const bf16_t*)w.data_ptr(),
                         (const bf16_t*)b.data_ptr(), labels.data_ptr<long>(),
                         (bf16_t*)dlogits.data_ptr(), (bf16_t*)dh.data_ptr(),
-                        loss.data_ptr<float>(), (float)scale, B, H, C,
-                        cur_stream());
+                        loss.data_ptr<float>(), dw2p, db2p, gf32,
+                        (float)scale, B, H, C, cur_stream());
   return {loss, dlogits, dh};
 }
 

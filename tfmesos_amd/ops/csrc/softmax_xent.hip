@@ -253,6 +253,11 @@ constexpr int HB = 128;        // padded B/H tile
 constexpr int HP = 136;        // hs row stride (8-elem pad, 16B aligned)
 constexpr int CP = 16;         // padded class dim
 
+// dw2/db2 (optional): the weight/bias grads of the classifier are
+// computed IN the same kernel (both operands already live in LDS),
+// removing the dW2 GEMM launch from the step entirely. GF32 selects
+// fp32 vs bf16 grad stores (the trainer's flat grad buffer dtype).
+template <bool GF32>
 __global__ __launch_bounds__(256)
 void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
                           const bf16_t* __restrict__ w,
@@ -261,13 +266,16 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
                           bf16_t* __restrict__ dlogits,
                           bf16_t* __restrict__ dh,
                           float* __restrict__ loss_out,
+                          void* __restrict__ dw2v,
+                          void* __restrict__ db2v,
                           float scale, int B, int H, int C) {
   __shared__ __align__(16) __bf16 hs[HB * HP];      // [row][k]
   // 32 rows (not CP): the B-fragment's lane index spans the full
   // 32-wide MFMA tile, so cols C..31 must read zeros, not neighbors
   __shared__ __align__(16) __bf16 wtp[32 * HP];     // [c][k]  (w^T)
   __shared__ __align__(16) __bf16 wpad[HB * CP];    // [hrow][c]
-  __shared__ __align__(16) __bf16 dls[HB * CP];     // [row][c]
+  __shared__ __align__(16) __bf16 dls[HB * 32];     // [row][c] (32 wide:
+                                       // fragment lanes span a full tile)
   __shared__ float ls[HB * CP];                     // logits fp32
   __shared__ float lsum[256];
   const int t = threadIdx.x;
@@ -278,10 +286,8 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   // tensors are tiny: 10 K + 1 K elements)
   for (int i = t; i < HB * HP; i += 256) hs[i] = (__bf16)0.f;
   for (int i = t; i < 32 * HP; i += 256) wtp[i] = (__bf16)0.f;
-  for (int i = t; i < HB * CP; i += 256) {
-    wpad[i] = (__bf16)0.f;
-    dls[i] = (__bf16)0.f;
-  }
+  for (int i = t; i < HB * CP; i += 256) wpad[i] = (__bf16)0.f;
+  for (int i = t; i < HB * 32; i += 256) dls[i] = (__bf16)0.f;
   __syncthreads();
   for (int i = t; i < B * H; i += 256)
     hs[(i / H) * HP + (i % H)] = *(const __bf16*)&h[i];
@@ -333,7 +339,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
     for (int c = 0; c < C; ++c) {
       const float p = e[c] * inv;
       const float d = (p - (c == label ? 1.f : 0.f)) * scale;
-      dls[t * CP + c] = (__bf16)d;
+      dls[t * 32 + c] = (__bf16)d;
       dlogits[(long)t * C + c] = f2bf(d);
       if (c == label) neglogp = -__logf(fmaxf(p, 1e-30f));
     }
@@ -345,7 +351,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   // 32wr..+31, loops the four 32-wide H column tiles; K = C (one MFMA)
   {
     const int arow = wr * 32 + (lane & 31);
-    bf16x8 a = *(const bf16x8*)&dls[arow * CP + ((lane >> 5) << 3)];
+    bf16x8 a = *(const bf16x8*)&dls[arow * 32 + ((lane >> 5) << 3)];
 #pragma unroll
     for (int ct = 0; ct < HB / 32; ++ct) {
       bf16x8 bv =
@@ -369,6 +375,42 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
     }
   }
 
+  // dW2[H,C] = h^T @ dls (K = B, one 32-row tile per wave) — both
+  // operands are in LDS already, so the step's separate dW2 GEMM
+  // launch disappears. Transposed A reads are scalar gathers (tiny).
+  if (dw2v != nullptr) {
+    f32x16v acc = {};
+#pragma unroll
+    for (int kh = 0; kh < HB / 16; ++kh) {
+      const int k0 = kh * 16 + ((lane >> 5) << 3);
+      bf16x8 a, bv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        a[j] = hs[(k0 + j) * HP + (wr * 32 + (lane & 31))];
+        bv[j] = dls[(k0 + j) * 32 + (lane & 31)];
+      }
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, bv, acc, 0, 0, 0);
+    }
+    const int c = lane & 31;
+    if (c < C) {
+#pragma unroll
+      for (int v = 0; v < 16; ++v) {
+        const int hrow = wr * 32 + ((v >> 2) << 3) + ((lane >> 5) << 2) +
+                         (v & 3);
+        if (hrow < H) {
+          if (GF32) ((float*)dw2v)[hrow * C + c] = acc[v];
+          else ((bf16_t*)dw2v)[hrow * C + c] = f2bf(acc[v]);
+        }
+      }
+    }
+    if (t < C) {
+      float s = 0.f;
+      for (int b2 = 0; b2 < B; ++b2) s += (float)dls[b2 * 32 + t];
+      if (GF32) ((float*)db2v)[t] = s;
+      else ((bf16_t*)db2v)[t] = f2bf(s);
+    }
+  }
+
   // mean loss
 #pragma unroll
   for (int s = 128; s > 0; s >>= 1) {
@@ -383,12 +425,18 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
 void launch_mlp_head_fused(const bf16_t* h, const bf16_t* w,
                            const bf16_t* bias, const long* labels,
                            bf16_t* dlogits, bf16_t* dh, float* loss,
+                           void* dw2, void* db2, bool grads_f32,
                            float scale, int B, int H, int C,
                            hipStream_t stream) {
   if (B <= 128 && H <= 128 && C <= 16) {
-    hipLaunchKernelGGL(mlp_head_mfma_kernel, dim3(1), dim3(256), 0, stream,
-                       h, w, bias, labels, dlogits, dh, loss, scale,
-                       B, H, C);
+    if (grads_f32)
+      hipLaunchKernelGGL((mlp_head_mfma_kernel<true>), dim3(1), dim3(256),
+                         0, stream, h, w, bias, labels, dlogits, dh, loss,
+                         dw2, db2, scale, B, H, C);
+    else
+      hipLaunchKernelGGL((mlp_head_mfma_kernel<false>), dim3(1), dim3(256),
+                         0, stream, h, w, bias, labels, dlogits, dh, loss,
+                         dw2, db2, scale, B, H, C);
     return;
   }
   hipLaunchKernelGGL(mlp_head_fused_kernel, dim3(1), dim3(512), 0, stream,
