@@ -619,3 +619,35 @@ def test_fused_cat_block_matches_cat():
     # runs — compare within a few ulps instead of bitwise
     for a, b in zip(g1, g2):
         assert torch.allclose(a.float(), b.float(), rtol=5e-2, atol=1e-6)
+
+
+def test_mlp_fwd_head_fused_matches_composed():
+    """Two-kernel fwd+head (split-K stripes consumed by the MFMA head)
+    == the composed pipeline, including the in-kernel dW2/db2 grads."""
+    from tfmesos_amd import ops
+    torch.manual_seed(51)
+    B, K, H, C = 100, 784, 100, 10
+    x = bf(torch.rand(B, K))
+    w1 = bf(torch.randn(K, H) * 0.03)
+    b1 = bf(torch.randn(H) * 0.1)
+    w2 = bf(torch.randn(H, C) * 0.1)
+    b2 = bf(torch.randn(C) * 0.1)
+    y = torch.randint(0, C, (B,), device=DEV)
+    dw2 = torch.zeros(H, C, device=DEV, dtype=torch.float32)
+    db2 = torch.zeros(C, device=DEV, dtype=torch.float32)
+    loss, dh = ops.mlp_fwd_head_fused(x, w1, b1, w2, b2, y,
+                                      dw2=dw2, db2=db2)
+
+    h = ops.gemm_bias_act(x, w1, b1, act="relu")
+    logits = ops.gemm_bias_act(h, w2, b2)
+    loss2, dl2 = ops.softmax_xent_fused(logits, y)
+    dh2 = ops.gemm_bias_act(dl2, w2, trans_b=True, act="relu_bwd", aux=h)
+    dw2r = torch.zeros(H, C, device=DEV, dtype=torch.float32)
+    db2r = torch.zeros(C, device=DEV, dtype=torch.float32)
+    ops.gemm_bias_act(h, dl2, trans_a=True, out=dw2r, colsum_out=db2r)
+
+    assert abs(float(loss) - float(loss2)) < 1e-2, (float(loss),
+                                                    float(loss2))
+    assert (dh.float() - dh2.float()).abs().max() < 3e-3
+    assert (dw2 - dw2r).abs().max() < 3e-3 * dw2r.abs().max() + 1e-4
+    assert (db2 - db2r).abs().max() < 3e-3 * db2r.abs().max() + 1e-4

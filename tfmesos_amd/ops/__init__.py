@@ -581,6 +581,28 @@ def mlp_head_fused(h, w, b, labels, scale=None, dw2=None, db2=None):
     return loss, db, dh.to(h.dtype)
 
 
+def mlp_fwd_head_fused(x, w1, b1, w2, b2, labels, scale=None,
+                       dw2=None, db2=None):
+    """Whole MLP fwd + classifier head in TWO kernels (GPU): split-K
+    GEMM stripes for h = relu(x@w1+b1), then the fused MFMA head
+    consumes the stripes directly — h never exists in global memory.
+    Returns (loss, dh); dw2/db2 classifier grads are written by the
+    head when given. Limits: B<=128, H<=128 (H%4==0), C<=16.
+    CPU: composed fp32 reference."""
+    B = x.shape[0]
+    s = float(scale if scale is not None else 1.0 / B)
+    if x.is_cuda:
+        e = torch.empty(0, device=x.device)
+        return _ext().mlp_fwd_head_fused(
+            x, w1, b1, w2, b2, labels, s,
+            dw2 if dw2 is not None else e,
+            db2 if db2 is not None else e)
+    h = torch.relu(x.float() @ w1.float() + b1.float()).to(x.dtype)
+    loss, _, dh = mlp_head_fused(h, w2, b2, labels, scale=s,
+                                 dw2=dw2, db2=db2)
+    return loss, dh
+
+
 class _AvgPool3x3Fn(torch.autograd.Function):
     """3x3 stride-1 pad-1 average pool (the Inception block pool) on a
     channels-last HIP stencil kernel; the stencil is symmetric so the

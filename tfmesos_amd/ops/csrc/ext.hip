@@ -27,6 +27,13 @@ void launch_softmax_xent_bwd(const bf16_t*, const long*, bf16_t*, float,
 void launch_mlp_head_fused(const bf16_t*, const bf16_t*, const bf16_t*,
                            const long*, bf16_t*, bf16_t*, float*, void*,
                            void*, bool, float, int, int, int, hipStream_t);
+void launch_mlp_fwd_head(const float*, int, const bf16_t*, const bf16_t*,
+                         const bf16_t*, const long*, bf16_t*, bf16_t*,
+                         float*, void*, void*, bool, float, int, int, int,
+                         hipStream_t);
+void launch_gemm_stripes(const bf16_t*, const bf16_t*, float*, int, int,
+                         int, int, int, int, int, int, int, int,
+                         hipStream_t);
 void launch_gather_bf16(const bf16_t*, const long*, bf16_t*, long, int, long,
                         hipStream_t);
 void launch_gather_f32(const float*, const long*, float*, long, int, long,
@@ -349,6 +356,93 @@ std::vector<torch::Tensor> mlp_head_fused(torch::Tensor h, torch::Tensor w,
                         loss.data_ptr<float>(), dw2p, db2p, gf32,
                         (float)scale, B, H, C, cur_stream());
   return {loss, dlogits, dh};
+}
+
+// Whole mnist fwd+head in TWO kernels: split-K GEMM stripes for
+// h = relu(x @ w1 + b1), then the fused head consumes the stripes
+// directly (h never exists in global memory). Returns (loss, dh);
+// dw2/db2 grads are written by the head. Falls back to a materialized
+// h + the normal head when split-K does not engage.
+std::vector<torch::Tensor> mlp_fwd_head_fused(
+    torch::Tensor x, torch::Tensor w1, torch::Tensor b1, torch::Tensor w2,
+    torch::Tensor b2, torch::Tensor labels, double scale,
+    torch::Tensor dw2, torch::Tensor db2) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous() &&
+              x.scalar_type() == torch::kBFloat16, "x must be bf16 [B,K]");
+  TORCH_CHECK(w1.is_contiguous() && w1.dim() == 2 &&
+              w1.size(0) == x.size(1) &&
+              w1.scalar_type() == torch::kBFloat16, "w1 must be bf16 [K,H]");
+  const int B = x.size(0), K1 = x.size(1), H = w1.size(1);
+  const int C = w2.size(1);
+  TORCH_CHECK(B <= 128 && H <= 128 && (H & 3) == 0 && C <= 16,
+              "mlp_fwd_head_fused limits: B<=128, H<=128 (H%4==0), C<=16");
+  TORCH_CHECK(b1.scalar_type() == torch::kBFloat16 && b1.numel() == H);
+  TORCH_CHECK(w2.is_contiguous() && w2.size(0) == H &&
+              w2.scalar_type() == torch::kBFloat16, "w2 must be bf16 [H,C]");
+  TORCH_CHECK(b2.scalar_type() == torch::kBFloat16 && b2.numel() == C);
+  TORCH_CHECK(labels.scalar_type() == torch::kInt64 && labels.numel() == B);
+  auto dlogits = torch::empty({B, C}, x.options());
+  auto dh = torch::empty({B, H}, x.options());
+  auto loss = torch::empty({}, x.options().dtype(torch::kFloat32));
+  void* dw2p = nullptr;
+  void* db2p = nullptr;
+  bool gf32 = true;
+  if (dw2.numel() > 0) {
+    TORCH_CHECK(dw2.is_contiguous() && dw2.numel() == (long)H * C &&
+                db2.is_contiguous() && db2.numel() == C &&
+                dw2.scalar_type() == db2.scalar_type() &&
+                (dw2.scalar_type() == torch::kFloat32 ||
+                 dw2.scalar_type() == torch::kBFloat16),
+                "dw2/db2 must be contiguous fp32 or bf16");
+    gf32 = dw2.scalar_type() == torch::kFloat32;
+    dw2p = dw2.data_ptr();
+    db2p = db2.data_ptr();
+  }
+  // same split-K policy as gemm_bias_act_out
+  const int nx = (H + 63) / 64, ny = (B + 63) / 64;
+  int nslice = 1, kc = 0;
+  if (nx * ny < 64 && K1 >= 256) {
+    int want = std::min(K1 / 64, 256 / (nx * ny));
+    if (want > 16) want = 16;
+    if (want > 1) {
+      kc = ((K1 + want - 1) / want + 31) / 32 * 32;
+      nslice = (K1 + kc - 1) / kc;
+    }
+  }
+  if (nslice > 1) {
+    int* cnt_unused;
+    float* ws = splitk_ws(x.device(), (long)B * H * nslice,
+                          (long)nx * ny, &cnt_unused);
+    launch_gemm_stripes((const bf16_t*)x.data_ptr(),
+                        (const bf16_t*)w1.data_ptr(), ws, kc, nslice,
+                        B, H, K1, K1, H, H,
+                        vec_level(x.data_ptr(), K1),
+                        vec_level(w1.data_ptr(), H), cur_stream());
+    launch_mlp_fwd_head(ws, nslice, (const bf16_t*)b1.data_ptr(),
+                        (const bf16_t*)w2.data_ptr(),
+                        (const bf16_t*)b2.data_ptr(),
+                        labels.data_ptr<long>(),
+                        (bf16_t*)dlogits.data_ptr(),
+                        (bf16_t*)dh.data_ptr(), loss.data_ptr<float>(),
+                        dw2p, db2p, gf32, (float)scale, B, H, C,
+                        cur_stream());
+    return {loss, dh};
+  }
+  // no split-K at this shape: materialize h, run the normal head
+  auto h = torch::empty({B, H}, x.options());
+  launch_gemm((const bf16_t*)x.data_ptr(), (const bf16_t*)w1.data_ptr(),
+              b1.data_ptr(), true, h.data_ptr(), false, nullptr, nullptr,
+              nullptr, nullptr, 0, 1, B, H, K1, K1, H, H, false, false, 1,
+              vec_level(x.data_ptr(), K1), vec_level(w1.data_ptr(), H),
+              cur_stream());
+  launch_mlp_head_fused((const bf16_t*)h.data_ptr(),
+                        (const bf16_t*)w2.data_ptr(),
+                        (const bf16_t*)b2.data_ptr(),
+                        labels.data_ptr<long>(),
+                        (bf16_t*)dlogits.data_ptr(), (bf16_t*)dh.data_ptr(),
+                        loss.data_ptr<float>(), dw2p, db2p, gf32,
+                        (float)scale, B, H, C, cur_stream());
+  return {loss, dh};
 }
 
 torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
@@ -691,6 +785,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "out (bf16 or fp32), split-K for deep skinny shapes");
   m.def("softmax_xent_fused", &softmax_xent_fused);
   m.def("mlp_head_fused", &mlp_head_fused);
+  m.def("mlp_fwd_head_fused", &mlp_fwd_head_fused);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("conv2d_fwd", &conv2d_fwd, "implicit-GEMM conv fwd (MFMA, bf16)");

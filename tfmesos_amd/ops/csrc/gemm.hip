@@ -315,3 +315,21 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
 #undef DISP_OUT
 #undef LAUNCH
 }
+
+// split-K phase 1 ONLY: store the fp32 partial stripes and return —
+// the caller fuses the stripe reduction into a consumer kernel (the
+// mnist fused head reduces them while staging h into LDS, so h never
+// exists in global memory and the reduce launch disappears).
+void launch_gemm_stripes(const bf16_t* A, const bf16_t* B, float* ws,
+                         int kc, int nslice, int M, int N, int K, int lda,
+                         int ldb, int ldc, int veca, int vecb,
+                         hipStream_t stream) {
+  dim3 grid(ceil_div(N, BN), ceil_div(M, BM), nslice);
+  dim3 block(256);
+  hipLaunchKernelGGL((gemm_kernel<false, false, 0, false, false, true,
+                                  false>),
+                     grid, block, 0, stream, (const __bf16*)A,
+                     (const __bf16*)B, nullptr, false, nullptr, nullptr,
+                     nullptr, ws, nullptr, M, N, K, lda, ldb, ldc, kc,
+                     veca, vecb);
+}

@@ -257,9 +257,16 @@ constexpr int CP = 16;         // padded class dim
 // computed IN the same kernel (both operands already live in LDS),
 // removing the dW2 GEMM launch from the step entirely. GF32 selects
 // fp32 vs bf16 grad stores (the trainer's flat grad buffer dtype).
-template <bool GF32>
+// FROMWS: h arrives as the fwd GEMM's fp32 split-K stripes (wsrc,
+// nslice of them, [B*H] each) instead of a materialized tensor — the
+// stripe reduce + bias1 + relu happen while staging hs into LDS, so h
+// never touches global memory and the separate reduce kernel is gone.
+// Requires H % 4 == 0 (f32x4 chunks stay within one row).
+template <bool GF32, bool FROMWS>
 __global__ __launch_bounds__(256)
 void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
+                          const float* __restrict__ wsrc, int nslice,
+                          const bf16_t* __restrict__ bias1,
                           const bf16_t* __restrict__ w,
                           const bf16_t* __restrict__ bias,
                           const long* __restrict__ labels,
@@ -289,8 +296,27 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   for (int i = t; i < HB * CP; i += 256) wpad[i] = (__bf16)0.f;
   for (int i = t; i < HB * 32; i += 256) dls[i] = (__bf16)0.f;
   __syncthreads();
-  for (int i = t; i < B * H; i += 256)
-    hs[(i / H) * HP + (i % H)] = *(const __bf16*)&h[i];
+  if (FROMWS) {
+    const int BH = B * H;
+    for (int i = t * 4; i < BH; i += 256 * 4) {
+      f32x4 v = {};
+      for (int z = 0; z < nslice; ++z) {
+        const f32x4 sv = *(const f32x4*)&wsrc[(long)z * BH + i];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) v[j] += sv[j];
+      }
+      const int row = i / H;
+      const int col = i - row * H;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float f = v[j] + (float)bias1[col + j];
+        hs[row * HP + col + j] = (__bf16)(f > 0.f ? f : 0.f);
+      }
+    }
+  } else {
+    for (int i = t; i < B * H; i += 256)
+      hs[(i / H) * HP + (i % H)] = *(const __bf16*)&h[i];
+  }
   for (int i = t; i < H * C; i += 256) {
     const int hr = i / C, c = i - hr * C;
     const __bf16 v = *(const __bf16*)&w[i];
@@ -430,15 +456,37 @@ void launch_mlp_head_fused(const bf16_t* h, const bf16_t* w,
                            hipStream_t stream) {
   if (B <= 128 && H <= 128 && C <= 16) {
     if (grads_f32)
-      hipLaunchKernelGGL((mlp_head_mfma_kernel<true>), dim3(1), dim3(256),
-                         0, stream, h, w, bias, labels, dlogits, dh, loss,
-                         dw2, db2, scale, B, H, C);
+      hipLaunchKernelGGL((mlp_head_mfma_kernel<true, false>), dim3(1),
+                         dim3(256), 0, stream, h, nullptr, 0, nullptr, w,
+                         bias, labels, dlogits, dh, loss, dw2, db2, scale,
+                         B, H, C);
     else
-      hipLaunchKernelGGL((mlp_head_mfma_kernel<false>), dim3(1), dim3(256),
-                         0, stream, h, w, bias, labels, dlogits, dh, loss,
-                         dw2, db2, scale, B, H, C);
+      hipLaunchKernelGGL((mlp_head_mfma_kernel<false, false>), dim3(1),
+                         dim3(256), 0, stream, h, nullptr, 0, nullptr, w,
+                         bias, labels, dlogits, dh, loss, dw2, db2, scale,
+                         B, H, C);
     return;
   }
   hipLaunchKernelGGL(mlp_head_fused_kernel, dim3(1), dim3(512), 0, stream,
                      h, w, bias, labels, dlogits, dh, loss, scale, B, H, C);
+}
+
+// head fed by the fwd GEMM's split-K stripes (see FROMWS above);
+// bounds checked by the binding: B<=128, H<=128 (H%4==0), C<=16
+void launch_mlp_fwd_head(const float* ws, int nslice, const bf16_t* b1,
+                         const bf16_t* w, const bf16_t* bias,
+                         const long* labels, bf16_t* dlogits, bf16_t* dh,
+                         float* loss, void* dw2, void* db2, bool grads_f32,
+                         float scale, int B, int H, int C,
+                         hipStream_t stream) {
+  if (grads_f32)
+    hipLaunchKernelGGL((mlp_head_mfma_kernel<true, true>), dim3(1),
+                       dim3(256), 0, stream, nullptr, ws, nslice, b1, w,
+                       bias, labels, dlogits, dh, loss, dw2, db2, scale,
+                       B, H, C);
+  else
+    hipLaunchKernelGGL((mlp_head_mfma_kernel<false, true>), dim3(1),
+                       dim3(256), 0, stream, nullptr, ws, nslice, b1, w,
+                       bias, labels, dlogits, dh, loss, dw2, db2, scale,
+                       B, H, C);
 }
