@@ -68,25 +68,28 @@ class MnistMLP(object):
         hid_w, hid_b = p("hid_w"), p("hid_b")
         sm_w, sm_b = p("sm_w"), p("sm_b")
 
+        # fwd hidden layer: split-K GEMM + bias + relu
+        h = ops.gemm_bias_act(x, hid_w, hid_b, act="relu")       # [B,H]
         if x.is_cuda and B <= 128 and self.hidden <= 128 \
-                and self.hidden % 4 == 0 and self.classes <= 16:
-            # the fwd GEMM emits split-K stripes and the fused MFMA
-            # head (csrc/softmax_xent.hip) consumes them directly:
-            # stripe-reduce + bias1 + relu + logits GEMM + softmax +
-            # loss + dlogits + relu-masked dh + dW2/db2 — FOUR kernel
-            # launches per step total (stripes, head, dW1, apply). At
-            # this size each launch is ~4-7 us of execution floor, so
-            # kernel COUNT is the step time. (A side-stream fork for a
-            # separate dW2 GEMM measured SLOWER: the event fork/join
-            # edges cost more than the 7 us GEMM they overlap.)
-            loss, dh = ops.mlp_fwd_head_fused(
-                x, hid_w, hid_b, sm_w, sm_b, y,
-                dw2=g("sm_w"), db2=g("sm_b"))
+                and self.classes <= 16:
+            # the whole classifier head in ONE MFMA kernel
+            # (csrc/softmax_xent.hip): logits GEMM + softmax + loss +
+            # dlogits + relu-masked dh + dW2/db2. At this size each
+            # launch is ~4-7 us of execution floor, so kernel COUNT is
+            # the step time: 5 launches total (fwd GEMM stripes, its
+            # reduce, head, dW1, apply). Two measured dead ends kept
+            # out: forking dW2 to a side stream (event edges cost more
+            # than the GEMM they overlap), and consuming the fwd
+            # split-K stripes directly in the head
+            # (ops.mlp_fwd_head_fused — a SINGLE workgroup pulling
+            # 360 KB of stripes from 8 XCDs' L2s took ~36 us; one-WG
+            # consumers must read tiny inputs, so the many-WG reduce
+            # kernel stays).
+            loss, dlogits, dh = ops.mlp_head_fused(
+                h, sm_w, sm_b, y, dw2=g("sm_w"), db2=g("sm_b"))
             ops.gemm_bias_act(x, dh, trans_a=True, out=g("hid_w"),
                               colsum_out=g("hid_b"))
             return loss
-        # fwd hidden layer: split-K GEMM + bias + relu
-        h = ops.gemm_bias_act(x, hid_w, hid_b, act="relu")       # [B,H]
         logits = ops.gemm_bias_act(h, sm_w, sm_b, act="none")
         loss, dlogits = ops.softmax_xent_fused(logits, y)
         dh = ops.gemm_bias_act(dlogits, sm_w, trans_b=True,
