@@ -348,7 +348,12 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   }
   __syncthreads();
 
-  // rowwise softmax + dlogits (one thread per row, C <= 16 scalar)
+  // rowwise softmax + dlogits (one thread per row, C <= 16 scalar).
+  // Two-workgroup split: WG0 owns {loss, dlogits store, dh}, WG1 owns
+  // {dW2, db2} — both recompute the cheap logits/softmax (22 KB of
+  // reads) so the two expensive output phases run in parallel.
+  const bool wg0 = blockIdx.x == 0;
+  const bool wg1 = blockIdx.x + 1 == gridDim.x;
   float neglogp = 0.f;
   if (t < B) {
     const float* lr = &ls[t * CP];
@@ -366,7 +371,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
       const float p = e[c] * inv;
       const float d = (p - (c == label ? 1.f : 0.f)) * scale;
       dls[t * 32 + c] = (__bf16)d;
-      dlogits[(long)t * C + c] = f2bf(d);
+      if (wg0) dlogits[(long)t * C + c] = f2bf(d);
       if (c == label) neglogp = -__logf(fmaxf(p, 1e-30f));
     }
   }
@@ -375,7 +380,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
 
   // dh[B,H] = dls @ w^T, relu-masked by h>0: wave wr owns rows
   // 32wr..+31, loops the four 32-wide H column tiles; K = C (one MFMA)
-  {
+  if (wg0) {
     const int arow = wr * 32 + (lane & 31);
     bf16x8 a = *(const bf16x8*)&dls[arow * 32 + ((lane >> 5) << 3)];
 #pragma unroll
@@ -404,7 +409,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   // dW2[H,C] = h^T @ dls (K = B, one 32-row tile per wave) — both
   // operands are in LDS already, so the step's separate dW2 GEMM
   // launch disappears. Transposed A reads are scalar gathers (tiny).
-  if (dw2v != nullptr) {
+  if (wg1 && dw2v != nullptr) {
     f32x16v acc = {};
 #pragma unroll
     for (int kh = 0; kh < HB / 16; ++kh) {
@@ -438,6 +443,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   }
 
   // mean loss
+  if (!wg0) return;
 #pragma unroll
   for (int s = 128; s > 0; s >>= 1) {
     if (t < s) lsum[t] += lsum[t + s];
@@ -455,13 +461,14 @@ void launch_mlp_head_fused(const bf16_t* h, const bf16_t* w,
                            float scale, int B, int H, int C,
                            hipStream_t stream) {
   if (B <= 128 && H <= 128 && C <= 16) {
+    dim3 hg(dw2 != nullptr ? 2 : 1);
     if (grads_f32)
-      hipLaunchKernelGGL((mlp_head_mfma_kernel<true, false>), dim3(1),
+      hipLaunchKernelGGL((mlp_head_mfma_kernel<true, false>), hg,
                          dim3(256), 0, stream, h, nullptr, 0, nullptr, w,
                          bias, labels, dlogits, dh, loss, dw2, db2, scale,
                          B, H, C);
     else
-      hipLaunchKernelGGL((mlp_head_mfma_kernel<false, false>), dim3(1),
+      hipLaunchKernelGGL((mlp_head_mfma_kernel<false, false>), hg,
                          dim3(256), 0, stream, h, nullptr, 0, nullptr, w,
                          bias, labels, dlogits, dh, loss, dw2, db2, scale,
                          B, H, C);
@@ -479,13 +486,14 @@ void launch_mlp_fwd_head(const float* ws, int nslice, const bf16_t* b1,
                          float* loss, void* dw2, void* db2, bool grads_f32,
                          float scale, int B, int H, int C,
                          hipStream_t stream) {
+  dim3 hg(dw2 != nullptr ? 2 : 1);
   if (grads_f32)
-    hipLaunchKernelGGL((mlp_head_mfma_kernel<true, true>), dim3(1),
+    hipLaunchKernelGGL((mlp_head_mfma_kernel<true, true>), hg,
                        dim3(256), 0, stream, nullptr, ws, nslice, b1, w,
                        bias, labels, dlogits, dh, loss, dw2, db2, scale,
                        B, H, C);
   else
-    hipLaunchKernelGGL((mlp_head_mfma_kernel<false, true>), dim3(1),
+    hipLaunchKernelGGL((mlp_head_mfma_kernel<false, true>), hg,
                        dim3(256), 0, stream, nullptr, ws, nslice, b1, w,
                        bias, labels, dlogits, dh, loss, dw2, db2, scale,
                        B, H, C);
