@@ -228,12 +228,13 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
     colsum_p = colsum_out.data_ptr();
   }
   // split-K when the plain tile grid can't feed the 256-CU chip and K
-  // has enough depth to slice
+  // has enough depth to slice (any transpose combo; not with the
+  // fused colsum/relu_bwd epilogues — those stay single-phase)
   const int nx = (N + 63) / 64, ny = (M + 63) / 64;
   int nslice = 1, kc = 0;
   float* ws = nullptr;
   int* cnt = nullptr;
-  if (!trans_a && !trans_b && act != 2 && nx * ny < 64 && Ka >= 256) {
+  if (act != 2 && colsum_p == nullptr && nx * ny < 256 && Ka >= 256) {
     int want = std::min((int)(Ka / 64), 256 / (nx * ny));
     if (want > 16) want = 16;
     if (want > 1) {

@@ -290,8 +290,14 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
     return;
   }
   if (sk) {
-    // phase 1: partials (epilogue template args unused in SK stores)
-    LAUNCH(false, false, 0, false, false, true, false);
+    // phase 1: partials (epilogue template args unused in SK stores);
+    // all four transpose combos (deep-K BACKWARD shapes with small
+    // tile grids — e.g. NMF dW [1000,200,k=1000] — were 1 wave/SIMD
+    // and LDS-latency-bound without slicing)
+    if (ta) { if (tb) LAUNCH(true, true, 0, false, false, true, false);
+              else    LAUNCH(true, false, 0, false, false, true, false); }
+    else    { if (tb) LAUNCH(false, true, 0, false, false, true, false);
+              else    LAUNCH(false, false, 0, false, false, true, false); }
     // phase 2: fixed-order stripe reduce + fused epilogue
     const long mn = (long)M * ldc;
     dim3 rgrid((unsigned)((mn / 4 + 255) / 256)), rblock(256);
