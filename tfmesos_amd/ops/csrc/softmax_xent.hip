@@ -289,12 +289,19 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   const int lane = t & 63;
   const int wr = t >> 6;
 
-  // zero the padded operand tiles, then stage h and w (scalar — the
-  // tensors are tiny: 10 K + 1 K elements)
-  for (int i = t; i < HB * HP; i += 256) hs[i] = (__bf16)0.f;
-  for (int i = t; i < 32 * HP; i += 256) wtp[i] = (__bf16)0.f;
-  for (int i = t; i < HB * CP; i += 256) wpad[i] = (__bf16)0.f;
-  for (int i = t; i < HB * 32; i += 256) dls[i] = (__bf16)0.f;
+  // zero the padded operand tiles (b128 stores — the scalar version
+  // was ~28K LDS writes), then stage h and w (scalar; tiny tensors)
+  {
+    const bf16x8 z8 = {};
+    for (int i = t * 8; i < HB * HP; i += 256 * 8)
+      *(bf16x8*)&hs[i] = z8;
+    for (int i = t * 8; i < 32 * HP; i += 256 * 8)
+      *(bf16x8*)&wtp[i] = z8;
+    for (int i = t * 8; i < HB * CP; i += 256 * 8)
+      *(bf16x8*)&wpad[i] = z8;
+    for (int i = t * 8; i < HB * 32; i += 256 * 8)
+      *(bf16x8*)&dls[i] = z8;
+  }
   __syncthreads();
   if (FROMWS) {
     const int BH = B * H;
