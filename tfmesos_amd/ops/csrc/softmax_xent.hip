@@ -321,8 +321,20 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
       }
     }
   } else {
-    for (int i = t; i < B * H; i += 256)
-      hs[(i / H) * HP + (i % H)] = *(const __bf16*)&h[i];
+    // b128 global loads over the flat [B*H] tensor; the 8-chunk can
+    // cross one row boundary (H >= 8), handled by the carry
+    const int BH = B * H;
+    for (int i = t * 8; i < BH; i += 256 * 8) {
+      bf16x8 v = {};
+      if (i + 8 <= BH) v = *(const bf16x8*)&h[i];
+      else for (int j = 0; i + j < BH; ++j) v[j] = *(const __bf16*)&h[i + j];
+      int row = i / H, col = i - (i / H) * H;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        hs[row * HP + col] = v[j];
+        if (++col == H) { col = 0; ++row; }
+      }
+    }
   }
   for (int i = t; i < H * C; i += 256) {
     const int hr = i / C, c = i - hr * C;
