@@ -5,8 +5,9 @@ Same architecture/hyperparameters as the reference benchmark workload
 stddev 1/sqrt(784), hidden 100, batch 100, lr 0.01). Forward AND backward
 are written explicitly against ``tfmesos_amd.ops`` (bf16 MFMA GEMM +
 fused softmax-xent + fused relu-bwd on GPU) — no autograd in the hot
-path, so the step is a short fixed kernel sequence that hipGraph can
-capture.
+path, so the step is a short fixed kernel sequence (5 kernels at the
+benchmark geometry; GraphedStep can capture it and self-tunes
+graph-vs-eager).
 """
 
 import math
@@ -24,9 +25,6 @@ class MnistMLP(object):
         self.inputs = image_pixels
         self.classes = classes
         self.seed = seed
-        self._side = None       # lazy side stream for the dW2 branch
-        self._ev_fork = None
-        self._ev_join = None
 
     def param_specs(self):
         return [
