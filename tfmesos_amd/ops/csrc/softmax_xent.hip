@@ -281,8 +281,11 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   // 32-wide MFMA tile, so cols C..31 must read zeros, not neighbors
   __shared__ __align__(16) __bf16 wtp[32 * HP];     // [c][k]  (w^T)
   __shared__ __align__(16) __bf16 wpad[HB * CP];    // [hrow][c]
-  __shared__ __align__(16) __bf16 dls[HB * 32];     // [row][c] (32 wide:
-                                       // fragment lanes span a full tile)
+  __shared__ __align__(16) __bf16 dls[HB * CP];     // [row][c]
+  // transposed copy for the dW2 B-fragments: [c][b] rows make the
+  // fragment reads aligned b128 instead of 8 latency-exposed scalar
+  // gathers per MFMA step (32 rows: fragment lanes span a full tile)
+  __shared__ __align__(16) __bf16 dlsT[32 * HP];
   __shared__ float ls[HB * CP];                     // logits fp32
   __shared__ float lsum[256];
   const int t = threadIdx.x;
@@ -299,8 +302,10 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
       *(bf16x8*)&wtp[i] = z8;
     for (int i = t * 8; i < HB * CP; i += 256 * 8)
       *(bf16x8*)&wpad[i] = z8;
-    for (int i = t * 8; i < HB * 32; i += 256 * 8)
+    for (int i = t * 8; i < HB * CP; i += 256 * 8)
       *(bf16x8*)&dls[i] = z8;
+    for (int i = t * 8; i < 32 * HP; i += 256 * 8)
+      *(bf16x8*)&dlsT[i] = z8;
   }
   __syncthreads();
   if (FROMWS) {
@@ -389,7 +394,8 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
     for (int c = 0; c < C; ++c) {
       const float p = e[c] * inv;
       const float d = (p - (c == label ? 1.f : 0.f)) * scale;
-      dls[t * 32 + c] = (__bf16)d;
+      dls[t * CP + c] = (__bf16)d;
+      dlsT[c * HP + t] = (__bf16)d;
       if (wg0) dlogits[(long)t * C + c] = f2bf(d);
       if (c == label) neglogp = -__logf(fmaxf(p, 1e-30f));
     }
@@ -401,7 +407,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   // 32wr..+31, loops the four 32-wide H column tiles; K = C (one MFMA)
   if (wg0) {
     const int arow = wr * 32 + (lane & 31);
-    bf16x8 a = *(const bf16x8*)&dls[arow * 32 + ((lane >> 5) << 3)];
+    bf16x8 a = *(const bf16x8*)&dls[arow * CP + ((lane >> 5) << 3)];
 #pragma unroll
     for (int ct = 0; ct < HB / 32; ++ct) {
       bf16x8 bv =
@@ -433,12 +439,11 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
 #pragma unroll
     for (int kh = 0; kh < HB / 16; ++kh) {
       const int k0 = kh * 16 + ((lane >> 5) << 3);
-      bf16x8 a, bv;
+      bf16x8 a;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
+      for (int j = 0; j < 8; ++j)
         a[j] = hs[(k0 + j) * HP + (wr * 32 + (lane & 31))];
-        bv[j] = dls[(k0 + j) * 32 + (lane & 31)];
-      }
+      bf16x8 bv = *(const bf16x8*)&dlsT[(lane & 31) * HP + k0];
       acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, bv, acc, 0, 0, 0);
     }
     const int c = lane & 31;
@@ -455,7 +460,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
     }
     if (t < C) {
       float s = 0.f;
-      for (int b2 = 0; b2 < B; ++b2) s += (float)dls[b2 * 32 + t];
+      for (int b2 = 0; b2 < B; ++b2) s += (float)dlsT[t * HP + b2];
       if (GF32) ((float*)db2v)[t] = s;
       else ((bf16_t*)db2v)[t] = f2bf(s);
     }
