@@ -234,15 +234,26 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
   int nslice = 1, kc = 0;
   float* ws = nullptr;
   int* cnt = nullptr;
-  if (act != 2 && colsum_p == nullptr && nx * ny < 256 && Ka >= 256) {
-    int want = std::min((int)(Ka / 64), 256 / (nx * ny));
+  // colsum-fused GEMMs (tn, no bias/act) slice too: their partials go
+  // to an extra [nslice, N] stripe area reduced in phase 2. The K gate
+  // is lower for them — the dW shapes are shallow (K = batch) and
+  // occupancy-bound without slicing (mnist dW1: 26 workgroups).
+  const bool cs_ok = colsum_p == nullptr ||
+                     (trans_a && !trans_b && bias_p == nullptr && act == 0);
+  const long kmin = colsum_p != nullptr ? 96 : 256;
+  if (act != 2 && cs_ok && nx * ny < 256 && Ka >= kmin) {
+    int want = std::min(colsum_p != nullptr ? (int)((Ka + 31) / 32)
+                                            : (int)(Ka / 64),
+                        256 / (nx * ny));
     if (want > 16) want = 16;
     if (want > 1) {
       kc = ((Ka + want - 1) / want + 31) / 32 * 32;
       nslice = (Ka + kc - 1) / kc;
       if (nslice > 1)
-        ws = splitk_ws(a.device(), (long)M * N * nslice, (long)nx * ny,
-                       &cnt);
+        ws = splitk_ws(a.device(),
+                       (long)M * N * nslice +
+                           (colsum_p != nullptr ? (long)nslice * N : 0),
+                       (long)nx * ny, &cnt);
     }
   }
   launch_gemm((const bf16_t*)a.data_ptr(), (const bf16_t*)b.data_ptr(),

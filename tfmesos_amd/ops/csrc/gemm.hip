@@ -172,7 +172,9 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
 #undef STAGE_AB
 
   if (CS && blockIdx.y == 0 && t < BN && tn0 + t < N) {
-    if (OUTF32) ((float*)colsum_out)[tn0 + t] = cs_acc;
+    if (SK)       // per-slice fp32 partial, summed by the reduce kernel
+      ((float*)colsum_out)[(long)blockIdx.z * N + tn0 + t] = cs_acc;
+    else if (OUTF32) ((float*)colsum_out)[tn0 + t] = cs_acc;
     else ((__bf16*)colsum_out)[tn0 + t] = (__bf16)cs_acc;
   }
 
@@ -230,9 +232,22 @@ template <int ACT, bool BIAS, bool OUTF32>
 __global__ __launch_bounds__(256)
 void splitk_reduce_kernel(const float* __restrict__ ws, const void* __restrict__ bias,
                           bool bias_bf16, void* __restrict__ Cout, long mn,
-                          int ldc, int nslice) {
+                          int ldc, int nslice,
+                          const float* __restrict__ cs_part,
+                          void* __restrict__ cs_out, bool cs_f32, int N) {
   const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
   if (i0 >= mn) return;
+  if (cs_part != nullptr && i0 < N) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const long n = i0 + j;
+      if (n >= N) break;
+      float s = 0.f;
+      for (int z = 0; z < nslice; ++z) s += cs_part[(long)z * N + n];
+      if (cs_f32) ((float*)cs_out)[n] = s;
+      else ((__bf16*)cs_out)[n] = (__bf16)s;
+    }
+  }
   f32x4 v = {};
   for (int z = 0; z < nslice; ++z) {
     const float* s = ws + (long)z * mn + i0;
@@ -278,15 +293,35 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
                      (const __bf16*)B, bias, bias_bf16, C,                  \
                      (const __bf16*)aux, colsum_out, ws, cnt, M, N, K, lda, \
                      ldb, ldc, kc, veca, vecb)
-  // constraint combos (relu_bwd only nt/bf16, colsum only tn, split-K only
-  // nn) are validated by the bindings in ext.hip
+  // constraint combos (relu_bwd only nt/bf16, colsum only tn) are
+  // validated by the bindings in ext.hip
   if (act == 2) {
     LAUNCH(false, true, 2, false, false, false, false);
     return;
   }
-  if (cs) {
+  if (cs && !sk) {
     if (out_f32) LAUNCH(true, false, 0, false, true, false, true);
     else         LAUNCH(true, false, 0, false, false, false, true);
+    return;
+  }
+  if (cs && sk) {
+    // phase 1: tn with per-slice output AND colsum partials; phase 2
+    // reduces both. cs_part lives at ws + mn*nslice (sized by binding)
+    const long mn = (long)M * ldc;
+    float* cs_part = ws + mn * nslice;
+    void* user_cs = colsum_out;
+    colsum_out = cs_part;    // phase 1 writes the partials
+    LAUNCH(true, false, 0, false, false, true, true);
+    colsum_out = user_cs;
+    dim3 rgrid((unsigned)((mn / 4 + 255) / 256)), rblock(256);
+    if (out_f32)
+      hipLaunchKernelGGL((splitk_reduce_kernel<0, false, true>), rgrid,
+                         rblock, 0, stream, ws, nullptr, false, C, mn, ldc,
+                         nslice, cs_part, colsum_out, out_f32, N);
+    else
+      hipLaunchKernelGGL((splitk_reduce_kernel<0, false, false>), rgrid,
+                         rblock, 0, stream, ws, nullptr, false, C, mn, ldc,
+                         nslice, cs_part, colsum_out, out_f32, N);
     return;
   }
   if (sk) {
@@ -304,7 +339,8 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
 #define RLAUNCH(ACTv, BIASv, OUTv)                                          \
     hipLaunchKernelGGL((splitk_reduce_kernel<ACTv, BIASv, OUTv>), rgrid,    \
                        rblock, 0, stream, ws, bias, bias_bf16, C, mn, ldc,  \
-                       nslice)
+                       nslice, (const float*)nullptr, (void*)nullptr,       \
+                       false, 0)
     if (act == 1) {
       if (has_bias) { if (out_f32) RLAUNCH(1, true, true);
                       else RLAUNCH(1, true, false); }
