@@ -234,13 +234,14 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
   int nslice = 1, kc = 0;
   float* ws = nullptr;
   int* cnt = nullptr;
-  // colsum-fused GEMMs (tn, no bias/act) slice too: their partials go
-  // to an extra [nslice, N] stripe area reduced in phase 2. The K gate
-  // is lower for them — the dW shapes are shallow (K = batch) and
-  // occupancy-bound without slicing (mnist dW1: 26 workgroups).
+  // colsum-fused GEMMs (tn, no bias/act) can slice too: partials go to
+  // an extra [nslice, N] stripe area reduced in phase 2. Same K gate as
+  // plain shapes — slicing SHALLOW colsum GEMMs (mnist dW1, K=100)
+  // measured slower: the second (reduce) launch costs more than the
+  // occupancy it buys.
   const bool cs_ok = colsum_p == nullptr ||
                      (trans_a && !trans_b && bias_p == nullptr && act == 0);
-  const long kmin = colsum_p != nullptr ? 96 : 256;
+  const long kmin = 256;
   if (act != 2 && cs_ok && nx * ny < 256 && Ka >= kmin) {
     int want = std::min(colsum_p != nullptr ? (int)((Ka + 31) / 32)
                                             : (int)(Ka / 64),
