@@ -651,3 +651,30 @@ def test_mlp_fwd_head_fused_matches_composed():
     assert (dh.float() - dh2.float()).abs().max() < 3e-3
     assert (dw2 - dw2r).abs().max() < 3e-3 * dw2r.abs().max() + 1e-4
     assert (db2 - db2r).abs().max() < 3e-3 * db2r.abs().max() + 1e-4
+
+
+def test_mlp_head_fused_odd_shapes():
+    """MFMA head path with non-multiple-of-32 B/H and small C (padding
+    rows/cols must stay zeroed through every phase)."""
+    from tfmesos_amd import ops
+    torch.manual_seed(52)
+    B, H, C = 37, 60, 7
+    h = torch.relu(bf(torch.randn(B, H)))
+    w = bf(torch.randn(H, C) * 0.1)
+    b = bf(torch.randn(C) * 0.1)
+    y = torch.randint(0, C, (B,), device=DEV)
+    dw2 = torch.zeros(H, C, device=DEV, dtype=torch.float32)
+    db2 = torch.zeros(C, device=DEV, dtype=torch.float32)
+    loss, dl, dh = ops.mlp_head_fused(h, w, b, y, dw2=dw2, db2=db2)
+
+    logits = ops.gemm_bias_act(h, w, b)
+    loss2, dl2 = ops.softmax_xent_fused(logits, y)
+    dh2 = ops.gemm_bias_act(dl2, w, trans_b=True, act="relu_bwd", aux=h)
+    dw2r = torch.zeros(H, C, device=DEV, dtype=torch.float32)
+    db2r = torch.zeros(C, device=DEV, dtype=torch.float32)
+    ops.gemm_bias_act(h, dl2, trans_a=True, out=dw2r, colsum_out=db2r)
+    assert abs(float(loss) - float(loss2)) < 1e-2
+    assert (dl.float() - dl2.float()).abs().max() < 1e-3
+    assert (dh.float() - dh2.float()).abs().max() < 2e-3
+    assert (dw2 - dw2r).abs().max() < 3e-3 * dw2r.abs().max() + 1e-4
+    assert (db2 - db2r).abs().max() < 3e-3 * db2r.abs().max() + 1e-4
