@@ -112,7 +112,9 @@ def main(argv=None):
     else:
         groups = make_pair_groups(roles)
         if roles.is_ps:
-            AsyncPSServer(trainer, groups).serve(args.train_steps)
+            # open-ended like the reference's async mode: serve until
+            # every worker sends its stop sentinel (README.rst:68-72)
+            AsyncPSServer(trainer, groups).serve()
         else:
             w = AsyncPSWorker(trainer, groups)
             for step in range(args.train_steps):
@@ -121,6 +123,7 @@ def main(argv=None):
                 timer.step()
                 if step % 20 == 0:
                     print("step %d loss %.4f" % (step, float(loss)))
+            w.stop()
     if device != "cpu":
         torch.cuda.synchronize()
     s = timer.summary()

@@ -25,6 +25,32 @@ def main():
     lr = 0.01 if optimizer == "adam" else 0.1
     if mode == "sync-adam":
         mode = "sync"
+
+    if mode == "sync-tiny":
+        # tiny model whose aligned total (512) yields an EMPTY shard at
+        # n_ps=3: collectives/applies must skip it (ADVICE round-1 low)
+        torch.manual_seed(7)
+        params = [("a", torch.randn(10, 10)), ("b", torch.randn(5))]
+        trainer = SyncReplicaTrainer(params, optimizer="sgd",
+                                     hparams={"lr": 0.1}, device="cpu",
+                                     n_ps=n_ps)
+        assert any(hi == lo for lo, hi in trainer.shards), trainer.shards
+        for _ in range(steps):
+            if trainer.roles.is_worker:
+                trainer.grad_view("a").fill_(0.01)
+                trainer.grad_view("b").fill_(0.02)
+            trainer.step()
+        trainer.sync_masters()
+        if trainer.rank == 0:
+            torch.save({n: trainer.store.view(n).clone()
+                        for n in trainer.store.names}, prefix + ".pt")
+            torch.save(trainer.store.global_step, prefix + ".step")
+        if trainer.world > 1:
+            import torch.distributed as dist
+            dist.barrier()
+            dist.destroy_process_group()
+        return
+
     model = MnistMLP()
     trainer = SyncReplicaTrainer(model.init_params(), optimizer=optimizer,
                                  hparams={"lr": lr}, device="cpu",
@@ -50,6 +76,20 @@ def main():
             for _ in range(steps):
                 model.fwd_bwd(trainer.param, x, y, trainer.grad_view)
                 w.step()
+    elif mode == "async-open":
+        # open-ended contract: no step count at the server; workers run
+        # UNEVEN step counts (steps + worker_index) then send the stop
+        # sentinel (reference async mode is open-ended, README.rst:68-72)
+        from tfmesos_amd.ps.replica import make_pair_groups
+        groups = make_pair_groups(roles)
+        if roles.is_ps:
+            AsyncPSServer(trainer, groups).serve()
+        else:
+            w = AsyncPSWorker(trainer, groups)
+            for _ in range(steps + roles.worker_index):
+                model.fwd_bwd(trainer.param, x, y, trainer.grad_view)
+                w.step()
+            w.stop()
     else:
         raise SystemExit("bad mode")
 

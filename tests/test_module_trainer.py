@@ -70,6 +70,64 @@ def test_module_trainer_checkpoint_roundtrip(tmp_path):
     assert tr2.store.global_step == 3
 
 
+def test_arena_layout_matches_store_offsets():
+    """The grad arena's slice offsets must equal the PStore flat
+    offsets (both 256-element aligned): the trainer's ONE bulk
+    flat_grad[:na].copy_(arena) is only correct under that invariant
+    (dense arena packing shifted 93/94 conv
+    gradients)."""
+    m = InceptionV3(num_classes=10)
+    tr = ModuleReplicaTrainer(m, hparams={"lr": 0.01})
+    assert tr._arena is not None
+    assert m._arena_offsets, "arena must be wired on every device"
+    for name, off in m._arena_offsets.items():
+        assert off == tr.store.offsets[name][0], name
+    # numerically: mark each conv's arena slice, run the bulk-copy step
+    # path, and check every grad_view sees its own mark
+    from tfmesos_amd.models.inception import Conv2d
+    convs = [(n + ".weight", mod) for n, mod in m.named_modules()
+             if isinstance(mod, Conv2d)]
+    for i, (_, mod) in enumerate(convs):
+        mod._dw_buf.fill_(float(i + 1))
+    tr.t.flat_grad[:tr._arena.numel()].copy_(tr._arena)
+    for i, (name, mod) in enumerate(convs):
+        gv = tr.t.grad_view(name)
+        assert torch.all(gv == float(i + 1)), name
+
+
+def test_store_load_preserves_views(tmp_path):
+    """Full-checkpoint load must copy INTO the existing flat buffers:
+    trainers hold live views (bf16 broadcast source, nn.Module .data
+    aliases) that a rebind would orphan."""
+    m = TinyCNN()
+    tr = ModuleReplicaTrainer(m, hparams={"lr": 0.05})
+    x = torch.rand(4, 3, 16, 16, dtype=torch.bfloat16)
+    y = torch.randint(0, 5, (4,))
+    for _ in range(3):
+        tr.zero_grad()
+        torch.nn.functional.cross_entropy(m(x), y).backward()
+        tr.step()
+    path = str(tmp_path / "ck.pt")
+    tr.save(path)
+    want = tr.store.flat.clone()
+    # train further, then restore — module params must SEE the restore
+    for _ in range(2):
+        tr.zero_grad()
+        torch.nn.functional.cross_entropy(m(x), y).backward()
+        tr.step()
+    assert not torch.equal(tr.store.flat, want)
+    flat_before = tr.store.flat
+    bf16_before = tr.store.flat_bf16
+    tr.load(path)
+    assert tr.store.flat is flat_before, "load must not rebind flat"
+    assert tr.store.flat_bf16 is bf16_before
+    assert torch.equal(tr.store.flat, want)
+    w = dict(m.named_parameters())["c1.conv.weight"]
+    assert torch.equal(
+        w.data.float().flatten(),
+        tr.store.view("c1.conv.weight").to(torch.bfloat16).float().flatten())
+
+
 def test_inception_v3_parameter_inventory():
     m = InceptionV3(num_classes=10)
     n = sum(p.numel() for p in m.parameters())

@@ -93,6 +93,32 @@ def test_async_world2_trains(tmp_path):
 
 
 @pytest.mark.timeout(240)
+def test_async_open_ended_uneven_workers(tmp_path):
+    """Open-ended async (the reference's contract, README.rst:68-72):
+    the server takes NO step count; workers run UNEVEN step counts and
+    send stop sentinels. Global step must equal the total worker
+    steps."""
+    prefix = str(tmp_path / "wopen")
+    _spawn_world(3, "async-open", 4, prefix)   # workers run 4 and 5 steps
+    step = torch.load(prefix + ".step", weights_only=True)
+    assert step == 4 + 5
+
+
+@pytest.mark.timeout(240)
+def test_empty_shards_small_model(tmp_path):
+    """n_ps=3 over a tiny model (total <= 2*256 after alignment): one
+    shard is EMPTY — collectives and applies must skip it instead of
+    issuing zero-numel ops."""
+    from tfmesos_amd.ps.replica import _shard_ranges
+    shards = _shard_ranges(512, 3)
+    assert any(hi == lo for lo, hi in shards), shards
+    prefix = str(tmp_path / "wempty")
+    _spawn_world(4, "sync-tiny", 5, prefix, n_ps=3)
+    got = torch.load(prefix + ".pt", weights_only=True)
+    assert all(torch.isfinite(v).all() for v in got.values())
+
+
+@pytest.mark.timeout(240)
 def test_sync_two_ps_shards_match_single_process(tmp_path):
     """2 PS shards + 1 worker: sharded reduce/apply/broadcast must equal
     the single-process reference exactly (multi-PS sharding, SURVEY.md

@@ -264,19 +264,28 @@ class InceptionV3(nn.Module):
         buffer with ONE fused cast-copy (a mixed-dtype _foreach_copy_
         de-batches into per-tensor hipMemcpys — measured +0.4 ms/step).
         Constraint: one forward per backward (no cross-step grad
-        accumulation) — which is the replica-trainer step pattern."""
+        accumulation) — which is the replica-trainer step pattern.
+
+        Slice starts are 256-element aligned to MATCH PStore.init_params'
+        flat layout (ps/store.py:_align): the trainer's one bulk
+        ``flat_grad[:na].copy_(arena)`` is only correct if every arena
+        param sits at the same offset in both buffers (the round-1 dense
+        packing silently shifted 93/94 conv-weight gradients)."""
+        from tfmesos_amd.ps.store import _align
         convs = [(n + ".weight", m) for n, m in self.named_modules()
                  if isinstance(m, Conv2d)]
-        total = sum(m.weight.numel() for _, m in convs)
+        total = sum(_align(m.weight.numel()) for _, m in convs)
         self._arena = torch.zeros(total, dtype=torch.float32,
                                   device=device)
         off, names = 0, []
+        self._arena_offsets = {}
         for name, m in convs:
             n = m.weight.numel()
             m._dw_buf = self._arena[off:off + n].view(m.weight.shape)
             m.weight._tfa_raw_grad = m._dw_buf
             names.append(name)
-            off += n
+            self._arena_offsets[name] = off
+            off += _align(n)
         return names, self._arena
 
     def _ensure_arena(self, device):

@@ -313,7 +313,7 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
     colsum_out = cs_part;    // phase 1 writes the partials
     LAUNCH(true, false, 0, false, false, true, true);
     colsum_out = user_cs;
-    dim3 rgrid((unsigned)((mn / 4 + 255) / 256)), rblock(256);
+    dim3 rgrid((unsigned)(((mn + 3) / 4 + 255) / 256)), rblock(256);
     if (out_f32)
       hipLaunchKernelGGL((splitk_reduce_kernel<0, false, true>), rgrid,
                          rblock, 0, stream, ws, nullptr, false, C, mn, ldc,
@@ -335,7 +335,7 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
               else    LAUNCH(false, false, 0, false, false, true, false); }
     // phase 2: fixed-order stripe reduce + fused epilogue
     const long mn = (long)M * ldc;
-    dim3 rgrid((unsigned)((mn / 4 + 255) / 256)), rblock(256);
+    dim3 rgrid((unsigned)(((mn + 3) / 4 + 255) / 256)), rblock(256);
 #define RLAUNCH(ACTv, BIASv, OUTv)                                          \
     hipLaunchKernelGGL((splitk_reduce_kernel<ACTv, BIASv, OUTv>), rgrid,    \
                        rblock, 0, stream, ws, bias, bias_bf16, C, mn, ldc,  \

@@ -33,8 +33,12 @@ class ModuleReplicaTrainer(object):
         # gather is a single fused cast-copy
         self._arena = None
         arena_names = []
-        if self.device.type == "cuda" and hasattr(module,
-                                                  "wire_grad_arena"):
+        if hasattr(module, "wire_grad_arena"):
+            # wired on every device so the flat layout (arena params
+            # first, 256-aligned) is identical on CPU and GPU; the
+            # kernels only write the arena on GPU (conv dw_out is
+            # is_cuda-gated), so on CPU the bulk copy moves zeros that
+            # the autograd foreach pass then overwrites
             arena_names, self._arena = module.wire_grad_arena(self.device)
         pdict = dict(named)
         order = arena_names + [n for n, _ in named
@@ -45,6 +49,20 @@ class ModuleReplicaTrainer(object):
                                     hparams=hparams, device=device,
                                     grad_dtype=torch.bfloat16, n_ps=n_ps,
                                     colocate_ps=colocate_ps)
+        if self._arena is not None:
+            # hard invariant behind the one bulk arena->flat_grad copy:
+            # every arena param must sit at the SAME offset in both
+            # buffers (store aligns starts to 256; wire_grad_arena must
+            # match)
+            offs = getattr(module, "_arena_offsets", {})
+            for name in arena_names:
+                if offs.get(name) != self.t.store.offsets[name][0]:
+                    raise RuntimeError(
+                        "grad-arena offset mismatch for %s: arena %s vs "
+                        "store %s — wire_grad_arena layout must match "
+                        "PStore alignment" % (
+                            name, offs.get(name),
+                            self.t.store.offsets[name][0]))
         self.roles = self.t.roles
         self.module = module
         module.to(device=self.device, dtype=torch.bfloat16)

@@ -149,7 +149,12 @@ class Roles(object):
 
 
 def _shard_ranges(total, n_ps, align=256):
-    """Even, 256-element-aligned contiguous slices of the flat buffer."""
+    """Even, 256-element-aligned contiguous slices of the flat buffer.
+
+    Small buffers with many PS ranks can yield EMPTY shards (lo == hi,
+    e.g. total<=512 with n_ps=3); every consumer skips collectives and
+    applies for those (zero-numel reduce/broadcast is error-prone on
+    gloo and pointless everywhere)."""
     bounds = [0]
     for i in range(1, n_ps):
         b = (total * i // n_ps + align - 1) // align * align
@@ -211,8 +216,9 @@ class SyncReplicaTrainer(object):
         if self.world == 1:
             return
         for i, (lo, hi) in enumerate(self.shards):
-            broadcast_(self.flat_params_bf16[lo:hi],
-                       src=self.roles.ps_ranks[i])
+            if hi > lo:
+                broadcast_(self.flat_params_bf16[lo:hi],
+                           src=self.roles.ps_ranks[i])
 
     # parameter views (bf16 working copies, refreshed in-place by pull)
     def param(self, name):
@@ -246,15 +252,18 @@ class SyncReplicaTrainer(object):
                 # (colocated ps ranks contribute their own worker grads)
                 self.flat_grad.zero_()
             for i, (lo, hi) in enumerate(self.shards):
-                reduce_(self.flat_grad[lo:hi], dst=self.roles.ps_ranks[i])
-        if self.my_shard is not None:
+                if hi > lo:
+                    reduce_(self.flat_grad[lo:hi],
+                            dst=self.roles.ps_ranks[i])
+        if self.my_shard is not None and self.my_shard[1] > self.my_shard[0]:
             lo, hi = self.my_shard
             self.store.apply_flat(self.flat_grad, grad_scale=scale,
                                   lo=lo, hi=hi)
         if self.world > 1:
             for i, (lo, hi) in enumerate(self.shards):
-                broadcast_(self.flat_params_bf16[lo:hi],
-                           src=self.roles.ps_ranks[i])
+                if hi > lo:
+                    broadcast_(self.flat_params_bf16[lo:hi],
+                               src=self.roles.ps_ranks[i])
         return self.store.global_step
 
     def sync_masters(self):
@@ -262,8 +271,9 @@ class SyncReplicaTrainer(object):
         complete master buffer (e.g. for a full checkpoint or eval)."""
         if self.world > 1:
             for i, (lo, hi) in enumerate(self.shards):
-                broadcast_(self.store.flat[lo:hi],
-                           src=self.roles.ps_ranks[i])
+                if hi > lo:
+                    broadcast_(self.store.flat[lo:hi],
+                               src=self.roles.ps_ranks[i])
         return self.store.flat
 
     def save(self, path):
@@ -288,16 +298,20 @@ class SyncReplicaTrainer(object):
 
 
 def make_pair_groups(roles):
-    """One process group per (ps, worker) pair — every rank must call
-    new_group for every pair (it is collective). These give async mode
-    independent channels: each PS serving thread blocks on its own
-    group, which is safe on both gloo and RCCL (one communicator per
-    thread)."""
-    groups = {}
-    for p in roles.ps_ranks:
-        for w in roles.worker_ranks:
-            groups[(p, w)] = dist.new_group([p, w])
-    return groups
+    """One Chan per (ps, worker) pair — ``new_group`` is collective, so
+    every rank calls this (with identical pair enumeration). Async mode
+    uses these as independent request/reply channels per pair."""
+    from tfmesos_amd.ps.chan import make_pair_chans
+    return make_pair_chans(roles.ps_ranks, roles.worker_ranks,
+                           my_rank=roles.rank)
+
+
+# async wire protocol: each interaction on a (ps, worker) channel is
+# [flag int64] then payload. Replicates the reference's OPEN-ENDED
+# async contract (workers train on their own clock until done,
+# README.rst:68-72): the PS never needs a step count up front.
+_FLAG_STEP = 1
+_FLAG_STOP = 0
 
 
 class AsyncPSWorker(object):
@@ -305,70 +319,143 @@ class AsyncPSWorker(object):
 
     Replicates the reference's default async mode (each worker pushes
     grads and pulls params on its own clock, no aggregation barrier —
-    README.rst:68-72)."""
+    README.rst:68-72). Call ``stop()`` when done: the server exits once
+    every worker has stopped (open-ended, like the reference)."""
 
-    def __init__(self, trainer, pair_groups):
+    def __init__(self, trainer, pair_chans):
         self.t = trainer
         assert trainer.roles.is_worker and trainer.world > 1, \
             "async mode needs separate PS rank(s)"
-        self.groups = pair_groups
+        self.chans = pair_chans
+        dev = trainer.flat_grad.device if dist.get_backend() == "nccl" \
+            else torch.device("cpu")
+        self._flag = {f: torch.tensor([f], dtype=torch.int64, device=dev)
+                      for f in (_FLAG_STEP, _FLAG_STOP)}
+
+    def _chan(self, ps_rank):
+        return self.chans[(ps_rank, self.t.rank)]
 
     def step(self):
         t = self.t
-        gloo = dist.get_backend() == "gloo"
         for i, (lo, hi) in enumerate(t.shards):
-            p = t.roles.ps_ranks[i]
-            g = t.flat_grad[lo:hi]
-            dist.send(g.cpu() if (gloo and g.is_cuda) else g, dst=p,
-                      group=self.groups[(p, t.rank)])
+            if hi <= lo:
+                continue
+            c = self._chan(t.roles.ps_ranks[i])
+            c.send(self._flag[_FLAG_STEP])
+            c.send(t.flat_grad[lo:hi])
         for i, (lo, hi) in enumerate(t.shards):
-            p = t.roles.ps_ranks[i]
-            dst = t.flat_params_bf16[lo:hi]
-            if gloo and dst.is_cuda:
-                buf = torch.empty_like(dst, device="cpu")
-                dist.recv(buf, src=p, group=self.groups[(p, t.rank)])
-                dst.copy_(buf)
-            else:
-                dist.recv(dst, src=p, group=self.groups[(p, t.rank)])
+            if hi <= lo:
+                continue
+            self._chan(t.roles.ps_ranks[i]).recv_into(
+                t.flat_params_bf16[lo:hi])
         return True
 
     def stop(self):
-        """Tell the PS this worker is done (a zero-length sentinel is not
-        expressible; the server counts steps instead)."""
+        """Tell every PS shard this worker is done (sentinel flag)."""
+        for i, (lo, hi) in enumerate(self.t.shards):
+            if hi > lo:
+                self._chan(self.t.roles.ps_ranks[i]).send(
+                    self._flag[_FLAG_STOP])
 
 
 class AsyncPSServer(object):
-    """PS-shard side: one serving thread per worker, apply-on-arrival
-    over this rank's shard only."""
+    """PS-shard side: apply-on-arrival over this rank's shard.
 
-    def __init__(self, trainer, pair_groups):
+    Two serving strategies by backend:
+
+    * **nccl/RCCL**: ONE polling loop over every worker channel (irecv
+      on the flag word, ``Work.is_completed`` is a CUDA-event query).
+      Single-threaded by design — N threads blocking on N RCCL
+      communicators of one device can deadlock.
+    * **gloo**: one blocking thread per worker channel (gloo's p2p
+      ``Work.is_completed`` does not flip without ``wait()``, so the
+      polling loop cannot make progress there; threads on gloo are
+      safe — it is the CPU/test backend).
+    """
+
+    def __init__(self, trainer, pair_chans):
         self.t = trainer
-        self.groups = pair_groups
+        self.chans = pair_chans
         assert trainer.roles.is_ps and trainer.world > 1
 
-    def _serve_one(self, worker_rank, steps):
+    def _handle(self, c, buf, full, lo, hi):
         t = self.t
-        lo, hi = t.my_shard
-        buf = torch.zeros(hi - lo, dtype=t.flat_grad.dtype,
-                          device=t.flat_grad.device)
-        group = self.groups[(t.rank, worker_rank)]
-        full = torch.zeros_like(t.flat_grad)
-        for _ in range(steps):
-            dist.recv(buf, src=worker_rank, group=group)
-            with t.store.lock:
-                full[lo:hi].copy_(buf)
-                t.store.apply_flat(full, lo=lo, hi=hi)
-                params = t.store.flat_bf16[lo:hi].clone()
-            dist.send(params, dst=worker_rank, group=group)
+        c.recv_into(buf)
+        with t.store.lock:
+            full[lo:hi].copy_(buf)
+            t.store.apply_flat(full, lo=lo, hi=hi)
+            params = t.store.flat_bf16[lo:hi].clone()
+        c.send(params)
 
-    def serve(self, steps_per_worker):
+    def serve(self, steps_per_worker=None):
+        """Serve until every worker sends the stop sentinel (default,
+        matching the reference's open-ended async mode — the reference
+        never needs a step count up front, README.rst:68-72), or until
+        each worker has been served ``steps_per_worker`` steps (bounded
+        variant for benchmarks)."""
+        t = self.t
+        lo, hi = t.my_shard if t.my_shard is not None else (0, 0)
+        if hi <= lo:
+            return t.store.global_step   # empty shard: nothing to serve
+        if dist.get_backend() == "nccl":
+            self._serve_polling(lo, hi, steps_per_worker)
+        else:
+            self._serve_threads(lo, hi, steps_per_worker)
+        return t.store.global_step
+
+    def _serve_threads(self, lo, hi, steps_per_worker):
         import threading
-        threads = [
-            threading.Thread(target=self._serve_one, args=(w, steps_per_worker))
-            for w in self.t.roles.worker_ranks
-        ]
+        t = self.t
+
+        def one(w):
+            c = self.chans[(t.rank, w)]
+            buf = torch.zeros(hi - lo, dtype=t.flat_grad.dtype,
+                              device=t.flat_grad.device)
+            full = torch.zeros_like(t.flat_grad)
+            flag = torch.zeros(1, dtype=torch.int64)
+            served = 0
+            while steps_per_worker is None or served < steps_per_worker:
+                c.recv_into(flag)
+                if int(flag.item()) == _FLAG_STOP:
+                    return
+                self._handle(c, buf, full, lo, hi)
+                served += 1
+
+        threads = [threading.Thread(target=one, args=(w,))
+                   for w in t.roles.worker_ranks]
         for th in threads:
             th.start()
         for th in threads:
             th.join()
-        return self.t.store.global_step
+
+    def _serve_polling(self, lo, hi, steps_per_worker):
+        import time as _time
+        t = self.t
+        dev = t.flat_grad.device
+        buf = torch.zeros(hi - lo, dtype=t.flat_grad.dtype, device=dev)
+        full = torch.zeros_like(t.flat_grad)
+        flags, pending, served = {}, {}, {}
+        chans = {w: self.chans[(t.rank, w)] for w in t.roles.worker_ranks}
+        for w, c in chans.items():
+            flags[w] = torch.zeros(1, dtype=torch.int64, device=dev)
+            pending[w] = c.irecv_into(flags[w])
+            served[w] = 0
+        while pending:
+            progress = False
+            for w in list(pending):
+                if not pending[w].is_completed():
+                    continue
+                progress = True
+                c = chans[w]
+                if int(flags[w].item()) == _FLAG_STOP:
+                    del pending[w]
+                    continue
+                self._handle(c, buf, full, lo, hi)
+                served[w] += 1
+                if steps_per_worker is not None \
+                        and served[w] >= steps_per_worker:
+                    del pending[w]   # bounded mode: no more irecvs
+                else:
+                    pending[w] = c.irecv_into(flags[w])
+            if not progress:
+                _time.sleep(1e-4)

@@ -189,13 +189,27 @@ class PStore(object):
                 self.names = ck["names"]
                 self.shapes = ck["shapes"]
                 self.offsets = ck["offsets"]
-                self.flat = ck["flat"].to(self.device)
-                self.flat_bf16 = self.flat.to(torch.bfloat16)
                 self.opt = ck["opt"]
                 self.hparams = ck["hparams"]
-                self.state = {k: v.to(self.device)
-                              for k, v in ck["state"].items()}
                 self.global_step = ck["global_step"]
+                # copy INTO the existing buffers when the layout matches:
+                # trainers hold live views of flat/flat_bf16 (broadcast
+                # source, nn.Module .data aliases) — rebinding would
+                # orphan them and train/broadcast stale memory
+                if (self.flat is not None
+                        and self.flat.numel() == ck["flat"].numel()):
+                    self.flat.copy_(ck["flat"].to(self.device))
+                    self.flat_bf16.copy_(self.flat.to(torch.bfloat16))
+                    for k, v in ck["state"].items():
+                        if k in self.state:
+                            self.state[k].copy_(v.to(self.device))
+                        else:
+                            self.state[k] = v.to(self.device)
+                else:
+                    self.flat = ck["flat"].to(self.device)
+                    self.flat_bf16 = self.flat.to(torch.bfloat16)
+                    self.state = {k: v.to(self.device)
+                                  for k, v in ck["state"].items()}
                 return
             cklo, ckhi = shard if shard is not None else (lo, hi)
             if self.flat is None:
