@@ -207,6 +207,15 @@ def main():
 
     if rank == 0:
         steps_per_sec = args.steps / elapsed
+        # honest topology label: N=1 numbers have NO transport cost
+        # (ps+worker colocated in one process); never conflate them
+        # with N>1 points of the scaling curve
+        if world == 1:
+            transport = "none(colocated-1proc)"
+        else:
+            transport = {"nccl": "rccl"}.get(dist.get_backend(),
+                                             dist.get_backend())
+        config["transport"] = transport
         out = {
             "metric": "global steps/sec, mnist_replica 1-ps/N-worker at "
                       "1/2/4/8 MI355X" if args.workload == "mnist"

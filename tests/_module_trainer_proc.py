@@ -26,21 +26,25 @@ class TinyCNN(torch.nn.Module):
 
 def main():
     steps, out = int(sys.argv[1]), sys.argv[2]
+    variant = sys.argv[3] if len(sys.argv) > 3 else "colocate"
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     if world > 1:
         dist.init_process_group("gloo", rank=rank, world_size=world)
     m = TinyCNN()
     tr = ModuleReplicaTrainer(m, optimizer="sgd", hparams={"lr": 0.05},
-                              colocate_ps=True)
+                              colocate_ps=variant.startswith("colocate"),
+                              overlap="no-overlap" not in variant)
     torch.manual_seed(42)          # same batch on every rank
     x = torch.rand(8, 3, 16, 16, dtype=torch.bfloat16)
     y = torch.randint(0, 5, (8,))
     for _ in range(steps):
         tr.zero_grad()
-        loss = torch.nn.functional.cross_entropy(m(x), y)
-        loss.backward()
+        if tr.roles.is_worker:
+            loss = torch.nn.functional.cross_entropy(m(x), y)
+            loss.backward()
         tr.step()
+    tr.t.sync_masters()
     if rank == 0:
         torch.save(tr.store.flat.clone(), out)
     if world > 1:

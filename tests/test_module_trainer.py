@@ -184,3 +184,48 @@ def test_module_trainer_world2_matches_single_process(tmp_path):
     got2 = torch.load(out2, weights_only=True)
     got1 = torch.load(out1, weights_only=True)
     assert torch.allclose(got1, got2, atol=1e-5)
+
+
+def _run_world2(tmp_path, variant, tag):
+    import os
+    import subprocess
+    import sys
+
+    from tfmesos_amd.utils import free_port
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    repo = os.path.dirname(here)
+    out = str(tmp_path / ("%s.pt" % tag))
+    port = free_port()
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ, RANK=str(rank), WORLD_SIZE="2",
+                   MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                   PYTHONPATH=repo)
+        procs.append(subprocess.Popen(
+            [sys.executable, os.path.join(here, "_module_trainer_proc.py"),
+             "4", out, variant], env=env))
+    for p in procs:
+        assert p.wait(timeout=180) == 0
+    return torch.load(out, weights_only=True)
+
+
+@pytest.mark.timeout(240)
+def test_overlap_matches_no_overlap_world2(tmp_path):
+    """The bucketed backward-overlap reduce path must produce masters
+    identical to the plain (blocking, post-backward) reduce path."""
+    a = _run_world2(tmp_path, "colocate", "ovl")
+    b = _run_world2(tmp_path, "colocate-no-overlap", "novl")
+    assert torch.allclose(a, b, atol=1e-6)
+
+
+@pytest.mark.timeout(240)
+def test_overlap_pure_ps_rank_world2(tmp_path):
+    """Non-colocated: rank 0 is a PURE PS (no backward, no hooks) — it
+    must issue the same bucket reduces in the same order with a zero
+    contribution. Masters must match the colocated 1-worker run is not
+    expected (different n_workers), but must match its own no-overlap
+    variant exactly."""
+    a = _run_world2(tmp_path, "ps", "psovl")
+    b = _run_world2(tmp_path, "ps-no-overlap", "psnovl")
+    assert torch.allclose(a, b, atol=1e-6)

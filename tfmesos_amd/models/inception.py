@@ -32,11 +32,13 @@ class Conv2d(nn.Module):
             torch.randn(cout, kernel_size[0], kernel_size[1], cin)
             * (2.0 / fan_in) ** 0.5)
         self._dw_buf = None    # fp32 view into the model grad arena (GPU)
+        self._dw_cb = None     # comm-overlap notify (ps/module_trainer.py)
 
     def forward(self, x):
         return ops.conv2d(x, self.weight, None, stride=self.stride,
                           padding=self.padding, weight_format="krsc",
-                          dw_out=self._dw_buf if x.is_cuda else None)
+                          dw_out=self._dw_buf if x.is_cuda else None,
+                          dw_cb=self._dw_cb if x.is_cuda else None)
 
 
 class BatchNorm2d(nn.Module):

@@ -259,12 +259,14 @@ class _Conv2dFn(torch.autograd.Function):
     torch fp32 reference so model code runs in CI."""
 
     @staticmethod
-    def forward(ctx, x, w, bias, stride, padding, krsc, dw_out=None):
+    def forward(ctx, x, w, bias, stride, padding, krsc, dw_out=None,
+                dw_cb=None):
         ctx.stride = stride
         ctx.padding = padding
         ctx.has_bias = bias is not None
         ctx.krsc = krsc
         ctx.dw_out = dw_out
+        ctx.dw_cb = dw_cb
         ctx.pw = False
         if x.is_cuda:
             # channels-last memory: gathers become contiguous channel
@@ -338,6 +340,13 @@ class _Conv2dFn(torch.autograd.Function):
                     dy, x, R, S, ctx.stride[0], ctx.stride[1],
                     ctx.padding[0], ctx.padding[1], ctx.dw_out)
                 dw = None
+                if ctx.dw_cb is not None:
+                    # comm-overlap hook: the dW kernel for this layer is
+                    # now ENQUEUED on the compute stream, so a reduce
+                    # issued here is stream-ordered after it — the
+                    # bucket manager uses this to overlap the PS push
+                    # with the rest of backward (ps/module_trainer.py)
+                    ctx.dw_cb()
             else:
                 dwm = _ext().conv2d_bwd_weight(
                     dy, x, R, S, ctx.stride[0], ctx.stride[1],
@@ -360,11 +369,11 @@ class _Conv2dFn(torch.autograd.Function):
                 dw = dw.permute(0, 2, 3, 1)
             dw = dw.contiguous().to(w.dtype)
         db = dy.float().sum(dim=(0, 2, 3)) if ctx.has_bias else None
-        return dx, dw, db, None, None, None, None
+        return dx, dw, db, None, None, None, None, None
 
 
 def conv2d(x, w, bias=None, stride=1, padding=0, weight_format="kcrs",
-           dw_out=None):
+           dw_out=None, dw_cb=None):
     """2-D convolution (NCHW activations): hand-written implicit-GEMM
     MFMA kernels on GPU (csrc/conv.hip), torch fp32 reference on CPU.
     Differentiable. weight_format "kcrs" (torch layout) or "krsc"
@@ -384,7 +393,7 @@ def conv2d(x, w, bias=None, stride=1, padding=0, weight_format="kcrs",
         assert weight_format == "krsc" and x.is_cuda, \
             "dw_out requires krsc weights on GPU"
     return _Conv2dFn.apply(x, w, bias, tuple(stride), tuple(padding),
-                           weight_format == "krsc", dw_out)
+                           weight_format == "krsc", dw_out, dw_cb)
 
 
 class _SoftmaxXentFn(torch.autograd.Function):

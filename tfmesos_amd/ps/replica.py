@@ -91,6 +91,36 @@ def all_reduce_(t):
         t.copy_(s)
 
 
+def reduce_batch(slices_dsts):
+    """Issue ALL per-shard reduces async, then wait once: shards with
+    different PS roots progress concurrently instead of serializing one
+    blocking collective per shard in a Python loop (round-1 weak #3 —
+    on xGMI each root's reduce uses that root's own links, so the
+    shard reduces are bandwidth-disjoint and should overlap)."""
+    works = []
+    for t, dst in slices_dsts:
+        s = _staged(t)
+        w = dist.reduce(s, dst=dst, op=dist.ReduceOp.SUM, async_op=True)
+        works.append((w, s, t))
+    for w, s, t in works:
+        w.wait()
+        if s is not t:
+            t.copy_(s)
+
+
+def broadcast_batch(slices_srcs):
+    """Async-issued per-shard broadcasts, one wait (see reduce_batch)."""
+    works = []
+    for t, src in slices_srcs:
+        s = _staged(t)
+        w = dist.broadcast(s, src=src, async_op=True)
+        works.append((w, s, t))
+    for w, s, t in works:
+        w.wait()
+        if s is not t:
+            t.copy_(s)
+
+
 class Roles(object):
     """Rank -> role mapping. Ranks [0, n_ps) are PS shards (reference:
     ps tasks come first in the jobs_def and hold param slices), the rest
@@ -215,10 +245,10 @@ class SyncReplicaTrainer(object):
     def _broadcast_params(self):
         if self.world == 1:
             return
-        for i, (lo, hi) in enumerate(self.shards):
-            if hi > lo:
-                broadcast_(self.flat_params_bf16[lo:hi],
-                           src=self.roles.ps_ranks[i])
+        broadcast_batch([(self.flat_params_bf16[lo:hi],
+                          self.roles.ps_ranks[i])
+                         for i, (lo, hi) in enumerate(self.shards)
+                         if hi > lo])
 
     # parameter views (bf16 working copies, refreshed in-place by pull)
     def param(self, name):
@@ -231,39 +261,40 @@ class SyncReplicaTrainer(object):
     def zero_grad(self):
         self.flat_grad.zero_()
 
-    def step(self, grad_scale=None):
+    def step(self, grad_scale=None, skip_reduce=False):
         """One global step: reduce grads per shard -> PS apply -> broadcast.
 
         The worker must have filled ``flat_grad`` (via ``grad_view``)
         before calling. grad_scale defaults to 1/n_workers (grad mean,
-        matching SyncReplicasOptimizer's averaging).
+        matching SyncReplicasOptimizer's averaging). ``skip_reduce``:
+        the caller already reduced flat_grad (the backward-overlap
+        bucket manager, ps/module_trainer.py) — go straight to apply.
         """
         scale = grad_scale if grad_scale is not None \
             else 1.0 / self.roles.n_workers
         if self.mode == "allreduce":
-            if self.world > 1:
+            if self.world > 1 and not skip_reduce:
                 all_reduce_(self.flat_grad)
             self.store.apply_flat(self.flat_grad, grad_scale=scale)
             return self.store.global_step
-        if self.world > 1:
+        if self.world > 1 and not skip_reduce:
             if self.roles.is_ps and not self.roles.is_worker:
                 # dist.reduce is in-place: a PURE ps buffer holds last
                 # step's sum and would be re-added — contribute zeros
                 # (colocated ps ranks contribute their own worker grads)
                 self.flat_grad.zero_()
-            for i, (lo, hi) in enumerate(self.shards):
-                if hi > lo:
-                    reduce_(self.flat_grad[lo:hi],
-                            dst=self.roles.ps_ranks[i])
+            reduce_batch([(self.flat_grad[lo:hi], self.roles.ps_ranks[i])
+                          for i, (lo, hi) in enumerate(self.shards)
+                          if hi > lo])
         if self.my_shard is not None and self.my_shard[1] > self.my_shard[0]:
             lo, hi = self.my_shard
             self.store.apply_flat(self.flat_grad, grad_scale=scale,
                                   lo=lo, hi=hi)
         if self.world > 1:
-            for i, (lo, hi) in enumerate(self.shards):
-                if hi > lo:
-                    broadcast_(self.flat_params_bf16[lo:hi],
-                               src=self.roles.ps_ranks[i])
+            broadcast_batch([(self.flat_params_bf16[lo:hi],
+                              self.roles.ps_ranks[i])
+                             for i, (lo, hi) in enumerate(self.shards)
+                             if hi > lo])
         return self.store.global_step
 
     def sync_masters(self):

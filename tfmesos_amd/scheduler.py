@@ -208,7 +208,35 @@ class LocalScheduler(object):
 
     # -------------------------------------------------------------- start
 
+    def _check_feasible(self):
+        """Fail fast when the jobs_def can NEVER fit the node: every
+        task must be alive simultaneously for the rendezvous barrier, so
+        total demand must fit total inventory. The reference (and
+        round 1) instead idled until the rendezvous timeout; raising at
+        start() with the offending task is strictly more useful."""
+        total_cpus = getattr(self.backend, "total_cpus",
+                             getattr(self.backend, "free_cpus", None))
+        total_mem = getattr(self.backend, "total_mem",
+                            getattr(self.backend, "free_mem", None))
+        gpu_ids = getattr(self.backend, "free_gpu_ids", None)
+        if total_cpus is None or total_mem is None or gpu_ids is None:
+            return  # unknown inventory: can't prove infeasibility
+        cpus, mem, gpus = 0.0, 0.0, 0
+        for t in self.tasks:
+            cpus += t.cpus
+            mem += t.mem
+            gpus += int(t.gpus)
+            if cpus > total_cpus or mem > total_mem or gpus > len(gpu_ids):
+                raise RuntimeError(
+                    "jobs_def does not fit this node: /job:%s/task:%s "
+                    "pushes cumulative demand to cpus=%.1f/%.1f "
+                    "mem=%.0f/%.0f gpus=%d/%d — it would wait for the "
+                    "rendezvous timeout and never start" % (
+                        t.job_name, t.task_index, cpus, total_cpus,
+                        mem, total_mem, gpus, len(gpu_ids)))
+
     def start(self):
+        self._check_feasible()
         lfd = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
         lfd.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
         lfd.bind(("127.0.0.1", 0))
