@@ -113,14 +113,17 @@ struct FwdPatchStage {
     wo = rem - ho * cs.Wo;
   }
 
-  DEVINL void stage(const __bf16* __restrict__ X, __bf16* Sm,
-                    const ConvShape cs, int k0, int KD, bool cvec) {
+  bf16x8 vv;   // in-flight tile (load() -> commit())
+
+  // split load/commit: load() issues the global reads into registers
+  // BEFORE the MFMA block (their latency overlaps compute); commit()
+  // is one b128 LDS store after the MFMAs. The fused form parked each
+  // wave on vmcnt ahead of its own MFMAs.
+  DEVINL void load(const __bf16* __restrict__ X, const ConvShape cs,
+                   int k0, int KD, bool cvec) {
     const int q0 = k0 + kk0;
-    if (q0 >= KD || !pok) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) srow(Sm, mx)[kk0 + j] = (__bf16)0.f;
-      return;
-    }
+    vv = {};
+    if (q0 >= KD || !pok) return;
     const int rs = fdiv(q0, cs.dC);    // tap (r,s) block (c0 % 8 == 0 when
     const int c0 = q0 - rs * cs.C;     //  C % 8 == 0, so the run stays
     const int r = fdiv(rs, cs.dS);     //  inside one (r,s))
@@ -132,10 +135,7 @@ struct FwdPatchStage {
       if (hi >= 0 && hi < cs.H && wi >= 0 && wi < cs.W) {
         const __bf16* src =
             X + (((long)n * cs.H + hi) * cs.W + wi) * cs.C + c0;
-        put8(&srow(Sm, mx)[kk0], src, true);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) srow(Sm, mx)[kk0 + j] = (__bf16)0.f;
+        vv = *(const bf16x8*)src;
       }
       return;
     }
@@ -150,28 +150,51 @@ struct FwdPatchStage {
         if (hij >= 0 && hij < cs.H && wij >= 0 && wij < cs.W)
           v = (float)X[(((long)n * cs.H + hij) * cs.W + wij) * cs.C + c];
       }
-      srow(Sm, mx)[kk0 + j] = (__bf16)v;
+      vv[j] = (__bf16)v;
       if (++c == cs.C) { c = 0; if (++ss == cs.S) { ss = 0; ++rr; } }
     }
+  }
+
+  DEVINL void commit(__bf16* Sm) {
+    *(bf16x8*)&srow(Sm, mx)[kk0] = vv;
   }
 };
 
 // Bs[k][kk]: weight memory [K][R*S*C] rows contiguous in tap order.
-DEVINL void stage_wrows(const __bf16* __restrict__ Wt, __bf16* Sn,
-                        int n0, int k0, int NROWS, int KD, int t, bool vec) {
-  const int x = t >> 2;
-  const int kk0 = (t & 3) * 8;
-  const int gx = n0 + x;
-  const int gk = k0 + kk0;
-  const __bf16* src = Wt + (long)gx * KD + gk;
-  if (gx < NROWS && gk + 8 <= KD) {
-    put8(&srow(Sn, x)[kk0], src, vec && ((gk & 7) == 0) && ((KD & 7) == 0));
-  } else {
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      srow(Sn, x)[kk0 + j] = (gx < NROWS && gk + j < KD) ? src[j] : (__bf16)0.f;
+// Split load/commit like the patch stagers.
+struct WrowStage {
+  int x, kk0;
+  bf16x8 vv;
+
+  DEVINL void init(int t) {
+    x = t >> 2;
+    kk0 = (t & 3) * 8;
   }
-}
+
+  DEVINL void load(const __bf16* __restrict__ Wt, int n0, int k0,
+                   int NROWS, int KD, bool vec) {
+    const int gx = n0 + x;
+    const int gk = k0 + kk0;
+    const __bf16* src = Wt + (long)gx * KD + gk;
+    vv = {};
+    if (gx < NROWS && gk + 8 <= KD) {
+      if (vec && ((gk & 7) == 0) && ((KD & 7) == 0)) {
+        vv = *(const bf16x8*)src;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vv[j] = src[j];
+      }
+    } else if (gx < NROWS) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (gk + j < KD) vv[j] = src[j];
+    }
+  }
+
+  DEVINL void commit(__bf16* Sn) {
+    *(bf16x8*)&srow(Sn, x)[kk0] = vv;
+  }
+};
 
 // SK: split the tap reduction over grid.z (small-spatial deep layers
 // otherwise leave most of the 256 CUs idle); slices store fp32 partial
@@ -205,15 +228,20 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
   // D: reg v -> row 8*(v>>2) + 4*(l>>5) + (v&3), col (l&31).
   f32x16 acc = {};
   FwdPatchStage pst;
+  WrowStage wst;
   pst.init(cs, tm0, M, t);
-  pst.stage(X, As[0], cs, ks, KD, cvec);
-  stage_wrows(Wt, Bs[0], tn0, ks, cs.K, KD, t, true);
+  wst.init(t);
+  pst.load(X, cs, ks, KD, cvec);
+  wst.load(Wt, tn0, ks, cs.K, KD, true);
+  pst.commit(As[0]);
+  wst.commit(Bs[0]);
   __syncthreads();
   int cur = 0;
   for (int k0 = ks; k0 < ke; k0 += BK, cur ^= 1) {
-    if (k0 + BK < ke) {
-      pst.stage(X, As[cur ^ 1], cs, k0 + BK, KD, cvec);
-      stage_wrows(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs.K, KD, t, true);
+    const bool more = k0 + BK < ke;
+    if (more) {
+      pst.load(X, cs, k0 + BK, KD, cvec);
+      wst.load(Wt, tn0, k0 + BK, cs.K, KD, true);
     }
 #pragma unroll
     for (int kh = 0; kh < 2; ++kh) {
@@ -221,6 +249,10 @@ void conv_fwd_kernel(const __bf16* __restrict__ X, const __bf16* __restrict__ Wt
       bf16x8 a = *(const bf16x8*)&srow(As[cur], wr * 32 + (lane & 31))[ko];
       bf16x8 b = *(const bf16x8*)&srow(Bs[cur], wc * 32 + (lane & 31))[ko];
       acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+    }
+    if (more) {
+      pst.commit(As[cur ^ 1]);
+      wst.commit(Bs[cur ^ 1]);
     }
     __syncthreads();
   }
@@ -313,21 +345,26 @@ struct BwddPatchStage {
     w = rem - h * cs.W;
   }
 
+  bf16x8 vv;   // in-flight tile (load() -> commit())
+
   template <int STRIDE>
-  DEVINL void stage(const __bf16* __restrict__ dY, __bf16* Sm,
-                    const ConvShape cs, int k0, int KD, bool kvec);
+  DEVINL void load(const __bf16* __restrict__ dY, const ConvShape cs,
+                   int k0, int KD, bool kvec);
+
+  DEVINL void commit(__bf16* Sm) {
+    *(bf16x8*)&srow(Sm, mx)[kk0] = vv;
+  }
 };
 
+// split load/commit (see FwdPatchStage): global reads land in
+// registers before the MFMA block; the b128 LDS store runs after it.
 template <int STRIDE>
-DEVINL void BwddPatchStage::stage(const __bf16* __restrict__ dY,
-                                  __bf16* Sm, const ConvShape cs,
-                                  int k0, int KD, bool kvec) {
+DEVINL void BwddPatchStage::load(const __bf16* __restrict__ dY,
+                                 const ConvShape cs,
+                                 int k0, int KD, bool kvec) {
   const int q0 = k0 + kk0;
-  if (q0 >= KD || !pok) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) srow(Sm, mx)[kk0 + j] = (__bf16)0.f;
-    return;
-  }
+  vv = {};
+  if (q0 >= KD || !pok) return;
   const int rs = fdiv(q0, cs.dK);
   const int kc0 = q0 - rs * cs.K;
   const int r = fdiv(rs, cs.dS);
@@ -352,10 +389,7 @@ DEVINL void BwddPatchStage::stage(const __bf16* __restrict__ dY,
     if (ok) {
       const __bf16* src =
           dY + (((long)n * cs.Ho + ho) * cs.Wo + wo) * cs.K + kc0;
-      put8(&srow(Sm, mx)[kk0], src, true);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) srow(Sm, mx)[kk0 + j] = (__bf16)0.f;
+      vv = *(const bf16x8*)src;
     }
     return;
   }
@@ -382,7 +416,7 @@ DEVINL void BwddPatchStage::stage(const __bf16* __restrict__ dY,
       if (ok)
         v = (float)dY[(((long)n * cs.Ho + ho) * cs.Wo + wo) * cs.K + kc];
     }
-    srow(Sm, mx)[kk0 + j] = (__bf16)v;
+    vv[j] = (__bf16)v;
     if (++kc == cs.K) { kc = 0; if (++ss2 == cs.S) { ss2 = 0; ++rr2; } }
   }
 }
@@ -391,33 +425,41 @@ DEVINL void BwddPatchStage::stage(const __bf16* __restrict__ dY,
 // tensor) — 8 contiguous c for one tap q=(r,s,k), transposed into LDS.
 // Kills the host-side W^T permute+copy the backward used to pay per
 // layer per step (~94 launch-bound copies across Inception).
-DEVINL void stage_w_krsc(const __bf16* __restrict__ W, __bf16* Sn,
-                         int c0blk, int q0, const ConvShape cs, int KD,
-                         int t) {
-  const int px = t >> 3;          // tap q within the BK tile (0..31)
-  const int kk0 = (t & 7) * 8;    // c chunk
-  const int q = q0 + px;
-  const int c = c0blk + kk0;
-  if (q >= KD || c >= cs.C) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) srow(Sn, kk0 + j)[px] = (__bf16)0.f;
-    return;
+// Split load/commit like the patch stagers.
+struct WkrscStage {
+  int px, kk0;
+  bf16x8 vv;
+
+  DEVINL void init(int t) {
+    px = t >> 3;          // tap q within the BK tile (0..31)
+    kk0 = (t & 7) * 8;    // c chunk
   }
-  const int rs = fdiv(q, cs.dK);
-  const int k = q - rs * cs.K;
-  const int r = fdiv(rs, cs.dS);
-  const int s = rs - r * cs.S;
-  const __bf16* src = W + (((long)k * cs.R + r) * cs.S + s) * cs.C + c;
-  if (((cs.C & 7) == 0) && c + 8 <= cs.C) {
-    bf16x8 v = *(const bf16x8*)src;
+
+  DEVINL void load(const __bf16* __restrict__ W, int c0blk, int q0,
+                   const ConvShape cs, int KD) {
+    const int q = q0 + px;
+    const int c = c0blk + kk0;
+    vv = {};
+    if (q >= KD || c >= cs.C) return;
+    const int rs = fdiv(q, cs.dK);
+    const int k = q - rs * cs.K;
+    const int r = fdiv(rs, cs.dS);
+    const int s = rs - r * cs.S;
+    const __bf16* src = W + (((long)k * cs.R + r) * cs.S + s) * cs.C + c;
+    if (((cs.C & 7) == 0) && c + 8 <= cs.C) {
+      vv = *(const bf16x8*)src;
+    } else {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) srow(Sn, kk0 + j)[px] = v[j];
-  } else {
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      srow(Sn, kk0 + j)[px] = (c + j < cs.C) ? src[j] : (__bf16)0.f;
+      for (int j = 0; j < 8; ++j)
+        if (c + j < cs.C) vv[j] = src[j];
+    }
   }
-}
+
+  DEVINL void commit(__bf16* Sn) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) srow(Sn, kk0 + j)[px] = vv[j];
+  }
+};
 
 template <int STRIDE, bool SK>
 __global__ __launch_bounds__(256)
@@ -445,15 +487,20 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
   // efficiency of the earlier 2x2 16x16x32 fragment scheme here).
   f32x16 acc = {};
   BwddPatchStage pst;
+  WkrscStage wst;
   pst.init(cs, tm0, M, t);
-  pst.stage<STRIDE>(dY, As[0], cs, ks, KD, kvec);
-  stage_w_krsc(Wt, Bs[0], tn0, ks, cs, KD, t);
+  wst.init(t);
+  pst.load<STRIDE>(dY, cs, ks, KD, kvec);
+  wst.load(Wt, tn0, ks, cs, KD);
+  pst.commit(As[0]);
+  wst.commit(Bs[0]);
   __syncthreads();
   int cur = 0;
   for (int k0 = ks; k0 < ke; k0 += BK, cur ^= 1) {
-    if (k0 + BK < ke) {
-      pst.stage<STRIDE>(dY, As[cur ^ 1], cs, k0 + BK, KD, kvec);
-      stage_w_krsc(Wt, Bs[cur ^ 1], tn0, k0 + BK, cs, KD, t);
+    const bool more = k0 + BK < ke;
+    if (more) {
+      pst.load<STRIDE>(dY, cs, k0 + BK, KD, kvec);
+      wst.load(Wt, tn0, k0 + BK, cs, KD);
     }
 #pragma unroll
     for (int kh = 0; kh < 2; ++kh) {
@@ -461,6 +508,10 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
       bf16x8 a = *(const bf16x8*)&srow(As[cur], wr * 32 + (lane & 31))[ko];
       bf16x8 b = *(const bf16x8*)&srow(Bs[cur], wc * 32 + (lane & 31))[ko];
       acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+    }
+    if (more) {
+      pst.commit(As[cur ^ 1]);
+      wst.commit(Bs[cur ^ 1]);
     }
     __syncthreads();
   }
@@ -507,6 +558,7 @@ struct DyBwdwStage {
   bool kok;
   long p;              // first pixel of this thread's chunk
   const __bf16* src;   // advances by BK*K per tile
+  bf16x8 v;            // in-flight tile (load() -> commit())
 
   DEVINL void init(const __bf16* __restrict__ dY, const ConvShape& cs,
                    int m0, long p0, int t, bool kvec) {
@@ -518,8 +570,13 @@ struct DyBwdwStage {
     src = dY + p * cs.K + (m0 + k);
   }
 
-  DEVINL void stage(__bf16* Sm, const ConvShape cs, long Ptot) {
-    bf16x8 v = {};
+  // split load/commit: load() issues the global reads into registers
+  // BEFORE the MFMA block so their latency is covered by compute;
+  // commit() (after the MFMAs) only pays the vmcnt drain + LDS write.
+  // The fused form parked every wave on vmcnt ahead of its own MFMAs
+  // (profiles/r01_inception_pmc.txt: SQ_WAIT 150x the MFMA count).
+  DEVINL void load(const ConvShape cs, long Ptot) {
+    v = {};
     if (kok) {
       const long left = Ptot - p;
       if (left >= 8) {
@@ -531,9 +588,12 @@ struct DyBwdwStage {
         for (int j = 0; j < (int)left; ++j) { v[j] = *s; s += cs.K; }
       }
     }
-    *(bf16x8*)&srow(Sm, k)[pxc] = v;
     p += BK;
     src += (long)BK * cs.K;
+  }
+
+  DEVINL void commit(__bf16* Sm) {
+    *(bf16x8*)&srow(Sm, k)[pxc] = v;
   }
 };
 
@@ -550,6 +610,7 @@ struct XBwdwStage {
   long p;            // current reduction pixel = p0 + px
   int n;             // p = n*HoWo + rem
   int rem, HoWo;
+  bf16x8 vv;         // in-flight tile (load() -> commit())
 
   DEVINL void init(const ConvShape& cs, int n0, long p0, int KD, int t,
                    bool cvec) {
@@ -569,29 +630,24 @@ struct XBwdwStage {
     rem = (int)(p - (long)n * HoWo);
   }
 
-  // stage the current tile then advance by BK pixels
-  DEVINL void stage(const __bf16* __restrict__ X, __bf16* Sn,
-                    const ConvShape cs, int KD, long Ptot) {
+  // split load/commit (see DyBwdwStage): global reads land in
+  // registers before the MFMA block; the transposed LDS scatter waits
+  // for them only AFTER the MFMAs are issued.
+  DEVINL void load(const __bf16* __restrict__ X, const ConvShape cs,
+                   int KD, long Ptot) {
     const int ho = fdiv(rem, cs.dWo);
     const int wo = rem - ho * cs.Wo;
     const bool pok = p < Ptot;
+    vv = {};
     if (vec8) {
       const int hi = ho * cs.U + r - cs.P;
       const int wi = wo * cs.V + s - cs.Q;
       if (pok && hi >= 0 && hi < cs.H && wi >= 0 && wi < cs.W) {
         const __bf16* src =
             X + (((long)n * cs.H + hi) * cs.W + wi) * cs.C + c0;
-        bf16x8 v = *(const bf16x8*)src;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) srow(Sn, qq0 + j)[px] = v[j];
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) srow(Sn, qq0 + j)[px] = (__bf16)0.f;
+        vv = *(const bf16x8*)src;
       }
-    } else if (!tapok || !pok) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) srow(Sn, qq0 + j)[px] = (__bf16)0.f;
-    } else {
+    } else if (tapok && pok) {
       // generic path: the 8 taps may straddle (r,s) blocks (small C)
       int c = c0, rr2 = r, ss2 = s;
 #pragma unroll
@@ -603,13 +659,20 @@ struct XBwdwStage {
           if (hij >= 0 && hij < cs.H && wij >= 0 && wij < cs.W)
             v = (float)X[(((long)n * cs.H + hij) * cs.W + wij) * cs.C + c];
         }
-        srow(Sn, qq0 + j)[px] = (__bf16)v;
+        vv[j] = (__bf16)v;
         if (++c == cs.C) { c = 0; if (++ss2 == cs.S) { ss2 = 0; ++rr2; } }
       }
     }
     p += BK;
     rem += BK;
     while (rem >= HoWo) { rem -= HoWo; ++n; }
+  }
+
+  DEVINL void commit(__bf16* Sn) {
+    // transposed scatter: 8 rows, same column px (8-row skew keeps the
+    // groups conflict-free)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) srow(Sn, qq0 + j)[px] = vv[j];
   }
 };
 
@@ -642,14 +705,20 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
   XBwdwStage xst;
   dst.init(dY, cs, tm0, ps, t, kvec);
   xst.init(cs, tn0, ps, KD, t, cvec);
-  dst.stage(As[0], cs, Ptot);
-  xst.stage(X, Bs[0], cs, KD, Ptot);
+  dst.load(cs, Ptot);
+  xst.load(X, cs, KD, Ptot);
+  dst.commit(As[0]);
+  xst.commit(Bs[0]);
   __syncthreads();
   int cur = 0;
   for (long p0 = ps; p0 < pe; p0 += BK, cur ^= 1) {
-    if (p0 + BK < pe) {
-      dst.stage(As[cur ^ 1], cs, Ptot);
-      xst.stage(X, Bs[cur ^ 1], cs, KD, Ptot);
+    const bool more = p0 + BK < pe;
+    if (more) {
+      // next tile's global reads first: their latency is covered by
+      // this tile's fragment reads + MFMAs (the fused stage parked on
+      // vmcnt before any MFMA issued)
+      dst.load(cs, Ptot);
+      xst.load(X, cs, KD, Ptot);
     }
     const int kfrag = (lane >> 4) * 8;
     bf16x8 bfrag[2];
@@ -665,6 +734,10 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
       for (int fn = 0; fn < 2; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a, bfrag[fn], acc[fm][fn], 0, 0, 0);
+    }
+    if (more) {
+      dst.commit(As[cur ^ 1]);
+      xst.commit(Bs[cur ^ 1]);
     }
     __syncthreads();
   }
