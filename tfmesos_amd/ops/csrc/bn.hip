@@ -308,7 +308,299 @@ void bn_bwd_apply_kernel(const __bf16* __restrict__ x,
   }
 }
 
+// ------------------------------------------------------------ row-wise path
+//
+// For C % 8 == 0 (every Inception BN layer): thread t owns a FIXED
+// 8-channel octet u = t % (C/8) and walks pixel rows. Per-channel
+// constants live in REGISTERS (loaded once per thread), the hot loop
+// is one b128 load (+1 store) per 8 elements with zero LDS traffic —
+// the flat-walk kernels below gathered 6 per-channel constants from
+// LDS per element and the PMC showed bank-conflict counts 10x the LDS
+// instruction count (profiles/r01_inception_pmc.txt), capping them at
+// 1.7-3 TB/s on an 6.3 TB/s chip.
+// Stats partials are written [c][z] (channel-major) so the finalize
+// lanes read CONSECUTIVE z — coalesced — instead of striding C.
+
+// grid (Z); block 256 = (C/8) channel octets x rpb rows
+__global__ __launch_bounds__(256)
+void bn_stats_rw_kernel(const __bf16* __restrict__ x,
+                        float* __restrict__ part,
+                        long P, int C, int Z) {
+  const int tpr = C >> 3;
+  const int u = (int)(threadIdx.x % tpr);
+  const int rl = (int)(threadIdx.x / tpr);
+  const int rpb = 256 / tpr < 1 ? 1 : 256 / tpr;
+  const int c0 = u * 8;
+  const int z = blockIdx.x;
+  float sum[8] = {}, sq[8] = {};
+  if (rl < rpb) {
+    const long rstep = (long)Z * rpb;
+    for (long r = (long)z * rpb + rl; r < P; r += rstep) {
+      const bf16x8 v = *(const bf16x8*)&x[r * C + c0];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = (float)v[j];
+        sum[j] += f;
+        sq[j] += f * f;
+      }
+    }
+  }
+  // cross-row reduce: rpb partial rows per channel octet
+  __shared__ float red[256 * 8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) red[threadIdx.x * 8 + j] = sum[j];
+  __syncthreads();
+  if (rl == 0) {
+    for (int rr = 1; rr < rpb; ++rr)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) red[u * 8 + j] += red[(rr * tpr + u) * 8 + j];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) sum[j] = red[u * 8 + j];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int j = 0; j < 8; ++j) red[threadIdx.x * 8 + j] = sq[j];
+  __syncthreads();
+  if (rl == 0) {
+    for (int rr = 1; rr < rpb; ++rr)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) red[u * 8 + j] += red[(rr * tpr + u) * 8 + j];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      // channel-major partials: [c][z]{sum, sq}
+      part[((long)(c0 + j) * Z + z) * 2] = sum[j];
+      part[((long)(c0 + j) * Z + z) * 2 + 1] = red[u * 8 + j];
+    }
+  }
+}
+
+// block 256 = 4 waves, one channel per wave; lanes stride z (coalesced
+// in the [c][z] layout)
+__global__ __launch_bounds__(256)
+void bn_finalize_rw_kernel(const float* __restrict__ part,
+                           float* __restrict__ mean,
+                           float* __restrict__ invstd,
+                           int C, int Z, float inv_count, float eps) {
+  const int c = blockIdx.x * 4 + (threadIdx.x >> 6);
+  const int l = threadIdx.x & 63;
+  if (c >= C) return;
+  float s = 0.f, q = 0.f;
+  for (int z = l; z < Z; z += 64) {
+    s += part[((long)c * Z + z) * 2];
+    q += part[((long)c * Z + z) * 2 + 1];
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s += __shfl_xor(s, off, 64);
+    q += __shfl_xor(q, off, 64);
+  }
+  if (l == 0) {
+    const float m = s * inv_count;
+    float var = q * inv_count - m * m;
+    if (var < 0.f) var = 0.f;
+    mean[c] = m;
+    invstd[c] = __frsqrt_rn(var + eps);
+  }
+}
+
+template <bool RELU>
+__global__ __launch_bounds__(256)
+void bn_apply_rw_kernel(const __bf16* __restrict__ x,
+                        const float* __restrict__ mean,
+                        const float* __restrict__ invstd,
+                        const __bf16* __restrict__ g,
+                        const __bf16* __restrict__ b,
+                        __bf16* __restrict__ y, long ldo, long P, int C) {
+  const int tpr = C >> 3;
+  const int u = (int)(threadIdx.x % tpr);
+  const int rl = (int)(threadIdx.x / tpr);
+  const int rpb = 256 / tpr < 1 ? 1 : 256 / tpr;
+  if (rl >= rpb) return;
+  const int c0 = u * 8;
+  float sc[8], sh[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const float s = (float)g[c0 + j] * invstd[c0 + j];
+    sc[j] = s;
+    sh[j] = (float)b[c0 + j] - mean[c0 + j] * s;
+  }
+  const long rstep = (long)gridDim.x * rpb;
+  for (long r = (long)blockIdx.x * rpb + rl; r < P; r += rstep) {
+    const bf16x8 v = *(const bf16x8*)&x[r * C + c0];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = (float)v[j] * sc[j] + sh[j];
+      if (RELU) f = f > 0.f ? f : 0.f;
+      o[j] = (__bf16)f;
+    }
+    *(bf16x8*)&y[r * ldo + c0] = o;
+  }
+}
+
+template <bool RELU>
+__global__ __launch_bounds__(256)
+void bn_bwd_stats_rw_kernel(const __bf16* __restrict__ x,
+                            const __bf16* __restrict__ dy, long ldy,
+                            const __bf16* __restrict__ g,
+                            const __bf16* __restrict__ b,
+                            const float* __restrict__ mean,
+                            const float* __restrict__ invstd,
+                            float* __restrict__ part, long P, int C, int Z) {
+  const int tpr = C >> 3;
+  const int u = (int)(threadIdx.x % tpr);
+  const int rl = (int)(threadIdx.x / tpr);
+  const int rpb = 256 / tpr < 1 ? 1 : 256 / tpr;
+  const int c0 = u * 8;
+  const int z = blockIdx.x;
+  float s1[8] = {}, s2[8] = {};
+  if (rl < rpb) {
+    float mu[8], is[8], sc[8], sh[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      mu[j] = mean[c0 + j];
+      is[j] = invstd[c0 + j];
+      const float s = RELU ? (float)g[c0 + j] * is[j] : 0.f;
+      sc[j] = s;
+      sh[j] = RELU ? (float)b[c0 + j] - mu[j] * s : 0.f;
+    }
+    const long rstep = (long)Z * rpb;
+    for (long r = (long)z * rpb + rl; r < P; r += rstep) {
+      const bf16x8 xv = *(const bf16x8*)&x[r * C + c0];
+      const bf16x8 dv = *(const bf16x8*)&dy[r * ldy + c0];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float xf = (float)xv[j];
+        float d = (float)dv[j];
+        // relu mask recomputed EXACTLY as the forward wrote it
+        if (RELU && (float)(__bf16)(xf * sc[j] + sh[j]) <= 0.f) d = 0.f;
+        s1[j] += d;
+        s2[j] += d * ((xf - mu[j]) * is[j]);
+      }
+    }
+  }
+  __shared__ float red[256 * 8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) red[threadIdx.x * 8 + j] = s1[j];
+  __syncthreads();
+  if (rl == 0) {
+    for (int rr = 1; rr < rpb; ++rr)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) red[u * 8 + j] += red[(rr * tpr + u) * 8 + j];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) s1[j] = red[u * 8 + j];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int j = 0; j < 8; ++j) red[threadIdx.x * 8 + j] = s2[j];
+  __syncthreads();
+  if (rl == 0) {
+    for (int rr = 1; rr < rpb; ++rr)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) red[u * 8 + j] += red[(rr * tpr + u) * 8 + j];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      part[((long)(c0 + j) * Z + z) * 2] = s1[j];
+      part[((long)(c0 + j) * Z + z) * 2 + 1] = red[u * 8 + j];
+    }
+  }
+}
+
+__global__ __launch_bounds__(256)
+void bn_bwd_finalize_rw_kernel(const float* __restrict__ part,
+                               __bf16* __restrict__ dgamma,
+                               __bf16* __restrict__ dbeta,
+                               float* __restrict__ s1n,
+                               float* __restrict__ s2n,
+                               int C, int Z, float inv_count) {
+  const int c = blockIdx.x * 4 + (threadIdx.x >> 6);
+  const int l = threadIdx.x & 63;
+  if (c >= C) return;
+  float s1 = 0.f, s2 = 0.f;
+  for (int z = l; z < Z; z += 64) {
+    s1 += part[((long)c * Z + z) * 2];
+    s2 += part[((long)c * Z + z) * 2 + 1];
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s1 += __shfl_xor(s1, off, 64);
+    s2 += __shfl_xor(s2, off, 64);
+  }
+  if (l == 0) {
+    dbeta[c] = (__bf16)s1;
+    dgamma[c] = (__bf16)s2;
+    s1n[c] = s1 * inv_count;
+    s2n[c] = s2 * inv_count;
+  }
+}
+
+template <bool RELU>
+__global__ __launch_bounds__(256)
+void bn_bwd_apply_rw_kernel(const __bf16* __restrict__ x,
+                            const __bf16* __restrict__ dy, long ldy,
+                            const __bf16* __restrict__ g,
+                            const __bf16* __restrict__ b,
+                            const float* __restrict__ mean,
+                            const float* __restrict__ invstd,
+                            const float* __restrict__ s1n,
+                            const float* __restrict__ s2n,
+                            __bf16* __restrict__ dx, long P, int C) {
+  const int tpr = C >> 3;
+  const int u = (int)(threadIdx.x % tpr);
+  const int rl = (int)(threadIdx.x / tpr);
+  const int rpb = 256 / tpr < 1 ? 1 : 256 / tpr;
+  if (rl >= rpb) return;
+  const int c0 = u * 8;
+  float gs[8], a[8], bb[8], mu[8], is[8], sh[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = c0 + j;
+    is[j] = invstd[c];
+    mu[j] = mean[c];
+    const float s = (float)g[c] * is[j];
+    gs[j] = s;
+    a[j] = s1n[c];
+    bb[j] = s2n[c];
+    sh[j] = (float)b[c] - mu[j] * s;
+  }
+  const long rstep = (long)gridDim.x * rpb;
+  for (long r = (long)blockIdx.x * rpb + rl; r < P; r += rstep) {
+    const bf16x8 xv = *(const bf16x8*)&x[r * C + c0];
+    const bf16x8 dv = *(const bf16x8*)&dy[r * ldy + c0];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float xf = (float)xv[j];
+      float d = (float)dv[j];
+      const float xh = (xf - mu[j]) * is[j];
+      if (RELU && (float)(__bf16)(xf * gs[j] + sh[j]) <= 0.f) d = 0.f;
+      o[j] = (__bf16)(gs[j] * (d - a[j] - xh * bb[j]));
+    }
+    *(bf16x8*)&dx[r * C + c0] = o;
+  }
+}
+
+inline bool rowwise_ok(int C) { return (C & 7) == 0 && C <= MAXC; }
+
+inline unsigned rw_grid(long P, int C) {
+  const int rpb = 256 / (C >> 3) < 1 ? 1 : 256 / (C >> 3);
+  long wgs = (P + rpb - 1) / rpb;
+  if (wgs > 4096) wgs = 4096;
+  if (wgs < 1) wgs = 1;
+  return (unsigned)wgs;
+}
+
 inline int stats_slices(long P, int C) {
+  if (rowwise_ok(C)) {
+    // row-wise stats: grid = Z blocks, each covering all C and rpb
+    // rows per pass; aim for >=4 passes per block, cap at 2048
+    const int t = 256 / (C >> 3);
+    const int rpb = t < 1 ? 1 : t;
+    long z = P / ((long)rpb * 4);
+    if (z > 2048) z = 2048;
+    if (z < 1) z = 1;
+    return (int)z;
+  }
   // target >=2048 workgroups across the (C/64) x Z grid, each slice
   // covering >=8 pixel rounds of 4 rows. The cap matters: narrow
   // layers (C=64 -> one channel column) need Z ~ 2048 to cover the
@@ -336,6 +628,23 @@ void launch_bn_fwd(const bf16_t* x, const bf16_t* g, const bf16_t* b,
                    bf16_t* y, long ldo, float* mean, float* invstd,
                    float* part, long P, int C, int Z, float eps, bool relu,
                    hipStream_t stream) {
+  if (rowwise_ok(C)) {
+    hipLaunchKernelGGL(bn_stats_rw_kernel, dim3(Z), dim3(256), 0, stream,
+                       (const __bf16*)x, part, P, C, Z);
+    hipLaunchKernelGGL(bn_finalize_rw_kernel, dim3(ceil_div(C, 4)),
+                       dim3(256), 0, stream, part, mean, invstd, C, Z,
+                       1.f / (float)P, eps);
+    dim3 ag(rw_grid(P, C)), ab(256);
+    if (relu)
+      hipLaunchKernelGGL((bn_apply_rw_kernel<true>), ag, ab, 0, stream,
+                         (const __bf16*)x, mean, invstd, (const __bf16*)g,
+                         (const __bf16*)b, (__bf16*)y, ldo, P, C);
+    else
+      hipLaunchKernelGGL((bn_apply_rw_kernel<false>), ag, ab, 0, stream,
+                         (const __bf16*)x, mean, invstd, (const __bf16*)g,
+                         (const __bf16*)b, (__bf16*)y, ldo, P, C);
+    return;
+  }
   dim3 sg(ceil_div(C, 64), Z), sb(256);
   hipLaunchKernelGGL(bn_stats_kernel, sg, sb, 0, stream, (const __bf16*)x,
                      part, P, C, Z);
@@ -365,6 +674,27 @@ void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, long ldy,
                    const float* invstd, bf16_t* dx, bf16_t* dgamma,
                    bf16_t* dbeta, float* part, float* s1n, float* s2n,
                    long P, int C, int Z, bool relu, hipStream_t stream) {
+  if (rowwise_ok(C)) {
+#define RWS(RELUv)                                                          \
+    hipLaunchKernelGGL((bn_bwd_stats_rw_kernel<RELUv>), dim3(Z), dim3(256), \
+                       0, stream, (const __bf16*)x, (const __bf16*)dy, ldy, \
+                       (const __bf16*)g, (const __bf16*)b, mean, invstd,    \
+                       part, P, C, Z)
+    if (relu) RWS(true); else RWS(false);
+#undef RWS
+    hipLaunchKernelGGL(bn_bwd_finalize_rw_kernel, dim3(ceil_div(C, 4)),
+                       dim3(256), 0, stream, part, (__bf16*)dgamma,
+                       (__bf16*)dbeta, s1n, s2n, C, Z, 1.f / (float)P);
+    dim3 ag(rw_grid(P, C)), ab(256);
+#define RWA(RELUv)                                                          \
+    hipLaunchKernelGGL((bn_bwd_apply_rw_kernel<RELUv>), ag, ab, 0, stream,  \
+                       (const __bf16*)x, (const __bf16*)dy, ldy,            \
+                       (const __bf16*)g, (const __bf16*)b, mean, invstd,    \
+                       s1n, s2n, (__bf16*)dx, P, C)
+    if (relu) RWA(true); else RWA(false);
+#undef RWA
+    return;
+  }
   dim3 sg(ceil_div(C, 64), Z), sb(256);
   if (relu)
     hipLaunchKernelGGL((bn_bwd_stats_kernel<true>), sg, sb, 0, stream,

@@ -700,7 +700,10 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
   const int wave = t >> 6;
   const int wr = wave >> 1, wc = wave & 1;
 
-  f32x4 acc[2][2] = {};
+  // 32x32x16 MFMA core (same scheme as conv_fwd/bwdd: one 32x32 wave
+  // tile, 2 MFMA + 4 fragment reads per BK=32 iteration — vs 4 MFMA +
+  // 6 reads with the earlier 2x2 16x16x32 fragments).
+  f32x16 acc = {};
   DyBwdwStage dst;
   XBwdwStage xst;
   dst.init(dY, cs, tm0, ps, t, kvec);
@@ -720,20 +723,12 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
       dst.load(cs, Ptot);
       xst.load(X, cs, KD, Ptot);
     }
-    const int kfrag = (lane >> 4) * 8;
-    bf16x8 bfrag[2];
 #pragma unroll
-    for (int fn = 0; fn < 2; ++fn)
-      bfrag[fn] =
-          *(const bf16x8*)&srow(Bs[cur], wc * 32 + fn * 16 + (lane & 15))[kfrag];
-#pragma unroll
-    for (int fm = 0; fm < 2; ++fm) {
-      bf16x8 a =
-          *(const bf16x8*)&srow(As[cur], wr * 32 + fm * 16 + (lane & 15))[kfrag];
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
-        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a, bfrag[fn], acc[fm][fn], 0, 0, 0);
+    for (int kh = 0; kh < 2; ++kh) {
+      const int ko = kh * 16 + ((lane >> 5) << 3);
+      bf16x8 a = *(const bf16x8*)&srow(As[cur], wr * 32 + (lane & 31))[ko];
+      bf16x8 b = *(const bf16x8*)&srow(Bs[cur], wc * 32 + (lane & 31))[ko];
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
     }
     if (more) {
       dst.commit(As[cur ^ 1]);
@@ -742,22 +737,21 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
     __syncthreads();
   }
 
+  // D: reg v -> A-row (k) 8*(v>>2) + 4*(lane>>5) + (v&3), B-col (q)
+  // lane&31 — q contiguous across lanes, so the dW stores coalesce
+  const int q = tn0 + wc * 32 + (lane & 31);
+  if (q < KD) {
 #pragma unroll
-  for (int fm = 0; fm < 2; ++fm)
-#pragma unroll
-    for (int fn = 0; fn < 2; ++fn) {
-      const int q = tn0 + wc * 32 + fn * 16 + (lane & 15);
-      if (q >= KD) continue;
-#pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const int k = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + rr;
-        if (k >= cs.K) continue;
-        if (gridDim.z == 1)
-          dW[(long)k * KD + q] = acc[fm][fn][rr];
-        else
-          unsafeAtomicAdd(&dW[(long)k * KD + q], acc[fm][fn][rr]);
-      }
+    for (int v = 0; v < 16; ++v) {
+      const int k = tm0 + wr * 32 + ((v >> 2) << 3) +
+                    ((lane >> 5) << 2) + (v & 3);
+      if (k >= cs.K) continue;
+      if (gridDim.z == 1)
+        dW[(long)k * KD + q] = acc[v];
+      else
+        unsafeAtomicAdd(&dW[(long)k * KD + q], acc[v]);
     }
+  }
 }
 
 }  // namespace
