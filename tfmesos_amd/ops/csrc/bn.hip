@@ -335,7 +335,23 @@ void bn_stats_rw_kernel(const __bf16* __restrict__ x,
   float sum[8] = {}, sq[8] = {};
   if (rl < rpb) {
     const long rstep = (long)Z * rpb;
-    for (long r = (long)z * rpb + rl; r < P; r += rstep) {
+    long r = (long)z * rpb + rl;
+    // 4 rows of loads in flight: the single-load loop was
+    // latency-bound (one dependent accumulate per HBM round trip)
+    for (; r + 3 * rstep < P; r += 4 * rstep) {
+      const bf16x8 v0 = *(const bf16x8*)&x[r * C + c0];
+      const bf16x8 v1 = *(const bf16x8*)&x[(r + rstep) * C + c0];
+      const bf16x8 v2 = *(const bf16x8*)&x[(r + 2 * rstep) * C + c0];
+      const bf16x8 v3 = *(const bf16x8*)&x[(r + 3 * rstep) * C + c0];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f0 = (float)v0[j], f1 = (float)v1[j];
+        const float f2 = (float)v2[j], f3 = (float)v3[j];
+        sum[j] += (f0 + f1) + (f2 + f3);
+        sq[j] += (f0 * f0 + f1 * f1) + (f2 * f2 + f3 * f3);
+      }
+    }
+    for (; r < P; r += rstep) {
       const bf16x8 v = *(const bf16x8*)&x[r * C + c0];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -425,7 +441,26 @@ void bn_apply_rw_kernel(const __bf16* __restrict__ x,
     sh[j] = (float)b[c0 + j] - mean[c0 + j] * s;
   }
   const long rstep = (long)gridDim.x * rpb;
-  for (long r = (long)blockIdx.x * rpb + rl; r < P; r += rstep) {
+  long r = (long)blockIdx.x * rpb + rl;
+  for (; r + rstep < P; r += 2 * rstep) {   // 2 loads in flight
+    const bf16x8 v0 = *(const bf16x8*)&x[r * C + c0];
+    const bf16x8 v1 = *(const bf16x8*)&x[(r + rstep) * C + c0];
+    bf16x8 o0, o1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f0 = (float)v0[j] * sc[j] + sh[j];
+      float f1 = (float)v1[j] * sc[j] + sh[j];
+      if (RELU) {
+        f0 = f0 > 0.f ? f0 : 0.f;
+        f1 = f1 > 0.f ? f1 : 0.f;
+      }
+      o0[j] = (__bf16)f0;
+      o1[j] = (__bf16)f1;
+    }
+    *(bf16x8*)&y[r * ldo + c0] = o0;
+    *(bf16x8*)&y[(r + rstep) * ldo + c0] = o1;
+  }
+  for (; r < P; r += rstep) {
     const bf16x8 v = *(const bf16x8*)&x[r * C + c0];
     bf16x8 o;
 #pragma unroll
@@ -465,14 +500,31 @@ void bn_bwd_stats_rw_kernel(const __bf16* __restrict__ x,
       sh[j] = RELU ? (float)b[c0 + j] - mu[j] * s : 0.f;
     }
     const long rstep = (long)Z * rpb;
-    for (long r = (long)z * rpb + rl; r < P; r += rstep) {
+    long r = (long)z * rpb + rl;
+    // 2 rows x (x, dy) = 4 loads in flight (see bn_stats_rw_kernel)
+    for (; r + rstep < P; r += 2 * rstep) {
+      const bf16x8 xv0 = *(const bf16x8*)&x[r * C + c0];
+      const bf16x8 dv0 = *(const bf16x8*)&dy[r * ldy + c0];
+      const bf16x8 xv1 = *(const bf16x8*)&x[(r + rstep) * C + c0];
+      const bf16x8 dv1 = *(const bf16x8*)&dy[(r + rstep) * ldy + c0];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float x0 = (float)xv0[j], x1 = (float)xv1[j];
+        float d0 = (float)dv0[j], d1 = (float)dv1[j];
+        // relu mask recomputed EXACTLY as the forward wrote it
+        if (RELU && (float)(__bf16)(x0 * sc[j] + sh[j]) <= 0.f) d0 = 0.f;
+        if (RELU && (float)(__bf16)(x1 * sc[j] + sh[j]) <= 0.f) d1 = 0.f;
+        s1[j] += d0 + d1;
+        s2[j] += d0 * ((x0 - mu[j]) * is[j]) + d1 * ((x1 - mu[j]) * is[j]);
+      }
+    }
+    for (; r < P; r += rstep) {
       const bf16x8 xv = *(const bf16x8*)&x[r * C + c0];
       const bf16x8 dv = *(const bf16x8*)&dy[r * ldy + c0];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const float xf = (float)xv[j];
         float d = (float)dv[j];
-        // relu mask recomputed EXACTLY as the forward wrote it
         if (RELU && (float)(__bf16)(xf * sc[j] + sh[j]) <= 0.f) d = 0.f;
         s1[j] += d;
         s2[j] += d * ((xf - mu[j]) * is[j]);
@@ -564,7 +616,28 @@ void bn_bwd_apply_rw_kernel(const __bf16* __restrict__ x,
     sh[j] = (float)b[c] - mu[j] * s;
   }
   const long rstep = (long)gridDim.x * rpb;
-  for (long r = (long)blockIdx.x * rpb + rl; r < P; r += rstep) {
+  long r = (long)blockIdx.x * rpb + rl;
+  for (; r + rstep < P; r += 2 * rstep) {   // 4 loads in flight
+    const bf16x8 xv0 = *(const bf16x8*)&x[r * C + c0];
+    const bf16x8 dv0 = *(const bf16x8*)&dy[r * ldy + c0];
+    const bf16x8 xv1 = *(const bf16x8*)&x[(r + rstep) * C + c0];
+    const bf16x8 dv1 = *(const bf16x8*)&dy[(r + rstep) * ldy + c0];
+    bf16x8 o0, o1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float x0 = (float)xv0[j], x1 = (float)xv1[j];
+      float d0 = (float)dv0[j], d1 = (float)dv1[j];
+      const float xh0 = (x0 - mu[j]) * is[j];
+      const float xh1 = (x1 - mu[j]) * is[j];
+      if (RELU && (float)(__bf16)(x0 * gs[j] + sh[j]) <= 0.f) d0 = 0.f;
+      if (RELU && (float)(__bf16)(x1 * gs[j] + sh[j]) <= 0.f) d1 = 0.f;
+      o0[j] = (__bf16)(gs[j] * (d0 - a[j] - xh0 * bb[j]));
+      o1[j] = (__bf16)(gs[j] * (d1 - a[j] - xh1 * bb[j]));
+    }
+    *(bf16x8*)&dx[r * C + c0] = o0;
+    *(bf16x8*)&dx[(r + rstep) * C + c0] = o1;
+  }
+  for (; r < P; r += rstep) {
     const bf16x8 xv = *(const bf16x8*)&x[r * C + c0];
     const bf16x8 dv = *(const bf16x8*)&dy[r * ldy + c0];
     bf16x8 o;
