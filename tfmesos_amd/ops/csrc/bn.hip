@@ -663,13 +663,26 @@ inline unsigned rw_grid(long P, int C) {
   return (unsigned)wgs;
 }
 
+// Row-wise stats only where the 8-channels-per-thread granularity
+// still yields enough BLOCKS to cover the chip (>=512 at 2 passes per
+// thread): small-P layers have 8x more parallelism at the old
+// one-channel-per-thread granularity and measured FASTER there (the
+// first rowwise cut regressed mid-size layers 2x on exactly this).
+inline bool rowwise_stats(long P, int C) {
+  if (!rowwise_ok(C)) return false;
+  const int t = 256 / (C >> 3);
+  const int rpb = t < 1 ? 1 : t;
+  return P / ((long)rpb * 2) >= 512;
+}
+
 inline int stats_slices(long P, int C) {
-  if (rowwise_ok(C)) {
+  if (rowwise_stats(P, C)) {
     // row-wise stats: grid = Z blocks, each covering all C and rpb
-    // rows per pass; aim for >=4 passes per block, cap at 2048
+    // rows per pass; ~2 passes per thread (2 loads in flight), more
+    // via the 4-deep unroll when P allows
     const int t = 256 / (C >> 3);
     const int rpb = t < 1 ? 1 : t;
-    long z = P / ((long)rpb * 4);
+    long z = (P + 2L * rpb - 1) / (2L * rpb);
     if (z > 2048) z = 2048;
     if (z < 1) z = 1;
     return (int)z;
@@ -701,12 +714,21 @@ void launch_bn_fwd(const bf16_t* x, const bf16_t* g, const bf16_t* b,
                    bf16_t* y, long ldo, float* mean, float* invstd,
                    float* part, long P, int C, int Z, float eps, bool relu,
                    hipStream_t stream) {
-  if (rowwise_ok(C)) {
+  if (rowwise_stats(P, C)) {
     hipLaunchKernelGGL(bn_stats_rw_kernel, dim3(Z), dim3(256), 0, stream,
                        (const __bf16*)x, part, P, C, Z);
     hipLaunchKernelGGL(bn_finalize_rw_kernel, dim3(ceil_div(C, 4)),
                        dim3(256), 0, stream, part, mean, invstd, C, Z,
                        1.f / (float)P, eps);
+  } else {
+    dim3 sg(ceil_div(C, 64), Z), sb(256);
+    hipLaunchKernelGGL(bn_stats_kernel, sg, sb, 0, stream, (const __bf16*)x,
+                       part, P, C, Z);
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64),
+                       0, stream, part, mean, invstd, C, Z, 1.f / (float)P,
+                       eps);
+  }
+  if (rowwise_ok(C)) {
     dim3 ag(rw_grid(P, C)), ab(256);
     if (relu)
       hipLaunchKernelGGL((bn_apply_rw_kernel<true>), ag, ab, 0, stream,
@@ -718,12 +740,6 @@ void launch_bn_fwd(const bf16_t* x, const bf16_t* g, const bf16_t* b,
                          (const __bf16*)b, (__bf16*)y, ldo, P, C);
     return;
   }
-  dim3 sg(ceil_div(C, 64), Z), sb(256);
-  hipLaunchKernelGGL(bn_stats_kernel, sg, sb, 0, stream, (const __bf16*)x,
-                     part, P, C, Z);
-  hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64),
-                     0, stream, part, mean, invstd, C, Z, 1.f / (float)P,
-                     eps);
   dim3 ag(ew_grid(P * (long)C)), ab(256);
   const bool vec = (C & 7) == 0;
   const bool strided = ldo != C;
@@ -747,7 +763,7 @@ void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, long ldy,
                    const float* invstd, bf16_t* dx, bf16_t* dgamma,
                    bf16_t* dbeta, float* part, float* s1n, float* s2n,
                    long P, int C, int Z, bool relu, hipStream_t stream) {
-  if (rowwise_ok(C)) {
+  if (rowwise_stats(P, C)) {
 #define RWS(RELUv)                                                          \
     hipLaunchKernelGGL((bn_bwd_stats_rw_kernel<RELUv>), dim3(Z), dim3(256), \
                        0, stream, (const __bf16*)x, (const __bf16*)dy, ldy, \
@@ -758,6 +774,23 @@ void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, long ldy,
     hipLaunchKernelGGL(bn_bwd_finalize_rw_kernel, dim3(ceil_div(C, 4)),
                        dim3(256), 0, stream, part, (__bf16*)dgamma,
                        (__bf16*)dbeta, s1n, s2n, C, Z, 1.f / (float)P);
+  } else {
+    dim3 sg(ceil_div(C, 64), Z), sb(256);
+    if (relu)
+      hipLaunchKernelGGL((bn_bwd_stats_kernel<true>), sg, sb, 0, stream,
+                         (const __bf16*)x, (const __bf16*)dy, ldy,
+                         (const __bf16*)g, (const __bf16*)b, mean, invstd,
+                         part, P, C, Z);
+    else
+      hipLaunchKernelGGL((bn_bwd_stats_kernel<false>), sg, sb, 0, stream,
+                         (const __bf16*)x, (const __bf16*)dy, ldy,
+                         (const __bf16*)g, (const __bf16*)b, mean, invstd,
+                         part, P, C, Z);
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(64), 0, stream,
+                       part, (__bf16*)dgamma, (__bf16*)dbeta, s1n, s2n, C, Z,
+                       1.f / (float)P);
+  }
+  if (rowwise_ok(C)) {
     dim3 ag(rw_grid(P, C)), ab(256);
 #define RWA(RELUv)                                                          \
     hipLaunchKernelGGL((bn_bwd_apply_rw_kernel<RELUv>), ag, ab, 0, stream,  \
@@ -768,20 +801,6 @@ void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, long ldy,
 #undef RWA
     return;
   }
-  dim3 sg(ceil_div(C, 64), Z), sb(256);
-  if (relu)
-    hipLaunchKernelGGL((bn_bwd_stats_kernel<true>), sg, sb, 0, stream,
-                       (const __bf16*)x, (const __bf16*)dy, ldy,
-                       (const __bf16*)g, (const __bf16*)b, mean, invstd,
-                       part, P, C, Z);
-  else
-    hipLaunchKernelGGL((bn_bwd_stats_kernel<false>), sg, sb, 0, stream,
-                       (const __bf16*)x, (const __bf16*)dy, ldy,
-                       (const __bf16*)g, (const __bf16*)b, mean, invstd,
-                       part, P, C, Z);
-  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(64), 0, stream,
-                     part, (__bf16*)dgamma, (__bf16*)dbeta, s1n, s2n, C, Z,
-                     1.f / (float)P);
   dim3 ag(ew_grid(P * (long)C)), ab(256);
   const bool vec = (C & 7) == 0;
   const FDiv dC = make_fd(C);
