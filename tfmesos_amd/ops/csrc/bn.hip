@@ -361,15 +361,26 @@ void bn_stats_rw_kernel(const __bf16* __restrict__ x,
       }
     }
   }
-  // cross-row reduce: rpb partial rows per channel octet
+  // cross-row tree reduce (log2(rpb) rounds, all rows participating —
+  // a serial per-thread loop here left tpr of 256 lanes active and
+  // dominated the big-layer kernels)
   __shared__ float red[256 * 8];
 #pragma unroll
-  for (int j = 0; j < 8; ++j) red[threadIdx.x * 8 + j] = sum[j];
+  for (int j = 0; j < 8; ++j) {
+    red[threadIdx.x * 8 + j] = sum[j];
+  }
   __syncthreads();
-  if (rl == 0) {
-    for (int rr = 1; rr < rpb; ++rr)
+  for (int cur = rpb; cur > 1;) {
+    const int half = cur >> 1;
+    if (rl < cur - half) {   // fold rows [half, cur) onto [0, cur-half)
 #pragma unroll
-      for (int j = 0; j < 8; ++j) red[u * 8 + j] += red[(rr * tpr + u) * 8 + j];
+      for (int j = 0; j < 8; ++j)
+        red[threadIdx.x * 8 + j] += red[((rl + half) * tpr + u) * 8 + j];
+    }
+    __syncthreads();
+    cur -= half;
+  }
+  if (rl == 0) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) sum[j] = red[u * 8 + j];
   }
@@ -377,10 +388,17 @@ void bn_stats_rw_kernel(const __bf16* __restrict__ x,
 #pragma unroll
   for (int j = 0; j < 8; ++j) red[threadIdx.x * 8 + j] = sq[j];
   __syncthreads();
-  if (rl == 0) {
-    for (int rr = 1; rr < rpb; ++rr)
+  for (int cur = rpb; cur > 1;) {
+    const int half = cur >> 1;
+    if (rl < cur - half) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) red[u * 8 + j] += red[(rr * tpr + u) * 8 + j];
+      for (int j = 0; j < 8; ++j)
+        red[threadIdx.x * 8 + j] += red[((rl + half) * tpr + u) * 8 + j];
+    }
+    __syncthreads();
+    cur -= half;
+  }
+  if (rl == 0) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       // channel-major partials: [c][z]{sum, sq}
@@ -390,18 +408,18 @@ void bn_stats_rw_kernel(const __bf16* __restrict__ x,
   }
 }
 
-// block 256 = 4 waves, one channel per wave; lanes stride z (coalesced
-// in the [c][z] layout)
+// block 256 = one channel; lanes stride z (coalesced in the [c][z]
+// layout — Z can be 2048, so a whole block per channel keeps enough
+// loads in flight)
 __global__ __launch_bounds__(256)
 void bn_finalize_rw_kernel(const float* __restrict__ part,
                            float* __restrict__ mean,
                            float* __restrict__ invstd,
                            int C, int Z, float inv_count, float eps) {
-  const int c = blockIdx.x * 4 + (threadIdx.x >> 6);
-  const int l = threadIdx.x & 63;
-  if (c >= C) return;
+  const int c = blockIdx.x;
+  const int t = threadIdx.x;
   float s = 0.f, q = 0.f;
-  for (int z = l; z < Z; z += 64) {
+  for (int z = t; z < Z; z += 256) {
     s += part[((long)c * Z + z) * 2];
     q += part[((long)c * Z + z) * 2 + 1];
   }
@@ -410,7 +428,15 @@ void bn_finalize_rw_kernel(const float* __restrict__ part,
     s += __shfl_xor(s, off, 64);
     q += __shfl_xor(q, off, 64);
   }
-  if (l == 0) {
+  __shared__ float ls[4], lq[4];
+  if ((t & 63) == 0) {
+    ls[t >> 6] = s;
+    lq[t >> 6] = q;
+  }
+  __syncthreads();
+  if (t == 0) {
+    s = ls[0] + ls[1] + ls[2] + ls[3];
+    q = lq[0] + lq[1] + lq[2] + lq[3];
     const float m = s * inv_count;
     float var = q * inv_count - m * m;
     if (var < 0.f) var = 0.f;
@@ -535,10 +561,17 @@ void bn_bwd_stats_rw_kernel(const __bf16* __restrict__ x,
 #pragma unroll
   for (int j = 0; j < 8; ++j) red[threadIdx.x * 8 + j] = s1[j];
   __syncthreads();
-  if (rl == 0) {
-    for (int rr = 1; rr < rpb; ++rr)
+  for (int cur = rpb; cur > 1;) {
+    const int half = cur >> 1;
+    if (rl < cur - half) {   // tree reduce (see bn_stats_rw_kernel)
 #pragma unroll
-      for (int j = 0; j < 8; ++j) red[u * 8 + j] += red[(rr * tpr + u) * 8 + j];
+      for (int j = 0; j < 8; ++j)
+        red[threadIdx.x * 8 + j] += red[((rl + half) * tpr + u) * 8 + j];
+    }
+    __syncthreads();
+    cur -= half;
+  }
+  if (rl == 0) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) s1[j] = red[u * 8 + j];
   }
@@ -546,10 +579,17 @@ void bn_bwd_stats_rw_kernel(const __bf16* __restrict__ x,
 #pragma unroll
   for (int j = 0; j < 8; ++j) red[threadIdx.x * 8 + j] = s2[j];
   __syncthreads();
-  if (rl == 0) {
-    for (int rr = 1; rr < rpb; ++rr)
+  for (int cur = rpb; cur > 1;) {
+    const int half = cur >> 1;
+    if (rl < cur - half) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) red[u * 8 + j] += red[(rr * tpr + u) * 8 + j];
+      for (int j = 0; j < 8; ++j)
+        red[threadIdx.x * 8 + j] += red[((rl + half) * tpr + u) * 8 + j];
+    }
+    __syncthreads();
+    cur -= half;
+  }
+  if (rl == 0) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       part[((long)(c0 + j) * Z + z) * 2] = s1[j];
@@ -565,11 +605,10 @@ void bn_bwd_finalize_rw_kernel(const float* __restrict__ part,
                                float* __restrict__ s1n,
                                float* __restrict__ s2n,
                                int C, int Z, float inv_count) {
-  const int c = blockIdx.x * 4 + (threadIdx.x >> 6);
-  const int l = threadIdx.x & 63;
-  if (c >= C) return;
+  const int c = blockIdx.x;
+  const int t = threadIdx.x;
   float s1 = 0.f, s2 = 0.f;
-  for (int z = l; z < Z; z += 64) {
+  for (int z = t; z < Z; z += 256) {
     s1 += part[((long)c * Z + z) * 2];
     s2 += part[((long)c * Z + z) * 2 + 1];
   }
@@ -578,7 +617,15 @@ void bn_bwd_finalize_rw_kernel(const float* __restrict__ part,
     s1 += __shfl_xor(s1, off, 64);
     s2 += __shfl_xor(s2, off, 64);
   }
-  if (l == 0) {
+  __shared__ float l1[4], l2[4];
+  if ((t & 63) == 0) {
+    l1[t >> 6] = s1;
+    l2[t >> 6] = s2;
+  }
+  __syncthreads();
+  if (t == 0) {
+    s1 = l1[0] + l1[1] + l1[2] + l1[3];
+    s2 = l2[0] + l2[1] + l2[2] + l2[3];
     dbeta[c] = (__bf16)s1;
     dgamma[c] = (__bf16)s2;
     s1n[c] = s1 * inv_count;
@@ -717,7 +764,7 @@ void launch_bn_fwd(const bf16_t* x, const bf16_t* g, const bf16_t* b,
   if (rowwise_stats(P, C)) {
     hipLaunchKernelGGL(bn_stats_rw_kernel, dim3(Z), dim3(256), 0, stream,
                        (const __bf16*)x, part, P, C, Z);
-    hipLaunchKernelGGL(bn_finalize_rw_kernel, dim3(ceil_div(C, 4)),
+    hipLaunchKernelGGL(bn_finalize_rw_kernel, dim3(C),
                        dim3(256), 0, stream, part, mean, invstd, C, Z,
                        1.f / (float)P, eps);
   } else {
@@ -771,7 +818,7 @@ void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, long ldy,
                        part, P, C, Z)
     if (relu) RWS(true); else RWS(false);
 #undef RWS
-    hipLaunchKernelGGL(bn_bwd_finalize_rw_kernel, dim3(ceil_div(C, 4)),
+    hipLaunchKernelGGL(bn_bwd_finalize_rw_kernel, dim3(C),
                        dim3(256), 0, stream, part, (__bf16*)dgamma,
                        (__bf16*)dbeta, s1n, s2n, C, Z, 1.f / (float)P);
   } else {
