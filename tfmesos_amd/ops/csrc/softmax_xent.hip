@@ -308,44 +308,51 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
       *(bf16x8*)&dlsT[i] = z8;
   }
   __syncthreads();
-  if (FROMWS) {
-    const int BH = B * H;
-    for (int i = t * 4; i < BH; i += 256 * 4) {
-      f32x4 v = {};
-      for (int z = 0; z < nslice; ++z) {
-        const f32x4 sv = *(const f32x4*)&wsrc[(long)z * BH + i];
+  // staging walks 4-element row chunks (H % 4 == 0, binding-checked):
+  // chunks never cross a row, so each is ONE aligned b64 global load +
+  // ONE b64 LDS store with a single division per chunk. The previous
+  // flat-index walk paid 2 integer divisions (~30 VALU each) plus an
+  // 8-scalar-store carry loop per chunk — the head's PMC showed ~6.7k
+  // issue slots per wave on a ~0.5 MFLOP kernel.
+  {
+    const int H4 = H >> 2;
+    const int nch = B * H4;
+    const int rstep = 256 / H4;          // uniform per-iteration walk:
+    const int cstep = 256 - rstep * H4;  // ONE division per thread total
+    int row = t / H4;
+    int c4i = t - row * H4;
+    for (int k = t; k < nch; k += 256) {
+      const int c4 = c4i * 4;
+      if (FROMWS) {
+        const int BH = B * H;
+        const int i = row * H + c4;
+        f32x4 v = {};
+        for (int z = 0; z < nslice; ++z) {
+          const f32x4 sv = *(const f32x4*)&wsrc[(long)z * BH + i];
 #pragma unroll
-        for (int j = 0; j < 4; ++j) v[j] += sv[j];
-      }
-      const int row = i / H;
-      const int col = i - row * H;
+          for (int j = 0; j < 4; ++j) v[j] += sv[j];
+        }
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const float f = v[j] + (float)bias1[col + j];
-        hs[row * HP + col + j] = (__bf16)(f > 0.f ? f : 0.f);
+        for (int j = 0; j < 4; ++j) {
+          const float f = v[j] + (float)bias1[c4 + j];
+          hs[row * HP + c4 + j] = (__bf16)(f > 0.f ? f : 0.f);
+        }
+      } else {
+        *(bf16x4*)&hs[row * HP + c4] =
+            *(const bf16x4*)&h[(long)row * H + c4];
       }
-    }
-  } else {
-    // b128 global loads over the flat [B*H] tensor; the 8-chunk can
-    // cross one row boundary (H >= 8), handled by the carry
-    const int BH = B * H;
-    for (int i = t * 8; i < BH; i += 256 * 8) {
-      bf16x8 v = {};
-      if (i + 8 <= BH) v = *(const bf16x8*)&h[i];
-      else for (int j = 0; i + j < BH; ++j) v[j] = *(const __bf16*)&h[i + j];
-      int row = i / H, col = i - (i / H) * H;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        hs[row * HP + col] = v[j];
-        if (++col == H) { col = 0; ++row; }
-      }
+      row += rstep;
+      c4i += cstep;
+      if (c4i >= H4) { c4i -= H4; ++row; }
     }
   }
-  for (int i = t; i < H * C; i += 256) {
-    const int hr = i / C, c = i - hr * C;
-    const __bf16 v = *(const __bf16*)&w[i];
-    wtp[c * HP + hr] = v;
-    wpad[hr * CP + c] = v;
+  // w rows are C elements: one thread per h-row, no divisions
+  for (int hr = t; hr < H; hr += 256) {
+    for (int c = 0; c < C; ++c) {
+      const __bf16 v = *(const __bf16*)&w[hr * C + c];
+      wtp[c * HP + hr] = v;
+      wpad[hr * CP + c] = v;
+    }
   }
   __syncthreads();
 
@@ -458,9 +465,20 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
         }
       }
     }
+    // db2[c] = sum_b dls[b][c]: 16 partial lanes per class (the serial
+    // B-loop was ~100 dependent LDS reads on 10 threads)
+    __shared__ float db2p[CP * 16];
+    if (t < C * 16) {
+      const int c = t >> 4, part = t & 15;
+      float s = 0.f;
+      for (int b2 = part; b2 < B; b2 += 16) s += (float)dlsT[c * HP + b2];
+      db2p[t] = s;
+    }
+    __syncthreads();
     if (t < C) {
       float s = 0.f;
-      for (int b2 = 0; b2 < B; ++b2) s += (float)dlsT[t * HP + b2];
+#pragma unroll
+      for (int p2 = 0; p2 < 16; ++p2) s += db2p[t * 16 + p2];
       if (GF32) ((float*)db2v)[t] = s;
       else ((bf16_t*)db2v)[t] = f2bf(s);
     }
@@ -484,7 +502,7 @@ void launch_mlp_head_fused(const bf16_t* h, const bf16_t* w,
                            void* dw2, void* db2, bool grads_f32,
                            float scale, int B, int H, int C,
                            hipStream_t stream) {
-  if (B <= 128 && H <= 128 && C <= 16) {
+  if (B <= 128 && H <= 128 && C <= 16 && (H & 3) == 0) {
     dim3 hg(dw2 != nullptr ? 2 : 1);
     if (grads_f32)
       hipLaunchKernelGGL((mlp_head_mfma_kernel<true, false>), hg,
