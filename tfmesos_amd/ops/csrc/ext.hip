@@ -27,6 +27,7 @@ void launch_softmax_xent_bwd(const bf16_t*, const long*, bf16_t*, float,
 void launch_mlp_head_fused(const bf16_t*, const bf16_t*, const bf16_t*,
                            const long*, bf16_t*, bf16_t*, float*, void*,
                            void*, bool, float, int, int, int, hipStream_t);
+void set_head_debug(void*);
 void launch_mlp_fwd_head(const float*, int, const bf16_t*, const bf16_t*,
                          const bf16_t*, const long*, bf16_t*, bf16_t*,
                          float*, void*, void*, bool, float, int, int, int,
@@ -798,6 +799,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "out (bf16 or fp32), split-K for deep skinny shapes");
   m.def("softmax_xent_fused", &softmax_xent_fused);
   m.def("mlp_head_fused", &mlp_head_fused);
+  m.def("head_debug", [](torch::Tensor t) {
+    set_head_debug(t.numel() ? t.data_ptr() : nullptr);
+  });
   m.def("mlp_fwd_head_fused", &mlp_fwd_head_fused);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);

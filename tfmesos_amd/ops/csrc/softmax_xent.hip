@@ -249,6 +249,15 @@ void mlp_head_fused_kernel(const bf16_t* __restrict__ h,
 // kernel.
 typedef __attribute__((ext_vector_type(16))) float f32x16v;
 
+// phase-stamp debug hook (tools/headbench.py --phases): wall_clock64
+// per phase per workgroup, written when the pointer is armed
+__device__ unsigned long long* g_head_dbg = nullptr;
+
+DEVINL void head_stamp(int t, int phase) {
+  if (g_head_dbg != nullptr && t == 0)
+    g_head_dbg[blockIdx.x * 16 + phase] = wall_clock64();
+}
+
 constexpr int HB = 128;        // padded B/H tile
 constexpr int HP = 136;        // hs row stride (8-elem pad, 16B aligned)
 constexpr int CP = 16;         // padded class dim
@@ -291,6 +300,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wr = t >> 6;
+  head_stamp(t, 0);
 
   // zero the padded operand tiles (b128 stores — the scalar version
   // was ~28K LDS writes), then stage h and w (scalar; tiny tensors)
@@ -308,6 +318,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
       *(bf16x8*)&dlsT[i] = z8;
   }
   __syncthreads();
+  head_stamp(t, 1);
   // staging walks 4-element row chunks (H % 4 == 0, binding-checked):
   // chunks never cross a row, so each is ONE aligned b64 global load +
   // ONE b64 LDS store with a single division per chunk. The previous
@@ -355,6 +366,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
     }
   }
   __syncthreads();
+  head_stamp(t, 2);
 
   // logits[B,C] = hs @ w: wave wr owns rows 32wr..32wr+31
   {
@@ -378,6 +390,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
     }
   }
   __syncthreads();
+  head_stamp(t, 3);
 
   // rowwise softmax + dlogits (one thread per row, C <= 16 scalar).
   // Two-workgroup split: WG0 owns {loss, dlogits store, dh}, WG1 owns
@@ -409,6 +422,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   }
   lsum[t] = neglogp;
   __syncthreads();
+  head_stamp(t, 4);
 
   // dh[B,H] = dls @ w^T, relu-masked by h>0: wave wr owns rows
   // 32wr..+31, loops the four 32-wide H column tiles; K = C (one MFMA)
@@ -437,6 +451,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
       }
     }
   }
+  head_stamp(t, 5);
 
   // dW2[H,C] = h^T @ dls (K = B, one 32-row tile per wave) — both
   // operands are in LDS already, so the step's separate dW2 GEMM
@@ -483,6 +498,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
       else ((bf16_t*)db2v)[t] = f2bf(s);
     }
   }
+  head_stamp(t, 6);
 
   // mean loss
   if (!wg0) return;
@@ -492,6 +508,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
     __syncthreads();
   }
   if (t == 0) *loss_out = lsum[0] / B;
+  head_stamp(t, 7);
 }
 
 }  // namespace
@@ -539,4 +556,9 @@ void launch_mlp_fwd_head(const float* ws, int nslice, const bf16_t* b1,
                        dim3(256), 0, stream, nullptr, ws, nslice, b1, w,
                        bias, labels, dlogits, dh, loss, dw2, db2, scale,
                        B, H, C);
+}
+
+void set_head_debug(void* ptr) {
+  // arm/disarm the per-phase wall_clock64 stamps (tools/headbench.py)
+  (void)hipMemcpyToSymbol(HIP_SYMBOL(g_head_dbg), &ptr, sizeof(ptr));
 }

@@ -59,6 +59,30 @@ def main():
     tiny = torch.zeros(256, device=dev)
     print("tiny fill (floor ref)  %7.2f us" % t_us(lambda: tiny.fill_(0.0)))
 
+    # per-phase wall_clock64 stamps (100 MHz-class constant clock)
+    from tfmesos_amd.ops import _ext
+    dbg = torch.zeros(32, dtype=torch.int64, device=dev)
+    _ext().head_debug(dbg)
+    ops.mlp_head_fused(h, w2, b2, y, dw2=dw2, db2=db2)
+    torch.cuda.synchronize()
+    _ext().head_debug(torch.empty(0, device=dev))
+    d = dbg.cpu().view(2, 16)
+    names = ["entry", "zero", "stage", "logits", "softmax", "dh/dw2",
+             "db2", "loss"]
+    for wgi in range(2):
+        row = d[wgi]
+        base = int(row[0])
+        out = []
+        prev = base
+        for ph in range(1, 8):
+            v = int(row[ph])
+            if v == 0:
+                continue
+            out.append("%s +%d" % (names[ph], v - prev))
+            prev = v
+        print("WG%d phase ticks (wall_clock64): %s  total %d"
+              % (wgi, "  ".join(out), prev - base))
+
 
 if __name__ == "__main__":
     main()
