@@ -295,7 +295,15 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   // fragment reads aligned b128 instead of 8 latency-exposed scalar
   // gathers per MFMA step (32 rows: fragment lanes span a full tile)
   __shared__ __align__(16) __bf16 dlsT[32 * HP];
-  __shared__ float ls[HB * CP];                     // logits fp32
+  // transposed h copy [col][row]: b64 relu-mask reads in the dh phase
+  // and b128 A-fragments in dW2 (both were 64 serial scalar LDS
+  // gathers per thread — the dh phase alone measured 4.4 us of the
+  // 11 us kernel, see tools/headbench.py --phases)
+  __shared__ __align__(16) __bf16 hsT[HB * HP];
+  // fp32 logits staging: stride 17 (a 16-dword stride put every lane
+  // of a softmax row-read on 2 of 32 banks — 32-way conflicts)
+  constexpr int LSP = CP + 1;
+  __shared__ float ls[HB * LSP];
   __shared__ float lsum[256];
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -316,6 +324,8 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
       *(bf16x8*)&dls[i] = z8;
     for (int i = t * 8; i < 32 * HP; i += 256 * 8)
       *(bf16x8*)&dlsT[i] = z8;
+    for (int i = t * 8; i < HB * HP; i += 256 * 8)
+      *(bf16x8*)&hsT[i] = z8;
   }
   __syncthreads();
   head_stamp(t, 1);
@@ -346,11 +356,15 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
           const float f = v[j] + (float)bias1[c4 + j];
-          hs[row * HP + c4 + j] = (__bf16)(f > 0.f ? f : 0.f);
+          const __bf16 r = (__bf16)(f > 0.f ? f : 0.f);
+          hs[row * HP + c4 + j] = r;
+          hsT[(c4 + j) * HP + row] = r;
         }
       } else {
-        *(bf16x4*)&hs[row * HP + c4] =
-            *(const bf16x4*)&h[(long)row * H + c4];
+        const bf16x4 v4 = *(const bf16x4*)&h[(long)row * H + c4];
+        *(bf16x4*)&hs[row * HP + c4] = v4;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) hsT[(c4 + j) * HP + row] = v4[j];
       }
       row += rstep;
       c4i += cstep;
@@ -385,7 +399,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
       for (int v = 0; v < 16; ++v) {
         const int row = wr * 32 + ((v >> 2) << 3) + ((lane >> 5) << 2) +
                         (v & 3);
-        ls[row * CP + col] = acc[v] + bc;
+        ls[row * LSP + col] = acc[v] + bc;
       }
     }
   }
@@ -400,7 +414,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
   const bool wg1 = blockIdx.x + 1 == gridDim.x;
   float neglogp = 0.f;
   if (t < B) {
-    const float* lr = &ls[t * CP];
+    const float* lr = &ls[t * LSP];
     float mx = -3.4e38f;
     for (int c = 0; c < C; ++c) mx = fmaxf(mx, lr[c]);
     float sum = 0.f;
@@ -438,13 +452,21 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
       acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, bv, acc, 0, 0, 0);
       const int colh = ct * 32 + (lane & 31);
       if (colh < H) {
+        // relu mask: 4 b64 reads of the transposed copy (one per
+        // 4-consecutive-row group of the D fragment) instead of 16
+        // serial scalar gathers across hs rows
+        const int rbase = wr * 32 + ((lane >> 5) << 2);
+        bf16x4 mk[4];
+#pragma unroll
+        for (int g = 0; g < 4; ++g)
+          mk[g] = *(const bf16x4*)&hsT[colh * HP + rbase + g * 8];
 #pragma unroll
         for (int v = 0; v < 16; ++v) {
           const int row = wr * 32 + ((v >> 2) << 3) + ((lane >> 5) << 2) +
                           (v & 3);
           if (row < B) {
             const float m =
-                (float)hs[row * HP + colh] > 0.f ? acc[v] : 0.f;
+                (float)mk[v >> 2][v & 3] > 0.f ? acc[v] : 0.f;
             dh[(long)row * H + colh] = f2bf(m);
           }
         }
@@ -461,10 +483,7 @@ void mlp_head_mfma_kernel(const bf16_t* __restrict__ h,
 #pragma unroll
     for (int kh = 0; kh < HB / 16; ++kh) {
       const int k0 = kh * 16 + ((lane >> 5) << 3);
-      bf16x8 a;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        a[j] = hs[(k0 + j) * HP + (wr * 32 + (lane & 31))];
+      bf16x8 a = *(const bf16x8*)&hsT[(wr * 32 + (lane & 31)) * HP + k0];
       bf16x8 bv = *(const bf16x8*)&dlsT[(lane & 31) * HP + k0];
       acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, bv, acc, 0, 0, 0);
     }
