@@ -412,11 +412,18 @@ std::vector<torch::Tensor> mlp_fwd_head_fused(
     dw2p = dw2.data_ptr();
     db2p = db2.data_ptr();
   }
-  // same split-K policy as gemm_bias_act_out
+  // split-K sliced LOW (default 4, TFA_HEAD_NSLICE to tune): the head
+  // workgroups re-read EVERY stripe (nslice x B*H fp32), so the
+  // generic 9-slice policy made the fused path read 360 KB per WG and
+  // lose to the separate reduce kernel (docs/KERNELS.md round-1 dead
+  // end); 4 fatter slices keep the gemm parallel enough while the
+  // head reads 160 KB
   const int nx = (H + 63) / 64, ny = (B + 63) / 64;
   int nslice = 1, kc = 0;
   if (nx * ny < 64 && K1 >= 256) {
-    int want = std::min(K1 / 64, 256 / (nx * ny));
+    int want = 4;
+    if (const char* e = getenv("TFA_HEAD_NSLICE")) want = atoi(e);
+    if (want > K1 / 64) want = K1 / 64;
     if (want > 16) want = 16;
     if (want > 1) {
       kc = ((K1 + want - 1) / want + 31) / 32 * 32;

@@ -52,9 +52,12 @@ def main():
           t_us(lambda: ops.mlp_head_fused(h, w2, b2, y)))
     print("head (+dw2, 2 WG)      %7.2f us" %
           t_us(lambda: ops.mlp_head_fused(h, w2, b2, y, dw2=dw2, db2=db2)))
-    print("fused fwd+head         %7.2f us" %
-          t_us(lambda: ops.mlp_fwd_head_fused(x, w1, b1, w2, b2, y,
-                                              dw2=dw2, db2=db2)))
+    for ns in (2, 3, 4, 6, 9):
+        os.environ["TFA_HEAD_NSLICE"] = str(ns)
+        print("fused fwd+head ns=%d    %7.2f us" %
+              (ns, t_us(lambda: ops.mlp_fwd_head_fused(x, w1, b1, w2, b2, y,
+                                                       dw2=dw2, db2=db2))))
+    os.environ.pop("TFA_HEAD_NSLICE")
     # launch floor reference: trivial elementwise kernel on a tiny tensor
     tiny = torch.zeros(256, device=dev)
     print("tiny fill (floor ref)  %7.2f us" % t_us(lambda: tiny.fill_(0.0)))
