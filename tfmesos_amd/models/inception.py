@@ -76,10 +76,15 @@ def _avg_pool(x, k, stride=1, padding=1):
                                           padding=padding)
 
 
-def _max_pool(x, k, stride):
+def _max_pool(x, k, stride, out=None):
     if k == 3 and stride == 2 and x.is_cuda:
-        return ops.max_pool3x3s2(x)   # hand-written (csrc/pool.hip)
-    return torch.nn.functional.max_pool2d(x, k, stride=stride)
+        # hand-written (csrc/pool.hip); out = concat-slice view (B/D)
+        return ops.max_pool3x3s2(x, out=out)
+    y = torch.nn.functional.max_pool2d(x, k, stride=stride)
+    if out is not None:
+        out.copy_(y)
+        return out
+    return y
 
 
 class _JoinViews(torch.autograd.Function):
@@ -104,13 +109,16 @@ class _JoinViews(torch.autograd.Function):
         return (None, *grads)
 
 
-def _fused_cat(x, specs):
+def _fused_cat(x, specs, hw=None):
     """specs: [(channels, builder)] — builder(out_view_or_None) returns
-    the branch output. GPU: each branch's terminal BN writes into its
-    slice of one buffer; CPU: plain torch.cat."""
+    the branch output. GPU: each branch's terminal BN (or maxpool)
+    writes into its slice of one buffer; CPU: plain torch.cat.
+    hw: output spatial dims when they differ from x's (the stride-2
+    reduction blocks B/D)."""
     if not x.is_cuda:
         return torch.cat([b(None) for _, b in specs], 1)
-    N, _, H, W = x.shape
+    N = x.shape[0]
+    H, W = hw if hw is not None else (x.shape[2], x.shape[3])
     ctot = sum(c for c, _ in specs)
     buf = torch.empty((N, ctot, H, W), device=x.device, dtype=x.dtype,
                       memory_format=torch.channels_last)
@@ -152,11 +160,13 @@ class InceptionB(nn.Module):
         self.b3x3dbl_3 = BasicConv2d(96, 96, kernel_size=3, stride=2)
 
     def forward(self, x):
-        return torch.cat([
-            self.b3x3(x),
-            self.b3x3dbl_3(self.b3x3dbl_2(self.b3x3dbl_1(x))),
-            _max_pool(x, 3, 2),
-        ], 1)
+        cin, h, w = x.shape[1], x.shape[2], x.shape[3]
+        return _fused_cat(x, [
+            (384, lambda o: self.b3x3(x, out=o)),
+            (96, lambda o: self.b3x3dbl_3(
+                self.b3x3dbl_2(self.b3x3dbl_1(x)), out=o)),
+            (cin, lambda o: _max_pool(x, 3, 2, out=o)),
+        ], hw=((h - 3) // 2 + 1, (w - 3) // 2 + 1))
 
 
 class InceptionC(nn.Module):
@@ -194,11 +204,13 @@ class InceptionD(nn.Module):
         self.b7_4 = BasicConv2d(192, 192, kernel_size=3, stride=2)
 
     def forward(self, x):
-        return torch.cat([
-            self.b3_2(self.b3_1(x)),
-            self.b7_4(self.b7_3(self.b7_2(self.b7_1(x)))),
-            _max_pool(x, 3, 2),
-        ], 1)
+        cin, h, w = x.shape[1], x.shape[2], x.shape[3]
+        return _fused_cat(x, [
+            (320, lambda o: self.b3_2(self.b3_1(x), out=o)),
+            (192, lambda o: self.b7_4(
+                self.b7_3(self.b7_2(self.b7_1(x))), out=o)),
+            (cin, lambda o: _max_pool(x, 3, 2, out=o)),
+        ], hw=((h - 3) // 2 + 1, (w - 3) // 2 + 1))
 
 
 class InceptionE(nn.Module):

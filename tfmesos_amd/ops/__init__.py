@@ -644,10 +644,12 @@ class _MaxPool3x3s2Fn(torch.autograd.Function):
     saves the winning tap; backward gathers deterministically."""
 
     @staticmethod
-    def forward(ctx, x):
+    def forward(ctx, x, out=None):
         if x.is_cuda:
             x = x.contiguous(memory_format=torch.channels_last)
-            y, idx = _ext().maxpool3x3s2_fwd(x)
+            e = out if out is not None \
+                else torch.empty(0, device=x.device, dtype=x.dtype)
+            y, idx = _ext().maxpool3x3s2_fwd(x, e)
             ctx.save_for_backward(idx)
             ctx.hw = (x.shape[2], x.shape[3])
             ctx.gpu = True
@@ -657,7 +659,11 @@ class _MaxPool3x3s2Fn(torch.autograd.Function):
             x.float(), 3, stride=2, return_indices=True)
         ctx.save_for_backward(idx)
         ctx.xshape = x.shape
-        return y.to(x.dtype)
+        y = y.to(x.dtype)
+        if out is not None:
+            out.copy_(y)
+            return out
+        return y
 
     @staticmethod
     def backward(ctx, dy):
@@ -665,12 +671,14 @@ class _MaxPool3x3s2Fn(torch.autograd.Function):
         if ctx.gpu:
             if not _is_cl_narrow(dy):
                 dy = dy.contiguous(memory_format=torch.channels_last)
-            return _ext().maxpool3x3s2_bwd(dy, idx, *ctx.hw)
+            return _ext().maxpool3x3s2_bwd(dy, idx, *ctx.hw), None
         return torch.nn.functional.max_unpool2d(
             dy.float(), idx, 3, stride=2,
-            output_size=ctx.xshape[2:]).to(dy.dtype)
+            output_size=ctx.xshape[2:]).to(dy.dtype), None
 
 
-def max_pool3x3s2(x):
-    """Differentiable 3x3/s2 max pool (no padding)."""
-    return _MaxPool3x3s2Fn.apply(x)
+def max_pool3x3s2(x, out=None):
+    """Differentiable 3x3/s2 max pool (no padding). out (GPU): a
+    channel-narrow channels-last view — the kernel stores the block's
+    concat slice directly (Inception B/D reduction blocks)."""
+    return _MaxPool3x3s2Fn.apply(x, out)

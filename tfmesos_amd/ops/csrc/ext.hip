@@ -68,7 +68,7 @@ void launch_bn_bwd(const bf16_t*, const bf16_t*, long, const bf16_t*,
 int bn_stats_slices(long, int);
 int bn_max_channels();
 void launch_avg3x3(const bf16_t*, bf16_t*, int, int, int, int, hipStream_t);
-void launch_maxpool3x3s2_fwd(const bf16_t*, bf16_t*, unsigned char*, int,
+void launch_maxpool3x3s2_fwd(const bf16_t*, bf16_t*, long, unsigned char*, int,
                              int, int, int, int, int, hipStream_t);
 void launch_maxpool3x3s2_bwd(const bf16_t*, long, const unsigned char*,
                              bf16_t*, int, int, int, int, int, int,
@@ -748,21 +748,35 @@ torch::Tensor avg_pool3x3(torch::Tensor x) {
   return y;
 }
 
-std::vector<torch::Tensor> maxpool3x3s2_fwd(torch::Tensor x) {
+std::vector<torch::Tensor> maxpool3x3s2_fwd(torch::Tensor x,
+                                            torch::Tensor out) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 && is_cl(x) &&
               x.scalar_type() == torch::kBFloat16,
               "maxpool3x3s2: bf16 channels-last NCHW");
   const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   const int Ho = (H - 3) / 2 + 1, Wo = (W - 3) / 2 + 1;
-  auto y = torch::empty({N, C, Ho, Wo},
-                        x.options().memory_format(
-                            torch::MemoryFormat::ChannelsLast));
+  long ldo = C;
+  torch::Tensor y;
+  if (out.numel() > 0) {
+    // channel-narrow view of the caller's concat buffer: the kernel
+    // stores strided and the aten cat copy disappears (like bn_fwd)
+    TORCH_CHECK(cl_narrow(out, &ldo) && out.size(0) == N &&
+                out.size(1) == C && out.size(2) == Ho && out.size(3) == Wo &&
+                out.scalar_type() == torch::kBFloat16,
+                "maxpool3x3s2_fwd: out must be a bf16 channels-last (or "
+                "channel-narrow) [N,C,Ho,Wo] tensor");
+    y = out;
+  } else {
+    y = torch::empty({N, C, Ho, Wo},
+                     x.options().memory_format(
+                         torch::MemoryFormat::ChannelsLast));
+  }
   auto idx = torch::empty({N, C, Ho, Wo},
                           x.options().dtype(torch::kUInt8).memory_format(
                               torch::MemoryFormat::ChannelsLast));
   launch_maxpool3x3s2_fwd((const bf16_t*)x.data_ptr(), (bf16_t*)y.data_ptr(),
-                          idx.data_ptr<unsigned char>(), N, H, W, C, Ho, Wo,
-                          cur_stream());
+                          ldo, idx.data_ptr<unsigned char>(), N, H, W, C, Ho,
+                          Wo, cur_stream());
   return {y, idx};
 }
 

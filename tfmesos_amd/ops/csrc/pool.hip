@@ -101,17 +101,21 @@ namespace {
 // bounds), channels-last. Forward saves the winning tap (0..8, u8);
 // backward GATHERS: each input pixel sums the dy of the <=2x2 windows
 // whose saved tap points at it — deterministic, no atomics.
+// ldo: output leading dim per pixel — a channel-narrow view of a wider
+// channels-last tensor writes the Inception B/D reduction-block concat
+// slice directly (no aten cat copy), mirroring the BN apply's strided
+// store (csrc/bn.hip).
 template <bool VEC>
 __global__ __launch_bounds__(256)
 void maxpool3x3s2_fwd_kernel(const __bf16* __restrict__ X,
-                             __bf16* __restrict__ Y,
+                             __bf16* __restrict__ Y, long ldo,
                              unsigned char* __restrict__ idx,
                              int N, int H, int W, int C, int Ho, int Wo) {
   const int nho = blockIdx.x;
   const int n = nho / Ho;
   const int ho = nho - n * Ho;
   const __bf16* base = X + ((long)n * H + ho * 2) * W * C;
-  __bf16* out = Y + ((long)n * Ho + ho) * Wo * C;
+  __bf16* out = Y + ((long)n * Ho + ho) * Wo * ldo;
   unsigned char* oi = idx + ((long)n * Ho + ho) * Wo * C;
   const int WoC = Wo * C;
   const int step = VEC ? 8 : 1;
@@ -133,7 +137,7 @@ void maxpool3x3s2_fwd_kernel(const __bf16* __restrict__ X,
                 (float)base[(long)r * W * C + col0 + (long)s * C + j];
             if (v > best) { best = v; bi = r * 3 + s; }
           }
-        out[i + j] = (__bf16)best;
+        out[(long)wo * ldo + c + j] = (__bf16)best;
         oi[i + j] = (unsigned char)bi;
       }
     } else {
@@ -146,7 +150,7 @@ void maxpool3x3s2_fwd_kernel(const __bf16* __restrict__ X,
           const float v = (float)base[(long)r * W * C + col0 + (long)s * C];
           if (v > best) { best = v; bi = r * 3 + s; }
         }
-      out[i] = (__bf16)best;
+      out[(long)wo * ldo + c] = (__bf16)best;
       oi[i] = (unsigned char)bi;
     }
   }
@@ -221,7 +225,8 @@ void maxpool3x3s2_bwd_kernel(const __bf16* __restrict__ dY, long ldy,
 
 }  // namespace
 
-void launch_maxpool3x3s2_fwd(const bf16_t* X, bf16_t* Y, unsigned char* idx,
+void launch_maxpool3x3s2_fwd(const bf16_t* X, bf16_t* Y, long ldo,
+                             unsigned char* idx,
                              int N, int H, int W, int C, int Ho, int Wo,
                              hipStream_t stream) {
   const long mrows = (long)N * Ho;
@@ -232,12 +237,12 @@ void launch_maxpool3x3s2_fwd(const bf16_t* X, bf16_t* Y, unsigned char* idx,
   dim3 grid((unsigned)mrows, (unsigned)mych), block(256);
   if ((C & 7) == 0)
     hipLaunchKernelGGL((maxpool3x3s2_fwd_kernel<true>), grid, block, 0,
-                       stream, (const __bf16*)X, (__bf16*)Y, idx, N, H, W, C,
-                       Ho, Wo);
+                       stream, (const __bf16*)X, (__bf16*)Y, ldo, idx, N, H,
+                       W, C, Ho, Wo);
   else
     hipLaunchKernelGGL((maxpool3x3s2_fwd_kernel<false>), grid, block, 0,
-                       stream, (const __bf16*)X, (__bf16*)Y, idx, N, H, W, C,
-                       Ho, Wo);
+                       stream, (const __bf16*)X, (__bf16*)Y, ldo, idx, N, H,
+                       W, C, Ho, Wo);
 }
 
 void launch_maxpool3x3s2_bwd(const bf16_t* dY, long ldy,
