@@ -678,3 +678,49 @@ def test_mlp_head_fused_odd_shapes():
     assert (dh.float() - dh2.float()).abs().max() < 2e-3
     assert (dw2 - dw2r).abs().max() < 3e-3 * dw2r.abs().max() + 1e-4
     assert (db2 - db2r).abs().max() < 3e-3 * db2r.abs().max() + 1e-4
+
+
+@pytest.mark.gpu
+def test_bn_group_matches_per_branch():
+    """Grouped BN (one launch triple for all block branches) must match
+    per-branch batch_norm_act exactly in both directions."""
+    import torch
+
+    from tfmesos_amd import ops
+
+    torch.manual_seed(5)
+    dev = "cuda:0"
+    N, H, W = 4, 9, 9
+    Cs = [64, 96, 32]
+    ctot = sum(Cs)
+    xs = [torch.randn(N, c, H, W, device=dev, dtype=torch.bfloat16)
+          .contiguous(memory_format=torch.channels_last)
+          .requires_grad_(True) for c in Cs]
+    ws = [torch.rand(c, device=dev, dtype=torch.bfloat16)
+          .requires_grad_(True) for c in Cs]
+    bs = [torch.randn(c, device=dev, dtype=torch.bfloat16)
+          .requires_grad_(True) for c in Cs]
+
+    buf = torch.empty(N, ctot, H, W, device=dev, dtype=torch.bfloat16,
+                      memory_format=torch.channels_last)
+    y = ops.bn_group_apply(buf.narrow(1, 0, ctot), xs, ws, bs,
+                           eps=1e-3, relu=True)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    off = 0
+    for i, c in enumerate(Cs):
+        x2 = xs[i].detach().clone().requires_grad_(True)
+        w2 = ws[i].detach().clone().requires_grad_(True)
+        b2 = bs[i].detach().clone().requires_grad_(True)
+        y2 = ops.batch_norm_act(x2, w2, b2, eps=1e-3, relu=True)
+        y2.backward(dy.narrow(1, off, c).contiguous(
+            memory_format=torch.channels_last))
+        assert torch.equal(y.narrow(1, off, c), y2), "fwd branch %d" % i
+        assert torch.allclose(xs[i].grad.float(), x2.grad.float(),
+                              atol=1e-3), "dx branch %d" % i
+        assert torch.allclose(ws[i].grad.float(), w2.grad.float(),
+                              rtol=1e-2, atol=1e-2), "dgamma branch %d" % i
+        assert torch.allclose(bs[i].grad.float(), b2.grad.float(),
+                              rtol=1e-2, atol=1e-2), "dbeta branch %d" % i
+        off += c

@@ -129,6 +129,37 @@ def _fused_cat(x, specs, hw=None):
     return _JoinViews.apply(buf, *views)
 
 
+def _group_cat(x, specs, hw=None, pool=None):
+    """Grouped-BN block concat. specs: [(C, BasicConv2d, conv_thunk)] —
+    conv_thunk() returns the branch's PRE-BN conv output; ONE grouped
+    BN (ops.bn_group_apply) normalizes every branch into the concat
+    buffer (6 kernels per BLOCK instead of per branch — the per-kernel
+    execution floor dominates at these layer sizes). pool: optional
+    (C, builder) no-BN branch (the B/D maxpool) writing its own slice.
+    CPU: per-branch BN + torch.cat."""
+    if not x.is_cuda:
+        parts = [m.bn(thunk()) for _, m, thunk in specs]
+        if pool is not None:
+            parts.append(pool[1](None))
+        return torch.cat(parts, 1)
+    N = x.shape[0]
+    H, W = hw if hw is not None else (x.shape[2], x.shape[3])
+    cbn = sum(c for c, _, _ in specs)
+    ctot = cbn + (pool[0] if pool is not None else 0)
+    buf = torch.empty((N, ctot, H, W), device=x.device, dtype=x.dtype,
+                      memory_format=torch.channels_last)
+    xs = [thunk() for _, _, thunk in specs]
+    y = ops.bn_group_apply(
+        buf.narrow(1, 0, cbn), xs,
+        [m.bn.weight for _, m, _ in specs],
+        [m.bn.bias for _, m, _ in specs],
+        eps=specs[0][1].bn.eps, relu=True)
+    views = [y]
+    if pool is not None:
+        views.append(pool[1](buf.narrow(1, cbn, pool[0])))
+    return _JoinViews.apply(buf, *views)
+
+
 class InceptionA(nn.Module):
     def __init__(self, cin, pool_features):
         super().__init__()
@@ -142,12 +173,12 @@ class InceptionA(nn.Module):
         self.pf = pool_features
 
     def forward(self, x):
-        return _fused_cat(x, [
-            (64, lambda o: self.b1x1(x, out=o)),
-            (64, lambda o: self.b5x5_2(self.b5x5_1(x), out=o)),
-            (96, lambda o: self.b3x3_3(self.b3x3_2(self.b3x3_1(x)),
-                                       out=o)),
-            (self.pf, lambda o: self.bpool(_avg_pool(x, 3), out=o)),
+        return _group_cat(x, [
+            (64, self.b1x1, lambda: self.b1x1.conv(x)),
+            (64, self.b5x5_2, lambda: self.b5x5_2.conv(self.b5x5_1(x))),
+            (96, self.b3x3_3,
+             lambda: self.b3x3_3.conv(self.b3x3_2(self.b3x3_1(x)))),
+            (self.pf, self.bpool, lambda: self.bpool.conv(_avg_pool(x, 3))),
         ])
 
 
@@ -161,12 +192,12 @@ class InceptionB(nn.Module):
 
     def forward(self, x):
         cin, h, w = x.shape[1], x.shape[2], x.shape[3]
-        return _fused_cat(x, [
-            (384, lambda o: self.b3x3(x, out=o)),
-            (96, lambda o: self.b3x3dbl_3(
-                self.b3x3dbl_2(self.b3x3dbl_1(x)), out=o)),
-            (cin, lambda o: _max_pool(x, 3, 2, out=o)),
-        ], hw=((h - 3) // 2 + 1, (w - 3) // 2 + 1))
+        return _group_cat(x, [
+            (384, self.b3x3, lambda: self.b3x3.conv(x)),
+            (96, self.b3x3dbl_3, lambda: self.b3x3dbl_3.conv(
+                self.b3x3dbl_2(self.b3x3dbl_1(x)))),
+        ], hw=((h - 3) // 2 + 1, (w - 3) // 2 + 1),
+            pool=(cin, lambda o: _max_pool(x, 3, 2, out=o)))
 
 
 class InceptionC(nn.Module):
@@ -184,12 +215,13 @@ class InceptionC(nn.Module):
         self.bpool = BasicConv2d(cin, 192, kernel_size=1)
 
     def forward(self, x):
-        return _fused_cat(x, [
-            (192, lambda o: self.b1x1(x, out=o)),
-            (192, lambda o: self.b7_3(self.b7_2(self.b7_1(x)), out=o)),
-            (192, lambda o: self.b7d_5(self.b7d_4(self.b7d_3(
-                self.b7d_2(self.b7d_1(x)))), out=o)),
-            (192, lambda o: self.bpool(_avg_pool(x, 3), out=o)),
+        return _group_cat(x, [
+            (192, self.b1x1, lambda: self.b1x1.conv(x)),
+            (192, self.b7_3,
+             lambda: self.b7_3.conv(self.b7_2(self.b7_1(x)))),
+            (192, self.b7d_5, lambda: self.b7d_5.conv(self.b7d_4(
+                self.b7d_3(self.b7d_2(self.b7d_1(x)))))),
+            (192, self.bpool, lambda: self.bpool.conv(_avg_pool(x, 3))),
         ])
 
 
@@ -205,12 +237,12 @@ class InceptionD(nn.Module):
 
     def forward(self, x):
         cin, h, w = x.shape[1], x.shape[2], x.shape[3]
-        return _fused_cat(x, [
-            (320, lambda o: self.b3_2(self.b3_1(x), out=o)),
-            (192, lambda o: self.b7_4(
-                self.b7_3(self.b7_2(self.b7_1(x))), out=o)),
-            (cin, lambda o: _max_pool(x, 3, 2, out=o)),
-        ], hw=((h - 3) // 2 + 1, (w - 3) // 2 + 1))
+        return _group_cat(x, [
+            (320, self.b3_2, lambda: self.b3_2.conv(self.b3_1(x))),
+            (192, self.b7_4, lambda: self.b7_4.conv(
+                self.b7_3(self.b7_2(self.b7_1(x))))),
+        ], hw=((h - 3) // 2 + 1, (w - 3) // 2 + 1),
+            pool=(cin, lambda o: _max_pool(x, 3, 2, out=o)))
 
 
 class InceptionE(nn.Module):
@@ -230,14 +262,14 @@ class InceptionE(nn.Module):
         b3 = self.b3_1(x)
         b3d = self.b3d_2(self.b3d_1(x))
         # the nested cats flatten: sub-branch slices are adjacent, so
-        # one buffer serves the whole block
-        return _fused_cat(x, [
-            (320, lambda o: self.b1x1(x, out=o)),
-            (384, lambda o: self.b3_2a(b3, out=o)),
-            (384, lambda o: self.b3_2b(b3, out=o)),
-            (384, lambda o: self.b3d_3a(b3d, out=o)),
-            (384, lambda o: self.b3d_3b(b3d, out=o)),
-            (192, lambda o: self.bpool(_avg_pool(x, 3), out=o)),
+        # one buffer (and ONE grouped BN) serves the whole block
+        return _group_cat(x, [
+            (320, self.b1x1, lambda: self.b1x1.conv(x)),
+            (384, self.b3_2a, lambda: self.b3_2a.conv(b3)),
+            (384, self.b3_2b, lambda: self.b3_2b.conv(b3)),
+            (384, self.b3d_3a, lambda: self.b3d_3a.conv(b3d)),
+            (384, self.b3d_3b, lambda: self.b3d_3b.conv(b3d)),
+            (192, self.bpool, lambda: self.bpool.conv(_avg_pool(x, 3))),
         ])
 
 

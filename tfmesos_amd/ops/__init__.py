@@ -548,6 +548,60 @@ class _BatchNormActFn(torch.autograd.Function):
                 None, None)
 
 
+class _GroupBNFn(torch.autograd.Function):
+    """Grouped train-mode BN(+relu) over the terminal branches of an
+    Inception block: ONE stats/finalize/apply launch triple for ALL
+    branches (per-channel math — results identical to per-branch BNs;
+    at these layer sizes the per-kernel execution floor made launch
+    count the dominant BN cost). Writes into a channel-narrow view of
+    the block's concat buffer and returns it (same .detach() contract
+    as _BatchNormActFn's out= path; see _JoinViews)."""
+
+    @staticmethod
+    def forward(ctx, out_view, eps, relu, n, *args):
+        xs = [a.contiguous(memory_format=torch.channels_last)
+              for a in args[:n]]
+        gs = [g.to(torch.bfloat16) for g in args[n:2 * n]]
+        bs = [b.to(torch.bfloat16) for b in args[2 * n:3 * n]]
+        mean, invstd = _ext().bn_group_fwd(xs, gs, bs, out_view,
+                                           eps, relu)
+        ctx.save_for_backward(mean, invstd, *xs, *gs, *bs)
+        ctx.n = n
+        ctx.relu = relu
+        return out_view.detach()
+
+    @staticmethod
+    def backward(ctx, dy):
+        n = ctx.n
+        saved = ctx.saved_tensors
+        mean, invstd = saved[0], saved[1]
+        xs = list(saved[2:2 + n])
+        gs = list(saved[2 + n:2 + 2 * n])
+        bs = list(saved[2 + 2 * n:2 + 3 * n])
+        if not _is_cl_narrow(dy):
+            dy = dy.contiguous(memory_format=torch.channels_last)
+        outs = _ext().bn_group_bwd(xs, dy, gs, bs, mean, invstd, ctx.relu)
+        dxs = outs[:n]
+        dgamma, dbeta = outs[n], outs[n + 1]
+        dgs, dbs = [], []
+        off = 0
+        for g in gs:
+            c = g.numel()
+            dgs.append(dgamma[off:off + c])
+            dbs.append(dbeta[off:off + c])
+            off += c
+        return (None, None, None, None, *dxs, *dgs, *dbs)
+
+
+def bn_group_apply(out_view, xs, weights, biases, eps=1e-3, relu=True):
+    """Grouped BN over parallel branches (GPU only): normalizes each
+    xs[i] with (weights[i], biases[i]) and writes the results into
+    consecutive channel slices of ``out_view`` (a channels-last or
+    channel-narrow view whose C == sum of branch channels)."""
+    return _GroupBNFn.apply(out_view, eps, relu, len(xs),
+                            *xs, *weights, *biases)
+
+
 def batch_norm_act(x, weight, bias, eps=1e-3, relu=False, out=None):
     """Differentiable fused train-mode BN (+relu); channels-last on GPU.
 

@@ -702,6 +702,214 @@ void bn_bwd_apply_rw_kernel(const __bf16* __restrict__ x,
 
 inline bool rowwise_ok(int C) { return (C & 7) == 0 && C <= MAXC; }
 
+// ------------------------------------------------------------ grouped path
+//
+// One launch per phase for ALL terminal BNs of an Inception block
+// (A/C/E blocks end in 4-6 independent conv->BN branches writing one
+// concat buffer): per-branch x/dy/param pointers ride a kernel arg
+// struct and grid.y (or a channel-block table) selects the branch.
+// BN is per-channel, so the grouped result is IDENTICAL to separate
+// per-branch BNs — only the launch count changes (6 kernels per
+// BRANCH chain -> 6 per BLOCK; the per-kernel execution floor at
+// these layer sizes made launch count the dominant BN cost).
+constexpr int BNG_MAX = 8;
+
+struct BNGroupArgs {
+  const __bf16* x[BNG_MAX];    // per-branch pre-BN conv output [P, C_b]
+  const __bf16* dy[BNG_MAX];   // bwd: dy slice base (already + coff)
+  __bf16* dx[BNG_MAX];         // bwd: per-branch dx out [P, C_b]
+  const __bf16* g[BNG_MAX];
+  const __bf16* b[BNG_MAX];
+  int C[BNG_MAX];
+  int coff[BNG_MAX];           // channel offset within the group
+  int cb0[BNG_MAX];            // first 64-wide channel block index
+  FDiv dC[BNG_MAX];            // magic divide by C (flat-walk decode)
+  int n;
+};
+
+DEVINL int bng_branch(const BNGroupArgs& a, int cb) {
+  int br = 0;
+  for (int i = 1; i < BNG_MAX; ++i)
+    if (i < a.n && cb >= a.cb0[i]) br = i;
+  return br;
+}
+
+// grid (cbtot, Z); column-walk stats (the grouped layers are the
+// small-P blocks where this granularity wins — see rowwise_stats)
+__global__ __launch_bounds__(256)
+void bng_stats_kernel(BNGroupArgs a, float* __restrict__ part,
+                      long P, int Ctot, int Z) {
+  const int cl = threadIdx.x & 63;
+  const int pr = threadIdx.x >> 6;
+  const int br = bng_branch(a, blockIdx.x);
+  const __bf16* x = a.x[br];
+  const int C = a.C[br];
+  const int c = (blockIdx.x - a.cb0[br]) * 64 + cl;
+  const int z = blockIdx.y;
+  float sum = 0.f, sq = 0.f;
+  if (c < C) {
+    const long step = (long)Z * 4;
+    long p = (long)z * 4 + pr;
+    for (; p + 3 * step < P; p += 4 * step) {
+      const float v0 = (float)x[p * C + c];
+      const float v1 = (float)x[(p + step) * C + c];
+      const float v2 = (float)x[(p + 2 * step) * C + c];
+      const float v3 = (float)x[(p + 3 * step) * C + c];
+      sum += v0 + v1 + v2 + v3;
+      sq += v0 * v0 + v1 * v1 + v2 * v2 + v3 * v3;
+    }
+    for (; p < P; p += step) {
+      const float v = (float)x[p * C + c];
+      sum += v;
+      sq += v * v;
+    }
+  }
+  __shared__ float ls[4][64], lq[4][64];
+  ls[pr][cl] = sum;
+  lq[pr][cl] = sq;
+  __syncthreads();
+  if (pr == 0 && c < C) {
+    sum = ls[0][cl] + ls[1][cl] + ls[2][cl] + ls[3][cl];
+    sq = lq[0][cl] + lq[1][cl] + lq[2][cl] + lq[3][cl];
+    const int gc = a.coff[br] + c;
+    part[((long)z * Ctot + gc) * 2] = sum;
+    part[((long)z * Ctot + gc) * 2 + 1] = sq;
+  }
+}
+
+// grid (ew blocks, n): per-branch flat walk, strided store into the
+// concat slice (out + coff), per-channel constants staged in LDS
+template <bool RELU>
+__global__ __launch_bounds__(256)
+void bng_apply_kernel(BNGroupArgs a, const float* __restrict__ mean,
+                      const float* __restrict__ invstd,
+                      __bf16* __restrict__ out, long ldo, long P) {
+  const int br = blockIdx.y;
+  const __bf16* x = a.x[br];
+  const int C = a.C[br];
+  const int coff = a.coff[br];
+  const __bf16* g = a.g[br];
+  const __bf16* bb = a.b[br];
+  __bf16* y = out + coff;
+  __shared__ float sc[512], sh[512];
+  for (int c = threadIdx.x; c < C; c += 256) {
+    const float s = (float)g[c] * invstd[coff + c];
+    sc[c] = s;
+    sh[c] = (float)bb[c] - mean[coff + c] * s;
+  }
+  __syncthreads();
+  const long total = P * C;
+  const long stride = (long)gridDim.x * 256 * 8;
+  const FDiv dC = a.dC[br];
+  for (long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8; i < total;
+       i += stride) {
+    const long prow = fd((unsigned)i, dC);  // C%8==0: chunk stays in-row
+    const int c0 = (int)(i - prow * C);
+    bf16x8 v = *(const bf16x8*)&x[i];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = (float)v[j] * sc[c0 + j] + sh[c0 + j];
+      if (RELU) f = f > 0.f ? f : 0.f;
+      o[j] = (__bf16)f;
+    }
+    *(bf16x8*)&y[prow * ldo + c0] = o;
+  }
+}
+
+template <bool RELU>
+__global__ __launch_bounds__(256)
+void bng_bwd_stats_kernel(BNGroupArgs a, long ldy,
+                          const float* __restrict__ mean,
+                          const float* __restrict__ invstd,
+                          float* __restrict__ part, long P, int Ctot,
+                          int Z) {
+  const int cl = threadIdx.x & 63;
+  const int pr = threadIdx.x >> 6;
+  const int br = bng_branch(a, blockIdx.x);
+  const __bf16* x = a.x[br];
+  const __bf16* dy = a.dy[br];
+  const int C = a.C[br];
+  const int c = (blockIdx.x - a.cb0[br]) * 64 + cl;
+  const int gc = a.coff[br] + c;
+  const int z = blockIdx.y;
+  float s1 = 0.f, s2 = 0.f;
+  if (c < C) {
+    const float mu = mean[gc], is = invstd[gc];
+    const float sc = RELU ? (float)a.g[br][c] * is : 0.f;
+    const float sh = RELU ? (float)a.b[br][c] - mu * sc : 0.f;
+    const long step = (long)Z * 4;
+    long p = (long)z * 4 + pr;
+    for (; p < P; p += step) {
+      const float xv = (float)x[p * C + c];
+      float d = (float)dy[p * ldy + c];
+      const float xh = (xv - mu) * is;
+      if (RELU && (float)(__bf16)(xv * sc + sh) <= 0.f) d = 0.f;
+      s1 += d;
+      s2 += d * xh;
+    }
+  }
+  __shared__ float l1[4][64], l2[4][64];
+  l1[pr][cl] = s1;
+  l2[pr][cl] = s2;
+  __syncthreads();
+  if (pr == 0 && c < C) {
+    s1 = l1[0][cl] + l1[1][cl] + l1[2][cl] + l1[3][cl];
+    s2 = l2[0][cl] + l2[1][cl] + l2[2][cl] + l2[3][cl];
+    part[((long)z * Ctot + gc) * 2] = s1;
+    part[((long)z * Ctot + gc) * 2 + 1] = s2;
+  }
+}
+
+template <bool RELU>
+__global__ __launch_bounds__(256)
+void bng_bwd_apply_kernel(BNGroupArgs a, long ldy,
+                          const float* __restrict__ mean,
+                          const float* __restrict__ invstd,
+                          const float* __restrict__ s1n,
+                          const float* __restrict__ s2n, long P) {
+  const int br = blockIdx.y;
+  const __bf16* x = a.x[br];
+  const __bf16* dy = a.dy[br];
+  __bf16* dx = a.dx[br];
+  const int C = a.C[br];
+  const int coff = a.coff[br];
+  __shared__ float lgs[512], la[512], lbb[512], lmu[512], lis[512];
+  __shared__ float lsh[512];
+  for (int c = threadIdx.x; c < C; c += 256) {
+    const int gc = coff + c;
+    const float s = (float)a.g[br][c] * invstd[gc];
+    lgs[c] = s;
+    la[c] = s1n[gc];
+    lbb[c] = s2n[gc];
+    lmu[c] = mean[gc];
+    lis[c] = invstd[gc];
+    lsh[c] = (float)a.b[br][c] - mean[gc] * s;
+  }
+  __syncthreads();
+  const long total = P * C;
+  const long stride = (long)gridDim.x * 256 * 8;
+  const FDiv dC = a.dC[br];
+  for (long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8; i < total;
+       i += stride) {
+    const long prow = fd((unsigned)i, dC);
+    const int c0 = (int)(i - prow * C);
+    const bf16x8 xv = *(const bf16x8*)&x[i];
+    const bf16x8 dv = *(const bf16x8*)&dy[prow * ldy + c0];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c0 + j;
+      const float x1 = (float)xv[j];
+      float d = (float)dv[j];
+      const float xh = (x1 - lmu[c]) * lis[c];
+      if (RELU && (float)(__bf16)(x1 * lgs[c] + lsh[c]) <= 0.f) d = 0.f;
+      o[j] = (__bf16)(lgs[c] * (d - la[c] - xh * lbb[c]));
+    }
+    *(bf16x8*)&dx[i] = o;
+  }
+}
+
 inline unsigned rw_grid(long P, int C) {
   const int rpb = 256 / (C >> 3) < 1 ? 1 : 256 / (C >> 3);
   long wgs = (P + rpb - 1) / rpb;
@@ -863,3 +1071,107 @@ void launch_bn_bwd(const bf16_t* x, const bf16_t* dy, long ldy,
 
 int bn_stats_slices(long P, int C) { return stats_slices(P, C); }
 int bn_max_channels() { return MAXC; }
+
+// -------------------------------------------------------- grouped launchers
+
+int bn_group_slices(long P, const int* Cs, int n) {
+  long cbtot = 0;
+  for (int i = 0; i < n; ++i) cbtot += ceil_div(Cs[i], 64);
+  long want = (2048 + cbtot - 1) / cbtot;
+  long per = P / (4 * 8);
+  long z = want < per ? want : per;
+  if (z < 1) z = 1;
+  if (z > 2048) z = 2048;
+  return (int)z;
+}
+
+namespace {
+
+BNGroupArgs fill_group_args(const bf16_t* const* xs, const bf16_t* const* gs,
+                            const bf16_t* const* bs, const int* Cs, int n,
+                            const bf16_t* dy_base, bf16_t* const* dxs,
+                            int* cbtot_out, int* ctot_out) {
+  BNGroupArgs a = {};
+  a.n = n;
+  int coff = 0, cb = 0;
+  for (int i = 0; i < n; ++i) {
+    a.x[i] = (const __bf16*)xs[i];
+    a.g[i] = (const __bf16*)gs[i];
+    a.b[i] = (const __bf16*)bs[i];
+    a.dy[i] = dy_base ? (const __bf16*)dy_base + coff : nullptr;
+    a.dx[i] = dxs ? (__bf16*)dxs[i] : nullptr;
+    a.C[i] = Cs[i];
+    a.coff[i] = coff;
+    a.cb0[i] = cb;
+    a.dC[i] = make_fd(Cs[i]);
+    coff += Cs[i];
+    cb += ceil_div(Cs[i], 64);
+  }
+  *cbtot_out = cb;
+  *ctot_out = coff;
+  return a;
+}
+
+inline unsigned bng_ew_grid(long P, int cmax) {
+  long wgs = ((P * cmax) / 8 + 255) / 256;
+  if (wgs > 2048) wgs = 2048;
+  if (wgs < 1) wgs = 1;
+  return (unsigned)wgs;
+}
+
+}  // namespace
+
+void launch_bn_group_fwd(const bf16_t* const* xs, const bf16_t* const* gs,
+                         const bf16_t* const* bs, const int* Cs, int n,
+                         bf16_t* out, long ldo, float* mean, float* invstd,
+                         float* part, long P, int Z, float eps, bool relu,
+                         hipStream_t stream) {
+  int cbtot, ctot;
+  BNGroupArgs a = fill_group_args(xs, gs, bs, Cs, n, nullptr, nullptr,
+                                  &cbtot, &ctot);
+  hipLaunchKernelGGL(bng_stats_kernel, dim3(cbtot, Z), dim3(256), 0, stream,
+                     a, part, P, ctot, Z);
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(ctot), dim3(64), 0, stream,
+                     part, mean, invstd, ctot, Z, 1.f / (float)P, eps);
+  int cmax = 0;
+  for (int i = 0; i < n; ++i) cmax = Cs[i] > cmax ? Cs[i] : cmax;
+  dim3 ag(bng_ew_grid(P, cmax), n), ab(256);
+  if (relu)
+    hipLaunchKernelGGL((bng_apply_kernel<true>), ag, ab, 0, stream, a, mean,
+                       invstd, (__bf16*)out, ldo, P);
+  else
+    hipLaunchKernelGGL((bng_apply_kernel<false>), ag, ab, 0, stream, a, mean,
+                       invstd, (__bf16*)out, ldo, P);
+}
+
+void launch_bn_group_bwd(const bf16_t* const* xs, const bf16_t* dy_base,
+                         long ldy, const bf16_t* const* gs,
+                         const bf16_t* const* bs, bf16_t* const* dxs,
+                         const int* Cs, int n, const float* mean,
+                         const float* invstd, bf16_t* dgamma, bf16_t* dbeta,
+                         float* s1n, float* s2n, float* part, long P, int Z,
+                         bool relu, hipStream_t stream) {
+  int cbtot, ctot;
+  BNGroupArgs a = fill_group_args(xs, gs, bs, Cs, n, dy_base, dxs,
+                                  &cbtot, &ctot);
+  if (relu)
+    hipLaunchKernelGGL((bng_bwd_stats_kernel<true>), dim3(cbtot, Z),
+                       dim3(256), 0, stream, a, ldy, mean, invstd, part, P,
+                       ctot, Z);
+  else
+    hipLaunchKernelGGL((bng_bwd_stats_kernel<false>), dim3(cbtot, Z),
+                       dim3(256), 0, stream, a, ldy, mean, invstd, part, P,
+                       ctot, Z);
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(ctot), dim3(64), 0, stream,
+                     part, (__bf16*)dgamma, (__bf16*)dbeta, s1n, s2n, ctot,
+                     Z, 1.f / (float)P);
+  int cmax = 0;
+  for (int i = 0; i < n; ++i) cmax = Cs[i] > cmax ? Cs[i] : cmax;
+  dim3 ag(bng_ew_grid(P, cmax), n), ab(256);
+  if (relu)
+    hipLaunchKernelGGL((bng_bwd_apply_kernel<true>), ag, ab, 0, stream, a,
+                       ldy, mean, invstd, s1n, s2n, P);
+  else
+    hipLaunchKernelGGL((bng_bwd_apply_kernel<false>), ag, ab, 0, stream, a,
+                       ldy, mean, invstd, s1n, s2n, P);
+}

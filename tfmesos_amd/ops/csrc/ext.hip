@@ -67,6 +67,16 @@ void launch_bn_bwd(const bf16_t*, const bf16_t*, long, const bf16_t*,
                    int, bool, hipStream_t);
 int bn_stats_slices(long, int);
 int bn_max_channels();
+int bn_group_slices(long, const int*, int);
+void launch_bn_group_fwd(const bf16_t* const*, const bf16_t* const*,
+                         const bf16_t* const*, const int*, int, bf16_t*,
+                         long, float*, float*, float*, long, int, float,
+                         bool, hipStream_t);
+void launch_bn_group_bwd(const bf16_t* const*, const bf16_t*, long,
+                         const bf16_t* const*, const bf16_t* const*,
+                         bf16_t* const*, const int*, int, const float*,
+                         const float*, bf16_t*, bf16_t*, float*, float*,
+                         float*, long, int, bool, hipStream_t);
 void launch_avg3x3(const bf16_t*, bf16_t*, int, int, int, int, hipStream_t);
 void launch_maxpool3x3s2_fwd(const bf16_t*, bf16_t*, long, unsigned char*, int,
                              int, int, int, int, int, hipStream_t);
@@ -738,6 +748,106 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   return {dx, dgamma, dbeta};
 }
 
+
+// grouped BN: ONE stats/finalize/apply launch triple for all terminal
+// BNs of an Inception block (per-channel math — identical results to
+// separate per-branch BNs, ~6x fewer launches per block)
+std::vector<torch::Tensor> bn_group_fwd(std::vector<torch::Tensor> xs,
+                                        std::vector<torch::Tensor> gs,
+                                        std::vector<torch::Tensor> bs,
+                                        torch::Tensor out, double eps,
+                                        bool relu) {
+  const int n = (int)xs.size();
+  TORCH_CHECK(n >= 1 && n <= 8 && (int)gs.size() == n &&
+              (int)bs.size() == n, "bn_group: 1..8 branches");
+  const bf16_t* xp[8];
+  const bf16_t* gp[8];
+  const bf16_t* bp[8];
+  int Cs[8];
+  int ctot = 0;
+  const long P = xs[0].numel() / xs[0].size(1);
+  for (int i = 0; i < n; ++i) {
+    auto& x = xs[i];
+    TORCH_CHECK(x.is_cuda() && x.dim() == 4 && is_cl(x) &&
+                x.scalar_type() == torch::kBFloat16 &&
+                x.numel() / x.size(1) == P && (x.size(1) & 7) == 0 &&
+                x.size(1) <= 512,
+                "bn_group_fwd: branch inputs must be bf16 channels-last "
+                "with equal spatial size, C%8==0, C<=512");
+    Cs[i] = (int)x.size(1);
+    TORCH_CHECK(gs[i].numel() == Cs[i] && bs[i].numel() == Cs[i] &&
+                gs[i].scalar_type() == torch::kBFloat16 &&
+                bs[i].scalar_type() == torch::kBFloat16, "bad g/b");
+    xp[i] = (const bf16_t*)x.data_ptr();
+    gp[i] = (const bf16_t*)gs[i].data_ptr();
+    bp[i] = (const bf16_t*)bs[i].data_ptr();
+    ctot += Cs[i];
+  }
+  long ldo = ctot;
+  TORCH_CHECK(cl_narrow(out, &ldo) && out.size(1) == ctot &&
+              out.numel() / out.size(1) == P &&
+              out.scalar_type() == torch::kBFloat16,
+              "bn_group_fwd: out must be a bf16 channels-last (or "
+              "channel-narrow) [N,Ctot,H,W] view");
+  const int Z = bn_group_slices(P, Cs, n);
+  auto opts = xs[0].options().dtype(torch::kFloat32);
+  auto mean = torch::empty({ctot}, opts);
+  auto invstd = torch::empty({ctot}, opts);
+  auto part = torch::empty({(long)Z * ctot * 2}, opts);
+  launch_bn_group_fwd(xp, gp, bp, Cs, n, (bf16_t*)out.data_ptr(), ldo,
+                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                      part.data_ptr<float>(), P, Z, (float)eps, relu,
+                      cur_stream());
+  return {mean, invstd};
+}
+
+std::vector<torch::Tensor> bn_group_bwd(std::vector<torch::Tensor> xs,
+                                        torch::Tensor dy,
+                                        std::vector<torch::Tensor> gs,
+                                        std::vector<torch::Tensor> bs,
+                                        torch::Tensor mean,
+                                        torch::Tensor invstd, bool relu) {
+  const int n = (int)xs.size();
+  TORCH_CHECK(n >= 1 && n <= 8, "bn_group: 1..8 branches");
+  const bf16_t* xp[8];
+  const bf16_t* gp[8];
+  const bf16_t* bp[8];
+  bf16_t* dxp[8];
+  int Cs[8];
+  int ctot = 0;
+  const long P = xs[0].numel() / xs[0].size(1);
+  std::vector<torch::Tensor> outs;
+  for (int i = 0; i < n; ++i) {
+    Cs[i] = (int)xs[i].size(1);
+    xp[i] = (const bf16_t*)xs[i].data_ptr();
+    gp[i] = (const bf16_t*)gs[i].data_ptr();
+    bp[i] = (const bf16_t*)bs[i].data_ptr();
+    auto dx = torch::empty_like(xs[i]);
+    dxp[i] = (bf16_t*)dx.data_ptr();
+    outs.push_back(dx);
+    ctot += Cs[i];
+  }
+  long ldy = ctot;
+  TORCH_CHECK(dy.is_cuda() && cl_narrow(dy, &ldy) && dy.size(1) == ctot,
+              "bn_group_bwd: dy must be channels-last or channel-narrow");
+  const int Z = bn_group_slices(P, Cs, n);
+  auto opts = xs[0].options().dtype(torch::kFloat32);
+  auto dgamma = torch::empty({ctot}, xs[0].options());
+  auto dbeta = torch::empty({ctot}, xs[0].options());
+  auto s1n = torch::empty({ctot}, opts);
+  auto s2n = torch::empty({ctot}, opts);
+  auto part = torch::empty({(long)Z * ctot * 2}, opts);
+  launch_bn_group_bwd(xp, (const bf16_t*)dy.data_ptr(), ldy, gp, bp, dxp,
+                      Cs, n, mean.data_ptr<float>(),
+                      invstd.data_ptr<float>(), (bf16_t*)dgamma.data_ptr(),
+                      (bf16_t*)dbeta.data_ptr(), s1n.data_ptr<float>(),
+                      s2n.data_ptr<float>(), part.data_ptr<float>(), P, Z,
+                      relu, cur_stream());
+  outs.push_back(dgamma);
+  outs.push_back(dbeta);
+  return outs;
+}
+
 torch::Tensor avg_pool3x3(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 && is_cl(x) &&
               x.scalar_type() == torch::kBFloat16,
@@ -830,6 +940,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd", &bn_fwd, "fused train-mode batch-norm (+relu) fwd");
   m.def("avg_pool3x3", &avg_pool3x3, "3x3 s1 p1 avg pool, channels-last");
   m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd);
+  m.def("bn_group_fwd", &bn_group_fwd);
+  m.def("bn_group_bwd", &bn_group_bwd);
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
   m.def("bn_bwd", &bn_bwd, "fused batch-norm (+relu mask) bwd");
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
