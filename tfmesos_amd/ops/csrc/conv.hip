@@ -575,17 +575,18 @@ struct DyBwdwStage {
   // commit() (after the MFMAs) only pays the vmcnt drain + LDS write.
   // The fused form parked every wave on vmcnt ahead of its own MFMAs
   // (profiles/r01_inception_pmc.txt: SQ_WAIT 150x the MFMA count).
+  // The 8 strided loads use j*K offsets off ONE base (a sequential
+  // pointer bump made each load's ADDRESS depend on the previous —
+  // PMC showed 42 VALU per MFMA, much of it serial address math).
   DEVINL void load(const ConvShape cs, long Ptot) {
     v = {};
     if (kok) {
       const long left = Ptot - p;
       if (left >= 8) {
-        const __bf16* s = src;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) { v[j] = *s; s += cs.K; }
+        for (int j = 0; j < 8; ++j) v[j] = src[(long)j * cs.K];
       } else if (left > 0) {
-        const __bf16* s = src;
-        for (int j = 0; j < (int)left; ++j) { v[j] = *s; s += cs.K; }
+        for (int j = 0; j < (int)left; ++j) v[j] = src[(long)j * cs.K];
       }
     }
     p += BK;
