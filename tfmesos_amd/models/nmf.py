@@ -135,8 +135,9 @@ class SparseNMF(object):
         b = self.batch
         I = torch.randint(0, self.n, (b,), generator=self._step_gen)
         J = torch.randint(0, self.n, (b,), generator=self._step_gen)
-        WI = self.client.pull("W", I)           # [b, r] bf16
-        HJ = self.client.pull("H", J)           # [b, r] bf16
+        # both pulls in flight at once (W and H live on different PS
+        # ranks — sequential pulls paid two full round trips)
+        WI, HJ = self.client.pull_many([("W", I), ("H", J)])  # [b,r] bf16
         X = (self.w0[I] @ self.h0[J].t() * self.scale).to(self.device)
         if self.device.type == "cuda":
             E = (ops.gemm_bias_act(WI, HJ, trans_b=True).float() - X)
@@ -148,8 +149,8 @@ class SparseNMF(object):
             E = WIf @ HJf.t() - X
             dW = 2.0 * (E @ HJf)
             dH = 2.0 * (E.t() @ WIf)
-        self.client.push("W", I, dW / self.batch)
-        self.client.push("H", J, dH / self.batch)
+        self.client.push_many([("W", I, dW / self.batch),
+                               ("H", J, dH / self.batch)])
         loss = float((E * E).mean())
         self.losses.append(loss)
         return loss

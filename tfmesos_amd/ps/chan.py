@@ -74,6 +74,13 @@ class Chan(object):
         dist.recv(buf, src=self.peer, group=self.group)
         return buf.to(device)
 
+    def isend(self, t):
+        """Non-blocking send; returns (work, staged) — keep ``staged``
+        alive until the work completes (the gloo+cuda path sends a
+        host staging copy)."""
+        s = self._out(t)
+        return dist.isend(s, dst=self.peer, group=self.group), s
+
     def irecv_into(self, t):
         """Non-blocking receive; returns the dist.Work handle. The
         caller polls ``work.is_completed()`` / ``work.wait()``. Used by

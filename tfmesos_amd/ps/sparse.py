@@ -34,6 +34,7 @@ import threading
 import time
 
 import torch
+import torch.distributed as dist
 
 from tfmesos_amd import ops
 from tfmesos_amd.ps.chan import Chan, make_pair_chans  # noqa: F401
@@ -121,6 +122,60 @@ class SparseWorkerClient(object):
         mine = sorted(n for n, h in self.homes.items()
                       if h == self.homes[name])
         return mine.index(name)
+
+    def pull_many(self, reqs):
+        """Concurrent pulls from multiple PS ranks: issue every
+        request's sends and row-irecvs before waiting (the sequential
+        pull() pair paid two full round trips per step — round-1 weak
+        #1: the distributed sparse step was transport-latency-bound).
+        reqs: [(name, ids)] — at most one request per table. Returns
+        rows in request order."""
+        keep, works = [], []
+        recv = []
+        for name, ids in reqs:
+            c = self._chan(name)
+            ids = ids.to(self.device)
+            for t in (self._hdr(ids.numel(), self._tidx(name)), ids):
+                w, s = c.isend(t)
+                works.append(w)
+                keep.append(s)
+            rows = torch.empty(ids.numel(), self.dims[name],
+                               dtype=torch.bfloat16, device=self.device)
+            recv.append((c, rows))
+        rworks = []
+        for c, rows in recv:
+            if c.device_only:
+                rworks.append((dist.irecv(rows, src=c.peer, group=c.group),
+                               rows, rows))
+            else:
+                buf = rows if rows.device.type == "cpu" else \
+                    torch.empty(rows.shape, dtype=rows.dtype, device="cpu")
+                rworks.append((dist.irecv(buf, src=c.peer, group=c.group),
+                               buf, rows))
+        for w in works:
+            w.wait()
+        out = []
+        for w, buf, rows in rworks:
+            w.wait()
+            if buf is not rows:
+                rows.copy_(buf)
+            out.append(rows)
+        return out
+
+    def push_many(self, reqs):
+        """Concurrent pushes: [(name, ids, grads)] — all sends issued
+        non-blocking, then awaited once."""
+        keep, works = [], []
+        for name, ids, grads in reqs:
+            c = self._chan(name)
+            ids = ids.to(self.device)
+            for t in (self._hdr(-ids.numel(), self._tidx(name)), ids,
+                      grads.to(device=self.device, dtype=torch.bfloat16)):
+                w, s = c.isend(t)
+                works.append(w)
+                keep.append(s)
+        for w in works:
+            w.wait()
 
     def done_all(self):
         """One shutdown marker per PS RANK (its serving loop drops this
