@@ -103,10 +103,14 @@ class SparseNMF(object):
         homes = {"W": 0, "H": ps_ranks[-1]}
         groups = make_sparse_pair_groups(ps_ranks, worker_ranks)
 
-        # planted low-rank target factors (same seed everywhere)
+        # planted low-rank target factors (same seed everywhere);
+        # device-resident so the per-step synthetic minibatch target
+        # X = w0[I] @ h0[J]^T is computed where the worker computes —
+        # as a host matmul it CPU-bound the whole distributed step on
+        # boxes with few/oversubscribed cores
         g = torch.Generator().manual_seed(seed)
-        self.w0 = torch.rand(n, factor_rank, generator=g)
-        self.h0 = torch.rand(n, factor_rank, generator=g)
+        self.w0 = torch.rand(n, factor_rank, generator=g).to(self.device)
+        self.h0 = torch.rand(n, factor_rank, generator=g).to(self.device)
         self.scale = 1.0 / factor_rank
 
         if self.is_ps:
@@ -138,7 +142,8 @@ class SparseNMF(object):
         # both pulls in flight at once (W and H live on different PS
         # ranks — sequential pulls paid two full round trips)
         WI, HJ = self.client.pull_many([("W", I), ("H", J)])  # [b,r] bf16
-        X = (self.w0[I] @ self.h0[J].t() * self.scale).to(self.device)
+        Id, Jd = I.to(self.device), J.to(self.device)
+        X = self.w0[Id] @ self.h0[Jd].t() * self.scale
         if self.device.type == "cuda":
             E = (ops.gemm_bias_act(WI, HJ, trans_b=True).float() - X)
             Eb = E.to(torch.bfloat16)
