@@ -310,7 +310,10 @@ class _Conv2dFn(torch.autograd.Function):
         x, w = ctx.saved_tensors
         krsc = ctx.krsc
         if x.is_cuda:
-            dy = dy.contiguous(memory_format=torch.channels_last)
+            # channel-narrow concat-grad views are read strided by the
+            # conv kernels (ConvShape.LDY) — no contiguous() copy
+            if not _is_cl_narrow(dy):
+                dy = dy.contiguous(memory_format=torch.channels_last)
             R, S = (w.shape[1], w.shape[2]) if krsc \
                 else (w.shape[2], w.shape[3])
             dx = None
@@ -321,7 +324,9 @@ class _Conv2dFn(torch.autograd.Function):
                 if ctx.pw:
                     # pointwise: dX = dY @ W on the tuned GEMM kernel
                     # (faster than the conv bwd-data path on every
-                    # Inception 1x1 shape — tools/bench_1x1.py)
+                    # Inception 1x1 shape — tools/bench_1x1.py); the
+                    # GEMM path needs a dense 2-D view
+                    dy = dy.contiguous(memory_format=torch.channels_last)
                     N, K, Ho, Wo = dy.shape
                     dy2 = dy.permute(0, 2, 3, 1).reshape(N * Ho * Wo, K)
                     dx2 = gemm_bias_act(dy2, wk.reshape(K, -1))

@@ -75,6 +75,8 @@ struct ConvShape {
   int U, V;           // stride
   int P, Q;           // pad
   FastDiv dC, dS, dK, dWo, dW2, dHoWo, dHW;   // divisors for gather decode
+  int LDY;   // dY row stride (== K, or wider when dY is a channel-
+             // narrow view of a concat-grad buffer — no contiguous copy)
 };
 
 // Load an 8-element channel run X[pix_base + c0 .. c0+8) into S[row][..],
@@ -388,7 +390,7 @@ DEVINL void BwddPatchStage::load(const __bf16* __restrict__ dY,
     }
     if (ok) {
       const __bf16* src =
-          dY + (((long)n * cs.Ho + ho) * cs.Wo + wo) * cs.K + kc0;
+          dY + (((long)n * cs.Ho + ho) * cs.Wo + wo) * cs.LDY + kc0;
       vv = *(const bf16x8*)src;
     }
     return;
@@ -414,7 +416,7 @@ DEVINL void BwddPatchStage::load(const __bf16* __restrict__ dY,
         ok = ok && ho < cs.Ho && wo < cs.Wo;
       }
       if (ok)
-        v = (float)dY[(((long)n * cs.Ho + ho) * cs.Wo + wo) * cs.K + kc];
+        v = (float)dY[(((long)n * cs.Ho + ho) * cs.Wo + wo) * cs.LDY + kc];
     }
     vv[j] = (__bf16)v;
     if (++kc == cs.K) { kc = 0; if (++ss2 == cs.S) { ss2 = 0; ++rr2; } }
@@ -567,7 +569,7 @@ struct DyBwdwStage {
     pxc = (t >> 6) * 8;    // 8-pixel chunk
     kok = m0 + k < cs.K;
     p = p0 + pxc;
-    src = dY + p * cs.K + (m0 + k);
+    src = dY + p * cs.LDY + (m0 + k);
   }
 
   // split load/commit: load() issues the global reads into registers
@@ -790,7 +792,7 @@ void launch_conv_fwd(const bf16_t* X, const bf16_t* W, const float* bias,
                      int P, int Q, bool relu, hipStream_t stream) {
   ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q,
                make_fdiv(C), make_fdiv(S), make_fdiv(K), make_fdiv(Wo),
-               make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd)};
+               make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd), K};
   const long M = (long)N * Ho * Wo;
   const int KD = R * S * C;
   const int z = conv_fwd_slices(N, K, Ho, Wo, C, R, S);
@@ -822,14 +824,15 @@ void launch_conv_fwd(const bf16_t* X, const bf16_t* W, const float* bias,
 #undef FL
 }
 
-void launch_conv_bwd_data(const bf16_t* dY, const bf16_t* Wt, bf16_t* dX,
+void launch_conv_bwd_data(const bf16_t* dY, long ldy, const bf16_t* Wt,
+                          bf16_t* dX,
                           float* ws, int N, int C, int H, int Wd, int K,
                           int R, int S, int Ho, int Wo, int U, int V, int P,
                           int Q, hipStream_t stream) {
-  // Wt: host-permuted W^T, memory [C][R*S*K]
   ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q,
                make_fdiv(C), make_fdiv(S), make_fdiv(K), make_fdiv(Wo),
-               make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd)};
+               make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd),
+               (int)ldy};
   const long M = (long)N * H * Wd;
   const int KD = R * S * K;
   const int z = conv_bwdd_slices(N, H, Wd, C, K, R, S);
@@ -877,7 +880,8 @@ int conv_bwdw_slices(int N, int C, int K, int R, int S, int Ho, int Wo) {
   return (int)((Ptot + pc - 1) / pc);
 }
 
-void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
+void launch_conv_bwd_weight(const bf16_t* dY, long ldy, const bf16_t* X,
+                            float* dW,
                             int N, int C, int H, int Wd, int K,
                             int R, int S, int Ho, int Wo, int U, int V,
                             int P, int Q, hipStream_t stream) {
@@ -885,7 +889,8 @@ void launch_conv_bwd_weight(const bf16_t* dY, const bf16_t* X, float* dW,
   (void)ws;
   ConvShape cs{N, C, H, Wd, K, R, S, Ho, Wo, U, V, P, Q,
                make_fdiv(C), make_fdiv(S), make_fdiv(K), make_fdiv(Wo),
-               make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd)};
+               make_fdiv(Wd), make_fdiv(Ho * Wo), make_fdiv(H * Wd),
+               (int)ldy};
   const int KD = C * R * S;
   const long Ptot = (long)N * Ho * Wo;
   int z = conv_bwdw_slices(N, C, K, R, S, Ho, Wo);

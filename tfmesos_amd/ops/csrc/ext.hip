@@ -50,11 +50,12 @@ void launch_conv_fwd(const bf16_t*, const bf16_t*, const float*, bf16_t*,
                      float*, int, int, int, int, int, int, int, int, int,
                      int, int, int, int, bool, hipStream_t);
 int conv_fwd_slices(int, int, int, int, int, int, int);
-void launch_conv_bwd_data(const bf16_t*, const bf16_t*, bf16_t*, float*,
+void launch_conv_bwd_data(const bf16_t*, long, const bf16_t*, bf16_t*,
+                          float*,
                           int, int, int, int, int, int, int, int, int, int,
                           int, int, int, hipStream_t);
 int conv_bwdd_slices(int, int, int, int, int, int, int);
-void launch_conv_bwd_weight(const bf16_t*, const bf16_t*, float*,
+void launch_conv_bwd_weight(const bf16_t*, long, const bf16_t*, float*,
                             int, int, int, int, int, int, int, int, int,
                             int, int, int, int, hipStream_t);
 int conv_bwdw_slices(int, int, int, int, int, int, int);
@@ -597,12 +598,19 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor wm,
 
 // dy channels-last; wt: [C,R,S,K] contiguous (W^T copy). Returns dx
 // channels-last.
+static bool cl_narrow(const torch::Tensor& t, long* ldy);
+
+// dy may be a channel-narrow view of the block concat-grad buffer
+// (_JoinViews backward): the kernels read it strided (ConvShape.LDY),
+// so the terminal convs of every Inception block skip a per-layer
+// contiguous() copy of their dy slice.
 torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor wt,
                               long H, long W, long stride_h, long stride_w,
                               long pad_h, long pad_w) {
-  TORCH_CHECK(dy.is_cuda() && dy.dim() == 4 && is_cl(dy) &&
+  long ldy = 0;
+  TORCH_CHECK(dy.is_cuda() && cl_narrow(dy, &ldy) &&
               dy.scalar_type() == torch::kBFloat16,
-              "dy must be bf16 channels-last");
+              "dy must be bf16 channels-last (or channel-narrow)");
   // weight in its NATIVE [K,R,S,C] layout — the kernel's transposing
   // stager reads it directly (no host-side W^T permute per call)
   TORCH_CHECK(wt.is_contiguous() && wt.dim() == 4 &&
@@ -618,7 +626,7 @@ torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor wt,
     ws = splitk_ws(dy.device(), (long)dy.size(0) * H * W * C * z, 1,
                    &cnt_unused);
   }
-  launch_conv_bwd_data((const bf16_t*)dy.data_ptr(),
+  launch_conv_bwd_data((const bf16_t*)dy.data_ptr(), ldy,
                        (const bf16_t*)wt.data_ptr(), (bf16_t*)dx.data_ptr(),
                        ws, dy.size(0), C, H, W, dy.size(1), R, S,
                        dy.size(2), dy.size(3), stride_h, stride_w,
@@ -630,15 +638,16 @@ torch::Tensor conv2d_bwd_data(torch::Tensor dy, torch::Tensor wt,
 torch::Tensor conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
                                 long R, long S, long stride_h, long stride_w,
                                 long pad_h, long pad_w) {
-  TORCH_CHECK(dy.is_cuda() && x.is_cuda() && is_cl(dy) && is_cl(x),
-              "dy/x must be channels-last");
+  long ldy = 0;
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda() && cl_narrow(dy, &ldy) && is_cl(x),
+              "dy must be channels-last (or channel-narrow), x channels-last");
   // zero-init only when the reduction is z-sliced (atomic accumulate)
   auto opts = x.options().dtype(torch::kFloat32);
   const int z = conv_bwdw_slices(x.size(0), x.size(1), dy.size(1), R, S,
                                  dy.size(2), dy.size(3));
   auto dw = z > 1 ? torch::zeros({dy.size(1), R, S, x.size(1)}, opts)
                   : torch::empty({dy.size(1), R, S, x.size(1)}, opts);
-  launch_conv_bwd_weight((const bf16_t*)dy.data_ptr(),
+  launch_conv_bwd_weight((const bf16_t*)dy.data_ptr(), ldy,
                          (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(),
                          x.size(0), x.size(1), x.size(2), x.size(3),
                          dy.size(1), R, S, dy.size(2), dy.size(3), stride_h,
@@ -653,14 +662,15 @@ torch::Tensor conv2d_bwd_weight(torch::Tensor dy, torch::Tensor x,
 void conv2d_bwd_weight_out(torch::Tensor dy, torch::Tensor x,
                            long R, long S, long stride_h, long stride_w,
                            long pad_h, long pad_w, torch::Tensor dw) {
-  TORCH_CHECK(dy.is_cuda() && x.is_cuda() && is_cl(dy) && is_cl(x),
-              "dy/x must be channels-last");
+  long ldy = 0;
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda() && cl_narrow(dy, &ldy) && is_cl(x),
+              "dy must be channels-last (or channel-narrow), x channels-last");
   TORCH_CHECK(dw.is_cuda() && dw.is_contiguous() &&
               dw.scalar_type() == torch::kFloat32 && dw.dim() == 4 &&
               dw.size(0) == dy.size(1) && dw.size(1) == R &&
               dw.size(2) == S && dw.size(3) == x.size(1),
               "dw must be contiguous fp32 [K,R,S,C]");
-  launch_conv_bwd_weight((const bf16_t*)dy.data_ptr(),
+  launch_conv_bwd_weight((const bf16_t*)dy.data_ptr(), ldy,
                          (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(),
                          x.size(0), x.size(1), x.size(2), x.size(3),
                          dy.size(1), R, S, dy.size(2), dy.size(3), stride_h,
