@@ -64,3 +64,39 @@ def test_tfa_run_replica_training_on_gpu(tmp_path):
         env=dict(os.environ, PYTHONPATH=repo), cwd=repo)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "validation xent" in r.stdout
+
+
+@pytest.mark.timeout(300)
+def test_sparse_nmf_multiproc_on_gpu():
+    """3-process sparse-PS NMF where every rank holds CUDA tensors but
+    the transport is gloo (ranks share one GPU on the test box): covers
+    the Chan host-staging path with REAL device tensors — isend staging
+    buffers must stay alive until the works complete."""
+    import os
+    import subprocess
+    import sys
+
+    from tfmesos_amd.utils import free_port
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    port = free_port()
+    procs = []
+    for rank in range(3):
+        env = dict(os.environ)
+        env.update({
+            "RANK": str(rank), "WORLD_SIZE": "3",
+            "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+            "PYTHONPATH": repo, "TFA_DIST_BACKEND": "gloo",
+            "OMP_NUM_THREADS": "4",
+        })
+        procs.append(subprocess.Popen(
+            [sys.executable, os.path.join(repo, "bench.py"),
+             "--workload", "nmf", "--steps", "30", "--warmup", "5",
+             "--nmf-n", "400", "--nmf-rank", "64", "--nmf-batch", "128"],
+            env=env, stdout=subprocess.PIPE, text=True))
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=240)
+        assert p.returncode == 0, out
+        outs.append(out)
+    assert any('"transport": "gloo"' in o for o in outs), outs
