@@ -724,3 +724,40 @@ def test_bn_group_matches_per_branch():
         assert torch.allclose(bs[i].grad.float(), b2.grad.float(),
                               rtol=1e-2, atol=1e-2), "dbeta branch %d" % i
         off += c
+
+
+@pytest.mark.gpu
+def test_bn_group_multi_matches_per_branch():
+    """Grouped BN with per-branch DENSE outputs (parallel inner-stage
+    branches) must match per-branch batch_norm_act exactly."""
+    import torch
+
+    from tfmesos_amd import ops
+
+    torch.manual_seed(6)
+    dev = "cuda:0"
+    N, H, W = 3, 7, 7
+    Cs = [48, 64]
+    xs = [torch.randn(N, c, H, W, device=dev, dtype=torch.bfloat16)
+          .contiguous(memory_format=torch.channels_last)
+          .requires_grad_(True) for c in Cs]
+    ws = [torch.rand(c, device=dev, dtype=torch.bfloat16)
+          .requires_grad_(True) for c in Cs]
+    bs = [torch.randn(c, device=dev, dtype=torch.bfloat16)
+          .requires_grad_(True) for c in Cs]
+    ys = ops.bn_group_multi(xs, ws, bs, eps=1e-3, relu=True)
+    dys = [torch.randn_like(y) for y in ys]
+    torch.autograd.backward(ys, dys)
+    for i, c in enumerate(Cs):
+        x2 = xs[i].detach().clone().requires_grad_(True)
+        w2 = ws[i].detach().clone().requires_grad_(True)
+        b2 = bs[i].detach().clone().requires_grad_(True)
+        y2 = ops.batch_norm_act(x2, w2, b2, eps=1e-3, relu=True)
+        y2.backward(dys[i])
+        assert torch.equal(ys[i], y2), "fwd branch %d" % i
+        assert torch.allclose(xs[i].grad.float(), x2.grad.float(),
+                              atol=1e-3), "dx branch %d" % i
+        assert torch.allclose(ws[i].grad.float(), w2.grad.float(),
+                              rtol=1e-2, atol=1e-2), i
+        assert torch.allclose(bs[i].grad.float(), b2.grad.float(),
+                              rtol=1e-2, atol=1e-2), i

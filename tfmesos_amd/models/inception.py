@@ -160,6 +160,16 @@ def _group_cat(x, specs, hw=None, pool=None):
     return _JoinViews.apply(buf, *views)
 
 
+def _bn_multi(mods, xs):
+    """Grouped BN for parallel inner-stage branches (one launch triple
+    for all of them; per-channel math identical to per-branch BN)."""
+    if not xs[0].is_cuda:
+        return [m.bn(x) for m, x in zip(mods, xs)]
+    return list(ops.bn_group_multi(
+        xs, [m.bn.weight for m in mods], [m.bn.bias for m in mods],
+        eps=mods[0].bn.eps, relu=True))
+
+
 class InceptionA(nn.Module):
     def __init__(self, cin, pool_features):
         super().__init__()
@@ -173,11 +183,16 @@ class InceptionA(nn.Module):
         self.pf = pool_features
 
     def forward(self, x):
+        if x.is_cuda:
+            y5, y3 = _bn_multi([self.b5x5_1, self.b3x3_1],
+                               [self.b5x5_1.conv(x), self.b3x3_1.conv(x)])
+        else:
+            y5, y3 = self.b5x5_1(x), self.b3x3_1(x)
         return _group_cat(x, [
             (64, self.b1x1, lambda: self.b1x1.conv(x)),
-            (64, self.b5x5_2, lambda: self.b5x5_2.conv(self.b5x5_1(x))),
+            (64, self.b5x5_2, lambda: self.b5x5_2.conv(y5)),
             (96, self.b3x3_3,
-             lambda: self.b3x3_3.conv(self.b3x3_2(self.b3x3_1(x)))),
+             lambda: self.b3x3_3.conv(self.b3x3_2(y3))),
             (self.pf, self.bpool, lambda: self.bpool.conv(_avg_pool(x, 3))),
         ])
 
@@ -215,12 +230,20 @@ class InceptionC(nn.Module):
         self.bpool = BasicConv2d(cin, 192, kernel_size=1)
 
     def forward(self, x):
+        if x.is_cuda:
+            s1a, s1b = _bn_multi([self.b7_1, self.b7d_1],
+                                 [self.b7_1.conv(x), self.b7d_1.conv(x)])
+            s2a, s2b = _bn_multi(
+                [self.b7_2, self.b7d_2],
+                [self.b7_2.conv(s1a), self.b7d_2.conv(s1b)])
+        else:
+            s2a = self.b7_2(self.b7_1(x))
+            s2b = self.b7d_2(self.b7d_1(x))
         return _group_cat(x, [
             (192, self.b1x1, lambda: self.b1x1.conv(x)),
-            (192, self.b7_3,
-             lambda: self.b7_3.conv(self.b7_2(self.b7_1(x)))),
+            (192, self.b7_3, lambda: self.b7_3.conv(s2a)),
             (192, self.b7d_5, lambda: self.b7d_5.conv(self.b7d_4(
-                self.b7d_3(self.b7d_2(self.b7d_1(x)))))),
+                self.b7d_3(s2b)))),
             (192, self.bpool, lambda: self.bpool.conv(_avg_pool(x, 3))),
         ])
 
@@ -237,10 +260,15 @@ class InceptionD(nn.Module):
 
     def forward(self, x):
         cin, h, w = x.shape[1], x.shape[2], x.shape[3]
+        if x.is_cuda:
+            y3, y7 = _bn_multi([self.b3_1, self.b7_1],
+                               [self.b3_1.conv(x), self.b7_1.conv(x)])
+        else:
+            y3, y7 = self.b3_1(x), self.b7_1(x)
         return _group_cat(x, [
-            (320, self.b3_2, lambda: self.b3_2.conv(self.b3_1(x))),
+            (320, self.b3_2, lambda: self.b3_2.conv(y3)),
             (192, self.b7_4, lambda: self.b7_4.conv(
-                self.b7_3(self.b7_2(self.b7_1(x))))),
+                self.b7_3(self.b7_2(y7)))),
         ], hw=((h - 3) // 2 + 1, (w - 3) // 2 + 1),
             pool=(cin, lambda o: _max_pool(x, 3, 2, out=o)))
 
@@ -259,8 +287,13 @@ class InceptionE(nn.Module):
         self.bpool = BasicConv2d(cin, 192, kernel_size=1)
 
     def forward(self, x):
-        b3 = self.b3_1(x)
-        b3d = self.b3d_2(self.b3d_1(x))
+        if x.is_cuda:
+            b3, y3d = _bn_multi([self.b3_1, self.b3d_1],
+                                [self.b3_1.conv(x), self.b3d_1.conv(x)])
+            b3d = self.b3d_2(y3d)
+        else:
+            b3 = self.b3_1(x)
+            b3d = self.b3d_2(self.b3d_1(x))
         # the nested cats flatten: sub-branch slices are adjacent, so
         # one buffer (and ONE grouped BN) serves the whole block
         return _group_cat(x, [

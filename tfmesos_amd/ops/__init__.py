@@ -598,6 +598,56 @@ class _GroupBNFn(torch.autograd.Function):
         return (None, None, None, None, *dxs, *dgs, *dbs)
 
 
+class _GroupBNMultiFn(torch.autograd.Function):
+    """Grouped BN over PARALLEL inner-stage branches with separate
+    dense outputs (e.g. the two 1x1 stems of an Inception block): one
+    stats/finalize/apply launch triple instead of one per branch."""
+
+    @staticmethod
+    def forward(ctx, eps, relu, n, *args):
+        xs = [a.contiguous(memory_format=torch.channels_last)
+              for a in args[:n]]
+        gs = [g.to(torch.bfloat16) for g in args[n:2 * n]]
+        bs = [b.to(torch.bfloat16) for b in args[2 * n:3 * n]]
+        outs = _ext().bn_group_fwd_multi(xs, gs, bs, eps, relu)
+        ys, mean, invstd = outs[:n], outs[n], outs[n + 1]
+        ctx.save_for_backward(mean, invstd, *xs, *gs, *bs)
+        ctx.n = n
+        ctx.relu = relu
+        return tuple(ys)
+
+    @staticmethod
+    def backward(ctx, *dys):
+        n = ctx.n
+        saved = ctx.saved_tensors
+        mean, invstd = saved[0], saved[1]
+        xs = list(saved[2:2 + n])
+        gs = list(saved[2 + n:2 + 2 * n])
+        bs = list(saved[2 + 2 * n:2 + 3 * n])
+        dys = [d if _is_cl_narrow(d)
+               else d.contiguous(memory_format=torch.channels_last)
+               for d in dys]
+        outs = _ext().bn_group_bwd_multi(xs, dys, gs, bs, mean, invstd,
+                                         ctx.relu)
+        dxs = outs[:n]
+        dgamma, dbeta = outs[n], outs[n + 1]
+        dgs, dbs = [], []
+        off = 0
+        for g in gs:
+            c = g.numel()
+            dgs.append(dgamma[off:off + c])
+            dbs.append(dbeta[off:off + c])
+            off += c
+        return (None, None, None, *dxs, *dgs, *dbs)
+
+
+def bn_group_multi(xs, weights, biases, eps=1e-3, relu=True):
+    """Grouped BN for parallel branches with separate outputs (GPU
+    only). Returns the per-branch normalized tensors."""
+    return _GroupBNMultiFn.apply(eps, relu, len(xs), *xs,
+                                 *weights, *biases)
+
+
 def bn_group_apply(out_view, xs, weights, biases, eps=1e-3, relu=True):
     """Grouped BN over parallel branches (GPU only): normalizes each
     xs[i] with (weights[i], biases[i]) and writes the results into

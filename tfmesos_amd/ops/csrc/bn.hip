@@ -716,8 +716,11 @@ constexpr int BNG_MAX = 8;
 
 struct BNGroupArgs {
   const __bf16* x[BNG_MAX];    // per-branch pre-BN conv output [P, C_b]
-  const __bf16* dy[BNG_MAX];   // bwd: dy slice base (already + coff)
+  const __bf16* dy[BNG_MAX];   // bwd: per-branch dy base
+  long dyld[BNG_MAX];          // bwd: per-branch dy row stride
   __bf16* dx[BNG_MAX];         // bwd: per-branch dx out [P, C_b]
+  __bf16* y[BNG_MAX];          // fwd: per-branch out base
+  long yld[BNG_MAX];           // fwd: per-branch out row stride
   const __bf16* g[BNG_MAX];
   const __bf16* b[BNG_MAX];
   int C[BNG_MAX];
@@ -782,15 +785,15 @@ void bng_stats_kernel(BNGroupArgs a, float* __restrict__ part,
 template <bool RELU>
 __global__ __launch_bounds__(256)
 void bng_apply_kernel(BNGroupArgs a, const float* __restrict__ mean,
-                      const float* __restrict__ invstd,
-                      __bf16* __restrict__ out, long ldo, long P) {
+                      const float* __restrict__ invstd, long P) {
   const int br = blockIdx.y;
   const __bf16* x = a.x[br];
   const int C = a.C[br];
   const int coff = a.coff[br];
   const __bf16* g = a.g[br];
   const __bf16* bb = a.b[br];
-  __bf16* y = out + coff;
+  __bf16* y = a.y[br];
+  const long ldo = a.yld[br];
   __shared__ float sc[512], sh[512];
   for (int c = threadIdx.x; c < C; c += 256) {
     const float s = (float)g[c] * invstd[coff + c];
@@ -819,7 +822,7 @@ void bng_apply_kernel(BNGroupArgs a, const float* __restrict__ mean,
 
 template <bool RELU>
 __global__ __launch_bounds__(256)
-void bng_bwd_stats_kernel(BNGroupArgs a, long ldy,
+void bng_bwd_stats_kernel(BNGroupArgs a,
                           const float* __restrict__ mean,
                           const float* __restrict__ invstd,
                           float* __restrict__ part, long P, int Ctot,
@@ -829,6 +832,7 @@ void bng_bwd_stats_kernel(BNGroupArgs a, long ldy,
   const int br = bng_branch(a, blockIdx.x);
   const __bf16* x = a.x[br];
   const __bf16* dy = a.dy[br];
+  const long ldy = a.dyld[br];
   const int C = a.C[br];
   const int c = (blockIdx.x - a.cb0[br]) * 64 + cl;
   const int gc = a.coff[br] + c;
@@ -863,7 +867,7 @@ void bng_bwd_stats_kernel(BNGroupArgs a, long ldy,
 
 template <bool RELU>
 __global__ __launch_bounds__(256)
-void bng_bwd_apply_kernel(BNGroupArgs a, long ldy,
+void bng_bwd_apply_kernel(BNGroupArgs a,
                           const float* __restrict__ mean,
                           const float* __restrict__ invstd,
                           const float* __restrict__ s1n,
@@ -871,6 +875,7 @@ void bng_bwd_apply_kernel(BNGroupArgs a, long ldy,
   const int br = blockIdx.y;
   const __bf16* x = a.x[br];
   const __bf16* dy = a.dy[br];
+  const long ldy = a.dyld[br];
   __bf16* dx = a.dx[br];
   const int C = a.C[br];
   const int coff = a.coff[br];
@@ -1089,7 +1094,9 @@ namespace {
 
 BNGroupArgs fill_group_args(const bf16_t* const* xs, const bf16_t* const* gs,
                             const bf16_t* const* bs, const int* Cs, int n,
-                            const bf16_t* dy_base, bf16_t* const* dxs,
+                            const bf16_t* const* dys, const long* dylds,
+                            bf16_t* const* dxs,
+                            bf16_t* const* ys, const long* ylds,
                             int* cbtot_out, int* ctot_out) {
   BNGroupArgs a = {};
   a.n = n;
@@ -1098,8 +1105,11 @@ BNGroupArgs fill_group_args(const bf16_t* const* xs, const bf16_t* const* gs,
     a.x[i] = (const __bf16*)xs[i];
     a.g[i] = (const __bf16*)gs[i];
     a.b[i] = (const __bf16*)bs[i];
-    a.dy[i] = dy_base ? (const __bf16*)dy_base + coff : nullptr;
+    a.dy[i] = dys ? (const __bf16*)dys[i] : nullptr;
+    a.dyld[i] = dylds ? dylds[i] : 0;
     a.dx[i] = dxs ? (__bf16*)dxs[i] : nullptr;
+    a.y[i] = ys ? (__bf16*)ys[i] : nullptr;
+    a.yld[i] = ylds ? ylds[i] : 0;
     a.C[i] = Cs[i];
     a.coff[i] = coff;
     a.cb0[i] = cb;
@@ -1123,12 +1133,13 @@ inline unsigned bng_ew_grid(long P, int cmax) {
 
 void launch_bn_group_fwd(const bf16_t* const* xs, const bf16_t* const* gs,
                          const bf16_t* const* bs, const int* Cs, int n,
-                         bf16_t* out, long ldo, float* mean, float* invstd,
+                         bf16_t* const* youts, const long* ylds,
+                         float* mean, float* invstd,
                          float* part, long P, int Z, float eps, bool relu,
                          hipStream_t stream) {
   int cbtot, ctot;
   BNGroupArgs a = fill_group_args(xs, gs, bs, Cs, n, nullptr, nullptr,
-                                  &cbtot, &ctot);
+                                  nullptr, youts, ylds, &cbtot, &ctot);
   hipLaunchKernelGGL(bng_stats_kernel, dim3(cbtot, Z), dim3(256), 0, stream,
                      a, part, P, ctot, Z);
   hipLaunchKernelGGL(bn_finalize_kernel, dim3(ctot), dim3(64), 0, stream,
@@ -1138,29 +1149,29 @@ void launch_bn_group_fwd(const bf16_t* const* xs, const bf16_t* const* gs,
   dim3 ag(bng_ew_grid(P, cmax), n), ab(256);
   if (relu)
     hipLaunchKernelGGL((bng_apply_kernel<true>), ag, ab, 0, stream, a, mean,
-                       invstd, (__bf16*)out, ldo, P);
+                       invstd, P);
   else
     hipLaunchKernelGGL((bng_apply_kernel<false>), ag, ab, 0, stream, a, mean,
-                       invstd, (__bf16*)out, ldo, P);
+                       invstd, P);
 }
 
-void launch_bn_group_bwd(const bf16_t* const* xs, const bf16_t* dy_base,
-                         long ldy, const bf16_t* const* gs,
+void launch_bn_group_bwd(const bf16_t* const* xs, const bf16_t* const* dys,
+                         const long* dylds, const bf16_t* const* gs,
                          const bf16_t* const* bs, bf16_t* const* dxs,
                          const int* Cs, int n, const float* mean,
                          const float* invstd, bf16_t* dgamma, bf16_t* dbeta,
                          float* s1n, float* s2n, float* part, long P, int Z,
                          bool relu, hipStream_t stream) {
   int cbtot, ctot;
-  BNGroupArgs a = fill_group_args(xs, gs, bs, Cs, n, dy_base, dxs,
-                                  &cbtot, &ctot);
+  BNGroupArgs a = fill_group_args(xs, gs, bs, Cs, n, dys, dylds, dxs,
+                                  nullptr, nullptr, &cbtot, &ctot);
   if (relu)
     hipLaunchKernelGGL((bng_bwd_stats_kernel<true>), dim3(cbtot, Z),
-                       dim3(256), 0, stream, a, ldy, mean, invstd, part, P,
+                       dim3(256), 0, stream, a, mean, invstd, part, P,
                        ctot, Z);
   else
     hipLaunchKernelGGL((bng_bwd_stats_kernel<false>), dim3(cbtot, Z),
-                       dim3(256), 0, stream, a, ldy, mean, invstd, part, P,
+                       dim3(256), 0, stream, a, mean, invstd, part, P,
                        ctot, Z);
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(ctot), dim3(64), 0, stream,
                      part, (__bf16*)dgamma, (__bf16*)dbeta, s1n, s2n, ctot,
@@ -1170,8 +1181,8 @@ void launch_bn_group_bwd(const bf16_t* const* xs, const bf16_t* dy_base,
   dim3 ag(bng_ew_grid(P, cmax), n), ab(256);
   if (relu)
     hipLaunchKernelGGL((bng_bwd_apply_kernel<true>), ag, ab, 0, stream, a,
-                       ldy, mean, invstd, s1n, s2n, P);
+                       mean, invstd, s1n, s2n, P);
   else
     hipLaunchKernelGGL((bng_bwd_apply_kernel<false>), ag, ab, 0, stream, a,
-                       ldy, mean, invstd, s1n, s2n, P);
+                       mean, invstd, s1n, s2n, P);
 }

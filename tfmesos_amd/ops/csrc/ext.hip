@@ -70,10 +70,11 @@ int bn_stats_slices(long, int);
 int bn_max_channels();
 int bn_group_slices(long, const int*, int);
 void launch_bn_group_fwd(const bf16_t* const*, const bf16_t* const*,
-                         const bf16_t* const*, const int*, int, bf16_t*,
-                         long, float*, float*, float*, long, int, float,
-                         bool, hipStream_t);
-void launch_bn_group_bwd(const bf16_t* const*, const bf16_t*, long,
+                         const bf16_t* const*, const int*, int,
+                         bf16_t* const*, const long*, float*, float*,
+                         float*, long, int, float, bool, hipStream_t);
+void launch_bn_group_bwd(const bf16_t* const*, const bf16_t* const*,
+                         const long*,
                          const bf16_t* const*, const bf16_t* const*,
                          bf16_t* const*, const int*, int, const float*,
                          const float*, bf16_t*, bf16_t*, float*, float*,
@@ -804,11 +805,66 @@ std::vector<torch::Tensor> bn_group_fwd(std::vector<torch::Tensor> xs,
   auto mean = torch::empty({ctot}, opts);
   auto invstd = torch::empty({ctot}, opts);
   auto part = torch::empty({(long)Z * ctot * 2}, opts);
-  launch_bn_group_fwd(xp, gp, bp, Cs, n, (bf16_t*)out.data_ptr(), ldo,
+  bf16_t* youts[8];
+  long ylds[8];
+  int coff = 0;
+  for (int i = 0; i < n; ++i) {
+    youts[i] = (bf16_t*)out.data_ptr() + coff;
+    ylds[i] = ldo;
+    coff += Cs[i];
+  }
+  launch_bn_group_fwd(xp, gp, bp, Cs, n, youts, ylds,
                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
                       part.data_ptr<float>(), P, Z, (float)eps, relu,
                       cur_stream());
   return {mean, invstd};
+}
+
+// per-branch DENSE outputs (parallel inner-stage BNs whose results
+// feed different consumers): same grouped kernels, one launch triple
+std::vector<torch::Tensor> bn_group_fwd_multi(std::vector<torch::Tensor> xs,
+                                              std::vector<torch::Tensor> gs,
+                                              std::vector<torch::Tensor> bs,
+                                              double eps, bool relu) {
+  const int n = (int)xs.size();
+  TORCH_CHECK(n >= 1 && n <= 8, "bn_group: 1..8 branches");
+  const bf16_t* xp[8];
+  const bf16_t* gp[8];
+  const bf16_t* bp[8];
+  bf16_t* youts[8];
+  long ylds[8];
+  int Cs[8];
+  int ctot = 0;
+  const long P = xs[0].numel() / xs[0].size(1);
+  std::vector<torch::Tensor> outs;
+  for (int i = 0; i < n; ++i) {
+    auto& x = xs[i];
+    TORCH_CHECK(x.is_cuda() && x.dim() == 4 && is_cl(x) &&
+                x.scalar_type() == torch::kBFloat16 &&
+                x.numel() / x.size(1) == P && (x.size(1) & 7) == 0 &&
+                x.size(1) <= 512, "bn_group_fwd_multi: bad branch input");
+    Cs[i] = (int)x.size(1);
+    xp[i] = (const bf16_t*)x.data_ptr();
+    gp[i] = (const bf16_t*)gs[i].data_ptr();
+    bp[i] = (const bf16_t*)bs[i].data_ptr();
+    auto y = torch::empty_like(x);
+    youts[i] = (bf16_t*)y.data_ptr();
+    ylds[i] = Cs[i];
+    outs.push_back(y);
+    ctot += Cs[i];
+  }
+  const int Z = bn_group_slices(P, Cs, n);
+  auto opts = xs[0].options().dtype(torch::kFloat32);
+  auto mean = torch::empty({ctot}, opts);
+  auto invstd = torch::empty({ctot}, opts);
+  auto part = torch::empty({(long)Z * ctot * 2}, opts);
+  launch_bn_group_fwd(xp, gp, bp, Cs, n, youts, ylds,
+                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                      part.data_ptr<float>(), P, Z, (float)eps, relu,
+                      cur_stream());
+  outs.push_back(mean);
+  outs.push_back(invstd);
+  return outs;
 }
 
 std::vector<torch::Tensor> bn_group_bwd(std::vector<torch::Tensor> xs,
@@ -840,6 +896,14 @@ std::vector<torch::Tensor> bn_group_bwd(std::vector<torch::Tensor> xs,
   long ldy = ctot;
   TORCH_CHECK(dy.is_cuda() && cl_narrow(dy, &ldy) && dy.size(1) == ctot,
               "bn_group_bwd: dy must be channels-last or channel-narrow");
+  const bf16_t* dys[8];
+  long dylds[8];
+  int coff = 0;
+  for (int i = 0; i < n; ++i) {
+    dys[i] = (const bf16_t*)dy.data_ptr() + coff;
+    dylds[i] = ldy;
+    coff += Cs[i];
+  }
   const int Z = bn_group_slices(P, Cs, n);
   auto opts = xs[0].options().dtype(torch::kFloat32);
   auto dgamma = torch::empty({ctot}, xs[0].options());
@@ -847,7 +911,60 @@ std::vector<torch::Tensor> bn_group_bwd(std::vector<torch::Tensor> xs,
   auto s1n = torch::empty({ctot}, opts);
   auto s2n = torch::empty({ctot}, opts);
   auto part = torch::empty({(long)Z * ctot * 2}, opts);
-  launch_bn_group_bwd(xp, (const bf16_t*)dy.data_ptr(), ldy, gp, bp, dxp,
+  launch_bn_group_bwd(xp, dys, dylds, gp, bp, dxp,
+                      Cs, n, mean.data_ptr<float>(),
+                      invstd.data_ptr<float>(), (bf16_t*)dgamma.data_ptr(),
+                      (bf16_t*)dbeta.data_ptr(), s1n.data_ptr<float>(),
+                      s2n.data_ptr<float>(), part.data_ptr<float>(), P, Z,
+                      relu, cur_stream());
+  outs.push_back(dgamma);
+  outs.push_back(dbeta);
+  return outs;
+}
+
+std::vector<torch::Tensor> bn_group_bwd_multi(std::vector<torch::Tensor> xs,
+                                              std::vector<torch::Tensor> dys_in,
+                                              std::vector<torch::Tensor> gs,
+                                              std::vector<torch::Tensor> bs,
+                                              torch::Tensor mean,
+                                              torch::Tensor invstd,
+                                              bool relu) {
+  const int n = (int)xs.size();
+  TORCH_CHECK(n >= 1 && n <= 8 && (int)dys_in.size() == n,
+              "bn_group: 1..8 branches");
+  const bf16_t* xp[8];
+  const bf16_t* gp[8];
+  const bf16_t* bp[8];
+  bf16_t* dxp[8];
+  const bf16_t* dys[8];
+  long dylds[8];
+  int Cs[8];
+  int ctot = 0;
+  const long P = xs[0].numel() / xs[0].size(1);
+  std::vector<torch::Tensor> outs;
+  for (int i = 0; i < n; ++i) {
+    Cs[i] = (int)xs[i].size(1);
+    xp[i] = (const bf16_t*)xs[i].data_ptr();
+    gp[i] = (const bf16_t*)gs[i].data_ptr();
+    bp[i] = (const bf16_t*)bs[i].data_ptr();
+    long ldy = Cs[i];
+    TORCH_CHECK(cl_narrow(dys_in[i], &ldy) && dys_in[i].size(1) == Cs[i],
+                "bn_group_bwd_multi: bad dy");
+    dys[i] = (const bf16_t*)dys_in[i].data_ptr();
+    dylds[i] = ldy;
+    auto dx = torch::empty_like(xs[i]);
+    dxp[i] = (bf16_t*)dx.data_ptr();
+    outs.push_back(dx);
+    ctot += Cs[i];
+  }
+  const int Z = bn_group_slices(P, Cs, n);
+  auto opts = xs[0].options().dtype(torch::kFloat32);
+  auto dgamma = torch::empty({ctot}, xs[0].options());
+  auto dbeta = torch::empty({ctot}, xs[0].options());
+  auto s1n = torch::empty({ctot}, opts);
+  auto s2n = torch::empty({ctot}, opts);
+  auto part = torch::empty({(long)Z * ctot * 2}, opts);
+  launch_bn_group_bwd(xp, dys, dylds, gp, bp, dxp,
                       Cs, n, mean.data_ptr<float>(),
                       invstd.data_ptr<float>(), (bf16_t*)dgamma.data_ptr(),
                       (bf16_t*)dbeta.data_ptr(), s1n.data_ptr<float>(),
@@ -952,6 +1069,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd);
   m.def("bn_group_fwd", &bn_group_fwd);
   m.def("bn_group_bwd", &bn_group_bwd);
+  m.def("bn_group_fwd_multi", &bn_group_fwd_multi);
+  m.def("bn_group_bwd_multi", &bn_group_bwd_multi);
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
   m.def("bn_bwd", &bn_bwd, "fused batch-norm (+relu mask) bwd");
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
