@@ -312,8 +312,14 @@ class InceptionV3(nn.Module):
     def __init__(self, num_classes=1000, seed=0):
         super().__init__()
         torch.manual_seed(seed)
+        # conv1 takes the input ZERO-PADDED from 3 to 8 channels (see
+        # forward): C=3 forced the conv stager's scalar-gather fallback
+        # and the single conv1 dispatch measured 155 us (~8 TF/s); at
+        # C=8 the b128 channel-run path engages (~5x). The 5 pad input
+        # channels of the weight see identically-zero activations, so
+        # their gradients are zero and the math is unchanged.
         self.stem = nn.ModuleList([
-            BasicConv2d(3, 32, kernel_size=3, stride=2),
+            BasicConv2d(8, 32, kernel_size=3, stride=2),
             BasicConv2d(32, 32, kernel_size=3),
             BasicConv2d(32, 64, kernel_size=3, padding=1),
         ])
@@ -376,6 +382,8 @@ class InceptionV3(nn.Module):
     def forward(self, x):
         if x.is_cuda:
             self._ensure_arena(x.device)
+        if x.shape[1] == 3:   # zero-pad to the conv1 weight's 8 channels
+            x = torch.nn.functional.pad(x, (0, 0, 0, 0, 0, 5))
         for m in self.stem:
             x = m(x)
         x = _max_pool(x, 3, 2)
