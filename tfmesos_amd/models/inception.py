@@ -35,6 +35,13 @@ class Conv2d(nn.Module):
         self._dw_cb = None     # comm-overlap notify (ps/module_trainer.py)
 
     def forward(self, x):
+        cin_w = self.weight.shape[3]
+        if x.shape[1] < cin_w:
+            # channel-padded weights (the 3-channel stem input): pad x
+            # with zero channels — identical math, and C=8 engages the
+            # stager's b128 channel-run path (C=3 forced scalar gathers)
+            x = torch.nn.functional.pad(
+                x, (0, 0, 0, 0, 0, cin_w - x.shape[1]))
         return ops.conv2d(x, self.weight, None, stride=self.stride,
                           padding=self.padding, weight_format="krsc",
                           dw_out=self._dw_buf if x.is_cuda else None,
@@ -382,8 +389,6 @@ class InceptionV3(nn.Module):
     def forward(self, x):
         if x.is_cuda:
             self._ensure_arena(x.device)
-        if x.shape[1] == 3:   # zero-pad to the conv1 weight's 8 channels
-            x = torch.nn.functional.pad(x, (0, 0, 0, 0, 0, 5))
         for m in self.stem:
             x = m(x)
         x = _max_pool(x, 3, 2)
