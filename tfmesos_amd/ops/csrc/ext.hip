@@ -35,6 +35,9 @@ void launch_mlp_fwd_head(const float*, int, const bf16_t*, const bf16_t*,
 void launch_gemm_stripes(const bf16_t*, const bf16_t*, float*, int, int,
                          int, int, int, int, int, int, int, int,
                          hipStream_t);
+void launch_gemm_small(const bf16_t*, const bf16_t*, const void*, bool,
+                       void*, bool, int, void*, bool, int, int, int, int,
+                       int, int, bool, hipStream_t);
 void launch_gather_bf16(const bf16_t*, const long*, bf16_t*, long, int, long,
                         hipStream_t);
 void launch_gather_f32(const float*, const long*, float*, long, int, long,
@@ -239,6 +242,24 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
                 colsum_out.scalar_type() == out.scalar_type(),
                 "colsum_out must match out dtype, [N]");
     colsum_p = colsum_out.data_ptr();
+  }
+  // small-tile ONE-kernel path for tiny tile grids (mnist fwd
+  // 100x100x784 and dW1 784x100x100): the 4 waves split K in-block
+  // (no split-K reduce pass — the second dispatch floor cost more
+  // than the whole GEMM at these sizes)
+  const long stiles = (long)((M + 31) / 32) * ((N + 31) / 32);
+  const bool cs_small_ok =
+      colsum_p == nullptr ||
+      (trans_a && !trans_b && bias_p == nullptr && act == 0 &&
+       colsum_out.scalar_type() == out.scalar_type());
+  if (act <= 1 && !trans_b && aux_p == nullptr && cs_small_ok &&
+      stiles <= 128) {
+    launch_gemm_small((const bf16_t*)a.data_ptr(),
+                      (const bf16_t*)b.data_ptr(), bias_p, bias_bf16,
+                      out.data_ptr(), out_f32, act == 1 ? 1 : 0, colsum_p,
+                      out_f32, M, N, Ka, a.size(1), b.size(1), N, trans_a,
+                      cur_stream());
+    return out;
   }
   // split-K when the plain tile grid can't feed the 256-CU chip and K
   // has enough depth to slice (any transpose combo; not with the

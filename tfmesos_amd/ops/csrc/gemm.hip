@@ -275,6 +275,148 @@ void splitk_reduce_kernel(const float* __restrict__ ws, const void* __restrict__
   }
 }
 
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+// ---------------------------------------------------------- small-tile path
+//
+// One-kernel GEMM for tiny tile grids (the mnist fwd 100x100x784 and
+// dW1 784x100x100): the two-phase split-K path paid TWO dispatch
+// floors (~13 us for ~16 MFLOP). 32x32 output tile per block; the 4
+// waves SPLIT K privately — each stages into its OWN LDS quarter, so
+// the k-loop has NO barrier — then one barrier, a cross-wave
+// accumulate by wave 0, and the fused epilogue (bias/relu, bf16/f32
+// out). colsum (db1) is summed during B staging and reduced by
+// blockIdx.x==0 only (it depends on K alone, so every m-tile block
+// would compute the same value; single-writer keeps it deterministic,
+// no atomics, overwrite semantics like the 64x64 path).
+// TA: op(A) = A^T (A stored [K,M] — transposed scatter staging).
+// B is always untransposed here ([K,N] memory).
+template <bool TA, bool CS>
+__global__ __launch_bounds__(256)
+void gemm_small_kernel(const __bf16* __restrict__ A,
+                       const __bf16* __restrict__ B,
+                       const void* __restrict__ bias, bool bias_bf16,
+                       void* __restrict__ Cout, bool out_f32, int relu,
+                       void* __restrict__ colsum_out, bool cs_f32,
+                       int M, int N, int K, int lda, int ldb, int ldc) {
+  __shared__ __align__(16) __bf16 As[4][32 * (BK + APAD) + 32];
+  __shared__ __align__(16) __bf16 Bs[4][32 * (BK + APAD) + 32];
+  __shared__ float red[4][64 * 16];
+  __shared__ float csred[4][32];
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int w = t >> 6;
+  const int m0 = blockIdx.x * 32;
+  const int n0 = blockIdx.y * 32;
+  // per-wave K span, BK-aligned
+  const int kq = (((K + 3) / 4) + BK - 1) / BK * BK;
+  const int ks = w * kq;
+  const int ke = min(ks + kq, K);
+  __bf16* Aw = As[w];
+  __bf16* Bw = Bs[w];
+  const int kk = lane & 31;        // staging k-row within the chunk
+  const int half = lane >> 5;      // staging x-half (16 cols)
+  f32x4 acc[4] = {};
+  float csp[16] = {};
+  for (int k0 = ks; k0 < ke; k0 += BK) {
+    // ---- stage A image [m][k] (32x40+skew)
+    if (TA) {
+      // A stored [K, M]: vector load 16 m's of one k, transposed scatter
+      const int gk = k0 + kk;
+      const int gm = m0 + half * 16;
+      const __bf16* src = A + (long)gk * lda + gm;
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        const __bf16 v = (gk < ke && gm + j < M) ? src[j] : (__bf16)0.f;
+        srow(Aw, half * 16 + j)[kk] = v;
+      }
+    } else {
+      // A stored [M, K]: row m = lane&31, 16 k's = half*16..+16
+      const int gm = m0 + (lane & 31);
+      const int gk0 = k0 + half * 16;
+      const __bf16* src = A + (long)gm * lda + gk0;
+#pragma unroll
+      for (int j = 0; j < 16; ++j)
+        srow(Aw, lane & 31)[half * 16 + j] =
+            (gm < M && gk0 + j < ke) ? src[j] : (__bf16)0.f;
+    }
+    // ---- stage B image [n][k]: vector load 16 n's of one k, scatter
+    {
+      const int gk = k0 + kk;
+      const int gn = n0 + half * 16;
+      const __bf16* src = B + (long)gk * ldb + gn;
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        const __bf16 v = (gk < ke && gn + j < N) ? src[j] : (__bf16)0.f;
+        srow(Bw, half * 16 + j)[kk] = v;
+        if (CS) csp[j] += (float)v;
+      }
+    }
+    // no barrier: each wave reads only its own LDS quarter, and LDS
+    // ops from one wave are ordered by the s_waitcnt the reads imply
+    __builtin_amdgcn_s_waitcnt(0);   // drain lgkm before fragment reads
+#pragma unroll
+    for (int kh = 0; kh < 2; ++kh) {
+      const int ko = kh * 16 + ((lane >> 5) << 3);
+      bf16x8 a = *(const bf16x8*)&srow(Aw, lane & 31)[ko];
+      bf16x8 b = *(const bf16x8*)&srow(Bw, lane & 31)[ko];
+      *(f32x16*)acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          a, b, *(f32x16*)acc, 0, 0, 0);
+    }
+  }
+  // cross-wave reduce (the only barrier)
+#pragma unroll
+  for (int v = 0; v < 16; ++v) red[w][lane * 16 + v] = ((float*)acc)[v];
+  if (CS && blockIdx.x == 0) {
+    // fold this wave's 32 k-lanes: lanes of one half share 16 n's
+    float c2[16];
+#pragma unroll
+    for (int j = 0; j < 16; ++j) c2[j] = csp[j];
+#pragma unroll
+    for (int off = 1; off < 32; off <<= 1)
+#pragma unroll
+      for (int j = 0; j < 16; ++j) c2[j] += __shfl_xor(c2[j], off, 64);
+    if (kk == 0) {
+      // lanes 0 and 32 hold the two n-halves; csred[w][n-within-tile]
+#pragma unroll
+      for (int j = 0; j < 16; ++j) csred[w][half * 16 + j] = c2[j];
+    }
+  }
+  __syncthreads();
+  if (w != 0) return;
+  f32x16 tot;
+#pragma unroll
+  for (int v = 0; v < 16; ++v)
+    tot[v] = red[0][lane * 16 + v] + red[1][lane * 16 + v] +
+             red[2][lane * 16 + v] + red[3][lane * 16 + v];
+  const int n = n0 + (lane & 31);
+  if (n < N) {
+    float bv = 0.f;
+    if (bias != nullptr)
+      bv = bias_bf16 ? (float)((const __bf16*)bias)[n]
+                     : ((const float*)bias)[n];
+#pragma unroll
+    for (int v = 0; v < 16; ++v) {
+      const int m = m0 + ((v >> 2) << 3) + ((lane >> 5) << 2) + (v & 3);
+      if (m >= M) continue;
+      float x = tot[v] + bv;
+      if (relu) x = x > 0.f ? x : 0.f;
+      if (out_f32) ((float*)Cout)[(long)m * ldc + n] = x;
+      else ((__bf16*)Cout)[(long)m * ldc + n] = (__bf16)x;
+    }
+  }
+  if (CS && blockIdx.x == 0 && lane < 32) {
+    const int nn = n0 + lane;
+    if (nn < N) {
+      const float sv = csred[0][lane] + csred[1][lane] + csred[2][lane] +
+                       csred[3][lane];
+      if (cs_f32) ((float*)colsum_out)[nn] = sv;
+      else ((__bf16*)colsum_out)[nn] = (__bf16)sv;
+    }
+  }
+}
+
 }  // namespace
 
 void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
@@ -375,6 +517,23 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
 // the caller fuses the stripe reduction into a consumer kernel (the
 // mnist fused head reduces them while staging h into LDS, so h never
 // exists in global memory and the reduce launch disappears).
+
+void launch_gemm_small(const bf16_t* A, const bf16_t* B, const void* bias,
+                       bool bias_bf16, void* C, bool out_f32, int relu,
+                       void* colsum_out, bool cs_f32, int M, int N, int K,
+                       int lda, int ldb, int ldc, bool ta,
+                       hipStream_t stream) {
+  dim3 grid(ceil_div(M, 32), ceil_div(N, 32)), block(256);
+#define GS(TAv, CSv)                                                        \
+  hipLaunchKernelGGL((gemm_small_kernel<TAv, CSv>), grid, block, 0,         \
+                     stream, (const __bf16*)A, (const __bf16*)B, bias,      \
+                     bias_bf16, C, out_f32, relu, colsum_out, cs_f32, M, N, \
+                     K, lda, ldb, ldc)
+  if (ta) { if (colsum_out) GS(true, true); else GS(true, false); }
+  else    { if (colsum_out) GS(false, true); else GS(false, false); }
+#undef GS
+}
+
 void launch_gemm_stripes(const bf16_t* A, const bf16_t* B, float* ws,
                          int kc, int nslice, int M, int N, int K, int lda,
                          int ldb, int ldc, int veca, int vecb,
