@@ -309,53 +309,110 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
   const int w = t >> 6;
   const int m0 = blockIdx.x * 32;
   const int n0 = blockIdx.y * 32;
-  // per-wave K span, BK-aligned
-  const int kq = (((K + 3) / 4) + BK - 1) / BK * BK;
+  const int kq = (((K + 3) / 4) + BK - 1) / BK * BK;  // per-wave K span
   const int ks = w * kq;
   const int ke = min(ks + kq, K);
   __bf16* Aw = As[w];
   __bf16* Bw = Bs[w];
-  const int kk = lane & 31;        // staging k-row within the chunk
-  const int half = lane >> 5;      // staging x-half (16 cols)
+  const int kk = lane & 31;
+  const int half = lane >> 5;
+  const bool av = (lda & 7) == 0;   // bf16x8-aligned rows
+  const bool bv = (ldb & 7) == 0;
   f32x4 acc[4] = {};
   float csp[16] = {};
-  for (int k0 = ks; k0 < ke; k0 += BK) {
-    // ---- stage A image [m][k] (32x40+skew)
+  bf16x8 ra0, ra1, rb0, rb1;        // in-flight chunk (load -> commit)
+
+  // load: global reads into REGISTERS (vector when the row allows —
+  // guarded scalar loads serialized the first cut at 2x the split-K
+  // path); commit: LDS stores. Registers double-buffer the single LDS
+  // image: the per-wave DS pipe is in-order, so this k-chunk's
+  // fragment reads are serviced before the next chunk's commit writes.
+  auto load_a = [&](int k0) {
+    ra0 = bf16x8{};
+    ra1 = bf16x8{};
     if (TA) {
-      // A stored [K, M]: vector load 16 m's of one k, transposed scatter
+      // A stored [K, M]: 16 consecutive m of one k
       const int gk = k0 + kk;
       const int gm = m0 + half * 16;
-      const __bf16* src = A + (long)gk * lda + gm;
+      if (gk < ke) {
+        const __bf16* src = A + (long)gk * lda + gm;
+        if (av && gm + 16 <= M && ((m0 & 7) == 0)) {
+          ra0 = *(const bf16x8*)src;
+          ra1 = *(const bf16x8*)(src + 8);
+        } else {
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const __bf16 v = (gk < ke && gm + j < M) ? src[j] : (__bf16)0.f;
-        srow(Aw, half * 16 + j)[kk] = v;
+          for (int j = 0; j < 8; ++j) {
+            if (gm + j < M) ra0[j] = src[j];
+            if (gm + 8 + j < M) ra1[j] = src[8 + j];
+          }
+        }
       }
     } else {
-      // A stored [M, K]: row m = lane&31, 16 k's = half*16..+16
-      const int gm = m0 + (lane & 31);
+      // A stored [M, K]: row m, 16 consecutive k
+      const int gm = m0 + kk;
       const int gk0 = k0 + half * 16;
-      const __bf16* src = A + (long)gm * lda + gk0;
+      if (gm < M) {
+        const __bf16* src = A + (long)gm * lda + gk0;
+        if (av && gk0 + 16 <= ke) {
+          ra0 = *(const bf16x8*)src;
+          ra1 = *(const bf16x8*)(src + 8);
+        } else {
 #pragma unroll
-      for (int j = 0; j < 16; ++j)
-        srow(Aw, lane & 31)[half * 16 + j] =
-            (gm < M && gk0 + j < ke) ? src[j] : (__bf16)0.f;
+          for (int j = 0; j < 8; ++j) {
+            if (gk0 + j < ke) ra0[j] = src[j];
+            if (gk0 + 8 + j < ke) ra1[j] = src[8 + j];
+          }
+        }
+      }
     }
-    // ---- stage B image [n][k]: vector load 16 n's of one k, scatter
+    // B stored [K, N]: 16 consecutive n of one k
+    rb0 = bf16x8{};
+    rb1 = bf16x8{};
     {
       const int gk = k0 + kk;
       const int gn = n0 + half * 16;
-      const __bf16* src = B + (long)gk * ldb + gn;
+      if (gk < ke) {
+        const __bf16* src = B + (long)gk * ldb + gn;
+        if (bv && gn + 16 <= N && ((n0 & 7) == 0)) {
+          rb0 = *(const bf16x8*)src;
+          rb1 = *(const bf16x8*)(src + 8);
+        } else {
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const __bf16 v = (gk < ke && gn + j < N) ? src[j] : (__bf16)0.f;
-        srow(Bw, half * 16 + j)[kk] = v;
-        if (CS) csp[j] += (float)v;
+          for (int j = 0; j < 8; ++j) {
+            if (gn + j < N) rb0[j] = src[j];
+            if (gn + 8 + j < N) rb1[j] = src[8 + j];
+          }
+        }
       }
     }
-    // no barrier: each wave reads only its own LDS quarter, and LDS
-    // ops from one wave are ordered by the s_waitcnt the reads imply
-    __builtin_amdgcn_s_waitcnt(0);   // drain lgkm before fragment reads
+  };
+  auto commit = [&]() {
+    if (TA) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        srow(Aw, half * 16 + j)[kk] = ra0[j];
+        srow(Aw, half * 16 + 8 + j)[kk] = ra1[j];
+      }
+    } else {
+      *(bf16x8*)&srow(Aw, kk)[half * 16] = ra0;
+      *(bf16x8*)&srow(Aw, kk)[half * 16 + 8] = ra1;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      srow(Bw, half * 16 + j)[kk] = rb0[j];
+      srow(Bw, half * 16 + 8 + j)[kk] = rb1[j];
+      if (CS) {
+        csp[j] += (float)rb0[j];
+        csp[8 + j] += (float)rb1[j];
+      }
+    }
+  };
+
+  load_a(ks);
+  commit();
+  for (int k0 = ks; k0 < ke; k0 += BK) {
+    const bool more = k0 + BK < ke;
+    if (more) load_a(k0 + BK);
 #pragma unroll
     for (int kh = 0; kh < 2; ++kh) {
       const int ko = kh * 16 + ((lane >> 5) << 3);
@@ -364,12 +421,13 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
       *(f32x16*)acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
           a, b, *(f32x16*)acc, 0, 0, 0);
     }
+    if (more) commit();
   }
-  // cross-wave reduce (the only barrier)
+
+  // cross-wave reduce (the only barrier) + epilogue by wave 0
 #pragma unroll
   for (int v = 0; v < 16; ++v) red[w][lane * 16 + v] = ((float*)acc)[v];
   if (CS && blockIdx.x == 0) {
-    // fold this wave's 32 k-lanes: lanes of one half share 16 n's
     float c2[16];
 #pragma unroll
     for (int j = 0; j < 16; ++j) c2[j] = csp[j];
@@ -378,7 +436,6 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
 #pragma unroll
       for (int j = 0; j < 16; ++j) c2[j] += __shfl_xor(c2[j], off, 64);
     if (kk == 0) {
-      // lanes 0 and 32 hold the two n-halves; csred[w][n-within-tile]
 #pragma unroll
       for (int j = 0; j < 16; ++j) csred[w][half * 16 + j] = c2[j];
     }
@@ -392,15 +449,15 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
              red[2][lane * 16 + v] + red[3][lane * 16 + v];
   const int n = n0 + (lane & 31);
   if (n < N) {
-    float bv = 0.f;
+    float bvv = 0.f;
     if (bias != nullptr)
-      bv = bias_bf16 ? (float)((const __bf16*)bias)[n]
-                     : ((const float*)bias)[n];
+      bvv = bias_bf16 ? (float)((const __bf16*)bias)[n]
+                      : ((const float*)bias)[n];
 #pragma unroll
     for (int v = 0; v < 16; ++v) {
       const int m = m0 + ((v >> 2) << 3) + ((lane >> 5) << 2) + (v & 3);
       if (m >= M) continue;
-      float x = tot[v] + bv;
+      float x = tot[v] + bvv;
       if (relu) x = x > 0.f ? x : 0.f;
       if (out_f32) ((float*)Cout)[(long)m * ldc + n] = x;
       else ((__bf16*)Cout)[(long)m * ldc + n] = (__bf16)x;
