@@ -48,21 +48,21 @@ class NMFWorkload(object):
         scale = 2.0 / self.X.numel()
         c = 2.0 * self.lam / self.X.numel()
         if self.use_bf16:
-            # 10 kernels/step: 3 MFMA GEMMs (fp32 grads via the fused
+            # 6 kernels/step: 3 MFMA GEMMs (fp32 grads via the fused
             # output-cast epilogue), one bf16 residual sub, and per
-            # factor {clamp, add, fused SGD apply + shadow refresh}
+            # factor ONE fused SGD apply (shadow refresh AND the soft
+            # nonnegativity penalty g += c*min(p,0) fold into the same
+            # pass — the per-factor clamp+add pair was 4 more launches)
             P = ops.gemm_bias_act(self.Wb, self.Hb)            # [n,n] bf16
             E = P.sub_(self.Xb)
             ops.gemm_bias_act(E, self.Hb, trans_b=True, out=self.dW)
             ops.gemm_bias_act(self.Wb, E, trans_a=True, out=self.dH)
-            # soft nonnegativity penalty folded into the grad buffers
-            # (alpha = c/scale so the apply's grad_scale restores c)
-            self.dW.add_(torch.clamp(self.W, max=0.0), alpha=c / scale)
-            self.dH.add_(torch.clamp(self.H, max=0.0), alpha=c / scale)
             ops.fused_sgd(self.W.view(-1), self.dW.view(-1), lr=self.lr,
-                          bf16_out=self.Wb.view(-1), grad_scale=scale)
+                          bf16_out=self.Wb.view(-1), grad_scale=scale,
+                          neg_decay=c)
             ops.fused_sgd(self.H.view(-1), self.dH.view(-1), lr=self.lr,
-                          bf16_out=self.Hb.view(-1), grad_scale=scale)
+                          bf16_out=self.Hb.view(-1), grad_scale=scale,
+                          neg_decay=c)
             return None
         E = self.W @ self.H - self.X
         dW = E @ self.H.t() * scale

@@ -47,12 +47,14 @@ def ext_available():
 
 @torch.no_grad()
 def fused_sgd(param, grad, lr, momentum=0.0, weight_decay=0.0, momentum_buf=None,
-              bf16_out=None, grad_scale=1.0):
+              bf16_out=None, grad_scale=1.0, neg_decay=0.0):
     """In-place SGD(+momentum, +wd) on an fp32 master param.
 
     grad may be bf16 or fp32 and is multiplied by grad_scale (worker-mean
     for sync replicas). If bf16_out is given, also writes the updated
-    param as bf16 (the broadcast copy) in the same pass.
+    param as bf16 (the broadcast copy) in the same pass. neg_decay folds
+    a soft nonnegativity penalty (g += neg_decay * min(p, 0) — the NMF
+    objective) into the same pass.
     """
     if param.is_cuda:
         _ext().fused_sgd(param, grad, momentum_buf if momentum_buf is not None
@@ -60,9 +62,11 @@ def fused_sgd(param, grad, lr, momentum=0.0, weight_decay=0.0, momentum_buf=None
                          bf16_out if bf16_out is not None
                          else torch.empty(0, dtype=torch.bfloat16, device=param.device),
                          float(lr), float(momentum), float(weight_decay),
-                         float(grad_scale))
+                         float(grad_scale), float(neg_decay))
         return
     g = grad.float() * grad_scale
+    if neg_decay:
+        g = g + neg_decay * torch.clamp(param, max=0.0)
     if weight_decay:
         g = g + weight_decay * param
     if momentum and momentum_buf is not None:

@@ -41,7 +41,8 @@ template <typename G, bool MOM, bool BF16OUT>
 __global__ void sgd_kernel(float* __restrict__ p, const G* __restrict__ g,
                            float* __restrict__ mbuf,
                            bf16_t* __restrict__ pbf, long n, float lr,
-                           float momentum, float wd, float gscale) {
+                           float momentum, float wd, float gscale,
+                           float negdecay) {
   long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 4;
   long stride = (long)gridDim.x * blockDim.x * 4;
   long i = i0;
@@ -51,6 +52,13 @@ __global__ void sgd_kernel(float* __restrict__ p, const G* __restrict__ g,
     f32x4 gv = load_g4(g, i) * gscale;
     f32x4 pv = *(const f32x4*)&p[i];
     if (wd != 0.f) gv += wd * pv;
+    if (negdecay != 0.f) {
+      // soft nonnegativity penalty (NMF): g += c * min(p, 0) — folds
+      // the per-factor clamp+add kernels into the apply pass
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        gv[j] += negdecay * fminf(pv[j], 0.f);
+    }
     if (MOM) {
       f32x4 m = *(const f32x4*)&mbuf[i] * momentum + gv;
       *(f32x4*)&mbuf[i] = m;
@@ -69,6 +77,7 @@ __global__ void sgd_kernel(float* __restrict__ p, const G* __restrict__ g,
     float gv = load_g(g, i) * gscale;
     float pv = p[i];
     if (wd != 0.f) gv += wd * pv;
+    if (negdecay != 0.f) gv += negdecay * fminf(pv, 0.f);
     if (MOM) {
       float m = mbuf[i] * momentum + gv;
       mbuf[i] = m;
@@ -171,11 +180,12 @@ inline dim3 apply_grid(long n) {
 
 void launch_sgd(float* p, const void* g, bool g_bf16, float* mbuf,
                 bf16_t* pbf, long n, float lr, float momentum, float wd,
-                float gscale, hipStream_t stream) {
+                float gscale, float negdecay, hipStream_t stream) {
   dim3 grid = apply_grid(n), block(256);
 #define DISP(GT, MOMV, OUTV)                                               \
   hipLaunchKernelGGL((sgd_kernel<GT, MOMV, OUTV>), grid, block, 0, stream, \
-                     p, (const GT*)g, mbuf, pbf, n, lr, momentum, wd, gscale)
+                     p, (const GT*)g, mbuf, pbf, n, lr, momentum, wd,     \
+                     gscale, negdecay)
   if (g_bf16) {
     if (mbuf) { if (pbf) DISP(bf16_t, true, true); else DISP(bf16_t, true, false); }
     else      { if (pbf) DISP(bf16_t, false, true); else DISP(bf16_t, false, false); }

@@ -9,7 +9,7 @@ typedef __hip_bfloat16 bf16_t;
 
 // launchers from the .hip translation units
 void launch_sgd(float*, const void*, bool, float*, bf16_t*, long, float,
-                float, float, float, hipStream_t);
+                float, float, float, float, hipStream_t);
 void launch_adam(float*, const void*, bool, float*, float*, bf16_t*, long,
                  long, float, float, float, float, float, float, hipStream_t);
 void launch_adagrad(float*, const void*, bool, float*, bf16_t*, long, float,
@@ -122,7 +122,8 @@ bf16_t* opt_bf16(torch::Tensor& t, long n, const char* name) {
 
 void fused_sgd(torch::Tensor param, torch::Tensor grad,
                torch::Tensor momentum_buf, torch::Tensor bf16_out, double lr,
-               double momentum, double weight_decay, double grad_scale) {
+               double momentum, double weight_decay, double grad_scale,
+               double neg_decay) {
   TORCH_CHECK(param.is_cuda() && param.is_contiguous() &&
               param.scalar_type() == torch::kFloat32,
               "param must be contiguous fp32 on GPU");
@@ -134,7 +135,7 @@ void fused_sgd(torch::Tensor param, torch::Tensor grad,
              opt_f32(momentum_buf, n, "momentum_buf"),
              opt_bf16(bf16_out, n, "bf16_out"), n, (float)lr,
              (float)momentum, (float)weight_decay, (float)grad_scale,
-             cur_stream());
+             (float)neg_decay, cur_stream());
 }
 
 void fused_adam(torch::Tensor param, torch::Tensor grad, torch::Tensor m,
@@ -1066,7 +1067,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd, "fused SGD apply + bf16 shadow",
         py::arg("param"), py::arg("grad"), py::arg("momentum_buf"),
         py::arg("bf16_out"), py::arg("lr"), py::arg("momentum") = 0.0,
-        py::arg("weight_decay") = 0.0, py::arg("grad_scale") = 1.0);
+        py::arg("weight_decay") = 0.0, py::arg("grad_scale") = 1.0,
+        py::arg("neg_decay") = 0.0);
   m.def("fused_adam", &fused_adam, "fused Adam apply + bf16 shadow",
         py::arg("param"), py::arg("grad"), py::arg("m"), py::arg("v"),
         py::arg("bf16_out"), py::arg("step"), py::arg("lr"),
