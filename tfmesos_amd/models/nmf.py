@@ -48,13 +48,13 @@ class NMFWorkload(object):
         scale = 2.0 / self.X.numel()
         c = 2.0 * self.lam / self.X.numel()
         if self.use_bf16:
-            # 6 kernels/step: 3 MFMA GEMMs (fp32 grads via the fused
+            # 5 kernels/step: 3 MFMA GEMMs (residual fused into the first) (fp32 grads via the fused
             # output-cast epilogue), one bf16 residual sub, and per
             # factor ONE fused SGD apply (shadow refresh AND the soft
             # nonnegativity penalty g += c*min(p,0) fold into the same
             # pass — the per-factor clamp+add pair was 4 more launches)
-            P = ops.gemm_bias_act(self.Wb, self.Hb)            # [n,n] bf16
-            E = P.sub_(self.Xb)
+            # E = W@H - X in ONE kernel (fused residual epilogue)
+            E = ops.gemm_bias_act(self.Wb, self.Hb, act="sub", aux=self.Xb)
             ops.gemm_bias_act(E, self.Hb, trans_b=True, out=self.dW)
             ops.gemm_bias_act(self.Wb, E, trans_a=True, out=self.dH)
             ops.fused_sgd(self.W.view(-1), self.dW.view(-1), lr=self.lr,

@@ -140,7 +140,7 @@ def gemm_bias_act(a, b, bias=None, act="none", trans_a=False, trans_b=False,
     GPU: hand-written MFMA kernel (csrc/gemm.hip) with split-K for deep
     skinny shapes. CPU: torch reference in fp32.
     """
-    act_code = {"none": 0, "relu": 1, "relu_bwd": 2}[act]
+    act_code = {"none": 0, "relu": 1, "relu_bwd": 2, "sub": 3}[act]
     if a.is_cuda:
         empty = torch.empty(0, device=a.device)
         ebias = bias if bias is not None else empty
@@ -163,6 +163,8 @@ def gemm_bias_act(a, b, bias=None, act="none", trans_a=False, trans_b=False,
         c = torch.relu(c)
     elif act == "relu_bwd":
         c = c * (aux.float() > 0)
+    elif act == "sub":
+        c = c - aux.float()
     if colsum_out is not None:
         colsum_out.copy_(y.sum(0).to(colsum_out.dtype))
     if out is not None:

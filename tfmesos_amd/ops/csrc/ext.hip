@@ -234,6 +234,13 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
     TORCH_CHECK(aux.scalar_type() == torch::kBFloat16 && aux.is_contiguous()
                 && aux.numel() == (long)M * N, "aux must be bf16 [M,N]");
     aux_p = (const bf16_t*)aux.data_ptr();
+  } else if (act == 3) {
+    // out = A@B - aux (the NMF residual, fused into the epilogue)
+    TORCH_CHECK(!trans_a && !trans_b && !out_f32 && bias_p == nullptr,
+                "act=sub needs nn, bf16 out, no bias");
+    TORCH_CHECK(aux.scalar_type() == torch::kBFloat16 && aux.is_contiguous()
+                && aux.numel() == (long)M * N, "aux must be bf16 [M,N]");
+    aux_p = (const bf16_t*)aux.data_ptr();
   }
   void* colsum_p = nullptr;
   if (colsum_out.numel() > 0) {
@@ -282,7 +289,7 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
   const bool cs_ok = colsum_p == nullptr ||
                      (trans_a && !trans_b && bias_p == nullptr && act == 0);
   const long kmin = 256;
-  if (act != 2 && cs_ok && nx * ny < 256 && Ka >= kmin) {
+  if (act <= 1 && cs_ok && nx * ny < 256 && Ka >= kmin) {
     int want = std::min(colsum_p != nullptr ? (int)((Ka + 31) / 32)
                                             : (int)(Ka / 64),
                         256 / (nx * ny));

@@ -217,6 +217,7 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
         v += bv;
         if (ACT == 1) v = v > 0.f ? v : 0.f;
         if (ACT == 2) v = (float)aux[idx] > 0.f ? v : 0.f;
+        if (ACT == 3) v -= (float)aux[idx];   // fused residual (NMF E)
         if (OUTF32)
           ((float*)Cout)[idx] = v;
         else
@@ -496,6 +497,10 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
   // validated by the bindings in ext.hip
   if (act == 2) {
     LAUNCH(false, true, 2, false, false, false, false);
+    return;
+  }
+  if (act == 3) {   // out = A@B - aux (nn, bf16 out; binding-gated)
+    LAUNCH(false, false, 3, false, false, false, false);
     return;
   }
   if (cs && !sk) {
