@@ -761,3 +761,37 @@ def test_bn_group_multi_matches_per_branch():
                               rtol=1e-2, atol=1e-2), i
         assert torch.allclose(bs[i].grad.float(), b2.grad.float(),
                               rtol=1e-2, atol=1e-2), i
+
+
+@pytest.mark.gpu
+def test_gemm_fused_sub_epilogue():
+    """act='sub' (out = A@B - aux, the NMF residual) vs fp32 torch."""
+    import torch
+
+    from tfmesos_amd import ops
+
+    torch.manual_seed(11)
+    a = torch.randn(257, 300, device="cuda:0", dtype=torch.bfloat16)
+    b = torch.randn(300, 190, device="cuda:0", dtype=torch.bfloat16)
+    x = torch.randn(257, 190, device="cuda:0", dtype=torch.bfloat16)
+    got = ops.gemm_bias_act(a, b, act="sub", aux=x)
+    ref = a.float() @ b.float() - x.float()
+    err = (got.float() - ref).abs().max() / ref.abs().max()
+    assert float(err) < 0.02, float(err)
+
+
+@pytest.mark.gpu
+def test_fused_sgd_neg_decay():
+    """neg_decay (g += c*min(p,0)) vs the explicit clamp+add form."""
+    import torch
+
+    from tfmesos_amd import ops
+
+    torch.manual_seed(12)
+    p1 = torch.randn(5000, device="cuda:0")
+    p2 = p1.clone()
+    g = torch.randn(5000, device="cuda:0")
+    ops.fused_sgd(p1, g, lr=0.1, grad_scale=0.5, neg_decay=0.3)
+    g2 = g * 0.5 + 0.3 * torch.clamp(p2, max=0.0)
+    p2 -= 0.1 * g2
+    assert torch.allclose(p1, p2, atol=1e-6)
