@@ -795,3 +795,38 @@ def test_fused_sgd_neg_decay():
     g2 = g * 0.5 + 0.3 * torch.clamp(p2, max=0.0)
     p2 -= 0.1 * g2
     assert torch.allclose(p1, p2, atol=1e-6)
+
+
+@pytest.mark.gpu
+def test_gemm_sgd_pair_simultaneous_semantics():
+    """Fused GEMM->SGD pair vs explicit fp32 reference: both gradient
+    GEMMs must read PRE-update factors (simultaneous update)."""
+    import torch
+
+    from tfmesos_amd import ops
+
+    torch.manual_seed(13)
+    dev = "cuda:0"
+    n, r, k = 300, 64, 300
+    W = torch.randn(n, r, device=dev) * 0.1
+    H = torch.randn(r, k, device=dev) * 0.1
+    Wb = W.to(torch.bfloat16)
+    Hb = H.to(torch.bfloat16)
+    E = torch.randn(n, k, device=dev, dtype=torch.bfloat16) * 0.1
+    W0, H0 = W.clone(), H.clone()
+    lr, scale, c = 0.05, 0.25, 0.1
+    ops.gemm_sgd_pair((E, Hb, W, Wb, False, True),
+                      (Wb, E, H, Hb, True, False),
+                      lr=lr, grad_scale=scale, neg_decay=c)
+    # reference: both grads from PRE-update values
+    Wb0 = W0.to(torch.bfloat16).float()
+    gW = E.float() @ Hb.float().t() * scale + c * torch.clamp(W0, max=0.0)
+    gH = Wb0.t() @ E.float() * scale + c * torch.clamp(H0, max=0.0)
+    refW = W0 - lr * gW
+    refH = H0 - lr * gH
+    assert torch.allclose(W, refW, atol=2e-2, rtol=1e-2), \
+        float((W - refW).abs().max())
+    assert torch.allclose(H, refH, atol=2e-2, rtol=1e-2), \
+        float((H - refH).abs().max())
+    assert torch.allclose(Wb.float(), W.to(torch.bfloat16).float())
+    assert torch.allclose(Hb.float(), H.to(torch.bfloat16).float())

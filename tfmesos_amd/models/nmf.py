@@ -30,13 +30,12 @@ class NMFWorkload(object):
         self.use_bf16 = self.device.type == "cuda"
         if self.use_bf16:
             self.Xb = self.X.to(torch.bfloat16)
-            # persistent bf16 shadows + fp32 grad buffers: the fused
-            # SGD apply refreshes the shadows in the same pass, so the
-            # step has NO per-step fp32<->bf16 casts of the big tensors
+            # persistent bf16 shadows: refreshed inside the fused
+            # GEMM->SGD reduce, so the step has NO per-step
+            # fp32<->bf16 casts of the big tensors and the factor
+            # gradients never materialize
             self.Wb = self.W.to(torch.bfloat16)
             self.Hb = self.H.to(torch.bfloat16)
-            self.dW = torch.empty_like(self.W)
-            self.dH = torch.empty_like(self.H)
 
     def loss(self):
         E = self.W @ self.H - self.X
@@ -48,21 +47,19 @@ class NMFWorkload(object):
         scale = 2.0 / self.X.numel()
         c = 2.0 * self.lam / self.X.numel()
         if self.use_bf16:
-            # 5 kernels/step: 3 MFMA GEMMs (residual fused into the first) (fp32 grads via the fused
-            # output-cast epilogue), one bf16 residual sub, and per
-            # factor ONE fused SGD apply (shadow refresh AND the soft
-            # nonnegativity penalty g += c*min(p,0) fold into the same
-            # pass — the per-factor clamp+add pair was 4 more launches)
-            # E = W@H - X in ONE kernel (fused residual epilogue)
+            # 5 dispatches/step: E = W@H - X in ONE kernel (fused
+            # residual epilogue), then each factor's GEMM+apply pair
+            # rides the fused GEMM->SGD path — the split-K reduce IS
+            # the optimizer update (shadow refresh + nonneg penalty
+            # folded in; the gradients never hit HBM as tensors)
             E = ops.gemm_bias_act(self.Wb, self.Hb, act="sub", aux=self.Xb)
-            ops.gemm_bias_act(E, self.Hb, trans_b=True, out=self.dW)
-            ops.gemm_bias_act(self.Wb, E, trans_a=True, out=self.dH)
-            ops.fused_sgd(self.W.view(-1), self.dW.view(-1), lr=self.lr,
-                          bf16_out=self.Wb.view(-1), grad_scale=scale,
-                          neg_decay=c)
-            ops.fused_sgd(self.H.view(-1), self.dH.view(-1), lr=self.lr,
-                          bf16_out=self.Hb.view(-1), grad_scale=scale,
-                          neg_decay=c)
+            # paired form: BOTH gradient GEMMs read the pre-update
+            # factors (simultaneous update, matching the reference's
+            # one-train-op apply), then both fused applies run
+            ops.gemm_sgd_pair(
+                (E, self.Hb, self.W, self.Wb, False, True),
+                (self.Wb, E, self.H, self.Hb, True, False),
+                lr=self.lr, grad_scale=scale, neg_decay=c)
             return None
         E = self.W @ self.H - self.X
         dW = E @ self.H.t() * scale

@@ -173,6 +173,55 @@ def gemm_bias_act(a, b, bias=None, act="none", trans_a=False, trans_b=False,
     return c.to(a.dtype)
 
 
+@torch.no_grad()
+def gemm_sgd(a, b, param, lr, shadow=None, grad_scale=1.0, neg_decay=0.0,
+             trans_a=False, trans_b=False):
+    """Fused GEMM -> SGD: p -= lr*(grad_scale*(op(A)@op(B)) +
+    neg_decay*min(p,0)); optional bf16 shadow refresh. The gradient
+    never materializes on GPU (the split-K reduce IS the apply)."""
+    if a.is_cuda:
+        e = torch.empty(0, dtype=torch.bfloat16, device=a.device)
+        _ext().gemm_sgd(a, b, bool(trans_a), bool(trans_b),
+                        param.view(-1), shadow.view(-1) if shadow is not None
+                        else e, float(lr), float(grad_scale),
+                        float(neg_decay))
+        return
+    x = a.float().t() if trans_a else a.float()
+    y = b.float().t() if trans_b else b.float()
+    g = (x @ y).reshape(param.shape)
+    fused_sgd(param, g, lr, grad_scale=grad_scale, neg_decay=neg_decay,
+              bf16_out=shadow)
+
+
+@torch.no_grad()
+def gemm_sgd_pair(spec1, spec2, lr, grad_scale=1.0, neg_decay=0.0):
+    """Two fused GEMM->SGD updates with SIMULTANEOUS semantics: both
+    gradient GEMMs read pre-update operands (their stripe phases run
+    before either apply). spec: (a, b, param, shadow, trans_a,
+    trans_b)."""
+    (a1, b1, p1, s1, ta1, tb1) = spec1
+    (a2, b2, p2, s2, ta2, tb2) = spec2
+    if a1.is_cuda:
+        e = torch.empty(0, dtype=torch.bfloat16, device=a1.device)
+        _ext().gemm_sgd_pair(
+            a1, b1, bool(ta1), bool(tb1), p1.view(-1),
+            s1.view(-1) if s1 is not None else e,
+            a2, b2, bool(ta2), bool(tb2), p2.view(-1),
+            s2.view(-1) if s2 is not None else e,
+            float(lr), float(grad_scale), float(neg_decay))
+        return
+    def grad(a, b, ta, tb, p):
+        x = a.float().t() if ta else a.float()
+        y = b.float().t() if tb else b.float()
+        return (x @ y).reshape(p.shape)
+    g1 = grad(a1, b1, ta1, tb1, p1)
+    g2 = grad(a2, b2, ta2, tb2, p2)
+    fused_sgd(p1, g1, lr, grad_scale=grad_scale, neg_decay=neg_decay,
+              bf16_out=s1)
+    fused_sgd(p2, g2, lr, grad_scale=grad_scale, neg_decay=neg_decay,
+              bf16_out=s2)
+
+
 def colsum(x, out=None):
     """out[n] = sum_m x[m,n] in fp32 (bias gradients)."""
     if x.is_cuda:

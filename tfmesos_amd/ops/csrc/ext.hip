@@ -38,6 +38,11 @@ void launch_gemm_stripes(const bf16_t*, const bf16_t*, float*, int, int,
 void launch_gemm_small(const bf16_t*, const bf16_t*, const void*, bool,
                        void*, bool, int, void*, bool, int, int, int, int,
                        int, int, bool, hipStream_t);
+void launch_gemm_stripes_any(const bf16_t*, const bf16_t*, float*, int, int,
+                             int, int, int, int, int, int, bool, bool, int,
+                             int, hipStream_t);
+void launch_splitk_reduce_sgd(const float*, int, float*, bf16_t*, long,
+                              float, float, float, hipStream_t);
 void launch_gather_bf16(const bf16_t*, const long*, bf16_t*, long, int, long,
                         hipStream_t);
 void launch_gather_f32(const float*, const long*, float*, long, int, long,
@@ -311,6 +316,156 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
               vec_level(a.data_ptr(), a.size(1)),
               vec_level(b.data_ptr(), b.size(1)), cur_stream());
   return out;
+}
+
+// GEMM -> SGD fusion for the NMF factor updates: the gradient
+// G = op(A)@op(B) never materializes — the split-K stripes feed
+// p -= lr*(gscale*G + nd*min(p,0)) and the bf16 shadow refresh in the
+// phase-2 reduce kernel (saves the separate apply launch AND the
+// gradient round trip). Falls back to plain GEMM + fused_sgd when
+// split-K does not engage at this shape.
+void gemm_sgd(torch::Tensor a, torch::Tensor b, bool trans_a, bool trans_b,
+              torch::Tensor param, torch::Tensor shadow, double lr,
+              double grad_scale, double neg_decay) {
+  TORCH_CHECK(a.is_cuda() && b.is_cuda() && a.is_contiguous() &&
+              b.is_contiguous() && a.dim() == 2 && b.dim() == 2 &&
+              a.scalar_type() == torch::kBFloat16 &&
+              b.scalar_type() == torch::kBFloat16, "gemm_sgd: bf16 2-D");
+  const int M = trans_a ? a.size(1) : a.size(0);
+  const int Ka = trans_a ? a.size(0) : a.size(1);
+  const int Kb = trans_b ? b.size(1) : b.size(0);
+  const int N = trans_b ? b.size(0) : b.size(1);
+  TORCH_CHECK(Ka == Kb, "gemm_sgd: inner dims mismatch");
+  TORCH_CHECK(param.is_contiguous() &&
+              param.scalar_type() == torch::kFloat32 &&
+              param.numel() == (long)M * N, "param must be fp32 [M*N]");
+  bf16_t* shp = nullptr;
+  if (shadow.numel() > 0) {
+    TORCH_CHECK(shadow.is_contiguous() &&
+                shadow.scalar_type() == torch::kBFloat16 &&
+                shadow.numel() == (long)M * N, "shadow must be bf16 [M*N]");
+    shp = (bf16_t*)shadow.data_ptr();
+  }
+  const int nx = (N + 63) / 64, ny = (M + 63) / 64;
+  int nslice = 1, kc = 0;
+  if (nx * ny < 256 && Ka >= 256) {
+    int want = std::min((int)(Ka / 64), 256 / (nx * ny));
+    if (want > 16) want = 16;
+    if (want > 1) {
+      kc = ((Ka + want - 1) / want + 31) / 32 * 32;
+      nslice = (Ka + kc - 1) / kc;
+    }
+  }
+  if (nslice > 1) {
+    int* cnt_unused;
+    float* ws = splitk_ws(a.device(), (long)M * N * nslice,
+                          (long)nx * ny, &cnt_unused);
+    launch_gemm_stripes_any((const bf16_t*)a.data_ptr(),
+                            (const bf16_t*)b.data_ptr(), ws, kc, nslice,
+                            M, N, Ka, a.size(1), b.size(1), N, trans_a,
+                            trans_b, vec_level(a.data_ptr(), a.size(1)),
+                            vec_level(b.data_ptr(), b.size(1)),
+                            cur_stream());
+    launch_splitk_reduce_sgd(ws, nslice, param.data_ptr<float>(), shp,
+                             (long)M * N, (float)lr, (float)grad_scale,
+                             (float)neg_decay, cur_stream());
+    return;
+  }
+  // no split-K at this shape: grad to a temp, then the fused apply
+  auto gtmp = torch::empty({M, N}, a.options().dtype(torch::kFloat32));
+  launch_gemm((const bf16_t*)a.data_ptr(), (const bf16_t*)b.data_ptr(),
+              nullptr, false, gtmp.data_ptr(), true, nullptr, nullptr,
+              nullptr, nullptr, 0, 1, M, N, Ka, a.size(1), b.size(1), N,
+              trans_a, trans_b, 0, vec_level(a.data_ptr(), a.size(1)),
+              vec_level(b.data_ptr(), b.size(1)), cur_stream());
+  launch_sgd(param.data_ptr<float>(), gtmp.data_ptr(), false, nullptr, shp,
+             (long)M * N, (float)lr, 0.f, 0.f, (float)grad_scale,
+             (float)neg_decay, cur_stream());
+}
+
+// Paired form preserving SIMULTANEOUS-update semantics (the reference
+// applies all gradients of a step at once): BOTH factors' stripe
+// phases launch before EITHER fused apply, reading only pre-update
+// values. One shared workspace allocation covers both.
+void gemm_sgd_pair(torch::Tensor a1, torch::Tensor b1, bool ta1, bool tb1,
+                   torch::Tensor p1, torch::Tensor s1,
+                   torch::Tensor a2, torch::Tensor b2, bool ta2, bool tb2,
+                   torch::Tensor p2, torch::Tensor s2,
+                   double lr, double grad_scale, double neg_decay) {
+  struct Half {
+    torch::Tensor a, b, p, s;
+    bool ta, tb;
+    int M, N, K, nslice, kc;
+    long ws_off;
+    torch::Tensor gtmp;
+  } h[2] = {{a1, b1, p1, s1, ta1, tb1}, {a2, b2, p2, s2, ta2, tb2}};
+  long ws_total = 0;
+  for (int i = 0; i < 2; ++i) {
+    auto& x = h[i];
+    x.M = x.ta ? x.a.size(1) : x.a.size(0);
+    x.K = x.ta ? x.a.size(0) : x.a.size(1);
+    x.N = x.tb ? x.b.size(0) : x.b.size(1);
+    TORCH_CHECK(x.p.is_contiguous() &&
+                x.p.scalar_type() == torch::kFloat32 &&
+                x.p.numel() == (long)x.M * x.N, "gemm_sgd_pair: bad param");
+    const int nx = (x.N + 63) / 64, ny = (x.M + 63) / 64;
+    x.nslice = 1;
+    x.kc = 0;
+    if (nx * ny < 256 && x.K >= 256) {
+      int want = std::min((int)(x.K / 64), 256 / (nx * ny));
+      if (want > 16) want = 16;
+      if (want > 1) {
+        x.kc = ((x.K + want - 1) / want + 31) / 32 * 32;
+        x.nslice = (x.K + x.kc - 1) / x.kc;
+      }
+    }
+    x.ws_off = ws_total;
+    if (x.nslice > 1) ws_total += (long)x.M * x.N * x.nslice;
+  }
+  torch::Tensor ws;
+  if (ws_total > 0)
+    ws = torch::empty({ws_total}, a1.options().dtype(torch::kFloat32));
+  // phase 1 for BOTH halves (reads only pre-update operands)
+  for (int i = 0; i < 2; ++i) {
+    auto& x = h[i];
+    if (x.nslice > 1) {
+      launch_gemm_stripes_any((const bf16_t*)x.a.data_ptr(),
+                              (const bf16_t*)x.b.data_ptr(),
+                              ws.data_ptr<float>() + x.ws_off, x.kc,
+                              x.nslice, x.M, x.N, x.K, x.a.size(1),
+                              x.b.size(1), x.N,
+                              x.ta, x.tb,
+                              vec_level(x.a.data_ptr(), x.a.size(1)),
+                              vec_level(x.b.data_ptr(), x.b.size(1)),
+                              cur_stream());
+    } else {
+      x.gtmp = torch::empty({x.M, x.N},
+                            a1.options().dtype(torch::kFloat32));
+      launch_gemm((const bf16_t*)x.a.data_ptr(),
+                  (const bf16_t*)x.b.data_ptr(), nullptr, false,
+                  x.gtmp.data_ptr(), true, nullptr, nullptr, nullptr,
+                  nullptr, 0, 1, x.M, x.N, x.K, x.a.size(1), x.b.size(1),
+                  x.N, x.ta, x.tb, 0,
+                  vec_level(x.a.data_ptr(), x.a.size(1)),
+                  vec_level(x.b.data_ptr(), x.b.size(1)), cur_stream());
+    }
+  }
+  // phase 2: fused applies
+  for (int i = 0; i < 2; ++i) {
+    auto& x = h[i];
+    bf16_t* shp = x.s.numel() ? (bf16_t*)x.s.data_ptr() : nullptr;
+    if (x.nslice > 1) {
+      launch_splitk_reduce_sgd(ws.data_ptr<float>() + x.ws_off, x.nslice,
+                               x.p.data_ptr<float>(), shp,
+                               (long)x.M * x.N, (float)lr,
+                               (float)grad_scale, (float)neg_decay,
+                               cur_stream());
+    } else {
+      launch_sgd(x.p.data_ptr<float>(), x.gtmp.data_ptr(), false, nullptr,
+                 shp, (long)x.M * x.N, (float)lr, 0.f, 0.f,
+                 (float)grad_scale, (float)neg_decay, cur_stream());
+    }
+  }
 }
 
 torch::Tensor gemm_bias_act(torch::Tensor a, torch::Tensor b,
@@ -1105,6 +1260,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_group_fwd", &bn_group_fwd);
   m.def("bn_group_bwd", &bn_group_bwd);
   m.def("bn_group_fwd_multi", &bn_group_fwd_multi);
+  m.def("gemm_sgd", &gemm_sgd);
+  m.def("gemm_sgd_pair", &gemm_sgd_pair);
   m.def("bn_group_bwd_multi", &bn_group_bwd_multi);
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
   m.def("bn_bwd", &bn_bwd, "fused batch-norm (+relu mask) bwd");

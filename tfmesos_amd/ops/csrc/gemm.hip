@@ -475,6 +475,42 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
   }
 }
 
+
+// split-K phase 2 fused with the SGD apply (the NMF factor update):
+// instead of materializing the fp32 gradient and re-reading it in a
+// separate apply kernel, the stripe sum feeds p -= lr*(gscale*g +
+// nd*min(p,0)) and the bf16 shadow refresh directly.
+__global__ __launch_bounds__(256)
+void splitk_reduce_sgd_kernel(const float* __restrict__ ws, int nslice,
+                              float* __restrict__ p,
+                              __bf16* __restrict__ shadow, long mn,
+                              float lr, float gscale, float nd) {
+  const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (i0 >= mn) return;
+  f32x4 v = {};
+  for (int z = 0; z < nslice; ++z) {
+    const float* sp = ws + (long)z * mn + i0;
+    if (i0 + 4 <= mn) {
+      const f32x4 sv = *(const f32x4*)sp;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] += sv[j];
+    } else {
+      for (int j = 0; i0 + j < mn; ++j) v[j] += sp[j];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const long i = i0 + j;
+    if (i >= mn) break;
+    float pv = p[i];
+    float g = gscale * v[j];
+    if (nd != 0.f) g += nd * fminf(pv, 0.f);
+    pv -= lr * g;
+    p[i] = pv;
+    if (shadow != nullptr) shadow[i] = (__bf16)pv;
+  }
+}
+
 }  // namespace
 
 void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
@@ -594,6 +630,35 @@ void launch_gemm_small(const bf16_t* A, const bf16_t* B, const void* bias,
   if (ta) { if (colsum_out) GS(true, true); else GS(true, false); }
   else    { if (colsum_out) GS(false, true); else GS(false, false); }
 #undef GS
+}
+
+
+// phase-1-only split-K stripes for ANY transpose combo (the fused
+// GEMM->SGD path reduces them with splitk_reduce_sgd_kernel)
+void launch_gemm_stripes_any(const bf16_t* A, const bf16_t* B, float* ws,
+                             int kc, int nslice, int M, int N, int K,
+                             int lda, int ldb, int ldc, bool ta, bool tb,
+                             int veca, int vecb, hipStream_t stream) {
+  const long Mtiles = ((long)M + BM - 1) / BM;
+  dim3 grid((unsigned)ceil_div(N, BN), (unsigned)Mtiles, nslice);
+  dim3 block(256);
+#define SLAUNCH(TAv, TBv)                                                   \
+  hipLaunchKernelGGL((gemm_kernel<TAv, TBv, 0, false, false, true, false>), \
+                     grid, block, 0, stream, (const __bf16*)A,              \
+                     (const __bf16*)B, nullptr, false, nullptr, nullptr,    \
+                     nullptr, ws, nullptr, M, N, K, lda, ldb, ldc, kc, veca,\
+                     vecb)
+  if (ta) { if (tb) SLAUNCH(true, true); else SLAUNCH(true, false); }
+  else    { if (tb) SLAUNCH(false, true); else SLAUNCH(false, false); }
+#undef SLAUNCH
+}
+
+void launch_splitk_reduce_sgd(const float* ws, int nslice, float* p,
+                              bf16_t* shadow, long mn, float lr,
+                              float gscale, float nd, hipStream_t stream) {
+  dim3 rgrid((unsigned)(((mn + 3) / 4 + 255) / 256)), rblock(256);
+  hipLaunchKernelGGL(splitk_reduce_sgd_kernel, rgrid, rblock, 0, stream,
+                     ws, nslice, p, (__bf16*)shadow, mn, lr, gscale, nd);
 }
 
 void launch_gemm_stripes(const bf16_t* A, const bf16_t* B, float* ws,
