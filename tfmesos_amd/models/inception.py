@@ -190,17 +190,20 @@ class InceptionA(nn.Module):
         self.pf = pool_features
 
     def forward(self, x):
+        # explicit fan-out: ONE n-way add sums the branch dx's in
+        # backward (autograd would chain n-1 pairwise adds)
+        xa, xb, xc, xd = ops.fan_out(x, 4)
         if x.is_cuda:
             y5, y3 = _bn_multi([self.b5x5_1, self.b3x3_1],
-                               [self.b5x5_1.conv(x), self.b3x3_1.conv(x)])
+                               [self.b5x5_1.conv(xb), self.b3x3_1.conv(xc)])
         else:
-            y5, y3 = self.b5x5_1(x), self.b3x3_1(x)
+            y5, y3 = self.b5x5_1(xb), self.b3x3_1(xc)
         return _group_cat(x, [
-            (64, self.b1x1, lambda: self.b1x1.conv(x)),
+            (64, self.b1x1, lambda: self.b1x1.conv(xa)),
             (64, self.b5x5_2, lambda: self.b5x5_2.conv(y5)),
             (96, self.b3x3_3,
              lambda: self.b3x3_3.conv(self.b3x3_2(y3))),
-            (self.pf, self.bpool, lambda: self.bpool.conv(_avg_pool(x, 3))),
+            (self.pf, self.bpool, lambda: self.bpool.conv(_avg_pool(xd, 3))),
         ])
 
 
@@ -214,12 +217,13 @@ class InceptionB(nn.Module):
 
     def forward(self, x):
         cin, h, w = x.shape[1], x.shape[2], x.shape[3]
+        xa, xb, xc = ops.fan_out(x, 3)
         return _group_cat(x, [
-            (384, self.b3x3, lambda: self.b3x3.conv(x)),
+            (384, self.b3x3, lambda: self.b3x3.conv(xa)),
             (96, self.b3x3dbl_3, lambda: self.b3x3dbl_3.conv(
-                self.b3x3dbl_2(self.b3x3dbl_1(x)))),
+                self.b3x3dbl_2(self.b3x3dbl_1(xb)))),
         ], hw=((h - 3) // 2 + 1, (w - 3) // 2 + 1),
-            pool=(cin, lambda o: _max_pool(x, 3, 2, out=o)))
+            pool=(cin, lambda o: _max_pool(xc, 3, 2, out=o)))
 
 
 class InceptionC(nn.Module):
@@ -237,21 +241,22 @@ class InceptionC(nn.Module):
         self.bpool = BasicConv2d(cin, 192, kernel_size=1)
 
     def forward(self, x):
+        xa, xb, xc, xd = ops.fan_out(x, 4)
         if x.is_cuda:
             s1a, s1b = _bn_multi([self.b7_1, self.b7d_1],
-                                 [self.b7_1.conv(x), self.b7d_1.conv(x)])
+                                 [self.b7_1.conv(xb), self.b7d_1.conv(xc)])
             s2a, s2b = _bn_multi(
                 [self.b7_2, self.b7d_2],
                 [self.b7_2.conv(s1a), self.b7d_2.conv(s1b)])
         else:
-            s2a = self.b7_2(self.b7_1(x))
-            s2b = self.b7d_2(self.b7d_1(x))
+            s2a = self.b7_2(self.b7_1(xb))
+            s2b = self.b7d_2(self.b7d_1(xc))
         return _group_cat(x, [
-            (192, self.b1x1, lambda: self.b1x1.conv(x)),
+            (192, self.b1x1, lambda: self.b1x1.conv(xa)),
             (192, self.b7_3, lambda: self.b7_3.conv(s2a)),
             (192, self.b7d_5, lambda: self.b7d_5.conv(self.b7d_4(
                 self.b7d_3(s2b)))),
-            (192, self.bpool, lambda: self.bpool.conv(_avg_pool(x, 3))),
+            (192, self.bpool, lambda: self.bpool.conv(_avg_pool(xd, 3))),
         ])
 
 
@@ -267,17 +272,18 @@ class InceptionD(nn.Module):
 
     def forward(self, x):
         cin, h, w = x.shape[1], x.shape[2], x.shape[3]
+        xa, xb, xc = ops.fan_out(x, 3)
         if x.is_cuda:
             y3, y7 = _bn_multi([self.b3_1, self.b7_1],
-                               [self.b3_1.conv(x), self.b7_1.conv(x)])
+                               [self.b3_1.conv(xa), self.b7_1.conv(xb)])
         else:
-            y3, y7 = self.b3_1(x), self.b7_1(x)
+            y3, y7 = self.b3_1(xa), self.b7_1(xb)
         return _group_cat(x, [
             (320, self.b3_2, lambda: self.b3_2.conv(y3)),
             (192, self.b7_4, lambda: self.b7_4.conv(
                 self.b7_3(self.b7_2(y7)))),
         ], hw=((h - 3) // 2 + 1, (w - 3) // 2 + 1),
-            pool=(cin, lambda o: _max_pool(x, 3, 2, out=o)))
+            pool=(cin, lambda o: _max_pool(xc, 3, 2, out=o)))
 
 
 class InceptionE(nn.Module):
@@ -294,22 +300,25 @@ class InceptionE(nn.Module):
         self.bpool = BasicConv2d(cin, 192, kernel_size=1)
 
     def forward(self, x):
+        xa, xb, xc, xd = ops.fan_out(x, 4)
         if x.is_cuda:
             b3, y3d = _bn_multi([self.b3_1, self.b3d_1],
-                                [self.b3_1.conv(x), self.b3d_1.conv(x)])
+                                [self.b3_1.conv(xb), self.b3d_1.conv(xc)])
             b3d = self.b3d_2(y3d)
         else:
-            b3 = self.b3_1(x)
-            b3d = self.b3d_2(self.b3d_1(x))
+            b3 = self.b3_1(xb)
+            b3d = self.b3d_2(self.b3d_1(xc))
+        b3a, b3b = ops.fan_out(b3, 2)
+        b3da, b3db = ops.fan_out(b3d, 2)
         # the nested cats flatten: sub-branch slices are adjacent, so
         # one buffer (and ONE grouped BN) serves the whole block
         return _group_cat(x, [
-            (320, self.b1x1, lambda: self.b1x1.conv(x)),
-            (384, self.b3_2a, lambda: self.b3_2a.conv(b3)),
-            (384, self.b3_2b, lambda: self.b3_2b.conv(b3)),
-            (384, self.b3d_3a, lambda: self.b3d_3a.conv(b3d)),
-            (384, self.b3d_3b, lambda: self.b3d_3b.conv(b3d)),
-            (192, self.bpool, lambda: self.bpool.conv(_avg_pool(x, 3))),
+            (320, self.b1x1, lambda: self.b1x1.conv(xa)),
+            (384, self.b3_2a, lambda: self.b3_2a.conv(b3a)),
+            (384, self.b3_2b, lambda: self.b3_2b.conv(b3b)),
+            (384, self.b3d_3a, lambda: self.b3d_3a.conv(b3da)),
+            (384, self.b3d_3b, lambda: self.b3d_3b.conv(b3db)),
+            (192, self.bpool, lambda: self.bpool.conv(_avg_pool(xd, 3))),
         ])
 
 

@@ -53,6 +53,7 @@ void launch_scatter_add_f32(float*, const long*, const float*, long, int,
                             long, hipStream_t);
 void launch_relu_bwd(const bf16_t*, const bf16_t*, bf16_t*, long,
                      hipStream_t);
+void launch_add_n(const bf16_t* const*, int, bf16_t*, long, hipStream_t);
 void launch_colsum(const bf16_t*, float*, int, int, hipStream_t);
 void launch_conv_fwd(const bf16_t*, const bf16_t*, const float*, bf16_t*,
                      float*, int, int, int, int, int, int, int, int, int,
@@ -466,6 +467,22 @@ void gemm_sgd_pair(torch::Tensor a1, torch::Tensor b1, bool ta1, bool tb1,
                  (float)grad_scale, (float)neg_decay, cur_stream());
     }
   }
+}
+
+torch::Tensor add_n(std::vector<torch::Tensor> xs) {
+  const int n = (int)xs.size();
+  TORCH_CHECK(n >= 1 && n <= 6, "add_n: 1..6 tensors");
+  const long total = xs[0].numel();
+  const bf16_t* srcs[6];
+  for (int i = 0; i < n; ++i) {
+    TORCH_CHECK(xs[i].is_cuda() && xs[i].is_contiguous() &&
+                xs[i].scalar_type() == torch::kBFloat16 &&
+                xs[i].numel() == total, "add_n: same-shape bf16");
+    srcs[i] = (const bf16_t*)xs[i].data_ptr();
+  }
+  auto out = torch::empty_like(xs[0]);
+  launch_add_n(srcs, n, (bf16_t*)out.data_ptr(), total, cur_stream());
+  return out;
 }
 
 torch::Tensor gemm_bias_act(torch::Tensor a, torch::Tensor b,
@@ -1262,6 +1279,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_group_fwd_multi", &bn_group_fwd_multi);
   m.def("gemm_sgd", &gemm_sgd);
   m.def("gemm_sgd_pair", &gemm_sgd_pair);
+  m.def("add_n", &add_n);
   m.def("bn_group_bwd_multi", &bn_group_bwd_multi);
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
   m.def("bn_bwd", &bn_bwd, "fused batch-norm (+relu mask) bwd");
