@@ -475,9 +475,13 @@ torch::Tensor add_n(std::vector<torch::Tensor> xs) {
   const long total = xs[0].numel();
   const bf16_t* srcs[6];
   for (int i = 0; i < n; ++i) {
-    TORCH_CHECK(xs[i].is_cuda() && xs[i].is_contiguous() &&
+    // dense in EITHER memory format (the Python wrapper makes all
+    // inputs the same layout, so flat elementwise math is valid)
+    TORCH_CHECK(xs[i].is_cuda() &&
+                (xs[i].is_contiguous() ||
+                 xs[i].is_contiguous(torch::MemoryFormat::ChannelsLast)) &&
                 xs[i].scalar_type() == torch::kBFloat16 &&
-                xs[i].numel() == total, "add_n: same-shape bf16");
+                xs[i].numel() == total, "add_n: same-shape dense bf16");
     srcs[i] = (const bf16_t*)xs[i].data_ptr();
   }
   auto out = torch::empty_like(xs[0]);
