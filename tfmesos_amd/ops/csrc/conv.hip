@@ -22,6 +22,8 @@
 //           native [k][r][s][c] layout by a transposing stager)
 //   bwd-w:  dW[k,r,s,c]   = sum_{n,ho,wo} dY[n,ho,wo,k] X[n,ho*U+r-P,...,c]
 //           GEMM  M = K,       Ncol = R*S*C, Kdim = N*Ho*Wo
+#include <cstdlib>
+
 #include "common.h"
 
 namespace {
@@ -757,6 +759,177 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
   }
 }
 
+// --------------------------------------------- bwd-weight, transpose-read
+// Pixel-major LDS image + ds_read_b64_tr_b16 fragment reads.
+//
+// The row-major (k-major) image above forces the staging to transpose
+// somewhere: DyBwdwStage pays 8 strided b16 GLOBAL loads per thread per
+// tile (k-coalesced but 8 separate instructions), XBwdwStage pays 8
+// sub-dword LDS scatter writes. Pixel-major storage makes BOTH stagers
+// a single b128 global load + a single b128 LDS write per thread per
+// tile, and the MFMA A/B fragments (8 reduction pixels for one column)
+// come out of the plain [pixel][column] image via two transpose-reads
+// each (tr reads cost no extra issue slots beside MFMAs —
+// MI355X_MICROARCH.md filler table).
+//
+// gfx950 ds_read_b64_tr_b16 semantics (measured, tools/trprobe.hip —
+// all four probe patterns fit): within each aligned 16-lane group,
+// provider lane x reads 8 B (4 bf16) at its own 8 B-aligned address
+// forming a 16x4 element matrix M, and consumer lane l receives
+// elem j = M[4*j + ((l>>2)&3)][l&3]. Feeding MFMA lane l with pixels
+// ko..ko+3 of its column m = 32*wr + (l&31) therefore takes provider
+// addresses  addr(l) = (ko + ((l&15)>>2))*TR_L + 32*wr +
+// 16*((l>>4)&1) + 4*(l&3)  — four consecutive image ROWS per group,
+// and a second read at +4*TR_L covers pixels ko+4..ko+7.
+//
+// Row stride TR_L = 96 elems (192 B ≡ 64 banks*3 mod 256 B): the four
+// rows a 32-lane service group touches land on bank quartets 0-15 /
+// 48-63 / 32-47 / 16-31 — a partition, so both tr reads and the b128
+// staging writes are conflict-free.
+constexpr int TR_L = 96;
+constexpr int TR_ELEMS = BK * TR_L;
+
+DEVINL int trindex(int p, int c) { return p * TR_L + c; }
+
+DEVINL bf16x4 tr_read(const __bf16* p) {
+  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) bf16x4*)p);
+}
+
+// As[p][k]: dY[n,ho,wo,k] — thread t owns ONE pixel (t>>3) and an
+// 8-channel chunk ((t&7)*8): the global load is one b128 (8 consecutive
+// k of one dY row) and the LDS store is one b128 into the pixel-major
+// image. Compare DyBwdwStage: same bytes, 8 strided loads + transposed
+// store.
+struct DyBwdwTrStage {
+  int so;              // LDS elem offset (fixed per thread)
+  int k0;
+  bool vec;
+  long p;
+  const __bf16* src;
+  bf16x8 v;
+
+  DEVINL void init(const __bf16* __restrict__ dY, const ConvShape& cs,
+                   int m0, long p0, int t, bool kvec) {
+    const int px = t >> 3, kc0 = (t & 7) * 8;
+    so = trindex(px, kc0);
+    k0 = m0 + kc0;
+    vec = kvec && (k0 + 8 <= cs.K);
+    p = p0 + px;
+    src = dY + p * cs.LDY + k0;
+  }
+
+  DEVINL void load(const ConvShape cs, long Ptot) {
+    v = {};
+    if (p < Ptot) {
+      if (vec) {
+        v = *(const bf16x8*)src;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (k0 + j < cs.K) v[j] = src[j];
+      }
+    }
+    p += BK;
+    src += (long)BK * cs.LDY;
+  }
+
+  DEVINL void commit(__bf16* Sm) { *(bf16x8*)&Sm[so] = v; }
+};
+
+// Bs[p][q]: X patches — identical thread mapping and load path to
+// XBwdwStage (one pixel, 8 consecutive taps), but the store is one
+// b128 into the pixel-major image instead of 8 b16 scatters.
+struct XBwdwTrStage : XBwdwStage {
+  int so;
+  DEVINL void init_tr(const ConvShape& cs, int n0, long p0, int KD, int t,
+                      bool cvec) {
+    init(cs, n0, p0, KD, t, cvec);
+    so = trindex(px, qq0);
+  }
+  DEVINL void commit_tr(__bf16* Sn) { *(bf16x8*)&Sn[so] = vv; }
+};
+
+__global__ __launch_bounds__(256)
+void conv_bwdw_kernel_tr(const __bf16* __restrict__ dY,
+                         const __bf16* __restrict__ X,
+                         float* __restrict__ dW, ConvShape cs, long pc) {
+  __shared__ __align__(16) __bf16 As[2][TR_ELEMS];
+  __shared__ __align__(16) __bf16 Bs[2][TR_ELEMS];
+  const int KD = cs.R * cs.S * cs.C;
+  const long Ptot = (long)cs.N * cs.Ho * cs.Wo;
+  const bool cvec = (cs.C & 7) == 0;
+  // b128 dY loads need the 8-k chunk 16 B aligned: base + p*LDY + k0
+  const bool kvec = ((cs.K & 7) == 0) && ((cs.LDY & 7) == 0) &&
+                    ((((unsigned long long)dY) & 15ULL) == 0);
+  const int tm0 = blockIdx.y * BM;
+  const int tn0 = blockIdx.x * BN;
+  const long ps = (long)blockIdx.z * pc;
+  const long pe = min(ps + pc, Ptot);
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+
+  f32x16 acc = {};
+  DyBwdwTrStage dst;
+  XBwdwTrStage xst;
+  dst.init(dY, cs, tm0, ps, t, kvec);
+  xst.init_tr(cs, tn0, ps, KD, t, cvec);
+  dst.load(cs, Ptot);
+  xst.load(X, cs, KD, Ptot);
+  dst.commit(As[0]);
+  xst.commit_tr(Bs[0]);
+  __syncthreads();
+
+  // provider-lane fragment addresses (fixed per lane; see header
+  // comment): row = pixel within the tile, column = this lane group's
+  // 4-element slice of the output column block
+  const int arow0 = ((lane & 15) >> 2) + (lane >> 5) * 8;
+  const int acoff = 16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+  const int abase = arow0 * TR_L + wr * 32 + acoff;
+  const int bbase = arow0 * TR_L + wc * 32 + acoff;
+
+  int cur = 0;
+  for (long p0 = ps; p0 < pe; p0 += BK, cur ^= 1) {
+    const bool more = p0 + BK < pe;
+    if (more) {
+      dst.load(cs, Ptot);
+      xst.load(X, cs, KD, Ptot);
+    }
+#pragma unroll
+    for (int kh = 0; kh < 2; ++kh) {
+      const int tp = kh * 16 * TR_L;
+      bf16x4 a0 = tr_read(&As[cur][abase + tp]);
+      bf16x4 a1 = tr_read(&As[cur][abase + tp + 4 * TR_L]);
+      bf16x4 b0 = tr_read(&Bs[cur][bbase + tp]);
+      bf16x4 b1 = tr_read(&Bs[cur][bbase + tp + 4 * TR_L]);
+      bf16x8 a = __builtin_shufflevector(a0, a1, 0, 1, 2, 3, 4, 5, 6, 7);
+      bf16x8 b = __builtin_shufflevector(b0, b1, 0, 1, 2, 3, 4, 5, 6, 7);
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+    }
+    if (more) {
+      dst.commit(As[cur ^ 1]);
+      xst.commit_tr(Bs[cur ^ 1]);
+    }
+    __syncthreads();
+  }
+
+  const int q = tn0 + wc * 32 + (lane & 31);
+  if (q < KD) {
+#pragma unroll
+    for (int v = 0; v < 16; ++v) {
+      const int k = tm0 + wr * 32 + ((v >> 2) << 3) +
+                    ((lane >> 5) << 2) + (v & 3);
+      if (k >= cs.K) continue;
+      if (gridDim.z == 1)
+        dW[(long)k * KD + q] = acc[v];
+      else
+        unsafeAtomicAdd(&dW[(long)k * KD + q], acc[v]);
+    }
+  }
+}
+
 }  // namespace
 
 int conv_fwd_slices(int N, int K, int Ho, int Wo, int C, int R, int S) {
@@ -899,6 +1072,19 @@ void launch_conv_bwd_weight(const bf16_t* dY, long ldy, const bf16_t* X,
   z = (int)((Ptot + pc - 1) / pc);
   dim3 grid(ceil_div(KD, BN), ceil_div(K, BM), z);
   dim3 block(256);
+  // TFA_BWDW_TR=0 falls back to the k-major staging kernel (A/B
+  // comparison; the transpose-read variant measured faster across the
+  // Inception shapes — docs/KERNELS.md round-2 log)
+  static int tr_mode = -1;
+  if (tr_mode < 0) {
+    const char* e = getenv("TFA_BWDW_TR");
+    tr_mode = e ? atoi(e) : 1;
+  }
+  if (tr_mode) {
+    hipLaunchKernelGGL(conv_bwdw_kernel_tr, grid, block, 0, stream,
+                       (const __bf16*)dY, (const __bf16*)X, dW, cs, pc);
+    return;
+  }
   hipLaunchKernelGGL(conv_bwdw_kernel, grid, block, 0, stream,
                      (const __bf16*)dY, (const __bf16*)X, dW, ws, cs, pc);
 }
