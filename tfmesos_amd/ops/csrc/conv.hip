@@ -850,7 +850,8 @@ struct XBwdwTrStage : XBwdwStage {
   DEVINL void commit_tr(__bf16* Sn) { *(bf16x8*)&Sn[so] = vv; }
 };
 
-__global__ __launch_bounds__(256)
+template <int MINB>
+__global__ __launch_bounds__(256, MINB)
 void conv_bwdw_kernel_tr(const __bf16* __restrict__ dY,
                          const __bf16* __restrict__ X,
                          float* __restrict__ dW, ConvShape cs, long pc) {
@@ -1080,8 +1081,13 @@ void launch_conv_bwd_weight(const bf16_t* dY, long ldy, const bf16_t* X,
     const char* e = getenv("TFA_BWDW_TR");
     tr_mode = e ? atoi(e) : 1;
   }
+  if (tr_mode == 2) {
+    hipLaunchKernelGGL((conv_bwdw_kernel_tr<5>), grid, block, 0, stream,
+                       (const __bf16*)dY, (const __bf16*)X, dW, cs, pc);
+    return;
+  }
   if (tr_mode) {
-    hipLaunchKernelGGL(conv_bwdw_kernel_tr, grid, block, 0, stream,
+    hipLaunchKernelGGL((conv_bwdw_kernel_tr<1>), grid, block, 0, stream,
                        (const __bf16*)dY, (const __bf16*)X, dW, cs, pc);
     return;
   }
