@@ -23,6 +23,7 @@
 //   bwd-w:  dW[k,r,s,c]   = sum_{n,ho,wo} dY[n,ho,wo,k] X[n,ho*U+r-P,...,c]
 //           GEMM  M = K,       Ncol = R*S*C, Kdim = N*Ho*Wo
 #include <cstdlib>
+#include <type_traits>
 
 #include "common.h"
 
@@ -430,6 +431,30 @@ DEVINL void BwddPatchStage::load(const __bf16* __restrict__ dY,
 // Kills the host-side W^T permute+copy the backward used to pay per
 // layer per step (~94 launch-bound copies across Inception).
 // Split load/commit like the patch stagers.
+// gfx950 ds_read_b64_tr_b16 semantics (measured, tools/trprobe.hip —
+// all four probe patterns fit): within each aligned 16-lane group,
+// provider lane x reads 8 B (4 bf16) at its own 8 B-aligned address
+// forming a 16x4 element matrix M, and consumer lane l receives
+// elem j = M[4*j + ((l>>2)&3)][l&3]. Feeding MFMA lane l with
+// reduction rows ko..ko+3 of its column m = 32*wr + (l&31) therefore
+// takes provider addresses  addr(l) = (ko + ((l&15)>>2))*TR_L +
+// 32*wr + 16*((l>>4)&1) + 4*(l&3)  — four consecutive image ROWS per
+// group, and a second read at +4*TR_L covers rows ko+4..ko+7.
+//
+// Row stride TR_L = 96 elems (192 B ≡ 64 banks*3 mod 256 B): the four
+// rows a 32-lane service group touches land on bank quartets 0-15 /
+// 48-63 / 32-47 / 16-31 — a partition, so both tr reads and the b128
+// staging writes are conflict-free.
+constexpr int TR_L = 96;
+constexpr int TR_ELEMS = BK * TR_L;
+
+DEVINL int trindex(int p, int c) { return p * TR_L + c; }
+
+DEVINL bf16x4 tr_read(const __bf16* p) {
+  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) bf16x4*)p);
+}
+
 struct WkrscStage {
   int px, kk0;
   bf16x8 vv;
@@ -465,14 +490,26 @@ struct WkrscStage {
   }
 };
 
-template <int STRIDE, bool SK>
+// Reduction-major variant of WkrscStage: one b128 LDS store into the
+// [tap][c] image (WkrscStage pays 8 sub-dword scatter writes to
+// transpose); the MFMA B fragments come back via ds_read_b64_tr_b16.
+struct WkrscTrStage : WkrscStage {
+  int so;
+  DEVINL void init(int t) {
+    WkrscStage::init(t);
+    so = trindex(px, kk0);
+  }
+  DEVINL void commit(__bf16* Sn) { *(bf16x8*)&Sn[so] = vv; }
+};
+
+template <int STRIDE, bool SK, bool TRB>
 __global__ __launch_bounds__(256)
 void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ Wt,
                       __bf16* __restrict__ dX, float* __restrict__ ws, int kc,
                       ConvShape cs) {
   // Wt memory: NATIVE [K][R*S*C] (the forward weight tensor)
   __shared__ __align__(16) __bf16 As[2][TILE_ELEMS];
-  __shared__ __align__(16) __bf16 Bs[2][TILE_ELEMS];
+  __shared__ __align__(16) __bf16 Bs[2][TRB ? TR_ELEMS : TILE_ELEMS];
   const long M = (long)cs.N * cs.H * cs.W;
   const int KD = cs.R * cs.S * cs.K;
   const bool kvec = (cs.K & 7) == 0;
@@ -491,9 +528,12 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
   // efficiency of the earlier 2x2 16x16x32 fragment scheme here).
   f32x16 acc = {};
   BwddPatchStage pst;
-  WkrscStage wst;
+  typename std::conditional<TRB, WkrscTrStage, WkrscStage>::type wst;
   pst.init(cs, tm0, M, t);
   wst.init(t);
+  // provider-lane fragment address for the tr image (unused when !TRB)
+  const int bbase = (((lane & 15) >> 2) + (lane >> 5) * 8) * TR_L +
+                    wc * 32 + 16 * ((lane >> 4) & 1) + 4 * (lane & 3);
   pst.load<STRIDE>(dY, cs, ks, KD, kvec);
   wst.load(Wt, tn0, ks, cs, KD);
   pst.commit(As[0]);
@@ -510,7 +550,14 @@ void conv_bwdd_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
     for (int kh = 0; kh < 2; ++kh) {
       const int ko = kh * 16 + ((lane >> 5) << 3);
       bf16x8 a = *(const bf16x8*)&srow(As[cur], wr * 32 + (lane & 31))[ko];
-      bf16x8 b = *(const bf16x8*)&srow(Bs[cur], wc * 32 + (lane & 31))[ko];
+      bf16x8 b;
+      if (TRB) {
+        bf16x4 b0 = tr_read(&Bs[cur][bbase + kh * 16 * TR_L]);
+        bf16x4 b1 = tr_read(&Bs[cur][bbase + kh * 16 * TR_L + 4 * TR_L]);
+        b = __builtin_shufflevector(b0, b1, 0, 1, 2, 3, 4, 5, 6, 7);
+      } else {
+        b = *(const bf16x8*)&srow(Bs[cur], wc * 32 + (lane & 31))[ko];
+      }
       acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
     }
     if (more) {
@@ -786,16 +833,6 @@ void conv_bwdw_kernel(const __bf16* __restrict__ dY, const __bf16* __restrict__ 
 // rows a 32-lane service group touches land on bank quartets 0-15 /
 // 48-63 / 32-47 / 16-31 — a partition, so both tr reads and the b128
 // staging writes are conflict-free.
-constexpr int TR_L = 96;
-constexpr int TR_ELEMS = BK * TR_L;
-
-DEVINL int trindex(int p, int c) { return p * TR_L + c; }
-
-DEVINL bf16x4 tr_read(const __bf16* p) {
-  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-      (__attribute__((address_space(3))) bf16x4*)p);
-}
-
 // As[p][k]: dY[n,ho,wo,k] — thread t owns ONE pixel (t>>3) and an
 // 8-channel chunk ((t&7)*8): the global load is one b128 (8 consecutive
 // k of one dY row) and the LDS store is one b128 into the pixel-major
@@ -998,6 +1035,15 @@ void launch_conv_fwd(const bf16_t* X, const bf16_t* W, const float* bias,
 #undef FL
 }
 
+static int bwdd_tr() {
+  static int m = -1;
+  if (m < 0) {
+    const char* e = getenv("TFA_BWDD_TR");
+    m = e ? atoi(e) : 1;
+  }
+  return m;
+}
+
 void launch_conv_bwd_data(const bf16_t* dY, long ldy, const bf16_t* Wt,
                           bf16_t* dX,
                           float* ws, int N, int C, int H, int Wd, int K,
@@ -1016,9 +1062,15 @@ void launch_conv_bwd_data(const bf16_t* dY, long ldy, const bf16_t* Wt,
     const int zr = ceil_div(KD, kc);
     dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(C, BN), zr);
 #define BL(STRIDEv)                                                         \
-    hipLaunchKernelGGL((conv_bwdd_kernel<STRIDEv, true>), grid, block, 0,   \
-                       stream, (const __bf16*)dY, (const __bf16*)Wt,        \
-                       (__bf16*)dX, ws, kc, cs)
+    do { if (bwdd_tr())                                                     \
+      hipLaunchKernelGGL((conv_bwdd_kernel<STRIDEv, true, true>), grid,     \
+                         block, 0, stream, (const __bf16*)dY,               \
+                         (const __bf16*)Wt, (__bf16*)dX, ws, kc, cs);       \
+    else                                                                    \
+      hipLaunchKernelGGL((conv_bwdd_kernel<STRIDEv, true, false>), grid,    \
+                         block, 0, stream, (const __bf16*)dY,               \
+                         (const __bf16*)Wt, (__bf16*)dX, ws, kc, cs);       \
+    } while (0)
     if (U == 1 && V == 1) BL(1);
     else if (U == 2 && V == 2) BL(2);
     else BL(0);
@@ -1031,9 +1083,15 @@ void launch_conv_bwd_data(const bf16_t* dY, long ldy, const bf16_t* Wt,
   }
   dim3 grid((unsigned)((M + BM - 1) / BM), ceil_div(C, BN));
 #define BL(STRIDEv)                                                         \
-  hipLaunchKernelGGL((conv_bwdd_kernel<STRIDEv, false>), grid, block, 0,    \
-                     stream, (const __bf16*)dY, (const __bf16*)Wt,          \
-                     (__bf16*)dX, nullptr, 0, cs)
+  do { if (bwdd_tr())                                                       \
+    hipLaunchKernelGGL((conv_bwdd_kernel<STRIDEv, false, true>), grid,      \
+                       block, 0, stream, (const __bf16*)dY,                 \
+                       (const __bf16*)Wt, (__bf16*)dX, nullptr, 0, cs);     \
+  else                                                                      \
+    hipLaunchKernelGGL((conv_bwdd_kernel<STRIDEv, false, false>), grid,     \
+                       block, 0, stream, (const __bf16*)dY,                 \
+                       (const __bf16*)Wt, (__bf16*)dX, nullptr, 0, cs);     \
+  } while (0)
   if (U == 1 && V == 1) BL(1);
   else if (U == 2 && V == 2) BL(2);
   else BL(0);
