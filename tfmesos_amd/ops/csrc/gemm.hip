@@ -75,9 +75,26 @@ DEVINL void stage_xk(const __bf16* __restrict__ P, __bf16* S,
   }
 }
 
-// Stage a [K,X]-stored operand tile (row k contiguous in x) into S[x][k].
-// Used for A when TA and B when !TB. Global access stays contiguous
-// (8 consecutive x per thread); the transpose happens on the LDS write.
+// Stage a [K,X]-stored operand tile (row k contiguous in x) into the
+// reduction-major transpose-read image S[k][x ^ 16*((k>>3)&1)].
+// Used for A when TA and B when !TB. The earlier k-major form paid 8
+// sub-dword LDS scatter writes per thread to transpose here; this one
+// is a single b128 store, and the MFMA fragments come back via
+// ds_read_b64_tr_b16 (semantics measured on-device, tools/trprobe.hip
+// — see conv.hip's transpose-read section). The 16-column XOR per
+// k-octet makes the two k-octets a 32-lane tr read touches land on
+// disjoint bank octets (16x16x32 fragments read rows kfrag..kfrag+7,
+// and rows r and r+8 alias banks at the 96-elem row stride).
+constexpr int TR_L = 96;
+constexpr int TR_ELEMS = BK * TR_L;
+
+DEVINL bf16x4 tr_read(const __bf16* p) {
+  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) bf16x4*)p);
+}
+
+DEVINL int tr_xor(int k) { return 16 * ((k >> 3) & 1); }
+
 DEVINL void stage_kx(const __bf16* __restrict__ P, __bf16* S,
                      int x0, int k0, int X, int K, int ld, int t, int vec) {
   const int k = t >> 3;          // 0..31 == BK
@@ -85,20 +102,30 @@ DEVINL void stage_kx(const __bf16* __restrict__ P, __bf16* S,
   const int gk = k0 + k;
   const int gx = x0 + xx0;
   const __bf16* src = P + (long)gk * ld + gx;
+  bf16x8 v = {};
   if (gk < K && gx + 8 <= X) {
     if (vec == 8) {
-      bf16x8 v = *(const bf16x8*)src;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) srow(S, xx0 + j)[k] = v[j];
+      v = *(const bf16x8*)src;
     } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) srow(S, xx0 + j)[k] = src[j];
+      for (int j = 0; j < 8; ++j) v[j] = src[j];
     }
-  } else {
+  } else if (gk < K) {
 #pragma unroll
     for (int j = 0; j < 8; ++j)
-      srow(S, xx0 + j)[k] = (gk < K && gx + j < X) ? src[j] : (__bf16)0.f;
+      if (gx + j < X) v[j] = src[j];
   }
+  *(bf16x8*)&S[k * TR_L + (xx0 ^ tr_xor(k))] = v;
+}
+
+// MFMA fragment (8 reduction elems for column cb + (lane&15)) out of
+// the stage_kx image via two transpose-reads.
+DEVINL bf16x8 tr_frag(const __bf16* S, int kfrag, int cb, int lane) {
+  const int a = (kfrag + ((lane & 15) >> 2)) * TR_L +
+                ((cb + 4 * (lane & 3)) ^ tr_xor(kfrag));
+  bf16x4 f0 = tr_read(&S[a]);
+  bf16x4 f1 = tr_read(&S[a + 4 * TR_L]);
+  return __builtin_shufflevector(f0, f1, 0, 1, 2, 3, 4, 5, 6, 7);
 }
 
 // ACT: 0 none, 1 relu, 2 relu-bwd (mask by aux>0, the saved activation).
@@ -117,8 +144,10 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
   // global-load latency -> barrier -> MFMA every iteration and ran
   // the deep backward shapes at ~1.1 us/iteration (11 TFLOP/s on the
   // NMF dW GEMM).
-  __shared__ __align__(16) __bf16 As[2][TILE_ELEMS];   // [m][k], skewed
-  __shared__ __align__(16) __bf16 Bs[2][TILE_ELEMS];   // [n][k], skewed
+  // xk-staged operands: [x][k] skewed rows; kx-staged (transposing)
+  // operands: reduction-major transpose-read image (TR_ELEMS)
+  __shared__ __align__(16) __bf16 As[2][TA ? TR_ELEMS : TILE_ELEMS];
+  __shared__ __align__(16) __bf16 Bs[2][TB ? TILE_ELEMS : TR_ELEMS];
 
   const int tm0 = blockIdx.y * BM;
   const int tn0 = blockIdx.x * BN;
@@ -149,19 +178,25 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
 
     if (CS && blockIdx.y == 0 && t < BN) {
 #pragma unroll
-      for (int kk = 0; kk < BK; ++kk) cs_acc += (float)srow(Bs[cur], t)[kk];
+      for (int kk = 0; kk < BK; ++kk)
+        cs_acc += TB ? (float)srow(Bs[cur], t)[kk]
+                     : (float)Bs[cur][kk * TR_L + (t ^ tr_xor(kk))];
     }
 
     const int kfrag = (lane >> 4) * 8;
     bf16x8 bfrag[2];
 #pragma unroll
     for (int fn = 0; fn < 2; ++fn)
-      bfrag[fn] = *(const bf16x8*)&srow(Bs[cur],
-                                        wc * 32 + fn * 16 + (lane & 15))[kfrag];
+      bfrag[fn] = TB
+          ? *(const bf16x8*)&srow(Bs[cur],
+                                  wc * 32 + fn * 16 + (lane & 15))[kfrag]
+          : tr_frag(Bs[cur], kfrag, wc * 32 + fn * 16, lane);
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
-      bf16x8 a = *(const bf16x8*)&srow(As[cur],
-                                       wr * 32 + fm * 16 + (lane & 15))[kfrag];
+      bf16x8 a = TA
+          ? tr_frag(As[cur], kfrag, wr * 32 + fm * 16, lane)
+          : *(const bf16x8*)&srow(As[cur],
+                                  wr * 32 + fm * 16 + (lane & 15))[kfrag];
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
