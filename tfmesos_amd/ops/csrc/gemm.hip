@@ -46,6 +46,21 @@ DEVINL const __bf16* srow(const __bf16* S, int r) {
   return S + r * (BK + APAD) + ((r >> 3) << 3);
 }
 
+// XOR-swizzled [64][32] tile for the 64x64 kernel's xk-staged
+// operands (same scheme as conv.hip: stride exactly 32 elems, 8-elem
+// chunk index XOR (r>>2)&3 — conflict-free for ds_read_b128's
+// non-contiguous 16-lane groups AND the b128 staging writes; verified
+// zero SQ_LDS_BANK_CONFLICT on the conv twins). gemm_small keeps the
+// srow pad+skew image (different, wave-private access pattern).
+constexpr int SWZ_ELEMS = BM * BK;
+
+DEVINL __bf16* sptr(__bf16* S, int r, int c) {
+  return S + r * BK + ((((c >> 3) ^ ((r >> 2) & 3)) << 3) | (c & 7));
+}
+DEVINL const __bf16* sptr(const __bf16* S, int r, int c) {
+  return S + r * BK + ((((c >> 3) ^ ((r >> 2) & 3)) << 3) | (c & 7));
+}
+
 typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
 
 // Stage a [X,K]-stored operand tile (row x contiguous in k) into S[x][k].
@@ -59,19 +74,19 @@ DEVINL void stage_xk(const __bf16* __restrict__ P, __bf16* S,
   const __bf16* src = P + (long)gx * ld + gk;
   if (gx < X && gk + 8 <= K) {
     if (vec == 8) {
-      *(bf16x8*)&srow(S, x)[kk0] = *(const bf16x8*)src;
+      *(bf16x8*)sptr(S, x, kk0) = *(const bf16x8*)src;
     } else if (vec == 2) {
 #pragma unroll
       for (int j = 0; j < 8; j += 2)
-        *(bf16x2*)&srow(S, x)[kk0 + j] = *(const bf16x2*)(src + j);
+        *(bf16x2*)sptr(S, x, kk0 + j) = *(const bf16x2*)(src + j);
     } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) srow(S, x)[kk0 + j] = src[j];
+      for (int j = 0; j < 8; ++j) *sptr(S, x, kk0 + j) = src[j];
     }
   } else {
 #pragma unroll
     for (int j = 0; j < 8; ++j)
-      srow(S, x)[kk0 + j] = (gx < X && gk + j < K) ? src[j] : (__bf16)0.f;
+      *sptr(S, x, kk0 + j) = (gx < X && gk + j < K) ? src[j] : (__bf16)0.f;
   }
 }
 
@@ -146,8 +161,8 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
   // NMF dW GEMM).
   // xk-staged operands: [x][k] skewed rows; kx-staged (transposing)
   // operands: reduction-major transpose-read image (TR_ELEMS)
-  __shared__ __align__(16) __bf16 As[2][TA ? TR_ELEMS : TILE_ELEMS];
-  __shared__ __align__(16) __bf16 Bs[2][TB ? TILE_ELEMS : TR_ELEMS];
+  __shared__ __align__(16) __bf16 As[2][TA ? TR_ELEMS : SWZ_ELEMS];
+  __shared__ __align__(16) __bf16 Bs[2][TB ? SWZ_ELEMS : TR_ELEMS];
 
   const int tm0 = blockIdx.y * BM;
   const int tn0 = blockIdx.x * BN;
@@ -179,7 +194,7 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     if (CS && blockIdx.y == 0 && t < BN) {
 #pragma unroll
       for (int kk = 0; kk < BK; ++kk)
-        cs_acc += TB ? (float)srow(Bs[cur], t)[kk]
+        cs_acc += TB ? (float)*sptr(Bs[cur], t, kk)
                      : (float)Bs[cur][kk * TR_L + (t ^ tr_xor(kk))];
     }
 
@@ -188,15 +203,15 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
 #pragma unroll
     for (int fn = 0; fn < 2; ++fn)
       bfrag[fn] = TB
-          ? *(const bf16x8*)&srow(Bs[cur],
-                                  wc * 32 + fn * 16 + (lane & 15))[kfrag]
+          ? *(const bf16x8*)sptr(Bs[cur],
+                                 wc * 32 + fn * 16 + (lane & 15), kfrag)
           : tr_frag(Bs[cur], kfrag, wc * 32 + fn * 16, lane);
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
       bf16x8 a = TA
           ? tr_frag(As[cur], kfrag, wr * 32 + fm * 16, lane)
-          : *(const bf16x8*)&srow(As[cur],
-                                  wr * 32 + fm * 16 + (lane & 15))[kfrag];
+          : *(const bf16x8*)sptr(As[cur],
+                                 wr * 32 + fm * 16 + (lane & 15), kfrag);
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
