@@ -133,6 +133,17 @@ def run_inception(args, device, rank, world):
             loss.backward()
         trainer.step()
 
+    if world == 1 and device.type == "cuda" and not args.no_graph:
+        # capture the whole fwd+bwd+apply step (~600 launches) as one
+        # hipGraph replay; GraphedStep self-tunes and keeps eager if
+        # replay doesn't clearly win, and we fall back if any captured
+        # op refuses graph capture
+        try:
+            from tfmesos_amd.utils.graphstep import GraphedStep
+            one_step = GraphedStep(one_step)
+        except Exception as e:
+            log("graph capture unavailable, running eager: %r" % (e,))
+
     return one_step, {
         "model": "inception_v3_%d" % args.inc_size,
         "global_batch": args.inc_batch * trainer.roles.n_workers,
