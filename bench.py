@@ -56,10 +56,18 @@ def run_mnist(args, device, rank, world):
     pview = (lambda n: trainer.store.view(n, bf16=True)) \
         if act_dtype == torch.bfloat16 else (lambda n: trainer.store.view(n))
 
-    def one_step():
-        if roles.is_worker:
-            model.fwd_bwd(pview, x, y, trainer.grad_view)
-        trainer.step()
+    if world == 1 and device.type == "cuda" and not args.no_fuse \
+            and model.supports_fused_apply(trainer, x):
+        # single-GPU fast path: the SGD apply rides the producing
+        # kernels' epilogues (4 launches/step instead of 5) — same
+        # update math, measured not skipped
+        def one_step():
+            model.fwd_bwd_apply(trainer, x, y, lr=args.lr)
+    else:
+        def one_step():
+            if roles.is_worker:
+                model.fwd_bwd(pview, x, y, trainer.grad_view)
+            trainer.step()
 
     if world == 1 and device.type == "cuda" and not args.no_graph:
         # single-GPU hot path: capture the whole 8-kernel step as ONE
@@ -168,6 +176,8 @@ def main():
                    help="PS shards (ranks 0..n_ps-1) when world>1")
     p.add_argument("--no-graph", action="store_true",
                    help="disable hipGraph step capture (world==1)")
+    p.add_argument("--no-fuse", action="store_true",
+                   help="disable the world==1 fused-apply fast paths")
     p.add_argument("--lr", type=float, default=0.01)
     p.add_argument("--nmf-n", type=int, default=1000)
     p.add_argument("--nmf-rank", type=int, default=200)

@@ -100,3 +100,46 @@ def test_sparse_nmf_multiproc_on_gpu():
         assert p.returncode == 0, out
         outs.append(out)
     assert any('"transport": "gloo"' in o for o in outs), outs
+
+
+@pytest.mark.timeout(300)
+def test_mnist_fused_apply_matches_eager_step():
+    """The world==1 fused-apply fast path (mlp_tail_sgd: dW1 GEMM with
+    all four SGD applies in the epilogue) must produce the same masters
+    and shadows as the eager fwd_bwd + trainer.step() sequence."""
+    from tfmesos_amd.models.mlp import MnistMLP, synthetic_batch
+    from tfmesos_amd.ps.replica import SyncReplicaTrainer
+
+    dev = torch.device("cuda", 0)
+    x, y = synthetic_batch(100, device=dev, dtype=torch.bfloat16, seed=7)
+
+    def make():
+        m = MnistMLP(hidden_units=100)
+        t = SyncReplicaTrainer(m.init_params(), optimizer="sgd",
+                               hparams={"lr": 0.01}, device=dev,
+                               grad_dtype=torch.bfloat16)
+        return m, t
+
+    m1, t1 = make()
+    assert m1.supports_fused_apply(t1, x)
+    for _ in range(5):
+        m1.fwd_bwd_apply(t1, x, y, lr=0.01)
+
+    m2, t2 = make()
+    pview = lambda n: t2.store.view(n, bf16=True)
+    for _ in range(5):
+        m2.fwd_bwd(pview, x, y, t2.grad_view)
+        t2.step()
+    torch.cuda.synchronize()
+
+    for name in ("hid_w", "hid_b", "sm_w", "sm_b"):
+        a = t1.store.view(name).float()
+        b = t2.store.view(name).float()
+        # identical inputs; the only drift source is fused fp32-acc
+        # apply vs bf16-grad-materialize apply for W1/b1
+        rel = float((a - b).abs().max()) / max(1e-6,
+                                               float(b.abs().max()))
+        assert rel < 2e-2, (name, rel)
+        sh = t1.store.view(name, bf16=True).float()
+        assert float((sh - a).abs().max()) <= \
+            float(a.abs().max()) * 0.01 + 0.01, name

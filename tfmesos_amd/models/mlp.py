@@ -100,6 +100,37 @@ class MnistMLP(object):
                           colsum_out=g("hid_b"))
         return loss
 
+    def supports_fused_apply(self, trainer, x):
+        """The single-kernel-tail fast path: world==1 colocated, plain
+        SGD, GPU, and the head-kernel shape limits."""
+        return (x.is_cuda and trainer.world == 1
+                and trainer.store.opt == "sgd"
+                and float(trainer.store.hparams.get("momentum", 0.0)) == 0.0
+                and float(trainer.store.hparams.get("weight_decay", 0.0)) == 0.0
+                and x.shape[0] <= 128 and self.hidden <= 128
+                and self.classes <= 16)
+
+    def fwd_bwd_apply(self, trainer, x, y, lr):
+        """One fused replica step (world==1 fast path): fwd GEMM ->
+        head (loss, dh, classifier grads) -> ONE tail kernel that
+        computes dW1 and applies SGD to all four params in its
+        epilogue (ops.mlp_tail_sgd). 4 kernels/step vs 5 — the flat
+        sgd apply launch disappears; same math as fwd_bwd + step
+        (plain SGD, grad mean over the single worker = 1).
+        """
+        st = trainer.store
+        p = lambda n: st.view(n, bf16=True)
+        g = trainer.grad_view
+        h = ops.gemm_bias_act(x, p("hid_w"), p("hid_b"), act="relu")
+        loss, _, dh = ops.mlp_head_fused(
+            h, p("sm_w"), p("sm_b"), y, dw2=g("sm_w"), db2=g("sm_b"))
+        ops.mlp_tail_sgd(
+            x, dh, st.view("hid_w"), p("hid_w"), st.view("hid_b"),
+            p("hid_b"), g("sm_w"), st.view("sm_w"), p("sm_w"),
+            g("sm_b"), st.view("sm_b"), p("sm_b"), lr)
+        st.global_step += 1
+        return loss
+
     def loss_only(self, p, x, y):
         h = ops.gemm_bias_act(x, p("hid_w"), p("hid_b"), act="relu")
         logits = ops.gemm_bias_act(h, p("sm_w"), p("sm_b"))

@@ -786,6 +786,23 @@ def mlp_head_fused(h, w, b, labels, scale=None, dw2=None, db2=None):
     return loss, db, dh.to(h.dtype)
 
 
+def mlp_tail_sgd(x, dh, w1_m, w1_s, b1_m, b1_s, g_w2, w2_m, w2_s,
+                 g_b2, b2_m, b2_s, lr):
+    """mnist single-GPU fused tail (one kernel): dW1 = x^T @ dh on the
+    small-tile GEMM with the plain-SGD apply of ALL FOUR params fused
+    into the epilogue — W1/b1 straight from the fp32 accumulators
+    (the gradient never materializes), W2/b2 by re-reading the small
+    bf16 classifier grads the head kernel wrote. Masters are fp32
+    flat-store views; shadows the matching bf16 views. Only valid for
+    sgd with no momentum/weight-decay and grad_scale 1 (the world==1
+    colocated bench config); the PS path applies via fused_sgd.
+    GPU-only (raises elsewhere — keep the HIP path the one that runs).
+    """
+    return _ext().mlp_tail_sgd(x, dh, w1_m, w1_s, b1_m, b1_s,
+                               g_w2, w2_m, w2_s, g_b2, b2_m, b2_s,
+                               float(lr))
+
+
 def mlp_fwd_head_fused(x, w1, b1, w2, b2, labels, scale=None,
                        dw2=None, db2=None):
     """Whole MLP fwd + classifier head in TWO kernels (GPU): split-K

@@ -19,6 +19,19 @@ DEVINL bf16_t f2bf(float v) { return __float2bfloat16(v); }
 
 static inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
+// host-side descriptor for gemm_small's fused SGD tail (gemm.hip) —
+// opaque pointers so the binding TU needs no device vector types
+struct SmallSgdArgs {
+  float* pmw; void* psw;      // W master fp32 / bf16 shadow [M,N]
+  float* pmb; void* psb;      // bias master / shadow [N]
+  const void* g2w;            // classifier W grad (bf16, from the head)
+  float* pm2w; void* ps2w;
+  const void* g2b;
+  float* pm2b; void* ps2b;
+  float lr;
+  int n2w, n2b;
+};
+
 #define HIP_CHECK(expr)                                                     \
   do {                                                                      \
     hipError_t _e = (expr);                                                 \

@@ -5,7 +5,8 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
-typedef __hip_bfloat16 bf16_t;
+#include "common.h"
+
 
 // launchers from the .hip translation units
 void launch_sgd(float*, const void*, bool, float*, bf16_t*, long, float,
@@ -37,7 +38,7 @@ void launch_gemm_stripes(const bf16_t*, const bf16_t*, float*, int, int,
                          hipStream_t);
 void launch_gemm_small(const bf16_t*, const bf16_t*, const void*, bool,
                        void*, bool, int, void*, bool, int, int, int, int,
-                       int, int, bool, hipStream_t);
+                       int, int, bool, const SmallSgdArgs*, hipStream_t);
 void launch_gemm_stripes_any(const bf16_t*, const bf16_t*, float*, int, int,
                              int, int, int, int, int, int, bool, bool, int,
                              int, hipStream_t);
@@ -277,7 +278,7 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
                       (const bf16_t*)b.data_ptr(), bias_p, bias_bf16,
                       out.data_ptr(), out_f32, act == 1 ? 1 : 0, colsum_p,
                       out_f32, M, N, Ka, a.size(1), b.size(1), N, trans_a,
-                      cur_stream());
+                      nullptr, cur_stream());
     return out;
   }
   // split-K when the plain tile grid can't feed the 256-CU chip and K
@@ -1246,6 +1247,60 @@ torch::Tensor maxpool3x3s2_bwd(torch::Tensor dy, torch::Tensor idx,
 
 }  // namespace
 
+// mnist single-GPU fused tail: dW1 = x^T @ dh with the SGD apply of
+// ALL FOUR params in the epilogue (W1/b1 from this GEMM's fp32
+// accumulators; W2/b2 by re-reading the small classifier grads the
+// head kernel just wrote). Replaces [dW1 GEMM, flat sgd_kernel] with
+// ONE launch; valid for plain SGD, grad_scale 1 (world==1 colocated).
+// Reference parity: the one-train-op apply of mnist_replica.py:146-147
+// collapsed into the producing kernels.
+void mlp_tail_sgd(torch::Tensor x, torch::Tensor dh,
+                  torch::Tensor w1m, torch::Tensor w1s,
+                  torch::Tensor b1m, torch::Tensor b1s,
+                  torch::Tensor gw2, torch::Tensor w2m, torch::Tensor w2s,
+                  torch::Tensor gb2, torch::Tensor b2m, torch::Tensor b2s,
+                  double lr) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous() &&
+              x.scalar_type() == torch::kBFloat16, "x must be bf16 [B,M]");
+  TORCH_CHECK(dh.is_cuda() && dh.dim() == 2 && dh.is_contiguous() &&
+              dh.scalar_type() == torch::kBFloat16 &&
+              dh.size(0) == x.size(0), "dh must be bf16 [B,N]");
+  const int B = (int)x.size(0), M = (int)x.size(1), N = (int)dh.size(1);
+  for (auto* t : {&w1m, &b1m, &w2m, &b2m})
+    TORCH_CHECK(t->is_cuda() && t->is_contiguous() &&
+                t->scalar_type() == torch::kFloat, "masters must be fp32");
+  for (auto* t : {&w1s, &b1s, &gw2, &w2s, &gb2, &b2s})
+    TORCH_CHECK(t->is_cuda() && t->is_contiguous() &&
+                t->scalar_type() == torch::kBFloat16,
+                "shadows/grads must be bf16");
+  TORCH_CHECK(w1m.numel() == (long)M * N && b1m.numel() == N,
+              "W1/b1 shape mismatch");
+  TORCH_CHECK(gw2.numel() == w2m.numel() && gb2.numel() == b2m.numel(),
+              "W2/b2 grad shape mismatch");
+  const long stiles = (long)((M + 31) / 32) * ((N + 31) / 32);
+  TORCH_CHECK(stiles >= 2 && stiles <= 512 && B <= 256,
+              "shape outside the small-GEMM regime");
+  SmallSgdArgs sga;
+  sga.pmw = (float*)w1m.data_ptr();
+  sga.psw = w1s.data_ptr();
+  sga.pmb = (float*)b1m.data_ptr();
+  sga.psb = b1s.data_ptr();
+  sga.g2w = gw2.data_ptr();
+  sga.pm2w = (float*)w2m.data_ptr();
+  sga.ps2w = w2s.data_ptr();
+  sga.g2b = gb2.data_ptr();
+  sga.pm2b = (float*)b2m.data_ptr();
+  sga.ps2b = b2s.data_ptr();
+  sga.lr = (float)lr;
+  sga.n2w = (int)w2m.numel();
+  sga.n2b = (int)b2m.numel();
+  launch_gemm_small((const bf16_t*)x.data_ptr(),
+                    (const bf16_t*)dh.data_ptr(), nullptr, false,
+                    nullptr, false, 0, nullptr, false,
+                    M, N, B, (int)x.size(1), (int)dh.size(1), N,
+                    /*ta=*/true, &sga, cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd, "fused SGD apply + bf16 shadow",
         py::arg("param"), py::arg("grad"), py::arg("momentum_buf"),
@@ -1282,6 +1337,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_group_bwd", &bn_group_bwd);
   m.def("bn_group_fwd_multi", &bn_group_fwd_multi);
   m.def("gemm_sgd", &gemm_sgd);
+  m.def("mlp_tail_sgd", &mlp_tail_sgd);
   m.def("gemm_sgd_pair", &gemm_sgd_pair);
   m.def("add_n", &add_n);
   m.def("bn_group_bwd_multi", &bn_group_bwd_multi);

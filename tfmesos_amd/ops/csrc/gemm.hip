@@ -343,6 +343,31 @@ typedef __attribute__((ext_vector_type(16))) float f32x16;
 // no atomics, overwrite semantics like the 64x64 path).
 // TA: op(A) = A^T (A stored [K,M] — transposed scatter staging).
 // B is always untransposed here ([K,N] memory).
+//
+// SmallSgd: optional fused optimizer tail (the mnist single-GPU fast
+// path). When pmw != null the epilogue APPLIES this GEMM's output as
+// an SGD gradient (master -= lr*g, refresh bf16 shadow) instead of
+// storing it, the colsum applies the bias grad the same way, and one
+// otherwise-idle wave of block (0,0) also applies the classifier
+// grads (g2w/g2b, written by the head kernel that ran just before) —
+// the step's separate flat-buffer sgd_kernel launch disappears.
+// Valid only for plain SGD (no momentum/decay) with grad_scale 1
+// (single worker); the distributed path keeps the PS apply.
+struct SmallSgd {
+  float* pmw;          // W master [M,N] fp32 (same layout as Cout)
+  bf16_t* psw;         // W bf16 shadow
+  float* pmb;          // bias master [N]
+  bf16_t* psb;
+  const bf16_t* g2w;   // classifier grads (from the head kernel)
+  float* pm2w;
+  bf16_t* ps2w;
+  const bf16_t* g2b;
+  float* pm2b;
+  bf16_t* ps2b;
+  float lr;
+  int n2w, n2b;
+};
+
 template <bool TA, bool CS>
 __global__ __launch_bounds__(256)
 void gemm_small_kernel(const __bf16* __restrict__ A,
@@ -350,7 +375,8 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
                        const void* __restrict__ bias, bool bias_bf16,
                        void* __restrict__ Cout, bool out_f32, int relu,
                        void* __restrict__ colsum_out, bool cs_f32,
-                       int M, int N, int K, int lda, int ldb, int ldc) {
+                       int M, int N, int K, int lda, int ldb, int ldc,
+                       SmallSgd sg) {
   __shared__ __align__(16) __bf16 As[4][32 * (BK + APAD) + 32];
   __shared__ __align__(16) __bf16 Bs[4][32 * (BK + APAD) + 32];
   __shared__ float red[4][64 * 16];
@@ -492,6 +518,21 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
     }
   }
   __syncthreads();
+  if (sg.pm2w != nullptr && w == 1 && blockIdx.x == 0 && blockIdx.y == 0) {
+    // classifier apply: ~1k elems on one idle wave; runs strictly
+    // after the head kernel wrote g2w/g2b (stream order), and nothing
+    // reads the classifier shadow until the next step's head
+    for (int i = lane; i < sg.n2w; i += 64) {
+      const float pv = sg.pm2w[i] - sg.lr * bf2f(sg.g2w[i]);
+      sg.pm2w[i] = pv;
+      sg.ps2w[i] = f2bf(pv);
+    }
+    for (int i = lane; i < sg.n2b; i += 64) {
+      const float pv = sg.pm2b[i] - sg.lr * bf2f(sg.g2b[i]);
+      sg.pm2b[i] = pv;
+      sg.ps2b[i] = f2bf(pv);
+    }
+  }
   if (w != 0) return;
   f32x16 tot;
 #pragma unroll
@@ -510,8 +551,16 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
       if (m >= M) continue;
       float x = tot[v] + bvv;
       if (relu) x = x > 0.f ? x : 0.f;
-      if (out_f32) ((float*)Cout)[(long)m * ldc + n] = x;
-      else ((__bf16*)Cout)[(long)m * ldc + n] = (__bf16)x;
+      if (sg.pmw != nullptr) {
+        const long idx = (long)m * ldc + n;
+        const float pv = sg.pmw[idx] - sg.lr * x;
+        sg.pmw[idx] = pv;
+        sg.psw[idx] = f2bf(pv);
+      } else if (out_f32) {
+        ((float*)Cout)[(long)m * ldc + n] = x;
+      } else {
+        ((__bf16*)Cout)[(long)m * ldc + n] = (__bf16)x;
+      }
     }
   }
   if (CS && blockIdx.x == 0 && lane < 32) {
@@ -519,8 +568,15 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
     if (nn < N) {
       const float sv = csred[0][lane] + csred[1][lane] + csred[2][lane] +
                        csred[3][lane];
-      if (cs_f32) ((float*)colsum_out)[nn] = sv;
-      else ((__bf16*)colsum_out)[nn] = (__bf16)sv;
+      if (sg.pmb != nullptr) {
+        const float pv = sg.pmb[nn] - sg.lr * sv;
+        sg.pmb[nn] = pv;
+        sg.psb[nn] = f2bf(pv);
+      } else if (cs_f32) {
+        ((float*)colsum_out)[nn] = sv;
+      } else {
+        ((__bf16*)colsum_out)[nn] = (__bf16)sv;
+      }
     }
   }
 }
@@ -670,15 +726,26 @@ void launch_gemm_small(const bf16_t* A, const bf16_t* B, const void* bias,
                        bool bias_bf16, void* C, bool out_f32, int relu,
                        void* colsum_out, bool cs_f32, int M, int N, int K,
                        int lda, int ldb, int ldc, bool ta,
-                       hipStream_t stream) {
+                       const SmallSgdArgs* sga, hipStream_t stream) {
   dim3 grid(ceil_div(M, 32), ceil_div(N, 32)), block(256);
+  SmallSgd sg = {};
+  if (sga) {
+    sg.pmw = sga->pmw; sg.psw = (bf16_t*)sga->psw;
+    sg.pmb = sga->pmb; sg.psb = (bf16_t*)sga->psb;
+    sg.g2w = (const bf16_t*)sga->g2w;
+    sg.pm2w = sga->pm2w; sg.ps2w = (bf16_t*)sga->ps2w;
+    sg.g2b = (const bf16_t*)sga->g2b;
+    sg.pm2b = sga->pm2b; sg.ps2b = (bf16_t*)sga->ps2b;
+    sg.lr = sga->lr; sg.n2w = sga->n2w; sg.n2b = sga->n2b;
+  }
+  const bool cs = colsum_out != nullptr || sg.pmb != nullptr;
 #define GS(TAv, CSv)                                                        \
   hipLaunchKernelGGL((gemm_small_kernel<TAv, CSv>), grid, block, 0,         \
                      stream, (const __bf16*)A, (const __bf16*)B, bias,      \
                      bias_bf16, C, out_f32, relu, colsum_out, cs_f32, M, N, \
-                     K, lda, ldb, ldc)
-  if (ta) { if (colsum_out) GS(true, true); else GS(true, false); }
-  else    { if (colsum_out) GS(false, true); else GS(false, false); }
+                     K, lda, ldb, ldc, sg)
+  if (ta) { if (cs) GS(true, true); else GS(true, false); }
+  else    { if (cs) GS(false, true); else GS(false, false); }
 #undef GS
 }
 
