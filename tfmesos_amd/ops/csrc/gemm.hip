@@ -153,7 +153,7 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
                  void* __restrict__ Cout, const __bf16* __restrict__ aux,
                  void* __restrict__ colsum_out, float* __restrict__ ws,
                  int* __restrict__ cnt, int M, int N, int K, int lda, int ldb,
-                 int ldc, int kc, int veca, int vecb) {
+                 int ldc, int kc, int veca, int vecb, int rflags) {
   // double-buffered: stage tile i+1 while MFMA consumes tile i (one
   // barrier per K-iteration). The single-buffered version serialized
   // global-load latency -> barrier -> MFMA every iteration and ran
@@ -230,7 +230,12 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
 
   if (SK) {
     // store this slice's fp32 partial tile into its workspace stripe;
-    // the follow-up reduce kernel sums stripes in fixed slice order
+    // the follow-up reduce kernel sums stripes in fixed slice order —
+    // or, when cnt != nullptr (small tile grids), the LAST-ARRIVING
+    // slice block sums them here and applies the epilogue itself
+    // (fixed z order, so still deterministic), saving the reduce
+    // launch + its dependent-kernel boundary. rflags: bit0 relu,
+    // bit1 fp32 out, bit2 bias.
     float* wslice = ws + (long)blockIdx.z * (long)M * ldc;
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm)
@@ -244,6 +249,40 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
           if (row < M) wslice[(long)row * ldc + col] = acc[fm][fn][r];
         }
       }
+    if (cnt != nullptr) {
+      __threadfence();
+      __shared__ int lastf;
+      if (t == 0)
+        lastf = (atomicAdd(&cnt[blockIdx.y * gridDim.x + blockIdx.x], 1) ==
+                 (int)gridDim.z - 1) ? 1 : 0;
+      __syncthreads();
+      if (!lastf) return;
+      if (t == 0) cnt[blockIdx.y * gridDim.x + blockIdx.x] = 0;
+      const int ns = gridDim.z;
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+          const int col = tn0 + wc * 32 + fn * 16 + (lane & 15);
+          if (col >= N) continue;
+          const float bvv = !(rflags & 4) ? 0.f
+              : (bias_bf16 ? (float)((const __bf16*)bias)[col]
+                           : ((const float*)bias)[col]);
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int row = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + r;
+            if (row >= M) continue;
+            const long idx = (long)row * ldc + col;
+            float v = 0.f;
+            for (int z2 = 0; z2 < ns; ++z2)
+              v += ws[(long)z2 * M * ldc + idx];
+            v += bvv;
+            if (rflags & 1) v = v > 0.f ? v : 0.f;
+            if (rflags & 2) ((float*)Cout)[idx] = v;
+            else ((__bf16*)Cout)[idx] = (__bf16)v;
+          }
+        }
+    }
     return;
   }
 
@@ -521,8 +560,52 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
   if (sg.pm2w != nullptr && w == 1 && blockIdx.x == 0 && blockIdx.y == 0) {
     // classifier apply: ~1k elems on one idle wave; runs strictly
     // after the head kernel wrote g2w/g2b (stream order), and nothing
-    // reads the classifier shadow until the next step's head
-    for (int i = lane; i < sg.n2w; i += 64) {
+    // reads the classifier shadow until the next step's head.
+    // Vectorized 4-deep with every load issued before any store — the
+    // first cut (scalar strided loop) serialized ~16 global round
+    // trips on this single wave and cost MORE than the sgd_kernel it
+    // replaced. Buffers are 256-elem-aligned flat-store slices, so
+    // the f32x4/bf16x4 accesses are aligned.
+    const int nv = sg.n2w >> 2;
+    f32x4 mv[4];
+    bf16x4 gv[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int i = lane + u * 64;
+      if (i < nv) {
+        gv[u] = ((const bf16x4*)sg.g2w)[i];
+        mv[u] = ((const f32x4*)sg.pm2w)[i];
+      }
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int i = lane + u * 64;
+      if (i < nv) {
+        f32x4 pv;
+        bf16x4 sv;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          pv[j] = mv[u][j] - sg.lr * (float)gv[u][j];
+          sv[j] = (__bf16)pv[j];
+        }
+        ((f32x4*)sg.pm2w)[i] = pv;
+        ((bf16x4*)sg.ps2w)[i] = sv;
+      }
+    }
+    for (int i = 256 + lane; i < nv; i += 64) {   // >64 KB W2 (not mnist)
+      f32x4 pv;
+      bf16x4 sv;
+      const bf16x4 g = ((const bf16x4*)sg.g2w)[i];
+      const f32x4 m = ((const f32x4*)sg.pm2w)[i];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        pv[j] = m[j] - sg.lr * (float)g[j];
+        sv[j] = (__bf16)pv[j];
+      }
+      ((f32x4*)sg.pm2w)[i] = pv;
+      ((bf16x4*)sg.ps2w)[i] = sv;
+    }
+    for (int i = (nv << 2) + lane; i < sg.n2w; i += 64) {   // n2w % 4
       const float pv = sg.pm2w[i] - sg.lr * bf2f(sg.g2w[i]);
       sg.pm2w[i] = pv;
       sg.ps2w[i] = f2bf(pv);
@@ -532,6 +615,38 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
       sg.pm2b[i] = pv;
       sg.ps2b[i] = f2bf(pv);
     }
+  }
+  if (sg.pmw != nullptr) {
+    // fused W1/b1 apply: ALL FOUR waves split the 16 output values
+    // (4 each, straight from the LDS partials) — the wave0-only form
+    // serialized the master-read/store latency on one wave per block
+    // and measured SLOWER than the separate wide sgd_kernel
+    const int n = n0 + (lane & 31);
+    if (n < N) {
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const int v = w * 4 + u;
+        const int m = m0 + ((v >> 2) << 3) + ((lane >> 5) << 2) + (v & 3);
+        if (m >= M) continue;
+        const float x = red[0][lane * 16 + v] + red[1][lane * 16 + v] +
+                        red[2][lane * 16 + v] + red[3][lane * 16 + v];
+        const long idx = (long)m * ldc + n;
+        const float pv = sg.pmw[idx] - sg.lr * x;
+        sg.pmw[idx] = pv;
+        sg.psw[idx] = f2bf(pv);
+      }
+    }
+    if (CS && w == 0 && blockIdx.x == 0 && lane < 32) {
+      const int nn = n0 + lane;
+      if (nn < N && sg.pmb != nullptr) {
+        const float sv = csred[0][lane] + csred[1][lane] + csred[2][lane] +
+                         csred[3][lane];
+        const float pv = sg.pmb[nn] - sg.lr * sv;
+        sg.pmb[nn] = pv;
+        sg.psb[nn] = f2bf(pv);
+      }
+    }
+    return;
   }
   if (w != 0) return;
   f32x16 tot;
@@ -551,16 +666,8 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
       if (m >= M) continue;
       float x = tot[v] + bvv;
       if (relu) x = x > 0.f ? x : 0.f;
-      if (sg.pmw != nullptr) {
-        const long idx = (long)m * ldc + n;
-        const float pv = sg.pmw[idx] - sg.lr * x;
-        sg.pmw[idx] = pv;
-        sg.psw[idx] = f2bf(pv);
-      } else if (out_f32) {
-        ((float*)Cout)[(long)m * ldc + n] = x;
-      } else {
-        ((__bf16*)Cout)[(long)m * ldc + n] = (__bf16)x;
-      }
+      if (out_f32) ((float*)Cout)[(long)m * ldc + n] = x;
+      else ((__bf16*)Cout)[(long)m * ldc + n] = (__bf16)x;
     }
   }
   if (CS && blockIdx.x == 0 && lane < 32) {
@@ -568,15 +675,8 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
     if (nn < N) {
       const float sv = csred[0][lane] + csred[1][lane] + csred[2][lane] +
                        csred[3][lane];
-      if (sg.pmb != nullptr) {
-        const float pv = sg.pmb[nn] - sg.lr * sv;
-        sg.pmb[nn] = pv;
-        sg.psb[nn] = f2bf(pv);
-      } else if (cs_f32) {
-        ((float*)colsum_out)[nn] = sv;
-      } else {
-        ((__bf16*)colsum_out)[nn] = (__bf16)sv;
-      }
+      if (cs_f32) ((float*)colsum_out)[nn] = sv;
+      else ((__bf16*)colsum_out)[nn] = (__bf16)sv;
     }
   }
 }
@@ -629,12 +729,19 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
   dim3 block(256);
   const bool has_bias = bias != nullptr;
   const bool cs = colsum_out != nullptr;
+  const int rflags = (act == 1 ? 1 : 0) | (out_f32 ? 2 : 0) |
+                     (has_bias ? 4 : 0);
+  // last-arriver epilogue only for plain-SK small tile grids (the
+  // wide reduce kernel wins when there are many tiles to spread over)
+  if (cs || act > 1 ||
+      (long)ceil_div(N, BN) * ceil_div(M, BM) > 64)
+    cnt = nullptr;
 #define LAUNCH(TAv, TBv, ACTv, BIASv, OUTv, SKv, CSv)                       \
   hipLaunchKernelGGL((gemm_kernel<TAv, TBv, ACTv, BIASv, OUTv, SKv, CSv>),  \
                      grid, block, 0, stream, (const __bf16*)A,              \
                      (const __bf16*)B, bias, bias_bf16, C,                  \
                      (const __bf16*)aux, colsum_out, ws, cnt, M, N, K, lda, \
-                     ldb, ldc, kc, veca, vecb)
+                     ldb, ldc, kc, veca, vecb, rflags)
   // constraint combos (relu_bwd only nt/bf16, colsum only tn) are
   // validated by the bindings in ext.hip
   if (act == 2) {
@@ -679,6 +786,7 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
               else    LAUNCH(true, false, 0, false, false, true, false); }
     else    { if (tb) LAUNCH(false, true, 0, false, false, true, false);
               else    LAUNCH(false, false, 0, false, false, true, false); }
+    if (cnt != nullptr) return;   // last-arriver epilogue ran in-kernel
     // phase 2: fixed-order stripe reduce + fused epilogue
     const long mn = (long)M * ldc;
     dim3 rgrid((unsigned)(((mn + 3) / 4 + 255) / 256)), rblock(256);
@@ -764,7 +872,7 @@ void launch_gemm_stripes_any(const bf16_t* A, const bf16_t* B, float* ws,
                      grid, block, 0, stream, (const __bf16*)A,              \
                      (const __bf16*)B, nullptr, false, nullptr, nullptr,    \
                      nullptr, ws, nullptr, M, N, K, lda, ldb, ldc, kc, veca,\
-                     vecb)
+                     vecb, 0)
   if (ta) { if (tb) SLAUNCH(true, true); else SLAUNCH(true, false); }
   else    { if (tb) SLAUNCH(false, true); else SLAUNCH(false, false); }
 #undef SLAUNCH
@@ -789,5 +897,5 @@ void launch_gemm_stripes(const bf16_t* A, const bf16_t* B, float* ws,
                      grid, block, 0, stream, (const __bf16*)A,
                      (const __bf16*)B, nullptr, false, nullptr, nullptr,
                      nullptr, ws, nullptr, M, N, K, lda, ldb, ldc, kc,
-                     veca, vecb);
+                     veca, vecb, 0);
 }
