@@ -27,6 +27,8 @@
 //    column strip in LDS anyway).
 //
 // Geometry: 64x64 block tile, 4 waves (2x2), 32x32 per wave, BK=32.
+#include <cstdlib>
+
 #include "common.h"
 
 namespace {
@@ -257,31 +259,44 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
                  (int)gridDim.z - 1) ? 1 : 0;
       __syncthreads();
       if (!lastf) return;
+      __threadfence();   // acquire: don't serve stale L2 for the stripes
       if (t == 0) cnt[blockIdx.y * gridDim.x + blockIdx.x] = 0;
       const int ns = gridDim.z;
+      // vectorized row-major sweep of the whole 64x64 tile: 4 f32x4
+      // positions per thread, every stripe's loads independent — the
+      // first cut walked the MFMA acc mapping with scalar loads and
+      // the lone consumer block serialized ~144 cross-XCD round trips
+      // (the one-WG-consumer trap, see mlp.py's FROMWS note)
+      const int crow = t >> 2;             // 0..63 within tile
+      const int ccol0 = (t & 3) * 16;      // 4 f32x4 groups per row
 #pragma unroll
-      for (int fm = 0; fm < 2; ++fm)
-#pragma unroll
-        for (int fn = 0; fn < 2; ++fn) {
-          const int col = tn0 + wc * 32 + fn * 16 + (lane & 15);
-          if (col >= N) continue;
-          const float bvv = !(rflags & 4) ? 0.f
-              : (bias_bf16 ? (float)((const __bf16*)bias)[col]
-                           : ((const float*)bias)[col]);
-#pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            const int row = tm0 + wr * 32 + fm * 16 + (lane >> 4) * 4 + r;
-            if (row >= M) continue;
-            const long idx = (long)row * ldc + col;
-            float v = 0.f;
-            for (int z2 = 0; z2 < ns; ++z2)
-              v += ws[(long)z2 * M * ldc + idx];
-            v += bvv;
-            if (rflags & 1) v = v > 0.f ? v : 0.f;
-            if (rflags & 2) ((float*)Cout)[idx] = v;
-            else ((__bf16*)Cout)[idx] = (__bf16)v;
+      for (int u = 0; u < 4; ++u) {
+        const int row = tm0 + crow;
+        const int col = tn0 + ccol0 + u * 4;
+        if (row >= M || col >= N) continue;
+        const long idx = (long)row * ldc + col;
+        f32x4 v = {};
+        for (int z2 = 0; z2 < ns; ++z2) {
+          const float* s2 = &ws[(long)z2 * M * ldc + idx];
+          if (col + 4 <= N) {
+            v += *(const f32x4*)s2;
+          } else {
+            for (int j = 0; j < 4; ++j)
+              if (col + j < N) v[j] += s2[j];
           }
         }
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          if (col + j >= N) continue;
+          float x = v[j];
+          if (rflags & 4)
+            x += bias_bf16 ? (float)((const __bf16*)bias)[col + j]
+                           : ((const float*)bias)[col + j];
+          if (rflags & 1) x = x > 0.f ? x : 0.f;
+          if (rflags & 2) ((float*)Cout)[idx + j] = x;
+          else ((__bf16*)Cout)[idx + j] = (__bf16)x;
+        }
+      }
     }
     return;
   }
@@ -731,9 +746,19 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
   const bool cs = colsum_out != nullptr;
   const int rflags = (act == 1 ? 1 : 0) | (out_f32 ? 2 : 0) |
                      (has_bias ? 4 : 0);
-  // last-arriver epilogue only for plain-SK small tile grids (the
-  // wide reduce kernel wins when there are many tiles to spread over)
-  if (cs || act > 1 ||
+  // last-arriver epilogue: MEASURED DEAD END on the mnist fwd shape
+  // (4 tiles x 9 slices): the 4 consumer blocks pull the stripes
+  // cross-XCD through the release/acquire fences and the fwd GEMM
+  // went 8.2 -> 48 us scalar / ~20 us vectorized, vs 12.9 us for
+  // stripes + the wide reduce kernel. Same lesson as the FROMWS head:
+  // few-WG consumers must not read many-XCD-producer output. Kept
+  // behind TFA_GEMM_LA=1 for re-measurement; default off.
+  static int la_on = -1;
+  if (la_on < 0) {
+    const char* e = getenv("TFA_GEMM_LA");
+    la_on = e ? atoi(e) : 0;
+  }
+  if (!la_on || cs || act > 1 ||
       (long)ceil_div(N, BN) * ceil_div(M, BM) > 64)
     cnt = nullptr;
 #define LAUNCH(TAv, TBv, ACTv, BIASv, OUTv, SKv, CSv)                       \
