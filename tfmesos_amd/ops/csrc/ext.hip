@@ -301,6 +301,12 @@ torch::Tensor gemm_bias_act_out(torch::Tensor a, torch::Tensor b,
                                             : (int)(Ka / 64),
                         256 / (nx * ny));
     if (want > 16) want = 16;
+    static int want_env = -2;
+    if (want_env == -2) {
+      const char* e = getenv("TFA_SK_WANT");   // tuning override
+      want_env = e ? atoi(e) : -1;
+    }
+    if (want_env > 0) want = want_env;
     if (want > 1) {
       kc = ((Ka + want - 1) / want + 31) / 32 * 32;
       nslice = (Ka + kc - 1) / kc;
