@@ -333,6 +333,29 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
 
 // split-K phase 2: out[i] = act(sum_z ws[z][i] + bias), f32x4-vectorized
 // over the flattened [M,ldc] output (ldc == N, contiguous).
+// 1-elem/thread variant for SMALL outputs (mnist fwd: mn = 10k gave
+// only 10 WGs at 4 elems/thread — the reduce was WG-count
+// latency-bound, not bandwidth-bound; 4x the blocks cut it ~2x)
+template <int ACT, bool BIAS, bool OUTF32>
+__global__ __launch_bounds__(256)
+void splitk_reduce_small_kernel(const float* __restrict__ ws,
+                                const void* __restrict__ bias,
+                                bool bias_bf16, void* __restrict__ Cout,
+                                long mn, int ldc, int nslice) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= mn) return;
+  float v = 0.f;
+  for (int z = 0; z < nslice; ++z) v += ws[(long)z * mn + i];
+  if (BIAS) {
+    const int col = (int)(i % ldc);
+    v += bias_bf16 ? (float)((const __bf16*)bias)[col]
+                   : ((const float*)bias)[col];
+  }
+  if (ACT == 1) v = v > 0.f ? v : 0.f;
+  if (OUTF32) ((float*)Cout)[i] = v;
+  else ((__bf16*)Cout)[i] = (__bf16)v;
+}
+
 template <int ACT, bool BIAS, bool OUTF32>
 __global__ __launch_bounds__(256)
 void splitk_reduce_kernel(const float* __restrict__ ws, const void* __restrict__ bias,
@@ -812,14 +835,24 @@ void launch_gemm(const bf16_t* A, const bf16_t* B, const void* bias,
     else    { if (tb) LAUNCH(false, true, 0, false, false, true, false);
               else    LAUNCH(false, false, 0, false, false, true, false); }
     if (cnt != nullptr) return;   // last-arriver epilogue ran in-kernel
-    // phase 2: fixed-order stripe reduce + fused epilogue
+    // phase 2: fixed-order stripe reduce + fused epilogue (small
+    // outputs: 1 elem/thread for 4x the workgroups)
     const long mn = (long)M * ldc;
-    dim3 rgrid((unsigned)(((mn + 3) / 4 + 255) / 256)), rblock(256);
+    const bool small_r = mn <= (1 << 16);
+    dim3 rgrid((unsigned)(small_r ? (mn + 255) / 256
+                                  : ((mn + 3) / 4 + 255) / 256));
+    dim3 rblock(256);
 #define RLAUNCH(ACTv, BIASv, OUTv)                                          \
-    hipLaunchKernelGGL((splitk_reduce_kernel<ACTv, BIASv, OUTv>), rgrid,    \
-                       rblock, 0, stream, ws, bias, bias_bf16, C, mn, ldc,  \
-                       nslice, (const float*)nullptr, (void*)nullptr,       \
-                       false, 0)
+    do { if (small_r)                                                       \
+      hipLaunchKernelGGL((splitk_reduce_small_kernel<ACTv, BIASv, OUTv>),   \
+                         rgrid, rblock, 0, stream, ws, bias, bias_bf16, C,  \
+                         mn, ldc, nslice);                                  \
+    else                                                                    \
+      hipLaunchKernelGGL((splitk_reduce_kernel<ACTv, BIASv, OUTv>), rgrid,  \
+                         rblock, 0, stream, ws, bias, bias_bf16, C, mn,     \
+                         ldc, nslice, (const float*)nullptr,                \
+                         (void*)nullptr, false, 0);                         \
+    } while (0)
     if (act == 1) {
       if (has_bias) { if (out_f32) RLAUNCH(1, true, true);
                       else RLAUNCH(1, true, false); }
