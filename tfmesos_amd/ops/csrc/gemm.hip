@@ -34,20 +34,6 @@
 namespace {
 
 constexpr int BM = 64, BN = 64, BK = 32;
-constexpr int APAD = 8;  // +16B: keeps b128 fragment reads aligned
-// 16B skew per 8-row group: the row stride (40 bf16 = 20 dwords) makes
-// banks repeat every 8 rows, so the transposing stagers' writes were
-// 8-way bank-conflicted (measured via PMC on the conv twins of these
-// stagers; same layout here).
-constexpr int TILE_ELEMS = BM * (BK + APAD) + (BM / 8) * 8;
-
-DEVINL __bf16* srow(__bf16* S, int r) {
-  return S + r * (BK + APAD) + ((r >> 3) << 3);
-}
-DEVINL const __bf16* srow(const __bf16* S, int r) {
-  return S + r * (BK + APAD) + ((r >> 3) << 3);
-}
-
 // XOR-swizzled [64][32] tile for the 64x64 kernel's xk-staged
 // operands (same scheme as conv.hip: stride exactly 32 elems, 8-elem
 // chunk index XOR (r>>2)&3 — conflict-free for ds_read_b128's
@@ -454,8 +440,13 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
                        void* __restrict__ colsum_out, bool cs_f32,
                        int M, int N, int K, int lda, int ldb, int ldc,
                        SmallSgd sg) {
-  __shared__ __align__(16) __bf16 As[4][32 * (BK + APAD) + 32];
-  __shared__ __align__(16) __bf16 Bs[4][32 * (BK + APAD) + 32];
+  // wave-private 32x32 images, stride exactly 32 + the sptr chunk
+  // swizzle: the transposing operands (A when TA, B always) store
+  // K-MAJOR [k][col] — commit is two b128 stores (the transposed
+  // scatter commit measured 3.5 extra LDS-conflict cycles per LDS
+  // instruction here) — and fragments come back via ds_read_b64_tr_b16
+  __shared__ __align__(16) __bf16 As[4][32 * BK];
+  __shared__ __align__(16) __bf16 Bs[4][32 * BK];
   __shared__ float red[4][64 * 16];
   __shared__ float csred[4][32];
   const int t = threadIdx.x;
@@ -541,25 +532,30 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
     }
   };
   auto commit = [&]() {
-    if (TA) {
+    // both layouts have row kk: TA-A/B rows are k (16 consecutive
+    // cols of one k per register), non-TA A rows are m (16
+    // consecutive k) — either way two b128 chunk-aligned stores
+    *(bf16x8*)sptr(Aw, kk, half * 16) = ra0;
+    *(bf16x8*)sptr(Aw, kk, half * 16 + 8) = ra1;
+    *(bf16x8*)sptr(Bw, kk, half * 16) = rb0;
+    *(bf16x8*)sptr(Bw, kk, half * 16 + 8) = rb1;
+    if (CS) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        srow(Aw, half * 16 + j)[kk] = ra0[j];
-        srow(Aw, half * 16 + 8 + j)[kk] = ra1[j];
-      }
-    } else {
-      *(bf16x8*)&srow(Aw, kk)[half * 16] = ra0;
-      *(bf16x8*)&srow(Aw, kk)[half * 16 + 8] = ra1;
-    }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      srow(Bw, half * 16 + j)[kk] = rb0[j];
-      srow(Bw, half * 16 + 8 + j)[kk] = rb1[j];
-      if (CS) {
         csp[j] += (float)rb0[j];
         csp[8 + j] += (float)rb1[j];
       }
     }
+  };
+
+  // transpose-read fragment out of a k-major wave image (see tr_frag;
+  // same provider-lane scheme, 32-wide rows via sptr)
+  auto trs_frag = [&](const __bf16* S, int ko) {
+    const int kq = ko + ((lane & 15) >> 2);
+    const int c = 16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+    bf16x4 f0 = tr_read(sptr((__bf16*)S, kq, c));
+    bf16x4 f1 = tr_read(sptr((__bf16*)S, kq + 4, c));
+    return __builtin_shufflevector(f0, f1, 0, 1, 2, 3, 4, 5, 6, 7);
   };
 
   load_a(ks);
@@ -570,8 +566,9 @@ void gemm_small_kernel(const __bf16* __restrict__ A,
 #pragma unroll
     for (int kh = 0; kh < 2; ++kh) {
       const int ko = kh * 16 + ((lane >> 5) << 3);
-      bf16x8 a = *(const bf16x8*)&srow(Aw, lane & 31)[ko];
-      bf16x8 b = *(const bf16x8*)&srow(Bw, lane & 31)[ko];
+      bf16x8 a = TA ? trs_frag(Aw, ko)
+                    : *(const bf16x8*)sptr(Aw, lane & 31, ko);
+      bf16x8 b = trs_frag(Bw, ko);
       *(f32x16*)acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
           a, b, *(f32x16*)acc, 0, 0, 0);
     }
