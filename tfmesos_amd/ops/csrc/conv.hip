@@ -89,17 +89,6 @@ struct ConvShape {
              // narrow view of a concat-grad buffer — no contiguous copy)
 };
 
-// Load an 8-element channel run X[pix_base + c0 .. c0+8) into S[row][..],
-// vectorized when aligned.
-DEVINL void put8(__bf16* dstrow, const __bf16* src, bool vec) {
-  if (vec) {
-    *(bf16x8*)dstrow = *(const bf16x8*)src;
-  } else {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) dstrow[j] = src[j];
-  }
-}
-
 // ---------------------------------------------------------------- forward
 
 // As[m][kk]: m = output pixel, kk = tap (r,s,c) with c innermost.

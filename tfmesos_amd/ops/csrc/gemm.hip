@@ -38,8 +38,8 @@ constexpr int BM = 64, BN = 64, BK = 32;
 // operands (same scheme as conv.hip: stride exactly 32 elems, 8-elem
 // chunk index XOR (r>>2)&3 — conflict-free for ds_read_b128's
 // non-contiguous 16-lane groups AND the b128 staging writes; verified
-// zero SQ_LDS_BANK_CONFLICT on the conv twins). gemm_small keeps the
-// srow pad+skew image (different, wave-private access pattern).
+// zero SQ_LDS_BANK_CONFLICT on the conv twins). gemm_small reuses
+// the same helper for its wave-private 32x32 images.
 constexpr int SWZ_ELEMS = BM * BK;
 
 DEVINL __bf16* sptr(__bf16* S, int r, int c) {
@@ -147,8 +147,8 @@ void gemm_kernel(const __bf16* __restrict__ A, const __bf16* __restrict__ B,
   // global-load latency -> barrier -> MFMA every iteration and ran
   // the deep backward shapes at ~1.1 us/iteration (11 TFLOP/s on the
   // NMF dW GEMM).
-  // xk-staged operands: [x][k] skewed rows; kx-staged (transposing)
-  // operands: reduction-major transpose-read image (TR_ELEMS)
+  // xk-staged operands: [x][k] XOR-swizzled tiles; kx-staged
+  // (transposing) operands: reduction-major transpose-read image
   __shared__ __align__(16) __bf16 As[2][TA ? TR_ELEMS : SWZ_ELEMS];
   __shared__ __align__(16) __bf16 Bs[2][TB ? SWZ_ELEMS : TR_ELEMS];
 
