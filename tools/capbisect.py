@@ -1,5 +1,6 @@
 import sys, torch
-sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
+import os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from tfmesos_amd import ops
 from tfmesos_amd.models.inception import InceptionV3, synthetic_images
 from tfmesos_amd.ps.module_trainer import ModuleReplicaTrainer
