@@ -229,3 +229,28 @@ def test_overlap_pure_ps_rank_world2(tmp_path):
     a = _run_world2(tmp_path, "ps", "psovl")
     b = _run_world2(tmp_path, "ps-no-overlap", "psnovl")
     assert torch.allclose(a, b, atol=1e-6)
+
+
+def test_fused_apply_gate_is_world1_sgd_gpu_only():
+    """The mnist fused-apply fast path must never engage on CPU, under
+    momentum, or for non-sgd optimizers (bench.py relies on this gate
+    to keep the PS path authoritative everywhere else)."""
+    import torch
+
+    from tfmesos_amd.models.mlp import MnistMLP, synthetic_batch
+    from tfmesos_amd.ps.replica import SyncReplicaTrainer
+
+    m = MnistMLP(hidden_units=100)
+    x, _ = synthetic_batch(100, device="cpu")
+    t = SyncReplicaTrainer(m.init_params(), optimizer="sgd",
+                           hparams={"lr": 0.01}, device="cpu")
+    assert not m.supports_fused_apply(t, x)          # CPU tensors
+
+    t2 = SyncReplicaTrainer(m.init_params(), optimizer="sgd",
+                            hparams={"lr": 0.01, "momentum": 0.9},
+                            device="cpu")
+    assert not m.supports_fused_apply(t2, x)         # momentum
+
+    t3 = SyncReplicaTrainer(m.init_params(), optimizer="adam",
+                            hparams={"lr": 0.01}, device="cpu")
+    assert not m.supports_fused_apply(t3, x)         # optimizer
