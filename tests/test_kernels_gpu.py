@@ -830,3 +830,29 @@ def test_gemm_sgd_pair_simultaneous_semantics():
         float((H - refH).abs().max())
     assert torch.allclose(Wb.float(), W.to(torch.bfloat16).float())
     assert torch.allclose(Hb.float(), H.to(torch.bfloat16).float())
+
+
+def test_conv2d_bwd_channel_narrow_dy_view():
+    """Concat backward hands each branch a channel-NARROW view of the
+    joint grad buffer (ldy > K). The transpose-read bwd kernels take
+    the row stride (cs.LDY) straight from the view — no contiguous
+    copy — so a strided dy must produce the same dX/dW as its
+    contiguous clone."""
+    from tfmesos_amd import ops
+    torch.manual_seed(99)
+    N, C, H, W, K, R, S = 2, 16, 13, 13, 24, 3, 3
+    x = bf(torch.randn(N, C, H, W)).requires_grad_(True)
+    w = bf(torch.randn(K, C, R, S) * 0.2).requires_grad_(True)
+    y = ops.conv2d(x, w, None, stride=1, padding=1)
+
+    big = bf(torch.randn(N, K + 40, H, W))      # concat-like buffer
+    dy_view = big.narrow(1, 8, K)               # strided channel slice
+    y.backward(dy_view)
+    dx_v, dw_v = x.grad.clone(), w.grad.clone()
+
+    x.grad = None
+    w.grad = None
+    y2 = ops.conv2d(x, w, None, stride=1, padding=1)
+    y2.backward(dy_view.contiguous(memory_format=torch.channels_last))
+    assert torch.equal(dx_v, x.grad), "dX differs for narrow-dy view"
+    assert torch.equal(dw_v, w.grad), "dW differs for narrow-dy view"
