@@ -65,8 +65,4 @@ class ResNet18(nn.Module):
         for b in self.blocks:
             x = b(x)
         x = x.mean((2, 3))                      # global average pool
-        if x.is_cuda:
-            return ops.gemm_bias_act(x.to(torch.bfloat16).contiguous(),
-                                     self.fc_w.to(torch.bfloat16),
-                                     self.fc_b)
-        return x.float() @ self.fc_w.float() + self.fc_b.float()
+        return ops.linear(x.contiguous(), self.fc_w, self.fc_b)
